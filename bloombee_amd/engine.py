@@ -1,0 +1,136 @@
+"""Local (single-process) inference engine: embeddings + block stack + LM head.
+
+This is the innermost compute loop shared by:
+  * bench.py at N=1 (the whole model on one MI355X),
+  * the per-worker stage runner in parallel/pipeline.py (a contiguous block
+    range per rank),
+  * the server backend (server/backend.py) which serves the same blocks over
+    the decentralized transport.
+
+The reference splits this across DistributedLlamaModel (client embeddings +
+lm_head, models/llama/model.py:45-118) and per-server block execution; here
+the same Blocks object serves both local and distributed paths.
+"""
+from __future__ import annotations
+
+from typing import List, Optional, Sequence
+
+import torch
+import torch.nn.functional as F
+
+from bloombee_amd.kv.paged import PagedKVCache, SessionHandle
+from bloombee_amd.models.auto import get_block_class
+from bloombee_amd.models.base import ModelConfig, resolve_config
+from bloombee_amd.models.llama.block import RopeTables
+from bloombee_amd.utils.logging import get_logger
+from bloombee_amd import ops
+
+logger = get_logger(__name__)
+
+
+class BlockStack(torch.nn.Module):
+    """A contiguous range [start, end) of transformer blocks on one device."""
+
+    def __init__(self, config: ModelConfig, start: int, end: int,
+                 device="cpu", seed: int = 0):
+        super().__init__()
+        self.config = config
+        self.start, self.end = start, end
+        block_cls = get_block_class(config.model_type)
+        self.rope = RopeTables(config)
+        self.blocks = torch.nn.ModuleList()
+        for i in range(start, end):
+            blk = block_cls(config, layer_index=i, rope=self.rope)
+            blk.init_random(seed=seed * 10_000 + i)
+            blk.layer_index = i - start  # KV page index local to this stack's pool
+            self.blocks.append(blk.to(device))
+        self.device = torch.device(device)
+
+    def make_kv(self, max_tokens: int) -> PagedKVCache:
+        return PagedKVCache(
+            num_layers=len(self.blocks),
+            num_kv_heads=self.config.num_key_value_heads,
+            head_dim=self.config.head_dim,
+            max_tokens=max_tokens,
+            device=self.device,
+            dtype=self.config.dtype,
+        )
+
+    @torch.no_grad()
+    def forward_inference(self, hidden: torch.Tensor, kv: SessionHandle,
+                          start_pos: torch.Tensor,
+                          position_ids: Optional[torch.Tensor] = None) -> torch.Tensor:
+        for blk in self.blocks:
+            hidden = blk.forward_inference(hidden, kv, start_pos, position_ids)
+        return hidden
+
+    def forward_train(self, hidden: torch.Tensor, start_pos: int = 0) -> torch.Tensor:
+        for blk in self.blocks:
+            hidden = blk.forward_train(hidden, start_pos)
+        return hidden
+
+
+class LocalEngine:
+    """Whole model in one process: greedy decode for bench/tests.
+
+    Client-side pieces (embeddings, final norm, LM head) follow the reference
+    client split (client/lm_head.py, models/llama/model.py:80-118).
+    """
+
+    def __init__(self, config_or_name, device="cpu", seed: int = 0,
+                 kv_max_tokens: int = 1 << 16):
+        cfg = (config_or_name if isinstance(config_or_name, ModelConfig)
+               else resolve_config(config_or_name))
+        self.config = cfg
+        self.device = torch.device(device)
+        gen = torch.Generator().manual_seed(seed)
+        dt = cfg.dtype
+        self.embed = (torch.randn(cfg.vocab_size, cfg.hidden_size, generator=gen)
+                      .mul_(0.02).to(dt).to(device))
+        self.final_norm_w = torch.ones(cfg.hidden_size, dtype=dt, device=device)
+        if cfg.tie_word_embeddings:
+            self.lm_head_w = self.embed
+        else:
+            self.lm_head_w = (torch.randn(cfg.vocab_size, cfg.hidden_size, generator=gen)
+                              .mul_(0.02).to(dt).to(device))
+        self.stack = BlockStack(cfg, 0, cfg.num_hidden_layers, device=device, seed=seed)
+        self.kv_pool = self.stack.make_kv(kv_max_tokens)
+
+    @torch.no_grad()
+    def logits_for(self, hidden_last: torch.Tensor) -> torch.Tensor:
+        """hidden_last: (B, H) -> (B, V)"""
+        y = ops.rms_norm(hidden_last, self.final_norm_w, self.config.rms_norm_eps)
+        return F.linear(y, self.lm_head_w)
+
+    @torch.no_grad()
+    def prefill(self, input_ids: torch.Tensor, kv: SessionHandle) -> torch.Tensor:
+        """input_ids: (B, T). Returns greedy next token ids (B,)."""
+        B, T = input_ids.shape
+        start = torch.tensor([s.l_spec for s in kv.seqs], dtype=torch.int32,
+                             device=self.device)
+        kv.extend(T)
+        hidden = F.embedding(input_ids.to(self.device), self.embed)
+        hidden = self.stack.forward_inference(hidden, kv, start)
+        return self.logits_for(hidden[:, -1]).argmax(-1)
+
+    @torch.no_grad()
+    def decode_step(self, input_ids: torch.Tensor, kv: SessionHandle) -> torch.Tensor:
+        """input_ids: (B,) last tokens. Returns next token ids (B,)."""
+        start = torch.tensor([s.l_spec for s in kv.seqs], dtype=torch.int32,
+                             device=self.device)
+        kv.extend(1)
+        hidden = F.embedding(input_ids.view(-1, 1).to(self.device), self.embed)
+        hidden = self.stack.forward_inference(hidden, kv, start)
+        return self.logits_for(hidden[:, -1]).argmax(-1)
+
+    @torch.no_grad()
+    def generate_greedy(self, input_ids: torch.Tensor, max_new_tokens: int) -> torch.Tensor:
+        B, T = input_ids.shape
+        kv = self.kv_pool.allocate(B, T + max_new_tokens + 1)
+        try:
+            toks = [self.prefill(input_ids, kv)]
+            for _ in range(max_new_tokens - 1):
+                toks.append(self.decode_step(toks[-1], kv))
+            return torch.stack(toks, dim=1)
+        finally:
+            kv.close()

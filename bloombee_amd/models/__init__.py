@@ -1,0 +1,10 @@
+from bloombee_amd.models.base import (  # noqa: F401
+    FamilyEntry,
+    ModelConfig,
+    get_family,
+    register_model_family,
+    resolve_config,
+)
+
+# Import families for registration side effects.
+import bloombee_amd.models.llama  # noqa: F401,E402

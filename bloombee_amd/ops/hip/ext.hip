@@ -1,0 +1,333 @@
+// Torch extension bindings for the bloombee_amd gfx950 kernel library.
+// Host-side shape checks, template dispatch, and stream plumbing only —
+// all compute lives in the .hip translation units.
+
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+#include "common.h"
+
+// Single translation unit: the template kernels must be visible at their
+// launch sites for instantiation.
+#include "elementwise.hip"
+#include "kvcache.hip"
+#include "attn_decode.hip"
+#include "attn_prefill.hip"
+#include "mfma_selftest.hip"
+#include "quant4.hip"
+
+typedef __attribute__((ext_vector_type(8))) short short8_h;
+
+#define CHECK_DEV(x) TORCH_CHECK(x.is_cuda(), #x " must be on device")
+#define CHECK_BF16(x) TORCH_CHECK(x.scalar_type() == at::kBFloat16, #x " must be bf16")
+#define CHECK_CONTIG(x) TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
+
+static inline const unsigned short* bf_ptr(const torch::Tensor& t) {
+  return reinterpret_cast<const unsigned short*>(t.data_ptr());
+}
+static inline unsigned short* bf_ptr_mut(torch::Tensor& t) {
+  return reinterpret_cast<unsigned short*>(t.data_ptr());
+}
+static inline hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+// ---------------------------------------------------------------------------
+// norms
+// ---------------------------------------------------------------------------
+
+template <typename F>
+static void dispatch_iters(int H, F f) {
+  constexpr int BLOCK = 256;
+  const int per_iter = BLOCK * 8;
+  if (H <= per_iter) f(std::integral_constant<int, 1>{});
+  else if (H <= 2 * per_iter) f(std::integral_constant<int, 2>{});
+  else if (H <= 4 * per_iter) f(std::integral_constant<int, 4>{});
+  else if (H <= 8 * per_iter) f(std::integral_constant<int, 8>{});
+  else TORCH_CHECK(false, "hidden size too large for norm kernel: ", H);
+}
+
+torch::Tensor rms_norm(torch::Tensor x, torch::Tensor w, double eps) {
+  CHECK_DEV(x); CHECK_BF16(x); CHECK_CONTIG(x);
+  const int H = x.size(-1);
+  TORCH_CHECK(H % 8 == 0, "H must be a multiple of 8");
+  const long N = x.numel() / H;
+  auto y = torch::empty_like(x);
+  dispatch_iters(H, [&](auto iters) {
+    rmsnorm_kernel<256, decltype(iters)::value><<<N, 256, 0, cur_stream()>>>(
+        bf_ptr(x), nullptr, bf_ptr(w), nullptr, bf_ptr_mut(y), H, (float)eps);
+  });
+  return y;
+}
+
+std::vector<torch::Tensor> rms_norm_residual(torch::Tensor x, torch::Tensor res,
+                                             torch::Tensor w, double eps) {
+  CHECK_DEV(x); CHECK_BF16(x); CHECK_CONTIG(x); CHECK_CONTIG(res);
+  const int H = x.size(-1);
+  const long N = x.numel() / H;
+  auto h = torch::empty_like(x);
+  auto y = torch::empty_like(x);
+  dispatch_iters(H, [&](auto iters) {
+    rmsnorm_kernel<256, decltype(iters)::value><<<N, 256, 0, cur_stream()>>>(
+        bf_ptr(x), bf_ptr(res), bf_ptr(w), bf_ptr_mut(h), bf_ptr_mut(y), H,
+        (float)eps);
+  });
+  return {h, y};
+}
+
+torch::Tensor layer_norm(torch::Tensor x, torch::Tensor w,
+                         c10::optional<torch::Tensor> bias, double eps) {
+  CHECK_DEV(x); CHECK_BF16(x); CHECK_CONTIG(x);
+  const int H = x.size(-1);
+  const long N = x.numel() / H;
+  auto y = torch::empty_like(x);
+  const unsigned short* bp = bias.has_value() ? bf_ptr(*bias) : nullptr;
+  dispatch_iters(H, [&](auto iters) {
+    layernorm_kernel<256, decltype(iters)::value><<<N, 256, 0, cur_stream()>>>(
+        bf_ptr(x), nullptr, bf_ptr(w), bp, nullptr, bf_ptr_mut(y), H, (float)eps);
+  });
+  return y;
+}
+
+// ---------------------------------------------------------------------------
+// rope / activations
+// ---------------------------------------------------------------------------
+
+void rope_apply_(torch::Tensor q, torch::Tensor k, torch::Tensor cos_t,
+                 torch::Tensor sin_t, torch::Tensor pos) {
+  CHECK_DEV(q); CHECK_BF16(q); CHECK_CONTIG(q); CHECK_CONTIG(k);
+  TORCH_CHECK(pos.scalar_type() == at::kInt, "pos must be int32");
+  const int B = q.size(0), Hq = q.size(1), T = q.size(2), D = q.size(3);
+  const int Hkv = k.size(1);
+  TORCH_CHECK(cos_t.scalar_type() == at::kFloat && cos_t.size(1) == D / 2);
+  dim3 grid(B * T, Hq + Hkv);
+  rope_kernel<<<grid, 64, 0, cur_stream()>>>(
+      bf_ptr_mut(q), bf_ptr_mut(k), cos_t.data_ptr<float>(),
+      sin_t.data_ptr<float>(), pos.data_ptr<int>(), B, Hq, Hkv, T, D);
+}
+
+torch::Tensor swiglu(torch::Tensor gu) {
+  CHECK_DEV(gu); CHECK_BF16(gu); CHECK_CONTIG(gu);
+  const long I = gu.size(-1) / 2;
+  const long N = gu.numel() / (2 * I);
+  TORCH_CHECK(I % 8 == 0);
+  auto sizes = gu.sizes().vec();
+  sizes.back() = I;
+  auto out = torch::empty(sizes, gu.options());
+  const long nvec = N * (I / 8);
+  const int grid = (int)std::min<long>((nvec + 255) / 256, 2048);
+  swiglu_kernel<<<grid, 256, 0, cur_stream()>>>(bf_ptr(gu), bf_ptr_mut(out), N, I);
+  return out;
+}
+
+torch::Tensor gelu_tanh(torch::Tensor x) {
+  CHECK_DEV(x); CHECK_BF16(x); CHECK_CONTIG(x);
+  auto out = torch::empty_like(x);
+  const long nvec = x.numel() / 8;
+  const int grid = (int)std::min<long>((nvec + 255) / 256, 2048);
+  gelu_kernel<<<grid, 256, 0, cur_stream()>>>(bf_ptr(x), bf_ptr_mut(out), x.numel());
+  return out;
+}
+
+torch::Tensor add_bf16(torch::Tensor a, torch::Tensor b) {
+  CHECK_DEV(a); CHECK_BF16(a); CHECK_CONTIG(a); CHECK_CONTIG(b);
+  auto out = torch::empty_like(a);
+  const long nvec = a.numel() / 8;
+  const int grid = (int)std::min<long>((nvec + 255) / 256, 2048);
+  add_kernel<<<grid, 256, 0, cur_stream()>>>(bf_ptr(a), bf_ptr(b), bf_ptr_mut(out),
+                                             a.numel());
+  return out;
+}
+
+// ---------------------------------------------------------------------------
+// paged KV
+// ---------------------------------------------------------------------------
+
+void kv_write(torch::Tensor k_new, torch::Tensor v_new, torch::Tensor k_pages,
+              torch::Tensor v_pages, torch::Tensor page_table,
+              torch::Tensor start_pos) {
+  CHECK_DEV(k_new); CHECK_BF16(k_new); CHECK_CONTIG(k_new); CHECK_CONTIG(v_new);
+  CHECK_CONTIG(k_pages); CHECK_CONTIG(v_pages);
+  TORCH_CHECK(page_table.scalar_type() == at::kInt && start_pos.scalar_type() == at::kInt);
+  const int B = k_new.size(0), Hkv = k_new.size(1), T = k_new.size(2), D = k_new.size(3);
+  const int P = k_pages.size(2), maxp = page_table.size(1);
+  TORCH_CHECK(k_pages.size(1) == Hkv && k_pages.size(3) == D);
+  const int threads = std::min(256, Hkv * D / 8);
+  kv_write_kernel<<<B * T, threads, 0, cur_stream()>>>(
+      bf_ptr(k_new), bf_ptr(v_new), bf_ptr_mut(k_pages), bf_ptr_mut(v_pages),
+      page_table.data_ptr<int>(), start_pos.data_ptr<int>(), B, Hkv, T, D, P, maxp);
+}
+
+std::vector<torch::Tensor> kv_gather(torch::Tensor k_pages, torch::Tensor v_pages,
+                                     torch::Tensor page_table, long batch_index,
+                                     long ctx) {
+  CHECK_DEV(k_pages);
+  const int Hkv = k_pages.size(1), P = k_pages.size(2), D = k_pages.size(3);
+  auto k = torch::empty({Hkv, ctx, D}, k_pages.options());
+  auto v = torch::empty({Hkv, ctx, D}, v_pages.options());
+  const long nvec = ctx * Hkv * (D / 8);
+  const int grid = (int)std::min<long>((nvec + 255) / 256, 2048);
+  kv_gather_kernel<<<grid, 256, 0, cur_stream()>>>(
+      bf_ptr(k_pages), bf_ptr(v_pages), bf_ptr_mut(k), bf_ptr_mut(v),
+      page_table.data_ptr<int>(), (int)batch_index, (int)ctx, Hkv, D, P,
+      page_table.size(1));
+  return {k, v};
+}
+
+// ---------------------------------------------------------------------------
+// attention
+// ---------------------------------------------------------------------------
+
+template <int D>
+static void attn_decode_launch(const torch::Tensor& q, const torch::Tensor& kp,
+                               const torch::Tensor& vp, const torch::Tensor& pt,
+                               const torch::Tensor& ctx, torch::Tensor& out,
+                               torch::Tensor& pml, torch::Tensor& pacc, int B,
+                               int Hkv, int G, int P, int maxp, int n_split,
+                               int window, float scale) {
+  dim3 grid(B * Hkv, n_split);
+  auto launch = [&](auto maxg) {
+    attn_decode_kernel<D, decltype(maxg)::value><<<grid, 256, 0, cur_stream()>>>(
+        bf_ptr(q), bf_ptr(kp), bf_ptr(vp), pt.data_ptr<int>(),
+        ctx.data_ptr<int>(), bf_ptr_mut(out), pml.data_ptr<float>(),
+        pacc.data_ptr<float>(), B, Hkv, G, P, maxp, n_split, window, scale);
+  };
+  if (G <= 4) launch(std::integral_constant<int, 4>{});
+  else if (G <= 8) launch(std::integral_constant<int, 8>{});
+  else TORCH_CHECK(false, "GQA group size > 8 unsupported: ", G);
+  if (n_split > 1) {
+    attn_decode_combine_kernel<D><<<B * Hkv, G * 16, 0, cur_stream()>>>(
+        pml.data_ptr<float>(), pacc.data_ptr<float>(), bf_ptr_mut(out), Hkv, G,
+        n_split);
+  }
+}
+
+torch::Tensor attn_decode(torch::Tensor q, torch::Tensor k_pages,
+                          torch::Tensor v_pages, torch::Tensor page_table,
+                          torch::Tensor ctx_lens, double scale, long window,
+                          long n_split_req) {
+  CHECK_DEV(q); CHECK_BF16(q); CHECK_CONTIG(q);
+  TORCH_CHECK(q.dim() == 4 && q.size(2) == 1, "attn_decode expects (B, Hq, 1, D)");
+  const int B = q.size(0), Hq = q.size(1), D = q.size(3);
+  const int Hkv = k_pages.size(1), P = k_pages.size(2), maxp = page_table.size(1);
+  const int G = Hq / Hkv;
+  TORCH_CHECK(Hq % Hkv == 0);
+  TORCH_CHECK(D < 256 || G <= 4, "D=256 decode supports GQA group size <= 4");
+  int n_split = (int)n_split_req;
+  if (n_split <= 0) {
+    // fill the chip: >= ~1024 workgroups (256 CUs x 8 XCDs, guide §1)
+    n_split = (int)std::max<long>(1, std::min<long>(32, 1024 / std::max(1, B * Hkv)));
+  }
+  auto out = torch::empty({B, Hq, 1, D}, q.options());
+  auto fopt = torch::TensorOptions().device(q.device()).dtype(at::kFloat);
+  torch::Tensor pml, pacc;
+  if (n_split > 1) {
+    pml = torch::empty({(long)B * Hkv * n_split, G, 2}, fopt);
+    pacc = torch::empty({(long)B * Hkv * n_split, G, D}, fopt);
+  } else {
+    pml = torch::empty({1}, fopt);
+    pacc = torch::empty({1}, fopt);
+  }
+  if (D == 128)
+    attn_decode_launch<128>(q, k_pages, v_pages, page_table, ctx_lens, out, pml,
+                            pacc, B, Hkv, G, P, maxp, n_split, (int)window,
+                            (float)scale);
+  else if (D == 64)
+    attn_decode_launch<64>(q, k_pages, v_pages, page_table, ctx_lens, out, pml,
+                           pacc, B, Hkv, G, P, maxp, n_split, (int)window,
+                           (float)scale);
+  else if (D == 256)
+    attn_decode_launch<256>(q, k_pages, v_pages, page_table, ctx_lens, out, pml,
+                            pacc, B, Hkv, G, P, maxp, n_split, (int)window,
+                            (float)scale);
+  else
+    TORCH_CHECK(false, "unsupported head_dim ", D);
+  return out;
+}
+
+torch::Tensor attn_prefill(torch::Tensor q, torch::Tensor k_pages,
+                           torch::Tensor v_pages, torch::Tensor page_table,
+                           torch::Tensor q_start, double scale, long window) {
+  CHECK_DEV(q); CHECK_BF16(q); CHECK_CONTIG(q);
+  TORCH_CHECK(q.dim() == 4, "attn_prefill expects (B, Hq, Tq, D)");
+  const int B = q.size(0), Hq = q.size(1), Tq = q.size(2), D = q.size(3);
+  const int Hkv = k_pages.size(1), P = k_pages.size(2), maxp = page_table.size(1);
+  const int G = Hq / Hkv;
+  auto out = torch::empty_like(q);
+  dim3 grid((Tq + 63) / 64, B * Hq);
+  auto launch = [&](auto d) {
+    attn_prefill_kernel<decltype(d)::value><<<grid, 256, 0, cur_stream()>>>(
+        bf_ptr(q), bf_ptr(k_pages), bf_ptr(v_pages), page_table.data_ptr<int>(),
+        q_start.data_ptr<int>(), bf_ptr_mut(out), B, Hq, G, Tq, P, maxp,
+        (int)window, (float)scale);
+  };
+  if (D == 128) launch(std::integral_constant<int, 128>{});
+  else if (D == 64) launch(std::integral_constant<int, 64>{});
+  else if (D == 256) launch(std::integral_constant<int, 256>{});
+  else TORCH_CHECK(false, "unsupported head_dim ", D);
+  return out;
+}
+
+// ---------------------------------------------------------------------------
+// quant4 / selftest
+// ---------------------------------------------------------------------------
+
+std::vector<torch::Tensor> quant4_pack(torch::Tensor x) {
+  CHECK_DEV(x); CHECK_BF16(x); CHECK_CONTIG(x);
+  const long ncols = x.size(-1);
+  TORCH_CHECK(ncols % 64 == 0, "quant4 needs cols % 64 == 0");
+  const long nrows = x.numel() / ncols;
+  auto packed = torch::empty({nrows, ncols / 2},
+                             torch::TensorOptions().device(x.device()).dtype(at::kByte));
+  auto hopt = torch::TensorOptions().device(x.device()).dtype(at::kHalf);
+  auto scale = torch::empty({nrows, ncols / 64}, hopt);
+  auto zero = torch::empty({nrows, ncols / 64}, hopt);
+  const long ngroups = nrows * (ncols / 64);
+  const int wpb = 4;  // waves per block
+  quant4_pack_kernel<64><<<(ngroups + wpb - 1) / wpb, wpb * 64, 0, cur_stream()>>>(
+      bf_ptr(x), packed.data_ptr<unsigned char>(),
+      reinterpret_cast<__half*>(scale.data_ptr()),
+      reinterpret_cast<__half*>(zero.data_ptr()), nrows, ncols);
+  return {packed, scale, zero};
+}
+
+torch::Tensor quant4_unpack(torch::Tensor packed, torch::Tensor scale,
+                            torch::Tensor zero) {
+  CHECK_DEV(packed);
+  const long nrows = packed.size(0), ncols = packed.size(1) * 2;
+  auto out = torch::empty({nrows, ncols},
+                          torch::TensorOptions().device(packed.device()).dtype(at::kBFloat16));
+  const long total = nrows * ncols;
+  const int grid = (int)std::min<long>((total + 255) / 256, 4096);
+  quant4_unpack_kernel<64><<<grid, 256, 0, cur_stream()>>>(
+      packed.data_ptr<unsigned char>(), reinterpret_cast<__half*>(scale.data_ptr()),
+      reinterpret_cast<__half*>(zero.data_ptr()), bf_ptr_mut(out), nrows, ncols);
+  return out;
+}
+
+torch::Tensor mfma_selftest(torch::Tensor A, torch::Tensor B) {
+  CHECK_DEV(A); CHECK_BF16(A); CHECK_CONTIG(A); CHECK_CONTIG(B);
+  TORCH_CHECK(A.size(0) == 16 && A.size(1) == 32 && B.size(0) == 32 && B.size(1) == 16);
+  auto C = torch::empty({16, 16}, torch::TensorOptions().device(A.device()).dtype(at::kFloat));
+  mfma_selftest_kernel<<<1, 64, 0, cur_stream()>>>(bf_ptr(A), bf_ptr(B),
+                                                   C.data_ptr<float>());
+  return C;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rms_norm", &rms_norm);
+  m.def("rms_norm_residual", &rms_norm_residual);
+  m.def("layer_norm", &layer_norm);
+  m.def("rope_apply_", &rope_apply_);
+  m.def("swiglu", &swiglu);
+  m.def("gelu_tanh", &gelu_tanh);
+  m.def("add_bf16", &add_bf16);
+  m.def("kv_write", &kv_write);
+  m.def("kv_gather", &kv_gather);
+  m.def("attn_decode", &attn_decode);
+  m.def("attn_prefill", &attn_prefill);
+  m.def("quant4_pack", &quant4_pack);
+  m.def("quant4_unpack", &quant4_unpack);
+  m.def("mfma_selftest", &mfma_selftest);
+}

@@ -1,0 +1,187 @@
+"""Dispatch layer between the gfx950 HIP extension and the torch reference.
+
+CPU tensors run the fp32 reference (bloombee_amd/ops/reference.py); device
+tensors run the in-tree _hip_ops extension and FAIL LOUDLY if it is absent —
+a silent eager fallback on a GPU box would invalidate every GPU test
+(the round-end driver records which .so the GPU processes actually load).
+"""
+from __future__ import annotations
+
+import importlib
+import math
+from typing import Optional, Tuple
+
+import torch
+
+from bloombee_amd.ops import reference as ref
+from bloombee_amd.ops.reference import rope_cos_sin  # noqa: F401  (host-side)
+from bloombee_amd.utils.logging import get_logger
+
+logger = get_logger(__name__)
+
+try:
+    hip_ops = importlib.import_module("bloombee_amd.ops._hip_ops")
+    HAVE_HIP_OPS = True
+except ImportError as e:  # pragma: no cover
+    hip_ops = None
+    HAVE_HIP_OPS = False
+    _import_error = e
+
+
+def _require_ext():
+    if not HAVE_HIP_OPS:
+        raise RuntimeError(
+            "bloombee_amd HIP extension is not built but a tensor is on GPU. "
+            "Run `python setup.py hip` (PYTORCH_ROCM_ARCH=gfx950). "
+            f"Import error was: {_import_error}"
+        )
+
+
+def _on_gpu(t: torch.Tensor) -> bool:
+    return t.is_cuda
+
+
+# ---------------------------------------------------------------------------
+
+
+def rms_norm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-5) -> torch.Tensor:
+    if _on_gpu(x):
+        _require_ext()
+        return hip_ops.rms_norm(x.contiguous(), weight.contiguous(), eps)
+    return ref.rms_norm(x, weight, eps)
+
+
+def rms_norm_residual(
+    x: torch.Tensor, residual: torch.Tensor, weight: torch.Tensor, eps: float = 1e-5
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    if _on_gpu(x):
+        _require_ext()
+        h, y = hip_ops.rms_norm_residual(x.contiguous(), residual.contiguous(),
+                                         weight.contiguous(), eps)
+        return h, y
+    return ref.rms_norm_residual(x, residual, weight, eps)
+
+
+def layer_norm(x, weight, bias=None, eps: float = 1e-5):
+    if _on_gpu(x):
+        _require_ext()
+        return hip_ops.layer_norm(x.contiguous(), weight.contiguous(),
+                                  bias.contiguous() if bias is not None else None, eps)
+    return ref.layer_norm(x, weight, bias, eps)
+
+
+def rope_apply_(q, k, cos, sin, position_ids):
+    """In-place RoPE on q, k: (B, H*, T, D). position_ids: (B, T) int32."""
+    if _on_gpu(q):
+        _require_ext()
+        hip_ops.rope_apply_(q, k, cos, sin, position_ids.int())
+        return q, k
+    q2, k2 = ref.rope_apply(q, k, cos, sin, position_ids)
+    q.copy_(q2)
+    k.copy_(k2)
+    return q, k
+
+
+def swiglu(gate_up: torch.Tensor) -> torch.Tensor:
+    """gate_up: (..., 2I) from the fused gate/up GEMM -> (..., I)."""
+    if _on_gpu(gate_up):
+        _require_ext()
+        return hip_ops.swiglu(gate_up.contiguous())
+    I = gate_up.shape[-1] // 2
+    return ref.swiglu(gate_up[..., :I], gate_up[..., I:])
+
+
+def gelu_tanh(x: torch.Tensor) -> torch.Tensor:
+    if _on_gpu(x):
+        _require_ext()
+        return hip_ops.gelu_tanh(x.contiguous())
+    return ref.gelu_tanh(x)
+
+
+def kv_write(k_new, v_new, k_pages, v_pages, page_table, start_pos):
+    if _on_gpu(k_new):
+        _require_ext()
+        hip_ops.kv_write(k_new.contiguous(), v_new.contiguous(), k_pages, v_pages,
+                         page_table, start_pos.int())
+        return
+    ref.kv_write(k_new, v_new, k_pages, v_pages, page_table, start_pos)
+
+
+def kv_gather(k_pages, v_pages, page_table, ctx_len: int, batch_index: int):
+    if _on_gpu(k_pages):
+        _require_ext()
+        k, v = hip_ops.kv_gather(k_pages, v_pages, page_table, batch_index, ctx_len)
+        return k, v
+    return ref.kv_gather(k_pages, v_pages, page_table, ctx_len, batch_index)
+
+
+def attn_decode(
+    q, k_pages, v_pages, page_table, ctx_lens, scale: Optional[float] = None,
+    window: int = 0, n_split: int = 0,
+) -> torch.Tensor:
+    """Single-token paged attention. q: (B, Hq, 1, D)."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if _on_gpu(q):
+        _require_ext()
+        return hip_ops.attn_decode(q.contiguous(), k_pages, v_pages, page_table,
+                                   ctx_lens.int(), scale, window, n_split)
+    q_start = ctx_lens.long() - 1
+    return ref.attn_paged(q, k_pages, v_pages, page_table, q_start, scale,
+                          sliding_window=window if window > 0 else None)
+
+
+def attn_prefill(
+    q, k_pages, v_pages, page_table, q_start, scale: Optional[float] = None,
+    window: int = 0,
+) -> torch.Tensor:
+    """Multi-token causal paged attention. q: (B, Hq, Tq, D); the new tokens'
+    K/V must already be in the pages (kv_write first)."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if _on_gpu(q):
+        _require_ext()
+        return hip_ops.attn_prefill(q.contiguous(), k_pages, v_pages, page_table,
+                                    q_start.int(), scale, window)
+    return ref.attn_paged(q, k_pages, v_pages, page_table, q_start, scale,
+                          sliding_window=window if window > 0 else None)
+
+
+def attn_paged(q, k_pages, v_pages, page_table, q_start, scale=None, window: int = 0,
+               tree_mask=None):
+    """Unified entry: picks decode vs prefill kernel by Tq."""
+    Tq = q.shape[2]
+    if tree_mask is not None:
+        # tree-attention (spec decode verify) — reference path for now
+        if scale is None:
+            scale = 1.0 / math.sqrt(q.shape[-1])
+        return ref.attn_paged(q.cpu(), k_pages.cpu(), v_pages.cpu(),
+                              page_table.cpu(), q_start.cpu(), scale,
+                              tree_mask=tree_mask.cpu()).to(q.device) \
+            if _on_gpu(q) else ref.attn_paged(q, k_pages, v_pages, page_table,
+                                              q_start, scale, tree_mask=tree_mask)
+    if Tq == 1:
+        ctx_lens = q_start + 1
+        return attn_decode(q, k_pages, v_pages, page_table, ctx_lens, scale, window)
+    return attn_prefill(q, k_pages, v_pages, page_table, q_start, scale, window)
+
+
+def quant4_pack(x: torch.Tensor, group_size: int = 64):
+    if _on_gpu(x):
+        _require_ext()
+        assert group_size == 64
+        packed, scale, zero = hip_ops.quant4_pack(x.contiguous())
+        return packed, scale, zero
+    return ref.quant4_pack(x, group_size)
+
+
+def quant4_unpack(packed, scale, zero, dtype=torch.bfloat16):
+    if _on_gpu(packed):
+        _require_ext()
+        return hip_ops.quant4_unpack(packed, scale, zero)
+    return ref.quant4_unpack(packed, scale, zero, dtype)
+
+
+def mfma_selftest(A: torch.Tensor, B: torch.Tensor) -> torch.Tensor:
+    _require_ext()
+    return hip_ops.mfma_selftest(A.contiguous(), B.contiguous())

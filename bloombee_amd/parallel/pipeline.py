@@ -1,0 +1,194 @@
+"""Pipeline-parallel decode over torch.distributed (RCCL on GPU, gloo on CPU).
+
+This is the single-node fast path of the reference's inter-server pipeline:
+on one 8xMI355X node the client->S1->...->SN->client activation hops become
+RCCL send/recv over xGMI (SURVEY.md §2.7 "Pipeline parallelism": xGMI is 7
+point-to-point links per GPU, so neighbor-only PP traffic rides one dedicated
+link), and the reference's server->server rpc_push + push-only-downstream
+decode (handler.py:2239-2760, inference_session.py:178-196) becomes
+micro-batch pipelining: stage r computes micro-batch j while j-1 is in
+flight to stage r+1.
+
+Topology: rank r hosts the contiguous block range [r*L/N, (r+1)*L/N); rank 0
+additionally holds the client role (embeddings, final norm, LM head) —
+matching the reference client split (client/lm_head.py, model.py:80-118).
+The ring is r -> r+1 -> ... -> N-1 -> 0.
+"""
+from __future__ import annotations
+
+import os
+from typing import List, Optional
+
+import torch
+import torch.distributed as dist
+import torch.nn.functional as F
+
+from bloombee_amd.engine import BlockStack
+from bloombee_amd.kv.paged import PagedKVCache
+from bloombee_amd.kv.views import SessionView
+from bloombee_amd.models.base import ModelConfig, resolve_config
+from bloombee_amd.utils.logging import get_logger
+
+logger = get_logger(__name__)
+
+
+def layer_range(num_layers: int, rank: int, world: int):
+    per = num_layers // world
+    extra = num_layers % world
+    start = rank * per + min(rank, extra)
+    end = start + per + (1 if rank < extra else 0)
+    return start, end
+
+
+class PipelineStage:
+    """One rank of the pipelined decoder."""
+
+    def __init__(self, config_or_name, device, global_batch: int,
+                 micro_batches: int = 0, seed: int = 0,
+                 kv_max_tokens: int = 1 << 17, max_session_len: int = 4096):
+        cfg = (config_or_name if isinstance(config_or_name, ModelConfig)
+               else resolve_config(config_or_name))
+        self.config = cfg
+        self.device = torch.device(device)
+        self.rank = dist.get_rank() if dist.is_initialized() else 0
+        self.world = dist.get_world_size() if dist.is_initialized() else 1
+        self.global_batch = global_batch
+        if micro_batches <= 0:
+            micro_batches = min(4 * self.world, global_batch) if self.world > 1 else 1
+        while global_batch % micro_batches != 0:
+            micro_batches -= 1
+        self.M = micro_batches
+        self.mb = global_batch // self.M
+
+        start, end = layer_range(cfg.num_hidden_layers, self.rank, self.world)
+        logger.info(f"rank {self.rank}/{self.world}: layers [{start}, {end})")
+        self.stack = BlockStack(cfg, start, end, device=self.device, seed=seed)
+        self.kv_pool = self.stack.make_kv(kv_max_tokens)
+        self.kv = self.kv_pool.allocate(global_batch, max_session_len)
+
+        self.is_client = self.rank == 0
+        if self.is_client:
+            gen = torch.Generator().manual_seed(seed)
+            dt = cfg.dtype
+            self.embed = (torch.randn(cfg.vocab_size, cfg.hidden_size, generator=gen)
+                          .mul_(0.02).to(dt).to(self.device))
+            self.final_norm_w = torch.ones(cfg.hidden_size, dtype=dt, device=self.device)
+            self.lm_head_w = (self.embed if cfg.tie_word_embeddings else
+                              (torch.randn(cfg.vocab_size, cfg.hidden_size, generator=gen)
+                               .mul_(0.02).to(dt).to(self.device)))
+        self.next_rank = (self.rank + 1) % self.world
+        self.prev_rank = (self.rank - 1) % self.world
+        # persistent recv buffers per micro-batch
+        self._recv_buf = [torch.empty(self.mb, 1, cfg.hidden_size, dtype=cfg.dtype,
+                                      device=self.device) for _ in range(self.M)]
+
+    # ------------------------------------------------------------------
+    def _run_local(self, hidden: torch.Tensor, view, start_pos: torch.Tensor):
+        return self.stack.forward_inference(hidden, view, start_pos)
+
+    def _views(self):
+        return [SessionView(self.kv, j * self.mb, (j + 1) * self.mb)
+                for j in range(self.M)]
+
+    @torch.no_grad()
+    def decode_round(self, ids: Optional[torch.Tensor]) -> Optional[torch.Tensor]:
+        """One decode step for the whole global batch, micro-batch pipelined.
+
+        rank 0: `ids` is (global_batch,) current tokens; returns next tokens.
+        other ranks: pass None; returns None.
+        """
+        cfg = self.config
+        B, M, mb = self.global_batch, self.M, self.mb
+        pos0 = self.kv.seqs[0].l_spec  # all sequences advance in lockstep
+        start_all = torch.full((B,), pos0, dtype=torch.int32, device=self.device)
+        self.kv.extend(1)
+        views = self._views()
+        pending = []
+
+        if self.is_client:
+            hid = F.embedding(ids.view(B, 1).to(self.device), self.embed)
+            outs: List[torch.Tensor] = []
+            for j in range(M):
+                h = self._run_local(hid[j * mb:(j + 1) * mb], views[j],
+                                    start_all[j * mb:(j + 1) * mb])
+                if self.world > 1:
+                    pending.append(dist.isend(h.contiguous(), self.next_rank))
+                else:
+                    outs.append(h)
+            if self.world > 1:
+                for j in range(M):
+                    dist.recv(self._recv_buf[j], self.prev_rank)
+                    outs.append(self._recv_buf[j])
+            for w in pending:
+                w.wait()
+            hidden = torch.cat(outs, dim=0)
+            return self._lm_head(hidden[:, -1])
+        else:
+            for j in range(M):
+                dist.recv(self._recv_buf[j], self.prev_rank)
+                h = self._run_local(self._recv_buf[j], views[j],
+                                    start_all[j * mb:(j + 1) * mb])
+                pending.append(dist.isend(h.contiguous(), self.next_rank))
+            for w in pending:
+                w.wait()
+            return None
+
+    @torch.no_grad()
+    def prefill_round(self, ids: Optional[torch.Tensor], T: int) -> Optional[torch.Tensor]:
+        """Prefill T prompt tokens (chunked by micro-batch over the batch dim)."""
+        cfg = self.config
+        B, M, mb = self.global_batch, self.M, self.mb
+        pos0 = self.kv.seqs[0].l_spec
+        start_all = torch.full((B,), pos0, dtype=torch.int32, device=self.device)
+        self.kv.extend(T)
+        views = self._views()
+        pending = []
+        bufs = [torch.empty(mb, T, cfg.hidden_size, dtype=cfg.dtype,
+                            device=self.device) for _ in range(M)]
+        if self.is_client:
+            hid = F.embedding(ids.to(self.device), self.embed)
+            outs = []
+            for j in range(M):
+                h = self._run_local(hid[j * mb:(j + 1) * mb], views[j],
+                                    start_all[j * mb:(j + 1) * mb])
+                if self.world > 1:
+                    pending.append(dist.isend(h.contiguous(), self.next_rank))
+                else:
+                    outs.append(h)
+            if self.world > 1:
+                for j in range(M):
+                    dist.recv(bufs[j], self.prev_rank)
+                    outs.append(bufs[j])
+            for w in pending:
+                w.wait()
+            hidden = torch.cat(outs, dim=0)
+            return self._lm_head(hidden[:, -1])
+        else:
+            for j in range(M):
+                dist.recv(bufs[j], self.prev_rank)
+                h = self._run_local(bufs[j], views[j], start_all[j * mb:(j + 1) * mb])
+                pending.append(dist.isend(h.contiguous(), self.next_rank))
+            for w in pending:
+                w.wait()
+            return None
+
+    def _lm_head(self, hidden_last: torch.Tensor) -> torch.Tensor:
+        from bloombee_amd import ops
+
+        y = ops.rms_norm(hidden_last, self.final_norm_w, self.config.rms_norm_eps)
+        return F.linear(y, self.lm_head_w).float().argmax(-1)
+
+
+def init_distributed(device_type: str = "auto") -> str:
+    """Initialize torch.distributed from torchrun env (one process per GPU,
+    RCCL over xGMI; gloo on CPU-only hosts/tests). Returns device string."""
+    if "RANK" not in os.environ:
+        return "cuda:0" if torch.cuda.is_available() else "cpu"
+    rank = int(os.environ["RANK"])
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    use_gpu = torch.cuda.is_available() and device_type != "cpu"
+    backend = "nccl" if use_gpu else "gloo"
+    if use_gpu:
+        torch.cuda.set_device(local_rank)
+    dist.init_process_group(backend=backend)
+    return f"cuda:{local_rank}" if use_gpu else "cpu"

@@ -1,0 +1,247 @@
+"""GPU kernel parity suite: every gfx950 kernel vs the fp32 torch reference
+(mirrors the reference's parity tests, e.g. test_mha_gen_llama_decode_parity.py,
+test_optimized_layers.py). All tests @pytest.mark.gpu."""
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from bloombee_amd import ops
+    from bloombee_amd.ops import reference as ref
+
+    assert ops.HAVE_HIP_OPS, (
+        "HIP extension must be built and loadable on a GPU box (fail-loud policy)"
+    )
+
+DEV = "cuda:0"
+
+
+def to_dev(*ts):
+    return [t.to(DEV) for t in ts]
+
+
+# ---------------------------------------------------------------------------
+
+
+def test_mfma_fragment_layout():
+    """Asymmetric-input check of the assumed A/B/C fragment maps (guide §3)."""
+    torch.manual_seed(0)
+    A = torch.randn(16, 32).to(torch.bfloat16)
+    B = torch.randn(32, 16).to(torch.bfloat16)
+    C = ops.mfma_selftest(A.to(DEV), B.to(DEV)).cpu()
+    expect = A.float() @ B.float()
+    assert torch.allclose(C, expect, atol=2e-2, rtol=1e-2), (
+        f"max err {(C - expect).abs().max()} — fragment layout is wrong"
+    )
+
+
+def test_rms_norm_parity():
+    torch.manual_seed(1)
+    for H in (256, 4096, 8192):
+        x = torch.randn(9, H).to(torch.bfloat16)
+        w = torch.randn(H).to(torch.bfloat16)
+        got = ops.rms_norm(x.to(DEV), w.to(DEV)).cpu().float()
+        want = ref.rms_norm(x, w).float()
+        assert torch.allclose(got, want, atol=2e-2), (got - want).abs().max()
+
+
+def test_rms_norm_residual_parity():
+    torch.manual_seed(2)
+    x = torch.randn(7, 4096).to(torch.bfloat16)
+    r = torch.randn(7, 4096).to(torch.bfloat16)
+    w = torch.randn(4096).to(torch.bfloat16)
+    hg, yg = ops.rms_norm_residual(x.to(DEV), r.to(DEV), w.to(DEV))
+    hw, yw = ref.rms_norm_residual(x, r, w)
+    assert torch.allclose(hg.cpu().float(), hw.float(), atol=2e-2)
+    assert torch.allclose(yg.cpu().float(), yw.float(), atol=2e-2)
+
+
+def test_layer_norm_parity():
+    torch.manual_seed(3)
+    x = torch.randn(5, 1024).to(torch.bfloat16)
+    w = torch.randn(1024).to(torch.bfloat16)
+    b = torch.randn(1024).to(torch.bfloat16)
+    got = ops.layer_norm(x.to(DEV), w.to(DEV), b.to(DEV)).cpu().float()
+    want = ref.layer_norm(x, w, b).float()
+    assert torch.allclose(got, want, atol=3e-2), (got - want).abs().max()
+
+
+def test_rope_parity():
+    torch.manual_seed(4)
+    B, Hq, Hkv, T, D = 2, 4, 2, 5, 128
+    cos, sin = ref.rope_cos_sin(D, 512)
+    q = torch.randn(B, Hq, T, D).to(torch.bfloat16)
+    k = torch.randn(B, Hkv, T, D).to(torch.bfloat16)
+    pos = torch.randint(0, 500, (B, T)).int()
+    qg, kg = q.clone().to(DEV), k.clone().to(DEV)
+    ops.rope_apply_(qg, kg, cos.to(DEV), sin.to(DEV), pos.to(DEV))
+    qw, kw = ref.rope_apply(q, k, cos, sin, pos)
+    assert torch.allclose(qg.cpu().float(), qw.float(), atol=2e-2)
+    assert torch.allclose(kg.cpu().float(), kw.float(), atol=2e-2)
+
+
+def test_swiglu_parity():
+    torch.manual_seed(5)
+    gu = torch.randn(6, 512).to(torch.bfloat16)
+    got = ops.swiglu(gu.to(DEV)).cpu().float()
+    want = ref.swiglu(gu[:, :256], gu[:, 256:]).float()
+    assert torch.allclose(got, want, atol=2e-2), (got - want).abs().max()
+
+
+def test_kv_write_gather_parity():
+    torch.manual_seed(6)
+    B, Hkv, T, D, P = 3, 2, 37, 128, 16
+    npages = 32
+    kp = torch.zeros(npages, Hkv, P, D).to(torch.bfloat16).to(DEV)
+    vp = torch.zeros(npages, Hkv, P, D).to(torch.bfloat16).to(DEV)
+    maxp = 8
+    pt = (torch.randperm(npages)[: B * maxp]).int().reshape(B, maxp).to(DEV)
+    k = torch.randn(B, Hkv, T, D).to(torch.bfloat16)
+    v = torch.randn(B, Hkv, T, D).to(torch.bfloat16)
+    start = torch.tensor([0, 3, 11], dtype=torch.int32)
+    ops.kv_write(k.to(DEV), v.to(DEV), kp, vp, pt, start.to(DEV))
+    for b in range(B):
+        ctx = int(start[b]) + T
+        kg, vg = ops.kv_gather(kp, vp, pt, ctx, b)
+        kg = kg.cpu()[:, int(start[b]):]
+        vg = vg.cpu()[:, int(start[b]):]
+        assert torch.equal(kg, k[b]), f"K mismatch seq {b}"
+        assert torch.equal(vg, v[b]), f"V mismatch seq {b}"
+
+
+def _paged_setup(B, Hq, Hkv, T, D, P=16, seed=7, dtype=torch.bfloat16):
+    torch.manual_seed(seed)
+    maxp = (T + P - 1) // P + 2
+    npages = B * maxp + 4
+    kp = torch.zeros(npages, Hkv, P, D, dtype=dtype)
+    vp = torch.zeros(npages, Hkv, P, D, dtype=dtype)
+    pt = torch.arange(B * maxp, dtype=torch.int32).reshape(B, maxp)
+    q = (torch.randn(B, Hq, T, D) / math.sqrt(D)).to(dtype)
+    k = torch.randn(B, Hkv, T, D).to(dtype)
+    v = torch.randn(B, Hkv, T, D).to(dtype)
+    start = torch.zeros(B, dtype=torch.int32)
+    ref.kv_write(k, v, kp, vp, pt, start)
+    return kp, vp, pt, q, k, v, start
+
+
+@pytest.mark.parametrize("shape", [
+    (2, 8, 2, 33, 128),    # GQA G=4
+    (1, 4, 4, 100, 128),   # MHA
+    (2, 8, 1, 57, 128),    # MQA G=8
+    (1, 2, 2, 300, 64),    # D=64
+    (1, 2, 1, 40, 256),    # D=256 (gemma-class)
+])
+def test_attn_decode_parity(shape):
+    B, Hq, Hkv, T, D = shape
+    kp, vp, pt, q, k, v, start = _paged_setup(B, Hq, Hkv, T, D)
+    ctx = torch.full((B,), T, dtype=torch.int32)
+    qd = q[:, :, -1:].contiguous()
+    want = ref.attn_paged(qd.float(), kp.float(), vp.float(), pt,
+                          (ctx - 1).long())
+    for n_split in (1, 4):
+        got = ops.attn_decode(qd.to(DEV), kp.to(DEV), vp.to(DEV), pt.to(DEV),
+                              ctx.to(DEV), n_split=n_split).cpu().float()
+        assert torch.allclose(got, want.float(), atol=2e-2), (
+            f"n_split={n_split} max err {(got - want).abs().max()}"
+        )
+
+
+@pytest.mark.parametrize("shape", [
+    (2, 8, 2, 128, 128),
+    (1, 4, 4, 200, 128),   # non-multiple of 64
+    (2, 8, 8, 64, 64),
+    (1, 2, 1, 96, 256),
+])
+def test_attn_prefill_parity(shape):
+    B, Hq, Hkv, T, D = shape
+    kp, vp, pt, q, k, v, start = _paged_setup(B, Hq, Hkv, T, D, seed=8)
+    want = ref.attn_paged(q.float(), kp.float(), vp.float(), pt, start.long())
+    got = ops.attn_prefill(q.to(DEV), kp.to(DEV), vp.to(DEV), pt.to(DEV),
+                           start.to(DEV)).cpu().float()
+    assert torch.allclose(got, want.float(), atol=2e-2), (
+        f"max err {(got - want).abs().max()}"
+    )
+
+
+def test_attn_prefill_with_prefix():
+    """Multi-turn: new chunk attends to existing cache prefix."""
+    B, Hq, Hkv, T, D = 1, 4, 2, 48, 128
+    kp, vp, pt, q, k, v, start = _paged_setup(B, Hq, Hkv, T, D, seed=9)
+    Tnew = 16
+    qn = q[:, :, -Tnew:].contiguous()
+    qs = torch.tensor([T - Tnew], dtype=torch.int32)
+    want = ref.attn_paged(qn.float(), kp.float(), vp.float(), pt, qs.long())
+    got = ops.attn_prefill(qn.to(DEV), kp.to(DEV), vp.to(DEV), pt.to(DEV),
+                           qs.to(DEV)).cpu().float()
+    assert torch.allclose(got, want.float(), atol=2e-2), (got - want).abs().max()
+
+
+def test_attn_decode_sliding_window():
+    B, Hq, Hkv, T, D = 1, 4, 2, 100, 128
+    kp, vp, pt, q, k, v, start = _paged_setup(B, Hq, Hkv, T, D, seed=10)
+    ctx = torch.full((B,), T, dtype=torch.int32)
+    qd = q[:, :, -1:].contiguous()
+    win = 32
+    want = ref.attn_paged(qd.float(), kp.float(), vp.float(), pt,
+                          (ctx - 1).long(), sliding_window=win)
+    got = ops.attn_decode(qd.to(DEV), kp.to(DEV), vp.to(DEV), pt.to(DEV),
+                          ctx.to(DEV), window=win).cpu().float()
+    assert torch.allclose(got, want.float(), atol=2e-2)
+
+
+def test_quant4_parity():
+    torch.manual_seed(11)
+    x = torch.randn(16, 256).to(torch.bfloat16)
+    pg, sg, zg = ops.quant4_pack(x.to(DEV))
+    pw, sw, zw = ref.quant4_pack(x)
+    assert torch.equal(pg.cpu(), pw.reshape(16, 128))
+    y = ops.quant4_unpack(pg, sg, zg).cpu()
+    yw = ref.quant4_unpack(pw, sw, zw)
+    assert torch.allclose(y.float(), yw.float(), atol=2e-2)
+
+
+def test_block_forward_gpu_vs_cpu():
+    """Whole llama block: GPU HIP path vs CPU reference path."""
+    from bloombee_amd.kv import PagedKVCache
+    from bloombee_amd.models.base import resolve_config
+    from bloombee_amd.models.llama.block import LlamaBlock
+
+    cfg = resolve_config("llama-mini-gpu")
+    torch.manual_seed(12)
+    blk = LlamaBlock(cfg, 0).init_random(3)
+    B, T = 2, 24
+    x = (torch.randn(B, T, cfg.hidden_size) * 0.1).to(torch.bfloat16)
+    start = torch.zeros(B, dtype=torch.int32)
+
+    pool_c = PagedKVCache(1, cfg.num_key_value_heads, cfg.head_dim,
+                          max_tokens=4096, device="cpu", dtype=torch.bfloat16)
+    kv_c = pool_c.allocate(B, 256)
+    kv_c.extend(T)
+    y_cpu = blk.forward_inference(x, kv_c, start)
+
+    blk_g = blk.to(DEV)
+    pool_g = PagedKVCache(1, cfg.num_key_value_heads, cfg.head_dim,
+                          max_tokens=4096, device=DEV, dtype=torch.bfloat16)
+    kv_g = pool_g.allocate(B, 256)
+    kv_g.extend(T)
+    y_gpu = blk_g.forward_inference(x.to(DEV), kv_g, start.to(DEV)).cpu()
+    assert torch.allclose(y_gpu.float(), y_cpu.float(), atol=5e-2), (
+        (y_gpu.float() - y_cpu.float()).abs().max()
+    )
+
+
+def test_generate_gpu_matches_cpu():
+    """End-to-end greedy decode: GPU engine tokens == CPU engine tokens."""
+    from bloombee_amd.engine import LocalEngine
+
+    ids = torch.randint(0, 1000, (2, 8), generator=torch.Generator().manual_seed(5))
+    eng_c = LocalEngine("llama-tiny", device="cpu", seed=4, kv_max_tokens=4096)
+    out_c = eng_c.generate_greedy(ids, 8)
+    eng_g = LocalEngine("llama-tiny", device=DEV, seed=4, kv_max_tokens=4096)
+    out_g = eng_g.generate_greedy(ids, 8)
+    # bf16 kernel-order differences may flip a late token; require the first
+    # few greedy tokens to match exactly (reference full-model test tolerance)
+    assert torch.equal(out_c[:, :4], out_g.cpu()[:, :4])

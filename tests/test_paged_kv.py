@@ -1,0 +1,121 @@
+"""Paged KV manager invariants (mirrors reference tests/test_paged_kv.py and
+test_cache.py: alloc semantics, budget backpressure, commit/rollback)."""
+import threading
+import time
+
+import pytest
+import torch
+
+from bloombee_amd.kv import AllocationFailed, PagedKVCache
+from bloombee_amd.ops import reference as ref
+
+
+def make_cache(max_tokens=1024, layers=2, heads=2, dim=16, page=16):
+    return PagedKVCache(layers, heads, dim, page_size=page, max_tokens=max_tokens,
+                        device="cpu", dtype=torch.float32)
+
+
+def test_alloc_and_release_budget():
+    c = make_cache(max_tokens=1024)
+    assert c.tokens_left == 1024
+    h = c.allocate(batch_size=2, max_length=256)
+    assert c.tokens_left == 1024 - 512
+    h.close()
+    assert c.tokens_left == 1024
+
+
+def test_alloc_too_large_raises():
+    c = make_cache(max_tokens=128)
+    with pytest.raises(AllocationFailed):
+        c.allocate(batch_size=2, max_length=128)
+
+
+def test_alloc_timeout_then_unblock():
+    c = make_cache(max_tokens=256)
+    h1 = c.allocate(2, 64)
+    with pytest.raises(AllocationFailed):
+        c.allocate(2, 96, timeout=0.05)
+
+    results = {}
+
+    def waiter():
+        results["h"] = c.allocate(2, 96, timeout=5.0)
+
+    t = threading.Thread(target=waiter)
+    t.start()
+    time.sleep(0.05)
+    h1.close()
+    t.join(timeout=5)
+    assert "h" in results
+    results["h"].close()
+
+
+def test_extend_allocates_pages_lazily():
+    c = make_cache(max_tokens=1024)
+    h = c.allocate(1, 256)
+    assert len(h.seqs[0].pages) == 0
+    h.extend(17)
+    assert len(h.seqs[0].pages) == 2  # 17 tokens -> 2 pages of 16
+    assert h.lengths == [17]
+    h.close()
+
+
+def test_commit_rollback_frees_pages():
+    c = make_cache(max_tokens=1024)
+    h = c.allocate(1, 256)
+    h.extend(16)                      # committed
+    free_before = len(c._free_pages)
+    h.extend(40, speculative=True)    # spec tokens -> 3 extra pages (56 tok)
+    assert h.lengths == [56] and h.committed_lengths == [16]
+    h.commit(accepted=[8])            # keep 8 of 40
+    assert h.committed_lengths == [24]
+    assert h.lengths == [24]
+    assert len(c._free_pages) == free_before - 1  # 24 tokens -> 2 pages
+    h.rollback()                      # no-op now
+    assert h.lengths == [24]
+    h.close()
+
+
+def test_truncate_failover():
+    c = make_cache(max_tokens=1024)
+    h = c.allocate(1, 256)
+    h.extend(100)
+    h.truncate([30])
+    assert h.committed_lengths == [30]
+    assert len(h.seqs[0].pages) == 2
+    h.close()
+
+
+def test_write_gather_roundtrip():
+    torch.manual_seed(0)
+    c = make_cache(max_tokens=1024, layers=1, heads=2, dim=16)
+    h = c.allocate(2, 64)
+    T = 20
+    start = torch.tensor([s.l_spec for s in h.seqs], dtype=torch.int32)
+    h.extend(T)
+    k = torch.randn(2, 2, T, 16)
+    v = torch.randn(2, 2, T, 16)
+    ref.kv_write(k, v, h.k_pages(0), h.v_pages(0), h.page_table(), start)
+    for b in range(2):
+        kg, vg = ref.kv_gather(h.k_pages(0), h.v_pages(0), h.page_table(), T, b)
+        assert torch.equal(kg, k[b])
+        assert torch.equal(vg, v[b])
+    # append one more token at position T
+    start2 = torch.tensor([T, T], dtype=torch.int32)
+    h.extend(1)
+    k2 = torch.randn(2, 2, 1, 16)
+    v2 = torch.randn(2, 2, 1, 16)
+    ref.kv_write(k2, v2, h.k_pages(0), h.v_pages(0), h.page_table(), start2)
+    kg, vg = ref.kv_gather(h.k_pages(0), h.v_pages(0), h.page_table(), T + 1, 0)
+    assert torch.equal(kg[:, -1], k2[0, :, 0])
+    h.close()
+
+
+def test_independent_page_lists_per_sequence():
+    c = make_cache(max_tokens=1024)
+    h = c.allocate(3, 128)
+    h.extend(33)
+    pages = [set(s.pages) for s in h.seqs]
+    assert pages[0] & pages[1] == set()
+    assert pages[1] & pages[2] == set()
+    h.close()
