@@ -106,29 +106,15 @@ class LlamaBlock(torch.nn.Module):
 
         x = ops.rms_norm(hidden, self.input_norm_w, cfg.rms_norm_eps)
         qkv = F.linear(x, self.qkv_w)                      # (B, T, (Hq+2Hkv)D)
-        if T == 1:
-            qkv = qkv.view(B, Hq + 2 * Hkv, 1, D)
-            q, k, v = qkv.split([Hq, Hkv, Hkv], dim=1)
-        else:
-            qkv = qkv.view(B, T, Hq + 2 * Hkv, D).permute(0, 2, 1, 3)
-            q, k, v = qkv.split([Hq, Hkv, Hkv], dim=1)
-            q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
-
-        if position_ids is None:
-            position_ids = start_pos.view(B, 1).int() + torch.arange(
-                T, device=hidden.device, dtype=torch.int32).view(1, T)
         cos, sin = self.rope.get(hidden.device)
-        q, k = ops.rope_apply_(q, k, cos, sin, position_ids)
-
-        ops.kv_write(k, v, kv.k_pages(self.layer_index), kv.v_pages(self.layer_index),
-                     kv.page_table(), start_pos)
-        attn = ops.attn_paged(q, kv.k_pages(self.layer_index),
-                              kv.v_pages(self.layer_index), kv.page_table(),
-                              start_pos, self.scale)
-        if T == 1:
-            attn = attn.view(B, 1, Hq * D)
-        else:
-            attn = attn.permute(0, 2, 1, 3).reshape(B, T, Hq * D)
+        kp = kv.k_pages(self.layer_index)
+        vp = kv.v_pages(self.layer_index)
+        pt = kv.page_table()
+        # fused RoPE + paged KV write on the raw GEMM output, then attention
+        # reads q in place and emits the O-projection input — no transposes
+        ops.rope_kv_write_(qkv, Hq, Hkv, cos, sin, position_ids, kp, vp, pt,
+                           start_pos)
+        attn = ops.attn_paged_qkv(qkv, Hq, Hkv, kp, vp, pt, start_pos, self.scale)
         a = F.linear(attn, self.o_w)
 
         # h2 = hidden + a fused into the post-attention norm

@@ -8,6 +8,7 @@ from bloombee_amd.ops.interface import (  # noqa: F401
     HAVE_HIP_OPS,
     attn_decode,
     attn_paged,
+    attn_paged_qkv,
     attn_prefill,
     gelu_tanh,
     hip_ops,
@@ -20,6 +21,7 @@ from bloombee_amd.ops.interface import (  # noqa: F401
     rms_norm,
     rms_norm_residual,
     rope_apply_,
+    rope_kv_write_,
     rope_cos_sin,
     swiglu,
 )

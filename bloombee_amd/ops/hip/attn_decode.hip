@@ -72,7 +72,9 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
     float* __restrict__ part_ml,                 // (B*Hkv*n_split, G, 2)
     float* __restrict__ part_acc,                // (B*Hkv*n_split, G, D)
     int B, int Hkv, int G, int P, int maxp, int n_split, int window,
-    float scale) {
+    float scale, long q_sb, long q_sh, long out_sb, long out_sh) {
+  // q_sb / out_sb: elements between consecutive batch rows of q / out —
+  // lets the kernel read the q section of a fused QKV GEMM output directly.
   constexpr int E = D / 16;  // elements per lane
   const int bh = blockIdx.x;
   const int split = blockIdx.y;
@@ -100,7 +102,7 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
 #pragma unroll
   for (int g = 0; g < MAXG; ++g) {
     if (g < G)
-      load_bf16_e<E>(q + ((long)b * Hkv * G + kvh * G + g) * D + li * E, qv[g]);
+      load_bf16_e<E>(q + (long)b * q_sb + (kvh * G + g) * q_sh + li * E, qv[g]);
   }
 
   float m2[MAXG], l[MAXG], acc[MAXG][E];
@@ -183,7 +185,7 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
       const float inv = (l[g] > 0.f) ? 1.f / l[g] : 0.f;
 #pragma unroll
       for (int j = 0; j < E; ++j) o[j] = acc[g][j] * inv;
-      store_bf16_e<E>(out + ((long)b * Hkv * G + kvh * G + g) * D + li * E, o);
+      store_bf16_e<E>(out + (long)b * out_sb + (kvh * G + g) * out_sh + li * E, o);
     }
   } else {
     const long pbase = ((long)bh * n_split + split);
@@ -205,7 +207,8 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
 template <int D>
 __global__ void attn_decode_combine_kernel(
     const float* __restrict__ part_ml, const float* __restrict__ part_acc,
-    unsigned short* __restrict__ out, int Hkv, int G, int n_split) {
+    unsigned short* __restrict__ out, int Hkv, int G, int n_split,
+    long out_sb, long out_sh) {
   constexpr int E = D / 16;
   const int bh = blockIdx.x;
   const int g = threadIdx.x / 16;
@@ -233,5 +236,5 @@ __global__ void attn_decode_combine_kernel(
 #pragma unroll
   for (int j = 0; j < E; ++j) o[j] = acc[j] * inv;
   const int b = bh / Hkv, kvh = bh % Hkv;
-  store_bf16_e<E>(out + ((long)b * Hkv * G + kvh * G + g) * D + li * E, o);
+  store_bf16_e<E>(out + (long)b * out_sb + (kvh * G + g) * out_sh + li * E, o);
 }

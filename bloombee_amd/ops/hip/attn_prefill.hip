@@ -32,7 +32,10 @@ __global__ __launch_bounds__(256) void attn_prefill_kernel(
     const int* __restrict__ page_table,          // (B, maxp)
     const int* __restrict__ q_start,             // (B,)
     unsigned short* __restrict__ out,            // (B, Hq, Tq, D)
-    int B, int Hq, int G, int Tq, int P, int maxp, int window, float scale) {
+    int B, int Hq, int G, int Tq, int P, int maxp, int window, float scale,
+    long q_sb, long q_st, long q_sh, long o_sb, long o_st, long o_sh) {
+  // q element (b, h, row, d) at b*q_sb + row*q_st + h*q_sh + d (same for
+  // out) — reads the q section of a fused (B, T, X, D) QKV output directly.
   constexpr int KROW_B = D * 2 + 16;  // K LDS row stride (pad kills conflicts)
   constexpr int VROW_B = KVBLK * 2 + 16;  // V^T LDS row stride (80 B)
   constexpr int PROW_B = KVBLK * 2 + 16;
@@ -65,7 +68,7 @@ __global__ __launch_bounds__(256) void attn_prefill_kernel(
 #pragma unroll
   for (int kk = 0; kk < NKK; ++kk) {
     qfrag[kk] = as_bf16x8(*reinterpret_cast<const short8*>(
-        q + (((long)bh) * Tq + my_qrow) * D + hi * 8 + 32 * kk));
+        q + (long)b * q_sb + (long)my_qrow * q_st + (long)h * q_sh + hi * 8 + 32 * kk));
   }
 
   // online-softmax state: this lane's 4 C-rows (r = hi*4 + reg)
@@ -181,7 +184,8 @@ __global__ __launch_bounds__(256) void attn_prefill_kernel(
     const float inv = (l[reg] > 0.f) ? 1.f / l[reg] : 0.f;
 #pragma unroll
     for (int n = 0; n < NDT; ++n) {
-      out[(((long)bh) * Tq + qrow) * D + li + 16 * n] = f2bf(acc_o[n][reg] * inv);
+      out[(long)b * o_sb + (long)qrow * o_st + (long)h * o_sh + li + 16 * n] =
+          f2bf(acc_o[n][reg] * inv);
     }
   }
 }

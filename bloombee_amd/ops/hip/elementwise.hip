@@ -40,7 +40,7 @@ __global__ __launch_bounds__(BLOCK) void rmsnorm_kernel(
         float r[8];
         load_bf16x8(rr + d, r);
 #pragma unroll
-        for (int j = 0; j < 8; ++j) v[it][j] += r[j];
+        for (int j = 0; j < 8; ++j) v[it][j] = bf2f(f2bf(v[it][j] + r[j]));
         if (out_h) store_bf16x8(out_h + row * (long)H + d, v[it]);
       }
 #pragma unroll
@@ -213,5 +213,53 @@ __global__ void add_kernel(const unsigned short* __restrict__ a,
 #pragma unroll
     for (int j = 0; j < 8; ++j) o[j] = va[j] + vb[j];
     store_bf16x8(out + idx * 8, o);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Fused RoPE + paged-KV write on the raw QKV GEMM output (B, T, (Hq+2Hkv)*D).
+// q heads: rotary in place. k heads: rotary -> k_pages (never written back).
+// v heads: straight copy -> v_pages. One launch replaces rope + kv_write and
+// removes every transpose copy between the QKV GEMM and attention.
+// grid: (B*T, Hq+2Hkv), block: 64.
+// ---------------------------------------------------------------------------
+
+__global__ void rope_kv_write_kernel(
+    unsigned short* __restrict__ qkv, const float* __restrict__ cos_t,
+    const float* __restrict__ sin_t, const int* __restrict__ pos,
+    unsigned short* __restrict__ k_pages, unsigned short* __restrict__ v_pages,
+    const int* __restrict__ page_table, const int* __restrict__ start_pos,
+    int B, int Hq, int Hkv, int T, int D, int P, int maxp) {
+  const int bt = blockIdx.x;
+  const int b = bt / T, t = bt % T;
+  const int h = blockIdx.y;
+  const int X = Hq + 2 * Hkv;
+  unsigned short* base = qkv + (((long)b * T + t) * X + h) * D;
+
+  const int abspos = start_pos[b] + t;
+  if (h >= Hq + Hkv) {  // v head: copy to pages
+    const int hv = h - Hq - Hkv;
+    const int page = page_table[b * maxp + abspos / P];
+    unsigned short* dst =
+        v_pages + (((long)page * Hkv + hv) * P + (abspos % P)) * D;
+    for (int i = threadIdx.x; i < D / 8; i += blockDim.x)
+      *reinterpret_cast<short8*>(dst + i * 8) =
+          *reinterpret_cast<const short8*>(base + i * 8);
+    return;
+  }
+  const int p = pos ? pos[b * T + t] : abspos;
+  const float* c = cos_t + (long)p * (D / 2);
+  const float* s = sin_t + (long)p * (D / 2);
+  unsigned short* dst = base;
+  if (h >= Hq) {  // k head: write rotated values to pages only
+    const int hk = h - Hq;
+    const int page = page_table[b * maxp + abspos / P];
+    dst = k_pages + (((long)page * Hkv + hk) * P + (abspos % P)) * D;
+  }
+  for (int i = threadIdx.x; i < D / 2; i += blockDim.x) {
+    float x1 = bf2f(base[i]), x2 = bf2f(base[i + D / 2]);
+    float cc = c[i], ss = s[i];
+    dst[i] = f2bf(x1 * cc - x2 * ss);
+    dst[i + D / 2] = f2bf(x2 * cc + x1 * ss);
   }
 }

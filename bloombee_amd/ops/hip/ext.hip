@@ -185,13 +185,15 @@ static void attn_decode_launch(const torch::Tensor& q, const torch::Tensor& kp,
                                const torch::Tensor& ctx, torch::Tensor& out,
                                torch::Tensor& pml, torch::Tensor& pacc, int B,
                                int Hkv, int G, int P, int maxp, int n_split,
-                               int window, float scale) {
+                               int window, float scale, long q_off, long q_sb,
+                               long q_sh, long out_sb, long out_sh) {
   dim3 grid(B * Hkv, n_split);
   auto launch = [&](auto maxg) {
     attn_decode_kernel<D, decltype(maxg)::value><<<grid, 256, 0, cur_stream()>>>(
-        bf_ptr(q), bf_ptr(kp), bf_ptr(vp), pt.data_ptr<int>(),
+        bf_ptr(q) + q_off, bf_ptr(kp), bf_ptr(vp), pt.data_ptr<int>(),
         ctx.data_ptr<int>(), bf_ptr_mut(out), pml.data_ptr<float>(),
-        pacc.data_ptr<float>(), B, Hkv, G, P, maxp, n_split, window, scale);
+        pacc.data_ptr<float>(), B, Hkv, G, P, maxp, n_split, window, scale,
+        q_sb, q_sh, out_sb, out_sh);
   };
   if (G <= 4) launch(std::integral_constant<int, 4>{});
   else if (G <= 8) launch(std::integral_constant<int, 8>{});
@@ -199,17 +201,16 @@ static void attn_decode_launch(const torch::Tensor& q, const torch::Tensor& kp,
   if (n_split > 1) {
     attn_decode_combine_kernel<D><<<B * Hkv, G * 16, 0, cur_stream()>>>(
         pml.data_ptr<float>(), pacc.data_ptr<float>(), bf_ptr_mut(out), Hkv, G,
-        n_split);
+        n_split, out_sb, out_sh);
   }
 }
 
-torch::Tensor attn_decode(torch::Tensor q, torch::Tensor k_pages,
-                          torch::Tensor v_pages, torch::Tensor page_table,
-                          torch::Tensor ctx_lens, double scale, long window,
-                          long n_split_req) {
-  CHECK_DEV(q); CHECK_BF16(q); CHECK_CONTIG(q);
-  TORCH_CHECK(q.dim() == 4 && q.size(2) == 1, "attn_decode expects (B, Hq, 1, D)");
-  const int B = q.size(0), Hq = q.size(1), D = q.size(3);
+static torch::Tensor attn_decode_core(
+    const torch::Tensor& q, const torch::Tensor& k_pages,
+    const torch::Tensor& v_pages, const torch::Tensor& page_table,
+    const torch::Tensor& ctx_lens, double scale, long window, long n_split_req,
+    int B, int Hq, int D, long q_off, long q_sb, long q_sh,
+    torch::Tensor out, long out_sb, long out_sh) {
   const int Hkv = k_pages.size(1), P = k_pages.size(2), maxp = page_table.size(1);
   const int G = Hq / Hkv;
   TORCH_CHECK(Hq % Hkv == 0);
@@ -219,7 +220,6 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor k_pages,
     // fill the chip: >= ~1024 workgroups (256 CUs x 8 XCDs, guide §1)
     n_split = (int)std::max<long>(1, std::min<long>(32, 1024 / std::max(1, B * Hkv)));
   }
-  auto out = torch::empty({B, Hq, 1, D}, q.options());
   auto fopt = torch::TensorOptions().device(q.device()).dtype(at::kFloat);
   torch::Tensor pml, pacc;
   if (n_split > 1) {
@@ -229,21 +229,51 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor k_pages,
     pml = torch::empty({1}, fopt);
     pacc = torch::empty({1}, fopt);
   }
-  if (D == 128)
-    attn_decode_launch<128>(q, k_pages, v_pages, page_table, ctx_lens, out, pml,
-                            pacc, B, Hkv, G, P, maxp, n_split, (int)window,
-                            (float)scale);
-  else if (D == 64)
-    attn_decode_launch<64>(q, k_pages, v_pages, page_table, ctx_lens, out, pml,
-                           pacc, B, Hkv, G, P, maxp, n_split, (int)window,
-                           (float)scale);
-  else if (D == 256)
-    attn_decode_launch<256>(q, k_pages, v_pages, page_table, ctx_lens, out, pml,
-                            pacc, B, Hkv, G, P, maxp, n_split, (int)window,
-                            (float)scale);
-  else
-    TORCH_CHECK(false, "unsupported head_dim ", D);
+  auto go = [&](auto d) {
+    attn_decode_launch<decltype(d)::value>(
+        q, k_pages, v_pages, page_table, ctx_lens, out, pml, pacc, B, Hkv, G,
+        P, maxp, n_split, (int)window, (float)scale, q_off, q_sb, q_sh,
+        out_sb, out_sh);
+  };
+  if (D == 128) go(std::integral_constant<int, 128>{});
+  else if (D == 64) go(std::integral_constant<int, 64>{});
+  else if (D == 256) go(std::integral_constant<int, 256>{});
+  else TORCH_CHECK(false, "unsupported head_dim ", D);
   return out;
+}
+
+torch::Tensor attn_decode(torch::Tensor q, torch::Tensor k_pages,
+                          torch::Tensor v_pages, torch::Tensor page_table,
+                          torch::Tensor ctx_lens, double scale, long window,
+                          long n_split_req) {
+  CHECK_DEV(q); CHECK_BF16(q); CHECK_CONTIG(q);
+  TORCH_CHECK(q.dim() == 4 && q.size(2) == 1, "attn_decode expects (B, Hq, 1, D)");
+  const int B = q.size(0), Hq = q.size(1), D = q.size(3);
+  auto out = torch::empty({B, Hq, 1, D}, q.options());
+  return attn_decode_core(q, k_pages, v_pages, page_table, ctx_lens, scale,
+                          window, n_split_req, B, Hq, D,
+                          /*q_off*/ 0, (long)Hq * D, (long)D,
+                          out, (long)Hq * D, (long)D);
+}
+
+// Fused-QKV decode: q section of qkv (B, 1, (Hq+2Hkv)*D) read in place,
+// out (B, 1, Hq*D) ready for the O-projection GEMM — zero transposes.
+torch::Tensor attn_decode_qkv(torch::Tensor qkv, long Hq, torch::Tensor k_pages,
+                              torch::Tensor v_pages, torch::Tensor page_table,
+                              torch::Tensor ctx_lens, double scale, long window,
+                              long n_split_req) {
+  CHECK_DEV(qkv); CHECK_BF16(qkv); CHECK_CONTIG(qkv);
+  const int Hkv = k_pages.size(1);
+  const int D = k_pages.size(3);
+  const int B = qkv.size(0);
+  TORCH_CHECK(qkv.dim() == 3 && qkv.size(1) == 1 &&
+              qkv.size(2) == (Hq + 2 * Hkv) * D, "bad fused qkv shape");
+  auto out = torch::empty({B, 1, Hq * D}, qkv.options());
+  return attn_decode_core(qkv, k_pages, v_pages, page_table, ctx_lens, scale,
+                          window, n_split_req, B, (int)Hq, D,
+                          /*q_off*/ 0, (long)(Hq + 2 * Hkv) * D, (long)D,
+                          out, (long)Hq * D, (long)D)
+      .view({B, 1, Hq * D});
 }
 
 torch::Tensor attn_prefill(torch::Tensor q, torch::Tensor k_pages,
@@ -260,13 +290,69 @@ torch::Tensor attn_prefill(torch::Tensor q, torch::Tensor k_pages,
     attn_prefill_kernel<decltype(d)::value><<<grid, 256, 0, cur_stream()>>>(
         bf_ptr(q), bf_ptr(k_pages), bf_ptr(v_pages), page_table.data_ptr<int>(),
         q_start.data_ptr<int>(), bf_ptr_mut(out), B, Hq, G, Tq, P, maxp,
-        (int)window, (float)scale);
+        (int)window, (float)scale,
+        (long)Hq * Tq * D, (long)D, (long)Tq * D,
+        (long)Hq * Tq * D, (long)D, (long)Tq * D);
   };
   if (D == 128) launch(std::integral_constant<int, 128>{});
   else if (D == 64) launch(std::integral_constant<int, 64>{});
   else if (D == 256) launch(std::integral_constant<int, 256>{});
   else TORCH_CHECK(false, "unsupported head_dim ", D);
   return out;
+}
+
+// Fused-QKV prefill: q read from qkv (B, Tq, (Hq+2Hkv)*D) in place,
+// out (B, Tq, Hq*D) ready for the O-projection GEMM.
+torch::Tensor attn_prefill_qkv(torch::Tensor qkv, long Hq_,
+                               torch::Tensor k_pages, torch::Tensor v_pages,
+                               torch::Tensor page_table, torch::Tensor q_start,
+                               double scale, long window) {
+  CHECK_DEV(qkv); CHECK_BF16(qkv); CHECK_CONTIG(qkv);
+  const int Hkv = k_pages.size(1), P = k_pages.size(2), D = k_pages.size(3);
+  const int maxp = page_table.size(1);
+  const int Hq = (int)Hq_;
+  const int B = qkv.size(0), Tq = qkv.size(1);
+  const int X = Hq + 2 * Hkv;
+  TORCH_CHECK(qkv.dim() == 3 && qkv.size(2) == (long)X * D, "bad fused qkv shape");
+  const int G = Hq / Hkv;
+  auto out = torch::empty({B, Tq, (long)Hq * D}, qkv.options());
+  dim3 grid((Tq + 63) / 64, B * Hq);
+  auto launch = [&](auto d) {
+    attn_prefill_kernel<decltype(d)::value><<<grid, 256, 0, cur_stream()>>>(
+        bf_ptr(qkv), bf_ptr(k_pages), bf_ptr(v_pages), page_table.data_ptr<int>(),
+        q_start.data_ptr<int>(), bf_ptr_mut(out), B, Hq, G, Tq, P, maxp,
+        (int)window, (float)scale,
+        (long)Tq * X * D, (long)X * D, (long)D,
+        (long)Tq * Hq * D, (long)Hq * D, (long)D);
+  };
+  if (D == 128) launch(std::integral_constant<int, 128>{});
+  else if (D == 64) launch(std::integral_constant<int, 64>{});
+  else if (D == 256) launch(std::integral_constant<int, 256>{});
+  else TORCH_CHECK(false, "unsupported head_dim ", D);
+  return out;
+}
+
+// Fused RoPE + paged-KV write on the raw QKV GEMM output.
+void rope_kv_write_(torch::Tensor qkv, long Hq_, long Hkv_, torch::Tensor cos_t,
+                    torch::Tensor sin_t, c10::optional<torch::Tensor> pos,
+                    torch::Tensor k_pages, torch::Tensor v_pages,
+                    torch::Tensor page_table, torch::Tensor start_pos) {
+  CHECK_DEV(qkv); CHECK_BF16(qkv); CHECK_CONTIG(qkv);
+  const int Hq = (int)Hq_, Hkv = (int)Hkv_;
+  const int P = k_pages.size(2), D = k_pages.size(3);
+  const int maxp = page_table.size(1);
+  const int B = qkv.size(0), T = qkv.size(1);
+  TORCH_CHECK(qkv.size(2) == (long)(Hq + 2 * Hkv) * D, "bad fused qkv shape");
+  const int* pos_ptr = nullptr;
+  if (pos.has_value()) {
+    TORCH_CHECK(pos->scalar_type() == at::kInt);
+    pos_ptr = pos->data_ptr<int>();
+  }
+  dim3 grid(B * T, Hq + 2 * Hkv);
+  rope_kv_write_kernel<<<grid, 64, 0, cur_stream()>>>(
+      bf_ptr_mut(qkv), cos_t.data_ptr<float>(), sin_t.data_ptr<float>(), pos_ptr,
+      bf_ptr_mut(k_pages), bf_ptr_mut(v_pages), page_table.data_ptr<int>(),
+      start_pos.data_ptr<int>(), B, Hq, Hkv, T, D, P, maxp);
 }
 
 // ---------------------------------------------------------------------------
@@ -326,6 +412,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("kv_write", &kv_write);
   m.def("kv_gather", &kv_gather);
   m.def("attn_decode", &attn_decode);
+  m.def("attn_decode_qkv", &attn_decode_qkv);
+  m.def("attn_prefill_qkv", &attn_prefill_qkv);
+  m.def("rope_kv_write_", &rope_kv_write_);
   m.def("attn_prefill", &attn_prefill);
   m.def("quant4_pack", &quant4_pack);
   m.def("quant4_unpack", &quant4_unpack);

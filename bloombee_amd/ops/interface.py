@@ -147,6 +147,57 @@ def attn_prefill(
                           sliding_window=window if window > 0 else None)
 
 
+def rope_kv_write_(qkv, Hq: int, Hkv: int, cos, sin, position_ids,
+                   k_pages, v_pages, page_table, start_pos):
+    """Fused RoPE + paged KV write on the raw QKV GEMM output.
+
+    qkv: (B, T, (Hq+2Hkv)*D) — q section roped in place; k roped -> pages;
+    v copied -> pages. position_ids None => start_pos[b] + t.
+    """
+    if _on_gpu(qkv):
+        _require_ext()
+        hip_ops.rope_kv_write_(qkv, Hq, Hkv, cos, sin,
+                               position_ids.int() if position_ids is not None else None,
+                               k_pages, v_pages, page_table, start_pos.int())
+        return
+    B, T, _ = qkv.shape
+    D = k_pages.shape[3]
+    q = qkv[..., : Hq * D].view(B, T, Hq, D).permute(0, 2, 1, 3)
+    k = qkv[..., Hq * D:(Hq + Hkv) * D].view(B, T, Hkv, D).permute(0, 2, 1, 3)
+    v = qkv[..., (Hq + Hkv) * D:].view(B, T, Hkv, D).permute(0, 2, 1, 3)
+    if position_ids is None:
+        position_ids = start_pos.view(B, 1).long() + torch.arange(T).view(1, T)
+    q2, k2 = ref.rope_apply(q, k, cos, sin, position_ids)
+    qkv[..., : Hq * D].copy_(q2.permute(0, 2, 1, 3).reshape(B, T, Hq * D))
+    ref.kv_write(k2, v.contiguous(), k_pages, v_pages, page_table, start_pos)
+
+
+def attn_paged_qkv(qkv, Hq: int, Hkv: int, k_pages, v_pages, page_table,
+                   q_start, scale=None, window: int = 0):
+    """Attention reading q straight from the fused QKV buffer.
+
+    qkv: (B, T, (Hq+2Hkv)*D) with the q section already roped and the new
+    tokens' K/V already in the pages (rope_kv_write_ first).
+    Returns (B, T, Hq*D) — the O-projection GEMM input, no transposes.
+    """
+    B, T, _ = qkv.shape
+    D = k_pages.shape[3]
+    if scale is None:
+        scale = 1.0 / math.sqrt(D)
+    if _on_gpu(qkv):
+        _require_ext()
+        if T == 1:
+            ctx = q_start + 1
+            return hip_ops.attn_decode_qkv(qkv, Hq, k_pages, v_pages, page_table,
+                                           ctx.int(), scale, window, 0)
+        return hip_ops.attn_prefill_qkv(qkv, Hq, k_pages, v_pages, page_table,
+                                        q_start.int(), scale, window)
+    q = qkv[..., : Hq * D].view(B, T, Hq, D).permute(0, 2, 1, 3).contiguous()
+    out = ref.attn_paged(q, k_pages, v_pages, page_table, q_start.long(), scale,
+                         sliding_window=window if window > 0 else None)
+    return out.permute(0, 2, 1, 3).reshape(B, T, Hq * D)
+
+
 def attn_paged(q, k_pages, v_pages, page_table, q_start, scale=None, window: int = 0,
                tree_mask=None):
     """Unified entry: picks decode vs prefill kernel by Tq."""

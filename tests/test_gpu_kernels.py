@@ -245,3 +245,39 @@ def test_generate_gpu_matches_cpu():
     # bf16 kernel-order differences may flip a late token; require the first
     # few greedy tokens to match exactly (reference full-model test tolerance)
     assert torch.equal(out_c[:, :4], out_g.cpu()[:, :4])
+
+
+def test_fused_qkv_path_parity():
+    """rope_kv_write_ + attn_paged_qkv on GPU vs the composed CPU reference."""
+    torch.manual_seed(20)
+    B, Hq, Hkv, D, P = 2, 8, 2, 128, 16
+    X = Hq + 2 * Hkv
+    maxp = 8
+    for T, start0 in [(1, 30), (40, 0)]:
+        kp_c = torch.zeros(B * maxp + 2, Hkv, P, D, dtype=torch.bfloat16)
+        vp_c = torch.zeros_like(kp_c)
+        pt = torch.arange(B * maxp, dtype=torch.int32).reshape(B, maxp)
+        qkv = (torch.randn(B, T, X * D) * 0.3).to(torch.bfloat16)
+        start = torch.full((B,), start0, dtype=torch.int32)
+        if start0:
+            # pre-fill some history
+            kh = torch.randn(B, Hkv, start0, D).to(torch.bfloat16)
+            vh = torch.randn(B, Hkv, start0, D).to(torch.bfloat16)
+            ref.kv_write(kh, vh, kp_c, vp_c, pt, torch.zeros(B, dtype=torch.int32))
+        kp_g, vp_g = kp_c.clone().to(DEV), vp_c.clone().to(DEV)
+        qkv_g = qkv.clone().to(DEV)
+        cos, sin = ref.rope_cos_sin(D, 512)
+
+        ops.rope_kv_write_(qkv, Hq, Hkv, cos, sin, None, kp_c, vp_c, pt, start)
+        out_c = ops.attn_paged_qkv(qkv, Hq, Hkv, kp_c, vp_c, pt, start.long())
+
+        ops.rope_kv_write_(qkv_g, Hq, Hkv, cos.to(DEV), sin.to(DEV), None,
+                           kp_g, vp_g, pt.to(DEV), start.to(DEV))
+        out_g = ops.attn_paged_qkv(qkv_g, Hq, Hkv, kp_g, vp_g, pt.to(DEV),
+                                   start.to(DEV))
+        assert torch.allclose(qkv_g.cpu().float(), qkv.float(), atol=2e-2), \
+            f"T={T}: roped q mismatch {(qkv_g.cpu().float()-qkv.float()).abs().max()}"
+        assert torch.allclose(kp_g.cpu().float(), kp_c.float(), atol=2e-2), \
+            f"T={T}: k pages mismatch"
+        assert torch.allclose(out_g.cpu().float(), out_c.float(), atol=3e-2), \
+            f"T={T}: attn out mismatch {(out_g.cpu().float()-out_c.float()).abs().max()}"
