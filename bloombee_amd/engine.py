@@ -40,10 +40,10 @@ class BlockStack(torch.nn.Module):
         self.rope = RopeTables(config)
         self.blocks = torch.nn.ModuleList()
         for i in range(start, end):
-            blk = block_cls(config, layer_index=i, rope=self.rope)
+            blk = block_cls(config, layer_index=i, rope=self.rope).to(device)
             blk.init_random(seed=seed * 10_000 + i)
             blk.layer_index = i - start  # KV page index local to this stack's pool
-            self.blocks.append(blk.to(device))
+            self.blocks.append(blk)
         self.device = torch.device(device)
 
     def make_kv(self, max_tokens: int) -> PagedKVCache:

@@ -78,15 +78,20 @@ class LlamaBlock(torch.nn.Module):
 
     @torch.no_grad()
     def init_random(self, seed: Optional[int] = None):
-        gen = torch.Generator().manual_seed(
-            seed if seed is not None else 1234 + self.layer_index)
+        """Random-init weights. Generates on the parameters' current device —
+        device-side RNG for resident blocks (seconds for 70B-class shards vs
+        minutes of host randn + transfer). Same (device kind, seed) => same
+        weights, which is what the multi-rank parity argument needs."""
+        s = seed if seed is not None else 1234 + self.layer_index
+        dev = self.input_norm_w.device
+        gen = torch.Generator(device=dev).manual_seed(s)
         std = 0.02 / math.sqrt(2 * self.config.num_hidden_layers)
         for name, w in self.named_parameters():
             if name.endswith("norm_w"):
                 w.fill_(1.0)
             else:
-                w.copy_(torch.randn(w.shape, generator=gen, dtype=torch.float32)
-                        .mul_(std).to(w.dtype))
+                w.copy_(torch.randn(w.shape, generator=gen, dtype=torch.float32,
+                                    device=dev).mul_(std).to(w.dtype))
         return self
 
     # ------------------------------------------------------------------
