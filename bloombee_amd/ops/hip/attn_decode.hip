@@ -131,13 +131,20 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
       for (int j = 0; j < E; ++j) dot += kv[j] * qv[g][j];
       dot = group16_reduce_sum(dot);
       const float s = dot * sc2;
-      const float mn = fmaxf(m2[g], s);
-      const float corr = fast_exp2(m2[g] - mn);
-      const float p = fast_exp2(s - mn);
-      l[g] = l[g] * corr + p;
+      // fast path: running max unchanged (the common case after the first
+      // few positions) — skip the acc rescale entirely (guide T13 class)
+      if (s <= m2[g]) {
+        const float p = fast_exp2(s - m2[g]);
+        l[g] += p;
 #pragma unroll
-      for (int j = 0; j < E; ++j) acc[g][j] = acc[g][j] * corr + p * vv[j];
-      m2[g] = mn;
+        for (int j = 0; j < E; ++j) acc[g][j] += p * vv[j];
+      } else {
+        const float corr = fast_exp2(m2[g] - s);
+        l[g] = l[g] * corr + 1.f;
+#pragma unroll
+        for (int j = 0; j < E; ++j) acc[g][j] = acc[g][j] * corr + vv[j];
+        m2[g] = s;
+      }
     }
   }
 

@@ -81,6 +81,17 @@ class PipelineStage:
         # persistent recv buffers per micro-batch
         self._recv_buf = [torch.empty(self.mb, 1, cfg.hidden_size, dtype=cfg.dtype,
                                       device=self.device) for _ in range(self.M)]
+        # hipGraph decode capture (single-rank fast path): the whole decode
+        # step — embed, 32 blocks, LM head, argmax, position bump — replays as
+        # one graph (reference utils/cuda_graphs.py analog; on MI355X the
+        # ~300-launch step is otherwise launch-gap bound)
+        from bloombee_amd.config import get_config
+        self._use_graphs = (get_config().use_hip_graphs and self.world == 1
+                            and self.device.type == "cuda")
+        self._graph = None
+        self._ids_buf: Optional[torch.Tensor] = None
+        self._pos_buf: Optional[torch.Tensor] = None
+        self._eager_steps = 0
 
     # ------------------------------------------------------------------
     def _run_local(self, hidden: torch.Tensor, view, start_pos: torch.Tensor):
@@ -90,6 +101,48 @@ class PipelineStage:
         return [SessionView(self.kv, j * self.mb, (j + 1) * self.mb)
                 for j in range(self.M)]
 
+    # ------------------------------------------------------------------
+    # single-rank graphed decode
+    # ------------------------------------------------------------------
+    def _client_step_local(self):
+        """Whole decode step on persistent buffers (graph-capturable: every
+        host-varying quantity lives in a device tensor)."""
+        B = self.global_batch
+        hid = F.embedding(self._ids_buf.view(B, 1), self.embed)
+        h = self.stack.forward_inference(hid, self.kv, self._pos_buf)
+        from bloombee_amd import ops
+
+        y = ops.rms_norm(h[:, -1], self.final_norm_w, self.config.rms_norm_eps)
+        nxt = F.linear(y, self.lm_head_w).float().argmax(-1)
+        self._ids_buf.copy_(nxt)
+        self._pos_buf += 1
+
+    def _decode_round_single(self, ids: torch.Tensor) -> torch.Tensor:
+        B = self.global_batch
+        if self._ids_buf is None:
+            self._ids_buf = torch.empty(B, dtype=torch.long, device=self.device)
+            self._pos_buf = torch.empty(B, dtype=torch.int32, device=self.device)
+            self._pos_buf.fill_(self.kv.seqs[0].l_spec)
+        if ids is not self._ids_buf:
+            self._ids_buf.copy_(ids.to(self.device))
+        self.kv.extend(1)
+        self.kv.page_table()  # flush any new pages to the device table
+        if not self._use_graphs:
+            self._client_step_local()
+            return self._ids_buf
+        if self._graph is None:
+            if self._eager_steps < 2:  # warm up allocator/kernels first
+                self._eager_steps += 1
+                self._client_step_local()
+                return self._ids_buf
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                self._client_step_local()
+            self._graph = g
+            logger.info("decode step captured as hipGraph")
+        self._graph.replay()
+        return self._ids_buf
+
     @torch.no_grad()
     def decode_round(self, ids: Optional[torch.Tensor]) -> Optional[torch.Tensor]:
         """One decode step for the whole global batch, micro-batch pipelined.
@@ -97,6 +150,8 @@ class PipelineStage:
         rank 0: `ids` is (global_batch,) current tokens; returns next tokens.
         other ranks: pass None; returns None.
         """
+        if self.world == 1 and self.device.type == "cuda":
+            return self._decode_round_single(ids)
         cfg = self.config
         B, M, mb = self.global_batch, self.M, self.mb
         pos0 = self.kv.seqs[0].l_spec  # all sequences advance in lockstep
