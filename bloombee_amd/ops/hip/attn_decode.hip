@@ -115,7 +115,56 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
   }
 
   const long head_slab = (long)kvh * P * D;
-  for (int pos = c0 + group_id; pos < c1; pos += 16) {
+  // Unroll the position walk 2-wide: two independent K/V rows, dots and
+  // exp2 chains in flight per iteration — the per-position shfl-reduce +
+  // transcendental chain is latency-bound, not bandwidth-bound, without it.
+  const int start_pos_it = c0 + group_id;
+  int pos = (start_pos_it < lo) ? (lo + group_id) : start_pos_it;
+  for (; pos + 16 < c1; pos += 32) {
+    const int pa = pos, pb = pos + 16;
+    const int page_a = page_table[b * maxp + pa / P];
+    const int page_b = page_table[b * maxp + pb / P];
+    const long off_a =
+        ((long)page_a * Hkv) * P * D + head_slab + (long)(pa % P) * D + li * E;
+    const long off_b =
+        ((long)page_b * Hkv) * P * D + head_slab + (long)(pb % P) * D + li * E;
+    float ka[E], va[E], kb[E], vb[E];
+    load_bf16_e<E>(k_pages + off_a, ka);
+    load_bf16_e<E>(k_pages + off_b, kb);
+    load_bf16_e<E>(v_pages + off_a, va);
+    load_bf16_e<E>(v_pages + off_b, vb);
+#pragma unroll
+    for (int g = 0; g < MAXG; ++g) {
+      if (g >= G) break;
+      float da = 0.f, db = 0.f;
+#pragma unroll
+      for (int j = 0; j < E; ++j) { da += ka[j] * qv[g][j]; db += kb[j] * qv[g][j]; }
+#pragma unroll
+      for (int m = 1; m < 16; m <<= 1) {
+        da += __shfl_xor(da, m);
+        db += __shfl_xor(db, m);
+      }
+      const float sa = da * sc2, sb = db * sc2;
+      const float mx = fmaxf(sa, sb);
+      if (mx <= m2[g]) {  // fast path: no rescale (T13 class)
+        const float pa_ = fast_exp2(sa - m2[g]);
+        const float pb_ = fast_exp2(sb - m2[g]);
+        l[g] += pa_ + pb_;
+#pragma unroll
+        for (int j = 0; j < E; ++j) acc[g][j] += pa_ * va[j] + pb_ * vb[j];
+      } else {
+        const float corr = fast_exp2(m2[g] - mx);
+        const float pa_ = fast_exp2(sa - mx);
+        const float pb_ = fast_exp2(sb - mx);
+        l[g] = l[g] * corr + pa_ + pb_;
+#pragma unroll
+        for (int j = 0; j < E; ++j)
+          acc[g][j] = acc[g][j] * corr + pa_ * va[j] + pb_ * vb[j];
+        m2[g] = mx;
+      }
+    }
+  }
+  for (; pos < c1; pos += 16) {  // tail (and sliding-window head skip)
     if (pos < lo) continue;
     const int page = page_table[b * maxp + pos / P];
     const long off =
@@ -131,8 +180,6 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
       for (int j = 0; j < E; ++j) dot += kv[j] * qv[g][j];
       dot = group16_reduce_sum(dot);
       const float s = dot * sc2;
-      // fast path: running max unchanged (the common case after the first
-      // few positions) — skip the acc rescale entirely (guide T13 class)
       if (s <= m2[g]) {
         const float p = fast_exp2(s - m2[g]);
         l[g] += p;
