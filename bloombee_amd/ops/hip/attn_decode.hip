@@ -118,8 +118,10 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
   // Unroll the position walk 2-wide: two independent K/V rows, dots and
   // exp2 chains in flight per iteration — the per-position shfl-reduce +
   // transcendental chain is latency-bound, not bandwidth-bound, without it.
-  const int start_pos_it = c0 + group_id;
-  int pos = (start_pos_it < lo) ? (lo + group_id) : start_pos_it;
+  int pos = c0 + group_id;
+  // sliding window: advance in full strides so the 16-group partition of
+  // positions is preserved (no double-processing across groups)
+  if (pos < lo) pos += ((lo - pos + 15) / 16) * 16;
   for (; pos + 16 < c1; pos += 32) {
     const int pa = pos, pb = pos + 16;
     const int page_a = page_table[b * maxp + pa / P];
