@@ -115,54 +115,86 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
   }
 
   const long head_slab = (long)kvh * P * D;
-  // Unroll the position walk 2-wide: two independent K/V rows, dots and
-  // exp2 chains in flight per iteration — the per-position shfl-reduce +
-  // transcendental chain is latency-bound, not bandwidth-bound, without it.
+  // Unroll the position walk 2-wide AND software-prefetch the next pair:
+  // the per-position shfl-reduce + transcendental chain is latency-bound;
+  // loads for iteration i+1 issue (raw bf16) before iteration i's compute so
+  // ~the whole compute phase covers the HBM latency (guide T14 pattern).
   int pos = c0 + group_id;
   // sliding window: advance in full strides so the 16-group partition of
   // positions is preserved (no double-processing across groups)
   if (pos < lo) pos += ((lo - pos + 15) / 16) * 16;
-  for (; pos + 16 < c1; pos += 32) {
-    const int pa = pos, pb = pos + 16;
-    const int page_a = page_table[b * maxp + pa / P];
-    const int page_b = page_table[b * maxp + pb / P];
-    const long off_a =
-        ((long)page_a * Hkv) * P * D + head_slab + (long)(pa % P) * D + li * E;
-    const long off_b =
-        ((long)page_b * Hkv) * P * D + head_slab + (long)(pb % P) * D + li * E;
-    float ka[E], va[E], kb[E], vb[E];
-    load_bf16_e<E>(k_pages + off_a, ka);
-    load_bf16_e<E>(k_pages + off_b, kb);
-    load_bf16_e<E>(v_pages + off_a, va);
-    load_bf16_e<E>(v_pages + off_b, vb);
+
+  auto row_off = [&](int p) {
+    const int page = page_table[b * maxp + p / P];
+    return ((long)page * Hkv) * P * D + head_slab + (long)(p % P) * D + li * E;
+  };
+  constexpr int NR = E / 8 > 0 ? E / 8 : 1;  // short8 regs per row slice
+  short8 kr_a[NR], vr_a[NR], kr_b[NR], vr_b[NR];
+  short8 kn_a[NR], vn_a[NR], kn_b[NR], vn_b[NR];
+
+  auto load_raw = [&](long off, short8* kr, short8* vr) {
 #pragma unroll
-    for (int g = 0; g < MAXG; ++g) {
-      if (g >= G) break;
-      float da = 0.f, db = 0.f;
+    for (int c = 0; c < NR; ++c) {
+      kr[c] = *reinterpret_cast<const short8*>(k_pages + off + c * 8);
+      vr[c] = *reinterpret_cast<const short8*>(v_pages + off + c * 8);
+    }
+  };
+  auto cvt = [&](const short8* r, float* out) {
 #pragma unroll
-      for (int j = 0; j < E; ++j) { da += ka[j] * qv[g][j]; db += kb[j] * qv[g][j]; }
+    for (int c = 0; c < NR; ++c)
 #pragma unroll
-      for (int m = 1; m < 16; m <<= 1) {
-        da += __shfl_xor(da, m);
-        db += __shfl_xor(db, m);
+      for (int j = 0; j < 8; ++j) out[c * 8 + j] = bf2f((unsigned short)r[c][j]);
+  };
+
+  if constexpr (E >= 8) {
+    if (pos + 16 < c1) {
+      load_raw(row_off(pos), kr_a, vr_a);
+      load_raw(row_off(pos + 16), kr_b, vr_b);
+    }
+    for (; pos + 16 < c1; pos += 32) {
+      const bool have_next = pos + 48 < c1;
+      if (have_next) {
+        load_raw(row_off(pos + 32), kn_a, vn_a);
+        load_raw(row_off(pos + 48), kn_b, vn_b);
       }
-      const float sa = da * sc2, sb = db * sc2;
-      const float mx = fmaxf(sa, sb);
-      if (mx <= m2[g]) {  // fast path: no rescale (T13 class)
-        const float pa_ = fast_exp2(sa - m2[g]);
-        const float pb_ = fast_exp2(sb - m2[g]);
-        l[g] += pa_ + pb_;
+      float ka[E], va[E], kb[E], vb[E];
+      cvt(kr_a, ka); cvt(kr_b, kb); cvt(vr_a, va); cvt(vr_b, vb);
 #pragma unroll
-        for (int j = 0; j < E; ++j) acc[g][j] += pa_ * va[j] + pb_ * vb[j];
-      } else {
-        const float corr = fast_exp2(m2[g] - mx);
-        const float pa_ = fast_exp2(sa - mx);
-        const float pb_ = fast_exp2(sb - mx);
-        l[g] = l[g] * corr + pa_ + pb_;
+      for (int g = 0; g < MAXG; ++g) {
+        if (g >= G) break;
+        float da = 0.f, db = 0.f;
 #pragma unroll
-        for (int j = 0; j < E; ++j)
-          acc[g][j] = acc[g][j] * corr + pa_ * va[j] + pb_ * vb[j];
-        m2[g] = mx;
+        for (int j = 0; j < E; ++j) { da += ka[j] * qv[g][j]; db += kb[j] * qv[g][j]; }
+#pragma unroll
+        for (int m = 1; m < 16; m <<= 1) {
+          da += __shfl_xor(da, m);
+          db += __shfl_xor(db, m);
+        }
+        const float sa = da * sc2, sb = db * sc2;
+        const float mx = fmaxf(sa, sb);
+        if (mx <= m2[g]) {  // fast path: no rescale (T13 class)
+          const float pa_ = fast_exp2(sa - m2[g]);
+          const float pb_ = fast_exp2(sb - m2[g]);
+          l[g] += pa_ + pb_;
+#pragma unroll
+          for (int j = 0; j < E; ++j) acc[g][j] += pa_ * va[j] + pb_ * vb[j];
+        } else {
+          const float corr = fast_exp2(m2[g] - mx);
+          const float pa_ = fast_exp2(sa - mx);
+          const float pb_ = fast_exp2(sb - mx);
+          l[g] = l[g] * corr + pa_ + pb_;
+#pragma unroll
+          for (int j = 0; j < E; ++j)
+            acc[g][j] = acc[g][j] * corr + pa_ * va[j] + pb_ * vb[j];
+          m2[g] = mx;
+        }
+      }
+      if (have_next) {
+#pragma unroll
+        for (int c = 0; c < NR; ++c) {
+          kr_a[c] = kn_a[c]; vr_a[c] = vn_a[c];
+          kr_b[c] = kn_b[c]; vr_b[c] = vn_b[c];
+        }
       }
     }
   }
