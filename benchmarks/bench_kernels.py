@@ -6,8 +6,11 @@ changes can be A/B'd without the end-to-end bench's noise.
     python benchmarks/bench_kernels.py [attn_decode attn_prefill norms ...]
 """
 import math
+import os
 import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
