@@ -5,6 +5,7 @@
 #include <torch/extension.h>
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_runtime.h>
+#include <cstdlib>
 
 #include "common.h"
 
@@ -13,6 +14,7 @@
 #include "elementwise.hip"
 #include "kvcache.hip"
 #include "attn_decode.hip"
+#include "attn_decode_mfma.hip"
 #include "attn_prefill.hip"
 #include "mfma_selftest.hip"
 #include "quant4.hip"
@@ -187,6 +189,7 @@ static void attn_decode_launch(const torch::Tensor& q, const torch::Tensor& kp,
                                int Hkv, int G, int P, int maxp, int n_split,
                                int window, float scale, long q_off, long q_sb,
                                long q_sh, long out_sb, long out_sh) {
+  static const bool use_valu = std::getenv("BBAMD_ATTN_VALU") != nullptr;
   dim3 grid(B * Hkv, n_split);
   auto launch = [&](auto maxg) {
     attn_decode_kernel<D, decltype(maxg)::value><<<grid, 256, 0, cur_stream()>>>(
@@ -195,9 +198,20 @@ static void attn_decode_launch(const torch::Tensor& q, const torch::Tensor& kp,
         pacc.data_ptr<float>(), B, Hkv, G, P, maxp, n_split, window, scale,
         q_sb, q_sh, out_sb, out_sh);
   };
-  if (G <= 4) launch(std::integral_constant<int, 4>{});
+  auto launch_mfma = [&](auto maxg) {
+    attn_decode_mfma_kernel<D, decltype(maxg)::value>
+        <<<grid, 256, 0, cur_stream()>>>(
+            bf_ptr(q) + q_off, bf_ptr(kp), bf_ptr(vp), pt.data_ptr<int>(),
+            ctx.data_ptr<int>(), bf_ptr_mut(out), pml.data_ptr<float>(),
+            pacc.data_ptr<float>(), B, Hkv, G, P, maxp, n_split, window, scale,
+            q_sb, q_sh, out_sb, out_sh);
+  };
+  if (!use_valu && G <= 16) {
+    if (G <= 4) launch_mfma(std::integral_constant<int, 4>{});
+    else launch_mfma(std::integral_constant<int, 16>{});
+  } else if (G <= 4) launch(std::integral_constant<int, 4>{});
   else if (G <= 8) launch(std::integral_constant<int, 8>{});
-  else TORCH_CHECK(false, "GQA group size > 8 unsupported: ", G);
+  else TORCH_CHECK(false, "GQA group size > 16 unsupported: ", G);
   if (n_split > 1) {
     attn_decode_combine_kernel<D><<<B * Hkv, G * 16, 0, cur_stream()>>>(
         pml.data_ptr<float>(), pacc.data_ptr<float>(), bf_ptr_mut(out), Hkv, G,
