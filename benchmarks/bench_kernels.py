@@ -64,6 +64,22 @@ def bench_attn_prefill():
               f"{t*1e6:8.1f} us  {flops/t/1e12:7.1f} TF/s")
 
 
+def bench_kv_stream():
+    from bloombee_amd.ops import hip_ops
+    B, Hkv, ctx, D, P = 32, 8, 2048, 128, 16
+    maxp = ctx // P
+    npages = B * maxp + 1
+    kp = torch.randn(npages, Hkv, P, D, dtype=torch.bfloat16, device=DEV)
+    vp = torch.randn_like(kp)
+    pt = torch.arange(B * maxp, dtype=torch.int32, device=DEV).reshape(B, maxp)
+    ctx_l = torch.full((B,), ctx, dtype=torch.int32, device=DEV)
+    bytes_ = B * Hkv * ctx * 2 * D * 2
+    for ns in (2, 4, 8, 16):
+        for nt in (False, True):
+            t = timeit(lambda: hip_ops.kv_stream_probe(kp, vp, pt, ctx_l, ns, nt))
+            print(f"kv_stream ns{ns} nt{int(nt)}: {t*1e6:7.1f} us  {bytes_/t/1e12:5.2f} TB/s")
+
+
 def bench_norms():
     for N, H in [(32, 4096), (16384, 4096)]:
         x = torch.randn(N, H, dtype=torch.bfloat16, device=DEV)
@@ -97,6 +113,7 @@ def bench_gemm():
 ALL = {
     "attn_decode": bench_attn_decode,
     "attn_prefill": bench_attn_prefill,
+    "kv_stream": bench_kv_stream,
     "norms": bench_norms,
     "swiglu": bench_swiglu,
     "gemm": bench_gemm,

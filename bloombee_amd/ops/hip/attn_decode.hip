@@ -326,3 +326,45 @@ __global__ void attn_decode_combine_kernel(
   const int b = bh / Hkv, kvh = bh % Hkv;
   store_bf16_e<E>(out + (long)b * out_sb + (kvh * G + g) * out_sh + li * E, o);
 }
+
+// Pure-streaming probe: same grid/walk/loads as attn_decode_kernel but no
+// softmax/acc — measures the access pattern's bandwidth ceiling. NT variant
+// uses nontemporal loads (KV is read-once-per-step; keep L2 for weights).
+template <int D, bool NT>
+__global__ __launch_bounds__(256) void kv_stream_probe_kernel(
+    const unsigned short* __restrict__ k_pages,
+    const unsigned short* __restrict__ v_pages,
+    const int* __restrict__ page_table, const int* __restrict__ ctx_lens,
+    float* __restrict__ out, int B, int Hkv, int P, int maxp, int n_split) {
+  constexpr int E = D / 16;
+  const int bh = blockIdx.x;
+  const int split = blockIdx.y;
+  const int b = bh / Hkv, kvh = bh % Hkv;
+  const int ctx = ctx_lens[b];
+  const int pages_total = (ctx + P - 1) / P;
+  const int pages_per_split = (pages_total + n_split - 1) / n_split;
+  const int c0 = split * pages_per_split * P;
+  const int c1 = min(ctx, c0 + pages_per_split * P);
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave = threadIdx.x / WAVE;
+  const int group_id = wave * 4 + lane / 16;
+  const int li = lane & 15;
+  const long head_slab = (long)kvh * P * D;
+  float acc = 0.f;
+  for (int pos = c0 + group_id; pos < c1; pos += 16) {
+    const int page = page_table[b * maxp + pos / P];
+    const long off =
+        ((long)page * Hkv) * P * D + head_slab + (long)(pos % P) * D + li * E;
+    const short8* kp8 = reinterpret_cast<const short8*>(k_pages + off);
+    const short8* vp8 = reinterpret_cast<const short8*>(v_pages + off);
+#pragma unroll
+    for (int c = 0; c < E / 8; ++c) {
+      short8 kv = NT ? __builtin_nontemporal_load(kp8 + c) : kp8[c];
+      short8 vv = NT ? __builtin_nontemporal_load(vp8 + c) : vp8[c];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) acc += bf2f((unsigned short)kv[j]) + bf2f((unsigned short)vv[j]);
+    }
+  }
+  acc = wave_reduce_sum(acc);
+  if (lane == 0) out[(blockIdx.y * gridDim.x + blockIdx.x) * 4 + wave] = acc;
+}

@@ -406,6 +406,28 @@ torch::Tensor quant4_unpack(torch::Tensor packed, torch::Tensor scale,
   return out;
 }
 
+torch::Tensor kv_stream_probe(torch::Tensor k_pages, torch::Tensor v_pages,
+                              torch::Tensor page_table, torch::Tensor ctx_lens,
+                              long n_split, bool nt) {
+  const int B = page_table.size(0), Hkv = k_pages.size(1);
+  const int P = k_pages.size(2), D = k_pages.size(3), maxp = page_table.size(1);
+  auto out = torch::empty({(long)B * Hkv * n_split * 4},
+                          torch::TensorOptions().device(k_pages.device()).dtype(at::kFloat));
+  dim3 grid(B * Hkv, n_split);
+  TORCH_CHECK(D == 128);
+  if (nt)
+    kv_stream_probe_kernel<128, true><<<grid, 256, 0, cur_stream()>>>(
+        bf_ptr(k_pages), bf_ptr(v_pages), page_table.data_ptr<int>(),
+        ctx_lens.data_ptr<int>(), out.data_ptr<float>(), B, Hkv, P, maxp,
+        (int)n_split);
+  else
+    kv_stream_probe_kernel<128, false><<<grid, 256, 0, cur_stream()>>>(
+        bf_ptr(k_pages), bf_ptr(v_pages), page_table.data_ptr<int>(),
+        ctx_lens.data_ptr<int>(), out.data_ptr<float>(), B, Hkv, P, maxp,
+        (int)n_split);
+  return out;
+}
+
 torch::Tensor mfma_selftest(torch::Tensor A, torch::Tensor B) {
   CHECK_DEV(A); CHECK_BF16(A); CHECK_CONTIG(A); CHECK_CONTIG(B);
   TORCH_CHECK(A.size(0) == 16 && A.size(1) == 32 && B.size(0) == 32 && B.size(1) == 16);
@@ -433,4 +455,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("quant4_pack", &quant4_pack);
   m.def("quant4_unpack", &quant4_unpack);
   m.def("mfma_selftest", &mfma_selftest);
+  m.def("kv_stream_probe", &kv_stream_probe);
 }
