@@ -81,27 +81,23 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
   const long head_slab = (long)kvh * P * D;
   const int tile0 = (max(c0, lo) - c0) / DKVBLK;  // window skip, tile-aligned
 
+  constexpr int VPASS = DKVBLK * D / (64 * 8);  // V loads per lane per tile
   for (int tb = c0 + (tile0 + wave) * DKVBLK; tb < c1; tb += 4 * DKVBLK) {
-    // ---- stage V tile transposed: V_lds[d][pos] ----
-    {
-      // 32 rows x D elems; this wave's 64 lanes: 8 elems/lane per pass
-      const int rows_per_pass = 64 * 8 / D;  // D=128 -> 4 rows/pass
+    // ---- T14 split: issue this tile's V loads FIRST (write to LDS later,
+    // after the QK^T phase has covered their HBM latency) ----
+    const int vrow0 = lane / (D / 8);
+    const int vd8 = (lane % (D / 8)) * 8;
+    short8 v_raw[VPASS];
 #pragma unroll
-      for (int pass = 0; pass < DKVBLK / (64 * 8 / D); ++pass) {
-        const int row = pass * rows_per_pass + lane / (D / 8);
-        const int d8 = (lane % (D / 8)) * 8;
-        const int pos = tb + row;
-        short8 vv{};
-        if (pos < c1) {
-          const int page = page_table[b * maxp + pos / P];
-          vv = *reinterpret_cast<const short8*>(
-              v_pages + ((long)page * Hkv) * P * D + head_slab +
-              (long)(pos % P) * D + d8);
-        }
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          *(unsigned short*)(v_lds + (d8 + j) * VROW_B + row * 2) =
-              (unsigned short)vv[j];
+    for (int pass = 0; pass < VPASS; ++pass) {
+      const int row = pass * (64 * 8 / D) + vrow0;
+      const int pos = tb + row;
+      v_raw[pass] = short8{};
+      if (pos < c1) {
+        const int page = page_table[b * maxp + pos / P];
+        v_raw[pass] = *reinterpret_cast<const short8*>(
+            v_pages + ((long)page * Hkv) * P * D + head_slab +
+            (long)(pos % P) * D + vd8);
       }
     }
 
@@ -122,6 +118,16 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
         s[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[kk], bfrag, s[n],
                                                        0, 0, 0);
       }
+    }
+
+    // ---- V raw regs -> transposed LDS image (loads have landed by now) ----
+#pragma unroll
+    for (int pass = 0; pass < VPASS; ++pass) {
+      const int row = pass * (64 * 8 / D) + vrow0;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        *(unsigned short*)(v_lds + (vd8 + j) * VROW_B + row * 2) =
+            (unsigned short)v_raw[pass][j];
     }
 
     // ---- mask + online softmax (all 16 rows share the same position) ----
