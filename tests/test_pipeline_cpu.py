@@ -10,12 +10,17 @@ import torch.multiprocessing as mp
 
 from bloombee_amd.engine import LocalEngine
 
-PORT = 29511
+def _free_port():
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
 
 
-def _worker(rank, world, q):
+def _worker(rank, world, q, port):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
-    os.environ["MASTER_PORT"] = str(PORT)
+    os.environ["MASTER_PORT"] = str(port)
     os.environ["RANK"] = str(rank)
     os.environ["WORLD_SIZE"] = str(world)
     dist.init_process_group("gloo", rank=rank, world_size=world)
@@ -43,7 +48,8 @@ def _worker(rank, world, q):
 def test_pipeline_matches_local_engine():
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    procs = [ctx.Process(target=_worker, args=(r, 2, q)) for r in range(2)]
+    port = _free_port()
+    procs = [ctx.Process(target=_worker, args=(r, 2, q, port)) for r in range(2)]
     for p in procs:
         p.start()
     got = q.get(timeout=150)
