@@ -38,7 +38,7 @@ def bench_attn_decode():
         maxp = (ctx + P - 1) // P
         npages = B * maxp + 1
         kp = torch.randn(npages, Hkv, P, D, dtype=torch.bfloat16, device=DEV)
-        vp = torch.randn_like(kp)
+        vp = torch.randn(kp.shape[0], Hkv, D, P, dtype=torch.bfloat16, device=DEV)
         pt = torch.arange(B * maxp, dtype=torch.int32, device=DEV).reshape(B, maxp)
         q = torch.randn(B, Hq, 1, D, dtype=torch.bfloat16, device=DEV) * 0.1
         ctx_l = torch.full((B,), ctx, dtype=torch.int32, device=DEV)
@@ -54,7 +54,7 @@ def bench_attn_prefill():
         maxp = (T + P - 1) // P
         npages = B * maxp + 1
         kp = torch.randn(npages, Hkv, P, D, dtype=torch.bfloat16, device=DEV)
-        vp = torch.randn_like(kp)
+        vp = torch.randn(kp.shape[0], Hkv, D, P, dtype=torch.bfloat16, device=DEV)
         pt = torch.arange(B * maxp, dtype=torch.int32, device=DEV).reshape(B, maxp)
         q = torch.randn(B, Hq, T, D, dtype=torch.bfloat16, device=DEV) * 0.1
         qs = torch.zeros(B, dtype=torch.int32, device=DEV)
@@ -70,7 +70,7 @@ def bench_kv_stream():
     maxp = ctx // P
     npages = B * maxp + 1
     kp = torch.randn(npages, Hkv, P, D, dtype=torch.bfloat16, device=DEV)
-    vp = torch.randn_like(kp)
+    vp = torch.randn(kp.shape[0], Hkv, D, P, dtype=torch.bfloat16, device=DEV)
     pt = torch.arange(B * maxp, dtype=torch.int32, device=DEV).reshape(B, maxp)
     ctx_l = torch.full((B,), ctx, dtype=torch.int32, device=DEV)
     bytes_ = B * Hkv * ctx * 2 * D * 2
@@ -106,8 +106,13 @@ def bench_gemm():
         x = torch.randn(M, K, dtype=torch.bfloat16, device=DEV)
         w = torch.randn(N, K, dtype=torch.bfloat16, device=DEV)
         t = timeit(lambda: torch.nn.functional.linear(x, w))
-        print(f"gemm {tag} {M}x{N}x{K}: {t*1e6:7.1f} us  "
+        print(f"gemm[blaslt] {tag} {M}x{N}x{K}: {t*1e6:7.1f} us  "
               f"{N*K*2/t/1e12:5.2f} TB/s(wt)  {2*M*N*K/t/1e12:6.1f} TF/s")
+        if N % 64 == 0:
+            for ks in (0, 1, 2, 4, 8, 16):
+                ts = timeit(lambda: ops.hip_ops.gemm_skinny(x, w, None, None, ks))
+                print(f"gemm[skinny ks={ks}] {tag}: {ts*1e6:7.1f} us  "
+                      f"{N*K*2/ts/1e12:5.2f} TB/s(wt)")
 
 
 ALL = {

@@ -92,10 +92,19 @@ class PagedKVCache:
             return self.max_tokens - self._reserved_tokens
 
     def k_pages(self, layer: int) -> torch.Tensor:
+        """K pages, token-major: (n_pages, Hkv, P, D)."""
         return self.pool[layer, 0]
 
     def v_pages(self, layer: int) -> torch.Tensor:
-        return self.pool[layer, 1]
+        """V pages, d-major: (n_pages, Hkv, D, P).
+
+        V is stored TRANSPOSED within each (page, head) slab so the decode
+        attention kernel's P.V MFMA B-fragments (8 consecutive positions at a
+        fixed d) are direct contiguous 16 B loads from HBM — no LDS transpose
+        staging. Same bytes as the K slab (P*D*2), reinterpreted.
+        """
+        v = self.pool[layer, 1]
+        return v.view(self.n_pages, self.num_kv_heads, self.head_dim, self.page_size)
 
     # -- admission --------------------------------------------------------
     def allocate(

@@ -110,7 +110,7 @@ class LlamaBlock(torch.nn.Module):
         cfg = self.config
 
         x = ops.rms_norm(hidden, self.input_norm_w, cfg.rms_norm_eps)
-        qkv = F.linear(x, self.qkv_w)                      # (B, T, (Hq+2Hkv)D)
+        qkv = ops.linear(x, self.qkv_w)                    # (B, T, (Hq+2Hkv)D)
         cos, sin = self.rope.get(hidden.device)
         kp = kv.k_pages(self.layer_index)
         vp = kv.v_pages(self.layer_index)
@@ -120,14 +120,13 @@ class LlamaBlock(torch.nn.Module):
         ops.rope_kv_write_(qkv, Hq, Hkv, cos, sin, position_ids, kp, vp, pt,
                            start_pos)
         attn = ops.attn_paged_qkv(qkv, Hq, Hkv, kp, vp, pt, start_pos, self.scale)
-        a = F.linear(attn, self.o_w)
+        a = ops.linear(attn, self.o_w)
 
         # h2 = hidden + a fused into the post-attention norm
         h2, y = ops.rms_norm_residual(a, hidden, self.post_norm_w, cfg.rms_norm_eps)
-        m = F.linear(ops.swiglu(F.linear(y, self.gate_up_w)), self.down_w)
-        if hidden.is_cuda and ops.HAVE_HIP_OPS:
-            return ops.hip_ops.add_bf16(m.contiguous(), h2)
-        return m + h2
+        # residual add fused into the down-projection epilogue
+        return ops.linear(ops.swiglu(ops.linear(y, self.gate_up_w)), self.down_w,
+                          residual=h2)
 
     # ------------------------------------------------------------------
     # training path (differentiable; full sequence, no KV cache)

@@ -15,6 +15,7 @@ from bloombee_amd.ops.interface import (  # noqa: F401
     kv_gather,
     kv_write,
     layer_norm,
+    linear,
     mfma_selftest,
     quant4_pack,
     quant4_unpack,

@@ -1,20 +1,25 @@
 // MFMA-based single-token paged attention for gfx950.
 //
-// The VALU formulation (attn_decode.hip) spends ~210 VALU instructions per
-// position pair (dots, shuffles, exp2, accumulate) and measures VALU-bound at
-// ~3 TB/s of KV streaming. Here the score dots and the P.V accumulation run
-// on the matrix cores: one 16x16x32 MFMA computes 16 positions x 16 rows at
-// once. The "q tile" is the G query heads of one (b, kv_head) GQA group —
-// rows >= G are garbage lanes that only pollute garbage output rows (C row r
-// depends only on A row r), so a 4-head group costs 4/16 MFMA efficiency and
-// the kernel is HBM-bound again, which is the point.
+// Score dots and the P.V accumulation run on the matrix cores: one 16x16x32
+// MFMA computes 16 positions x 16 rows at once. The "q tile" is the G query
+// heads of one (b, kv_head) GQA group — rows >= G are garbage lanes that only
+// pollute garbage output rows (C row r depends only on A row r), so a 4-head
+// group costs 4/16 MFMA efficiency and the kernel is HBM-bound, which is the
+// point.
 //
-// Work decomposition: grid = (B*Hkv, n_split), 4 waves per workgroup; wave w
-// owns KV tiles (32 positions) w, w+4, w+8, ... of the chunk, with a final
-// 4-way LDS merge of (m, l, acc). K fragments are read DIRECTLY from the
-// paged pool (B-fragment address pattern covers each (page, head) slab in
-// aligned 64 B pieces exactly once — no LDS round trip); V is transpose-
-// staged through per-wave LDS for the P.V operand.
+// BOTH operand streams read straight from the paged pool — no LDS staging:
+//  * K pages are token-major (np, Hkv, P, D): a QK^T B-fragment (8 consecutive
+//    d at fixed position) is a contiguous 16 B load.
+//  * V pages are d-major (np, Hkv, D, P): a P.V B-fragment (8 consecutive
+//    positions at fixed d) is a contiguous 16 B load. The transpose happened
+//    once at kv-write time.
+// LDS is only used for the P C->A fragment relayout (tiny) and the final
+// 4-wave merge, so occupancy is VGPR-limited, not LDS-limited.
+//
+// Work decomposition: grid = (B*Hkv, n_split) flash-decode splits, 4 waves
+// per workgroup; wave w owns KV tiles (32 positions) w, w+4, w+8, ... of its
+// split, with a final 4-way LDS merge of (m, l, acc). V fragment loads are
+// issued BEFORE the QK^T/softmax phase so their HBM latency is covered.
 
 #include "common.h"
 
@@ -24,7 +29,7 @@ template <int D, int MAXG>
 __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
     const unsigned short* __restrict__ q,        // strided, see q_sb/q_sh
     const unsigned short* __restrict__ k_pages,  // (np, Hkv, P, D)
-    const unsigned short* __restrict__ v_pages,
+    const unsigned short* __restrict__ v_pages,  // (np, Hkv, D, P) d-major
     const int* __restrict__ page_table,          // (B, maxp)
     const int* __restrict__ ctx_lens,            // (B,)
     unsigned short* __restrict__ out,
@@ -34,7 +39,6 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
     float scale, long q_sb, long q_sh, long out_sb, long out_sh) {
   constexpr int NKK = D / 32;   // QK^T k-slices
   constexpr int NDT = D / 16;   // PV d-tiles
-  constexpr int VROW_B = DKVBLK * 2 + 16;  // transposed V row stride (80 B)
   constexpr int PROW_B = DKVBLK * 2 + 16;
 
   const int bh = blockIdx.x;
@@ -56,12 +60,11 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
   const int hi = lane >> 4;
   const float sc2 = scale * LOG2E;
 
-  // per-wave LDS: transposed V tile + P tile + merge scratch
+  // per-wave LDS: P tile (C->A relayout) + merge scratch
   __shared__ __attribute__((aligned(16))) unsigned char lds[
-      4 * (D * VROW_B + 16 * PROW_B) + 3 * (MAXG * (D + 2) * 4)];
-  unsigned char* v_lds = lds + wave * (D * VROW_B + 16 * PROW_B);
-  unsigned char* p_lds = v_lds + D * VROW_B;
-  float* merge_lds = (float*)(lds + 4 * (D * VROW_B + 16 * PROW_B));
+      4 * 16 * PROW_B + 3 * (MAXG * (D + 2) * 4)];
+  unsigned char* p_lds = lds + wave * 16 * PROW_B;
+  float* merge_lds = (float*)(lds + 4 * 16 * PROW_B);
 
   // Q fragments: A[row][k] with row = head g (rows >= G harmless garbage)
   const int qh = (li < G) ? li : G - 1;
@@ -78,27 +81,26 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
 #pragma unroll
   for (int n = 0; n < NDT; ++n) acc_o[n] = (f32x4){0.f, 0.f, 0.f, 0.f};
 
-  const long head_slab = (long)kvh * P * D;
+  const long head_slab_k = (long)kvh * P * D;  // same bytes, both layouts
   const int tile0 = (max(c0, lo) - c0) / DKVBLK;  // window skip, tile-aligned
 
-  constexpr int VPASS = DKVBLK * D / (64 * 8);  // V loads per lane per tile
   for (int tb = c0 + (tile0 + wave) * DKVBLK; tb < c1; tb += 4 * DKVBLK) {
-    // ---- T14 split: issue this tile's V loads FIRST (write to LDS later,
-    // after the QK^T phase has covered their HBM latency) ----
-    const int vrow0 = lane / (D / 8);
-    const int vd8 = (lane % (D / 8)) * 8;
-    short8 v_raw[VPASS];
+    // ---- V fragments for this tile: direct B-layout loads, issued FIRST so
+    // the QK^T phase covers their latency. B[k=pos][j=d]: lane (li -> d tile
+    // col, hi -> position octet); 8 consecutive positions at fixed d are
+    // contiguous in the d-major pool. tb is 32-aligned and P | 32, so each
+    // octet sits in one page.
+    const int vpos = tb + hi * 8;
+    int vpage = -1;
+    if (vpos < c1) vpage = page_table[b * maxp + vpos / P];
+    const long vrow = ((long)vpage * Hkv + kvh) * D;
+    bf16x8 vfrag[NDT];
 #pragma unroll
-    for (int pass = 0; pass < VPASS; ++pass) {
-      const int row = pass * (64 * 8 / D) + vrow0;
-      const int pos = tb + row;
-      v_raw[pass] = short8{};
-      if (pos < c1) {
-        const int page = page_table[b * maxp + pos / P];
-        v_raw[pass] = *reinterpret_cast<const short8*>(
-            v_pages + ((long)page * Hkv) * P * D + head_slab +
-            (long)(pos % P) * D + vd8);
-      }
+    for (int n = 0; n < NDT; ++n) {
+      vfrag[n] = as_bf16x8(short8{});
+      if (vpage >= 0)
+        vfrag[n] = as_bf16x8(*reinterpret_cast<const short8*>(
+            v_pages + (vrow + li + 16 * n) * P + (vpos % P)));
     }
 
     // ---- S = Q K^T : K fragments straight from the paged pool ----
@@ -109,7 +111,7 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
       const int cpos = min(pos, c1 - 1);
       const int page = page_table[b * maxp + cpos / P];
       const unsigned short* krow = k_pages + ((long)page * Hkv) * P * D +
-                                   head_slab + (long)(cpos % P) * D;
+                                   head_slab_k + (long)(cpos % P) * D;
       s[n] = (f32x4){0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int kk = 0; kk < NKK; ++kk) {
@@ -118,16 +120,6 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
         s[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[kk], bfrag, s[n],
                                                        0, 0, 0);
       }
-    }
-
-    // ---- V raw regs -> transposed LDS image (loads have landed by now) ----
-#pragma unroll
-    for (int pass = 0; pass < VPASS; ++pass) {
-      const int row = pass * (64 * 8 / D) + vrow0;
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        *(unsigned short*)(v_lds + (vd8 + j) * VROW_B + row * 2) =
-            (unsigned short)v_raw[pass][j];
     }
 
     // ---- mask + online softmax (all 16 rows share the same position) ----
@@ -170,14 +162,11 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
     bf16x8 pfrag = as_bf16x8(
         *reinterpret_cast<const short8*>(p_lds + li * PROW_B + hi * 16));
 
-    // ---- acc += P V ----
+    // ---- acc += P V (B-fragments were loaded at tile start) ----
 #pragma unroll
-    for (int n = 0; n < NDT; ++n) {
-      bf16x8 vfrag = as_bf16x8(*reinterpret_cast<const short8*>(
-          v_lds + (li + 16 * n) * VROW_B + hi * 16));
-      acc_o[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pfrag, vfrag,
+    for (int n = 0; n < NDT; ++n)
+      acc_o[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pfrag, vfrag[n],
                                                          acc_o[n], 0, 0, 0);
-    }
   }
 
   // ---- merge the 4 waves ----

@@ -87,24 +87,35 @@ __global__ __launch_bounds__(256) void attn_prefill_kernel(
 
   for (int kbase = kstart; kbase < kmax; kbase += KVBLK) {
     __syncthreads();  // everyone done reading previous tile
-    // ---- stage K tile (row-major, padded) and V tile (transposed) ----
+    // ---- stage K tile (row-major, padded) ----
     for (int idx = tid; idx < KVBLK * (D / 8); idx += 256) {
       const int t = idx / (D / 8);
       const int d8 = (idx % (D / 8)) * 8;
       const int kpos = kbase + t;
-      short8 kv{}, vv{};
+      short8 kv{};
       if (kpos < ctx) {
         const int page = page_table[b * maxp + kpos / P];
         const long off =
             (((long)page * (long)(Hq / G) + kvh) * P + (kpos % P)) * D + d8;
         kv = *reinterpret_cast<const short8*>(k_pages + off);
-        vv = *reinterpret_cast<const short8*>(v_pages + off);
       }
       *reinterpret_cast<short8*>((unsigned char*)k_lds + t * KROW_B + d8 * 2) = kv;
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        *(unsigned short*)((unsigned char*)v_lds + (d8 + j) * VROW_B + t * 2) =
-            (unsigned short)vv[j];
+    }
+    // ---- stage V tile: pool is d-major (np, Hkv, D, P), so an 8-position
+    // piece at fixed d is one contiguous 16 B load AND one contiguous 16 B
+    // LDS store (kbase is KVBLK-aligned => each octet sits in one page) ----
+    for (int idx = tid; idx < D * (KVBLK / 8); idx += 256) {
+      const int d = idx / (KVBLK / 8);
+      const int o8 = (idx % (KVBLK / 8)) * 8;
+      const int kpos = kbase + o8;
+      short8 vv{};
+      if (kpos < ctx) {
+        const int page = page_table[b * maxp + kpos / P];
+        vv = *reinterpret_cast<const short8*>(
+            v_pages + (((long)page * (long)(Hq / G) + kvh) * D + d) * P +
+            (kpos % P));
+      }
+      *reinterpret_cast<short8*>((unsigned char*)v_lds + d * VROW_B + o8 * 2) = vv;
     }
     __syncthreads();
 

@@ -237,14 +237,13 @@ __global__ void rope_kv_write_kernel(
   unsigned short* base = qkv + (((long)b * T + t) * X + h) * D;
 
   const int abspos = start_pos[b] + t;
-  if (h >= Hq + Hkv) {  // v head: copy to pages
-    const int hv = h - Hq - Hkv;
+  if (h >= Hq + Hkv) {  // v head: copy to pages (d-major V layout: transpose
+    const int hv = h - Hq - Hkv;  // on write, scattered 2 B stores, once/token)
     const int page = page_table[b * maxp + abspos / P];
-    unsigned short* dst =
-        v_pages + (((long)page * Hkv + hv) * P + (abspos % P)) * D;
-    for (int i = threadIdx.x; i < D / 8; i += blockDim.x)
-      *reinterpret_cast<short8*>(dst + i * 8) =
-          *reinterpret_cast<const short8*>(base + i * 8);
+    const int slot = abspos % P;
+    unsigned short* dst = v_pages + ((long)page * Hkv + hv) * D * P + slot;
+    for (int i = threadIdx.x; i < D; i += blockDim.x)
+      dst[(long)i * P] = base[i];
     return;
   }
   const int p = pos ? pos[b * T + t] : abspos;

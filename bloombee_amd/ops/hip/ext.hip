@@ -16,6 +16,7 @@
 #include "attn_decode.hip"
 #include "attn_decode_mfma.hip"
 #include "attn_prefill.hip"
+#include "gemm_skinny.hip"
 #include "mfma_selftest.hip"
 #include "quant4.hip"
 
@@ -155,6 +156,8 @@ void kv_write(torch::Tensor k_new, torch::Tensor v_new, torch::Tensor k_pages,
   const int B = k_new.size(0), Hkv = k_new.size(1), T = k_new.size(2), D = k_new.size(3);
   const int P = k_pages.size(2), maxp = page_table.size(1);
   TORCH_CHECK(k_pages.size(1) == Hkv && k_pages.size(3) == D);
+  TORCH_CHECK(v_pages.size(2) == D && v_pages.size(3) == P,
+              "v_pages must be d-major (np, Hkv, D, P)");
   const int threads = std::min(256, Hkv * D / 8);
   kv_write_kernel<<<B * T, threads, 0, cur_stream()>>>(
       bf_ptr(k_new), bf_ptr(v_new), bf_ptr_mut(k_pages), bf_ptr_mut(v_pages),
@@ -189,15 +192,7 @@ static void attn_decode_launch(const torch::Tensor& q, const torch::Tensor& kp,
                                int Hkv, int G, int P, int maxp, int n_split,
                                int window, float scale, long q_off, long q_sb,
                                long q_sh, long out_sb, long out_sh) {
-  static const bool use_valu = std::getenv("BBAMD_ATTN_VALU") != nullptr;
   dim3 grid(B * Hkv, n_split);
-  auto launch = [&](auto maxg) {
-    attn_decode_kernel<D, decltype(maxg)::value><<<grid, 256, 0, cur_stream()>>>(
-        bf_ptr(q) + q_off, bf_ptr(kp), bf_ptr(vp), pt.data_ptr<int>(),
-        ctx.data_ptr<int>(), bf_ptr_mut(out), pml.data_ptr<float>(),
-        pacc.data_ptr<float>(), B, Hkv, G, P, maxp, n_split, window, scale,
-        q_sb, q_sh, out_sb, out_sh);
-  };
   auto launch_mfma = [&](auto maxg) {
     attn_decode_mfma_kernel<D, decltype(maxg)::value>
         <<<grid, 256, 0, cur_stream()>>>(
@@ -206,11 +201,8 @@ static void attn_decode_launch(const torch::Tensor& q, const torch::Tensor& kp,
             pacc.data_ptr<float>(), B, Hkv, G, P, maxp, n_split, window, scale,
             q_sb, q_sh, out_sb, out_sh);
   };
-  if (!use_valu && G <= 16) {
-    if (G <= 4) launch_mfma(std::integral_constant<int, 4>{});
-    else launch_mfma(std::integral_constant<int, 16>{});
-  } else if (G <= 4) launch(std::integral_constant<int, 4>{});
-  else if (G <= 8) launch(std::integral_constant<int, 8>{});
+  if (G <= 4) launch_mfma(std::integral_constant<int, 4>{});
+  else if (G <= 16) launch_mfma(std::integral_constant<int, 16>{});
   else TORCH_CHECK(false, "GQA group size > 16 unsupported: ", G);
   if (n_split > 1) {
     attn_decode_combine_kernel<D><<<B * Hkv, G * 16, 0, cur_stream()>>>(
@@ -370,6 +362,65 @@ void rope_kv_write_(torch::Tensor qkv, long Hq_, long Hkv_, torch::Tensor cos_t,
 }
 
 // ---------------------------------------------------------------------------
+// skinny-M GEMM (decode projections)
+// ---------------------------------------------------------------------------
+
+// C(M,N) = A(M,K) @ W(N,K)^T [+ residual (M,N)] [+ bias (N,)], M <= 32.
+torch::Tensor gemm_skinny(torch::Tensor A, torch::Tensor W,
+                          c10::optional<torch::Tensor> residual,
+                          c10::optional<torch::Tensor> bias, long ksplit_req) {
+  CHECK_DEV(A); CHECK_BF16(A); CHECK_CONTIG(A);
+  CHECK_DEV(W); CHECK_BF16(W); CHECK_CONTIG(W);
+  const int K = A.size(-1);
+  const int M = A.numel() / K;
+  const int N = W.size(0);
+  TORCH_CHECK(M <= 32, "gemm_skinny is for M <= 32, got ", M);
+  TORCH_CHECK(W.size(1) == K, "A/W K mismatch");
+  TORCH_CHECK(K % 32 == 0 && N % 64 == 0, "K%32, N%64 required");
+  auto sizes = A.sizes().vec();
+  sizes.back() = N;
+  auto C = torch::empty(sizes, A.options());
+  const unsigned short* rp = nullptr;
+  if (residual.has_value()) {
+    TORCH_CHECK(residual->is_contiguous() && residual->numel() == (long)M * N);
+    rp = bf_ptr(*residual);
+  }
+  const unsigned short* bp = nullptr;
+  if (bias.has_value()) {
+    TORCH_CHECK(bias->is_contiguous() && bias->numel() == N);
+    bp = bf_ptr(*bias);
+  }
+  int ksplit = (int)ksplit_req;
+  if (ksplit <= 0) {
+    // enough workgroups to fill 256 CUs across 8 XCDs, K slices >= 512
+    ksplit = (int)std::min<long>(K / 512, std::max<long>(1, 1024 / (N / 64)));
+    ksplit = std::max(1, ksplit);
+  }
+  torch::Tensor part;
+  float* pp = nullptr;
+  if (ksplit > 1) {
+    part = torch::empty({(long)ksplit, (long)M, (long)N},
+                        torch::TensorOptions().device(A.device()).dtype(at::kFloat));
+    pp = part.data_ptr<float>();
+  }
+  dim3 grid(N / 64, ksplit);
+  auto launch = [&](auto mt) {
+    gemm_skinny_kernel<decltype(mt)::value><<<grid, 256, 0, cur_stream()>>>(
+        bf_ptr(A), bf_ptr(W), ksplit == 1 ? rp : nullptr,
+        ksplit == 1 ? bp : nullptr, bf_ptr_mut(C), pp, M, N, K, ksplit);
+  };
+  if (M <= 16) launch(std::integral_constant<int, 1>{});
+  else launch(std::integral_constant<int, 2>{});
+  if (ksplit > 1) {
+    const long total = (long)M * N;
+    const int cgrid = (int)std::min<long>((total + 255) / 256, 2048);
+    gemm_skinny_combine_kernel<<<cgrid, 256, 0, cur_stream()>>>(
+        pp, rp, bp, bf_ptr_mut(C), M, N, ksplit);
+  }
+  return C;
+}
+
+// ---------------------------------------------------------------------------
 // quant4 / selftest
 // ---------------------------------------------------------------------------
 
@@ -452,6 +503,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_prefill_qkv", &attn_prefill_qkv);
   m.def("rope_kv_write_", &rope_kv_write_);
   m.def("attn_prefill", &attn_prefill);
+  m.def("gemm_skinny", &gemm_skinny);
   m.def("quant4_pack", &quant4_pack);
   m.def("quant4_unpack", &quant4_unpack);
   m.def("mfma_selftest", &mfma_selftest);

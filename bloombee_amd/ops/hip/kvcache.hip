@@ -7,8 +7,11 @@
 // kv_gather_kernel  — gather a prefix back to dense (debug/failover path,
 //                     ref paged_kv.py:265-316 `gather_prefix`).
 //
-// Pages: (n_pages, Hkv, P, D) bf16 — a (page, head) slab is P*D*2 bytes
-// contiguous (4 KB at P=16, D=128): one coalesced burst per wave.
+// Layouts: K pages (n_pages, Hkv, P, D) token-major; V pages
+// (n_pages, Hkv, D, P) d-major. V is transposed on WRITE (once per token,
+// scattered 2 B stores are cheap) so the decode attention P.V MFMA
+// B-fragments — 8 consecutive positions at a fixed d — are direct contiguous
+// 16 B READS from HBM every step, with no LDS transpose staging.
 
 #include "common.h"
 
@@ -27,11 +30,13 @@ __global__ void kv_write_kernel(
     const int h = i / (D / 8);
     const int d = (i % (D / 8)) * 8;
     const long src = (((long)b * Hkv + h) * T + t) * D + d;
-    const long dst = (((long)page * Hkv + h) * P + slot) * D + d;
-    *reinterpret_cast<short8*>(k_pages + dst) =
+    const long kdst = (((long)page * Hkv + h) * P + slot) * D + d;
+    *reinterpret_cast<short8*>(k_pages + kdst) =
         *reinterpret_cast<const short8*>(k_new + src);
-    *reinterpret_cast<short8*>(v_pages + dst) =
-        *reinterpret_cast<const short8*>(v_new + src);
+    const short8 vv = *reinterpret_cast<const short8*>(v_new + src);
+    const long vbase = (((long)page * Hkv + h) * D + d) * P + slot;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) v_pages[vbase + (long)j * P] = (unsigned short)vv[j];
   }
 }
 
@@ -48,11 +53,14 @@ __global__ void kv_gather_kernel(
     const int h = (int)((i / (D / 8)) % Hkv);
     const int pos = (int)(i / ((D / 8) * (long)Hkv));
     const int page = page_table[batch_index * maxp + pos / P];
-    const long src = (((long)page * Hkv + h) * P + (pos % P)) * D + d;
+    const long ksrc = (((long)page * Hkv + h) * P + (pos % P)) * D + d;
     const long dst = (((long)h * ctx) + pos) * D + d;
     *reinterpret_cast<short8*>(k_out + dst) =
-        *reinterpret_cast<const short8*>(k_pages + src);
-    *reinterpret_cast<short8*>(v_out + dst) =
-        *reinterpret_cast<const short8*>(v_pages + src);
+        *reinterpret_cast<const short8*>(k_pages + ksrc);
+    const long vbase = (((long)page * Hkv + h) * D + d) * P + (pos % P);
+    short8 vv;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) vv[j] = (short)v_pages[vbase + (long)j * P];
+    *reinterpret_cast<short8*>(v_out + dst) = vv;
   }
 }

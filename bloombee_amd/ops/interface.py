@@ -217,6 +217,23 @@ def attn_paged(q, k_pages, v_pages, page_table, q_start, scale=None, window: int
     return attn_prefill(q, k_pages, v_pages, page_table, q_start, scale, window)
 
 
+def linear(x: torch.Tensor, w: torch.Tensor, residual: Optional[torch.Tensor] = None,
+           bias: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """y = x @ w^T (+bias) (+residual). Decode-shaped GEMMs (<=32 rows) on GPU
+    go through the skinny-M MFMA kernel (gemm_skinny.hip) — hipBLASLt leaves
+    ~2x weight-stream bandwidth on the table at M<=32; everything else uses
+    hipBLASLt via F.linear."""
+    M = x.numel() // x.shape[-1]
+    if (_on_gpu(x) and M <= 32 and x.dtype == torch.bfloat16
+            and x.shape[-1] % 32 == 0 and w.shape[0] % 64 == 0):
+        _require_ext()
+        return hip_ops.gemm_skinny(x.contiguous(), w, residual, bias, 0)
+    y = torch.nn.functional.linear(x, w, bias)
+    if residual is not None:
+        y = y + residual.view_as(y)
+    return y
+
+
 def quant4_pack(x: torch.Tensor, group_size: int = 64):
     if _on_gpu(x):
         _require_ext()

@@ -158,7 +158,9 @@ def kv_write(
 ) -> None:
     """Scatter new K/V into the paged cache, in place.
 
-    k_new/v_new: (B, Hkv, T, D); pages: (n_pages, Hkv, P, D);
+    k_new/v_new: (B, Hkv, T, D); k_pages: (n_pages, Hkv, P, D) token-major;
+    v_pages: (n_pages, Hkv, D, P) d-major (transposed for direct MFMA
+    B-fragment loads in the decode kernel — see PagedKVCache.v_pages);
     page_table: (B, max_pages) int32; start_pos: (B,) int32 — absolute position
     of k_new[:, :, 0]. (ref paged_kv.py:137-204 `write`, MI-native layout.)
     """
@@ -171,7 +173,7 @@ def kv_write(
             page = int(page_table[b, pos // P])
             slot = pos % P
             k_pages[page, :, slot, :] = k_new[b, :, t, :]
-            v_pages[page, :, slot, :] = v_new[b, :, t, :]
+            v_pages[page, :, :, slot] = v_new[b, :, t, :]
 
 
 def kv_gather(
@@ -184,7 +186,8 @@ def kv_gather(
     n = (ctx_len + P - 1) // P
     pages = page_table[batch_index, :n].long()
     k = k_pages[pages].permute(1, 0, 2, 3).reshape(k_pages.shape[1], n * P, -1)[:, :ctx_len]
-    v = v_pages[pages].permute(1, 0, 2, 3).reshape(v_pages.shape[1], n * P, -1)[:, :ctx_len]
+    # v_pages is d-major (n_pages, Hkv, D, P) -> (Hkv, n, P, D) -> (Hkv, ctx, D)
+    v = v_pages[pages].permute(1, 0, 3, 2).reshape(v_pages.shape[1], n * P, -1)[:, :ctx_len]
     return k, v
 
 

@@ -96,7 +96,7 @@ def test_kv_write_gather_parity():
     B, Hkv, T, D, P = 3, 2, 37, 128, 16
     npages = 32
     kp = torch.zeros(npages, Hkv, P, D).to(torch.bfloat16).to(DEV)
-    vp = torch.zeros(npages, Hkv, P, D).to(torch.bfloat16).to(DEV)
+    vp = torch.zeros(npages, Hkv, D, P).to(torch.bfloat16).to(DEV)
     maxp = 8
     pt = (torch.randperm(npages)[: B * maxp]).int().reshape(B, maxp).to(DEV)
     k = torch.randn(B, Hkv, T, D).to(torch.bfloat16)
@@ -117,7 +117,7 @@ def _paged_setup(B, Hq, Hkv, T, D, P=16, seed=7, dtype=torch.bfloat16):
     maxp = (T + P - 1) // P + 2
     npages = B * maxp + 4
     kp = torch.zeros(npages, Hkv, P, D, dtype=dtype)
-    vp = torch.zeros(npages, Hkv, P, D, dtype=dtype)
+    vp = torch.zeros(npages, Hkv, D, P, dtype=dtype)
     pt = torch.arange(B * maxp, dtype=torch.int32).reshape(B, maxp)
     q = (torch.randn(B, Hq, T, D) / math.sqrt(D)).to(dtype)
     k = torch.randn(B, Hkv, T, D).to(dtype)
@@ -257,7 +257,7 @@ def test_fused_qkv_path_parity():
     maxp = 8
     for T, start0 in [(1, 30), (40, 0)]:
         kp_c = torch.zeros(B * maxp + 2, Hkv, P, D, dtype=torch.bfloat16)
-        vp_c = torch.zeros_like(kp_c)
+        vp_c = torch.zeros(B * maxp + 2, Hkv, D, P, dtype=torch.bfloat16)
         pt = torch.arange(B * maxp, dtype=torch.int32).reshape(B, maxp)
         qkv = (torch.randn(B, T, X * D) * 0.3).to(torch.bfloat16)
         start = torch.full((B,), start0, dtype=torch.int32)
@@ -283,3 +283,28 @@ def test_fused_qkv_path_parity():
             f"T={T}: k pages mismatch"
         assert torch.allclose(out_g.cpu().float(), out_c.float(), atol=3e-2), \
             f"T={T}: attn out mismatch {(out_g.cpu().float()-out_c.float()).abs().max()}"
+
+
+@pytest.mark.parametrize("M,N,K,ksplit", [
+    (32, 4096, 4096, 0), (32, 6144, 4096, 1), (17, 4096, 4096, 8),
+    (8, 256, 14336, 0), (32, 28672, 4096, 2), (1, 4096, 4096, 0),
+])
+def test_gemm_skinny_parity(M, N, K, ksplit):
+    torch.manual_seed(11)
+    x = (torch.randn(M, K) / math.sqrt(K)).to(torch.bfloat16).to(DEV)
+    w = torch.randn(N, K).to(torch.bfloat16).to(DEV)
+    r = torch.randn(M, N).to(torch.bfloat16).to(DEV)
+    bias = torch.randn(N).to(torch.bfloat16).to(DEV)
+    want = torch.nn.functional.linear(x.float(), w.float(), bias.float()) + r.float()
+    got = ops.hip_ops.gemm_skinny(x, w, r, bias, ksplit).float()
+    assert torch.allclose(got.cpu(), want.cpu(), atol=5e-2, rtol=5e-2), \
+        (got.cpu() - want.cpu()).abs().max()
+
+
+def test_linear_dispatch_matches_f_linear():
+    torch.manual_seed(3)
+    x = (torch.randn(2, 1, 512) * 0.1).to(torch.bfloat16).to(DEV)
+    w = torch.randn(256, 512).to(torch.bfloat16).to(DEV)
+    got = ops.linear(x, w)
+    want = torch.nn.functional.linear(x, w)
+    assert torch.allclose(got.float().cpu(), want.float().cpu(), atol=2e-2)

@@ -62,7 +62,7 @@ def test_attn_paged_matches_sdpa_prefill():
     B, Hq, Hkv, T, D, P = 2, 4, 2, 23, 32, 16
     npages = 64
     kp = torch.zeros(npages, Hkv, P, D)
-    vp = torch.zeros(npages, Hkv, P, D)
+    vp = torch.zeros(npages, Hkv, D, P)
     maxp = 4
     pt = torch.arange(B * maxp, dtype=torch.int32).reshape(B, maxp)
     q = torch.randn(B, Hq, T, D)
@@ -86,7 +86,7 @@ def test_attn_paged_decode_step_matches_full():
     torch.manual_seed(2)
     B, Hq, Hkv, D, P = 1, 4, 4, 16, 16
     kp = torch.zeros(8, Hkv, P, D)
-    vp = torch.zeros(8, Hkv, P, D)
+    vp = torch.zeros(8, Hkv, D, P)
     pt = torch.arange(8, dtype=torch.int32).reshape(1, 8)
     T = 21
     q = torch.randn(B, Hq, T, D)
@@ -104,7 +104,7 @@ def test_attn_sliding_window():
     torch.manual_seed(3)
     B, H, T, D, P = 1, 2, 40, 16, 16
     kp = torch.zeros(8, H, P, D)
-    vp = torch.zeros(8, H, P, D)
+    vp = torch.zeros(8, H, D, P)
     pt = torch.arange(8, dtype=torch.int32).reshape(1, 8)
     q = torch.randn(B, H, T, D)
     k = torch.randn(B, H, T, D)
@@ -129,7 +129,7 @@ def test_tree_mask_attention():
     torch.manual_seed(4)
     B, H, D, P = 1, 2, 16, 16
     kp = torch.zeros(8, H, P, D)
-    vp = torch.zeros(8, H, P, D)
+    vp = torch.zeros(8, H, D, P)
     pt = torch.arange(8, dtype=torch.int32).reshape(1, 8)
     Tpre, Ttree = 5, 3
     k = torch.randn(B, H, Tpre + Ttree, D)
