@@ -42,10 +42,11 @@ def bench_attn_decode():
         pt = torch.arange(B * maxp, dtype=torch.int32, device=DEV).reshape(B, maxp)
         q = torch.randn(B, Hq, 1, D, dtype=torch.bfloat16, device=DEV) * 0.1
         ctx_l = torch.full((B,), ctx, dtype=torch.int32, device=DEV)
-        t = timeit(lambda: ops.attn_decode(q, kp, vp, pt, ctx_l))
         bytes_ = B * Hkv * ctx * 2 * D * 2
-        print(f"attn_decode B{B} Hq{Hq}/{Hkv} ctx{ctx} D{D}: "
-              f"{t*1e6:8.1f} us  {bytes_/t/1e12:6.2f} TB/s")
+        for ns in (0, 2, 4, 8, 16):
+            t = timeit(lambda: ops.attn_decode(q, kp, vp, pt, ctx_l, n_split=ns))
+            print(f"attn_decode B{B} Hq{Hq}/{Hkv} ctx{ctx} D{D} ns={ns}: "
+                  f"{t*1e6:8.1f} us  {bytes_/t/1e12:6.2f} TB/s")
 
 
 def bench_attn_prefill():

@@ -396,6 +396,16 @@ torch::Tensor gemm_skinny(torch::Tensor A, torch::Tensor W,
     ksplit = (int)std::min<long>(K / 512, std::max<long>(1, 1024 / (N / 64)));
     ksplit = std::max(1, ksplit);
   }
+  const bool v2 = (K % 256) == 0;
+  int kchunk = K;
+  if (v2) {
+    // v2 stages 256-element chunks: round the K slice to a stage multiple and
+    // drop empty trailing splits.
+    const int nch = K / 256;
+    const int per = (nch + ksplit - 1) / ksplit;
+    ksplit = (nch + per - 1) / per;
+    kchunk = per * 256;
+  }
   torch::Tensor part;
   float* pp = nullptr;
   if (ksplit > 1) {
@@ -405,9 +415,14 @@ torch::Tensor gemm_skinny(torch::Tensor A, torch::Tensor W,
   }
   dim3 grid(N / 64, ksplit);
   auto launch = [&](auto mt) {
-    gemm_skinny_kernel<decltype(mt)::value><<<grid, 256, 0, cur_stream()>>>(
-        bf_ptr(A), bf_ptr(W), ksplit == 1 ? rp : nullptr,
-        ksplit == 1 ? bp : nullptr, bf_ptr_mut(C), pp, M, N, K, ksplit);
+    if (v2)
+      gemm_skinny_v2_kernel<decltype(mt)::value><<<grid, 256, 0, cur_stream()>>>(
+          bf_ptr(A), bf_ptr(W), ksplit == 1 ? rp : nullptr,
+          ksplit == 1 ? bp : nullptr, bf_ptr_mut(C), pp, M, N, K, kchunk, ksplit);
+    else
+      gemm_skinny_kernel<decltype(mt)::value><<<grid, 256, 0, cur_stream()>>>(
+          bf_ptr(A), bf_ptr(W), ksplit == 1 ? rp : nullptr,
+          ksplit == 1 ? bp : nullptr, bf_ptr_mut(C), pp, M, N, K, ksplit);
   };
   if (M <= 16) launch(std::integral_constant<int, 1>{});
   else launch(std::integral_constant<int, 2>{});

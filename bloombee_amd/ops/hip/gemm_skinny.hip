@@ -104,6 +104,108 @@ __global__ __launch_bounds__(256) void gemm_skinny_kernel(
   }
 }
 
+// v2: LDS-staged W streaming. v1's B-fragment loads touch 16 W rows per wave
+// instruction in 64 B pieces (rows are K*2 bytes apart) — DRAM sees half-line
+// granules and the kernel capped at ~2.6 TB/s measured. Here a stage
+// cooperatively copies a (64 rows x KSTEP) W tile to LDS with 512 B-dense
+// per-row wave loads (one wave instruction = 1 KB contiguous), and the MFMA
+// B-fragments read from padded LDS conflict-free. Register double-buffering:
+// stage s+1's loads are issued before stage s's MFMAs so HBM latency hides
+// behind compute.
+template <int MT>
+__global__ __launch_bounds__(256) void gemm_skinny_v2_kernel(
+    const unsigned short* __restrict__ A,   // (M, K) bf16
+    const unsigned short* __restrict__ W,   // (N, K) bf16
+    const unsigned short* __restrict__ R,   // (M, N) bf16 residual or null
+    const unsigned short* __restrict__ bias,  // (N,) bf16 or null
+    unsigned short* __restrict__ C,         // (M, N) bf16   (ksplit == 1)
+    float* __restrict__ Cpart,              // (ksplit, M, N) f32 (ksplit > 1)
+    int M, int N, int K, int kchunk, int ksplit) {
+  constexpr int KSTEP = 256;               // elements staged per stage
+  constexpr int NLOAD = KSTEP / 32;        // 16 B pieces per thread per stage
+  constexpr int ROWS = 64;                 // W rows per workgroup
+  constexpr int RSTRIDE = KSTEP + 8;       // LDS row stride (16 B pad)
+  __shared__ __attribute__((aligned(16))) unsigned short wtile[ROWS * RSTRIDE];
+
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int li = lane & 15;
+  const int hi = lane >> 4;
+  const int nblk = blockIdx.x * ROWS;
+  const int split = blockIdx.y;
+  const int k0 = split * kchunk;
+  const int k1 = min(K, k0 + kchunk);
+
+  int arow[MT];
+#pragma unroll
+  for (int t = 0; t < MT; ++t) arow[t] = min(t * 16 + li, M - 1);
+
+  f32x4 acc[MT];
+#pragma unroll
+  for (int t = 0; t < MT; ++t) acc[t] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+  // stage loader: thread's j-th piece = row (j*4 + wave), 16 B at lane*8
+  short8 regs[NLOAD];
+  auto load_stage = [&](int ks) {
+#pragma unroll
+    for (int j = 0; j < NLOAD; ++j) {
+      const int row = j * 4 + wave;
+      regs[j] = *reinterpret_cast<const short8*>(
+          W + (long)(nblk + row) * K + ks + lane * 8);
+    }
+  };
+  load_stage(k0);
+
+  for (int ks = k0; ks < k1; ks += KSTEP) {
+    __syncthreads();  // LDS free (previous stage consumed)
+#pragma unroll
+    for (int j = 0; j < NLOAD; ++j)
+      *reinterpret_cast<short8*>(wtile + (j * 4 + wave) * RSTRIDE + lane * 8) =
+          regs[j];
+    __syncthreads();
+    if (ks + KSTEP < k1) load_stage(ks + KSTEP);  // prefetch next stage
+
+    // wave owns LDS rows [wave*16, wave*16+16) -> C cols nblk + wave*16 + li
+#pragma unroll
+    for (int kk = 0; kk < KSTEP; kk += 32) {
+      bf16x8 bfrag = as_bf16x8(*reinterpret_cast<const short8*>(
+          wtile + (wave * 16 + li) * RSTRIDE + kk + hi * 8));
+#pragma unroll
+      for (int t = 0; t < MT; ++t) {
+        bf16x8 afrag = as_bf16x8(*reinterpret_cast<const short8*>(
+            A + (long)arow[t] * K + ks + kk + hi * 8));
+        acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[t],
+                                                         0, 0, 0);
+      }
+    }
+  }
+
+  const int n0 = nblk + wave * 16;
+  if (ksplit == 1) {
+#pragma unroll
+    for (int t = 0; t < MT; ++t)
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int m = t * 16 + hi * 4 + reg;
+        if (m >= M) continue;
+        float v = acc[t][reg];
+        if (bias) v += bf2f(bias[n0 + li]);
+        if (R) v += bf2f(R[(long)m * N + n0 + li]);
+        C[(long)m * N + n0 + li] = f2bf(v);
+      }
+  } else {
+    float* dst = Cpart + (long)split * M * N;
+#pragma unroll
+    for (int t = 0; t < MT; ++t)
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int m = t * 16 + hi * 4 + reg;
+        if (m >= M) continue;
+        dst[(long)m * N + n0 + li] = acc[t][reg];
+      }
+  }
+}
+
 __global__ void gemm_skinny_combine_kernel(
     const float* __restrict__ Cpart, const unsigned short* __restrict__ R,
     const unsigned short* __restrict__ bias, unsigned short* __restrict__ C,

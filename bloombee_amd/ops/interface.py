@@ -225,7 +225,7 @@ def linear(x: torch.Tensor, w: torch.Tensor, residual: Optional[torch.Tensor] = 
     hipBLASLt via F.linear."""
     M = x.numel() // x.shape[-1]
     if (_on_gpu(x) and M <= 32 and x.dtype == torch.bfloat16
-            and x.shape[-1] % 32 == 0 and w.shape[0] % 64 == 0):
+            and x.shape[-1] % 256 == 0 and w.shape[0] % 64 == 0):
         _require_ext()
         return hip_ops.gemm_skinny(x.contiguous(), w, residual, bias, 0)
     y = torch.nn.functional.linear(x, w, bias)
