@@ -104,14 +104,14 @@ __global__ __launch_bounds__(256) void gemm_skinny_kernel(
   }
 }
 
-// v2: LDS-staged W streaming. v1's B-fragment loads touch 16 W rows per wave
-// instruction in 64 B pieces (rows are K*2 bytes apart) — DRAM sees half-line
-// granules and the kernel capped at ~2.6 TB/s measured. Here a stage
-// cooperatively copies a (64 rows x KSTEP) W tile to LDS with 512 B-dense
-// per-row wave loads (one wave instruction = 1 KB contiguous), and the MFMA
-// B-fragments read from padded LDS conflict-free. Register double-buffering:
-// stage s+1's loads are issued before stage s's MFMAs so HBM latency hides
-// behind compute.
+// v2: software-pipelined direct-VGPR W streaming. The guide's verdict for
+// M<=32 decode-weight GEMMs (cdna_hip_programming.md "GEMV / M <= 16 decode
+// weights" row): the W operand is streamed once and not shared across waves,
+// so an LDS round trip is pure overhead — load straight to VGPRs with a deep
+// unroll and late vmcnt. Two register sets alternate: while set s's MFMAs
+// consume, set s^1's 128-k-deep loads (4 KB of W per wave) are in flight, so
+// each wave keeps ~one HBM-latency's worth of bytes outstanding.
+// Host guarantees (k1 - k0) % 256 == 0 so the set pairing needs no tail.
 template <int MT>
 __global__ __launch_bounds__(256) void gemm_skinny_v2_kernel(
     const unsigned short* __restrict__ A,   // (M, K) bf16
@@ -121,21 +121,18 @@ __global__ __launch_bounds__(256) void gemm_skinny_v2_kernel(
     unsigned short* __restrict__ C,         // (M, N) bf16   (ksplit == 1)
     float* __restrict__ Cpart,              // (ksplit, M, N) f32 (ksplit > 1)
     int M, int N, int K, int kchunk, int ksplit) {
-  constexpr int KSTEP = 256;               // elements staged per stage
-  constexpr int NLOAD = KSTEP / 32;        // 16 B pieces per thread per stage
-  constexpr int ROWS = 64;                 // W rows per workgroup
-  constexpr int RSTRIDE = KSTEP + 8;       // LDS row stride (16 B pad)
-  __shared__ __attribute__((aligned(16))) unsigned short wtile[ROWS * RSTRIDE];
-
+  constexpr int U = 4;                     // k-slices per pipeline set (128 k)
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int li = lane & 15;
   const int hi = lane >> 4;
-  const int nblk = blockIdx.x * ROWS;
+  const int n0 = blockIdx.x * 64 + wave * 16;
+  if (n0 >= N) return;
   const int split = blockIdx.y;
   const int k0 = split * kchunk;
   const int k1 = min(K, k0 + kchunk);
 
+  const unsigned short* wrow = W + (long)(n0 + li) * K;
   int arow[MT];
 #pragma unroll
   for (int t = 0; t < MT; ++t) arow[t] = min(t * 16 + li, M - 1);
@@ -144,43 +141,55 @@ __global__ __launch_bounds__(256) void gemm_skinny_v2_kernel(
 #pragma unroll
   for (int t = 0; t < MT; ++t) acc[t] = (f32x4){0.f, 0.f, 0.f, 0.f};
 
-  // stage loader: thread's j-th piece = row (j*4 + wave), 16 B at lane*8
-  short8 regs[NLOAD];
-  auto load_stage = [&](int ks) {
+  bf16x8 bB0[U], bB1[U], bA0[U][MT], bA1[U][MT];
+  auto issue0 = [&](int k) {
 #pragma unroll
-    for (int j = 0; j < NLOAD; ++j) {
-      const int row = j * 4 + wave;
-      regs[j] = *reinterpret_cast<const short8*>(
-          W + (long)(nblk + row) * K + ks + lane * 8);
+    for (int u = 0; u < U; ++u) {
+      const int kk = k + u * 32 + hi * 8;
+      bB0[u] = as_bf16x8(*reinterpret_cast<const short8*>(wrow + kk));
+#pragma unroll
+      for (int t = 0; t < MT; ++t)
+        bA0[u][t] = as_bf16x8(
+            *reinterpret_cast<const short8*>(A + (long)arow[t] * K + kk));
     }
   };
-  load_stage(k0);
-
-  for (int ks = k0; ks < k1; ks += KSTEP) {
-    __syncthreads();  // LDS free (previous stage consumed)
+  auto issue1 = [&](int k) {
 #pragma unroll
-    for (int j = 0; j < NLOAD; ++j)
-      *reinterpret_cast<short8*>(wtile + (j * 4 + wave) * RSTRIDE + lane * 8) =
-          regs[j];
-    __syncthreads();
-    if (ks + KSTEP < k1) load_stage(ks + KSTEP);  // prefetch next stage
-
-    // wave owns LDS rows [wave*16, wave*16+16) -> C cols nblk + wave*16 + li
+    for (int u = 0; u < U; ++u) {
+      const int kk = k + u * 32 + hi * 8;
+      bB1[u] = as_bf16x8(*reinterpret_cast<const short8*>(wrow + kk));
 #pragma unroll
-    for (int kk = 0; kk < KSTEP; kk += 32) {
-      bf16x8 bfrag = as_bf16x8(*reinterpret_cast<const short8*>(
-          wtile + (wave * 16 + li) * RSTRIDE + kk + hi * 8));
-#pragma unroll
-      for (int t = 0; t < MT; ++t) {
-        bf16x8 afrag = as_bf16x8(*reinterpret_cast<const short8*>(
-            A + (long)arow[t] * K + ks + kk + hi * 8));
-        acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[t],
-                                                         0, 0, 0);
-      }
+      for (int t = 0; t < MT; ++t)
+        bA1[u][t] = as_bf16x8(
+            *reinterpret_cast<const short8*>(A + (long)arow[t] * K + kk));
     }
+  };
+  auto mfma0 = [&]() {
+#pragma unroll
+    for (int u = 0; u < U; ++u)
+#pragma unroll
+      for (int t = 0; t < MT; ++t)
+        acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(bA0[u][t], bB0[u],
+                                                         acc[t], 0, 0, 0);
+  };
+  auto mfma1 = [&]() {
+#pragma unroll
+    for (int u = 0; u < U; ++u)
+#pragma unroll
+      for (int t = 0; t < MT; ++t)
+        acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(bA1[u][t], bB1[u],
+                                                         acc[t], 0, 0, 0);
+  };
+
+  // (k1 - k0) is a multiple of 256 = two sets; pipeline pairs of sets.
+  issue0(k0);
+  for (int k = k0; k < k1; k += 256) {
+    issue1(k + 128);
+    mfma0();
+    if (k + 256 < k1) issue0(k + 256);
+    mfma1();
   }
 
-  const int n0 = nblk + wave * 16;
   if (ksplit == 1) {
 #pragma unroll
     for (int t = 0; t < MT; ++t)
