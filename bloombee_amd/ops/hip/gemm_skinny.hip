@@ -104,13 +104,17 @@ __global__ __launch_bounds__(256) void gemm_skinny_kernel(
   }
 }
 
-// v2: software-pipelined direct-VGPR W streaming. The guide's verdict for
-// M<=32 decode-weight GEMMs (cdna_hip_programming.md "GEMV / M <= 16 decode
-// weights" row): the W operand is streamed once and not shared across waves,
-// so an LDS round trip is pure overhead — load straight to VGPRs with a deep
-// unroll and late vmcnt. Two register sets alternate: while set s's MFMAs
-// consume, set s^1's 128-k-deep loads (4 KB of W per wave) are in flight, so
-// each wave keeps ~one HBM-latency's worth of bytes outstanding.
+// v2: software-pipelined direct-VGPR W streaming + A through LDS.
+// The guide's verdicts for this shape class (cdna_hip_programming.md):
+//  * "GEMV / M <= 16 decode weights" — W is streamed once and not shared
+//    across waves: no LDS round trip, load straight to VGPRs, deep unroll,
+//    late vmcnt. Two register sets alternate so each wave keeps a
+//    HBM-latency's worth of W bytes in flight.
+//  * "the x operand ... through LDS in full 128-B lines, NOT fragment-shaped
+//    loads straight to VGPRs" — per-fragment A loads double TA pressure
+//    (+18..45% measured in the guide); here the (M x 256) A stage is copied
+//    to LDS once per stage with full-line loads and A fragments come from
+//    conflict-free padded ds_read_b128.
 // Host guarantees (k1 - k0) % 256 == 0 so the set pairing needs no tail.
 template <int MT>
 __global__ __launch_bounds__(256) void gemm_skinny_v2_kernel(
@@ -122,72 +126,93 @@ __global__ __launch_bounds__(256) void gemm_skinny_v2_kernel(
     float* __restrict__ Cpart,              // (ksplit, M, N) f32 (ksplit > 1)
     int M, int N, int K, int kchunk, int ksplit) {
   constexpr int U = 4;                     // k-slices per pipeline set (128 k)
+  constexpr int KSTEP = 256;               // A elements staged per stage
+  constexpr int RSTRIDE = KSTEP + 8;       // padded LDS row stride (elements)
+  __shared__ __attribute__((aligned(16))) unsigned short atile[32 * RSTRIDE];
+
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int li = lane & 15;
   const int hi = lane >> 4;
   const int n0 = blockIdx.x * 64 + wave * 16;
-  if (n0 >= N) return;
   const int split = blockIdx.y;
   const int k0 = split * kchunk;
   const int k1 = min(K, k0 + kchunk);
 
   const unsigned short* wrow = W + (long)(n0 + li) * K;
-  int arow[MT];
-#pragma unroll
-  for (int t = 0; t < MT; ++t) arow[t] = min(t * 16 + li, M - 1);
 
   f32x4 acc[MT];
 #pragma unroll
   for (int t = 0; t < MT; ++t) acc[t] = (f32x4){0.f, 0.f, 0.f, 0.f};
 
-  bf16x8 bB0[U], bB1[U], bA0[U][MT], bA1[U][MT];
+  // ---- A staging: the whole workgroup copies A[0:M][ks:ks+256] with
+  // full-line loads; rows beyond M-1 are clamped (their C rows are dropped).
+  // Thread piece j: row = (tid + j*256)/32 clamped, 8 elements at
+  // ((tid + j*256)%32)*8 within the stage.
+  short8 aregs[(32 * KSTEP) / (256 * 8)];  // 4 pieces per thread
+  auto load_a = [&](int ks) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const int flat = threadIdx.x + j * 256;
+      const int row = min(flat / 32, M - 1);
+      const int off = (flat % 32) * 8;
+      aregs[j] = *reinterpret_cast<const short8*>(A + (long)row * K + ks + off);
+    }
+  };
+  auto store_a = [&]() {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const int flat = threadIdx.x + j * 256;
+      *reinterpret_cast<short8*>(atile + (flat / 32) * RSTRIDE + (flat % 32) * 8) =
+          aregs[j];
+    }
+  };
+
+  bf16x8 bB0[U], bB1[U];
   auto issue0 = [&](int k) {
 #pragma unroll
-    for (int u = 0; u < U; ++u) {
-      const int kk = k + u * 32 + hi * 8;
-      bB0[u] = as_bf16x8(*reinterpret_cast<const short8*>(wrow + kk));
-#pragma unroll
-      for (int t = 0; t < MT; ++t)
-        bA0[u][t] = as_bf16x8(
-            *reinterpret_cast<const short8*>(A + (long)arow[t] * K + kk));
-    }
+    for (int u = 0; u < U; ++u)
+      bB0[u] = as_bf16x8(
+          *reinterpret_cast<const short8*>(wrow + k + u * 32 + hi * 8));
   };
   auto issue1 = [&](int k) {
 #pragma unroll
-    for (int u = 0; u < U; ++u) {
-      const int kk = k + u * 32 + hi * 8;
-      bB1[u] = as_bf16x8(*reinterpret_cast<const short8*>(wrow + kk));
-#pragma unroll
-      for (int t = 0; t < MT; ++t)
-        bA1[u][t] = as_bf16x8(
-            *reinterpret_cast<const short8*>(A + (long)arow[t] * K + kk));
-    }
+    for (int u = 0; u < U; ++u)
+      bB1[u] = as_bf16x8(
+          *reinterpret_cast<const short8*>(wrow + k + u * 32 + hi * 8));
   };
-  auto mfma0 = [&]() {
+  // A fragment for m-tile t, k-slice u of the staged 256: ds_read_b128 at
+  // row (t*16 + li), element (u*32 + hi*8) — row stride 528 B => lanes hit
+  // distinct banks.
+  auto mfma_set = [&](bf16x8* bB, int half) {
 #pragma unroll
     for (int u = 0; u < U; ++u)
 #pragma unroll
-      for (int t = 0; t < MT; ++t)
-        acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(bA0[u][t], bB0[u],
-                                                         acc[t], 0, 0, 0);
-  };
-  auto mfma1 = [&]() {
-#pragma unroll
-    for (int u = 0; u < U; ++u)
-#pragma unroll
-      for (int t = 0; t < MT; ++t)
-        acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(bA1[u][t], bB1[u],
-                                                         acc[t], 0, 0, 0);
+      for (int t = 0; t < MT; ++t) {
+        bf16x8 afrag = as_bf16x8(*reinterpret_cast<const short8*>(
+            atile + (t * 16 + li) * RSTRIDE + half * 128 + u * 32 + hi * 8));
+        acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bB[u], acc[t],
+                                                         0, 0, 0);
+      }
   };
 
-  // (k1 - k0) is a multiple of 256 = two sets; pipeline pairs of sets.
+  // Pipeline: stage s's A sits in LDS while its two B register sets stream;
+  // stage s+1's A loads issue right after the barrier that frees them.
+  load_a(k0);
   issue0(k0);
-  for (int k = k0; k < k1; k += 256) {
+  store_a();
+  __syncthreads();
+  for (int k = k0; k < k1; k += KSTEP) {
+    if (k + KSTEP < k1) load_a(k + KSTEP);
     issue1(k + 128);
-    mfma0();
-    if (k + 256 < k1) issue0(k + 256);
-    mfma1();
+    mfma_set(bB0, 0);
+    if (k + KSTEP < k1) issue0(k + KSTEP);
+    mfma_set(bB1, 1);
+    if (k + KSTEP < k1) {
+      __syncthreads();  // everyone done reading stage s
+      store_a();
+      __syncthreads();
+    }
   }
 
   if (ksplit == 1) {
