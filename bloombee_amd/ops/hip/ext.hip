@@ -392,8 +392,13 @@ torch::Tensor gemm_skinny(torch::Tensor A, torch::Tensor W,
   }
   int ksplit = (int)ksplit_req;
   if (ksplit <= 0) {
-    // enough workgroups to fill 256 CUs across 8 XCDs, K slices >= 512
-    ksplit = (int)std::min<long>(K / 512, std::max<long>(1, 1024 / (N / 64)));
+    // Measured optima (benchmarks/bench_kernels.py gemm): big-N shapes are
+    // BW-bound — once N/64 covers the 256 CUs extra splits only add combine
+    // traffic; small-N shapes are per-block-latency-bound and want
+    // (N/64)*ksplit ~ 512-1024 blocks, capped at 8.
+    if (N / 64 >= 256) ksplit = 1;
+    else ksplit = (int)std::min<long>(K / 512,
+                                      std::max<long>(1, std::min<long>(8, 1024 / (N / 64))));
     ksplit = std::max(1, ksplit);
   }
   const bool v2 = (K % 256) == 0;

@@ -134,7 +134,19 @@ __global__ __launch_bounds__(256) void gemm_skinny_v2_kernel(
   const int lane = threadIdx.x & (WAVE - 1);
   const int li = lane & 15;
   const int hi = lane >> 4;
-  const int n0 = blockIdx.x * 64 + wave * 16;
+  // Bijective XCD-aware remap: the dispatcher places block b on XCD b%8, so
+  // without the remap memory-adjacent N-tiles land on different XCDs. Group
+  // them instead: each XCD streams one contiguous W region (guide: +10-12%
+  // when HBM-bound).
+  int tile = blockIdx.x;
+  {
+    const int nwg = gridDim.x;
+    const int xcd = tile % 8, orig8 = tile / 8;
+    const int q = nwg / 8, r = nwg % 8;
+    if (nwg >= 8)
+      tile = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + orig8;
+  }
+  const int n0 = tile * 64 + wave * 16;
   const int split = blockIdx.y;
   const int k0 = split * kchunk;
   const int k1 = min(K, k0 + kchunk);
