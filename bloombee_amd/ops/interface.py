@@ -224,8 +224,14 @@ def linear(x: torch.Tensor, w: torch.Tensor, residual: Optional[torch.Tensor] = 
     ~2x weight-stream bandwidth on the table at M<=32; everything else uses
     hipBLASLt via F.linear."""
     M = x.numel() // x.shape[-1]
-    if (_on_gpu(x) and M <= 32 and x.dtype == torch.bfloat16
-            and x.shape[-1] % 256 == 0 and w.shape[0] % 64 == 0):
+    N, K = w.shape[0], x.shape[-1]
+    # Measured rule (benchmarks/bench_kernels.py gemm, profiles/): the skinny
+    # kernel beats hipBLASLt where blaslt is grid/latency-bound (small N*K)
+    # and on huge-N streams (lm_head); blaslt keeps the two mid-size MLP
+    # shapes (gate_up/down) until the v3 kernel closes the last ~15%.
+    use_skinny = (N * K <= (1 << 25)) or (N >= 65536)
+    if (_on_gpu(x) and M <= 32 and x.dtype == torch.bfloat16 and use_skinny
+            and K % 256 == 0 and N % 64 == 0):
         _require_ext()
         return hip_ops.gemm_skinny(x.contiguous(), w, residual, bias, 0)
     y = torch.nn.functional.linear(x, w, bias)
