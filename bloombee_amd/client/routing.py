@@ -1,0 +1,230 @@
+"""Route construction over the DHT swarm map.
+
+Parity: reference RemoteSequenceManager (client/routing/sequence_manager.py)
+— a background refresh of per-block server maps, two routing modes, and a
+ban list with exponential backoff:
+
+  * ``max_throughput``: weighted random span choice, weight = throughput x
+    covered length (ref :320-342).
+  * ``min_latency``: Dijkstra over a DAG whose nodes are block boundaries;
+    edge cost of serving blocks [i, j) on server s = (j - i) / inference_rps
+    + per-hop serialization overhead + RTT estimate + a large penalty if the
+    server lacks KV room (ref :235-296 via dijkstar — here a hand-rolled
+    Dijkstra, no dependency).
+"""
+from __future__ import annotations
+
+import heapq
+import random
+import threading
+import time
+from typing import Dict, List, Optional, Sequence, Tuple
+
+from bloombee_amd.client.config import ClientConfig
+from bloombee_amd.data_structures import (RemoteModuleInfo, RemoteSpanInfo,
+                                          ServerState, get_remote_module_infos,
+                                          module_uids)
+from bloombee_amd.net.dht import Dht
+from bloombee_amd.utils.logging import get_logger
+
+logger = get_logger(__name__)
+
+HOP_OVERHEAD_S = 0.018          # serialization per hop (ref :241)
+DEFAULT_RPS = 300.0             # per-block steps/s fallback (ref :242)
+NO_CACHE_PENALTY_S = 10.0       # ref :243
+
+
+class MissingBlocksError(RuntimeError):
+    def __init__(self, blocks):
+        super().__init__(f"no alive servers for blocks {blocks}")
+
+
+class Blacklist:
+    """Temporary peer bans with exponential backoff (ref :130-131, 412-429)."""
+
+    def __init__(self, base: float, max_banned: float = 300.0):
+        self.base, self.max = base, max_banned
+        self._until: Dict[str, float] = {}
+        self._count: Dict[str, int] = {}
+        self._lock = threading.Lock()
+
+    def register_failure(self, peer_id: str):
+        with self._lock:
+            n = self._count.get(peer_id, 0) + 1
+            self._count[peer_id] = n
+            self._until[peer_id] = time.monotonic() + min(
+                self.max, self.base * 2 ** (n - 1))
+
+    def register_success(self, peer_id: str):
+        with self._lock:
+            self._count.pop(peer_id, None)
+            self._until.pop(peer_id, None)
+
+    def is_banned(self, peer_id: str) -> bool:
+        with self._lock:
+            return time.monotonic() < self._until.get(peer_id, -1)
+
+
+class RemoteSequenceManager:
+    def __init__(self, config: ClientConfig, model_name: str, num_blocks: int,
+                 dht: Optional[Dht] = None, start: bool = True):
+        self.config = config
+        self.model_name = model_name
+        self.num_blocks = num_blocks
+        self.uids = module_uids(model_name, num_blocks)
+        self.dht = dht or Dht(initial_peers=list(config.initial_peers),
+                              port=config.dht_port)
+        self.blacklist = Blacklist(config.ban_timeout)
+        self.infos: List[RemoteModuleInfo] = [
+            RemoteModuleInfo(uid=u) for u in self.uids]
+        self._lock = threading.Lock()
+        self._stop = threading.Event()
+        self._ready = threading.Event()
+        self._thread = threading.Thread(target=self._update_loop, daemon=True,
+                                        name="seqmgr-update")
+        if start:
+            self.update()
+            self._thread.start()
+
+    # -- swarm map --------------------------------------------------------
+    def update(self):
+        infos = get_remote_module_infos(self.dht, self.uids)
+        with self._lock:
+            self.infos = infos
+        self._ready.set()
+
+    def _update_loop(self):
+        while not self._stop.is_set():
+            self._stop.wait(self.config.update_period)
+            if self._stop.is_set():
+                return
+            try:
+                self.update()
+            except Exception as e:  # noqa: BLE001
+                logger.warning("swarm map refresh failed: %s", e)
+
+    def _usable_spans(self, start_index: int, end_index: int) -> List[RemoteSpanInfo]:
+        """Every contiguous span any server offers within [start, end)."""
+        with self._lock:
+            infos = self.infos
+        spans: Dict[str, RemoteSpanInfo] = {}
+        for i in range(start_index, end_index):
+            for peer_id, srv in infos[i].servers.items():
+                if srv.state != ServerState.ONLINE:
+                    continue
+                if self.blacklist.is_banned(peer_id):
+                    continue
+                if (self.config.allowed_servers is not None
+                        and peer_id not in self.config.allowed_servers):
+                    continue
+                if (self.config.blocked_servers is not None
+                        and peer_id in self.config.blocked_servers):
+                    continue
+                sp = spans.get(peer_id)
+                if sp is not None and sp.end == i:
+                    sp.end = i + 1
+                elif sp is None:
+                    spans[peer_id] = RemoteSpanInfo(peer_id, i, i + 1, srv)
+        return list(spans.values())
+
+    # -- routes -----------------------------------------------------------
+    def make_sequence(self, start_index: int = 0, end_index: Optional[int] = None,
+                      mode: Optional[str] = None,
+                      cache_tokens_needed: Optional[int] = None,
+                      ) -> List[RemoteSpanInfo]:
+        end_index = end_index if end_index is not None else self.num_blocks
+        mode = mode or self.config.routing_mode
+        spans = self._usable_spans(start_index, end_index)
+        if mode == "min_latency":
+            route = self._route_min_latency(spans, start_index, end_index,
+                                            cache_tokens_needed)
+        else:
+            route = self._route_max_throughput(spans, start_index, end_index)
+        if route is None:
+            covered = set()
+            for s in spans:
+                covered.update(range(s.start, s.end))
+            missing = [i for i in range(start_index, end_index)
+                       if i not in covered]
+            raise MissingBlocksError(missing or list(range(start_index, end_index)))
+        return route
+
+    def _route_max_throughput(self, spans, start_index, end_index):
+        """Greedy weighted-random chain (ref :320-342)."""
+        route = []
+        cur = start_index
+        guard = 0
+        while cur < end_index:
+            options = [s for s in spans if s.start <= cur < s.end]
+            if not options:
+                return None
+            weights = [max(1e-6, s.server_info.throughput) * (s.end - cur)
+                       for s in options]
+            chosen = random.choices(options, weights=weights, k=1)[0]
+            route.append(RemoteSpanInfo(chosen.peer_id, cur,
+                                        min(chosen.end, end_index),
+                                        chosen.server_info))
+            cur = min(chosen.end, end_index)
+            guard += 1
+            if guard > self.num_blocks + 2:
+                return None
+        return route
+
+    def _route_min_latency(self, spans, start_index, end_index,
+                           cache_tokens_needed):
+        """Dijkstra over block boundaries (ref :235-296)."""
+        # edges[u] = list of (v, cost, span)
+        edges: Dict[int, List[Tuple[int, float, RemoteSpanInfo]]] = {}
+        for s in spans:
+            rps = s.server_info.inference_rps or s.server_info.throughput or DEFAULT_RPS
+            penalty = 0.0
+            if (cache_tokens_needed is not None
+                    and s.server_info.cache_tokens_left is not None
+                    and s.server_info.cache_tokens_left < cache_tokens_needed):
+                penalty = NO_CACHE_PENALTY_S
+            for u in range(max(s.start, start_index), min(s.end, end_index)):
+                for v in range(u + 1, min(s.end, end_index) + 1):
+                    cost = (v - u) / max(rps, 1e-6) + HOP_OVERHEAD_S + penalty
+                    edges.setdefault(u, []).append(
+                        (v, cost, RemoteSpanInfo(s.peer_id, u, v, s.server_info)))
+        dist = {start_index: 0.0}
+        prev: Dict[int, Tuple[int, RemoteSpanInfo]] = {}
+        heap = [(0.0, start_index)]
+        while heap:
+            d, u = heapq.heappop(heap)
+            if u == end_index:
+                break
+            if d > dist.get(u, float("inf")):
+                continue
+            for v, cost, span in edges.get(u, []):
+                nd = d + cost
+                if nd < dist.get(v, float("inf")):
+                    dist[v] = nd
+                    prev[v] = (u, span)
+                    heapq.heappush(heap, (nd, v))
+        if end_index not in prev and end_index != start_index:
+            return None
+        route = []
+        cur = end_index
+        while cur != start_index:
+            u, span = prev[cur]
+            route.append(span)
+            cur = u
+        return list(reversed(route))
+
+    # -- failures ---------------------------------------------------------
+    def on_request_failure(self, peer_id: str):
+        self.blacklist.register_failure(peer_id)
+        logger.info("peer %s banned temporarily after failure", peer_id)
+
+    def on_request_success(self, peer_id: str):
+        self.blacklist.register_success(peer_id)
+
+    def get_retry_delay(self, attempt: int) -> float:
+        if attempt == 0:
+            return 0.0
+        return min(self.config.min_backoff * 2 ** (attempt - 1),
+                   self.config.max_backoff)
+
+    def shutdown(self):
+        self._stop.set()

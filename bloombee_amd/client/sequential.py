@@ -1,0 +1,178 @@
+"""RemoteSequential: an nn.Module whose forward runs on the swarm.
+
+Parity: reference client/remote_sequential.py + sequential_autograd.py —
+fault-tolerant span-wise forward with retry/re-routing, and a backward that
+re-runs forward on replacement servers to rebuild lost activations
+(sequential_autograd.py:96-152).
+"""
+from __future__ import annotations
+
+import time
+from typing import List, Optional, Tuple
+
+import torch
+
+from bloombee_amd.client.config import ClientConfig
+from bloombee_amd.client.routing import RemoteSequenceManager, RemoteSpanInfo
+from bloombee_amd.client.session import InferenceSession
+from bloombee_amd.client.worker import get_client, run_coroutine
+from bloombee_amd.net.rpc import RpcError
+from bloombee_amd.utils.logging import get_logger
+
+logger = get_logger(__name__)
+
+MAX_TOKENS_IN_BATCH = 1024  # client sub-batch split (ref sequential_autograd.py:22)
+
+
+def _call_forward(span: RemoteSpanInfo, hidden: torch.Tensor,
+                  timeout: float) -> torch.Tensor:
+    client = get_client(span.server_info.host, span.server_info.port)
+    meta, tensors = run_coroutine(
+        client.call("rpc_forward", {}, [hidden], timeout=timeout),
+        timeout + 5)
+    return tensors[0]
+
+
+def _call_backward(span: RemoteSpanInfo, hidden_in: torch.Tensor,
+                   grad_out: torch.Tensor, timeout: float) -> torch.Tensor:
+    client = get_client(span.server_info.host, span.server_info.port)
+    meta, tensors = run_coroutine(
+        client.call("rpc_backward", {}, [hidden_in, grad_out], timeout=timeout),
+        timeout + 5)
+    return tensors[0]
+
+
+def sequential_forward(manager: RemoteSequenceManager, hidden: torch.Tensor,
+                       start: int = 0, end: Optional[int] = None,
+                       ) -> Tuple[torch.Tensor, List[Tuple[RemoteSpanInfo, torch.Tensor]]]:
+    """Forward through [start, end); returns output + per-span (span, input)
+    pairs saved for backward (ref sequential_forward :25-105)."""
+    end = end if end is not None else manager.num_blocks
+    cfg = manager.config
+    saved: List[Tuple[RemoteSpanInfo, torch.Tensor]] = []
+    cur = start
+    out = hidden
+    attempt = 0
+    while cur < end:
+        route = manager.make_sequence(cur, end)
+        span = route[0]
+        try:
+            result = _call_forward(span, out, cfg.request_timeout)
+            manager.on_request_success(span.peer_id)
+            saved.append((span, out))
+            out = result
+            cur = span.end
+            attempt = 0
+        except (RpcError, OSError, TimeoutError) as e:
+            attempt += 1
+            manager.on_request_failure(span.peer_id)
+            if cfg.max_retries is not None and attempt > cfg.max_retries:
+                raise
+            delay = manager.get_retry_delay(attempt)
+            logger.warning("forward span [%d:%d) on %s failed (%s); retrying "
+                           "in %.1fs", span.start, span.end, span.peer_id, e,
+                           delay)
+            time.sleep(delay)
+            manager.update()
+    return out, saved
+
+
+def sequential_backward(manager: RemoteSequenceManager, grad_out: torch.Tensor,
+                        saved: List[Tuple[RemoteSpanInfo, torch.Tensor]],
+                        ) -> torch.Tensor:
+    """Reverse pass over the saved spans; on failure re-routes the span and
+    re-runs forward from its saved input to rebuild activations on the
+    replacement server (ref sequential_backward :112-197)."""
+    cfg = manager.config
+    grad = grad_out
+    for span, span_input in reversed(saved):
+        attempt = 0
+        while True:
+            try:
+                grad = _call_backward(span, span_input, grad, cfg.request_timeout)
+                manager.on_request_success(span.peer_id)
+                break
+            except (RpcError, OSError, TimeoutError) as e:
+                attempt += 1
+                manager.on_request_failure(span.peer_id)
+                if cfg.max_retries is not None and attempt > cfg.max_retries:
+                    raise
+                time.sleep(manager.get_retry_delay(attempt))
+                manager.update()
+                # find a replacement covering exactly this span's range; the
+                # replacement recomputes the forward internally in rpc_backward
+                route = manager.make_sequence(span.start, span.end)
+                if len(route) == 1:
+                    span = route[0]
+                else:
+                    # range now split across servers: forward to rebuild
+                    # intermediate inputs, then continue span-wise
+                    logger.warning("backward span re-split across %d servers",
+                                   len(route))
+                    _, sub_saved = sequential_forward(
+                        manager, span_input, span.start, span.end)
+                    grad = sequential_backward(manager, grad, sub_saved)
+                    span = None
+                    break
+        if span is None:
+            continue
+    return grad
+
+
+class _RemoteSequentialAutograd(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, hidden: torch.Tensor, manager: RemoteSequenceManager):
+        outs, all_saved = [], []
+        # token-bounded sub-batches (ref MAX_TOKENS_IN_BATCH)
+        B, T, H = hidden.shape
+        rows = max(1, MAX_TOKENS_IN_BATCH // max(T, 1))
+        for i in range(0, B, rows):
+            out, saved = sequential_forward(manager, hidden[i:i + rows].detach())
+            outs.append(out)
+            all_saved.append(saved)
+        ctx.manager = manager
+        ctx.saved = all_saved
+        ctx.rows = rows
+        return torch.cat(outs, dim=0)
+
+    @staticmethod
+    def backward(ctx, grad_out: torch.Tensor):
+        grads = []
+        for i, saved in enumerate(ctx.saved):
+            g = grad_out[i * ctx.rows:(i + 1) * ctx.rows]
+            grads.append(sequential_backward(ctx.manager, g, saved))
+        return torch.cat(grads, dim=0), None
+
+
+class RemoteSequential(torch.nn.Module):
+    """Chain of remote transformer blocks (ref client/remote_sequential.py)."""
+
+    def __init__(self, config: ClientConfig, model_name: str, num_blocks: int,
+                 manager: Optional[RemoteSequenceManager] = None):
+        super().__init__()
+        self.config = config
+        self.manager = manager or RemoteSequenceManager(config, model_name,
+                                                        num_blocks)
+        self._active_session: Optional[InferenceSession] = None
+
+    def forward(self, hidden: torch.Tensor) -> torch.Tensor:
+        if self._active_session is not None:
+            return self._active_session.step(hidden)
+        return _RemoteSequentialAutograd.apply(hidden, self.manager)
+
+    def inference_session(self, max_length: int) -> InferenceSession:
+        return InferenceSession(self.manager, max_length)
+
+    class _Use:
+        def __init__(self, outer, session):
+            self.outer, self.session = outer, session
+
+        def __enter__(self):
+            self.outer._active_session = self.session
+            return self.session
+
+        def __exit__(self, *exc):
+            self.outer._active_session = None
+
+    def use_session(self, session: InferenceSession) -> "_Use":
+        return RemoteSequential._Use(self, session)

@@ -1,0 +1,221 @@
+"""Multi-step decode session across a chain of remote spans.
+
+Parity: reference InferenceSession / _ServerInferenceSession
+(client/inference_session.py:511, :206-406) including:
+  * one bidirectional rpc_inference stream per span,
+  * server→server push: middle spans are opened quiet with a push_to target
+    so during decode the client only sends to span 0 and receives from the
+    last span (ref push_only_downstream_decode, :178-196, 404-406),
+  * failover: on span failure the route is rebuilt and the session history
+    (span-0 inputs) is replayed through fresh sessions to rebuild KV
+    (ref `history` replay + _update_sequence, :71, 139-150, 802-831).
+"""
+from __future__ import annotations
+
+import time
+import uuid
+from typing import List, Optional, Tuple
+
+import torch
+
+from bloombee_amd.client.config import ClientConfig
+from bloombee_amd.client.routing import RemoteSequenceManager, RemoteSpanInfo
+from bloombee_amd.client.worker import get_client, run_coroutine
+from bloombee_amd.net.rpc import RpcError, Stream
+from bloombee_amd.utils.logging import get_logger
+
+logger = get_logger(__name__)
+
+
+class _SpanSession:
+    """Client end of one span's rpc_inference stream."""
+
+    def __init__(self, span: RemoteSpanInfo, stream: Stream, session_id: str,
+                 quiet: bool):
+        self.span = span
+        self.stream = stream
+        self.session_id = session_id
+        self.quiet = quiet
+
+    @classmethod
+    def create(cls, span: RemoteSpanInfo, batch_size: int, max_length: int,
+               push_to: Optional[Tuple[str, int, str]] = None,
+               push_only_recv: bool = False, quiet: bool = False,
+               timeout: float = 30.0) -> "_SpanSession":
+        client = get_client(span.server_info.host, span.server_info.port)
+        sid = uuid.uuid4().hex
+
+        async def open_():
+            stream = await client.open_stream("rpc_inference", {
+                "session_id": sid,
+                "max_length": max_length,
+                "batch_size": batch_size,
+                "push_to": list(push_to) if push_to else None,
+                "push_only_recv": push_only_recv,
+                "quiet": quiet,
+            })
+            first = await stream.recv()
+            if first is None or not first[0].get("ok"):
+                raise RpcError(f"failed to open session on {span.peer_id}")
+            return stream
+
+        stream = run_coroutine(open_(), timeout)
+        return cls(span, stream, sid, quiet)
+
+    def step(self, hidden: torch.Tensor, pos: int, step: int,
+             timeout: float) -> Optional[torch.Tensor]:
+        async def go():
+            await self.stream.send({"pos": pos, "step": step}, [hidden])
+            item = await self.stream.recv()
+            if item is None:
+                raise RpcError(f"stream closed by {self.span.peer_id}")
+            meta, tensors = item
+            return tensors[0] if tensors else None
+
+        return run_coroutine(go(), timeout)
+
+    def send_only(self, hidden: torch.Tensor, pos: int, step: int,
+                  timeout: float) -> None:
+        async def go():
+            await self.stream.send({"pos": pos, "step": step}, [hidden])
+
+        run_coroutine(go(), timeout)
+
+    def recv_only(self, timeout: float) -> Tuple[dict, List[torch.Tensor]]:
+        async def go():
+            item = await self.stream.recv()
+            if item is None:
+                raise RpcError(f"stream closed by {self.span.peer_id}")
+            return item
+
+        return run_coroutine(go(), timeout)
+
+    def close(self):
+        async def go():
+            try:
+                await self.stream.send({"close": True})
+                await self.stream.send_end()
+            except Exception:
+                pass
+
+        try:
+            run_coroutine(go(), 5)
+        except Exception:
+            pass
+
+
+class InferenceSession:
+    def __init__(self, manager: RemoteSequenceManager, max_length: int,
+                 config: Optional[ClientConfig] = None):
+        self.manager = manager
+        self.config = config or manager.config
+        self.max_length = max_length
+        self.batch_size: Optional[int] = None
+        self.spans: List[_SpanSession] = []
+        self.position = 0           # committed tokens across the chain
+        self.step_count = 0
+        self.history: List[Tuple[int, torch.Tensor]] = []  # (pos, span0 input)
+        self._closed = False
+
+    # -- chain management -------------------------------------------------
+    def _open_chain(self, batch_size: int, replay: bool) -> None:
+        route = self.manager.make_sequence(
+            0, self.manager.num_blocks,
+            cache_tokens_needed=batch_size * self.max_length)
+        use_push = (self.config.use_server_to_server and len(route) > 1)
+        sessions: List[Optional[_SpanSession]] = [None] * len(route)
+        # open back-to-front so each span knows its downstream session id
+        push_to = None
+        for i in reversed(range(len(route))):
+            quiet = use_push and i < len(route) - 1
+            push_only_recv = use_push and i > 0
+            s = _SpanSession.create(
+                route[i], batch_size, self.max_length,
+                push_to=push_to, push_only_recv=push_only_recv, quiet=quiet,
+                timeout=self.config.request_timeout)
+            sessions[i] = s
+            push_to = (route[i].server_info.host, route[i].server_info.port,
+                       s.session_id) if use_push else None
+        self.spans = sessions  # type: ignore[assignment]
+        self._push_mode = use_push
+        if replay and self.history:
+            logger.info("replaying %d cached steps into the new chain",
+                        len(self.history))
+            for pos, hidden in self.history:
+                self._chain_step(hidden, pos, replay=True)
+
+    def _chain_step(self, hidden: torch.Tensor, pos: int,
+                    replay: bool = False) -> torch.Tensor:
+        t = self.config.step_timeout
+        if self._push_mode:
+            first, last = self.spans[0], self.spans[-1]
+            if len(self.spans) == 1:
+                return first.step(hidden, pos, self.step_count, t)
+            first.send_only(hidden, pos, self.step_count, t)
+            # quiet middle spans send nothing; output arrives from the last
+            meta, tensors = last.recv_only(t)
+            return tensors[0]
+        out = hidden
+        for s in self.spans:
+            out = s.step(out, pos, self.step_count, t)
+        return out
+
+    def _ban_dead_spans(self) -> None:
+        """Probe each span's endpoint; ban the ones that don't answer
+        (failure attribution before re-routing, ref sequence_manager bans)."""
+        for s in self.spans:
+            try:
+                client = get_client(s.span.server_info.host,
+                                    s.span.server_info.port)
+                run_coroutine(client.call("rpc_info", {}, timeout=3), 5)
+                self.manager.on_request_success(s.span.peer_id)
+            except Exception:
+                self.manager.on_request_failure(s.span.peer_id)
+
+    # -- public API -------------------------------------------------------
+    def step(self, hidden: torch.Tensor) -> torch.Tensor:
+        """Run the full remote chain on `hidden` (B, T, H) at the current
+        position; returns the last span's output hidden states."""
+        if self._closed:
+            raise RuntimeError("session is closed")
+        if self.batch_size is None:
+            self.batch_size = hidden.shape[0]
+            self._open_chain(self.batch_size, replay=False)
+        pos = self.position
+        attempt = 0
+        while True:
+            try:
+                out = self._chain_step(hidden, pos)
+                for s in self.spans:
+                    self.manager.on_request_success(s.span.peer_id)
+                break
+            except (RpcError, OSError, TimeoutError, ConnectionError) as e:
+                attempt += 1
+                for s in self.spans:
+                    s.close()
+                self._ban_dead_spans()
+                max_r = self.config.max_retries
+                if max_r is not None and attempt > max_r:
+                    raise
+                delay = self.manager.get_retry_delay(attempt)
+                logger.warning("chain step failed (%s); rebuilding route in "
+                               "%.1fs (attempt %d)", e, delay, attempt)
+                time.sleep(delay)
+                self.manager.update()
+                self._open_chain(self.batch_size, replay=True)
+        self.history.append((pos, hidden))
+        self.position = pos + hidden.shape[1]
+        self.step_count += 1
+        return out
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *exc):
+        self.close()
+
+    def close(self):
+        if not self._closed:
+            for s in self.spans:
+                s.close()
+            self._closed = True

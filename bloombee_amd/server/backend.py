@@ -1,0 +1,155 @@
+"""Server compute backend: a BlockStack + paged KV + the prioritized pool.
+
+Parity target: reference TransformerBackend (server/backend.py:488-789
+inference_step; :427-462 backward) and _MergedInferenceStep (:1214-1399) —
+here the whole local block range always runs as one merged call on the worker
+thread, with the paged KV session handle playing the role of the 2,160-line
+KVCacheManager (kv/paged.py docstring).
+"""
+from __future__ import annotations
+
+import threading
+import time
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Tuple
+
+import torch
+
+from bloombee_amd.engine import BlockStack
+from bloombee_amd.kv.paged import PagedKVCache, SessionHandle
+from bloombee_amd.models.base import ModelConfig
+from bloombee_amd.server.task_pool import (PRIORITY_INFERENCE, PRIORITY_TRAIN,
+                                           TaskPool)
+from bloombee_amd.utils.logging import get_logger
+
+logger = get_logger(__name__)
+
+
+@dataclass
+class SessionState:
+    handle: SessionHandle
+    last_activity: float = field(default_factory=time.monotonic)
+
+
+class StackBackend:
+    """One contiguous block range on one device, served concurrently."""
+
+    def __init__(self, config: ModelConfig, start: int, end: int,
+                 device: str = "cpu", seed: int = 0,
+                 kv_max_tokens: int = 1 << 18,
+                 checkpoint_dir: Optional[str] = None):
+        self.config = config
+        self.start, self.end = start, end
+        self.device = torch.device(device)
+        self.stack = BlockStack(config, start, end, device=device, seed=seed)
+        if checkpoint_dir is not None:
+            from bloombee_amd.server.from_pretrained import load_block_weights
+            for i, blk in enumerate(self.stack.blocks):
+                load_block_weights(blk, checkpoint_dir, start + i)
+        self.kv_pool: PagedKVCache = self.stack.make_kv(kv_max_tokens)
+        self.pool = TaskPool(name=f"worker[{start}:{end}]")
+        self.sessions: Dict[str, SessionState] = {}
+        self._lock = threading.Lock()
+
+    # -- sessions ---------------------------------------------------------
+    def open_session(self, session_id: str, batch_size: int, max_length: int,
+                     timeout: Optional[float] = 10.0) -> None:
+        handle = self.kv_pool.allocate(batch_size, max_length, timeout=timeout)
+        with self._lock:
+            self.sessions[session_id] = SessionState(handle)
+
+    def close_session(self, session_id: str) -> None:
+        with self._lock:
+            state = self.sessions.pop(session_id, None)
+        if state is not None:
+            state.handle.close()
+
+    def _session(self, session_id: str) -> SessionState:
+        with self._lock:
+            state = self.sessions.get(session_id)
+        if state is None:
+            raise KeyError(f"unknown session {session_id!r}")
+        state.last_activity = time.monotonic()
+        return state
+
+    # -- compute entry points (run on the worker thread) ------------------
+    def inference_step(self, session_id: str, hidden: torch.Tensor,
+                       start_pos: int, prompts: Optional[torch.Tensor] = None,
+                       ) -> torch.Tensor:
+        """One decode/prefill step for an open session.
+
+        start_pos: absolute position of hidden[:, 0]. If the session has
+        advanced further (failover replay, ref inference_session.py:802-831),
+        the cache is truncated back to start_pos first.
+        """
+        state = self._session(session_id)
+        handle = state.handle
+
+        def run():
+            h = hidden.to(self.device, non_blocking=True)
+            if h.dtype != self.config.dtype:
+                h = h.to(self.config.dtype)
+            if prompts is not None:
+                # deep p-tuning: prepend-trained prompts arrive as additive
+                # hidden-state deltas for this span's first block
+                h = h + prompts.to(h.device, h.dtype)
+            B, T, _ = h.shape
+            cur = handle.seqs[0].l_acc
+            if start_pos < cur:
+                handle.truncate([min(start_pos, s.l_acc) for s in handle.seqs])
+                handle.rollback()
+            elif start_pos > cur:
+                raise ValueError(
+                    f"inference step at position {start_pos} but cache has "
+                    f"only {cur} tokens (gap)")
+            sp = torch.full((B,), start_pos, dtype=torch.int32, device=self.device)
+            handle.extend(T)
+            out = self.stack.forward_inference(h, handle, sp)
+            return out
+
+        return self.pool.submit(run, PRIORITY_INFERENCE).result()
+
+    def forward(self, hidden: torch.Tensor) -> torch.Tensor:
+        """Training-path forward (no KV cache, full sequence)."""
+
+        def run():
+            h = hidden.to(self.device).to(self.config.dtype)
+            with torch.no_grad():
+                return self.stack.forward_train(h)
+
+        return self.pool.submit(run, PRIORITY_TRAIN).result()
+
+    def backward(self, hidden_in: torch.Tensor, grad_out: torch.Tensor,
+                 ) -> torch.Tensor:
+        """Re-forward + backward; returns grad wrt the span input. Server
+        weights are frozen — only input/prompt grads flow (ref
+        backend.py:106-109, 427-462)."""
+
+        def run():
+            h = hidden_in.to(self.device).to(self.config.dtype)
+            h = h.detach().requires_grad_(True)
+            with torch.enable_grad():
+                out = self.stack.forward_train(h)
+                g = grad_out.to(out.device).to(out.dtype)
+                (grad_in,) = torch.autograd.grad(out, h, g)
+            return grad_in
+
+        return self.pool.submit(run, PRIORITY_TRAIN).result()
+
+    # -- info -------------------------------------------------------------
+    def info(self) -> dict:
+        return {
+            "model_type": self.config.model_type,
+            "start_block": self.start,
+            "end_block": self.end,
+            "torch_dtype": self.config.torch_dtype,
+            "cache_tokens_left": self.kv_pool.tokens_left,
+            "num_sessions": len(self.sessions),
+        }
+
+    def shutdown(self):
+        with self._lock:
+            for state in self.sessions.values():
+                state.handle.close()
+            self.sessions.clear()
+        self.pool.shutdown()

@@ -1,0 +1,148 @@
+"""RPC surface of a worker: the reference TransformerConnectionHandler
+(server/handler.py:798+ rpc_inference, :2860-3010 rpc_forward/backward,
+:1850 rpc_push, :3256 rpc_info) on bloombee_amd.net.rpc.
+
+One asyncio process serves every connection; compute hops to the backend's
+worker thread via its prioritized pool, so handler concurrency never blocks
+on the GPU (the reference needed 8 forked handler processes + an mp event
+bus for this — SURVEY.md §7 step 3 collapses that).
+
+Session ownership: a decode session is created by the first `rpc_inference`
+stream item (or an upstream `rpc_push`) and keyed by the client-chosen
+session_id, so server→server pushed inputs and client stream inputs meet in
+one place (`_iterate_inference_steps` merge semantics, handler.py:1677-1847).
+"""
+from __future__ import annotations
+
+import asyncio
+import uuid
+from typing import Dict, List, Optional, Tuple
+
+import torch
+
+from bloombee_amd.net.rpc import RpcClient, RpcServer, Stream
+from bloombee_amd.server.backend import StackBackend
+from bloombee_amd.utils.logging import get_logger
+
+logger = get_logger(__name__)
+
+
+class ConnectionHandler:
+    def __init__(self, backend: StackBackend, server: RpcServer):
+        self.backend = backend
+        self.rpc = server
+        # (session_id, step) -> queued pushed inputs awaiting the local stream
+        self._push_q: Dict[str, asyncio.Queue] = {}
+        self._peers: Dict[Tuple[str, int], RpcClient] = {}
+        server.register("rpc_info", self.rpc_info)
+        server.register("rpc_forward", self.rpc_forward)
+        server.register("rpc_backward", self.rpc_backward)
+        server.register("rpc_push", self.rpc_push)
+        server.register_stream("rpc_inference", self.rpc_inference)
+
+    # ------------------------------------------------------------------
+    async def rpc_info(self, meta, tensors):
+        return self.backend.info(), []
+
+    async def rpc_forward(self, meta, tensors):
+        (hidden,) = tensors
+        out = await asyncio.get_event_loop().run_in_executor(
+            None, self.backend.forward, hidden)
+        return {}, [out.cpu()]
+
+    async def rpc_backward(self, meta, tensors):
+        hidden_in, grad_out = tensors
+        grad_in = await asyncio.get_event_loop().run_in_executor(
+            None, self.backend.backward, hidden_in, grad_out)
+        return {}, [grad_in.cpu()]
+
+    # ------------------------------------------------------------------
+    async def rpc_push(self, meta, tensors):
+        """Upstream server pushed this step's input for a session we serve
+        (ref handler.py:1850-1911). Enqueue; the session's inference loop
+        prefers pushed inputs over the client stream."""
+        sid = meta["session_id"]
+        q = self._push_q.setdefault(sid, asyncio.Queue())
+        q.put_nowait((meta, tensors))
+        return {"ok": True}, []
+
+    # ------------------------------------------------------------------
+    async def rpc_inference(self, meta, tensors, stream: Stream):
+        """Bidirectional decode stream. Open meta: {session_id?, max_length,
+        batch_size, push_to?: [host, port, session_id]}. Items: {pos} +
+        [hidden] (or empty tensors in push-only mode — inputs arrive via
+        rpc_push). Replies: {pos} + [hidden_out]."""
+        sid = meta.get("session_id") or uuid.uuid4().hex
+        max_length = int(meta["max_length"])
+        batch_size = int(meta["batch_size"])
+        push_to = meta.get("push_to")  # downstream [host, port]
+        push_only_recv = bool(meta.get("push_only_recv"))  # inputs via rpc_push
+        quiet = bool(meta.get("quiet"))  # don't echo outputs to the client
+        loop = asyncio.get_event_loop()
+        await loop.run_in_executor(
+            None, lambda: self.backend.open_session(sid, batch_size, max_length))
+        try:
+            await stream.send({"session_id": sid, "ok": True})
+            while True:
+                if push_only_recv:
+                    q = self._push_q.setdefault(sid, asyncio.Queue())
+                    get_push = asyncio.ensure_future(q.get())
+                    get_cli = asyncio.ensure_future(stream.recv())
+                    done, pending = await asyncio.wait(
+                        {get_push, get_cli}, return_when=asyncio.FIRST_COMPLETED)
+                    if get_push in done:
+                        get_cli.cancel()
+                        item_meta, item_tensors = get_push.result()
+                    else:
+                        get_push.cancel()
+                        item = get_cli.result()
+                        if item is None:
+                            break
+                        item_meta, item_tensors = item
+                        if item_meta.get("close"):
+                            break
+                else:
+                    item = await stream.recv()
+                    if item is None:
+                        break
+                    item_meta, item_tensors = item
+                    if item_meta.get("close"):
+                        break
+                pos = int(item_meta["pos"])
+                hidden = item_tensors[0]
+                prompts = item_tensors[1] if len(item_tensors) > 1 else None
+                out = await loop.run_in_executor(
+                    None, lambda: self.backend.inference_step(sid, hidden, pos,
+                                                              prompts))
+                out_cpu = out.cpu()
+                if push_to is not None:
+                    await self._push_downstream(push_to, pos, out_cpu,
+                                                item_meta)
+                if not quiet:
+                    await stream.send({"pos": pos, "step": item_meta.get("step")},
+                                      [out_cpu])
+                # quiet spans send nothing: during push-mode decode the client
+                # reads outputs from the LAST span only (push_only_downstream,
+                # ref inference_session.py:178-196)
+        finally:
+            await loop.run_in_executor(None,
+                                       lambda: self.backend.close_session(sid))
+            self._push_q.pop(sid, None)
+            try:
+                await stream.send_end()
+            except Exception:
+                pass
+
+    async def _push_downstream(self, push_to, pos: int, hidden: torch.Tensor,
+                               item_meta: dict) -> None:
+        host, port, down_sid = push_to[0], int(push_to[1]), push_to[2]
+        key = (host, port)
+        if key not in self._peers:
+            self._peers[key] = RpcClient(host, port)
+        try:
+            await self._peers[key].call(
+                "rpc_push", {"session_id": down_sid, "pos": pos,
+                             "step": item_meta.get("step")}, [hidden],
+                timeout=30)
+        except Exception as e:  # noqa: BLE001 — client will fall back
+            logger.warning("s2s push to %s failed: %s", push_to, e)

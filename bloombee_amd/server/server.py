@@ -1,0 +1,158 @@
+"""Worker orchestrator: DHT + RPC handler + backend + announcer.
+
+Parity target: reference Server / ModuleContainer / ModuleAnnouncerThread
+(server/server.py:103-1007). One process per worker: an asyncio loop hosts
+the RPC endpoint and DHT node; compute runs on the backend's worker thread.
+"""
+from __future__ import annotations
+
+import asyncio
+import threading
+import time
+import uuid
+from typing import List, Optional, Sequence, Tuple
+
+import torch
+
+from bloombee_amd.data_structures import (ServerInfo, ServerState,
+                                          declare_active_modules,
+                                          get_remote_module_infos, module_uids)
+from bloombee_amd.models.base import ModelConfig, resolve_config
+from bloombee_amd.net.dht import Dht
+from bloombee_amd.net.rpc import RpcServer
+from bloombee_amd.server.backend import StackBackend
+from bloombee_amd.server.block_selection import choose_best_blocks
+from bloombee_amd.server.handler import ConnectionHandler
+from bloombee_amd.utils.logging import get_logger
+
+logger = get_logger(__name__)
+
+
+class Server:
+    def __init__(
+        self,
+        model: str | ModelConfig,
+        *,
+        initial_peers: Optional[Sequence[Tuple[str, int]]] = None,
+        host: str = "127.0.0.1",
+        port: int = 0,
+        dht_port: int = 0,
+        block_indices: Optional[Tuple[int, int]] = None,
+        num_blocks: Optional[int] = None,
+        device: str = "cpu",
+        seed: int = 0,
+        kv_max_tokens: int = 1 << 18,
+        update_period: float = 30.0,
+        expiration: Optional[float] = None,
+        checkpoint_dir: Optional[str] = None,
+        throughput: float = 1.0,
+        model_name: Optional[str] = None,
+    ):
+        self.config = model if isinstance(model, ModelConfig) else resolve_config(model)
+        self.model_name = model_name or (model if isinstance(model, str)
+                                         else self.config.model_type)
+        self.device = device
+        self.peer_id = uuid.uuid4().hex[:16]
+        self.update_period = update_period
+        self.expiration = expiration or max(60.0, 2 * update_period)
+        self.throughput = throughput
+        self.host = host
+        self.dht = Dht(initial_peers=list(initial_peers or []), host=host,
+                       port=dht_port)
+
+        L = self.config.num_hidden_layers
+        if block_indices is None:
+            nb = num_blocks or L
+            infos = get_remote_module_infos(
+                self.dht, module_uids(self.model_name, L))
+            blocks = choose_best_blocks(nb, infos)
+            block_indices = (blocks[0], blocks[-1] + 1)
+        self.block_range = block_indices
+        self.uids = [f"{self.model_name}.{i}"
+                     for i in range(block_indices[0], block_indices[1])]
+
+        self.backend = StackBackend(self.config, block_indices[0],
+                                    block_indices[1], device=device, seed=seed,
+                                    kv_max_tokens=kv_max_tokens,
+                                    checkpoint_dir=checkpoint_dir)
+        self.rpc = RpcServer(host, port)
+        self.handler = ConnectionHandler(self.backend, self.rpc)
+        self._loop: Optional[asyncio.AbstractEventLoop] = None
+        self._thread: Optional[threading.Thread] = None
+        self._announcer: Optional[threading.Thread] = None
+        self._stop = threading.Event()
+        self.endpoint: Optional[Tuple[str, int]] = None
+
+    # ------------------------------------------------------------------
+    def _server_info(self) -> ServerInfo:
+        return ServerInfo(
+            state=ServerState.ONLINE,
+            host=self.endpoint[0], port=self.endpoint[1],
+            throughput=self.throughput,
+            cache_tokens_left=self.backend.kv_pool.tokens_left,
+            torch_dtype=self.config.torch_dtype,
+            device=self.device,
+            start_block=self.block_range[0], end_block=self.block_range[1],
+        )
+
+    def _announce_loop(self):
+        while not self._stop.is_set():
+            try:
+                declare_active_modules(self.dht, self.uids, self.peer_id,
+                                       self._server_info(),
+                                       time.time() + self.expiration)
+            except Exception as e:  # noqa: BLE001
+                logger.warning("announce failed: %s", e)
+            self._stop.wait(self.update_period)
+
+    # ------------------------------------------------------------------
+    def run_in_background(self) -> Tuple[str, int]:
+        """Start the RPC endpoint + announcer; returns the block endpoint."""
+        started = threading.Event()
+
+        def run():
+            loop = asyncio.new_event_loop()
+            asyncio.set_event_loop(loop)
+            self._loop = loop
+            self.endpoint = loop.run_until_complete(self.rpc.start())
+            started.set()
+            loop.run_forever()
+
+        self._thread = threading.Thread(target=run, daemon=True,
+                                        name=f"server-{self.peer_id[:6]}")
+        self._thread.start()
+        if not started.wait(timeout=30):
+            raise RuntimeError("server failed to start")
+        declare_active_modules(self.dht, self.uids, self.peer_id,
+                               self._server_info(), time.time() + self.expiration)
+        self._announcer = threading.Thread(target=self._announce_loop,
+                                           daemon=True, name="announcer")
+        self._announcer.start()
+        logger.info("server %s serving %s.[%d:%d) at %s", self.peer_id,
+                    self.model_name, *self.block_range, self.endpoint)
+        return self.endpoint
+
+    def run(self):
+        """Foreground serve (CLI entry)."""
+        self.run_in_background()
+        try:
+            while not self._stop.is_set():
+                time.sleep(1.0)
+        except KeyboardInterrupt:
+            pass
+        finally:
+            self.shutdown()
+
+    def shutdown(self):
+        self._stop.set()
+        if self._loop is not None:
+            fut = asyncio.run_coroutine_threadsafe(self.rpc.stop(), self._loop)
+            try:
+                fut.result(timeout=5)
+            except Exception:
+                pass
+            self._loop.call_soon_threadsafe(self._loop.stop)
+        if self._thread is not None:
+            self._thread.join(timeout=5)
+        self.backend.shutdown()
+        self.dht.shutdown()

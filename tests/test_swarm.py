@@ -1,0 +1,141 @@
+"""Loopback swarm integration: 2 servers + DHT + client, exact-match decode
+vs the single-process LocalEngine (mirrors the reference's
+test_full_model.py / test_chained_calls.py tier, but self-contained — no
+out-of-band swarm bootstrap required)."""
+import time
+
+import pytest
+import torch
+
+from bloombee_amd.client import ClientConfig
+from bloombee_amd.engine import LocalEngine
+from bloombee_amd.models.auto import AutoDistributedModelForCausalLM
+from bloombee_amd.net.dht import Dht
+from bloombee_amd.server import Server
+
+MODEL = "llama-tiny"
+SEED = 0
+
+
+@pytest.fixture(scope="module")
+def swarm():
+    boot = Dht()
+    servers = []
+    s1 = Server(MODEL, initial_peers=[boot.endpoint], block_indices=(0, 2),
+                device="cpu", seed=SEED, kv_max_tokens=1 << 14,
+                update_period=2.0)
+    s1.run_in_background()
+    servers.append(s1)
+    s2 = Server(MODEL, initial_peers=[boot.endpoint], block_indices=(2, 4),
+                device="cpu", seed=SEED, kv_max_tokens=1 << 14,
+                update_period=2.0)
+    s2.run_in_background()
+    servers.append(s2)
+    yield boot, servers
+    for s in servers:
+        s.shutdown()
+    boot.shutdown()
+
+
+def _local_tokens(prompt, new_tokens):
+    eng = LocalEngine(MODEL, device="cpu", seed=SEED, kv_max_tokens=1 << 14)
+    kv = eng.kv_pool.allocate(prompt.shape[0], 64)
+    toks = [eng.prefill(prompt, kv)]
+    for _ in range(new_tokens - 1):
+        toks.append(eng.decode_step(toks[-1], kv))
+    kv.close()
+    return torch.stack(toks, 1)
+
+
+def _make_model(boot, **kw):
+    cfg = ClientConfig(initial_peers=[boot.endpoint], **kw)
+    return AutoDistributedModelForCausalLM.from_pretrained(
+        MODEL, client_config=cfg, seed=SEED)
+
+
+def test_swarm_greedy_matches_local(swarm):
+    boot, _ = swarm
+    model = _make_model(boot)
+    gen = torch.Generator().manual_seed(5)
+    prompt = torch.randint(0, 1000, (2, 7), generator=gen)
+    out = model.generate(prompt, max_new_tokens=6)
+    expect = _local_tokens(prompt, 6)
+    assert torch.equal(out[:, 7:], expect), (out[:, 7:], expect)
+    model.remote.manager.shutdown()
+
+
+def test_swarm_greedy_no_push_matches_local(swarm):
+    boot, _ = swarm
+    model = _make_model(boot, use_server_to_server=False)
+    gen = torch.Generator().manual_seed(5)
+    prompt = torch.randint(0, 1000, (2, 7), generator=gen)
+    out = model.generate(prompt, max_new_tokens=6)
+    expect = _local_tokens(prompt, 6)
+    assert torch.equal(out[:, 7:], expect)
+    model.remote.manager.shutdown()
+
+
+def test_forward_backward_grads(swarm):
+    boot, _ = swarm
+    model = _make_model(boot)
+    model.transformer.pre_seq_len = 0
+    gen = torch.Generator().manual_seed(6)
+    ids = torch.randint(0, 1000, (1, 5), generator=gen)
+    h = model.transformer.embed(ids).detach().requires_grad_(True)
+    out = model.transformer.remote(h)
+    loss = out.float().square().mean()
+    loss.backward()
+    assert h.grad is not None and torch.isfinite(h.grad).all()
+    assert h.grad.abs().sum() > 0
+    model.remote.manager.shutdown()
+
+
+def test_failover_rebuilds_chain():
+    """A mid-chain server dying mid-session must trigger history replay on a
+    replacement (ref inference_session.py:802-831)."""
+    boot = Dht()
+    s1 = Server(MODEL, initial_peers=[boot.endpoint], block_indices=(0, 2),
+                device="cpu", seed=SEED, kv_max_tokens=1 << 14, update_period=1.0)
+    s1.run_in_background()
+    dying = Server(MODEL, initial_peers=[boot.endpoint], block_indices=(2, 4),
+                   device="cpu", seed=SEED, kv_max_tokens=1 << 14,
+                   update_period=1.0)
+    dying.run_in_background()
+    backup = Server(MODEL, initial_peers=[boot.endpoint], block_indices=(2, 4),
+                    device="cpu", seed=SEED, kv_max_tokens=1 << 14,
+                    update_period=1.0)
+    try:
+        cfg = ClientConfig(initial_peers=[boot.endpoint],
+                           blocked_servers=[backup.peer_id],
+                           min_backoff=0.1, step_timeout=10.0)
+        model = AutoDistributedModelForCausalLM.from_pretrained(
+            MODEL, client_config=cfg, seed=SEED)
+        gen = torch.Generator().manual_seed(5)
+        prompt = torch.randint(0, 1000, (1, 7), generator=gen)
+        session = model.remote.inference_session(40)
+        h = model.embed(prompt)
+        out1 = session.step(h)
+        tok = model.lm_head(model.final_norm(out1[:, -1:]))[:, -1].argmax(-1)
+
+        # kill the tail server, bring the backup online, unblock it
+        dying.shutdown()
+        backup.run_in_background()
+        model.remote.manager.config.blocked_servers = None
+        time.sleep(0.2)
+        model.remote.manager.update()
+
+        toks = [tok]
+        for _ in range(3):
+            h = model.embed(toks[-1].view(1, 1))
+            out = session.step(h)
+            toks.append(model.lm_head(model.final_norm(out[:, -1:]))[:, -1].argmax(-1))
+        session.close()
+
+        expect = _local_tokens(prompt, 4)
+        got = torch.stack(toks, 1)
+        assert torch.equal(got, expect), (got, expect)
+        model.remote.manager.shutdown()
+    finally:
+        for s in (s1, backup):
+            s.shutdown()
+        boot.shutdown()
