@@ -1,0 +1,73 @@
+"""Worker entrypoint (parity: reference cli/run_server.py).
+
+    python -m bloombee_amd.cli.run_server llama-3-8b \
+        --initial-peers 127.0.0.1:31337 --block-indices 0:16 --device cuda:0
+"""
+from __future__ import annotations
+
+import argparse
+
+from bloombee_amd.server import Server
+from bloombee_amd.utils.logging import get_logger
+
+logger = get_logger(__name__)
+
+
+def parse_endpoint(s: str):
+    host, port = s.rsplit(":", 1)
+    return host, int(port)
+
+
+def main():
+    ap = argparse.ArgumentParser(description=__doc__)
+    ap.add_argument("model", help="model preset name or local checkpoint dir")
+    ap.add_argument("--initial-peers", nargs="*", default=[])
+    ap.add_argument("--host", default="127.0.0.1")
+    ap.add_argument("--port", type=int, default=0)
+    ap.add_argument("--dht-port", type=int, default=0)
+    ap.add_argument("--block-indices", default=None,
+                    help="start:end block range (default: auto-select)")
+    ap.add_argument("--num-blocks", type=int, default=None)
+    ap.add_argument("--device", default=None,
+                    help="cuda:0 / cpu (default: cuda if available)")
+    ap.add_argument("--seed", type=int, default=0)
+    ap.add_argument("--attn-cache-tokens", type=int, default=1 << 18)
+    ap.add_argument("--update-period", type=float, default=30.0)
+    ap.add_argument("--checkpoint-dir", default=None,
+                    help="per-block .npy weight dir (random init if absent)")
+    ap.add_argument("--throughput", type=float, default=None,
+                    help="announced rps (default: measured)")
+    args = ap.parse_args()
+
+    import torch
+
+    device = args.device or ("cuda:0" if torch.cuda.is_available() else "cpu")
+    block_indices = None
+    if args.block_indices:
+        a, b = args.block_indices.split(":")
+        block_indices = (int(a), int(b))
+
+    throughput = args.throughput
+    if throughput is None:
+        from bloombee_amd.models.base import resolve_config
+        from bloombee_amd.server.throughput import get_server_throughput
+
+        cfg = resolve_config(args.model)
+        nb = (block_indices[1] - block_indices[0]) if block_indices \
+            else (args.num_blocks or cfg.num_hidden_layers)
+        throughput = get_server_throughput(cfg, device, nb)["throughput"]
+
+    server = Server(
+        args.model,
+        initial_peers=[parse_endpoint(p) for p in args.initial_peers],
+        host=args.host, port=args.port, dht_port=args.dht_port,
+        block_indices=block_indices, num_blocks=args.num_blocks,
+        device=device, seed=args.seed, kv_max_tokens=args.attn_cache_tokens,
+        update_period=args.update_period, checkpoint_dir=args.checkpoint_dir,
+        throughput=throughput,
+    )
+    server.run()
+
+
+if __name__ == "__main__":
+    main()

@@ -85,9 +85,17 @@ class LocalEngine:
         self.device = torch.device(device)
         gen = torch.Generator().manual_seed(seed)
         dt = cfg.dtype
+        # family head shape: bloom/falcon use LayerNorm heads; bloom also
+        # layer-norms the embedding output (HF word_embeddings_layernorm)
+        self.embed_ln = cfg.model_type == "bloom"
+        self.ln_final = cfg.model_type in ("bloom", "falcon")
         self.embed = (torch.randn(cfg.vocab_size, cfg.hidden_size, generator=gen)
                       .mul_(0.02).to(dt).to(device))
         self.final_norm_w = torch.ones(cfg.hidden_size, dtype=dt, device=device)
+        self.final_norm_b = torch.zeros(cfg.hidden_size, dtype=dt, device=device)
+        if self.embed_ln:
+            self.embed_ln_w = torch.ones(cfg.hidden_size, dtype=dt, device=device)
+            self.embed_ln_b = torch.zeros(cfg.hidden_size, dtype=dt, device=device)
         if cfg.tie_word_embeddings:
             self.lm_head_w = self.embed
         else:
@@ -99,8 +107,19 @@ class LocalEngine:
     @torch.no_grad()
     def logits_for(self, hidden_last: torch.Tensor) -> torch.Tensor:
         """hidden_last: (B, H) -> (B, V)"""
-        y = ops.rms_norm(hidden_last, self.final_norm_w, self.config.rms_norm_eps)
+        if self.ln_final:
+            y = ops.layer_norm(hidden_last, self.final_norm_w, self.final_norm_b,
+                               self.config.layer_norm_epsilon)
+        else:
+            y = ops.rms_norm(hidden_last, self.final_norm_w, self.config.rms_norm_eps)
         return F.linear(y, self.lm_head_w)
+
+    def _embed(self, input_ids: torch.Tensor) -> torch.Tensor:
+        h = F.embedding(input_ids.to(self.device), self.embed)
+        if self.embed_ln:
+            h = ops.layer_norm(h, self.embed_ln_w, self.embed_ln_b,
+                               self.config.layer_norm_epsilon)
+        return h
 
     @torch.no_grad()
     def prefill(self, input_ids: torch.Tensor, kv: SessionHandle) -> torch.Tensor:
@@ -109,7 +128,7 @@ class LocalEngine:
         start = torch.tensor([s.l_spec for s in kv.seqs], dtype=torch.int32,
                              device=self.device)
         kv.extend(T)
-        hidden = F.embedding(input_ids.to(self.device), self.embed)
+        hidden = self._embed(input_ids)
         hidden = self.stack.forward_inference(hidden, kv, start)
         return self.logits_for(hidden[:, -1]).argmax(-1)
 
@@ -119,7 +138,7 @@ class LocalEngine:
         start = torch.tensor([s.l_spec for s in kv.seqs], dtype=torch.int32,
                              device=self.device)
         kv.extend(1)
-        hidden = F.embedding(input_ids.view(-1, 1).to(self.device), self.embed)
+        hidden = self._embed(input_ids.view(-1, 1))
         hidden = self.stack.forward_inference(hidden, kv, start)
         return self.logits_for(hidden[:, -1]).argmax(-1)
 

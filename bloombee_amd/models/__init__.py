@@ -8,3 +8,4 @@ from bloombee_amd.models.base import (  # noqa: F401
 
 # Import families for registration side effects.
 import bloombee_amd.models.llama  # noqa: F401,E402
+import bloombee_amd.models.bloom  # noqa: F401,E402

@@ -6,6 +6,7 @@ instead of silently falling back to eager PyTorch.
 """
 from bloombee_amd.ops.interface import (  # noqa: F401
     HAVE_HIP_OPS,
+    alibi_slopes_for,
     attn_decode,
     attn_paged,
     attn_paged_qkv,

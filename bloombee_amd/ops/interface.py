@@ -14,6 +14,7 @@ from typing import Optional, Tuple
 import torch
 
 from bloombee_amd.ops import reference as ref
+from bloombee_amd.ops.reference import alibi_slopes as alibi_slopes_for  # noqa: F401
 from bloombee_amd.ops.reference import rope_cos_sin  # noqa: F401  (host-side)
 from bloombee_amd.utils.logging import get_logger
 
@@ -199,9 +200,20 @@ def attn_paged_qkv(qkv, Hq: int, Hkv: int, k_pages, v_pages, page_table,
 
 
 def attn_paged(q, k_pages, v_pages, page_table, q_start, scale=None, window: int = 0,
-               tree_mask=None):
+               tree_mask=None, alibi_slopes=None):
     """Unified entry: picks decode vs prefill kernel by Tq."""
     Tq = q.shape[2]
+    if alibi_slopes is not None:
+        if _on_gpu(q):
+            _require_ext()
+            raise NotImplementedError(
+                "alibi attention has no gfx950 kernel yet — bloom GPU serving "
+                "pending; CPU path is complete")
+        if scale is None:
+            scale = 1.0 / math.sqrt(q.shape[-1])
+        return ref.attn_paged(q, k_pages, v_pages, page_table, q_start, scale,
+                              alibi_slopes=alibi_slopes,
+                              sliding_window=window if window > 0 else None)
     if tree_mask is not None:
         # tree-attention (spec decode verify) — reference path for now
         if scale is None:

@@ -203,6 +203,7 @@ def attn_paged(
     scale: Optional[float] = None,
     tree_mask: Optional[torch.Tensor] = None,
     sliding_window: Optional[int] = None,
+    alibi_slopes: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
     """Causal attention of q against the paged KV cache.
 
@@ -235,6 +236,11 @@ def attn_paged(
         kg = kf.repeat_interleave(G, dim=0)
         vg = vf.repeat_interleave(G, dim=0)
         scores = torch.einsum("htd,hcd->htc", qf, kg) * scale
+        if alibi_slopes is not None:
+            # ALiBi (bloom family): softmax-shift-invariant absolute form
+            # bias[h, :, j] = slope_h * j (HF bloom build_alibi_tensor)
+            scores = scores + alibi_slopes.float().view(-1, 1, 1) * \
+                torch.arange(ctx, dtype=torch.float32).view(1, 1, ctx)
         pos_q = torch.arange(s, s + Tq).unsqueeze(1)  # (Tq, 1)
         pos_k = torch.arange(ctx).unsqueeze(0)        # (1, ctx)
         mask = pos_k <= pos_q
@@ -282,3 +288,16 @@ def quant4_unpack(
     q = torch.stack([lo, hi], dim=-1).reshape(*packed.shape[:-1], -1)
     x = q * scale.float().unsqueeze(-1) + zero.float().unsqueeze(-1)
     return x.reshape(*packed.shape[:-2], -1).to(dtype)
+
+
+def alibi_slopes(n_heads: int) -> torch.Tensor:
+    """HF bloom build_alibi_tensor slopes (negative, per head)."""
+    import math as _m
+
+    closest = 2 ** _m.floor(_m.log2(n_heads))
+    base = 2.0 ** (-(2.0 ** -(_m.log2(closest) - 3)))
+    slopes = [base ** (i + 1) for i in range(closest)]
+    if closest < n_heads:
+        extra_base = 2.0 ** (-(2.0 ** -(_m.log2(2 * closest) - 3)))
+        slopes += [extra_base ** (2 * i + 1) for i in range(n_heads - closest)]
+    return -torch.tensor(slopes, dtype=torch.float32)

@@ -139,3 +139,37 @@ def test_failover_rebuilds_chain():
         for s in (s1, backup):
             s.shutdown()
         boot.shutdown()
+
+
+def test_bloom_swarm_matches_local():
+    """BASELINE.json config 1: bloom-family plumbing — 2 CPU workers over
+    loopback DHT, greedy decode exact-match vs local."""
+    boot = Dht()
+    s1 = Server("bloom-tiny", initial_peers=[boot.endpoint], block_indices=(0, 2),
+                device="cpu", seed=SEED, kv_max_tokens=1 << 14, update_period=5.0)
+    s2 = Server("bloom-tiny", initial_peers=[boot.endpoint], block_indices=(2, 4),
+                device="cpu", seed=SEED, kv_max_tokens=1 << 14, update_period=5.0)
+    s1.run_in_background()
+    s2.run_in_background()
+    try:
+        cfg = ClientConfig(initial_peers=[boot.endpoint])
+        model = AutoDistributedModelForCausalLM.from_pretrained(
+            "bloom-tiny", client_config=cfg, seed=SEED)
+        gen = torch.Generator().manual_seed(9)
+        prompt = torch.randint(0, 1000, (2, 6), generator=gen)
+        out = model.generate(prompt, max_new_tokens=5)
+
+        eng = LocalEngine("bloom-tiny", device="cpu", seed=SEED,
+                          kv_max_tokens=1 << 14)
+        kv = eng.kv_pool.allocate(2, 64)
+        toks = [eng.prefill(prompt, kv)]
+        for _ in range(4):
+            toks.append(eng.decode_step(toks[-1], kv))
+        kv.close()
+        expect = torch.stack(toks, 1)
+        assert torch.equal(out[:, 6:], expect), (out[:, 6:], expect)
+        model.remote.manager.shutdown()
+    finally:
+        s1.shutdown()
+        s2.shutdown()
+        boot.shutdown()
