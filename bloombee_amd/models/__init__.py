@@ -9,3 +9,6 @@ from bloombee_amd.models.base import (  # noqa: F401
 # Import families for registration side effects.
 import bloombee_amd.models.llama  # noqa: F401,E402
 import bloombee_amd.models.bloom  # noqa: F401,E402
+import bloombee_amd.models.falcon  # noqa: F401,E402
+import bloombee_amd.models.qwen3  # noqa: F401,E402
+import bloombee_amd.models.mixtral  # noqa: F401,E402

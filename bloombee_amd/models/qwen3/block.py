@@ -1,0 +1,94 @@
+"""Server-side Qwen3 block (parity: reference models/qwen3/block.py
+WrappedQwen3Block :18-181): LLaMA-style GQA block plus per-head RMS q/k norms
+applied on the raw QKV GEMM output before RoPE. head_dim is decoupled from
+hidden_size (Qwen3-8B: D=128 with Hq*D != hidden)."""
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+
+from bloombee_amd import ops
+from bloombee_amd.kv.paged import SessionHandle
+from bloombee_amd.models.base import ModelConfig
+from bloombee_amd.models.llama.block import LlamaBlock, RopeTables
+
+
+class Qwen3Block(LlamaBlock):
+    def __init__(self, config: ModelConfig, layer_index: int = 0,
+                 rope: Optional[RopeTables] = None):
+        super().__init__(config, layer_index, rope)
+        D = config.head_dim
+        dt = config.dtype
+        self.q_norm_w = torch.nn.Parameter(torch.empty(D, dtype=dt),
+                                           requires_grad=False)
+        self.k_norm_w = torch.nn.Parameter(torch.empty(D, dtype=dt),
+                                           requires_grad=False)
+
+    @torch.no_grad()
+    def forward_inference(self, hidden: torch.Tensor, kv: SessionHandle,
+                          start_pos: torch.Tensor,
+                          position_ids=None) -> torch.Tensor:
+        B, T, H = hidden.shape
+        Hq, Hkv, D = self.Hq, self.Hkv, self.D
+        cfg = self.config
+
+        x = ops.rms_norm(hidden, self.input_norm_w, cfg.rms_norm_eps)
+        qkv = ops.linear(x, self.qkv_w)
+        # per-head q/k RMS norm on the fused buffer (contiguous D-sized rows)
+        qk = qkv[..., :(Hq + Hkv) * D]
+        q_flat = qkv[..., :Hq * D].reshape(-1, D)
+        k_flat = qkv[..., Hq * D:(Hq + Hkv) * D].reshape(-1, D)
+        qkv[..., :Hq * D] = ops.rms_norm(q_flat, self.q_norm_w,
+                                         cfg.rms_norm_eps).view(B, T, Hq * D)
+        qkv[..., Hq * D:(Hq + Hkv) * D] = ops.rms_norm(
+            k_flat, self.k_norm_w, cfg.rms_norm_eps).view(B, T, Hkv * D)
+        cos, sin = self.rope.get(hidden.device)
+        kp = kv.k_pages(self.layer_index)
+        vp = kv.v_pages(self.layer_index)
+        pt = kv.page_table()
+        ops.rope_kv_write_(qkv, Hq, Hkv, cos, sin, position_ids, kp, vp, pt,
+                           start_pos)
+        attn = ops.attn_paged_qkv(qkv, Hq, Hkv, kp, vp, pt, start_pos, self.scale)
+        a = ops.linear(attn, self.o_w)
+        h2, y = ops.rms_norm_residual(a, hidden, self.post_norm_w, cfg.rms_norm_eps)
+        return ops.linear(ops.swiglu(ops.linear(y, self.gate_up_w)), self.down_w,
+                          residual=h2)
+
+    def forward_train(self, hidden: torch.Tensor, start_pos: int = 0) -> torch.Tensor:
+        B, T, H = hidden.shape
+        Hq, Hkv, D = self.Hq, self.Hkv, self.D
+        cfg = self.config
+        G = Hq // Hkv
+
+        def rms(x, w):
+            xf = x.float()
+            return (xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True)
+                                     + cfg.rms_norm_eps)).to(x.dtype) * w
+
+        x = rms(hidden, self.input_norm_w)
+        qkv = torch.nn.functional.linear(x, self.qkv_w)
+        qkv = qkv.view(B, T, Hq + 2 * Hkv, D).permute(0, 2, 1, 3)
+        q, k, v = qkv.split([Hq, Hkv, Hkv], dim=1)
+        q = rms(q, self.q_norm_w)
+        k = rms(k, self.k_norm_w)
+        cos, sin = self.rope.get(hidden.device)
+        pos = torch.arange(start_pos, start_pos + T).view(1, T).expand(B, T)
+        from bloombee_amd.ops import reference as refops
+        q, k = refops.rope_apply(q, k, cos, sin, pos)
+        k = k.repeat_interleave(G, dim=1)
+        v = v.repeat_interleave(G, dim=1)
+        scores = torch.matmul(q.float(), k.float().transpose(-1, -2)) * self.scale
+        mask = torch.ones(T, T, dtype=torch.bool).tril()
+        scores = scores.masked_fill(~mask.to(scores.device), float("-inf"))
+        p = torch.softmax(scores, dim=-1)
+        attn = torch.matmul(p, v.float()).to(hidden.dtype)
+        attn = attn.permute(0, 2, 1, 3).reshape(B, T, Hq * D)
+        h2 = hidden + torch.nn.functional.linear(attn, self.o_w)
+        y = rms(h2, self.post_norm_w)
+        gu = torch.nn.functional.linear(y, self.gate_up_w)
+        g, u = gu.split([self.I, self.I], dim=-1)
+        m = torch.nn.functional.linear(
+            torch.nn.functional.silu(g.float()).to(u.dtype) * u, self.down_w)
+        return h2 + m

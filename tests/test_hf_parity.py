@@ -1,0 +1,185 @@
+"""Numerical parity vs HuggingFace transformers blocks with copied weights
+(the reference's strongest parity tier: test_optimized_layers.py /
+test_qwen3_block_parity.py — random weights, fp32, atol 1e-4 class)."""
+import pytest
+import torch
+
+from bloombee_amd.models.base import resolve_config
+
+
+def _mk_block(model, seed=3):
+    from bloombee_amd.engine import BlockStack
+
+    cfg = resolve_config(model)
+    cfg.torch_dtype = "float32"
+    stack = BlockStack(cfg, 0, 1, device="cpu", seed=seed)
+    return cfg, stack.blocks[0], stack
+
+
+def test_llama_block_matches_hf():
+    from transformers.models.llama.configuration_llama import LlamaConfig as HFCfg
+    from transformers.models.llama.modeling_llama import (LlamaDecoderLayer,
+                                                          LlamaRotaryEmbedding)
+
+    cfg, blk, stack = _mk_block("llama-tiny")
+    hf_cfg = HFCfg(
+        hidden_size=cfg.hidden_size, intermediate_size=cfg.intermediate_size,
+        num_attention_heads=cfg.num_attention_heads,
+        num_key_value_heads=cfg.num_key_value_heads,
+        head_dim=cfg.head_dim, rms_norm_eps=cfg.rms_norm_eps,
+        rope_theta=cfg.rope_theta, attention_bias=False, attention_dropout=0.0,
+        max_position_embeddings=cfg.max_position_embeddings,
+        vocab_size=cfg.vocab_size, attn_implementation="eager",
+    )
+    layer = LlamaDecoderLayer(hf_cfg, layer_idx=0).eval()
+    Hq, Hkv, D = cfg.num_attention_heads, cfg.num_key_value_heads, cfg.head_dim
+    with torch.no_grad():
+        layer.input_layernorm.weight.copy_(blk.input_norm_w)
+        layer.post_attention_layernorm.weight.copy_(blk.post_norm_w)
+        qkv = blk.qkv_w
+        layer.self_attn.q_proj.weight.copy_(qkv[:Hq * D])
+        layer.self_attn.k_proj.weight.copy_(qkv[Hq * D:(Hq + Hkv) * D])
+        layer.self_attn.v_proj.weight.copy_(qkv[(Hq + Hkv) * D:])
+        layer.self_attn.o_proj.weight.copy_(blk.o_w)
+        I = cfg.intermediate_size
+        layer.mlp.gate_proj.weight.copy_(blk.gate_up_w[:I])
+        layer.mlp.up_proj.weight.copy_(blk.gate_up_w[I:])
+        layer.mlp.down_proj.weight.copy_(blk.down_w)
+
+    torch.manual_seed(0)
+    B, T = 2, 11
+    h = torch.randn(B, T, cfg.hidden_size) * 0.3
+    rotary = LlamaRotaryEmbedding(hf_cfg)
+    pos = torch.arange(T).unsqueeze(0).expand(B, T)
+    cos_sin = rotary(h, pos)
+    mask = torch.full((T, T), float("-inf")).triu(1).view(1, 1, T, T).expand(B, 1, T, T)
+    with torch.no_grad():
+        hf_out = layer(h, attention_mask=mask, position_ids=pos,
+                       position_embeddings=cos_sin)
+        if isinstance(hf_out, tuple):
+            hf_out = hf_out[0]
+
+    kv = stack.make_kv(1024).allocate(B, 64)
+    kv.extend(T)
+    ours = blk.forward_inference(h.clone(), kv,
+                                 torch.zeros(B, dtype=torch.int32))
+    kv.close()
+    diff = (ours - hf_out).abs().max().item()
+    assert diff < 1e-4, f"llama block vs HF: max diff {diff}"
+
+
+def test_qwen3_block_matches_hf():
+    from transformers.models.qwen3.configuration_qwen3 import Qwen3Config as HFCfg
+    from transformers.models.qwen3.modeling_qwen3 import (Qwen3DecoderLayer,
+                                                          Qwen3RotaryEmbedding)
+
+    cfg, blk, stack = _mk_block("qwen3-tiny")
+    hf_cfg = HFCfg(
+        hidden_size=cfg.hidden_size, intermediate_size=cfg.intermediate_size,
+        num_attention_heads=cfg.num_attention_heads,
+        num_key_value_heads=cfg.num_key_value_heads,
+        head_dim=cfg.head_dim, rms_norm_eps=cfg.rms_norm_eps,
+        rope_theta=cfg.rope_theta, attention_bias=False, attention_dropout=0.0,
+        max_position_embeddings=cfg.max_position_embeddings,
+        vocab_size=cfg.vocab_size, attn_implementation="eager",
+    )
+    layer = Qwen3DecoderLayer(hf_cfg, layer_idx=0).eval()
+    Hq, Hkv, D = cfg.num_attention_heads, cfg.num_key_value_heads, cfg.head_dim
+    with torch.no_grad():
+        layer.input_layernorm.weight.copy_(blk.input_norm_w)
+        layer.post_attention_layernorm.weight.copy_(blk.post_norm_w)
+        qkv = blk.qkv_w
+        layer.self_attn.q_proj.weight.copy_(qkv[:Hq * D])
+        layer.self_attn.k_proj.weight.copy_(qkv[Hq * D:(Hq + Hkv) * D])
+        layer.self_attn.v_proj.weight.copy_(qkv[(Hq + Hkv) * D:])
+        layer.self_attn.o_proj.weight.copy_(blk.o_w)
+        layer.self_attn.q_norm.weight.copy_(blk.q_norm_w)
+        layer.self_attn.k_norm.weight.copy_(blk.k_norm_w)
+        I = cfg.intermediate_size
+        layer.mlp.gate_proj.weight.copy_(blk.gate_up_w[:I])
+        layer.mlp.up_proj.weight.copy_(blk.gate_up_w[I:])
+        layer.mlp.down_proj.weight.copy_(blk.down_w)
+
+    torch.manual_seed(0)
+    B, T = 2, 9
+    h = torch.randn(B, T, cfg.hidden_size) * 0.3
+    rotary = Qwen3RotaryEmbedding(hf_cfg)
+    pos = torch.arange(T).unsqueeze(0).expand(B, T)
+    cos_sin = rotary(h, pos)
+    mask = torch.full((T, T), float("-inf")).triu(1).view(1, 1, T, T).expand(B, 1, T, T)
+    with torch.no_grad():
+        hf_out = layer(h, attention_mask=mask, position_ids=pos,
+                       position_embeddings=cos_sin)
+        if isinstance(hf_out, tuple):
+            hf_out = hf_out[0]
+
+    kv = stack.make_kv(1024).allocate(B, 64)
+    kv.extend(T)
+    ours = blk.forward_inference(h.clone(), kv,
+                                 torch.zeros(B, dtype=torch.int32))
+    kv.close()
+    diff = (ours - hf_out).abs().max().item()
+    assert diff < 1e-4, f"qwen3 block vs HF: max diff {diff}"
+
+
+def test_mixtral_block_matches_hf():
+    from transformers.models.mixtral.configuration_mixtral import \
+        MixtralConfig as HFCfg
+    from transformers.models.mixtral.modeling_mixtral import (
+        MixtralDecoderLayer, MixtralRotaryEmbedding)
+
+    cfg, blk, stack = _mk_block("mixtral-tiny")
+    hf_cfg = HFCfg(
+        hidden_size=cfg.hidden_size, intermediate_size=cfg.intermediate_size,
+        num_attention_heads=cfg.num_attention_heads,
+        num_key_value_heads=cfg.num_key_value_heads,
+        rms_norm_eps=cfg.rms_norm_eps, rope_theta=cfg.rope_theta,
+        max_position_embeddings=cfg.max_position_embeddings,
+        vocab_size=cfg.vocab_size, num_local_experts=blk.E,
+        num_experts_per_tok=blk.topk, attn_implementation="eager",
+        attention_dropout=0.0,
+    )
+    layer = MixtralDecoderLayer(hf_cfg, layer_idx=0).eval()
+    Hq, Hkv, D = cfg.num_attention_heads, cfg.num_key_value_heads, cfg.head_dim
+    I = cfg.intermediate_size
+    with torch.no_grad():
+        layer.input_layernorm.weight.copy_(blk.input_norm_w)
+        layer.post_attention_layernorm.weight.copy_(blk.post_norm_w)
+        qkv = blk.qkv_w
+        layer.self_attn.q_proj.weight.copy_(qkv[:Hq * D])
+        layer.self_attn.k_proj.weight.copy_(qkv[Hq * D:(Hq + Hkv) * D])
+        layer.self_attn.v_proj.weight.copy_(qkv[(Hq + Hkv) * D:])
+        layer.self_attn.o_proj.weight.copy_(blk.o_w)
+        mlp = layer.block_sparse_moe if hasattr(layer, "block_sparse_moe") else layer.mlp
+        (mlp.gate if hasattr(mlp, "gate") else mlp.router).weight.copy_(blk.router_w)
+        experts = mlp.experts
+        if hasattr(experts, "gate_up_proj"):  # transformers 5.x batched experts
+            experts.gate_up_proj.copy_(blk.expert_gate_up_w)
+            experts.down_proj.copy_(blk.expert_down_w)
+        else:
+            for e in range(blk.E):
+                ex = experts[e]
+                ex.w1.weight.copy_(blk.expert_gate_up_w[e][:I])
+                ex.w3.weight.copy_(blk.expert_gate_up_w[e][I:])
+                ex.w2.weight.copy_(blk.expert_down_w[e])
+
+    torch.manual_seed(0)
+    B, T = 2, 9
+    h = torch.randn(B, T, cfg.hidden_size) * 0.3
+    rotary = MixtralRotaryEmbedding(hf_cfg)
+    pos = torch.arange(T).unsqueeze(0).expand(B, T)
+    cos_sin = rotary(h, pos)
+    mask = torch.full((T, T), float("-inf")).triu(1).view(1, 1, T, T).expand(B, 1, T, T)
+    with torch.no_grad():
+        hf_out = layer(h, attention_mask=mask, position_ids=pos,
+                       position_embeddings=cos_sin)
+        if isinstance(hf_out, tuple):
+            hf_out = hf_out[0]
+
+    kv = stack.make_kv(1024).allocate(B, 64)
+    kv.extend(T)
+    ours = blk.forward_inference(h.clone(), kv,
+                                 torch.zeros(B, dtype=torch.int32))
+    kv.close()
+    diff = (ours - hf_out).abs().max().item()
+    assert diff < 1e-4, f"mixtral block vs HF: max diff {diff}"
