@@ -47,10 +47,16 @@ class BlockStack(torch.nn.Module):
         self.device = torch.device(device)
 
     def make_kv(self, max_tokens: int) -> PagedKVCache:
+        # per-layer KV geometry hook (gemma-4 heterogeneous head_dim/kv heads)
+        if hasattr(self.config, "kv_geometry"):
+            hkv, dims = self.config.kv_geometry(self.start, self.end)
+        else:
+            hkv = self.config.num_key_value_heads
+            dims = self.config.head_dim
         return PagedKVCache(
             num_layers=len(self.blocks),
-            num_kv_heads=self.config.num_key_value_heads,
-            head_dim=self.config.head_dim,
+            num_kv_heads=hkv,
+            head_dim=dims,
             max_tokens=max_tokens,
             device=self.device,
             dtype=self.config.dtype,
@@ -89,6 +95,8 @@ class LocalEngine:
         # layer-norms the embedding output (HF word_embeddings_layernorm)
         self.embed_ln = cfg.model_type == "bloom"
         self.ln_final = cfg.model_type in ("bloom", "falcon")
+        self.embed_scale = (cfg.hidden_size ** 0.5
+                            if cfg.model_type == "gemma4" else 1.0)
         self.embed = (torch.randn(cfg.vocab_size, cfg.hidden_size, generator=gen)
                       .mul_(0.02).to(dt).to(device))
         self.final_norm_w = torch.ones(cfg.hidden_size, dtype=dt, device=device)
@@ -116,6 +124,8 @@ class LocalEngine:
 
     def _embed(self, input_ids: torch.Tensor) -> torch.Tensor:
         h = F.embedding(input_ids.to(self.device), self.embed)
+        if self.embed_scale != 1.0:
+            h = h * self.embed_scale
         if self.embed_ln:
             h = ops.layer_norm(h, self.embed_ln_w, self.embed_ln_b,
                                self.config.layer_norm_epsilon)

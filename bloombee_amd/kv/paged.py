@@ -55,28 +55,44 @@ class PagedKVCache:
     def __init__(
         self,
         num_layers: int,
-        num_kv_heads: int,
-        head_dim: int,
+        num_kv_heads,
+        head_dim,
         *,
         page_size: int = 16,
         max_tokens: int = 1 << 20,
         device: torch.device | str = "cpu",
         dtype: torch.dtype = torch.bfloat16,
+        layout=None,
     ):
+        """num_kv_heads / head_dim may be ints (uniform) or per-layer lists —
+        gemma-4 style heterogeneous KV geometry (sliding layers D=256, full
+        layers D=512; ref server/backend.py:291-306 per-block KV descriptor
+        dispatch). `layout[l]` may be "kv" (default) or "k" — k-only layers
+        (gemma-4 attention_k_eq_v full-attention layers) allocate no V plane.
+        """
         self.num_layers = num_layers
-        self.num_kv_heads = num_kv_heads
-        self.head_dim = head_dim
+        as_list = (lambda v: list(v) if isinstance(v, (list, tuple))
+                   else [v] * num_layers)
+        self.num_kv_heads_per_layer = as_list(num_kv_heads)
+        self.head_dim_per_layer = as_list(head_dim)
+        self.layout = as_list(layout if layout is not None else "kv")
+        self.num_kv_heads = self.num_kv_heads_per_layer[0]
+        self.head_dim = self.head_dim_per_layer[0]
         self.page_size = page_size
         self.device = torch.device(device)
         self.dtype = dtype
         self.n_pages = max(1, max_tokens // page_size)
         self.max_tokens = self.n_pages * page_size
 
-        # One allocation; pool[l, 0] = K pages of layer l, pool[l, 1] = V.
-        self.pool = torch.zeros(
-            num_layers, 2, self.n_pages, num_kv_heads, page_size, head_dim,
-            device=self.device, dtype=dtype,
-        )
+        # Per-layer page planes; pools[l][0] = K pages, pools[l][1] = V (or
+        # aliased to K for "k" layout). A page id indexes every layer's plane.
+        self.pools = []
+        for l in range(num_layers):
+            hkv, d = self.num_kv_heads_per_layer[l], self.head_dim_per_layer[l]
+            k = torch.zeros(self.n_pages, hkv, page_size, d,
+                            device=self.device, dtype=dtype)
+            v = k if self.layout[l] == "k" else torch.zeros_like(k)
+            self.pools.append((k, v))
         self._free_pages: List[int] = list(range(self.n_pages - 1, -1, -1))
         self._reserved_tokens = 0  # admission-control reservation
         self._lock = threading.Condition()
@@ -93,7 +109,7 @@ class PagedKVCache:
 
     def k_pages(self, layer: int) -> torch.Tensor:
         """K pages, token-major: (n_pages, Hkv, P, D)."""
-        return self.pool[layer, 0]
+        return self.pools[layer][0]
 
     def v_pages(self, layer: int) -> torch.Tensor:
         """V pages, d-major: (n_pages, Hkv, D, P).
@@ -102,9 +118,15 @@ class PagedKVCache:
         attention kernel's P.V MFMA B-fragments (8 consecutive positions at a
         fixed d) are direct contiguous 16 B loads from HBM — no LDS transpose
         staging. Same bytes as the K slab (P*D*2), reinterpreted.
+
+        For "k"-layout layers (gemma-4 k==v aliasing) there is no V plane;
+        callers must read K instead (the block handles it).
         """
-        v = self.pool[layer, 1]
-        return v.view(self.n_pages, self.num_kv_heads, self.head_dim, self.page_size)
+        if self.layout[layer] == "k":
+            raise PagedKVError(f"layer {layer} is k-only (v aliases k)")
+        v = self.pools[layer][1]
+        hkv, d = self.num_kv_heads_per_layer[layer], self.head_dim_per_layer[layer]
+        return v.view(self.n_pages, hkv, d, self.page_size)
 
     # -- admission --------------------------------------------------------
     def allocate(

@@ -12,3 +12,4 @@ import bloombee_amd.models.bloom  # noqa: F401,E402
 import bloombee_amd.models.falcon  # noqa: F401,E402
 import bloombee_amd.models.qwen3  # noqa: F401,E402
 import bloombee_amd.models.mixtral  # noqa: F401,E402
+import bloombee_amd.models.gemma4  # noqa: F401,E402

@@ -7,14 +7,15 @@ import torch
 from bloombee_amd.engine import LocalEngine
 
 FAMILIES = ["llama-tiny", "bloom-tiny", "falcon-tiny", "qwen3-tiny",
-            "mixtral-tiny"]
+            "mixtral-tiny", "gemma4-tiny"]
 
 
 @pytest.mark.parametrize("model", FAMILIES)
 def test_decode_equals_prefill(model):
     torch.manual_seed(0)
     eng = LocalEngine(model, device="cpu", seed=1, kv_max_tokens=4096)
-    ids = torch.randint(0, 900, (2, 9), generator=torch.Generator().manual_seed(3))
+    ids = torch.randint(0, min(900, eng.config.vocab_size), (2, 9),
+                        generator=torch.Generator().manual_seed(3))
     # full prefill
     kv1 = eng.kv_pool.allocate(2, 64)
     t_full = eng.prefill(ids, kv1)
@@ -49,7 +50,8 @@ def test_train_path_matches_inference(model):
 @pytest.mark.parametrize("model", FAMILIES)
 def test_generate_runs(model):
     eng = LocalEngine(model, device="cpu", seed=0, kv_max_tokens=4096)
-    ids = torch.randint(0, 900, (1, 5), generator=torch.Generator().manual_seed(1))
+    ids = torch.randint(0, min(900, eng.config.vocab_size), (1, 5),
+                        generator=torch.Generator().manual_seed(1))
     out = eng.generate_greedy(ids, 4)
     assert out.shape == (1, 4)
     assert (out >= 0).all() and (out < eng.config.vocab_size).all()

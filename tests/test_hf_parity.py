@@ -183,3 +183,67 @@ def test_mixtral_block_matches_hf():
     kv.close()
     diff = (ours - hf_out).abs().max().item()
     assert diff < 1e-4, f"mixtral block vs HF: max diff {diff}"
+
+
+def test_gemma4_block_matches_hf():
+    from transformers.models.gemma4.configuration_gemma4 import Gemma4TextConfig
+    from transformers.models.gemma4.modeling_gemma4 import (
+        Gemma4TextDecoderLayer, Gemma4TextRotaryEmbedding)
+
+    cfg, blk, stack = _mk_block("gemma4-tiny")
+    li = 0  # sliding layer
+    hf_cfg = Gemma4TextConfig(
+        hidden_size=cfg.hidden_size, num_hidden_layers=cfg.num_hidden_layers,
+        num_attention_heads=cfg.num_attention_heads,
+        num_key_value_heads=cfg.num_key_value_heads,
+        intermediate_size=cfg.intermediate_size, vocab_size=cfg.vocab_size,
+        head_dim=cfg.head_dim,
+        global_head_dim=int(cfg.extras["global_head_dim"]),
+        hidden_size_per_layer_input=0, sliding_window=int(cfg.extras["sliding_window"]),
+        rms_norm_eps=cfg.rms_norm_eps, attn_implementation="eager",
+        attention_dropout=0.0,
+    )
+    layer = Gemma4TextDecoderLayer(hf_cfg, layer_idx=li).eval()
+    Hq, Hkv, D = blk.Hq, blk.Hkv, blk.D
+    with torch.no_grad():
+        layer.input_layernorm.weight.copy_(blk.input_norm_w)
+        layer.post_attention_layernorm.weight.copy_(blk.post_attn_norm_w)
+        layer.pre_feedforward_layernorm.weight.copy_(blk.pre_ffn_norm_w)
+        layer.post_feedforward_layernorm.weight.copy_(blk.post_ffn_norm_w)
+        layer.self_attn.q_proj.weight.copy_(blk.q_w)
+        layer.self_attn.k_proj.weight.copy_(blk.k_w)
+        layer.self_attn.v_proj.weight.copy_(blk.v_w)
+        layer.self_attn.o_proj.weight.copy_(blk.o_w)
+        layer.self_attn.q_norm.weight.copy_(blk.q_norm_w)
+        layer.self_attn.k_norm.weight.copy_(blk.k_norm_w)
+        I = cfg.intermediate_size
+        layer.mlp.gate_proj.weight.copy_(blk.gate_up_w[:I])
+        layer.mlp.up_proj.weight.copy_(blk.gate_up_w[I:])
+        layer.mlp.down_proj.weight.copy_(blk.down_w)
+
+    torch.manual_seed(0)
+    B, T = 1, 12  # > sliding_window=8 so the window path is exercised
+    h = torch.randn(B, T, cfg.hidden_size) * 0.3
+    rotary = Gemma4TextRotaryEmbedding(hf_cfg)
+    pos = torch.arange(T).unsqueeze(0).expand(B, T)
+    pos_emb = rotary(h, pos, layer_type="sliding_attention")
+    if isinstance(pos_emb, dict):
+        pos_emb = pos_emb["sliding_attention"]
+    qpos = torch.arange(T).view(T, 1)
+    kpos = torch.arange(T).view(1, T)
+    w = int(cfg.extras["sliding_window"])
+    keep = (kpos <= qpos) & (kpos > qpos - w)
+    mask = torch.where(keep, 0.0, float("-inf")).view(1, 1, T, T)
+    with torch.no_grad():
+        hf_out = layer(h, shared_kv_states={}, position_embeddings=pos_emb,
+                       attention_mask=mask, position_ids=pos)
+        if isinstance(hf_out, tuple):
+            hf_out = hf_out[0]
+
+    kv = stack.make_kv(1024).allocate(B, 64)
+    kv.extend(T)
+    ours = blk.forward_inference(h.clone(), kv,
+                                 torch.zeros(B, dtype=torch.int32))
+    kv.close()
+    diff = (ours - hf_out).abs().max().item()
+    assert diff < 1e-4, f"gemma4 block vs HF: max diff {diff}"
