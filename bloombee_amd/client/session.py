@@ -106,9 +106,11 @@ class _SpanSession:
 
 class InferenceSession:
     def __init__(self, manager: RemoteSequenceManager, max_length: int,
-                 config: Optional[ClientConfig] = None):
+                 config: Optional[ClientConfig] = None,
+                 allow_push: bool = True):
         self.manager = manager
         self.config = config or manager.config
+        self.allow_push = allow_push
         self.max_length = max_length
         self.batch_size: Optional[int] = None
         self.spans: List[_SpanSession] = []
@@ -122,7 +124,8 @@ class InferenceSession:
         route = self.manager.make_sequence(
             0, self.manager.num_blocks,
             cache_tokens_needed=batch_size * self.max_length)
-        use_push = (self.config.use_server_to_server and len(route) > 1)
+        use_push = (self.allow_push and self.config.use_server_to_server
+                    and len(route) > 1)
         sessions: List[Optional[_SpanSession]] = [None] * len(route)
         # open back-to-front so each span knows its downstream session id
         push_to = None
@@ -207,6 +210,49 @@ class InferenceSession:
         self.position = pos + hidden.shape[1]
         self.step_count += 1
         return out
+
+    # -- speculative decoding --------------------------------------------
+    def spec_step(self, hidden: torch.Tensor, position_ids: torch.Tensor,
+                  tree_mask: torch.Tensor) -> torch.Tensor:
+        """Tree-verify step: hidden (B, T, H) linearized tree nodes with
+        per-node absolute positions + ancestor mask. KV written speculatively
+        on every span; follow with spec_commit(keep). Spec sessions are
+        opened with allow_push=False (the tree tensors ride the client
+        stream span by span)."""
+        if self.batch_size is None:
+            self.batch_size = hidden.shape[0]
+            self._open_chain(self.batch_size, replay=False)
+        t = self.config.step_timeout
+        out = hidden
+        for s in self.spans:
+            async def go(s=s, out=out):
+                await s.stream.send({"pos": self.position, "spec": True,
+                                     "step": self.step_count},
+                                    [out, position_ids.int(), tree_mask])
+                item = await s.stream.recv()
+                if item is None:
+                    raise RpcError(f"stream closed by {s.span.peer_id}")
+                return item[1][0]
+
+            out = run_coroutine(go(), t)
+        self.step_count += 1
+        return out
+
+    def spec_commit(self, keep) -> int:
+        """Accept tree nodes `keep[b]` (ascending linear indices); every span
+        compacts + commits its KV. Returns the accepted length."""
+        t = self.config.step_timeout
+        for s in self.spans:
+            async def go(s=s):
+                await s.stream.send({"spec_commit": [list(k) for k in keep]})
+                item = await s.stream.recv()
+                if item is None or not item[0].get("committed"):
+                    raise RpcError(f"spec commit failed on {s.span.peer_id}")
+
+            run_coroutine(go(), t)
+        n = len(keep[0])
+        self.position += n
+        return n
 
     def __enter__(self):
         return self

@@ -65,9 +65,14 @@ class BlockStack(torch.nn.Module):
     @torch.no_grad()
     def forward_inference(self, hidden: torch.Tensor, kv: SessionHandle,
                           start_pos: torch.Tensor,
-                          position_ids: Optional[torch.Tensor] = None) -> torch.Tensor:
+                          position_ids: Optional[torch.Tensor] = None,
+                          tree_mask: Optional[torch.Tensor] = None) -> torch.Tensor:
         for blk in self.blocks:
-            hidden = blk.forward_inference(hidden, kv, start_pos, position_ids)
+            if tree_mask is not None:
+                hidden = blk.forward_inference(hidden, kv, start_pos,
+                                               position_ids, tree_mask=tree_mask)
+            else:
+                hidden = blk.forward_inference(hidden, kv, start_pos, position_ids)
         return hidden
 
     def forward_train(self, hidden: torch.Tensor, start_pos: int = 0) -> torch.Tensor:

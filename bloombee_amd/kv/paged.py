@@ -273,6 +273,37 @@ class SessionHandle:
             # freed slots in the host table are stale but unreachable (ctx_len
             # bounds every kernel read); no rewrite needed.
 
+    def reorder_and_commit(self, keep: List[List[int]]) -> None:
+        """Speculative-tree accept: compact the kept nodes' K/V.
+
+        keep[b] = ascending linear indices (within this round's speculative
+        region, i.e. relative to l_acc) of the accepted tree nodes. Their
+        K/V rows move to contiguous positions l_acc..l_acc+len(keep[b]) and
+        are committed; the rest of the speculative region is rolled back
+        (ref memory_cache_manager.py:1749-2040 update_cache_and_async_reorder
+        — here a paged in-place compaction, no async side channel needed).
+        """
+        P = self.cache.page_size
+        for b, idx in enumerate(keep):
+            s = self.seqs[b]
+            base = s.l_acc
+            for d, rel in enumerate(idx):
+                src, dst = base + rel, base + d
+                if src == dst:
+                    continue
+                if src < dst or base + rel >= s.l_spec:
+                    raise PagedKVError("keep indices must be ascending and "
+                                       "inside the speculative region")
+                pg_s = int(self._page_table_host[b, src // P])
+                pg_d = int(self._page_table_host[b, dst // P])
+                for l in range(self.cache.num_layers):
+                    kp = self.cache.k_pages(l)
+                    kp[pg_d, :, dst % P, :] = kp[pg_s, :, src % P, :]
+                    if self.cache.layout[l] != "k":
+                        vp = self.cache.v_pages(l)
+                        vp[pg_d, :, :, dst % P] = vp[pg_s, :, :, src % P]
+        self.commit([len(k) for k in keep])
+
     def truncate(self, new_lengths: List[int]) -> None:
         """Failover / history-replay support: cut sequences back to
         `new_lengths` committed tokens (ref inference_session.py:802-831)."""

@@ -104,6 +104,7 @@ class LlamaBlock(torch.nn.Module):
         kv: SessionHandle,
         start_pos: torch.Tensor,           # (B,) int32 — absolute pos of token 0
         position_ids: Optional[torch.Tensor] = None,  # (B, T) int32
+        tree_mask: Optional[torch.Tensor] = None,     # (B, T, T) bool, spec verify
     ) -> torch.Tensor:
         B, T, H = hidden.shape
         Hq, Hkv, D = self.Hq, self.Hkv, self.D
@@ -119,7 +120,17 @@ class LlamaBlock(torch.nn.Module):
         # reads q in place and emits the O-projection input — no transposes
         ops.rope_kv_write_(qkv, Hq, Hkv, cos, sin, position_ids, kp, vp, pt,
                            start_pos)
-        attn = ops.attn_paged_qkv(qkv, Hq, Hkv, kp, vp, pt, start_pos, self.scale)
+        if tree_mask is not None:
+            # tree-structured verify step (spec decoding): unfused q view +
+            # the tree-mask attention path
+            q = (qkv[..., : Hq * D].view(B, T, Hq, D)
+                 .permute(0, 2, 1, 3).contiguous())
+            attn = ops.attn_paged(q, kp, vp, pt, start_pos.long(), self.scale,
+                                  tree_mask=tree_mask)
+            attn = attn.permute(0, 2, 1, 3).reshape(B, T, Hq * D)
+        else:
+            attn = ops.attn_paged_qkv(qkv, Hq, Hkv, kp, vp, pt, start_pos,
+                                      self.scale)
         a = ops.linear(attn, self.o_w)
 
         # h2 = hidden + a fused into the post-attention norm

