@@ -37,7 +37,8 @@ class StackBackend:
     def __init__(self, config: ModelConfig, start: int, end: int,
                  device: str = "cpu", seed: int = 0,
                  kv_max_tokens: int = 1 << 18,
-                 checkpoint_dir: Optional[str] = None):
+                 checkpoint_dir: Optional[str] = None,
+                 offload_policy=None):
         self.config = config
         self.start, self.end = start, end
         self.device = torch.device(device)
@@ -46,6 +47,9 @@ class StackBackend:
             from bloombee_amd.server.from_pretrained import load_block_weights
             for i, blk in enumerate(self.stack.blocks):
                 load_block_weights(blk, checkpoint_dir, start + i)
+        if offload_policy is not None and offload_policy.offloads_weights:
+            from bloombee_amd.offload.weights import OffloadedBlockStack
+            self.stack = OffloadedBlockStack(self.stack, offload_policy)
         self.kv_pool: PagedKVCache = self.stack.make_kv(kv_max_tokens)
         self.pool = TaskPool(name=f"worker[{start}:{end}]")
         self.sessions: Dict[str, SessionState] = {}

@@ -47,6 +47,7 @@ class Server:
         checkpoint_dir: Optional[str] = None,
         throughput: float = 1.0,
         model_name: Optional[str] = None,
+        offload_policy=None,
     ):
         self.config = model if isinstance(model, ModelConfig) else resolve_config(model)
         self.model_name = model_name or (model if isinstance(model, str)
@@ -74,7 +75,8 @@ class Server:
         self.backend = StackBackend(self.config, block_indices[0],
                                     block_indices[1], device=device, seed=seed,
                                     kv_max_tokens=kv_max_tokens,
-                                    checkpoint_dir=checkpoint_dir)
+                                    checkpoint_dir=checkpoint_dir,
+                                    offload_policy=offload_policy)
         self.rpc = RpcServer(host, port)
         self.handler = ConnectionHandler(self.backend, self.rpc)
         self._loop: Optional[asyncio.AbstractEventLoop] = None
