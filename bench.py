@@ -31,6 +31,8 @@ def main():
                     help="prompt length prefillled before the timed decode steps")
     ap.add_argument("--device", default=None, help="override (e.g. cpu for tests)")
     ap.add_argument("--micro-batches", type=int, default=0)
+    ap.add_argument("--tp", type=int, default=1,
+                    help="tensor-parallel degree within the pipeline (world %% tp == 0)")
     args = ap.parse_args()
 
     import torch.distributed as dist
@@ -52,7 +54,8 @@ def main():
 
     stage = PipelineStage(args.model, device, global_batch,
                           micro_batches=args.micro_batches,
-                          kv_max_tokens=kv_tokens, max_session_len=session_len)
+                          kv_max_tokens=kv_tokens, max_session_len=session_len,
+                          tp=args.tp)
 
     gen = torch.Generator().manual_seed(42)
     V = stage.config.vocab_size
@@ -122,7 +125,8 @@ def main():
                 "model": args.model,
                 "global_batch": global_batch,
                 "seq_len": args.prompt,
-                "parallelism": f"pp{world}",
+                "parallelism": (f"pp{world // args.tp}xtp{args.tp}" if args.tp > 1
+                                else f"pp{world}"),
                 "micro_batches": stage.M,
                 "p50_step_ms": round(p50, 3),
             },

@@ -45,24 +45,46 @@ class PipelineStage:
 
     def __init__(self, config_or_name, device, global_batch: int,
                  micro_batches: int = 0, seed: int = 0,
-                 kv_max_tokens: int = 1 << 17, max_session_len: int = 4096):
+                 kv_max_tokens: int = 1 << 17, max_session_len: int = 4096,
+                 tp: int = 1):
         cfg = (config_or_name if isinstance(config_or_name, ModelConfig)
                else resolve_config(config_or_name))
         self.config = cfg
         self.device = torch.device(device)
         self.rank = dist.get_rank() if dist.is_initialized() else 0
         self.world = dist.get_world_size() if dist.is_initialized() else 1
+        assert self.world % tp == 0, f"world {self.world} not divisible by tp {tp}"
+        self.tp = tp
+        self.pp_world = self.world // tp
+        self.pp_stage = self.rank // tp
+        self.tp_rank = self.rank % tp
+        # per-stage tensor-parallel groups (RCCL all-reduce over xGMI inside
+        # a worker — BASELINE.json config 4; created on every rank)
+        self.tp_group = None
+        if tp > 1:
+            for st in range(self.pp_world):
+                g = dist.new_group(list(range(st * tp, (st + 1) * tp)))
+                if st == self.pp_stage:
+                    self.tp_group = g
         self.global_batch = global_batch
         if micro_batches <= 0:
-            micro_batches = min(4 * self.world, global_batch) if self.world > 1 else 1
+            micro_batches = (min(4 * self.pp_world, global_batch)
+                             if self.pp_world > 1 else 1)
         while global_batch % micro_batches != 0:
             micro_batches -= 1
         self.M = micro_batches
         self.mb = global_batch // self.M
 
-        start, end = layer_range(cfg.num_hidden_layers, self.rank, self.world)
-        logger.info(f"rank {self.rank}/{self.world}: layers [{start}, {end})")
-        self.stack = BlockStack(cfg, start, end, device=self.device, seed=seed)
+        start, end = layer_range(cfg.num_hidden_layers, self.pp_stage,
+                                 self.pp_world)
+        logger.info(f"rank {self.rank}/{self.world} (pp {self.pp_stage} tp "
+                    f"{self.tp_rank}): layers [{start}, {end})")
+        if tp > 1:
+            from bloombee_amd.parallel.tensor import TPBlockStack
+            self.stack = TPBlockStack(cfg, start, end, device=self.device,
+                                      seed=seed, group=self.tp_group)
+        else:
+            self.stack = BlockStack(cfg, start, end, device=self.device, seed=seed)
         self.kv_pool = self.stack.make_kv(kv_max_tokens)
         self.kv = self.kv_pool.allocate(global_batch, max_session_len)
 
@@ -76,8 +98,11 @@ class PipelineStage:
             self.lm_head_w = (self.embed if cfg.tie_word_embeddings else
                               (torch.randn(cfg.vocab_size, cfg.hidden_size, generator=gen)
                                .mul_(0.02).to(dt).to(self.device)))
-        self.next_rank = (self.rank + 1) % self.world
-        self.prev_rank = (self.rank - 1) % self.world
+        # inter-stage chain runs between tp leaders (tp_rank 0); activations
+        # are replicated within a stage after each block's all-reduce
+        self.next_rank = ((self.pp_stage + 1) % self.pp_world) * tp
+        self.prev_rank = ((self.pp_stage - 1) % self.pp_world) * tp
+        self.is_leader = self.tp_rank == 0
         # persistent recv buffers per micro-batch
         self._recv_buf = [torch.empty(self.mb, 1, cfg.hidden_size, dtype=cfg.dtype,
                                       device=self.device) for _ in range(self.M)]
@@ -160,30 +185,44 @@ class PipelineStage:
         views = self._views()
         pending = []
 
-        if self.is_client:
-            hid = F.embedding(ids.view(B, 1).to(self.device), self.embed)
+        if self.pp_stage == 0:
+            if self.is_client:
+                hid = F.embedding(ids.view(B, 1).to(self.device), self.embed)
+            else:
+                hid = torch.empty(B, 1, cfg.hidden_size, dtype=cfg.dtype,
+                                  device=self.device)
+            if self.tp > 1:
+                dist.broadcast(hid, src=0, group=self.tp_group)
             outs: List[torch.Tensor] = []
             for j in range(M):
                 h = self._run_local(hid[j * mb:(j + 1) * mb], views[j],
                                     start_all[j * mb:(j + 1) * mb])
-                if self.world > 1:
+                if self.pp_world > 1 and self.is_leader:
                     pending.append(dist.isend(h.contiguous(), self.next_rank))
-                else:
+                elif self.pp_world == 1:
                     outs.append(h)
-            if self.world > 1:
+            if self.pp_world > 1 and self.is_client:
                 for j in range(M):
                     dist.recv(self._recv_buf[j], self.prev_rank)
                     outs.append(self._recv_buf[j])
             for w in pending:
                 w.wait()
+            if not self.is_client:
+                return None
             hidden = torch.cat(outs, dim=0)
             return self._lm_head(hidden[:, -1])
         else:
             for j in range(M):
-                dist.recv(self._recv_buf[j], self.prev_rank)
+                if self.is_leader:
+                    dist.recv(self._recv_buf[j], self.prev_rank)
+                if self.tp > 1:
+                    dist.broadcast(self._recv_buf[j],
+                                   src=self.pp_stage * self.tp,
+                                   group=self.tp_group)
                 h = self._run_local(self._recv_buf[j], views[j],
                                     start_all[j * mb:(j + 1) * mb])
-                pending.append(dist.isend(h.contiguous(), self.next_rank))
+                if self.is_leader:
+                    pending.append(dist.isend(h.contiguous(), self.next_rank))
             for w in pending:
                 w.wait()
             return None
@@ -200,29 +239,42 @@ class PipelineStage:
         pending = []
         bufs = [torch.empty(mb, T, cfg.hidden_size, dtype=cfg.dtype,
                             device=self.device) for _ in range(M)]
-        if self.is_client:
-            hid = F.embedding(ids.to(self.device), self.embed)
+        if self.pp_stage == 0:
+            if self.is_client:
+                hid = F.embedding(ids.to(self.device), self.embed)
+            else:
+                hid = torch.empty(B, T, cfg.hidden_size, dtype=cfg.dtype,
+                                  device=self.device)
+            if self.tp > 1:
+                dist.broadcast(hid, src=0, group=self.tp_group)
             outs = []
             for j in range(M):
                 h = self._run_local(hid[j * mb:(j + 1) * mb], views[j],
                                     start_all[j * mb:(j + 1) * mb])
-                if self.world > 1:
+                if self.pp_world > 1 and self.is_leader:
                     pending.append(dist.isend(h.contiguous(), self.next_rank))
-                else:
+                elif self.pp_world == 1:
                     outs.append(h)
-            if self.world > 1:
+            if self.pp_world > 1 and self.is_client:
                 for j in range(M):
                     dist.recv(bufs[j], self.prev_rank)
                     outs.append(bufs[j])
             for w in pending:
                 w.wait()
+            if not self.is_client:
+                return None
             hidden = torch.cat(outs, dim=0)
             return self._lm_head(hidden[:, -1])
         else:
             for j in range(M):
-                dist.recv(bufs[j], self.prev_rank)
+                if self.is_leader:
+                    dist.recv(bufs[j], self.prev_rank)
+                if self.tp > 1:
+                    dist.broadcast(bufs[j], src=self.pp_stage * self.tp,
+                                   group=self.tp_group)
                 h = self._run_local(bufs[j], views[j], start_all[j * mb:(j + 1) * mb])
-                pending.append(dist.isend(h.contiguous(), self.next_rank))
+                if self.is_leader:
+                    pending.append(dist.isend(h.contiguous(), self.next_rank))
             for w in pending:
                 w.wait()
             return None
