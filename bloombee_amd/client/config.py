@@ -25,5 +25,6 @@ class ClientConfig:
     update_period: float = 30.0
     ban_timeout: float = 15.0
 
+    active_adapter: "Optional[str]" = None  # LoRA adapter name servers apply
     use_server_to_server: bool = True      # s2s activation push during decode
     push_only_downstream_decode: bool = True

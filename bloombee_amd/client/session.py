@@ -41,7 +41,8 @@ class _SpanSession:
     def create(cls, span: RemoteSpanInfo, batch_size: int, max_length: int,
                push_to: Optional[Tuple[str, int, str]] = None,
                push_only_recv: bool = False, quiet: bool = False,
-               timeout: float = 30.0) -> "_SpanSession":
+               timeout: float = 30.0, adapter: Optional[str] = None,
+               ) -> "_SpanSession":
         client = get_client(span.server_info.host, span.server_info.port)
         sid = uuid.uuid4().hex
 
@@ -53,6 +54,7 @@ class _SpanSession:
                 "push_to": list(push_to) if push_to else None,
                 "push_only_recv": push_only_recv,
                 "quiet": quiet,
+                "adapter": adapter,
             })
             first = await stream.recv()
             if first is None or not first[0].get("ok"):
@@ -135,7 +137,8 @@ class InferenceSession:
             s = _SpanSession.create(
                 route[i], batch_size, self.max_length,
                 push_to=push_to, push_only_recv=push_only_recv, quiet=quiet,
-                timeout=self.config.request_timeout)
+                timeout=self.config.request_timeout,
+                adapter=getattr(self.config, "active_adapter", None))
             sessions[i] = s
             push_to = (route[i].server_info.host, route[i].server_info.port,
                        s.session_id) if use_push else None

@@ -94,6 +94,16 @@ class LlamaBlock(torch.nn.Module):
                                     device=dev).mul_(std).to(w.dtype))
         return self
 
+    def _lin(self, x, w, name, **kw):
+        """ops.linear + optional active-LoRA delta (utils/peft.py)."""
+        y = ops.linear(x, w, **kw)
+        ld = getattr(self, "lora_delta", None)
+        if ld is not None:
+            d = ld(name, x)
+            if d is not None:
+                y = y + d
+        return y
+
     # ------------------------------------------------------------------
     # inference path (HIP kernels, paged KV)
     # ------------------------------------------------------------------
@@ -111,7 +121,7 @@ class LlamaBlock(torch.nn.Module):
         cfg = self.config
 
         x = ops.rms_norm(hidden, self.input_norm_w, cfg.rms_norm_eps)
-        qkv = ops.linear(x, self.qkv_w)                    # (B, T, (Hq+2Hkv)D)
+        qkv = self._lin(x, self.qkv_w, "qkv_w")            # (B, T, (Hq+2Hkv)D)
         cos, sin = self.rope.get(hidden.device)
         kp = kv.k_pages(self.layer_index)
         vp = kv.v_pages(self.layer_index)
@@ -131,7 +141,7 @@ class LlamaBlock(torch.nn.Module):
         else:
             attn = ops.attn_paged_qkv(qkv, Hq, Hkv, kp, vp, pt, start_pos,
                                       self.scale)
-        a = ops.linear(attn, self.o_w)
+        a = self._lin(attn, self.o_w, "o_w")
 
         # h2 = hidden + a fused into the post-attention norm
         h2, y = ops.rms_norm_residual(a, hidden, self.post_norm_w, cfg.rms_norm_eps)

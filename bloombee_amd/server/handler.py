@@ -73,6 +73,7 @@ class ConnectionHandler:
         [hidden] (or empty tensors in push-only mode — inputs arrive via
         rpc_push). Replies: {pos} + [hidden_out]."""
         sid = meta.get("session_id") or uuid.uuid4().hex
+        adapter = meta.get("adapter")
         max_length = int(meta["max_length"])
         batch_size = int(meta["batch_size"])
         push_to = meta.get("push_to")  # downstream [host, port]
@@ -124,10 +125,14 @@ class ConnectionHandler:
                     rest = rest[2:]
                 if rest:
                     prompts = rest[0]
-                out = await loop.run_in_executor(
-                    None, lambda: self.backend.inference_step(
-                        sid, hidden, pos, prompts, position_ids, tree_mask,
-                        speculative=spec))
+                def _step():
+                    from bloombee_amd.utils.peft import using_adapter
+                    with using_adapter(adapter):
+                        return self.backend.inference_step(
+                            sid, hidden, pos, prompts, position_ids, tree_mask,
+                            speculative=spec)
+
+                out = await loop.run_in_executor(None, _step)
                 out_cpu = out.cpu()
                 if push_to is not None:
                     await self._push_downstream(push_to, pos, out_cpu,
