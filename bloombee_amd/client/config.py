@@ -26,5 +26,9 @@ class ClientConfig:
     ban_timeout: float = 15.0
 
     active_adapter: "Optional[str]" = None  # LoRA adapter name servers apply
+    # activation wire codec: "raw" | "zlib" | "bsplit+zlib" (byte-split lanes
+    # before DEFLATE — the reference lossless transport's best default,
+    # lossless_transport.py:1604-1667)
+    wire_codec: str = "raw"
     use_server_to_server: bool = True      # s2s activation push during decode
     push_only_downstream_decode: bool = True

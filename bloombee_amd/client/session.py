@@ -31,18 +31,19 @@ class _SpanSession:
     """Client end of one span's rpc_inference stream."""
 
     def __init__(self, span: RemoteSpanInfo, stream: Stream, session_id: str,
-                 quiet: bool):
+                 quiet: bool, codec: str = "raw"):
         self.span = span
         self.stream = stream
         self.session_id = session_id
         self.quiet = quiet
+        self.codec = codec
 
     @classmethod
     def create(cls, span: RemoteSpanInfo, batch_size: int, max_length: int,
                push_to: Optional[Tuple[str, int, str]] = None,
                push_only_recv: bool = False, quiet: bool = False,
                timeout: float = 30.0, adapter: Optional[str] = None,
-               ) -> "_SpanSession":
+               codec: str = "raw") -> "_SpanSession":
         client = get_client(span.server_info.host, span.server_info.port)
         sid = uuid.uuid4().hex
 
@@ -55,6 +56,7 @@ class _SpanSession:
                 "push_only_recv": push_only_recv,
                 "quiet": quiet,
                 "adapter": adapter,
+                "codec": codec,
             })
             first = await stream.recv()
             if first is None or not first[0].get("ok"):
@@ -62,12 +64,13 @@ class _SpanSession:
             return stream
 
         stream = run_coroutine(open_(), timeout)
-        return cls(span, stream, sid, quiet)
+        return cls(span, stream, sid, quiet, codec)
 
     def step(self, hidden: torch.Tensor, pos: int, step: int,
              timeout: float) -> Optional[torch.Tensor]:
         async def go():
-            await self.stream.send({"pos": pos, "step": step}, [hidden])
+            await self.stream.send({"pos": pos, "step": step}, [hidden],
+                                   codec=self.codec)
             item = await self.stream.recv()
             if item is None:
                 raise RpcError(f"stream closed by {self.span.peer_id}")
@@ -79,7 +82,8 @@ class _SpanSession:
     def send_only(self, hidden: torch.Tensor, pos: int, step: int,
                   timeout: float) -> None:
         async def go():
-            await self.stream.send({"pos": pos, "step": step}, [hidden])
+            await self.stream.send({"pos": pos, "step": step}, [hidden],
+                                   codec=self.codec)
 
         run_coroutine(go(), timeout)
 
@@ -138,7 +142,8 @@ class InferenceSession:
                 route[i], batch_size, self.max_length,
                 push_to=push_to, push_only_recv=push_only_recv, quiet=quiet,
                 timeout=self.config.request_timeout,
-                adapter=getattr(self.config, "active_adapter", None))
+                adapter=getattr(self.config, "active_adapter", None),
+                codec=getattr(self.config, "wire_codec", "raw"))
             sessions[i] = s
             push_to = (route[i].server_info.host, route[i].server_info.port,
                        s.session_id) if use_push else None

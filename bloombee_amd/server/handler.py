@@ -74,6 +74,7 @@ class ConnectionHandler:
         rpc_push). Replies: {pos} + [hidden_out]."""
         sid = meta.get("session_id") or uuid.uuid4().hex
         adapter = meta.get("adapter")
+        codec = meta.get("codec", "raw")
         max_length = int(meta["max_length"])
         batch_size = int(meta["batch_size"])
         push_to = meta.get("push_to")  # downstream [host, port]
@@ -139,7 +140,7 @@ class ConnectionHandler:
                                                 item_meta)
                 if not quiet:
                     await stream.send({"pos": pos, "step": item_meta.get("step")},
-                                      [out_cpu])
+                                      [out_cpu], codec=codec)
                 # quiet spans send nothing: during push-mode decode the client
                 # reads outputs from the LAST span only (push_only_downstream,
                 # ref inference_session.py:178-196)

@@ -173,3 +173,14 @@ def test_bloom_swarm_matches_local():
         s1.shutdown()
         s2.shutdown()
         boot.shutdown()
+
+
+def test_swarm_compressed_wire_matches_local(swarm):
+    boot, _ = swarm
+    model = _make_model(boot, wire_codec="bsplit+zlib")
+    gen = torch.Generator().manual_seed(5)
+    prompt = torch.randint(0, 1000, (2, 7), generator=gen)
+    out = model.generate(prompt, max_new_tokens=6)
+    expect = _local_tokens(prompt, 6)
+    assert torch.equal(out[:, 7:], expect)  # codec is lossless
+    model.remote.manager.shutdown()
