@@ -49,20 +49,24 @@ DEVINL void store_bf16_e(unsigned short* p, const float* in) {
 template <int D>
 __global__ void attn_decode_combine_kernel(
     const float* __restrict__ part_ml, const float* __restrict__ part_acc,
-    unsigned short* __restrict__ out, int Hkv, int G, int n_split,
-    long out_sb, long out_sh) {
+    unsigned short* __restrict__ out, int Hkv, int G, int nch, int maxg,
+    int n_split, long out_sb, long out_sh) {
   constexpr int E = D / 16;
-  const int bh = blockIdx.x;
+  const int bh = blockIdx.x;           // (b, kvh, chunk)
+  const int b = bh / (Hkv * nch);
+  const int rem = bh % (Hkv * nch);
+  const int kvh = rem / nch;
+  const int g0 = (rem % nch) * maxg;
   const int g = threadIdx.x / 16;
   const int li = threadIdx.x & 15;
-  if (g >= G) return;
+  if (g0 + g >= G || g >= maxg) return;
   float m2 = NEG_BIG, l = 0.f, acc[E];
 #pragma unroll
   for (int j = 0; j < E; ++j) acc[j] = 0.f;
   for (int s = 0; s < n_split; ++s) {
     const long pbase = (long)bh * n_split + s;
-    const float mo = part_ml[(pbase * G + g) * 2 + 0];
-    const float lo = part_ml[(pbase * G + g) * 2 + 1];
+    const float mo = part_ml[(pbase * maxg + g) * 2 + 0];
+    const float lo = part_ml[(pbase * maxg + g) * 2 + 1];
     if (lo == 0.f) continue;
     const float mn = fmaxf(m2, mo);
     const float c1f = fast_exp2(m2 - mn);
@@ -70,15 +74,15 @@ __global__ void attn_decode_combine_kernel(
     l = l * c1f + lo * c2f;
 #pragma unroll
     for (int j = 0; j < E; ++j)
-      acc[j] = acc[j] * c1f + part_acc[(pbase * G + g) * D + li * E + j] * c2f;
+      acc[j] = acc[j] * c1f + part_acc[(pbase * maxg + g) * D + li * E + j] * c2f;
     m2 = mn;
   }
   float o[E];
   const float inv = (l > 0.f) ? 1.f / l : 0.f;
 #pragma unroll
   for (int j = 0; j < E; ++j) o[j] = acc[j] * inv;
-  const int b = bh / Hkv, kvh = bh % Hkv;
-  store_bf16_e<E>(out + (long)b * out_sb + (kvh * G + g) * out_sh + li * E, o);
+  store_bf16_e<E>(out + (long)b * out_sb + (kvh * G + g0 + g) * out_sh + li * E,
+                  o);
 }
 
 // Pure-streaming probe: same grid/walk/loads as attn_decode_kernel but no

@@ -25,6 +25,9 @@
 
 static constexpr int DKVBLK = 32;  // positions per KV tile
 
+// nch: GQA group chunks per kv head (ceil(G/16)) — MQA groups wider than the
+// 16-row MFMA q-tile (falcon-7b G=71) split into chunks sharing the kv head.
+// alibi: per-query-head slopes (bloom family), or null.
 template <int D, int MAXG>
 __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
     const unsigned short* __restrict__ q,        // strided, see q_sb/q_sh
@@ -32,10 +35,11 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
     const unsigned short* __restrict__ v_pages,  // (np, Hkv, D, P) d-major
     const int* __restrict__ page_table,          // (B, maxp)
     const int* __restrict__ ctx_lens,            // (B,)
+    const float* __restrict__ alibi,             // (Hq,) slopes or null
     unsigned short* __restrict__ out,
-    float* __restrict__ part_ml,                 // (B*Hkv*n_split, G, 2)
-    float* __restrict__ part_acc,                // (B*Hkv*n_split, G, D)
-    int B, int Hkv, int G, int P, int maxp, int n_split, int window,
+    float* __restrict__ part_ml,                 // (B*Hkv*nch*n_split, MAXG, 2)
+    float* __restrict__ part_acc,                // (..., MAXG, D)
+    int B, int Hkv, int G, int nch, int P, int maxp, int n_split, int window,
     float scale, long q_sb, long q_sh, long out_sb, long out_sh) {
   constexpr int NKK = D / 32;   // QK^T k-slices
   constexpr int NDT = D / 16;   // PV d-tiles
@@ -43,7 +47,11 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
 
   const int bh = blockIdx.x;
   const int split = blockIdx.y;
-  const int b = bh / Hkv, kvh = bh % Hkv;
+  const int b = bh / (Hkv * nch);
+  const int rem = bh % (Hkv * nch);
+  const int kvh = rem / nch;
+  const int g0 = (rem % nch) * MAXG;   // this chunk's first group row
+  const int Gl = min(G - g0, MAXG);    // live rows in this chunk
   const int ctx = ctx_lens[b];
 
   const int pages_total = (ctx + P - 1) / P;
@@ -66,13 +74,20 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
   unsigned char* p_lds = lds + wave * 16 * PROW_B;
   float* merge_lds = (float*)(lds + 4 * 16 * PROW_B);
 
-  // Q fragments: A[row][k] with row = head g (rows >= G harmless garbage)
-  const int qh = (li < G) ? li : G - 1;
+  // Q fragments: A[row][k] with row = head g (rows >= Gl harmless garbage)
+  const int qh = g0 + ((li < Gl) ? li : Gl - 1);
   bf16x8 qfrag[NKK];
 #pragma unroll
   for (int kk = 0; kk < NKK; ++kk)
     qfrag[kk] = as_bf16x8(*reinterpret_cast<const short8*>(
         q + (long)b * q_sb + (long)(kvh * G + qh) * q_sh + hi * 8 + 32 * kk));
+  // per-row alibi slopes (row r = hi*4 + reg -> head kvh*G + g0 + r)
+  float aslope[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int g = g0 + hi * 4 + r;
+    aslope[r] = (alibi != nullptr && g < G) ? alibi[kvh * G + g] : 0.f;
+  }
 
   float m2[4], l[4];
   f32x4 acc_o[NDT];
@@ -130,7 +145,7 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
 #pragma unroll
       for (int n = 0; n < DKVBLK / 16; ++n) {
         const int kpos = tb + li + 16 * n;
-        float sv = s[n][reg] * sc2;
+        float sv = s[n][reg] * sc2 + aslope[reg] * LOG2E * (float)kpos;
         const bool dead = (kpos >= c1) | (kpos < lo);
         sv = dead ? NEG_BIG : sv;
         p[n][reg] = sv;
@@ -170,8 +185,8 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
   }
 
   // ---- merge the 4 waves ----
-  // C row r = hi*4 + reg holds head g = r (valid for g < G, so MQA G up to
-  // 16 works). Waves 1..3 publish (acc rows, m, l) to LDS; wave 0 folds them.
+  // C row r = hi*4 + reg holds chunk-local head g = r (valid for g < Gl).
+  // Waves 1..3 publish (acc rows, m, l) to LDS; wave 0 folds them.
   auto wave_slot = [&](int slot) { return merge_lds + slot * MAXG * (D + 2); };
   __syncthreads();
   if (wave > 0) {
@@ -179,7 +194,7 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
 #pragma unroll
     for (int reg = 0; reg < 4; ++reg) {
       const int g = hi * 4 + reg;
-      if (g < G) {
+      if (g < Gl) {
 #pragma unroll
         for (int n = 0; n < NDT; ++n) dst[g * (D + 2) + li + 16 * n] = acc_o[n][reg];
         if (li == 0) {
@@ -196,7 +211,7 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
 #pragma unroll
     for (int reg = 0; reg < 4; ++reg) {
       const int g = hi * 4 + reg;
-      if (g >= G) continue;
+      if (g >= Gl) continue;
       const float mo = src[g * (D + 2) + D];
       const float lo2 = src[g * (D + 2) + D + 1];
       const float mn = fmaxf(m2[reg], mo);
@@ -214,11 +229,11 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
 #pragma unroll
     for (int reg = 0; reg < 4; ++reg) {
       const int g = hi * 4 + reg;
-      if (g >= G) continue;
+      if (g >= Gl) continue;
       const float inv = (l[reg] > 0.f) ? 1.f / l[reg] : 0.f;
 #pragma unroll
       for (int n = 0; n < NDT; ++n)
-        out[(long)b * out_sb + (long)(kvh * G + g) * out_sh + li + 16 * n] =
+        out[(long)b * out_sb + (long)(kvh * G + g0 + g) * out_sh + li + 16 * n] =
             f2bf(acc_o[n][reg] * inv);
     }
   } else {
@@ -226,13 +241,13 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
 #pragma unroll
     for (int reg = 0; reg < 4; ++reg) {
       const int g = hi * 4 + reg;
-      if (g >= G) continue;
+      if (g >= Gl) continue;
 #pragma unroll
       for (int n = 0; n < NDT; ++n)
-        part_acc[(pbase * G + g) * D + li + 16 * n] = acc_o[n][reg];
+        part_acc[(pbase * MAXG + g) * D + li + 16 * n] = acc_o[n][reg];
       if (li == 0) {
-        part_ml[(pbase * G + g) * 2 + 0] = m2[reg];
-        part_ml[(pbase * G + g) * 2 + 1] = l[reg];
+        part_ml[(pbase * MAXG + g) * 2 + 0] = m2[reg];
+        part_ml[(pbase * MAXG + g) * 2 + 1] = l[reg];
       }
     }
   }

@@ -31,6 +31,7 @@ __global__ __launch_bounds__(256) void attn_prefill_kernel(
     const unsigned short* __restrict__ v_pages,
     const int* __restrict__ page_table,          // (B, maxp)
     const int* __restrict__ q_start,             // (B,)
+    const float* __restrict__ alibi,             // (Hq,) slopes or null
     unsigned short* __restrict__ out,            // (B, Hq, Tq, D)
     int B, int Hq, int G, int Tq, int P, int maxp, int window, float scale,
     long q_sb, long q_st, long q_sh, long o_sb, long o_st, long o_sh) {
@@ -61,6 +62,7 @@ __global__ __launch_bounds__(256) void attn_prefill_kernel(
   const int qs = q_start[b];
   const int ctx = qs + Tq;
   const float sc2 = scale * LOG2E;
+  const float aslope = (alibi != nullptr) ? alibi[h] * LOG2E : 0.f;
 
   // ---- Q fragments (load once): row = li, k = hi*8 + j (+32*kk) ----
   const int my_qrow = min(qbase + wave * 16 + li, Tq - 1);
@@ -142,7 +144,7 @@ __global__ __launch_bounds__(256) void attn_prefill_kernel(
 #pragma unroll
       for (int n = 0; n < KVBLK / 16; ++n) {
         const int kpos = kbase + li + 16 * n;
-        float sv = s[n][reg] * sc2;
+        float sv = s[n][reg] * sc2 + aslope * (float)kpos;
         const bool dead = (qrow >= Tq) | (kpos > qpos) | (kpos >= ctx) |
                           (window > 0 && kpos <= qpos - window);
         sv = dead ? NEG_BIG : sv;

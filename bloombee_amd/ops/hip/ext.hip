@@ -187,59 +187,70 @@ std::vector<torch::Tensor> kv_gather(torch::Tensor k_pages, torch::Tensor v_page
 template <int D>
 static void attn_decode_launch(const torch::Tensor& q, const torch::Tensor& kp,
                                const torch::Tensor& vp, const torch::Tensor& pt,
-                               const torch::Tensor& ctx, torch::Tensor& out,
+                               const torch::Tensor& ctx,
+                               const float* alibi_ptr, torch::Tensor& out,
                                torch::Tensor& pml, torch::Tensor& pacc, int B,
-                               int Hkv, int G, int P, int maxp, int n_split,
-                               int window, float scale, long q_off, long q_sb,
-                               long q_sh, long out_sb, long out_sh) {
-  dim3 grid(B * Hkv, n_split);
-  auto launch_mfma = [&](auto maxg) {
-    attn_decode_mfma_kernel<D, decltype(maxg)::value>
+                               int Hkv, int G, int nch, int P, int maxp,
+                               int n_split, int window, float scale, long q_off,
+                               long q_sb, long q_sh, long out_sb, long out_sh) {
+  dim3 grid(B * Hkv * nch, n_split);
+  const int maxg = (G <= 4 && nch == 1) ? 4 : 16;
+  auto launch_mfma = [&](auto mg) {
+    attn_decode_mfma_kernel<D, decltype(mg)::value>
         <<<grid, 256, 0, cur_stream()>>>(
             bf_ptr(q) + q_off, bf_ptr(kp), bf_ptr(vp), pt.data_ptr<int>(),
-            ctx.data_ptr<int>(), bf_ptr_mut(out), pml.data_ptr<float>(),
-            pacc.data_ptr<float>(), B, Hkv, G, P, maxp, n_split, window, scale,
-            q_sb, q_sh, out_sb, out_sh);
+            ctx.data_ptr<int>(), alibi_ptr, bf_ptr_mut(out),
+            pml.data_ptr<float>(), pacc.data_ptr<float>(), B, Hkv, G, nch, P,
+            maxp, n_split, window, scale, q_sb, q_sh, out_sb, out_sh);
   };
-  if (G <= 4) launch_mfma(std::integral_constant<int, 4>{});
-  else if (G <= 16) launch_mfma(std::integral_constant<int, 16>{});
-  else TORCH_CHECK(false, "GQA group size > 16 unsupported: ", G);
+  if (maxg == 4) launch_mfma(std::integral_constant<int, 4>{});
+  else launch_mfma(std::integral_constant<int, 16>{});
   if (n_split > 1) {
-    attn_decode_combine_kernel<D><<<B * Hkv, G * 16, 0, cur_stream()>>>(
-        pml.data_ptr<float>(), pacc.data_ptr<float>(), bf_ptr_mut(out), Hkv, G,
-        n_split, out_sb, out_sh);
+    attn_decode_combine_kernel<D>
+        <<<B * Hkv * nch, maxg * 16, 0, cur_stream()>>>(
+            pml.data_ptr<float>(), pacc.data_ptr<float>(), bf_ptr_mut(out),
+            Hkv, G, nch, maxg, n_split, out_sb, out_sh);
   }
 }
 
 static torch::Tensor attn_decode_core(
     const torch::Tensor& q, const torch::Tensor& k_pages,
     const torch::Tensor& v_pages, const torch::Tensor& page_table,
-    const torch::Tensor& ctx_lens, double scale, long window, long n_split_req,
+    const torch::Tensor& ctx_lens, const c10::optional<torch::Tensor>& alibi,
+    double scale, long window, long n_split_req,
     int B, int Hq, int D, long q_off, long q_sb, long q_sh,
     torch::Tensor out, long out_sb, long out_sh) {
   const int Hkv = k_pages.size(1), P = k_pages.size(2), maxp = page_table.size(1);
   const int G = Hq / Hkv;
   TORCH_CHECK(Hq % Hkv == 0);
   TORCH_CHECK(D < 256 || G <= 4, "D=256 decode supports GQA group size <= 4");
+  const int nch = (G + 15) / 16;  // MQA chunks (falcon G=71 -> 5)
+  const float* alibi_ptr = nullptr;
+  if (alibi.has_value()) {
+    TORCH_CHECK(alibi->scalar_type() == at::kFloat && alibi->numel() == Hq);
+    alibi_ptr = alibi->data_ptr<float>();
+  }
   int n_split = (int)n_split_req;
   if (n_split <= 0) {
     // fill the chip: >= ~1024 workgroups (256 CUs x 8 XCDs, guide §1)
-    n_split = (int)std::max<long>(1, std::min<long>(32, 1024 / std::max(1, B * Hkv)));
+    n_split = (int)std::max<long>(
+        1, std::min<long>(32, 1024 / std::max(1, B * Hkv * nch)));
   }
+  const int maxg = (G <= 4 && nch == 1) ? 4 : 16;
   auto fopt = torch::TensorOptions().device(q.device()).dtype(at::kFloat);
   torch::Tensor pml, pacc;
   if (n_split > 1) {
-    pml = torch::empty({(long)B * Hkv * n_split, G, 2}, fopt);
-    pacc = torch::empty({(long)B * Hkv * n_split, G, D}, fopt);
+    pml = torch::empty({(long)B * Hkv * nch * n_split, maxg, 2}, fopt);
+    pacc = torch::empty({(long)B * Hkv * nch * n_split, maxg, D}, fopt);
   } else {
     pml = torch::empty({1}, fopt);
     pacc = torch::empty({1}, fopt);
   }
   auto go = [&](auto d) {
     attn_decode_launch<decltype(d)::value>(
-        q, k_pages, v_pages, page_table, ctx_lens, out, pml, pacc, B, Hkv, G,
-        P, maxp, n_split, (int)window, (float)scale, q_off, q_sb, q_sh,
-        out_sb, out_sh);
+        q, k_pages, v_pages, page_table, ctx_lens, alibi_ptr, out, pml, pacc,
+        B, Hkv, G, nch, P, maxp, n_split, (int)window, (float)scale, q_off,
+        q_sb, q_sh, out_sb, out_sh);
   };
   if (D == 128) go(std::integral_constant<int, 128>{});
   else if (D == 64) go(std::integral_constant<int, 64>{});
@@ -251,13 +262,14 @@ static torch::Tensor attn_decode_core(
 torch::Tensor attn_decode(torch::Tensor q, torch::Tensor k_pages,
                           torch::Tensor v_pages, torch::Tensor page_table,
                           torch::Tensor ctx_lens, double scale, long window,
-                          long n_split_req) {
+                          long n_split_req,
+                          c10::optional<torch::Tensor> alibi) {
   CHECK_DEV(q); CHECK_BF16(q); CHECK_CONTIG(q);
   TORCH_CHECK(q.dim() == 4 && q.size(2) == 1, "attn_decode expects (B, Hq, 1, D)");
   const int B = q.size(0), Hq = q.size(1), D = q.size(3);
   auto out = torch::empty({B, Hq, 1, D}, q.options());
-  return attn_decode_core(q, k_pages, v_pages, page_table, ctx_lens, scale,
-                          window, n_split_req, B, Hq, D,
+  return attn_decode_core(q, k_pages, v_pages, page_table, ctx_lens, alibi,
+                          scale, window, n_split_req, B, Hq, D,
                           /*q_off*/ 0, (long)Hq * D, (long)D,
                           out, (long)Hq * D, (long)D);
 }
@@ -275,28 +287,34 @@ torch::Tensor attn_decode_qkv(torch::Tensor qkv, long Hq, torch::Tensor k_pages,
   TORCH_CHECK(qkv.dim() == 3 && qkv.size(1) == 1 &&
               qkv.size(2) == (Hq + 2 * Hkv) * D, "bad fused qkv shape");
   auto out = torch::empty({B, 1, Hq * D}, qkv.options());
-  return attn_decode_core(qkv, k_pages, v_pages, page_table, ctx_lens, scale,
-                          window, n_split_req, B, (int)Hq, D,
-                          /*q_off*/ 0, (long)(Hq + 2 * Hkv) * D, (long)D,
+  return attn_decode_core(qkv, k_pages, v_pages, page_table, ctx_lens,
+                          c10::nullopt, scale, window, n_split_req, B, (int)Hq,
+                          D, /*q_off*/ 0, (long)(Hq + 2 * Hkv) * D, (long)D,
                           out, (long)Hq * D, (long)D)
       .view({B, 1, Hq * D});
 }
 
 torch::Tensor attn_prefill(torch::Tensor q, torch::Tensor k_pages,
                            torch::Tensor v_pages, torch::Tensor page_table,
-                           torch::Tensor q_start, double scale, long window) {
+                           torch::Tensor q_start, double scale, long window,
+                           c10::optional<torch::Tensor> alibi) {
   CHECK_DEV(q); CHECK_BF16(q); CHECK_CONTIG(q);
   TORCH_CHECK(q.dim() == 4, "attn_prefill expects (B, Hq, Tq, D)");
   const int B = q.size(0), Hq = q.size(1), Tq = q.size(2), D = q.size(3);
   const int Hkv = k_pages.size(1), P = k_pages.size(2), maxp = page_table.size(1);
   const int G = Hq / Hkv;
+  const float* alibi_ptr = nullptr;
+  if (alibi.has_value()) {
+    TORCH_CHECK(alibi->scalar_type() == at::kFloat && alibi->numel() == Hq);
+    alibi_ptr = alibi->data_ptr<float>();
+  }
   auto out = torch::empty_like(q);
   dim3 grid((Tq + 63) / 64, B * Hq);
   auto launch = [&](auto d) {
     attn_prefill_kernel<decltype(d)::value><<<grid, 256, 0, cur_stream()>>>(
         bf_ptr(q), bf_ptr(k_pages), bf_ptr(v_pages), page_table.data_ptr<int>(),
-        q_start.data_ptr<int>(), bf_ptr_mut(out), B, Hq, G, Tq, P, maxp,
-        (int)window, (float)scale,
+        q_start.data_ptr<int>(), alibi_ptr, bf_ptr_mut(out), B, Hq, G, Tq, P,
+        maxp, (int)window, (float)scale,
         (long)Hq * Tq * D, (long)D, (long)Tq * D,
         (long)Hq * Tq * D, (long)D, (long)Tq * D);
   };
@@ -326,8 +344,8 @@ torch::Tensor attn_prefill_qkv(torch::Tensor qkv, long Hq_,
   auto launch = [&](auto d) {
     attn_prefill_kernel<decltype(d)::value><<<grid, 256, 0, cur_stream()>>>(
         bf_ptr(qkv), bf_ptr(k_pages), bf_ptr(v_pages), page_table.data_ptr<int>(),
-        q_start.data_ptr<int>(), bf_ptr_mut(out), B, Hq, G, Tq, P, maxp,
-        (int)window, (float)scale,
+        q_start.data_ptr<int>(), nullptr, bf_ptr_mut(out), B, Hq, G, Tq, P,
+        maxp, (int)window, (float)scale,
         (long)Tq * X * D, (long)X * D, (long)D,
         (long)Tq * Hq * D, (long)Hq * D, (long)D);
   };

@@ -118,23 +118,26 @@ def kv_gather(k_pages, v_pages, page_table, ctx_len: int, batch_index: int):
 
 def attn_decode(
     q, k_pages, v_pages, page_table, ctx_lens, scale: Optional[float] = None,
-    window: int = 0, n_split: int = 0,
+    window: int = 0, n_split: int = 0, alibi_slopes=None,
 ) -> torch.Tensor:
     """Single-token paged attention. q: (B, Hq, 1, D)."""
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
     if _on_gpu(q):
         _require_ext()
+        al = (alibi_slopes.float().to(q.device).contiguous()
+              if alibi_slopes is not None else None)
         return hip_ops.attn_decode(q.contiguous(), k_pages, v_pages, page_table,
-                                   ctx_lens.int(), scale, window, n_split)
+                                   ctx_lens.int(), scale, window, n_split, al)
     q_start = ctx_lens.long() - 1
     return ref.attn_paged(q, k_pages, v_pages, page_table, q_start, scale,
-                          sliding_window=window if window > 0 else None)
+                          sliding_window=window if window > 0 else None,
+                          alibi_slopes=alibi_slopes)
 
 
 def attn_prefill(
     q, k_pages, v_pages, page_table, q_start, scale: Optional[float] = None,
-    window: int = 0,
+    window: int = 0, alibi_slopes=None,
 ) -> torch.Tensor:
     """Multi-token causal paged attention. q: (B, Hq, Tq, D); the new tokens'
     K/V must already be in the pages (kv_write first)."""
@@ -142,10 +145,13 @@ def attn_prefill(
         scale = 1.0 / math.sqrt(q.shape[-1])
     if _on_gpu(q):
         _require_ext()
+        al = (alibi_slopes.float().to(q.device).contiguous()
+              if alibi_slopes is not None else None)
         return hip_ops.attn_prefill(q.contiguous(), k_pages, v_pages, page_table,
-                                    q_start.int(), scale, window)
+                                    q_start.int(), scale, window, al)
     return ref.attn_paged(q, k_pages, v_pages, page_table, q_start, scale,
-                          sliding_window=window if window > 0 else None)
+                          sliding_window=window if window > 0 else None,
+                          alibi_slopes=alibi_slopes)
 
 
 def rope_kv_write_(qkv, Hq: int, Hkv: int, cos, sin, position_ids,
@@ -203,12 +209,7 @@ def attn_paged(q, k_pages, v_pages, page_table, q_start, scale=None, window: int
                tree_mask=None, alibi_slopes=None):
     """Unified entry: picks decode vs prefill kernel by Tq."""
     Tq = q.shape[2]
-    if alibi_slopes is not None:
-        if _on_gpu(q):
-            _require_ext()
-            raise NotImplementedError(
-                "alibi attention has no gfx950 kernel yet — bloom GPU serving "
-                "pending; CPU path is complete")
+    if alibi_slopes is not None and not _on_gpu(q):
         if scale is None:
             scale = 1.0 / math.sqrt(q.shape[-1])
         return ref.attn_paged(q, k_pages, v_pages, page_table, q_start, scale,
@@ -225,8 +226,10 @@ def attn_paged(q, k_pages, v_pages, page_table, q_start, scale=None, window: int
                                               q_start, scale, tree_mask=tree_mask)
     if Tq == 1:
         ctx_lens = q_start + 1
-        return attn_decode(q, k_pages, v_pages, page_table, ctx_lens, scale, window)
-    return attn_prefill(q, k_pages, v_pages, page_table, q_start, scale, window)
+        return attn_decode(q, k_pages, v_pages, page_table, ctx_lens, scale,
+                           window, alibi_slopes=alibi_slopes)
+    return attn_prefill(q, k_pages, v_pages, page_table, q_start, scale, window,
+                        alibi_slopes=alibi_slopes)
 
 
 def linear(x: torch.Tensor, w: torch.Tensor, residual: Optional[torch.Tensor] = None,

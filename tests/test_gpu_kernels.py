@@ -333,3 +333,59 @@ def test_family_engine_gpu_matches_cpu_tokens(model):
     # bf16 kernel vs fp32-accum reference: argmax tokens may diverge late on
     # random-init models; require the first 6 to agree exactly
     assert torch.equal(got[:, :6], want[:, :6]), (got, want)
+
+
+def test_attn_decode_alibi_parity():
+    B, Hq, Hkv, T, D = 3, 8, 8, 70, 64
+    kp, vp, pt, q, k, v, start = _paged_setup(B, Hq, Hkv, T, D, seed=13)
+    slopes = ops.alibi_slopes_for(Hq)
+    ctx = torch.full((B,), T, dtype=torch.int32)
+    want = ref.attn_paged(q[:, :, -1:], kp, vp, pt, (ctx - 1).long(),
+                          1.0 / math.sqrt(D), alibi_slopes=slopes)
+    got = ops.attn_decode(q[:, :, -1:].to(DEV), kp.to(DEV), vp.to(DEV),
+                          pt.to(DEV), ctx.to(DEV), alibi_slopes=slopes)
+    assert torch.allclose(got.cpu().float(), want.float(), atol=3e-2), \
+        (got.cpu().float() - want.float()).abs().max()
+
+
+def test_attn_prefill_alibi_parity():
+    B, Hq, Hkv, T, D = 2, 4, 4, 33, 64
+    kp, vp, pt, q, k, v, start = _paged_setup(B, Hq, Hkv, T, D, seed=14)
+    slopes = ops.alibi_slopes_for(Hq)
+    qs = torch.zeros(B, dtype=torch.int64)
+    want = ref.attn_paged(q, kp, vp, pt, qs, 1.0 / math.sqrt(D),
+                          alibi_slopes=slopes)
+    got = ops.attn_prefill(q.to(DEV), kp.to(DEV), vp.to(DEV), pt.to(DEV),
+                           qs.int().to(DEV), alibi_slopes=slopes)
+    assert torch.allclose(got.cpu().float(), want.float(), atol=3e-2), \
+        (got.cpu().float() - want.float()).abs().max()
+
+
+def test_attn_decode_wide_mqa_group():
+    """GQA group wider than the 16-row MFMA q-tile (falcon-7b G=71 class):
+    chunked launch must match the reference."""
+    B, Hq, Hkv, T, D = 2, 24, 1, 50, 64
+    kp, vp, pt, q, k, v, start = _paged_setup(B, Hq, Hkv, T, D, seed=15)
+    ctx = torch.full((B,), T, dtype=torch.int32)
+    want = ref.attn_paged(q[:, :, -1:], kp, vp, pt, (ctx - 1).long(),
+                          1.0 / math.sqrt(D))
+    for ns in (1, 4):
+        got = ops.attn_decode(q[:, :, -1:].to(DEV), kp.to(DEV), vp.to(DEV),
+                              pt.to(DEV), ctx.to(DEV), n_split=ns)
+        assert torch.allclose(got.cpu().float(), want.float(), atol=3e-2), \
+            (ns, (got.cpu().float() - want.float()).abs().max())
+
+
+def test_bloom_engine_gpu_matches_cpu_tokens():
+    from bloombee_amd.engine import LocalEngine
+
+    ids = torch.randint(0, 900, (2, 10), generator=torch.Generator().manual_seed(3))
+    cpu = LocalEngine("bloom-tiny", device="cpu", seed=0, kv_max_tokens=8192)
+    want = cpu.generate_greedy(ids, 6)
+    gpu = LocalEngine("bloom-tiny", device=DEV, seed=0, kv_max_tokens=8192)
+    gpu.embed.copy_(cpu.embed.to(DEV))
+    for gb, cb in zip(gpu.stack.blocks, cpu.stack.blocks):
+        for (n1, pg), (n2, pc) in zip(gb.named_parameters(), cb.named_parameters()):
+            pg.data.copy_(pc.data.to(DEV))
+    got = gpu.generate_greedy(ids, 6).cpu()
+    assert torch.equal(got[:, :4], want[:, :4]), (got, want)
