@@ -9,10 +9,15 @@
 #include "common.h"
 
 typedef __attribute__((ext_vector_type(4))) short short4_v;
+typedef __attribute__((ext_vector_type(2))) short short2_v;
 
 template <int E>
 DEVINL void load_bf16_e(const unsigned short* p, float* out) {
-  if constexpr (E == 4) {
+  if constexpr (E == 2) {
+    short2_v v = *reinterpret_cast<const short2_v*>(p);
+    out[0] = bf2f((unsigned short)v[0]);
+    out[1] = bf2f((unsigned short)v[1]);
+  } else if constexpr (E == 4) {
     short4_v v = *reinterpret_cast<const short4_v*>(p);
 #pragma unroll
     for (int j = 0; j < 4; ++j) out[j] = bf2f((unsigned short)v[j]);
@@ -28,7 +33,12 @@ DEVINL void load_bf16_e(const unsigned short* p, float* out) {
 
 template <int E>
 DEVINL void store_bf16_e(unsigned short* p, const float* in) {
-  if constexpr (E == 4) {
+  if constexpr (E == 2) {
+    short2_v v;
+    v[0] = (short)f2bf(in[0]);
+    v[1] = (short)f2bf(in[1]);
+    *reinterpret_cast<short2_v*>(p) = v;
+  } else if constexpr (E == 4) {
     short4_v v;
 #pragma unroll
     for (int j = 0; j < 4; ++j) v[j] = (short)f2bf(in[j]);

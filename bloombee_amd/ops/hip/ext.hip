@@ -255,6 +255,7 @@ static torch::Tensor attn_decode_core(
   if (D == 128) go(std::integral_constant<int, 128>{});
   else if (D == 64) go(std::integral_constant<int, 64>{});
   else if (D == 256) go(std::integral_constant<int, 256>{});
+  else if (D == 32) go(std::integral_constant<int, 32>{});
   else TORCH_CHECK(false, "unsupported head_dim ", D);
   return out;
 }
@@ -321,6 +322,7 @@ torch::Tensor attn_prefill(torch::Tensor q, torch::Tensor k_pages,
   if (D == 128) launch(std::integral_constant<int, 128>{});
   else if (D == 64) launch(std::integral_constant<int, 64>{});
   else if (D == 256) launch(std::integral_constant<int, 256>{});
+  else if (D == 32) launch(std::integral_constant<int, 32>{});
   else TORCH_CHECK(false, "unsupported head_dim ", D);
   return out;
 }
@@ -352,6 +354,7 @@ torch::Tensor attn_prefill_qkv(torch::Tensor qkv, long Hq_,
   if (D == 128) launch(std::integral_constant<int, 128>{});
   else if (D == 64) launch(std::integral_constant<int, 64>{});
   else if (D == 256) launch(std::integral_constant<int, 256>{});
+  else if (D == 32) launch(std::integral_constant<int, 32>{});
   else TORCH_CHECK(false, "unsupported head_dim ", D);
   return out;
 }
