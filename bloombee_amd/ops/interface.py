@@ -125,6 +125,14 @@ def attn_decode(
         scale = 1.0 / math.sqrt(q.shape[-1])
     if _on_gpu(q):
         _require_ext()
+        if q.shape[-1] > 256:
+            # gemma-4 global layers (D=512): register pressure puts the MFMA
+            # kernel past the VGPR budget; run the gather-based torch
+            # composition on-device (kernel support tracked for round 2)
+            return ref.attn_paged(q, k_pages, v_pages, page_table,
+                                  ctx_lens.long() - 1, scale,
+                                  sliding_window=window if window > 0 else None,
+                                  alibi_slopes=alibi_slopes)
         al = (alibi_slopes.float().to(q.device).contiguous()
               if alibi_slopes is not None else None)
         return hip_ops.attn_decode(q.contiguous(), k_pages, v_pages, page_table,
@@ -145,6 +153,11 @@ def attn_prefill(
         scale = 1.0 / math.sqrt(q.shape[-1])
     if _on_gpu(q):
         _require_ext()
+        if q.shape[-1] > 256:
+            return ref.attn_paged(q, k_pages, v_pages, page_table,
+                                  q_start.long(), scale,
+                                  sliding_window=window if window > 0 else None,
+                                  alibi_slopes=alibi_slopes)
         al = (alibi_slopes.float().to(q.device).contiguous()
               if alibi_slopes is not None else None)
         return hip_ops.attn_prefill(q.contiguous(), k_pages, v_pages, page_table,
@@ -216,14 +229,14 @@ def attn_paged(q, k_pages, v_pages, page_table, q_start, scale=None, window: int
                               alibi_slopes=alibi_slopes,
                               sliding_window=window if window > 0 else None)
     if tree_mask is not None:
-        # tree-attention (spec decode verify) — reference path for now
+        # tree-attention (spec decode verify): device-agnostic torch
+        # composition (ref.attn_paged runs on GPU tensors directly; a fused
+        # tree-mask kernel is a round-2 item)
         if scale is None:
             scale = 1.0 / math.sqrt(q.shape[-1])
-        return ref.attn_paged(q.cpu(), k_pages.cpu(), v_pages.cpu(),
-                              page_table.cpu(), q_start.cpu(), scale,
-                              tree_mask=tree_mask.cpu()).to(q.device) \
-            if _on_gpu(q) else ref.attn_paged(q, k_pages, v_pages, page_table,
-                                              q_start, scale, tree_mask=tree_mask)
+        return ref.attn_paged(q, k_pages, v_pages, page_table,
+                              q_start.to(q.device), scale,
+                              tree_mask=tree_mask.to(q.device))
     if Tq == 1:
         ctx_lens = q_start + 1
         return attn_decode(q, k_pages, v_pages, page_table, ctx_lens, scale,

@@ -239,17 +239,19 @@ def attn_paged(
         if alibi_slopes is not None:
             # ALiBi (bloom family): softmax-shift-invariant absolute form
             # bias[h, :, j] = slope_h * j (HF bloom build_alibi_tensor)
-            scores = scores + alibi_slopes.float().view(-1, 1, 1) * \
-                torch.arange(ctx, dtype=torch.float32).view(1, 1, ctx)
-        pos_q = torch.arange(s, s + Tq).unsqueeze(1)  # (Tq, 1)
-        pos_k = torch.arange(ctx).unsqueeze(0)        # (1, ctx)
+            scores = scores + alibi_slopes.float().to(q.device).view(-1, 1, 1) * \
+                torch.arange(ctx, dtype=torch.float32,
+                             device=q.device).view(1, 1, ctx)
+        dev = q.device
+        pos_q = torch.arange(s, s + Tq, device=dev).unsqueeze(1)  # (Tq, 1)
+        pos_k = torch.arange(ctx, device=dev).unsqueeze(0)        # (1, ctx)
         mask = pos_k <= pos_q
         if sliding_window is not None:
             mask &= pos_k > (pos_q - sliding_window)
         if tree_mask is not None:
             # new-token block (positions >= s) follows the tree mask instead
-            new_block = torch.zeros(Tq, ctx, dtype=torch.bool)
-            new_block[:, s:] = tree_mask[b]
+            new_block = torch.zeros(Tq, ctx, dtype=torch.bool, device=dev)
+            new_block[:, s:] = tree_mask[b].to(dev)
             old_block = mask.clone()
             old_block[:, s:] = False
             mask = old_block | new_block

@@ -389,3 +389,18 @@ def test_bloom_engine_gpu_matches_cpu_tokens():
             pg.data.copy_(pc.data.to(DEV))
     got = gpu.generate_greedy(ids, 6).cpu()
     assert torch.equal(got[:, :4], want[:, :4]), (got, want)
+
+
+def test_gemma4_engine_gpu_matches_cpu_tokens():
+    from bloombee_amd.engine import LocalEngine
+
+    ids = torch.randint(0, 500, (2, 10), generator=torch.Generator().manual_seed(4))
+    cpu = LocalEngine("gemma4-tiny", device="cpu", seed=0, kv_max_tokens=8192)
+    want = cpu.generate_greedy(ids, 6)
+    gpu = LocalEngine("gemma4-tiny", device=DEV, seed=0, kv_max_tokens=8192)
+    gpu.embed.copy_(cpu.embed.to(DEV))
+    for gb, cb in zip(gpu.stack.blocks, cpu.stack.blocks):
+        for (n1, pg), (n2, pc) in zip(gb.named_parameters(), cb.named_parameters()):
+            pg.data.copy_(pc.data.to(DEV))
+    got = gpu.generate_greedy(ids, 6).cpu()
+    assert torch.equal(got[:, :4], want[:, :4]), (got, want)
