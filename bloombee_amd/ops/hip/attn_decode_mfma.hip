@@ -28,8 +28,11 @@ static constexpr int DKVBLK = 32;  // positions per KV tile
 // nch: GQA group chunks per kv head (ceil(G/16)) — MQA groups wider than the
 // 16-row MFMA q-tile (falcon-7b G=71) split into chunks sharing the kv head.
 // alibi: per-query-head slopes (bloom family), or null.
+// min 2 workgroups/CU: at the natural 232-VGPR allocation the kernel sat at
+// 1 wave/SIMD with no latency hiding (measured ~67 us vs the ~19 us KV-stream
+// floor); capping registers doubles resident waves.
 template <int D, int MAXG>
-__global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
+__global__ __launch_bounds__(256, 2) void attn_decode_mfma_kernel(
     const unsigned short* __restrict__ q,        // strided, see q_sb/q_sh
     const unsigned short* __restrict__ k_pages,  // (np, Hkv, P, D)
     const unsigned short* __restrict__ v_pages,  // (np, Hkv, D, P) d-major
