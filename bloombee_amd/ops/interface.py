@@ -253,12 +253,13 @@ def linear(x: torch.Tensor, w: torch.Tensor, residual: Optional[torch.Tensor] = 
     hipBLASLt via F.linear."""
     M = x.numel() // x.shape[-1]
     N, K = w.shape[0], x.shape[-1]
-    # Measured rule (benchmarks/bench_kernels.py gemm, profiles/): the skinny
-    # kernel beats hipBLASLt where blaslt is grid/latency-bound (small N*K)
-    # and on huge-N streams (lm_head); blaslt keeps the two mid-size MLP
-    # shapes (gate_up/down) until the v3 kernel closes the last ~15%.
-    use_skinny = (N * K <= (1 << 25)) or (N >= 65536)
-    if (_on_gpu(x) and M <= 32 and x.dtype == torch.bfloat16 and use_skinny
+    # All decode-shaped GEMMs route to the skinny kernel: although hipBLASLt
+    # wins the two MLP shapes in an isolated microbench, in the real decode
+    # step it runs ~1.5x slower (cold L2 / per-stream heuristics, see
+    # profiles/), while gemm_skinny runs faster in-context AND fuses the
+    # residual/bias epilogue the blaslt path would re-emit as a separate
+    # eager kernel.
+    if (_on_gpu(x) and M <= 32 and x.dtype == torch.bfloat16
             and K % 256 == 0 and N % 64 == 0):
         _require_ext()
         return hip_ops.gemm_skinny(x.contiguous(), w, residual, bias, 0)
