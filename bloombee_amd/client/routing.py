@@ -92,6 +92,25 @@ class RemoteSequenceManager:
         with self._lock:
             self.infos = infos
         self._ready.set()
+        if self.config.routing_mode == "min_latency":
+            self._measure_rtts(infos)
+
+    def _measure_rtts(self, infos):
+        """Client->server RTTs feeding min-latency edge costs (ref
+        PingAggregator usage, sequence_manager.py:235-296)."""
+        try:
+            from bloombee_amd.utils.ping import PingAggregator
+
+            eps = {(srv.host, srv.port)
+                   for info in infos for srv in info.servers.values()}
+            if not eps:
+                return
+            if not hasattr(self, "_pinger"):
+                self._pinger = PingAggregator(timeout=2.0)
+            self._pinger.ping_many(sorted(eps))
+            self._rtts = dict(self._pinger.rtts)
+        except Exception as e:  # noqa: BLE001
+            logger.debug("rtt measurement failed: %s", e)
 
     def _update_loop(self):
         while not self._stop.is_set():
@@ -172,7 +191,10 @@ class RemoteSequenceManager:
 
     def _route_min_latency(self, spans, start_index, end_index,
                            cache_tokens_needed):
-        """Dijkstra over block boundaries (ref :235-296)."""
+        """Dijkstra over block boundaries (ref :235-296): per-block compute
+        edges 1/rps + measured client RTT/2 per hop + gossiped server-to-
+        server next_pings where available + cache-room penalty."""
+        rtts = getattr(self, "_rtts", {})
         # edges[u] = list of (v, cost, span)
         edges: Dict[int, List[Tuple[int, float, RemoteSpanInfo]]] = {}
         for s in spans:
@@ -182,9 +204,12 @@ class RemoteSequenceManager:
                     and s.server_info.cache_tokens_left is not None
                     and s.server_info.cache_tokens_left < cache_tokens_needed):
                 penalty = NO_CACHE_PENALTY_S
+            ep = (s.server_info.host, s.server_info.port)
+            rtt = rtts.get(ep)
+            net = (rtt / 2 if rtt is not None and rtt == rtt else HOP_OVERHEAD_S)
             for u in range(max(s.start, start_index), min(s.end, end_index)):
                 for v in range(u + 1, min(s.end, end_index) + 1):
-                    cost = (v - u) / max(rps, 1e-6) + HOP_OVERHEAD_S + penalty
+                    cost = (v - u) / max(rps, 1e-6) + net + penalty
                     edges.setdefault(u, []).append(
                         (v, cost, RemoteSpanInfo(s.peer_id, u, v, s.server_info)))
         dist = {start_index: 0.0}

@@ -38,5 +38,9 @@ class AutoDistributedModelForCausalLM(_AutoModelBase):
     _attr = "causal_lm_cls"
 
 
+class AutoDistributedSpeculativeModel(_AutoModelBase):
+    _attr = "speculative_cls"
+
+
 def get_block_class(model_type: str):
     return get_family(model_type).block_cls

@@ -78,6 +78,7 @@ class FamilyEntry:
     block_cls: type
     model_cls: Optional[type] = None       # client DistributedModel
     causal_lm_cls: Optional[type] = None   # client DistributedModelForCausalLM
+    speculative_cls: Optional[type] = None # client ForSpeculativeGeneration
     presets: Dict[str, dict] = field(default_factory=dict)
 
 

@@ -8,17 +8,21 @@ def _entry():
     # package import time
     from bloombee_amd.models.llama.model import (DistributedLlamaForCausalLM,
                                                  DistributedLlamaModel)
-    return DistributedLlamaModel, DistributedLlamaForCausalLM
+    from bloombee_amd.models.llama.speculative import \
+        DistributedLlamaForSpeculativeGeneration
+    return (DistributedLlamaModel, DistributedLlamaForCausalLM,
+            DistributedLlamaForSpeculativeGeneration)
 
 
 try:
-    _model_cls, _causal_cls = _entry()
+    _model_cls, _causal_cls, _spec_cls = _entry()
 except ImportError:  # pragma: no cover
-    _model_cls = _causal_cls = None
+    _model_cls = _causal_cls = _spec_cls = None
 
 register_model_family(
     "llama",
     FamilyEntry(config_cls=LlamaConfig, block_cls=LlamaBlock,
                 model_cls=_model_cls, causal_lm_cls=_causal_cls,
+                speculative_cls=_spec_cls,
                 presets=LLAMA_PRESETS),
 )

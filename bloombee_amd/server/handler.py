@@ -23,6 +23,7 @@ import torch
 from bloombee_amd.net.rpc import RpcClient, RpcServer, Stream
 from bloombee_amd.server.backend import StackBackend
 from bloombee_amd.utils.logging import get_logger
+from bloombee_amd.utils.telemetry import StageTimes
 
 logger = get_logger(__name__)
 
@@ -81,6 +82,7 @@ class ConnectionHandler:
         push_only_recv = bool(meta.get("push_only_recv"))  # inputs via rpc_push
         quiet = bool(meta.get("quiet"))  # don't echo outputs to the client
         loop = asyncio.get_event_loop()
+        times = StageTimes()
         await loop.run_in_executor(
             None, lambda: self.backend.open_session(sid, batch_size, max_length))
         try:
@@ -133,18 +135,25 @@ class ConnectionHandler:
                             sid, hidden, pos, prompts, position_ids, tree_mask,
                             speculative=spec)
 
-                out = await loop.run_in_executor(None, _step)
-                out_cpu = out.cpu()
+                with times.span("compute"):
+                    out = await loop.run_in_executor(None, _step)
+                    out_cpu = out.cpu()
+                times.bump_step()
                 if push_to is not None:
-                    await self._push_downstream(push_to, pos, out_cpu,
-                                                item_meta)
+                    with times.span("push"):
+                        await self._push_downstream(push_to, pos, out_cpu,
+                                                    item_meta)
                 if not quiet:
-                    await stream.send({"pos": pos, "step": item_meta.get("step")},
-                                      [out_cpu], codec=codec)
+                    with times.span("reply"):
+                        await stream.send(
+                            {"pos": pos, "step": item_meta.get("step")},
+                            [out_cpu], codec=codec)
                 # quiet spans send nothing: during push-mode decode the client
                 # reads outputs from the LAST span only (push_only_downstream,
                 # ref inference_session.py:178-196)
         finally:
+            if times.steps:
+                logger.info("session %s closed\n%s", sid[:8], times.table())
             await loop.run_in_executor(None,
                                        lambda: self.backend.close_session(sid))
             self._push_q.pop(sid, None)

@@ -79,6 +79,9 @@ class Server:
                                     offload_policy=offload_policy)
         self.rpc = RpcServer(host, port)
         self.handler = ConnectionHandler(self.backend, self.rpc)
+        from bloombee_amd.server.reachability import attach_reachability
+        attach_reachability(self.rpc)
+        self._next_pings = {}
         self._loop: Optional[asyncio.AbstractEventLoop] = None
         self._thread: Optional[threading.Thread] = None
         self._announcer: Optional[threading.Thread] = None
@@ -95,11 +98,33 @@ class Server:
             torch_dtype=self.config.torch_dtype,
             device=self.device,
             start_block=self.block_range[0], end_block=self.block_range[1],
+            next_pings=dict(self._next_pings),
         )
+
+    def _measure_next_pings(self) -> None:
+        """RTTs to servers hosting the blocks after ours — gossiped as
+        next_pings for the client's min-latency router (ref ModuleAnnouncer
+        next-server pings, server/server.py:957-1007)."""
+        try:
+            from bloombee_amd.utils.ping import PingAggregator
+
+            L = self.config.num_hidden_layers
+            nxt = self.block_range[1] % L
+            infos = get_remote_module_infos(
+                self.dht, [f"{self.model_name}.{nxt}"])
+            eps = [(srv.host, srv.port) for srv in infos[0].servers.values()
+                   if (srv.host, srv.port) != tuple(self.endpoint or ())]
+            if eps:
+                agg = PingAggregator(timeout=2.0)
+                agg.ping_many(eps)
+                self._next_pings = agg.to_dict()
+        except Exception as e:  # noqa: BLE001
+            logger.debug("next-ping measurement failed: %s", e)
 
     def _announce_loop(self):
         while not self._stop.is_set():
             try:
+                self._measure_next_pings()
                 declare_active_modules(self.dht, self.uids, self.peer_id,
                                        self._server_info(),
                                        time.time() + self.expiration)
