@@ -22,6 +22,7 @@ Key choices (MI355X-native):
 """
 from __future__ import annotations
 
+import contextlib
 import threading
 import time
 from dataclasses import dataclass, field
@@ -303,6 +304,84 @@ class SessionHandle:
                         vp = self.cache.v_pages(l)
                         vp[pg_d, :, :, dst % P] = vp[pg_s, :, :, src % P]
         self.commit([len(k) for k in keep])
+
+    # -- host offload (session multiplexing) ------------------------------
+    def swap_out(self) -> None:
+        """Offload this session's KV to pinned host buffers and release its
+        device pages (reference micro-batch KV offload: GPU working slots +
+        CPU snapshots on dedicated streams, memory_cache_manager.py:944-1371).
+        Another session can use the freed pages; swap_in restores the data
+        (possibly into different physical pages — the page table rewrites).
+        """
+        if getattr(self, "_swapped", None) is not None:
+            return
+        cache = self.cache
+        on_gpu = cache.device.type == "cuda"
+        stream = torch.cuda.Stream(cache.device) if on_gpu else None
+        snap = []  # per seq: per layer: (k_host, v_host) stacked page data
+        with torch.cuda.stream(stream) if on_gpu else contextlib.nullcontext():
+            for b, s in enumerate(self.seqs):
+                pages = torch.tensor(s.pages, dtype=torch.long,
+                                     device=cache.device)
+                per_layer = []
+                for l in range(cache.num_layers):
+                    k = cache.k_pages(l)[pages]
+                    kh = torch.empty_like(k, device="cpu",
+                                          pin_memory=on_gpu)
+                    kh.copy_(k, non_blocking=on_gpu)
+                    if cache.layout[l] != "k":
+                        v = cache.v_pages(l)[pages]
+                        vh = torch.empty_like(v, device="cpu",
+                                              pin_memory=on_gpu)
+                        vh.copy_(v, non_blocking=on_gpu)
+                    else:
+                        vh = None
+                    per_layer.append((kh, vh))
+                snap.append(per_layer)
+        if on_gpu:
+            stream.synchronize()
+        self._swapped = snap
+        freed = []
+        for s in self.seqs:
+            freed.extend(s.pages)
+            s.pages = list(s.pages)  # keep COUNT for re-alloc; ids stale
+        # release the physical pages but keep the reservation (the session
+        # still owns its token budget)
+        self.cache._give_pages(freed)
+
+    def swap_in(self) -> None:
+        """Restore a swapped-out session: take fresh pages, copy the host
+        snapshot back, rewrite the page table."""
+        snap = getattr(self, "_swapped", None)
+        if snap is None:
+            return
+        cache = self.cache
+        on_gpu = cache.device.type == "cuda"
+        stream = torch.cuda.Stream(cache.device) if on_gpu else None
+        with torch.cuda.stream(stream) if on_gpu else contextlib.nullcontext():
+            for b, s in enumerate(self.seqs):
+                n = len(s.pages)
+                new_pages = self.cache._take_pages(n)
+                s.pages = new_pages
+                for j, pg in enumerate(new_pages):
+                    self._page_table_host[b, j] = pg
+                idx = torch.tensor(new_pages, dtype=torch.long,
+                                   device=cache.device)
+                for l in range(cache.num_layers):
+                    kh, vh = snap[b][l]
+                    cache.k_pages(l)[idx] = kh.to(cache.device,
+                                                  non_blocking=on_gpu)
+                    if vh is not None:
+                        cache.v_pages(l)[idx] = vh.to(cache.device,
+                                                      non_blocking=on_gpu)
+        if on_gpu:
+            stream.synchronize()
+        self._table_dirty = True
+        self._swapped = None
+
+    @property
+    def is_swapped(self) -> bool:
+        return getattr(self, "_swapped", None) is not None
 
     def truncate(self, new_lengths: List[int]) -> None:
         """Failover / history-replay support: cut sequences back to

@@ -58,9 +58,28 @@ class StackBackend:
     # -- sessions ---------------------------------------------------------
     def open_session(self, session_id: str, batch_size: int, max_length: int,
                      timeout: Optional[float] = 10.0) -> None:
-        handle = self.kv_pool.allocate(batch_size, max_length, timeout=timeout)
+        from bloombee_amd.kv.paged import AllocationFailed
+
+        try:
+            handle = self.kv_pool.allocate(batch_size, max_length, timeout=0.5)
+        except AllocationFailed:
+            # KV pressure: offload the least-recently-active idle session to
+            # host (ref micro-batch KV multiplexing) and retry
+            self._swap_out_idle()
+            handle = self.kv_pool.allocate(batch_size, max_length,
+                                           timeout=timeout)
         with self._lock:
             self.sessions[session_id] = SessionState(handle)
+
+    def _swap_out_idle(self) -> None:
+        with self._lock:
+            candidates = sorted(
+                ((sid, st) for sid, st in self.sessions.items()
+                 if not st.handle.is_swapped),
+                key=lambda kv: kv[1].last_activity)
+        for sid, st in candidates[:1]:
+            logger.info("KV pressure: swapping out idle session %s", sid[:8])
+            self.pool.submit(st.handle.swap_out, PRIORITY_TRAIN).result()
 
     def close_session(self, session_id: str) -> None:
         with self._lock:
@@ -96,6 +115,8 @@ class StackBackend:
         handle = state.handle
 
         def run():
+            if handle.is_swapped:
+                handle.swap_in()
             h = hidden.to(self.device, non_blocking=True)
             if h.dtype != self.config.dtype:
                 h = h.to(self.config.dtype)

@@ -119,3 +119,48 @@ def test_independent_page_lists_per_sequence():
     assert pages[0] & pages[1] == set()
     assert pages[1] & pages[2] == set()
     h.close()
+
+
+def test_session_swap_out_in_roundtrip():
+    """KV host offload for session multiplexing (ref micro-batch KV
+    offload/prefetch): swap out -> pages reusable by others -> swap in ->
+    identical contents at (possibly) different physical pages."""
+    import torch
+
+    from bloombee_amd.kv.paged import PagedKVCache
+    from bloombee_amd.ops import reference as ref
+
+    pool = PagedKVCache(num_layers=2, num_kv_heads=2, head_dim=16,
+                        page_size=4, max_tokens=64)
+    h1 = pool.allocate(1, 16)
+    k = torch.randn(1, 2, 10, 16).to(torch.bfloat16)
+    v = torch.randn(1, 2, 10, 16).to(torch.bfloat16)
+    h1.extend(10)
+    for l in range(2):
+        ref.kv_write(k, v, h1.k_pages(l), h1.v_pages(l), h1.page_table(),
+                     torch.zeros(1, dtype=torch.int32))
+    before = [ref.kv_gather(h1.k_pages(l), h1.v_pages(l), h1.page_table(), 10, 0)
+              for l in range(2)]
+    old_pages = list(h1.seqs[0].pages)
+
+    h1.swap_out()
+    assert h1.is_swapped
+    # freed pages are usable by another session
+    h2 = pool.allocate(1, 16)
+    h2.extend(12)
+    k2 = torch.randn(1, 2, 12, 16).to(torch.bfloat16)
+    for l in range(2):
+        ref.kv_write(k2, k2, h2.k_pages(l), h2.v_pages(l), h2.page_table(),
+                     torch.zeros(1, dtype=torch.int32))
+
+    h1.swap_in()
+    assert not h1.is_swapped
+    after = [ref.kv_gather(h1.k_pages(l), h1.v_pages(l), h1.page_table(), 10, 0)
+             for l in range(2)]
+    for (kb, vb), (ka, va) in zip(before, after):
+        assert torch.equal(kb, ka)
+        assert torch.equal(vb, va)
+    # decode continues: extend past the swap
+    h1.extend(2)
+    h2.close()
+    h1.close()
