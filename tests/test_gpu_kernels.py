@@ -404,3 +404,27 @@ def test_gemma4_engine_gpu_matches_cpu_tokens():
             pg.data.copy_(pc.data.to(DEV))
     got = gpu.generate_greedy(ids, 6).cpu()
     assert torch.equal(got[:, :4], want[:, :4]), (got, want)
+
+
+def test_kv_swap_roundtrip_gpu():
+    """Pinned-host swap-out/in on device (micro-batch KV offload parity)."""
+    from bloombee_amd.kv.paged import PagedKVCache
+
+    pool = PagedKVCache(num_layers=2, num_kv_heads=2, head_dim=64,
+                        page_size=16, max_tokens=1024, device=DEV)
+    h = pool.allocate(2, 128)
+    k = torch.randn(2, 2, 40, 64).to(torch.bfloat16).to(DEV)
+    v = torch.randn(2, 2, 40, 64).to(torch.bfloat16).to(DEV)
+    h.extend(40)
+    for l in range(2):
+        ops.kv_write(k, v, h.k_pages(l), h.v_pages(l), h.page_table(),
+                     torch.zeros(2, dtype=torch.int32, device=DEV))
+    before = [ops.kv_gather(h.k_pages(l), h.v_pages(l), h.page_table(), 40, 1)
+              for l in range(2)]
+    h.swap_out()
+    h.swap_in()
+    after = [ops.kv_gather(h.k_pages(l), h.v_pages(l), h.page_table(), 40, 1)
+             for l in range(2)]
+    for (kb, vb), (ka, va) in zip(before, after):
+        assert torch.equal(kb, ka) and torch.equal(vb, va)
+    h.close()
