@@ -166,8 +166,12 @@ __global__ __launch_bounds__(256, 2) void attn_decode_mfma_kernel(
       psum = group16_reduce_sum(psum);
       l[reg] = l[reg] * corr + psum;
       m2[reg] = mn;
+      // steady-state decode rarely raises the max after the first tiles:
+      // skip the NDT-wide rescale when no lane needs it (wave-uniform branch)
+      if (__builtin_amdgcn_ballot_w64(corr < 0.9999f)) {
 #pragma unroll
-      for (int n = 0; n < NDT; ++n) acc_o[n][reg] *= corr;
+        for (int n = 0; n < NDT; ++n) acc_o[n][reg] *= corr;
+      }
     }
 
     // ---- P through LDS into A-fragment layout ----
