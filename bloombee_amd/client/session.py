@@ -221,7 +221,7 @@ class InferenceSession:
 
     # -- speculative decoding --------------------------------------------
     def spec_step(self, hidden: torch.Tensor, position_ids: torch.Tensor,
-                  tree_mask: torch.Tensor) -> torch.Tensor:
+                  tree_mask: torch.Tensor, tree: Optional[dict] = None):
         """Tree-verify step: hidden (B, T, H) linearized tree nodes with
         per-node absolute positions + ancestor mask. KV written speculatively
         on every span; follow with spec_commit(keep). Spec sessions are
@@ -232,19 +232,20 @@ class InferenceSession:
             self._open_chain(self.batch_size, replay=False)
         t = self.config.step_timeout
         out = hidden
+        keep = None
         for s in self.spans:
             async def go(s=s, out=out):
                 await s.stream.send({"pos": self.position, "spec": True,
-                                     "step": self.step_count},
+                                     "step": self.step_count, "tree": tree},
                                     [out, position_ids.int(), tree_mask])
                 item = await s.stream.recv()
                 if item is None:
                     raise RpcError(f"stream closed by {s.span.peer_id}")
-                return item[1][0]
+                return item[0].get("keep"), item[1][0]
 
-            out = run_coroutine(go(), t)
+            keep, out = run_coroutine(go(), t)
         self.step_count += 1
-        return out
+        return out, keep
 
     def spec_commit(self, keep) -> int:
         """Accept tree nodes `keep[b]` (ascending linear indices); every span

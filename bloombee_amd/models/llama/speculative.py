@@ -22,6 +22,28 @@ from bloombee_amd.utils.logging import get_logger
 logger = get_logger(__name__)
 
 
+class _RestrictedTree:
+    """Tree view whose children() hides pruned nodes (they can never be
+    accepted); tokens/probs/roots pass through."""
+
+    def __init__(self, tree: TokenTree, allowed):
+        self._t = tree
+        self._allowed = allowed
+        self.tokens = tree.tokens
+        self.probs = tree.probs
+        self.parents = tree.parents
+
+    def roots(self):
+        return [r for r in self._t.roots() if r in self._allowed]
+
+    def children(self, i):
+        return [c for c in self._t.children(i) if c in self._allowed]
+
+
+def _restrict(tree: TokenTree, allowed) -> "_RestrictedTree":
+    return _RestrictedTree(tree, allowed)
+
+
 class DistributedLlamaForSpeculativeGeneration(DistributedLlamaForCausalLM):
     """Adds `generate_speculative` on top of the distributed causal LM.
 
@@ -75,16 +97,31 @@ class DistributedLlamaForSpeculativeGeneration(DistributedLlamaForCausalLM):
                 pos_ids = tree.position_ids(prefix_len).view(1, -1)
                 mask = tree.attention_mask().unsqueeze(0)
                 h = self.embed(toks)
-                h = session.spec_step(h, pos_ids.int(), mask)
-                logits = self.lm_head(self.final_norm(h))[0].float()  # (T, V)
+                h, keep = session.spec_step(
+                    h, pos_ids.int(), mask,
+                    tree={"tokens": tree.tokens, "parents": tree.parents})
+                logits_k = self.lm_head(self.final_norm(h))[0].float()
+                if keep is not None:
+                    # last-block pruner flattened to kept rows: scatter back;
+                    # pruned nodes can never be accepted (restricted walk,
+                    # ref _restore_hidden_states, inference_session.py:696-800)
+                    T = len(tree)
+                    V = logits_k.shape[-1]
+                    logits = torch.zeros(T, V)
+                    logits[keep] = logits_k
+                    allowed = set(keep)
+                    pruned_tree = _restrict(tree, allowed)
+                else:
+                    logits = logits_k
+                    pruned_tree = tree
 
                 # ---- accept ----
                 # node 0 (pending) is committed by construction; verify walks
                 # its subtree
                 sub_accepted, bonus = (
-                    verify_tree_greedy(tree, logits, logits[0])
+                    verify_tree_greedy(pruned_tree, logits, logits[0])
                     if not do_sample else
-                    verify_tree_sampling(tree, logits, logits[0]))
+                    verify_tree_sampling(pruned_tree, logits, logits[0]))
                 accepted = [0] + sub_accepted
                 session.spec_commit([accepted])
 

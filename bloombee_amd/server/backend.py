@@ -51,6 +51,20 @@ class StackBackend:
             from bloombee_amd.offload.weights import OffloadedBlockStack
             self.stack = OffloadedBlockStack(self.stack, offload_policy)
         self.kv_pool: PagedKVCache = self.stack.make_kv(kv_max_tokens)
+        self.is_last_block = (end == config.num_hidden_layers)
+        self.pruner = None
+        if self.is_last_block:
+            from bloombee_amd.spec.pruner import (MidLMHead, PruningMethod,
+                                                  create_pruner)
+            import os as _os
+
+            method = PruningMethod(_os.environ.get("BBAMD_SPEC_PRUNING",
+                                                   "none"))
+            if method != PruningMethod.NONE:
+                head = MidLMHead(config.hidden_size, config.vocab_size,
+                                 seed=seed, dtype=config.dtype, device=device)
+                self.pruner = create_pruner(method, head)
+                logger.info("mid-network tree pruner enabled: %s", method)
         self.pool = TaskPool(name=f"worker[{start}:{end}]")
         self.sessions: Dict[str, SessionState] = {}
         self._lock = threading.Lock()
@@ -141,7 +155,24 @@ class StackBackend:
                    if position_ids is not None else None)
             tm = tree_mask.to(self.device) if tree_mask is not None else None
             out = self.stack.forward_inference(h, handle, sp, pos, tree_mask=tm)
+            from bloombee_amd.utils import activation_dumper
+            if activation_dumper.enabled():
+                activation_dumper.capture_activation(
+                    f"blocks{self.start}_{self.end}", out)
             return out
+
+        return self.pool.submit(run, PRIORITY_INFERENCE).result()
+
+    def prune_tree(self, hidden: torch.Tensor, tokens: list,
+                   parents: list):
+        """Last-block mid-network pruning (ref backend.py:763-777): returns
+        (kept_hidden, keep_indices) or (hidden, None) when disabled."""
+        if self.pruner is None or hidden.shape[0] != 1:
+            return hidden, None
+
+        def run():
+            keep = self.pruner.keep_indices(hidden[0], tokens, parents)
+            return hidden[:, keep], keep
 
         return self.pool.submit(run, PRIORITY_INFERENCE).result()
 

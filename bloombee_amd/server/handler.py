@@ -137,7 +137,14 @@ class ConnectionHandler:
 
                 with times.span("compute"):
                     out = await loop.run_in_executor(None, _step)
-                    out_cpu = out.cpu()
+                keep = None
+                tree = item_meta.get("tree")
+                if (spec and tree is not None
+                        and self.backend.pruner is not None):
+                    out, keep = await loop.run_in_executor(
+                        None, lambda: self.backend.prune_tree(
+                            out, tree["tokens"], tree["parents"]))
+                out_cpu = out.cpu()
                 times.bump_step()
                 if push_to is not None:
                     with times.span("push"):
@@ -146,7 +153,8 @@ class ConnectionHandler:
                 if not quiet:
                     with times.span("reply"):
                         await stream.send(
-                            {"pos": pos, "step": item_meta.get("step")},
+                            {"pos": pos, "step": item_meta.get("step"),
+                             "keep": keep},
                             [out_cpu], codec=codec)
                 # quiet spans send nothing: during push-mode decode the client
                 # reads outputs from the LAST span only (push_only_downstream,

@@ -175,3 +175,46 @@ def test_speculative_swarm_matches_greedy():
         s1.shutdown()
         s2.shutdown()
         boot.shutdown()
+
+
+def test_speculative_with_pruner_still_greedy(monkeypatch):
+    """Mid-network pruning at the last block: pruned subtrees must only
+    REDUCE acceptance, never change emitted tokens (greedy invariant holds
+    because pruned nodes are simply never accepted)."""
+    import os
+
+    from bloombee_amd.client import ClientConfig
+    from bloombee_amd.engine import LocalEngine
+    from bloombee_amd.models.llama.speculative import \
+        DistributedLlamaForSpeculativeGeneration
+    from bloombee_amd.net.dht import Dht
+    from bloombee_amd.server import Server
+
+    monkeypatch.setenv("BBAMD_SPEC_PRUNING", "probability")
+    boot = Dht()
+    s1 = Server("llama-tiny", initial_peers=[boot.endpoint], block_indices=(0, 2),
+                device="cpu", seed=0, kv_max_tokens=1 << 14, update_period=5.0)
+    s2 = Server("llama-tiny", initial_peers=[boot.endpoint], block_indices=(2, 4),
+                device="cpu", seed=0, kv_max_tokens=1 << 14, update_period=5.0)
+    assert s2.backend.pruner is not None  # last-block server got the pruner
+    s1.run_in_background()
+    s2.run_in_background()
+    try:
+        cfg = ClientConfig(initial_peers=[boot.endpoint])
+        model = DistributedLlamaForSpeculativeGeneration.from_pretrained(
+            "llama-tiny", client_config=cfg, seed=0)
+        draft = LocalEngine("llama-tiny", device="cpu", seed=7,
+                            kv_max_tokens=1 << 13)
+        model.set_drafter(draft, node_budget=6, max_depth=3)
+        gen = torch.Generator().manual_seed(5)
+        prompt = torch.randint(0, 1000, (1, 7), generator=gen)
+        out = model.generate_speculative(prompt, max_new_tokens=8)
+        eng = LocalEngine("llama-tiny", device="cpu", seed=0,
+                          kv_max_tokens=1 << 14)
+        expect = eng.generate_greedy(prompt, 8)
+        assert torch.equal(out, expect), (out, expect)
+        model.remote.manager.shutdown()
+    finally:
+        s1.shutdown()
+        s2.shutdown()
+        boot.shutdown()

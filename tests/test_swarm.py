@@ -184,3 +184,20 @@ def test_swarm_compressed_wire_matches_local(swarm):
     expect = _local_tokens(prompt, 6)
     assert torch.equal(out[:, 7:], expect)  # codec is lossless
     model.remote.manager.shutdown()
+
+
+def test_multi_call_generate_session_reuse(swarm):
+    """Session resume across .generate() calls (ref RemotePastKeyValues
+    multi-call resume, remote_generation.py:183-216)."""
+    boot, _ = swarm
+    model = _make_model(boot)
+    gen = torch.Generator().manual_seed(5)
+    prompt = torch.randint(0, 1000, (1, 7), generator=gen)
+    sess = model.remote.inference_session(64)
+    out1 = model.generate(prompt, max_new_tokens=3, session=sess)
+    out2 = model.generate(out1[:, -1:], max_new_tokens=3, session=sess)
+    sess.close()
+    expect = _local_tokens(prompt, 6)
+    got = torch.cat([out1[:, 7:], out2[:, 1:]], dim=1)
+    assert torch.equal(got, expect), (got, expect)
+    model.remote.manager.shutdown()
