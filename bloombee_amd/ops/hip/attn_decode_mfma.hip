@@ -117,8 +117,9 @@ __global__ __launch_bounds__(256, 2) void attn_decode_mfma_kernel(
     for (int n = 0; n < NDT; ++n) {
       vfrag[n] = as_bf16x8(short8{});
       if (vpage >= 0)
-        vfrag[n] = as_bf16x8(*reinterpret_cast<const short8*>(
-            v_pages + (vrow + li + 16 * n) * P + (vpos % P)));
+        vfrag[n] = as_bf16x8(__builtin_nontemporal_load(
+            reinterpret_cast<const short8*>(
+                v_pages + (vrow + li + 16 * n) * P + (vpos % P))));
     }
 
     // ---- S = Q K^T : K fragments straight from the paged pool ----
@@ -133,8 +134,10 @@ __global__ __launch_bounds__(256, 2) void attn_decode_mfma_kernel(
       s[n] = (f32x4){0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int kk = 0; kk < NKK; ++kk) {
-        bf16x8 bfrag = as_bf16x8(
-            *reinterpret_cast<const short8*>(krow + hi * 8 + 32 * kk));
+        // KV is read once per decode step: nontemporal keeps L2 for the
+        // GEMM weight streams that follow in the same step
+        bf16x8 bfrag = as_bf16x8(__builtin_nontemporal_load(
+            reinterpret_cast<const short8*>(krow + hi * 8 + 32 * kk)));
         s[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[kk], bfrag, s[n],
                                                        0, 0, 0);
       }
