@@ -114,7 +114,8 @@ class StackBackend:
                        start_pos: int, prompts: Optional[torch.Tensor] = None,
                        position_ids: Optional[torch.Tensor] = None,
                        tree_mask: Optional[torch.Tensor] = None,
-                       speculative: bool = False) -> torch.Tensor:
+                       speculative: bool = False,
+                       batch_offset: Optional[int] = None) -> torch.Tensor:
         """One decode/prefill step for an open session.
 
         start_pos: absolute position of hidden[:, 0]. If the session has
@@ -140,6 +141,21 @@ class StackBackend:
                 h = h + prompts.to(h.device, h.dtype)
             B, T, _ = h.shape
             cur = handle.seqs[0].l_acc
+            if batch_offset is not None:
+                # micro-batch slice of the session (server-side split or
+                # upstream per-MB push): the FIRST slice of a step extends
+                # the whole session; later slices reuse the extension
+                if handle.seqs[0].l_spec == start_pos:
+                    handle.extend(T)
+                elif handle.seqs[0].l_spec != start_pos + T:
+                    raise ValueError(
+                        f"micro-batch at position {start_pos} inconsistent "
+                        f"with cache length {handle.seqs[0].l_spec}")
+                from bloombee_amd.kv.views import SessionView
+                view = SessionView(handle, batch_offset, batch_offset + B)
+                sp = torch.full((B,), start_pos, dtype=torch.int32,
+                                device=self.device)
+                return self.stack.forward_inference(h, view, sp)
             if start_pos < cur:
                 handle.truncate([min(start_pos, s.l_acc) for s in handle.seqs])
                 handle.rollback()

@@ -28,13 +28,49 @@ from bloombee_amd.utils.telemetry import StageTimes
 logger = get_logger(__name__)
 
 
+class AdaptivePushConcurrency:
+    """Self-tuning in-flight limit for server->server pushes (parity:
+    reference handler.py:255-370): grows the window while pushes succeed
+    quickly, shrinks it multiplicatively on failures/slowdowns."""
+
+    def __init__(self, initial: int = 2, lo: int = 1, hi: int = 16,
+                 slow_s: float = 0.25):
+        self.limit = initial
+        self.lo, self.hi = lo, hi
+        self.slow_s = slow_s
+        self._sem = asyncio.Semaphore(initial)
+
+    async def __aenter__(self):
+        await self._sem.acquire()
+        return self
+
+    async def __aexit__(self, *exc):
+        self._sem.release()
+
+    def feedback(self, elapsed_s: float, ok: bool):
+        old = self.limit
+        if not ok or elapsed_s > self.slow_s:
+            self.limit = max(self.lo, self.limit // 2)
+        elif elapsed_s < self.slow_s / 4:
+            self.limit = min(self.hi, self.limit + 1)
+        for _ in range(self.limit - old):
+            self._sem.release()
+        for _ in range(old - self.limit):
+            # shrink lazily: steal permits as they come back
+            asyncio.ensure_future(self._sem.acquire())
+
+
 class ConnectionHandler:
+    MICROBATCH_MIN_BATCH = 8     # split threshold (ref should_split_batch)
+    MICROBATCH_SIZE = 4          # sequences per micro-batch (ref default 2-4)
+
     def __init__(self, backend: StackBackend, server: RpcServer):
         self.backend = backend
         self.rpc = server
         # (session_id, step) -> queued pushed inputs awaiting the local stream
         self._push_q: Dict[str, asyncio.Queue] = {}
         self._peers: Dict[Tuple[str, int], RpcClient] = {}
+        self._push_limiter = AdaptivePushConcurrency()
         server.register("rpc_info", self.rpc_info)
         server.register("rpc_forward", self.rpc_forward)
         server.register("rpc_backward", self.rpc_backward)
@@ -83,6 +119,7 @@ class ConnectionHandler:
         quiet = bool(meta.get("quiet"))  # don't echo outputs to the client
         loop = asyncio.get_event_loop()
         times = StageTimes()
+        mb_buffers: Dict[int, dict] = {}  # pos -> {offset: out_cpu}
         await loop.run_in_executor(
             None, lambda: self.backend.open_session(sid, batch_size, max_length))
         try:
@@ -135,6 +172,67 @@ class ConnectionHandler:
                             sid, hidden, pos, prompts, position_ids, tree_mask,
                             speculative=spec)
 
+                mbinfo = item_meta.get("mb")
+                if mbinfo is not None and not spec:
+                    # pushed micro-batch slice: compute immediately (cross-
+                    # stage overlap, ref handler.py:1677-1847 push-preferred
+                    # merge), forward the slice downstream, reply merged once
+                    # every slice of this step has arrived
+                    off = int(mbinfo["offset"])
+                    part = await loop.run_in_executor(
+                        None, lambda: self.backend.inference_step(
+                            sid, hidden, pos, prompts, batch_offset=off))
+                    part_cpu = part.cpu()
+                    if push_to is not None:
+                        asyncio.ensure_future(self._push_downstream(
+                            push_to, pos, part_cpu, item_meta, mb=mbinfo))
+                    buf = mb_buffers.setdefault(pos, {})
+                    buf[off] = part_cpu
+                    got = sum(t.shape[0] for t in buf.values())
+                    times.bump_step()
+                    if got >= int(mbinfo["total"]):
+                        out_cpu = torch.cat(
+                            [buf[o] for o in sorted(buf)], dim=0)
+                        del mb_buffers[pos]
+                        if not quiet:
+                            await stream.send(
+                                {"pos": pos, "step": item_meta.get("step"),
+                                 "keep": None}, [out_cpu], codec=codec)
+                    continue
+                B = hidden.shape[0]
+                can_split = (not spec and push_to is not None
+                             and B >= self.MICROBATCH_MIN_BATCH
+                             and B % self.MICROBATCH_SIZE == 0)
+                if can_split:
+                    # reference micro-batch overlap (block_functions.py:
+                    # 2055-2460): compute micro-batch j while j-1 is in
+                    # flight downstream; the client reply carries the merged
+                    # batch from the LAST span as usual
+                    mbs = self.MICROBATCH_SIZE
+                    outs = []
+                    push_tasks = []
+                    with times.span("compute"):
+                        for j in range(0, B, mbs):
+                            part = await loop.run_in_executor(
+                                None, lambda j=j: self.backend.inference_step(
+                                    sid, hidden[j:j + mbs], pos, prompts,
+                                    batch_offset=j))
+                            part_cpu = part.cpu()
+                            outs.append(part_cpu)
+                            push_tasks.append(asyncio.ensure_future(
+                                self._push_downstream(
+                                    push_to, pos, part_cpu, item_meta,
+                                    mb={"offset": j, "total": B})))
+                        out_cpu = torch.cat(outs, dim=0)
+                        for t_ in push_tasks:
+                            await t_
+                    times.bump_step()
+                    if not quiet:
+                        with times.span("reply"):
+                            await stream.send(
+                                {"pos": pos, "step": item_meta.get("step"),
+                                 "keep": None}, [out_cpu], codec=codec)
+                    continue
                 with times.span("compute"):
                     out = await loop.run_in_executor(None, _step)
                 keep = None
@@ -171,15 +269,24 @@ class ConnectionHandler:
                 pass
 
     async def _push_downstream(self, push_to, pos: int, hidden: torch.Tensor,
-                               item_meta: dict) -> None:
+                               item_meta: dict, mb: Optional[dict] = None,
+                               ) -> None:
+        import time as _time
+
         host, port, down_sid = push_to[0], int(push_to[1]), push_to[2]
         key = (host, port)
         if key not in self._peers:
             self._peers[key] = RpcClient(host, port)
+        t0 = _time.monotonic()
+        ok = True
         try:
-            await self._peers[key].call(
-                "rpc_push", {"session_id": down_sid, "pos": pos,
-                             "step": item_meta.get("step")}, [hidden],
-                timeout=30)
+            async with self._push_limiter:
+                await self._peers[key].call(
+                    "rpc_push", {"session_id": down_sid, "pos": pos,
+                                 "step": item_meta.get("step"),
+                                 "mb": mb}, [hidden],
+                    timeout=30)
         except Exception as e:  # noqa: BLE001 — client will fall back
+            ok = False
             logger.warning("s2s push to %s failed: %s", push_to, e)
+        self._push_limiter.feedback(_time.monotonic() - t0, ok)

@@ -201,3 +201,17 @@ def test_multi_call_generate_session_reuse(swarm):
     got = torch.cat([out1[:, 7:], out2[:, 1:]], dim=1)
     assert torch.equal(got, expect), (got, expect)
     model.remote.manager.shutdown()
+
+
+def test_swarm_microbatch_split_matches_local(swarm):
+    """Batch >= split threshold: servers split into micro-batches and push
+    slice-by-slice downstream (cross-stage overlap); outputs must still be
+    exact (ref block_functions.py:2055-2460)."""
+    boot, _ = swarm
+    model = _make_model(boot)
+    gen = torch.Generator().manual_seed(11)
+    prompt = torch.randint(0, 1000, (8, 6), generator=gen)
+    out = model.generate(prompt, max_new_tokens=5)
+    expect = _local_tokens(prompt, 5)
+    assert torch.equal(out[:, 6:], expect), (out[:, 6:], expect)
+    model.remote.manager.shutdown()
