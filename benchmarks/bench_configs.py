@@ -1,0 +1,165 @@
+"""BASELINE.json progression-config evidence at single-GPU scale.
+
+The driver's bench.py measures config 2 (llama-3-8b pipeline decode). This
+harness exercises the remaining configs' machinery on one MI355X:
+
+  offload  — llama-2-70b-class shard with FlexGen-style host offload
+             (weight_gpu_percent, 4-bit host compression optional)
+  mixtral  — MoE decode (skinny grouped expert GEMMs)
+  spec     — tree speculative decoding, local target+draft, acceptance rate
+
+    python benchmarks/bench_configs.py offload mixtral spec
+"""
+from __future__ import annotations
+
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch  # noqa: E402
+
+DEV = "cuda:0" if torch.cuda.is_available() else "cpu"
+
+
+def _decode_loop(eng, B, prompt_len, steps):
+    ids = torch.randint(0, eng.config.vocab_size, (B, prompt_len),
+                        generator=torch.Generator().manual_seed(0))
+    kv = eng.kv_pool.allocate(B, prompt_len + steps + 4)
+    tok = eng.prefill(ids, kv)
+    if DEV.startswith("cuda"):
+        torch.cuda.synchronize()
+    t0 = time.monotonic()
+    for _ in range(steps):
+        tok = eng.decode_step(tok, kv)
+    if DEV.startswith("cuda"):
+        torch.cuda.synchronize()
+    dt = time.monotonic() - t0
+    kv.close()
+    return B * steps / dt, dt / steps * 1000
+
+
+def bench_offload(steps=16):
+    """llama-2-70b shard (10 blocks ~17 GB bf16) with 50% of blocks streamed
+    from pinned host memory through the double-buffered arena."""
+    from bloombee_amd.engine import BlockStack
+    from bloombee_amd.models.base import resolve_config
+    from bloombee_amd.offload import OffloadPolicy, OffloadedBlockStack
+
+    cfg = resolve_config("llama-2-70b")
+    nblocks = 10
+    stack = BlockStack(cfg, 0, nblocks, device=DEV, seed=0)
+    off = OffloadedBlockStack(stack, OffloadPolicy(weight_gpu_percent=50.0))
+    kv = off.make_kv(1 << 16)
+    B, T, steps_ = 8, 128, steps
+    h = kv.allocate(B, T + steps_ + 4)
+    x = torch.randint(0, cfg.vocab_size, (B, T))
+    hid = (torch.randn(B, T, cfg.hidden_size) * 0.02).to(cfg.dtype).to(DEV)
+    h.extend(T)
+    off.forward_inference(hid, h, torch.zeros(B, dtype=torch.int32, device=DEV))
+    if DEV.startswith("cuda"):
+        torch.cuda.synchronize()
+    t0 = time.monotonic()
+    one = (torch.randn(B, 1, cfg.hidden_size) * 0.02).to(cfg.dtype).to(DEV)
+    for i in range(steps_):
+        sp = torch.full((B,), T + i, dtype=torch.int32, device=DEV)
+        h.extend(1)
+        off.forward_inference(one, h, sp)
+    if DEV.startswith("cuda"):
+        torch.cuda.synchronize()
+    dt = (time.monotonic() - t0) / steps_
+    h.close()
+    print(json.dumps({"config": "llama-2-70b shard offload",
+                      "blocks": nblocks, "resident": off.resident,
+                      "ms_per_step": round(dt * 1e3, 2),
+                      "note": "weights streamed host->HBM each step for "
+                              "offloaded blocks"}))
+
+
+def bench_mixtral(steps=24):
+    from bloombee_amd.engine import LocalEngine
+
+    eng = LocalEngine("mixtral-tiny" if DEV == "cpu" else "mixtral-8x7b-4l",
+                      device=DEV, seed=0, kv_max_tokens=1 << 16)
+    tps, ms = _decode_loop(eng, B=16, prompt_len=64, steps=steps)
+    print(json.dumps({"config": "mixtral MoE decode", "tokens_per_s": round(tps, 1),
+                      "ms_per_step": round(ms, 2)}))
+
+
+def bench_spec(steps=32, self_draft=False):
+    """Local speculative decoding: llama-3-8b target + tiny draft; reports
+    tokens/round and acceptance."""
+    from bloombee_amd.engine import LocalEngine
+    from bloombee_amd.spec.drafter import MultiDrafter
+    from bloombee_amd.spec.tree import TokenTree
+    from bloombee_amd.spec.verify import verify_tree_greedy
+
+    tgt = LocalEngine("llama-3-8b" if DEV != "cpu" else "llama-tiny",
+                      device=DEV, seed=0, kv_max_tokens=1 << 15)
+    # self_draft: draft == target (upper-bound acceptance; random-init models
+    # with different weights rarely agree, so the realistic-acceptance case
+    # needs real checkpoints this offline environment lacks)
+    if self_draft:
+        draft = LocalEngine("llama-3-8b" if DEV != "cpu" else "llama-tiny",
+                            device=DEV, seed=0, kv_max_tokens=1 << 14)
+    else:
+        draft = LocalEngine("llama-mini-gpu" if DEV != "cpu" else "llama-tiny",
+                            device=DEV, seed=3, kv_max_tokens=1 << 14)
+    drafter = MultiDrafter(draft, node_budget=8, max_depth=4)
+    prompt = torch.randint(0, 1000, (1, 32),
+                           generator=torch.Generator().manual_seed(1))
+    kv = tgt.kv_pool.allocate(1, 512)
+    tok = tgt.prefill(prompt, kv)
+    history = prompt[0].tolist()
+    pending = int(tok)
+    emitted = 0
+    rounds = 0
+    t0 = time.monotonic()
+    while emitted < steps:
+        sub = drafter.build_tree(torch.tensor(history + [pending]))
+        tree = TokenTree()
+        tree.add(pending, -1, 1.0)
+        for i in range(len(sub)):
+            tree.add(sub.tokens[i],
+                     0 if sub.parents[i] == -1 else sub.parents[i] + 1,
+                     sub.probs[i])
+        prefix = kv.seqs[0].l_acc
+        toks = tree.token_tensor().view(1, -1).to(tgt.device)
+        pos = tree.position_ids(prefix).view(1, -1)
+        mask = tree.attention_mask().unsqueeze(0)
+        kv.rollback()
+        kv.extend(len(tree), speculative=True)
+        hid = tgt._embed(toks)
+        sp = torch.full((1,), prefix, dtype=torch.int32, device=tgt.device)
+        h = tgt.stack.forward_inference(hid, kv, sp, pos.int().to(tgt.device),
+                                        tree_mask=mask.to(tgt.device))
+        logits = tgt.logits_for(h[0]).float().cpu()
+        acc, bonus = verify_tree_greedy(tree, logits, logits[0], start=0)
+        accepted = [0] + acc
+        kv.reorder_and_commit([accepted])
+        emit = [tree.tokens[i] for i in accepted]
+        history += emit
+        emitted += len(emit)
+        pending = bonus
+        rounds += 1
+        drafter.record_result(len(acc), offered_depth=4)
+    if DEV.startswith("cuda"):
+        torch.cuda.synchronize()
+    dt = time.monotonic() - t0
+    kv.close()
+    print(json.dumps({"config": "speculative decode (tree verify)"
+                                + (" self-draft" if self_draft else ""),
+                      "tokens": emitted, "rounds": rounds,
+                      "tokens_per_round": round(emitted / rounds, 2),
+                      "tokens_per_s": round(emitted / dt, 1)}))
+
+
+ALL = {"offload": bench_offload, "mixtral": bench_mixtral,
+       "spec": bench_spec,
+       "spec_selfdraft": lambda: bench_spec(self_draft=True)}
+
+if __name__ == "__main__":
+    for name in (sys.argv[1:] or list(ALL)):
+        ALL[name]()

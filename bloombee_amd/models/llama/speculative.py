@@ -119,9 +119,10 @@ class DistributedLlamaForSpeculativeGeneration(DistributedLlamaForCausalLM):
                 # node 0 (pending) is committed by construction; verify walks
                 # its subtree
                 sub_accepted, bonus = (
-                    verify_tree_greedy(pruned_tree, logits, logits[0])
+                    verify_tree_greedy(pruned_tree, logits, logits[0], start=0)
                     if not do_sample else
-                    verify_tree_sampling(pruned_tree, logits, logits[0]))
+                    verify_tree_sampling(pruned_tree, logits, logits[0],
+                                         start=0))
                 accepted = [0] + sub_accepted
                 session.spec_commit([accepted])
 

@@ -11,6 +11,7 @@ from bloombee_amd.spec.tree import TokenTree
 
 def verify_tree_greedy(tree: TokenTree, logits: torch.Tensor,
                        prefix_logits: torch.Tensor,
+                       start: Optional[int] = None,
                        ) -> Tuple[List[int], int]:
     """Greedy target verification.
 
@@ -25,7 +26,7 @@ def verify_tree_greedy(tree: TokenTree, logits: torch.Tensor,
     the bonus token (ref verify_path :58-106).
     """
     accepted: List[int] = []
-    cur_children = tree.roots()
+    cur_children = tree.children(start) if start is not None else tree.roots()
     cur_logits = prefix_logits
     while True:
         want = int(cur_logits.argmax(-1))
@@ -41,6 +42,7 @@ def verify_tree_sampling(tree: TokenTree, logits: torch.Tensor,
                          prefix_logits: torch.Tensor,
                          generator: Optional[torch.Generator] = None,
                          temperature: float = 1.0,
+                         start: Optional[int] = None,
                          ) -> Tuple[List[int], int]:
     """SpecInfer-style stochastic verification (ref verify_edge :108-154).
 
@@ -51,7 +53,7 @@ def verify_tree_sampling(tree: TokenTree, logits: torch.Tensor,
     residual distribution.
     """
     accepted: List[int] = []
-    cur_children = tree.roots()
+    cur_children = tree.children(start) if start is not None else tree.roots()
     cur_logits = prefix_logits
     while True:
         p = torch.softmax(cur_logits.float() / max(temperature, 1e-6), -1)
