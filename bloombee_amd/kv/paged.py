@@ -203,7 +203,11 @@ class SessionHandle:
         self._page_table = torch.zeros(
             batch_size, self._max_pages, device=cache.device, dtype=torch.int32
         )
-        self._page_table_host = torch.zeros(batch_size, self._max_pages, dtype=torch.int32)
+        # pinned host mirror: the dirty-flush H2D is async on the compute
+        # stream; pageable async copies are a correctness hazard
+        self._page_table_host = torch.zeros(
+            batch_size, self._max_pages, dtype=torch.int32,
+            pin_memory=(cache.device.type == "cuda"))
         self._table_dirty = False
         self._closed = False
 

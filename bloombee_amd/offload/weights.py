@@ -139,6 +139,9 @@ class OffloadedBlockStack:
                     self._copy_stream.wait_event(done)
                 for name, hp in self._host[i].items():
                     hp.copy_into(arena[name])
+                    # arena tensors were allocated on the default stream; tell
+                    # the caching allocator they are used on the copy stream
+                    arena[name].record_stream(self._copy_stream)
                 ev = torch.cuda.Event()
                 ev.record(self._copy_stream)
                 self._events[i] = ev
