@@ -23,6 +23,11 @@
 typedef __attribute__((ext_vector_type(8))) short short8_h;
 
 #define CHECK_DEV(x) TORCH_CHECK(x.is_cuda(), #x " must be on device")
+// a CPU tensor pointer reaching a kernel is a GPU memory fault at a random
+// later point — check EVERY tensor argument, not just the big ones
+#define CHECK_DEV_ALL1(a) CHECK_DEV(a)
+#define CHECK_DEV_ALL2(a, b) { CHECK_DEV(a); CHECK_DEV(b); }
+#define CHECK_DEV_ALL3(a, b, c) { CHECK_DEV(a); CHECK_DEV(b); CHECK_DEV(c); }
 #define CHECK_BF16(x) TORCH_CHECK(x.scalar_type() == at::kBFloat16, #x " must be bf16")
 #define CHECK_CONTIG(x) TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
 
@@ -100,6 +105,7 @@ torch::Tensor layer_norm(torch::Tensor x, torch::Tensor w,
 void rope_apply_(torch::Tensor q, torch::Tensor k, torch::Tensor cos_t,
                  torch::Tensor sin_t, torch::Tensor pos) {
   CHECK_DEV(q); CHECK_BF16(q); CHECK_CONTIG(q); CHECK_CONTIG(k);
+  CHECK_DEV_ALL3(cos_t, sin_t, pos);
   TORCH_CHECK(pos.scalar_type() == at::kInt, "pos must be int32");
   const int B = q.size(0), Hq = q.size(1), T = q.size(2), D = q.size(3);
   const int Hkv = k.size(1);
@@ -153,6 +159,7 @@ void kv_write(torch::Tensor k_new, torch::Tensor v_new, torch::Tensor k_pages,
   CHECK_DEV(k_new); CHECK_BF16(k_new); CHECK_CONTIG(k_new); CHECK_CONTIG(v_new);
   CHECK_CONTIG(k_pages); CHECK_CONTIG(v_pages);
   TORCH_CHECK(page_table.scalar_type() == at::kInt && start_pos.scalar_type() == at::kInt);
+  CHECK_DEV_ALL2(page_table, start_pos);
   const int B = k_new.size(0), Hkv = k_new.size(1), T = k_new.size(2), D = k_new.size(3);
   const int P = k_pages.size(2), maxp = page_table.size(1);
   TORCH_CHECK(k_pages.size(1) == Hkv && k_pages.size(3) == D);
@@ -167,7 +174,7 @@ void kv_write(torch::Tensor k_new, torch::Tensor v_new, torch::Tensor k_pages,
 std::vector<torch::Tensor> kv_gather(torch::Tensor k_pages, torch::Tensor v_pages,
                                      torch::Tensor page_table, long batch_index,
                                      long ctx) {
-  CHECK_DEV(k_pages);
+  CHECK_DEV(k_pages); CHECK_DEV(page_table);
   const int Hkv = k_pages.size(1), P = k_pages.size(2), D = k_pages.size(3);
   auto k = torch::empty({Hkv, ctx, D}, k_pages.options());
   auto v = torch::empty({Hkv, ctx, D}, v_pages.options());
@@ -268,6 +275,8 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor k_pages,
                           long n_split_req,
                           c10::optional<torch::Tensor> alibi) {
   CHECK_DEV(q); CHECK_BF16(q); CHECK_CONTIG(q);
+  CHECK_DEV_ALL3(k_pages, page_table, ctx_lens);
+  if (alibi.has_value()) CHECK_DEV((*alibi));
   TORCH_CHECK(q.dim() == 4 && q.size(2) == 1, "attn_decode expects (B, Hq, 1, D)");
   const int B = q.size(0), Hq = q.size(1), D = q.size(3);
   auto out = torch::empty({B, Hq, 1, D}, q.options());
@@ -284,6 +293,7 @@ torch::Tensor attn_decode_qkv(torch::Tensor qkv, long Hq, torch::Tensor k_pages,
                               torch::Tensor ctx_lens, double scale, long window,
                               long n_split_req) {
   CHECK_DEV(qkv); CHECK_BF16(qkv); CHECK_CONTIG(qkv);
+  CHECK_DEV_ALL3(k_pages, page_table, ctx_lens);
   const int Hkv = k_pages.size(1);
   const int D = k_pages.size(3);
   const int B = qkv.size(0);
@@ -302,6 +312,8 @@ torch::Tensor attn_prefill(torch::Tensor q, torch::Tensor k_pages,
                            torch::Tensor q_start, double scale, long window,
                            c10::optional<torch::Tensor> alibi) {
   CHECK_DEV(q); CHECK_BF16(q); CHECK_CONTIG(q);
+  CHECK_DEV_ALL3(k_pages, page_table, q_start);
+  if (alibi.has_value()) CHECK_DEV((*alibi));
   TORCH_CHECK(q.dim() == 4, "attn_prefill expects (B, Hq, Tq, D)");
   const int B = q.size(0), Hq = q.size(1), Tq = q.size(2), D = q.size(3);
   const int Hkv = k_pages.size(1), P = k_pages.size(2), maxp = page_table.size(1);
@@ -336,6 +348,7 @@ torch::Tensor attn_prefill_qkv(torch::Tensor qkv, long Hq_,
                                torch::Tensor page_table, torch::Tensor q_start,
                                double scale, long window) {
   CHECK_DEV(qkv); CHECK_BF16(qkv); CHECK_CONTIG(qkv);
+  CHECK_DEV_ALL3(k_pages, page_table, q_start);
   const int Hkv = k_pages.size(1), P = k_pages.size(2), D = k_pages.size(3);
   const int maxp = page_table.size(1);
   const int Hq = (int)Hq_;
@@ -367,6 +380,9 @@ void rope_kv_write_(torch::Tensor qkv, long Hq_, long Hkv_, torch::Tensor cos_t,
                     torch::Tensor k_pages, torch::Tensor v_pages,
                     torch::Tensor page_table, torch::Tensor start_pos) {
   CHECK_DEV(qkv); CHECK_BF16(qkv); CHECK_CONTIG(qkv);
+  CHECK_DEV_ALL3(k_pages, page_table, start_pos);
+  CHECK_DEV_ALL2(cos_t, sin_t);
+  if (pos.has_value()) CHECK_DEV((*pos));
   const int Hq = (int)Hq_, Hkv = (int)Hkv_;
   const int P = k_pages.size(2), D = k_pages.size(3);
   const int maxp = page_table.size(1);

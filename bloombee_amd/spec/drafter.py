@@ -69,7 +69,8 @@ class MultiDrafter:
                             pos = kv.seqs[0].l_spec
                             kv.extend(1, speculative=True)
                             h = eng._embed(torch.tensor([[chain_tok]]))
-                            sp = torch.tensor([pos], dtype=torch.int32)
+                            sp = torch.tensor([pos], dtype=torch.int32,
+                                              device=eng.device)
                             h = eng.stack.forward_inference(h, kv, sp)
                             lg = eng.logits_for(h[:, -1]).float()[0]
                         p = torch.softmax(lg, -1)
