@@ -44,7 +44,7 @@ class MultiDrafter:
             kv = eng.kv_pool.allocate(1, ids.shape[1] + self.max_depth + 4)
             try:
                 # committed prefill of the draft model
-                start = torch.zeros(1, dtype=torch.int32)
+                start = torch.zeros(1, dtype=torch.int32, device=eng.device)
                 kv.extend(ids.shape[1])
                 hidden = eng._embed(ids)
                 hidden = eng.stack.forward_inference(hidden, kv, start)
