@@ -34,10 +34,17 @@ class MultiDrafter:
         self.stats = AcceptanceStats(max_depth=max_depth + 2)
 
     def build_tree(self, prompt_ids: torch.Tensor) -> TokenTree:
-        """prompt_ids: (T,) full committed token history of the sequence."""
+        """prompt_ids: (T,) full committed token history of the sequence.
+
+        The drafter assumes a token-compatible draft model (same tokenizer,
+        ref spec_decoding_drafter SSMs). Out-of-vocab ids (e.g. a larger
+        target vocab in random-init testing) are clamped so the draft
+        embedding never indexes out of bounds — such tokens simply draft
+        badly instead of crashing the device."""
         widths = plan_tree_shape(self.stats, self.node_budget,
                                  max_depth=self.max_depth)
         eng = self.draft
+        prompt_ids = prompt_ids.clamp(0, eng.config.vocab_size - 1)
         tree = TokenTree()
         ids = prompt_ids.view(1, -1)
         with torch.no_grad():
