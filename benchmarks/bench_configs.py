@@ -7,9 +7,6 @@ harness exercises the remaining configs' machinery on one MI355X:
              (weight_gpu_percent, 4-bit host compression optional)
   mixtral  — MoE decode (skinny grouped expert GEMMs)
   spec     — tree speculative decoding, local target+draft, acceptance rate
-             (KNOWN ISSUE on GPU: the multi-round drafting loop trips an HSA
-             hardware exception after ~15 rounds — single tree-verify steps
-             and the full CPU spec swarm suite pass; under investigation)
 
     python benchmarks/bench_configs.py offload mixtral spec
 """
