@@ -5,10 +5,10 @@
 // paged cache (new tokens are kv_write'n first, then attended — one code path
 // for first-prefill, multi-turn continuation and chunked prefill).
 //
-// Structure (v1, correctness-first per the guide's §5 ladder; the 8-wave
-// 256-row swizzled structure is the round-2 upgrade):
-//   grid  = (ceil(Tq/64), B*Hq); block = 256 (4 waves)
-//   each wave owns 16 q rows; KV tiles of 32 keys stage through LDS
+// Structure (v2): grid = (ceil(Tq/128), B*Hq); block = 512 (8 waves), each
+//   wave owns 16 q rows — 128 rows per workgroup halves the KV re-streaming
+//   of the 64-row v1 (every staged tile feeds twice the MFMA work).
+//   KV tiles of 32 keys stage through LDS
 //   (K row-major padded +16 B -> conflict-free 16-lane ds_read_b128;
 //    V transposed [d][key] at 80 B row stride — same property)
 //   QK^T and P.V on v_mfma_f32_16x16x32_bf16; online softmax in f32
@@ -25,7 +25,7 @@ static constexpr int KVBLK = 32;
 
 
 template <int D>
-__global__ __launch_bounds__(256) void attn_prefill_kernel(
+__global__ __launch_bounds__(512) void attn_prefill_kernel(
     const unsigned short* __restrict__ q,        // (B, Hq, Tq, D)
     const unsigned short* __restrict__ k_pages,  // (np, Hkv, P, D)
     const unsigned short* __restrict__ v_pages,
@@ -44,7 +44,7 @@ __global__ __launch_bounds__(256) void attn_prefill_kernel(
   constexpr int NDT = D / 16;   // PV d-tiles
 
   __shared__ __attribute__((aligned(16))) unsigned char lds[
-      KVBLK * KROW_B + D * VROW_B + 4 * 16 * PROW_B];
+      KVBLK * KROW_B + D * VROW_B + 8 * 16 * PROW_B];
   unsigned short* k_lds = (unsigned short*)lds;
   unsigned short* v_lds = (unsigned short*)(lds + KVBLK * KROW_B);
   unsigned short* p_lds = (unsigned short*)(lds + KVBLK * KROW_B + D * VROW_B);
@@ -53,7 +53,7 @@ __global__ __launch_bounds__(256) void attn_prefill_kernel(
   const int bh = blockIdx.y;
   const int b = bh / Hq, h = bh % Hq;
   const int kvh = h / G;
-  const int qbase = qtile * 64;
+  const int qbase = qtile * 128;
   const int tid = threadIdx.x;
   const int wave = tid / WAVE;
   const int lane = tid & (WAVE - 1);
@@ -82,7 +82,7 @@ __global__ __launch_bounds__(256) void attn_prefill_kernel(
   for (int n = 0; n < NDT; ++n) acc_o[n] = (f32x4){0.f, 0.f, 0.f, 0.f};
 
   // key range this q-tile can see: [lo_min, kmax)
-  const int qpos_max = qs + min(qbase + 63, Tq - 1);
+  const int qpos_max = qs + min(qbase + 127, Tq - 1);
   const int kmax = min(ctx, qpos_max + 1);
   int kstart = 0;
   if (window > 0) kstart = max(0, ((qs + qbase) - window + 1) / KVBLK * KVBLK);
@@ -90,7 +90,7 @@ __global__ __launch_bounds__(256) void attn_prefill_kernel(
   for (int kbase = kstart; kbase < kmax; kbase += KVBLK) {
     __syncthreads();  // everyone done reading previous tile
     // ---- stage K tile (row-major, padded) ----
-    for (int idx = tid; idx < KVBLK * (D / 8); idx += 256) {
+    for (int idx = tid; idx < KVBLK * (D / 8); idx += 512) {
       const int t = idx / (D / 8);
       const int d8 = (idx % (D / 8)) * 8;
       const int kpos = kbase + t;
@@ -106,7 +106,7 @@ __global__ __launch_bounds__(256) void attn_prefill_kernel(
     // ---- stage V tile: pool is d-major (np, Hkv, D, P), so an 8-position
     // piece at fixed d is one contiguous 16 B load AND one contiguous 16 B
     // LDS store (kbase is KVBLK-aligned => each octet sits in one page) ----
-    for (int idx = tid; idx < D * (KVBLK / 8); idx += 256) {
+    for (int idx = tid; idx < D * (KVBLK / 8); idx += 512) {
       const int d = idx / (KVBLK / 8);
       const int o8 = (idx % (KVBLK / 8)) * 8;
       const int kpos = kbase + o8;

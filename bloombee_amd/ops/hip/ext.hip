@@ -324,9 +324,9 @@ torch::Tensor attn_prefill(torch::Tensor q, torch::Tensor k_pages,
     alibi_ptr = alibi->data_ptr<float>();
   }
   auto out = torch::empty_like(q);
-  dim3 grid((Tq + 63) / 64, B * Hq);
+  dim3 grid((Tq + 127) / 128, B * Hq);
   auto launch = [&](auto d) {
-    attn_prefill_kernel<decltype(d)::value><<<grid, 256, 0, cur_stream()>>>(
+    attn_prefill_kernel<decltype(d)::value><<<grid, 512, 0, cur_stream()>>>(
         bf_ptr(q), bf_ptr(k_pages), bf_ptr(v_pages), page_table.data_ptr<int>(),
         q_start.data_ptr<int>(), alibi_ptr, bf_ptr_mut(out), B, Hq, G, Tq, P,
         maxp, (int)window, (float)scale,
@@ -357,9 +357,9 @@ torch::Tensor attn_prefill_qkv(torch::Tensor qkv, long Hq_,
   TORCH_CHECK(qkv.dim() == 3 && qkv.size(2) == (long)X * D, "bad fused qkv shape");
   const int G = Hq / Hkv;
   auto out = torch::empty({B, Tq, (long)Hq * D}, qkv.options());
-  dim3 grid((Tq + 63) / 64, B * Hq);
+  dim3 grid((Tq + 127) / 128, B * Hq);
   auto launch = [&](auto d) {
-    attn_prefill_kernel<decltype(d)::value><<<grid, 256, 0, cur_stream()>>>(
+    attn_prefill_kernel<decltype(d)::value><<<grid, 512, 0, cur_stream()>>>(
         bf_ptr(qkv), bf_ptr(k_pages), bf_ptr(v_pages), page_table.data_ptr<int>(),
         q_start.data_ptr<int>(), nullptr, bf_ptr_mut(out), B, Hq, G, Tq, P,
         maxp, (int)window, (float)scale,
