@@ -1,0 +1,75 @@
+"""CLI smoke tests: run_dht + run_server boot a servable worker
+(mirror the reference's out-of-band swarm bring-up, but self-contained)."""
+import os
+import signal
+import subprocess
+import sys
+import time
+
+import pytest
+import torch
+
+
+def test_cli_dht_and_server_serve_requests(tmp_path):
+    env = dict(os.environ)
+    env["PYTHONPATH"] = os.getcwd()
+    dht = subprocess.Popen(
+        [sys.executable, "-m", "bloombee_amd.cli.run_dht", "--port", "0"],
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True, env=env)
+    try:
+        # parse the announced endpoint from the log line
+        ep = None
+        t0 = time.monotonic()
+        while time.monotonic() - t0 < 30:
+            line = dht.stdout.readline()
+            if "DHT bootstrap node at" in line:
+                part = line.split("DHT bootstrap node at")[1].split("—")[0]
+                host, port = part.strip().rsplit(":", 1)
+                ep = (host, int(port))
+                break
+        assert ep is not None, "bootstrap endpoint not announced"
+
+        srv = subprocess.Popen(
+            [sys.executable, "-m", "bloombee_amd.cli.run_server", "llama-tiny",
+             "--initial-peers", f"{ep[0]}:{ep[1]}", "--block-indices", "0:4",
+             "--device", "cpu", "--throughput", "1.0",
+             "--attn-cache-tokens", "8192"],
+            stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True, env=env)
+        try:
+            from bloombee_amd.client import ClientConfig
+            from bloombee_amd.models.auto import AutoDistributedModelForCausalLM
+
+            cfg = ClientConfig(initial_peers=[ep])
+            model = None
+            t0 = time.monotonic()
+            last = None
+            while time.monotonic() - t0 < 60:
+                try:
+                    model = AutoDistributedModelForCausalLM.from_pretrained(
+                        "llama-tiny", client_config=cfg, seed=0)
+                    model.remote.manager.make_sequence()
+                    break
+                except Exception as e:  # noqa: BLE001
+                    last = e
+                    if model is not None:
+                        model.remote.manager.shutdown()
+                        model = None
+                    time.sleep(1.0)
+            assert model is not None, f"swarm never became routable: {last}"
+            ids = torch.randint(0, 1000, (1, 5),
+                                generator=torch.Generator().manual_seed(0))
+            out = model.generate(ids, max_new_tokens=3)
+            assert out.shape == (1, 8)
+            model.remote.manager.shutdown()
+        finally:
+            srv.send_signal(signal.SIGINT)
+            try:
+                srv.wait(timeout=10)
+            except subprocess.TimeoutExpired:
+                srv.kill()
+    finally:
+        dht.terminate()
+        try:
+            dht.wait(timeout=5)
+        except subprocess.TimeoutExpired:
+            dht.kill()

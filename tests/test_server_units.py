@@ -1,0 +1,93 @@
+"""Server-side unit tests: block selection, throughput measurement,
+telemetry, reachability (mirror reference test_server_stats/aux tiers)."""
+import asyncio
+import time
+
+import pytest
+import torch
+
+from bloombee_amd.data_structures import (RemoteModuleInfo, ServerInfo,
+                                          ServerState)
+from bloombee_amd.server.block_selection import (block_throughputs,
+                                                 choose_best_blocks,
+                                                 should_choose_other_blocks)
+from bloombee_amd.server.throughput import measure_compute_rps
+from bloombee_amd.utils.telemetry import StageTimes
+
+
+def _infos(cov):
+    """cov: list of per-block total throughput."""
+    out = []
+    for i, t in enumerate(cov):
+        servers = {}
+        if t > 0:
+            servers[f"p{i}"] = ServerInfo(throughput=t)
+        out.append(RemoteModuleInfo(uid=f"m.{i}", servers=servers))
+    return out
+
+
+def test_choose_best_blocks_picks_weakest_window():
+    infos = _infos([5, 5, 0, 0, 5, 5])
+    assert choose_best_blocks(2, infos) == [2, 3]
+    assert block_throughputs(infos) == [5, 5, 0, 0, 5, 5]
+
+
+def test_should_choose_other_blocks():
+    # our server p0 sits on well-covered blocks while block 3 starves
+    infos = _infos([9, 9, 9, 0.5])
+    infos[0].servers["me"] = ServerInfo(throughput=3.0)
+    infos[1].servers["me"] = ServerInfo(throughput=3.0)
+    assert should_choose_other_blocks("me", infos, balance_quality=0.9)
+    # balanced swarm: no move
+    infos2 = _infos([5, 5, 5, 5])
+    infos2[0].servers["me"] = ServerInfo(throughput=3.0)
+    assert not should_choose_other_blocks("me", infos2, balance_quality=0.75)
+
+
+def test_measure_compute_rps_runs():
+    from bloombee_amd.models.base import resolve_config
+
+    rps = measure_compute_rps(resolve_config("llama-tiny"), "cpu",
+                              n_tokens=1, n_steps=3)
+    assert rps > 0
+
+
+def test_stage_times_summary():
+    st = StageTimes()
+    with st.span("compute"):
+        time.sleep(0.01)
+    st.bump_step()
+    s = st.summary()
+    assert s["steps"] == 1 and s["compute"]["count"] == 1
+    assert s["compute"]["mean_ms"] >= 10
+    assert "compute" in st.table()
+
+
+def test_reachability_probe():
+    from bloombee_amd.net.rpc import RpcServer
+    from bloombee_amd.server.reachability import (attach_reachability,
+                                                  check_direct_reachability)
+
+    async def run():
+        target = RpcServer()
+
+        async def info(meta, tensors):
+            return {"ok": True}, []
+
+        target.register("rpc_info", info)
+        t_ep = await target.start()
+
+        helper = RpcServer()
+        attach_reachability(helper)
+        h_ep = await helper.start()
+
+        ok = await check_direct_reachability(t_ep, [h_ep])
+        assert ok is True
+        bad = await check_direct_reachability(("127.0.0.1", 1), [h_ep])
+        assert bad is False
+        none = await check_direct_reachability(t_ep, [("127.0.0.1", 1)])
+        assert none is None
+        await target.stop()
+        await helper.stop()
+
+    asyncio.run(run())
