@@ -293,7 +293,9 @@ def quant4_unpack(
 
 
 def alibi_slopes(n_heads: int) -> torch.Tensor:
-    """HF bloom build_alibi_tensor slopes (negative, per head)."""
+    """HF bloom build_alibi_tensor slopes (positive, per head; the absolute
+    bias slope*j is softmax-shift-equivalent to the -slope*(i-j) distance
+    penalty)."""
     import math as _m
 
     closest = 2 ** _m.floor(_m.log2(n_heads))
@@ -302,4 +304,4 @@ def alibi_slopes(n_heads: int) -> torch.Tensor:
     if closest < n_heads:
         extra_base = 2.0 ** (-(2.0 ** -(_m.log2(2 * closest) - 3)))
         slopes += [extra_base ** (2 * i + 1) for i in range(n_heads - closest)]
-    return -torch.tensor(slopes, dtype=torch.float32)
+    return torch.tensor(slopes, dtype=torch.float32)

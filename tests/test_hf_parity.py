@@ -247,3 +247,107 @@ def test_gemma4_block_matches_hf():
     kv.close()
     diff = (ours - hf_out).abs().max().item()
     assert diff < 1e-4, f"gemma4 block vs HF: max diff {diff}"
+
+
+def test_bloom_block_matches_hf():
+    from transformers.models.bloom.configuration_bloom import BloomConfig as HFCfg
+    from transformers.models.bloom.modeling_bloom import (BloomBlock,
+                                                          build_alibi_tensor)
+
+    cfg, blk, stack = _mk_block("bloom-tiny")
+    Hq, D, H = blk.Hq, blk.D, cfg.hidden_size
+    hf_cfg = HFCfg(hidden_size=H, n_head=Hq,
+                   n_layer=cfg.num_hidden_layers, vocab_size=cfg.vocab_size,
+                   layer_norm_epsilon=cfg.layer_norm_epsilon,
+                   attn_implementation="eager", hidden_dropout=0.0,
+                   attention_dropout=0.0)
+    layer = BloomBlock(hf_cfg, layer_idx=0).eval()
+    with torch.no_grad():
+        layer.input_layernorm.weight.copy_(blk.ln1_w)
+        layer.input_layernorm.bias.copy_(blk.ln1_b)
+        layer.post_attention_layernorm.weight.copy_(blk.ln2_w)
+        layer.post_attention_layernorm.bias.copy_(blk.ln2_b)
+        # HF bloom fuses qkv per-head interleaved [h: q|k|v]; ours is q|k|v
+        qkv_hf = torch.empty_like(blk.qkv_w)
+        qkvb_hf = torch.empty_like(blk.qkv_b)
+        for h in range(Hq):
+            for j, sec in enumerate(range(3)):
+                qkv_hf[(h * 3 + j) * D:(h * 3 + j + 1) * D] = \
+                    blk.qkv_w[j * Hq * D + h * D: j * Hq * D + (h + 1) * D]
+                qkvb_hf[(h * 3 + j) * D:(h * 3 + j + 1) * D] = \
+                    blk.qkv_b[j * Hq * D + h * D: j * Hq * D + (h + 1) * D]
+        layer.self_attention.query_key_value.weight.copy_(qkv_hf)
+        layer.self_attention.query_key_value.bias.copy_(qkvb_hf)
+        layer.self_attention.dense.weight.copy_(blk.dense_w)
+        layer.self_attention.dense.bias.copy_(blk.dense_b)
+        layer.mlp.dense_h_to_4h.weight.copy_(blk.up_w)
+        layer.mlp.dense_h_to_4h.bias.copy_(blk.up_b)
+        layer.mlp.dense_4h_to_h.weight.copy_(blk.down_w)
+        layer.mlp.dense_4h_to_h.bias.copy_(blk.down_b)
+
+    torch.manual_seed(0)
+    B, T = 2, 9
+    h = torch.randn(B, T, H) * 0.3
+    mask2d = torch.ones(B, T, dtype=torch.long)
+    alibi = build_alibi_tensor(mask2d, Hq, dtype=torch.float32)
+    causal = torch.full((T, T), float("-inf")).triu(1).view(1, 1, T, T).expand(B, 1, T, T)
+    with torch.no_grad():
+        hf_out = layer(h, alibi=alibi, attention_mask=causal)
+        if isinstance(hf_out, tuple):
+            hf_out = hf_out[0]
+
+    kv = stack.make_kv(1024).allocate(B, 64)
+    kv.extend(T)
+    ours = blk.forward_inference(h.clone(), kv,
+                                 torch.zeros(B, dtype=torch.int32))
+    kv.close()
+    diff = (ours - hf_out).abs().max().item()
+    assert diff < 1e-4, f"bloom block vs HF: max diff {diff}"
+
+
+def test_falcon_block_matches_hf():
+    from transformers.models.falcon.configuration_falcon import \
+        FalconConfig as HFCfg
+    from transformers.models.falcon.modeling_falcon import (
+        FalconDecoderLayer, FalconRotaryEmbedding)
+
+    cfg, blk, stack = _mk_block("falcon-tiny")
+    Hq, Hkv, D, H = blk.Hq, blk.Hkv, blk.D, cfg.hidden_size
+    hf_cfg = HFCfg(hidden_size=H, num_attention_heads=Hq,
+                   num_hidden_layers=cfg.num_hidden_layers,
+                   vocab_size=cfg.vocab_size, multi_query=True,
+                   parallel_attn=True, bias=False,
+                   new_decoder_architecture=False,
+                   layer_norm_epsilon=cfg.layer_norm_epsilon,
+                   rope_theta=cfg.rope_theta,
+                   attn_implementation="eager", hidden_dropout=0.0,
+                   attention_dropout=0.0)
+    layer = FalconDecoderLayer(hf_cfg, layer_idx=0).eval()
+    with torch.no_grad():
+        layer.input_layernorm.weight.copy_(blk.ln_w)
+        layer.input_layernorm.bias.copy_(blk.ln_b)
+        layer.self_attention.query_key_value.weight.copy_(blk.qkv_w)
+        layer.self_attention.dense.weight.copy_(blk.o_w)
+        layer.mlp.dense_h_to_4h.weight.copy_(blk.up_w)
+        layer.mlp.dense_4h_to_h.weight.copy_(blk.down_w)
+
+    torch.manual_seed(0)
+    B, T = 2, 9
+    h = torch.randn(B, T, H) * 0.3
+    rotary = FalconRotaryEmbedding(hf_cfg)
+    pos = torch.arange(T).unsqueeze(0).expand(B, T)
+    cos_sin = rotary(h, pos)
+    mask = torch.full((T, T), float("-inf")).triu(1).view(1, 1, T, T).expand(B, 1, T, T)
+    with torch.no_grad():
+        hf_out = layer(h, position_embeddings=cos_sin, attention_mask=mask,
+                       alibi=None, position_ids=pos)
+        if isinstance(hf_out, tuple):
+            hf_out = hf_out[0]
+
+    kv = stack.make_kv(1024).allocate(B, 64)
+    kv.extend(T)
+    ours = blk.forward_inference(h.clone(), kv,
+                                 torch.zeros(B, dtype=torch.int32))
+    kv.close()
+    diff = (ours - hf_out).abs().max().item()
+    assert diff < 1e-4, f"falcon block vs HF: max diff {diff}"
