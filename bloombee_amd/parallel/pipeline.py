@@ -68,8 +68,14 @@ class PipelineStage:
                     self.tp_group = g
         self.global_batch = global_batch
         if micro_batches <= 0:
-            micro_batches = (min(4 * self.pp_world, global_batch)
-                             if self.pp_world > 1 else 1)
+            # enough micro-batches to overlap the stage chain, but keep each
+            # slice >= 8 sequences: BW-bound decode kernels lose ~2x at B8
+            # vs B32 (benchmarks/bench_kernels.py attn_decode)
+            if self.pp_world > 1:
+                micro_batches = max(1, min(2 * self.pp_world,
+                                           global_batch // 8))
+            else:
+                micro_batches = 1
         while global_batch % micro_batches != 0:
             micro_batches -= 1
         self.M = micro_batches

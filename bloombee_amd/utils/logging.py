@@ -22,3 +22,26 @@ def get_logger(name: str) -> logging.Logger:
         root.propagate = False
         _configured = True
     return logging.getLogger(name if name.startswith("bloombee_amd") else f"bloombee_amd.{name}")
+
+
+# ---------------------------------------------------------------------------
+# Per-channel debug switches (parity: reference utils/debug_config.py —
+# BLOOMBEE_DEBUG master + per-group toggles). BBAMD_DEBUG=1 enables all;
+# BBAMD_DEBUG_KV / _MICROBATCH / _COMPRESSION / _INFERENCE narrow it.
+# ---------------------------------------------------------------------------
+import os as _os
+
+_MASTER = _os.environ.get("BBAMD_DEBUG", "") not in ("", "0", "false")
+_CHANNELS = {
+    ch: _os.environ.get(f"BBAMD_DEBUG_{ch.upper()}", "") not in ("", "0", "false")
+    for ch in ("kv", "microbatch", "compression", "inference")
+}
+
+
+def is_log_channel_enabled(channel: str) -> bool:
+    return _MASTER or _CHANNELS.get(channel, False)
+
+
+def debug_log(channel: str, logger, msg: str, *args) -> None:
+    if is_log_channel_enabled(channel):
+        logger.info(f"[{channel}] {msg}", *args)
