@@ -20,6 +20,12 @@ import time
 import torch
 
 
+def stage_max_pos(model: str) -> int:
+    from bloombee_amd.models.base import resolve_config
+
+    return resolve_config(model).max_position_embeddings
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -49,8 +55,11 @@ def main():
 
     global_batch = args.batch_per_gpu * world
     session_len = args.prompt + args.steps + args.warmup + 8
-    # KV budget: whole session for every sequence on every rank
-    kv_tokens = global_batch * session_len + 1024
+    assert session_len <= stage_max_pos(args.model), (
+        f"prompt+steps {session_len} exceeds the model's "
+        f"max_position_embeddings — shorten --prompt or --steps")
+    # KV budget: whole session for every sequence on every rank (page-rounded)
+    kv_tokens = global_batch * ((session_len + 15) // 16 + 1) * 16 + 1024
 
     stage = PipelineStage(args.model, device, global_batch,
                           micro_batches=args.micro_batches,

@@ -142,14 +142,18 @@ __global__ __launch_bounds__(BLOCK) void layernorm_kernel(
 __global__ void rope_kernel(
     unsigned short* __restrict__ q, unsigned short* __restrict__ k,
     const float* __restrict__ cos_t, const float* __restrict__ sin_t,
-    const int* __restrict__ pos, int B, int Hq, int Hkv, int T, int D) {
+    const int* __restrict__ pos, int B, int Hq, int Hkv, int T, int D,
+    int max_pos) {
   const int bt = blockIdx.x;
   const int b = bt / T, t = bt % T;
   const int h = blockIdx.y;
   unsigned short* base =
       (h < Hq) ? q + (((long)b * Hq + h) * T + t) * D
                : k + (((long)b * Hkv + (h - Hq)) * T + t) * D;
-  const int p = pos[b * T + t];
+  // clamp: positions past the rope table (model max_position_embeddings)
+  // must not read out of bounds — the python layer guards the supported
+  // range, this is the device-safety net
+  const int p = min(max(pos[b * T + t], 0), max_pos - 1);
   const float* c = cos_t + (long)p * (D / 2);
   const float* s = sin_t + (long)p * (D / 2);
   for (int i = threadIdx.x; i < D / 2; i += blockDim.x) {
@@ -229,7 +233,7 @@ __global__ void rope_kv_write_kernel(
     const float* __restrict__ sin_t, const int* __restrict__ pos,
     unsigned short* __restrict__ k_pages, unsigned short* __restrict__ v_pages,
     const int* __restrict__ page_table, const int* __restrict__ start_pos,
-    int B, int Hq, int Hkv, int T, int D, int P, int maxp) {
+    int B, int Hq, int Hkv, int T, int D, int P, int maxp, int max_pos) {
   const int bt = blockIdx.x;
   const int b = bt / T, t = bt % T;
   const int h = blockIdx.y;
@@ -246,7 +250,7 @@ __global__ void rope_kv_write_kernel(
       dst[(long)i * P] = base[i];
     return;
   }
-  const int p = pos ? pos[b * T + t] : abspos;
+  const int p = min(max(pos ? pos[b * T + t] : abspos, 0), max_pos - 1);
   const float* c = cos_t + (long)p * (D / 2);
   const float* s = sin_t + (long)p * (D / 2);
   unsigned short* dst = base;

@@ -113,7 +113,8 @@ void rope_apply_(torch::Tensor q, torch::Tensor k, torch::Tensor cos_t,
   dim3 grid(B * T, Hq + Hkv);
   rope_kernel<<<grid, 64, 0, cur_stream()>>>(
       bf_ptr_mut(q), bf_ptr_mut(k), cos_t.data_ptr<float>(),
-      sin_t.data_ptr<float>(), pos.data_ptr<int>(), B, Hq, Hkv, T, D);
+      sin_t.data_ptr<float>(), pos.data_ptr<int>(), B, Hq, Hkv, T, D,
+      (int)cos_t.size(0));
 }
 
 torch::Tensor swiglu(torch::Tensor gu) {
@@ -397,7 +398,8 @@ void rope_kv_write_(torch::Tensor qkv, long Hq_, long Hkv_, torch::Tensor cos_t,
   rope_kv_write_kernel<<<grid, 64, 0, cur_stream()>>>(
       bf_ptr_mut(qkv), cos_t.data_ptr<float>(), sin_t.data_ptr<float>(), pos_ptr,
       bf_ptr_mut(k_pages), bf_ptr_mut(v_pages), page_table.data_ptr<int>(),
-      start_pos.data_ptr<int>(), B, Hq, Hkv, T, D, P, maxp);
+      start_pos.data_ptr<int>(), B, Hq, Hkv, T, D, P, maxp,
+      (int)cos_t.size(0));
 }
 
 // ---------------------------------------------------------------------------
