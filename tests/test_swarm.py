@@ -215,3 +215,45 @@ def test_swarm_microbatch_split_matches_local(swarm):
     expect = _local_tokens(prompt, 5)
     assert torch.equal(out[:, 6:], expect), (out[:, 6:], expect)
     model.remote.manager.shutdown()
+
+
+@pytest.mark.parametrize("family_model", ["qwen3-tiny", "falcon-tiny",
+                                          "mixtral-tiny", "gemma4-tiny"])
+def test_family_swarm_matches_local(family_model):
+    """Every registered family must decode exactly over a 2-server swarm
+    (client model classes + server blocks + per-family KV geometry)."""
+    from bloombee_amd.models.base import resolve_config
+    n_layers = resolve_config(family_model).num_hidden_layers
+    half = n_layers // 2
+    boot = Dht()
+    s1 = Server(family_model, initial_peers=[boot.endpoint],
+                block_indices=(0, half),
+                device="cpu", seed=SEED, kv_max_tokens=1 << 14, update_period=5.0)
+    s2 = Server(family_model, initial_peers=[boot.endpoint],
+                block_indices=(half, n_layers),
+                device="cpu", seed=SEED, kv_max_tokens=1 << 14, update_period=5.0)
+    s1.run_in_background()
+    s2.run_in_background()
+    try:
+        cfg = ClientConfig(initial_peers=[boot.endpoint])
+        model = AutoDistributedModelForCausalLM.from_pretrained(
+            family_model, client_config=cfg, seed=SEED)
+        V = model.config.vocab_size
+        gen = torch.Generator().manual_seed(9)
+        prompt = torch.randint(0, min(900, V), (2, 6), generator=gen)
+        out = model.generate(prompt, max_new_tokens=4)
+
+        eng = LocalEngine(family_model, device="cpu", seed=SEED,
+                          kv_max_tokens=1 << 14)
+        kv = eng.kv_pool.allocate(2, 64)
+        toks = [eng.prefill(prompt, kv)]
+        for _ in range(3):
+            toks.append(eng.decode_step(toks[-1], kv))
+        kv.close()
+        expect = torch.stack(toks, 1)
+        assert torch.equal(out[:, 6:], expect), (family_model, out[:, 6:], expect)
+        model.remote.manager.shutdown()
+    finally:
+        s1.shutdown()
+        s2.shutdown()
+        boot.shutdown()
