@@ -27,13 +27,26 @@ class RemoteGenerationMixin:
                  top_k: Optional[int] = None, top_p: Optional[float] = None,
                  session=None) -> torch.Tensor:
         B, T = input_ids.shape
+        # p-tuning at inference (ref remote_generation + ptune): trained
+        # prompt embeds are prepended once at prefill; deep per-block prompts
+        # ride the first step's item and are re-applied by every span
+        pe = getattr(self.transformer, "prompt_embeds", None) \
+            if hasattr(self, "transformer") else None
+        dp = getattr(self.transformer, "deep_prompts", None) \
+            if hasattr(self, "transformer") else None
+        pre = pe.shape[0] if pe is not None else 0
         own_session = session is None
         if own_session:
-            session = self.remote.inference_session(T + max_new_tokens + 1)
+            session = self.remote.inference_session(
+                T + pre + max_new_tokens + 1)
         out_tokens = [input_ids]
         try:
             hidden = self.embed(input_ids)
-            hidden = session.step(hidden)
+            if pe is not None and session.position == 0:
+                p = pe.detach().to(hidden.dtype).unsqueeze(0).expand(B, -1, -1)
+                hidden = torch.cat([p, hidden], dim=1)
+            hidden = session.step(
+                hidden, prompts=dp.detach() if dp is not None else None)
             logits = self.lm_head(self.final_norm(hidden[:, -1:]))[:, -1]
             next_tok = self._pick(logits, do_sample, temperature, top_k, top_p)
             out_tokens.append(next_tok.view(B, 1))

@@ -25,28 +25,37 @@ MAX_TOKENS_IN_BATCH = 1024  # client sub-batch split (ref sequential_autograd.py
 
 
 def _call_forward(span: RemoteSpanInfo, hidden: torch.Tensor,
-                  timeout: float) -> torch.Tensor:
+                  timeout: float,
+                  prompts: Optional[torch.Tensor] = None) -> torch.Tensor:
     client = get_client(span.server_info.host, span.server_info.port)
+    payload = [hidden] if prompts is None else [hidden, prompts]
     meta, tensors = run_coroutine(
-        client.call("rpc_forward", {}, [hidden], timeout=timeout),
+        client.call("rpc_forward", {}, payload, timeout=timeout),
         timeout + 5)
     return tensors[0]
 
 
 def _call_backward(span: RemoteSpanInfo, hidden_in: torch.Tensor,
-                   grad_out: torch.Tensor, timeout: float) -> torch.Tensor:
+                   grad_out: torch.Tensor, timeout: float,
+                   prompts: Optional[torch.Tensor] = None,
+                   ) -> Tuple[torch.Tensor, Optional[torch.Tensor]]:
     client = get_client(span.server_info.host, span.server_info.port)
+    payload = ([hidden_in, grad_out] if prompts is None
+               else [hidden_in, grad_out, prompts])
     meta, tensors = run_coroutine(
-        client.call("rpc_backward", {}, [hidden_in, grad_out], timeout=timeout),
+        client.call("rpc_backward", {}, payload, timeout=timeout),
         timeout + 5)
-    return tensors[0]
+    return tensors[0], (tensors[1] if len(tensors) > 1 else None)
 
 
 def sequential_forward(manager: RemoteSequenceManager, hidden: torch.Tensor,
                        start: int = 0, end: Optional[int] = None,
+                       prompts: Optional[torch.Tensor] = None,
                        ) -> Tuple[torch.Tensor, List[Tuple[RemoteSpanInfo, torch.Tensor]]]:
     """Forward through [start, end); returns output + per-span (span, input)
-    pairs saved for backward (ref sequential_forward :25-105)."""
+    pairs saved for backward (ref sequential_forward :25-105). prompts:
+    full-depth deep p-tune tensor (num_blocks, pre, H); each span receives
+    its block slice."""
     end = end if end is not None else manager.num_blocks
     cfg = manager.config
     saved: List[Tuple[RemoteSpanInfo, torch.Tensor]] = []
@@ -57,7 +66,9 @@ def sequential_forward(manager: RemoteSequenceManager, hidden: torch.Tensor,
         route = manager.make_sequence(cur, end)
         span = route[0]
         try:
-            result = _call_forward(span, out, cfg.request_timeout)
+            p = (prompts[span.start:span.end].detach()
+                 if prompts is not None else None)
+            result = _call_forward(span, out, cfg.request_timeout, prompts=p)
             manager.on_request_success(span.peer_id)
             saved.append((span, out))
             out = result
@@ -79,17 +90,26 @@ def sequential_forward(manager: RemoteSequenceManager, hidden: torch.Tensor,
 
 def sequential_backward(manager: RemoteSequenceManager, grad_out: torch.Tensor,
                         saved: List[Tuple[RemoteSpanInfo, torch.Tensor]],
+                        prompts: Optional[torch.Tensor] = None,
+                        grad_prompts: Optional[torch.Tensor] = None,
                         ) -> torch.Tensor:
     """Reverse pass over the saved spans; on failure re-routes the span and
     re-runs forward from its saved input to rebuild activations on the
-    replacement server (ref sequential_backward :112-197)."""
+    replacement server (ref sequential_backward :112-197). When prompts is
+    given, per-span deep-prompt grads are accumulated into grad_prompts
+    (full-depth, pre-zeroed) in place."""
     cfg = manager.config
     grad = grad_out
     for span, span_input in reversed(saved):
         attempt = 0
         while True:
             try:
-                grad = _call_backward(span, span_input, grad, cfg.request_timeout)
+                p = (prompts[span.start:span.end].detach()
+                     if prompts is not None else None)
+                grad, gp = _call_backward(span, span_input, grad,
+                                          cfg.request_timeout, prompts=p)
+                if gp is not None and grad_prompts is not None:
+                    grad_prompts[span.start:span.end] += gp.to(grad_prompts.dtype)
                 manager.on_request_success(span.peer_id)
                 break
             except (RpcError, OSError, TimeoutError) as e:
@@ -110,8 +130,11 @@ def sequential_backward(manager: RemoteSequenceManager, grad_out: torch.Tensor,
                     logger.warning("backward span re-split across %d servers",
                                    len(route))
                     _, sub_saved = sequential_forward(
-                        manager, span_input, span.start, span.end)
-                    grad = sequential_backward(manager, grad, sub_saved)
+                        manager, span_input, span.start, span.end,
+                        prompts=prompts)
+                    grad = sequential_backward(manager, grad, sub_saved,
+                                               prompts=prompts,
+                                               grad_prompts=grad_prompts)
                     span = None
                     break
         if span is None:
@@ -121,27 +144,36 @@ def sequential_backward(manager: RemoteSequenceManager, grad_out: torch.Tensor,
 
 class _RemoteSequentialAutograd(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, hidden: torch.Tensor, manager: RemoteSequenceManager):
+    def forward(ctx, hidden: torch.Tensor,
+                prompts: Optional[torch.Tensor],
+                manager: RemoteSequenceManager):
         outs, all_saved = [], []
         # token-bounded sub-batches (ref MAX_TOKENS_IN_BATCH)
         B, T, H = hidden.shape
         rows = max(1, MAX_TOKENS_IN_BATCH // max(T, 1))
         for i in range(0, B, rows):
-            out, saved = sequential_forward(manager, hidden[i:i + rows].detach())
+            out, saved = sequential_forward(manager, hidden[i:i + rows].detach(),
+                                            prompts=prompts)
             outs.append(out)
             all_saved.append(saved)
         ctx.manager = manager
         ctx.saved = all_saved
         ctx.rows = rows
+        ctx.prompts = prompts
         return torch.cat(outs, dim=0)
 
     @staticmethod
     def backward(ctx, grad_out: torch.Tensor):
         grads = []
+        gp = (torch.zeros_like(ctx.prompts, dtype=torch.float32)
+              if ctx.prompts is not None else None)
         for i, saved in enumerate(ctx.saved):
             g = grad_out[i * ctx.rows:(i + 1) * ctx.rows]
-            grads.append(sequential_backward(ctx.manager, g, saved))
-        return torch.cat(grads, dim=0), None
+            grads.append(sequential_backward(ctx.manager, g, saved,
+                                             prompts=ctx.prompts,
+                                             grad_prompts=gp))
+        gp_out = gp.to(ctx.prompts.dtype) if gp is not None else None
+        return torch.cat(grads, dim=0), gp_out, None
 
 
 class RemoteSequential(torch.nn.Module):
@@ -155,10 +187,11 @@ class RemoteSequential(torch.nn.Module):
                                                         num_blocks)
         self._active_session: Optional[InferenceSession] = None
 
-    def forward(self, hidden: torch.Tensor) -> torch.Tensor:
+    def forward(self, hidden: torch.Tensor,
+                prompts: Optional[torch.Tensor] = None) -> torch.Tensor:
         if self._active_session is not None:
             return self._active_session.step(hidden)
-        return _RemoteSequentialAutograd.apply(hidden, self.manager)
+        return _RemoteSequentialAutograd.apply(hidden, prompts, self.manager)
 
     def inference_session(self, max_length: int) -> InferenceSession:
         return InferenceSession(self.manager, max_length)

@@ -67,9 +67,11 @@ class _SpanSession:
         return cls(span, stream, sid, quiet, codec)
 
     def step(self, hidden: torch.Tensor, pos: int, step: int,
-             timeout: float) -> Optional[torch.Tensor]:
+             timeout: float,
+             prompts: Optional[torch.Tensor] = None) -> Optional[torch.Tensor]:
         async def go():
-            await self.stream.send({"pos": pos, "step": step}, [hidden],
+            payload = [hidden] if prompts is None else [hidden, prompts]
+            await self.stream.send({"pos": pos, "step": step}, payload,
                                    codec=self.codec)
             item = await self.stream.recv()
             if item is None:
@@ -80,9 +82,11 @@ class _SpanSession:
         return run_coroutine(go(), timeout)
 
     def send_only(self, hidden: torch.Tensor, pos: int, step: int,
-                  timeout: float) -> None:
+                  timeout: float,
+                  prompts: Optional[torch.Tensor] = None) -> None:
         async def go():
-            await self.stream.send({"pos": pos, "step": step}, [hidden],
+            payload = [hidden] if prompts is None else [hidden, prompts]
+            await self.stream.send({"pos": pos, "step": step}, payload,
                                    codec=self.codec)
 
         run_coroutine(go(), timeout)
@@ -122,11 +126,15 @@ class InferenceSession:
         self.spans: List[_SpanSession] = []
         self.position = 0           # committed tokens across the chain
         self.step_count = 0
-        self.history: List[Tuple[int, torch.Tensor]] = []  # (pos, span0 input)
+        # (pos, span0 input, deep prompts or None)
+        self.history: List[Tuple[int, torch.Tensor, Optional[torch.Tensor]]] = []
         self._closed = False
 
     # -- chain management -------------------------------------------------
-    def _open_chain(self, batch_size: int, replay: bool) -> None:
+    def _open_chain(self, batch_size: int, replay: bool,
+                    allow_push: Optional[bool] = None) -> None:
+        if allow_push is not None:
+            self.allow_push = self.allow_push and allow_push
         route = self.manager.make_sequence(
             0, self.manager.num_blocks,
             cache_tokens_needed=batch_size * self.max_length)
@@ -152,23 +160,34 @@ class InferenceSession:
         if replay and self.history:
             logger.info("replaying %d cached steps into the new chain",
                         len(self.history))
-            for pos, hidden in self.history:
-                self._chain_step(hidden, pos, replay=True)
+            for pos, hidden, pr in self.history:
+                self._chain_step(hidden, pos, replay=True, prompts=pr)
 
     def _chain_step(self, hidden: torch.Tensor, pos: int,
-                    replay: bool = False) -> torch.Tensor:
+                    replay: bool = False,
+                    prompts: Optional[torch.Tensor] = None) -> torch.Tensor:
         t = self.config.step_timeout
+
+        def _slice(span):
+            # deep p-tune: full-depth (num_blocks, pre, H) -> this span's rows
+            return (prompts[span.start:span.end]
+                    if prompts is not None else None)
+
         if self._push_mode:
             first, last = self.spans[0], self.spans[-1]
             if len(self.spans) == 1:
-                return first.step(hidden, pos, self.step_count, t)
-            first.send_only(hidden, pos, self.step_count, t)
+                return first.step(hidden, pos, self.step_count, t,
+                                  prompts=_slice(first.span))
+            # push mode with deep prompts would need per-hop prompt routing;
+            # sessions with prompts are opened pushless by step() below
+            first.send_only(hidden, pos, self.step_count, t,
+                            prompts=_slice(first.span))
             # quiet middle spans send nothing; output arrives from the last
             meta, tensors = last.recv_only(t)
             return tensors[0]
         out = hidden
         for s in self.spans:
-            out = s.step(out, pos, self.step_count, t)
+            out = s.step(out, pos, self.step_count, t, prompts=_slice(s.span))
         return out
 
     def _ban_dead_spans(self) -> None:
@@ -184,19 +203,23 @@ class InferenceSession:
                 self.manager.on_request_failure(s.span.peer_id)
 
     # -- public API -------------------------------------------------------
-    def step(self, hidden: torch.Tensor) -> torch.Tensor:
+    def step(self, hidden: torch.Tensor,
+             prompts: Optional[torch.Tensor] = None) -> torch.Tensor:
         """Run the full remote chain on `hidden` (B, T, H) at the current
-        position; returns the last span's output hidden states."""
+        position; returns the last span's output hidden states. prompts:
+        optional full-depth deep p-tune tensor (num_blocks, pre, H); each
+        span receives its block-range slice with the step item."""
         if self._closed:
             raise RuntimeError("session is closed")
         if self.batch_size is None:
             self.batch_size = hidden.shape[0]
-            self._open_chain(self.batch_size, replay=False)
+            self._open_chain(self.batch_size, replay=False,
+                             allow_push=prompts is None)
         pos = self.position
         attempt = 0
         while True:
             try:
-                out = self._chain_step(hidden, pos)
+                out = self._chain_step(hidden, pos, prompts=prompts)
                 for s in self.spans:
                     self.manager.on_request_success(s.span.peer_id)
                 break
@@ -214,7 +237,7 @@ class InferenceSession:
                 time.sleep(delay)
                 self.manager.update()
                 self._open_chain(self.batch_size, replay=True)
-        self.history.append((pos, hidden))
+        self.history.append((pos, hidden, prompts))
         self.position = pos + hidden.shape[1]
         self.step_count += 1
         return out

@@ -66,8 +66,20 @@ class BlockStack(torch.nn.Module):
     def forward_inference(self, hidden: torch.Tensor, kv: SessionHandle,
                           start_pos: torch.Tensor,
                           position_ids: Optional[torch.Tensor] = None,
-                          tree_mask: Optional[torch.Tensor] = None) -> torch.Tensor:
-        for blk in self.blocks:
+                          tree_mask: Optional[torch.Tensor] = None,
+                          deep_prompts: Optional[torch.Tensor] = None,
+                          ) -> torch.Tensor:
+        # deep p-tuning (ref client/ptune.py deep mode): deep_prompts is
+        # (n_local_blocks, pre, H); block i's slice is added to the prompt
+        # positions (< pre) of its input. Decode steps past the prompt
+        # region skip the addition entirely.
+        sp0 = int(start_pos[0]) if deep_prompts is not None else 0
+        pre = deep_prompts.shape[1] if deep_prompts is not None else 0
+        n_over = min(pre - sp0, hidden.shape[1]) if sp0 < pre else 0
+        for i, blk in enumerate(self.blocks):
+            if n_over > 0:
+                dp = deep_prompts[i, sp0:sp0 + n_over].to(hidden.dtype)
+                hidden[:, :n_over] += dp
             if tree_mask is not None:
                 hidden = blk.forward_inference(hidden, kv, start_pos,
                                                position_ids, tree_mask=tree_mask)
@@ -75,8 +87,14 @@ class BlockStack(torch.nn.Module):
                 hidden = blk.forward_inference(hidden, kv, start_pos, position_ids)
         return hidden
 
-    def forward_train(self, hidden: torch.Tensor, start_pos: int = 0) -> torch.Tensor:
-        for blk in self.blocks:
+    def forward_train(self, hidden: torch.Tensor, start_pos: int = 0,
+                      deep_prompts: Optional[torch.Tensor] = None) -> torch.Tensor:
+        pre = deep_prompts.shape[1] if deep_prompts is not None else 0
+        for i, blk in enumerate(self.blocks):
+            if pre > 0 and hidden.shape[1] >= pre:
+                # functional (not in-place) so prompt grads flow in backward
+                dp = deep_prompts[i].to(hidden.dtype).unsqueeze(0)
+                hidden = torch.cat([hidden[:, :pre] + dp, hidden[:, pre:]], 1)
             hidden = blk.forward_train(hidden, start_pos)
         return hidden
 

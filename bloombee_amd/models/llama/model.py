@@ -25,7 +25,8 @@ class DistributedLlamaModel(torch.nn.Module):
     def __init__(self, config: ModelConfig, client_config: ClientConfig,
                  model_name: str, seed: int = 0, device: str = "cpu",
                  manager: Optional[RemoteSequenceManager] = None,
-                 pre_seq_len: int = 0):
+                 pre_seq_len: int = 0,
+                 deep_ptune: bool = False):
         super().__init__()
         self.config = config
         self.device_ = torch.device(device)
@@ -44,7 +45,10 @@ class DistributedLlamaModel(torch.nn.Module):
                                        config.num_hidden_layers,
                                        manager=manager)
         # p-tuning: trainable prompt embeddings kept fp32 on the client
-        # (ref client/ptune.py:23-41)
+        # (ref client/ptune.py:23-41). deep_ptune additionally trains one
+        # additive prompt per transformer block (ref ptune.py deep mode:
+        # intermediate_prompts added to the prompt positions at every
+        # block input; grads ride back on rpc_backward).
         self.pre_seq_len = pre_seq_len
         if pre_seq_len > 0:
             self.prompt_embeds = torch.nn.Parameter(
@@ -52,6 +56,12 @@ class DistributedLlamaModel(torch.nn.Module):
                 .mul_(0.02))
         else:
             self.prompt_embeds = None
+        if pre_seq_len > 0 and deep_ptune:
+            self.deep_prompts = torch.nn.Parameter(
+                torch.zeros(config.num_hidden_layers, pre_seq_len,
+                            config.hidden_size, device=device))
+        else:
+            self.deep_prompts = None
 
     def embed(self, input_ids: torch.Tensor) -> torch.Tensor:
         h = F.embedding(input_ids.to(self.device_), self.embed_tokens)
@@ -67,7 +77,7 @@ class DistributedLlamaModel(torch.nn.Module):
             B = h.shape[0]
             p = self.prompt_embeds.to(h.dtype).unsqueeze(0).expand(B, -1, -1)
             h = torch.cat([p, h], dim=1)
-        h = self.remote(h)
+        h = self.remote(h, prompts=self.deep_prompts)
         return self.final_norm(h)
 
 
@@ -92,13 +102,15 @@ class DistributedLlamaForCausalLM(RemoteGenerationMixin, torch.nn.Module):
     def __init__(self, config: ModelConfig, client_config: ClientConfig,
                  model_name: str, seed: int = 0, device: str = "cpu",
                  manager: Optional[RemoteSequenceManager] = None,
-                 pre_seq_len: int = 0):
+                 pre_seq_len: int = 0,
+                 deep_ptune: bool = False):
         super().__init__()
         self.config = config
         self.transformer = DistributedLlamaModel(config, client_config,
                                                  model_name, seed=seed,
                                                  device=device, manager=manager,
-                                                 pre_seq_len=pre_seq_len)
+                                                 pre_seq_len=pre_seq_len,
+                                                 deep_ptune=deep_ptune)
         self.lm_head = LMHead(config, self.transformer.embed_tokens,
                               gen=self.transformer._gen, device=device)
 

@@ -167,8 +167,12 @@ class OffloadedBlockStack:
     @torch.no_grad()
     def forward_inference(self, hidden: torch.Tensor, kv: SessionHandle,
                           start_pos: torch.Tensor,
-                          position_ids=None, tree_mask=None) -> torch.Tensor:
+                          position_ids=None, tree_mask=None,
+                          deep_prompts=None) -> torch.Tensor:
         n = len(self.stack.blocks)
+        sp0 = int(start_pos[0]) if deep_prompts is not None else 0
+        pre = deep_prompts.shape[1] if deep_prompts is not None else 0
+        n_over = min(pre - sp0, hidden.shape[1]) if sp0 < pre else 0
         # prefetch the first offloaded block before the resident prefix runs
         if self.resident < n and self.policy.overlap:
             if self._arena_owner[self.resident % 2] != self.resident:
@@ -180,6 +184,8 @@ class OffloadedBlockStack:
                     nxt = i + 1
                     if self._arena_owner[nxt % 2] != nxt:
                         self._fill(nxt, nxt % 2)
+            if n_over > 0:
+                hidden[:, :n_over] += deep_prompts[i, sp0:sp0 + n_over].to(hidden.dtype)
             if tree_mask is not None:
                 hidden = blk.forward_inference(hidden, kv, start_pos,
                                                position_ids, tree_mask=tree_mask)
@@ -192,10 +198,15 @@ class OffloadedBlockStack:
                 self._done_events[i % 2] = ev
         return hidden
 
-    def forward_train(self, hidden: torch.Tensor, start_pos: int = 0):
+    def forward_train(self, hidden: torch.Tensor, start_pos: int = 0,
+                      deep_prompts=None):
+        pre = deep_prompts.shape[1] if deep_prompts is not None else 0
         for i, blk in enumerate(self.stack.blocks):
             if i >= self.resident:
                 self._ensure(i)
+            if pre > 0 and hidden.shape[1] >= pre:
+                dp = deep_prompts[i].to(hidden.dtype).unsqueeze(0)
+                hidden = torch.cat([hidden[:, :pre] + dp, hidden[:, pre:]], 1)
             hidden = blk.forward_train(hidden, start_pos)
         return hidden
 

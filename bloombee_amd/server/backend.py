@@ -135,10 +135,16 @@ class StackBackend:
             h = hidden.to(self.device, non_blocking=True)
             if h.dtype != self.config.dtype:
                 h = h.to(self.config.dtype)
+            deep = None
             if prompts is not None:
-                # deep p-tuning: prepend-trained prompts arrive as additive
-                # hidden-state deltas for this span's first block
-                h = h + prompts.to(h.device, h.dtype)
+                p = prompts.to(h.device)
+                if p.dim() == 3 and p.shape[0] == len(self.stack.blocks):
+                    # deep p-tuning: (n_local_blocks, pre, H) added to the
+                    # prompt positions at every block input
+                    deep = p
+                else:
+                    # legacy shallow form: additive delta for the span input
+                    h = h + p.to(h.dtype)
             B, T, _ = h.shape
             cur = handle.seqs[0].l_acc
             if batch_offset is not None:
@@ -155,7 +161,8 @@ class StackBackend:
                 view = SessionView(handle, batch_offset, batch_offset + B)
                 sp = torch.full((B,), start_pos, dtype=torch.int32,
                                 device=self.device)
-                return self.stack.forward_inference(h, view, sp)
+                return self.stack.forward_inference(h, view, sp,
+                                                    deep_prompts=deep)
             if start_pos < cur:
                 handle.truncate([min(start_pos, s.l_acc) for s in handle.seqs])
                 handle.rollback()
@@ -170,7 +177,8 @@ class StackBackend:
             pos = (position_ids.to(self.device).int()
                    if position_ids is not None else None)
             tm = tree_mask.to(self.device) if tree_mask is not None else None
-            out = self.stack.forward_inference(h, handle, sp, pos, tree_mask=tm)
+            out = self.stack.forward_inference(h, handle, sp, pos, tree_mask=tm,
+                                               deep_prompts=deep)
             from bloombee_amd.utils import activation_dumper
             if activation_dumper.enabled():
                 activation_dumper.capture_activation(
@@ -198,30 +206,39 @@ class StackBackend:
         self.pool.submit(lambda: handle.reorder_and_commit(keep),
                          PRIORITY_INFERENCE).result()
 
-    def forward(self, hidden: torch.Tensor) -> torch.Tensor:
-        """Training-path forward (no KV cache, full sequence)."""
+    def forward(self, hidden: torch.Tensor,
+                prompts: Optional[torch.Tensor] = None) -> torch.Tensor:
+        """Training-path forward (no KV cache, full sequence). prompts:
+        optional deep p-tune tensor (n_local_blocks, pre, H) added to the
+        prompt positions at every block input (ref client/ptune.py deep)."""
 
         def run():
             h = hidden.to(self.device).to(self.config.dtype)
+            dp = prompts.to(self.device) if prompts is not None else None
             with torch.no_grad():
-                return self.stack.forward_train(h)
+                return self.stack.forward_train(h, deep_prompts=dp)
 
         return self.pool.submit(run, PRIORITY_TRAIN).result()
 
     def backward(self, hidden_in: torch.Tensor, grad_out: torch.Tensor,
-                 ) -> torch.Tensor:
-        """Re-forward + backward; returns grad wrt the span input. Server
-        weights are frozen — only input/prompt grads flow (ref
-        backend.py:106-109, 427-462)."""
+                 prompts: Optional[torch.Tensor] = None,
+                 ) -> Tuple[torch.Tensor, Optional[torch.Tensor]]:
+        """Re-forward + backward; returns (grad wrt span input, grad wrt deep
+        prompts or None). Server weights are frozen — only input/prompt grads
+        flow (ref backend.py:106-109, 427-462)."""
 
         def run():
             h = hidden_in.to(self.device).to(self.config.dtype)
             h = h.detach().requires_grad_(True)
+            dp = None
+            if prompts is not None:
+                dp = prompts.to(self.device).detach().requires_grad_(True)
             with torch.enable_grad():
-                out = self.stack.forward_train(h)
+                out = self.stack.forward_train(h, deep_prompts=dp)
                 g = grad_out.to(out.device).to(out.dtype)
-                (grad_in,) = torch.autograd.grad(out, h, g)
-            return grad_in
+                leaves = (h,) if dp is None else (h, dp)
+                grads = torch.autograd.grad(out, leaves, g)
+            return (grads[0], grads[1] if dp is not None else None)
 
         return self.pool.submit(run, PRIORITY_TRAIN).result()
 

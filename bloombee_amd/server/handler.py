@@ -82,16 +82,24 @@ class ConnectionHandler:
         return self.backend.info(), []
 
     async def rpc_forward(self, meta, tensors):
-        (hidden,) = tensors
+        # tensors: [hidden] or [hidden, deep_prompts(n_local_blocks, pre, H)]
+        hidden = tensors[0]
+        prompts = tensors[1] if len(tensors) > 1 else None
         out = await asyncio.get_event_loop().run_in_executor(
-            None, self.backend.forward, hidden)
+            None, self.backend.forward, hidden, prompts)
         return {}, [out.cpu()]
 
     async def rpc_backward(self, meta, tensors):
-        hidden_in, grad_out = tensors
-        grad_in = await asyncio.get_event_loop().run_in_executor(
-            None, self.backend.backward, hidden_in, grad_out)
-        return {}, [grad_in.cpu()]
+        # tensors: [hidden_in, grad_out] or [..., deep_prompts]; replies
+        # [grad_in] or [grad_in, grad_prompts]
+        hidden_in, grad_out = tensors[0], tensors[1]
+        prompts = tensors[2] if len(tensors) > 2 else None
+        grad_in, grad_p = await asyncio.get_event_loop().run_in_executor(
+            None, self.backend.backward, hidden_in, grad_out, prompts)
+        outs = [grad_in.cpu()]
+        if grad_p is not None:
+            outs.append(grad_p.cpu())
+        return {}, outs
 
     # ------------------------------------------------------------------
     async def rpc_push(self, meta, tensors):
