@@ -20,6 +20,14 @@ def get_logger(name: str) -> logging.Logger:
         root.addHandler(handler)
         root.setLevel(level)
         root.propagate = False
+        # per-module overrides: BBAMD_LOG="client.session=debug,server=warning"
+        # (parity: reference debug_config per-group toggles, generalized to
+        # any logger subtree)
+        for spec in filter(None, os.environ.get("BBAMD_LOG", "").split(",")):
+            mod, _, lvl = spec.partition("=")
+            if mod and lvl:
+                logging.getLogger(f"bloombee_amd.{mod.strip()}") \
+                    .setLevel(lvl.strip().upper())
         _configured = True
     return logging.getLogger(name if name.startswith("bloombee_amd") else f"bloombee_amd.{name}")
 

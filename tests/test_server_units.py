@@ -91,3 +91,17 @@ def test_reachability_probe():
         await helper.stop()
 
     asyncio.run(run())
+
+
+def test_per_module_log_levels(monkeypatch):
+    """BBAMD_LOG per-module level overrides (ref utils/debug_config.py)."""
+    import importlib
+    import logging as stdlog
+
+    from bloombee_amd.utils import logging as bl
+
+    monkeypatch.setenv("BBAMD_LOG", "client.session=debug, server=warning")
+    monkeypatch.setattr(bl, "_configured", False)
+    bl.get_logger("client.session")
+    assert stdlog.getLogger("bloombee_amd.client.session").level == stdlog.DEBUG
+    assert stdlog.getLogger("bloombee_amd.server").level == stdlog.WARNING
