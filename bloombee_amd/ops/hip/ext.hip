@@ -240,11 +240,12 @@ static torch::Tensor attn_decode_core(
   }
   int n_split = (int)n_split_req;
   if (n_split <= 0) {
-    // measured optimum (benchmarks/bench_kernels.py attn_decode): the kernel
-    // runs 2 workgroups/CU (launch_bounds 256,2), so ~512 total workgroups =
-    // exactly full residency; more splits only add merge traffic
+    // measured optimum (benchmarks/bench_kernels.py attn_decode): the S^T
+    // kernel runs at 118 VGPRs -> 4 workgroups/CU resident, so ~1024 total
+    // workgroups = full residency (B32/ctx2048: ns=4 57us vs ns=2 63us);
+    // more splits only add merge traffic
     n_split = (int)std::max<long>(
-        1, std::min<long>(32, 512 / std::max(1, B * Hkv * nch)));
+        1, std::min<long>(32, 1024 / std::max(1, B * Hkv * nch)));
   }
   const int maxg = (G <= 4 && nch == 1) ? 4 : 16;
   auto fopt = torch::TensorOptions().device(q.device()).dtype(at::kFloat);
