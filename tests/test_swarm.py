@@ -322,3 +322,23 @@ def test_deep_ptune_generation_uses_prompts(swarm):
     # trained prompts shift the distribution; tiny chance of coincidence
     assert not torch.equal(out1[:, 5:], plain)
     model.remote.manager.shutdown()
+
+
+def test_deep_ptune_microbatch_split_matches_small_batch(swarm):
+    """Deep prompts must survive the server-side micro-batch split (B>=8
+    prefill splits into slices; every slice re-applies the per-block
+    prompts). Row 0 of a B=8 batch must match the B=1 result."""
+    boot, _ = swarm
+    cfg = ClientConfig(initial_peers=[boot.endpoint])
+    model = AutoDistributedModelForCausalLM.from_pretrained(
+        "llama-tiny", client_config=cfg, seed=SEED,
+        pre_seq_len=4, deep_ptune=True)
+    with torch.no_grad():
+        model.transformer.deep_prompts.normal_(0, 0.3)
+        model.transformer.prompt_embeds.normal_(0, 0.3)
+    gen = torch.Generator().manual_seed(13)
+    prompt = torch.randint(0, 1000, (8, 5), generator=gen)
+    out_b8 = model.generate(prompt, max_new_tokens=4)
+    out_b1 = model.generate(prompt[:1], max_new_tokens=4)
+    assert torch.equal(out_b8[:1], out_b1), (out_b8[:1], out_b1)
+    model.remote.manager.shutdown()
