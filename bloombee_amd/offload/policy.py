@@ -18,6 +18,10 @@ class OffloadPolicy:
     pin_weight: bool = True
     compress_weight: bool = False        # 4-bit group quant on the host tier
     compress_cache: bool = False
+    # top-k sparse decode attention: fraction of cache positions whose V
+    # rows join the weighted sum (ref Policy.attn_sparsity + the
+    # _sparse_attention_value path); 1.0 = exact dense
+    attn_sparsity: float = 1.0
     overlap: bool = True                 # prefetch block i+1 during block i
     prefetch_depth: int = 1
 

@@ -25,5 +25,6 @@ from bloombee_amd.ops.interface import (  # noqa: F401
     rope_apply_,
     rope_kv_write_,
     rope_cos_sin,
+    set_attn_sparsity,
     swiglu,
 )

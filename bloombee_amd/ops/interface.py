@@ -116,6 +116,18 @@ def kv_gather(k_pages, v_pages, page_table, ctx_len: int, batch_index: int):
     return ref.kv_gather(k_pages, v_pages, page_table, ctx_len, batch_index)
 
 
+_attn_sparsity = 1.0  # Policy.attn_sparsity (ref flexgen policy :10-55)
+
+
+def set_attn_sparsity(frac: float) -> None:
+    """Enable top-k sparse decode attention globally (parity: reference
+    Policy.attn_sparsity — a server-wide policy knob, not per-call). frac
+    is the fraction of cache positions whose V rows participate; 1.0
+    restores exact dense attention."""
+    global _attn_sparsity
+    _attn_sparsity = float(frac)
+
+
 def attn_decode(
     q, k_pages, v_pages, page_table, ctx_lens, scale: Optional[float] = None,
     window: int = 0, n_split: int = 0, alibi_slopes=None,
@@ -123,6 +135,10 @@ def attn_decode(
     """Single-token paged attention. q: (B, Hq, 1, D)."""
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
+    if _attn_sparsity < 1.0 and window <= 0:
+        return ref.attn_paged_topk(q, k_pages, v_pages, page_table, ctx_lens,
+                                   _attn_sparsity, scale,
+                                   alibi_slopes=alibi_slopes)
     if _on_gpu(q):
         _require_ext()
         if q.shape[-1] > 256:

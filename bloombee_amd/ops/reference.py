@@ -305,3 +305,45 @@ def alibi_slopes(n_heads: int) -> torch.Tensor:
         extra_base = 2.0 ** (-(2.0 ** -(_m.log2(2 * closest) - 3)))
         slopes += [extra_base ** (2 * i + 1) for i in range(n_heads - closest)]
     return torch.tensor(slopes, dtype=torch.float32)
+
+
+def attn_paged_topk(
+    q: torch.Tensor,
+    k_pages: torch.Tensor, v_pages: torch.Tensor,
+    page_table: torch.Tensor,
+    ctx_lens: torch.Tensor,
+    sparsity: float,
+    scale: Optional[float] = None,
+    alibi_slopes: Optional[torch.Tensor] = None,
+) -> torch.Tensor:
+    """Top-k sparse decode attention (parity: reference
+    pytorch_backend.py:935-968 `_sparse_attention_value` driven by
+    Policy.attn_sparsity): scores are computed against the full K cache,
+    then only the top ceil(sparsity*ctx) positions' V rows participate in
+    the weighted sum — the V fetch (the expensive half when the cache is
+    host-resident) touches a fraction of the cache. sparsity >= 1 is exact
+    dense attention. q: (B, Hq, 1, D)."""
+    B, Hq, Tq, D = q.shape
+    assert Tq == 1, "top-k sparse path is decode-only (one query token)"
+    Hkv = k_pages.shape[1]
+    G = Hq // Hkv
+    if scale is None:
+        scale = 1.0 / math.sqrt(D)
+    out = torch.empty_like(q)
+    for b in range(B):
+        ctx = int(ctx_lens[b])
+        k, v = kv_gather(k_pages, v_pages, page_table, ctx, b)  # (Hkv, ctx, D)
+        kg = k.float().repeat_interleave(G, dim=0)              # (Hq, ctx, D)
+        vg = v.float().repeat_interleave(G, dim=0)
+        qf = q[b, :, 0].float()                                 # (Hq, D)
+        scores = torch.einsum("hd,hcd->hc", qf, kg) * scale     # (Hq, ctx)
+        if alibi_slopes is not None:
+            scores = scores + alibi_slopes.float().to(q.device).view(-1, 1) * \
+                torch.arange(ctx, dtype=torch.float32, device=q.device)
+        kkeep = max(1, min(ctx, math.ceil(sparsity * ctx)))
+        top = scores.topk(kkeep, dim=-1)                        # (Hq, kkeep)
+        w = torch.softmax(top.values, dim=-1)                   # renormalized
+        vsel = torch.gather(
+            vg, 1, top.indices.unsqueeze(-1).expand(-1, -1, D))  # (Hq, kk, D)
+        out[b, :, 0] = torch.einsum("hk,hkd->hd", w, vsel).to(q.dtype)
+    return out

@@ -50,6 +50,9 @@ class StackBackend:
         if offload_policy is not None and offload_policy.offloads_weights:
             from bloombee_amd.offload.weights import OffloadedBlockStack
             self.stack = OffloadedBlockStack(self.stack, offload_policy)
+        if offload_policy is not None and offload_policy.attn_sparsity < 1.0:
+            from bloombee_amd import ops as _ops
+            _ops.set_attn_sparsity(offload_policy.attn_sparsity)
         self.kv_pool: PagedKVCache = self.stack.make_kv(kv_max_tokens)
         self.is_last_block = (end == config.num_hidden_layers)
         self.pruner = None
@@ -130,6 +133,11 @@ class StackBackend:
         handle = state.handle
 
         def run():
+            from bloombee_amd.utils.trace import trace_range
+            with trace_range(f"infer[{self.start}:{self.end}] pos={start_pos}"):
+                return _compute()
+
+        def _compute():
             if handle.is_swapped:
                 handle.swap_in()
             h = hidden.to(self.device, non_blocking=True)

@@ -162,3 +162,25 @@ def create_pruner(method: PruningMethod, head: MidLMHead, **kw):
     if method == PruningMethod.ADAPTIVE:
         return AdaptiveNeuralPruner(head, **kw)
     raise ValueError(method)
+
+
+def save_pruner(pruner, path: str) -> None:
+    """Checkpoint a pruner's trained state (parity: reference
+    speculative_pruner/lm_head_trainer.py:87 LM-head trainer checkpoints).
+    Saves the mid-network head and, for the adaptive pruner, the scorer MLP
+    + optimizer so online training resumes where it left off."""
+    state = {"head": pruner.head.state_dict()}
+    if isinstance(pruner, AdaptiveNeuralPruner):
+        state["scorer"] = pruner.net.state_dict()
+        state["opt"] = pruner.opt.state_dict()
+    torch.save(state, path)
+
+
+def load_pruner(pruner, path: str):
+    """Restore state saved by save_pruner into a same-shape pruner."""
+    state = torch.load(path, map_location="cpu", weights_only=False)
+    pruner.head.load_state_dict(state["head"])
+    if isinstance(pruner, AdaptiveNeuralPruner) and "scorer" in state:
+        pruner.net.load_state_dict(state["scorer"])
+        pruner.opt.load_state_dict(state["opt"])
+    return pruner

@@ -61,3 +61,50 @@ def test_offloaded_compressed_close():
     # 4-bit weights: lossy but bounded
     rel = (got - want).norm() / want.norm()
     assert rel < 0.2, rel
+
+
+def test_attn_sparsity_topk():
+    """Top-k sparse decode attention (ref _sparse_attention_value +
+    Policy.attn_sparsity): full fraction is exactly dense; a concentrated
+    distribution survives aggressive sparsity nearly unchanged."""
+    import math
+
+    import torch
+
+    from bloombee_amd import ops
+    from bloombee_amd.ops import reference as ref
+
+    torch.manual_seed(0)
+    B, Hq, Hkv, D, ctx, P = 2, 4, 2, 32, 64, 16
+    npages = B * (ctx // P + 1)
+    kp = torch.randn(npages, Hkv, P, D) * 0.1
+    vp = torch.randn(npages, Hkv, D, P) * 0.1
+    pt = torch.arange(npages, dtype=torch.int32).view(B, -1)
+    ctx_l = torch.full((B,), ctx, dtype=torch.int32)
+    q = torch.randn(B, Hq, 1, D) * 0.1
+
+    dense = ref.attn_paged(q.float(), kp.float(), vp.float(), pt,
+                           ctx_l.long() - 1)
+    full = ref.attn_paged_topk(q.float(), kp.float(), vp.float(), pt,
+                               ctx_l, sparsity=1.0)
+    assert torch.allclose(dense, full, atol=1e-5)
+
+    # concentrate attention mass on position 7 (MHA shapes so every query
+    # head has an aligned K row): top-k must match dense closely
+    q1 = torch.randn(B, Hkv, 1, D) * 0.1
+    kp2 = kp.clone().float()
+    for b in range(B):
+        for h in range(Hkv):
+            kp2[pt[b, 0], h, 7] = q1[b, h, 0].float() * 400
+    dense2 = ref.attn_paged(q1.float(), kp2, vp.float(), pt, ctx_l.long() - 1)
+    sparse2 = ref.attn_paged_topk(q1.float(), kp2, vp.float(), pt, ctx_l,
+                                  sparsity=0.25)
+    assert torch.allclose(dense2, sparse2, atol=1e-3)
+
+    # global policy switch routes ops.attn_decode through the top-k path
+    ops.set_attn_sparsity(0.25)
+    try:
+        routed = ops.attn_decode(q1.float(), kp2, vp.float(), pt, ctx_l)
+        assert torch.allclose(routed, sparse2, atol=1e-5)
+    finally:
+        ops.set_attn_sparsity(1.0)

@@ -218,3 +218,30 @@ def test_speculative_with_pruner_still_greedy(monkeypatch):
         s1.shutdown()
         s2.shutdown()
         boot.shutdown()
+
+
+def test_pruner_checkpoint_roundtrip(tmp_path):
+    """save_pruner/load_pruner restore head + scorer + optimizer state
+    (ref speculative_pruner/lm_head_trainer.py checkpoints)."""
+    import torch
+
+    from bloombee_amd.spec.pruner import (AdaptiveNeuralPruner, MidLMHead,
+                                          load_pruner, save_pruner)
+
+    head = MidLMHead(32, 100, seed=1, dtype=torch.float32)
+    p = AdaptiveNeuralPruner(head)
+    hidden = torch.randn(5, 32)
+    p.keep_indices(hidden, [1, 2, 3, 4, 5], [-1, 0, 0, 1, 1])
+    p.train_step([0, 1])
+    path = str(tmp_path / "pruner.pt")
+    save_pruner(p, path)
+
+    head2 = MidLMHead(32, 100, seed=7, dtype=torch.float32)  # different init
+    p2 = AdaptiveNeuralPruner(head2)
+    load_pruner(p2, path)
+    for a, b in zip(p.net.parameters(), p2.net.parameters()):
+        assert torch.equal(a, b)
+    assert torch.equal(p.head.weight, p2.head.weight)
+    k1 = p.keep_indices(hidden, [1, 2, 3, 4, 5], [-1, 0, 0, 1, 1])
+    k2 = p2.keep_indices(hidden, [1, 2, 3, 4, 5], [-1, 0, 0, 1, 1])
+    assert k1 == k2
