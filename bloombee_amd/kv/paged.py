@@ -23,6 +23,7 @@ Key choices (MI355X-native):
 from __future__ import annotations
 
 import contextlib
+import os
 import threading
 import time
 from dataclasses import dataclass, field
@@ -310,12 +311,18 @@ class SessionHandle:
         self.commit([len(k) for k in keep])
 
     # -- host offload (session multiplexing) ------------------------------
-    def swap_out(self) -> None:
+    def swap_out(self, to_disk: bool = False,
+                 disk_dir: Optional[str] = None) -> None:
         """Offload this session's KV to pinned host buffers and release its
         device pages (reference micro-batch KV offload: GPU working slots +
         CPU snapshots on dedicated streams, memory_cache_manager.py:944-1371).
         Another session can use the freed pages; swap_in restores the data
         (possibly into different physical pages — the page table rewrites).
+
+        to_disk: spill the snapshot to a file instead of keeping it in host
+        RAM (the reference's third KV tier, TorchMixedDevice GPU/CPU/disk
+        partition, pytorch_backend.py:1207-1237; here whole-session
+        granularity under the same page-table-rewrite contract).
         """
         if getattr(self, "_swapped", None) is not None:
             return
@@ -344,7 +351,15 @@ class SessionHandle:
                 snap.append(per_layer)
         if on_gpu:
             stream.synchronize()
-        self._swapped = snap
+        if to_disk:
+            import tempfile
+            fd, path = tempfile.mkstemp(
+                suffix=".kvswap.pt", dir=disk_dir)
+            os.close(fd)
+            torch.save(snap, path)
+            self._swapped = ("disk", path)
+        else:
+            self._swapped = snap
         freed = []
         for s in self.seqs:
             freed.extend(s.pages)
@@ -359,6 +374,10 @@ class SessionHandle:
         snap = getattr(self, "_swapped", None)
         if snap is None:
             return
+        if isinstance(snap, tuple) and snap[0] == "disk":
+            path = snap[1]
+            snap = torch.load(path, map_location="cpu", weights_only=False)
+            os.unlink(path)
         cache = self.cache
         on_gpu = cache.device.type == "cuda"
         stream = torch.cuda.Stream(cache.device) if on_gpu else None

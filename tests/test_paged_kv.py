@@ -164,3 +164,39 @@ def test_session_swap_out_in_roundtrip():
     h1.extend(2)
     h2.close()
     h1.close()
+
+
+def test_swap_out_to_disk_roundtrip(tmp_path):
+    """Disk KV tier (ref TorchMixedDevice GPU/CPU/disk partition): a session
+    swapped to a file restores byte-identical KV into fresh pages."""
+    import torch
+
+    from bloombee_amd.kv.paged import PagedKVCache
+
+    cache = PagedKVCache(num_layers=2, num_kv_heads=2, head_dim=16,
+                         max_tokens=1 << 10, device="cpu",
+                         dtype=torch.float32)
+    h = cache.allocate(2, 128)
+    h.extend(40)
+    for l in range(2):
+        cache.k_pages(l).normal_()
+        cache.v_pages(l).normal_()
+    before = [(cache.k_pages(l)[torch.tensor(h.seqs[0].pages)].clone(),
+               cache.v_pages(l)[torch.tensor(h.seqs[0].pages)].clone())
+              for l in range(2)]
+    h.swap_out(to_disk=True, disk_dir=str(tmp_path))
+    assert h.is_swapped
+    assert any(p.suffix == ".pt" for p in tmp_path.iterdir())
+    # scribble over the whole pool while the session is on disk
+    for l in range(2):
+        cache.k_pages(l).zero_()
+        cache.v_pages(l).zero_()
+    h.swap_in()
+    assert not h.is_swapped
+    assert not any(p.suffix == ".pt" for p in tmp_path.iterdir())
+    for l in range(2):
+        k = cache.k_pages(l)[torch.tensor(h.seqs[0].pages)]
+        v = cache.v_pages(l)[torch.tensor(h.seqs[0].pages)]
+        assert torch.equal(k, before[l][0])
+        assert torch.equal(v, before[l][1])
+    h.close()
