@@ -347,3 +347,61 @@ def attn_paged_topk(
             vg, 1, top.indices.unsqueeze(-1).expand(-1, -1, D))  # (Hq, kk, D)
         out[b, :, 0] = torch.einsum("hk,hkd->hd", w, vsel).to(q.dtype)
     return out
+
+
+def attn_paged_mixed(
+    q: torch.Tensor,
+    k_pages: torch.Tensor, v_pages: torch.Tensor,
+    page_table: torch.Tensor,
+    ctx_lens: torch.Tensor,
+    host_k: torch.Tensor, host_v: torch.Tensor,
+    scale: Optional[float] = None,
+) -> torch.Tensor:
+    """Mixed-device decode attention (parity: reference
+    pytorch_backend.py:969-1014 `_mixed_device_attention` — KV split into a
+    device-resident recent segment and a host-resident old segment; the host
+    part is computed fp32 on CPU and the two partial softmaxes are merged).
+
+    host_k/host_v: (B, Hkv, S_host, D) on CPU — positions [0, S_host).
+    The paged pool holds positions [S_host, S_host + ctx_lens[b]).
+    q: (B, Hq, 1, D) on any device. Merge is exact (log-sum-exp), so the
+    result equals dense attention over the concatenated cache.
+    """
+    B, Hq, Tq, D = q.shape
+    assert Tq == 1, "mixed-device path is decode-only"
+    Hkv = k_pages.shape[1]
+    G = Hq // Hkv
+    S_host = host_k.shape[2]
+    if scale is None:
+        scale = 1.0 / math.sqrt(D)
+    out = torch.empty_like(q)
+    for b in range(B):
+        qf = q[b, :, 0].float()  # (Hq, D)
+        # host segment on CPU (fp32)
+        qc = qf.cpu()
+        kh = host_k[b].float().repeat_interleave(G, dim=0)  # (Hq, S_host, D)
+        vh = host_v[b].float().repeat_interleave(G, dim=0)
+        s_h = torch.einsum("hd,hcd->hc", qc, kh) * scale
+        m_h = s_h.max(-1, keepdim=True).values
+        p_h = torch.exp(s_h - m_h)
+        l_h = p_h.sum(-1)                                   # (Hq,)
+        o_h = torch.einsum("hc,hcd->hd", p_h, vh)
+        # device-resident recent segment via the paged pool
+        ctx = int(ctx_lens[b])
+        k, v = kv_gather(k_pages, v_pages, page_table, ctx, b)
+        kg = k.float().repeat_interleave(G, dim=0)
+        vg = v.float().repeat_interleave(G, dim=0)
+        s_d = torch.einsum("hd,hcd->hc", qf, kg) * scale
+        m_d = s_d.max(-1, keepdim=True).values
+        p_d = torch.exp(s_d - m_d)
+        l_d = p_d.sum(-1)
+        o_d = torch.einsum("hc,hcd->hd", p_d, vg)
+        # exact merge of the two partial softmaxes
+        m_hd, m_dd = m_h[:, 0].to(q.device), m_d[:, 0]
+        o_hd, l_hd = o_h.to(q.device), l_h.to(q.device)
+        m = torch.maximum(m_hd, m_dd)
+        c_h = torch.exp(m_hd - m).unsqueeze(-1)
+        c_d = torch.exp(m_dd - m).unsqueeze(-1)
+        denom = l_hd.unsqueeze(-1) * c_h + l_d.unsqueeze(-1) * c_d
+        out[b, :, 0] = ((o_hd * c_h + o_d * c_d) / denom).to(q.dtype)
+    return out

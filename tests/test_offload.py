@@ -108,3 +108,46 @@ def test_attn_sparsity_topk():
         assert torch.allclose(routed, sparse2, atol=1e-5)
     finally:
         ops.set_attn_sparsity(1.0)
+
+
+def test_mixed_device_attention_exact():
+    """Mixed GPU/host KV attention must equal dense attention over the
+    concatenated cache (ref _mixed_device_attention, exact LSE merge)."""
+    import torch
+
+    from bloombee_amd.ops import reference as ref
+
+    torch.manual_seed(1)
+    B, Hq, Hkv, D, P = 2, 4, 2, 32, 16
+    S_host, S_dev = 48, 32
+    # full cache in pages for the dense reference
+    np_full = B * ((S_host + S_dev) // P + 1)
+    kp = torch.randn(np_full, Hkv, P, D) * 0.3
+    vp = torch.randn(np_full, Hkv, D, P) * 0.3
+    pt = torch.arange(np_full, dtype=torch.int32).view(B, -1)
+    ctx_l = torch.full((B,), S_host + S_dev, dtype=torch.int32)
+    q = torch.randn(B, Hq, 1, D) * 0.3
+    dense = ref.attn_paged(q.float(), kp.float(), vp.float(), pt,
+                           ctx_l.long() - 1)
+
+    # split: first S_host positions -> host tensors, rest -> a second pool
+    hk = torch.empty(B, Hkv, S_host, D)
+    hv = torch.empty(B, Hkv, S_host, D)
+    for b in range(B):
+        k, v = ref.kv_gather(kp.float(), vp.float(), pt, S_host + S_dev, b)
+        hk[b], hv[b] = k[:, :S_host], v[:, :S_host]
+        dk, dv = k[:, S_host:], v[:, S_host:]
+        if b == 0:
+            np_dev = B * (S_dev // P + 1)
+            kp2 = torch.zeros(np_dev, Hkv, P, D)
+            vp2 = torch.zeros(np_dev, Hkv, D, P)
+            pt2 = torch.arange(np_dev, dtype=torch.int32).view(B, -1)
+        for t in range(S_dev):
+            pg = pt2[b, t // P]
+            kp2[pg, :, t % P] = dk[:, t]
+            vp2[pg, :, :, t % P] = dv[:, t]
+    mixed = ref.attn_paged_mixed(
+        q.float(), kp2, vp2, pt2, torch.full((B,), S_dev, dtype=torch.int32),
+        hk, hv)
+    assert torch.allclose(dense, mixed, atol=1e-4), \
+        (dense - mixed).abs().max()
