@@ -46,9 +46,10 @@ DEVINL float quad16_reduce_sum(float x) {
 // nch: GQA group chunks per kv head (ceil(G/16)) — MQA groups wider than the
 // 16-col MFMA q-tile (falcon-7b G=71) split into chunks sharing the kv head.
 // alibi: per-query-head slopes (bloom family), or null.
-// min 2 workgroups/CU: at the natural VGPR allocation the kernel sat at
-// 1 wave/SIMD with no latency hiding (measured ~67 us vs the ~19 us KV-stream
-// floor); capping registers doubles resident waves.
+// The S^T form compiles to 118 VGPRs / 0 spills -> 4 waves/SIMD resident
+// (the S-form needed 208 VGPR + 32 AGPR and sat at 2); launch_bounds keeps
+// the floor at 2 WGs/CU. Host-side n_split sizes the grid for ~1024
+// workgroups = 4 resident WGs on all 256 CUs (profiles/r01_st_attention.md).
 template <int D, int MAXG>
 __global__ __launch_bounds__(256, 2) void attn_decode_mfma_kernel(
     const unsigned short* __restrict__ q,        // strided, see q_sb/q_sh
