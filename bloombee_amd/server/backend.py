@@ -98,6 +98,19 @@ class StackBackend:
             logger.info("KV pressure: swapping out idle session %s", sid[:8])
             self.pool.submit(st.handle.swap_out, PRIORITY_TRAIN).result()
 
+    def reap_idle_sessions(self, max_idle_s: float = 600.0) -> int:
+        """Close sessions idle past max_idle_s (ref handler session GC —
+        a client that vanished without closing its stream must not pin KV
+        pages forever). Returns the number reaped."""
+        now = time.monotonic()
+        with self._lock:
+            dead = [sid for sid, st in self.sessions.items()
+                    if now - st.last_activity > max_idle_s]
+        for sid in dead:
+            logger.info("reaping idle session %s", sid[:8])
+            self.close_session(sid)
+        return len(dead)
+
     def close_session(self, session_id: str) -> None:
         with self._lock:
             state = self.sessions.pop(session_id, None)

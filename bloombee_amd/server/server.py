@@ -43,6 +43,7 @@ class Server:
         seed: int = 0,
         kv_max_tokens: int = 1 << 18,
         update_period: float = 30.0,
+        session_max_idle: float = 600.0,
         expiration: Optional[float] = None,
         checkpoint_dir: Optional[str] = None,
         throughput: float = 1.0,
@@ -55,6 +56,7 @@ class Server:
         self.device = device
         self.peer_id = uuid.uuid4().hex[:16]
         self.update_period = update_period
+        self.session_max_idle = session_max_idle
         self.expiration = expiration or max(60.0, 2 * update_period)
         self.throughput = throughput
         self.host = host
@@ -128,6 +130,7 @@ class Server:
                 declare_active_modules(self.dht, self.uids, self.peer_id,
                                        self._server_info(),
                                        time.time() + self.expiration)
+                self.backend.reap_idle_sessions(self.session_max_idle)
             except Exception as e:  # noqa: BLE001
                 logger.warning("announce failed: %s", e)
             self._stop.wait(self.update_period)

@@ -105,3 +105,23 @@ def test_per_module_log_levels(monkeypatch):
     bl.get_logger("client.session")
     assert stdlog.getLogger("bloombee_amd.client.session").level == stdlog.DEBUG
     assert stdlog.getLogger("bloombee_amd.server").level == stdlog.WARNING
+
+
+def test_idle_session_reaper():
+    """Sessions idle past the threshold are closed and their KV freed
+    (crashed-client leak protection)."""
+    import torch
+
+    from bloombee_amd.models.base import resolve_config
+    from bloombee_amd.server.backend import StackBackend
+
+    cfg = resolve_config("llama-tiny")
+    be = StackBackend(cfg, 0, 2, device="cpu", seed=0, kv_max_tokens=1 << 12)
+    be.open_session("s1", batch_size=1, max_length=64)
+    left_before = be.kv_pool.tokens_left
+    assert be.reap_idle_sessions(max_idle_s=3600) == 0   # fresh: kept
+    be.sessions["s1"].last_activity -= 7200
+    assert be.reap_idle_sessions(max_idle_s=3600) == 1   # idle: reaped
+    assert "s1" not in be.sessions
+    assert be.kv_pool.tokens_left > left_before
+    be.shutdown()
