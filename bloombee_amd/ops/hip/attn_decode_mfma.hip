@@ -285,3 +285,210 @@ __global__ __launch_bounds__(256, 2) void attn_decode_mfma_kernel(
     }
   }
 }
+
+// ---------------------------------------------------------------------------
+// Wide-head decode attention (D = 512: gemma-4 global layers).
+//
+// Same S^T formulation, but the head dim is split ACROSS the 4 waves instead
+// of splitting positions: wave w owns d-slice [128w, 128w+128). All waves
+// walk the SAME 32-position KV tile per iteration:
+//   1. wave w computes partial scores over its d-slice (4 MFMAs per 16-pos
+//      subtile) and publishes its C regs to LDS;
+//   2. after a barrier every wave sums the four partials IN THE SAME ORDER
+//      (bitwise-identical scores -> identical m/l/corr in all waves, so the
+//      per-wave normalization of the d-partitioned output is consistent);
+//   3. wave 0 writes P^T to LDS; every wave PV-accumulates its own 8 d-tiles
+//      (32 acc VGPRs instead of the 128 a monolithic D=512 tile would need).
+// No output merge: d-slices are disjoint. KV bytes read once, like D<=256.
+// ---------------------------------------------------------------------------
+
+template <int MAXG>
+__global__ __launch_bounds__(256, 2) void attn_decode_wide_kernel(
+    const unsigned short* __restrict__ q,
+    const unsigned short* __restrict__ k_pages,  // (np, Hkv, P, 512)
+    const unsigned short* __restrict__ v_pages,  // (np, Hkv, 512, P) d-major
+    const int* __restrict__ page_table, const int* __restrict__ ctx_lens,
+    const float* __restrict__ alibi, unsigned short* __restrict__ out,
+    float* __restrict__ part_ml, float* __restrict__ part_acc,
+    int B, int Hkv, int G, int nch, int P, int maxp, int n_split, int window,
+    float scale, long q_sb, long q_sh, long out_sb, long out_sh) {
+  constexpr int D = 512;
+  constexpr int NKK_W = 4;          // QK k-slices per wave (128 d)
+  constexpr int NDT_W = 8;          // PV d-tiles per wave (128 d)
+  constexpr int NPT = DKVBLK / 16;  // position subtiles
+  constexpr int PROW_B = DKVBLK * 2 + 16;
+
+  const int bh = blockIdx.x;
+  const int split = blockIdx.y;
+  const int b = bh / (Hkv * nch);
+  const int rem = bh % (Hkv * nch);
+  const int kvh = rem / nch;
+  const int g0 = (rem % nch) * MAXG;
+  const int Gl = min(G - g0, MAXG);
+  const int ctx = ctx_lens[b];
+
+  const int pages_total = (ctx + P - 1) / P;
+  const int pages_per_split = (pages_total + n_split - 1) / n_split;
+  const int c0 = split * pages_per_split * P;
+  const int c1 = min(ctx, c0 + pages_per_split * P);
+  int lo = 0;
+  if (window > 0) lo = max(0, ctx - window);
+
+  const int tid = threadIdx.x;
+  const int wave = tid / WAVE;
+  const int lane = tid & (WAVE - 1);
+  const int li = lane & 15;
+  const int hi = lane >> 4;
+  const int d0 = wave * 128;        // this wave's d-slice
+  const float sc2 = scale * LOG2E;
+
+  // LDS: per-wave partial-score slots (4 x 64 lanes x 8 floats) + one shared
+  // P^T tile (16 head rows x 32 positions, bf16)
+  __shared__ __attribute__((aligned(16))) float sc_lds[4][WAVE][NPT * 4];
+  __shared__ __attribute__((aligned(16))) unsigned char p_lds[16 * PROW_B];
+
+  const int qh = g0 + ((li < Gl) ? li : Gl - 1);
+  bf16x8 qfrag[NKK_W];
+#pragma unroll
+  for (int kk = 0; kk < NKK_W; ++kk)
+    qfrag[kk] = as_bf16x8(*reinterpret_cast<const short8*>(
+        q + (long)b * q_sb + (long)(kvh * G + qh) * q_sh + d0 + hi * 8 +
+        32 * kk));
+  const int ag = g0 + li;
+  const float aslope =
+      (alibi != nullptr && ag < G) ? alibi[kvh * G + ag] * LOG2E : 0.f;
+
+  float m2 = NEG_BIG, l = 0.f;
+  f32x4 acc_o[NDT_W];
+#pragma unroll
+  for (int n = 0; n < NDT_W; ++n) acc_o[n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+  const long head_slab_k = (long)kvh * P * D;
+  const int tile0 = (max(c0, lo) - c0) / DKVBLK;
+
+  for (int tb = c0 + tile0 * DKVBLK; tb < c1; tb += DKVBLK) {
+    // V^T fragments for this wave's d-slice (issued first, latency cover)
+    const int vpos = tb + hi * 8;
+    int vpage = -1;
+    if (vpos < c1) vpage = page_table[b * maxp + vpos / P];
+    const long vrow = ((long)vpage * Hkv + kvh) * D + d0;
+    bf16x8 vfrag[NDT_W];
+#pragma unroll
+    for (int n = 0; n < NDT_W; ++n) {
+      vfrag[n] = as_bf16x8(short8{});
+      if (vpage >= 0)
+        vfrag[n] = as_bf16x8(__builtin_nontemporal_load(
+            reinterpret_cast<const short8*>(
+                v_pages + (vrow + li + 16 * n) * P + (vpos % P))));
+    }
+
+    // partial S^T over this wave's 128 d
+    f32x4 st[NPT];
+#pragma unroll
+    for (int n = 0; n < NPT; ++n) {
+      const int pos = tb + li + 16 * n;
+      const int cpos = min(pos, c1 - 1);
+      const int page = page_table[b * maxp + cpos / P];
+      const unsigned short* krow = k_pages + ((long)page * Hkv) * P * D +
+                                   head_slab_k + (long)(cpos % P) * D + d0;
+      st[n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int kk = 0; kk < NKK_W; ++kk) {
+        bf16x8 kfrag = as_bf16x8(__builtin_nontemporal_load(
+            reinterpret_cast<const short8*>(krow + hi * 8 + 32 * kk)));
+        st[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kfrag, qfrag[kk],
+                                                        st[n], 0, 0, 0);
+      }
+    }
+    __syncthreads();  // previous iteration's P^T reads are done
+#pragma unroll
+    for (int n = 0; n < NPT; ++n)
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg)
+        sc_lds[wave][lane][n * 4 + reg] = st[n][reg];
+    __syncthreads();
+    // full scores: fixed summation order -> bitwise identical in all waves
+    float p[NPT][4];
+    float rm = NEG_BIG;
+#pragma unroll
+    for (int n = 0; n < NPT; ++n)
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        float sv = sc_lds[0][lane][n * 4 + reg];
+#pragma unroll
+        for (int w = 1; w < 4; ++w) sv += sc_lds[w][lane][n * 4 + reg];
+        const int kpos = tb + 16 * n + hi * 4 + reg;
+        sv = sv * sc2 + aslope * (float)kpos;
+        const bool dead = (kpos >= c1) | (kpos < lo);
+        sv = dead ? NEG_BIG : sv;
+        p[n][reg] = sv;
+        rm = fmaxf(rm, sv);
+      }
+    rm = quad16_reduce_max(rm);
+    const float mn = fmaxf(m2, rm);
+    const float corr = fast_exp2(m2 - mn);
+    float psum = 0.f;
+#pragma unroll
+    for (int n = 0; n < NPT; ++n)
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        p[n][reg] = fast_exp2(p[n][reg] - mn);
+        psum += p[n][reg];
+      }
+    psum = quad16_reduce_sum(psum);
+    l = l * corr + psum;
+    m2 = mn;
+    if (__builtin_amdgcn_ballot_w64(corr < 0.9999f)) {
+#pragma unroll
+      for (int n = 0; n < NDT_W; ++n)
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg) acc_o[n][reg] *= corr;
+    }
+
+    // one shared P^T tile (identical p in every wave; wave 0 publishes)
+    if (wave == 0) {
+#pragma unroll
+      for (int n = 0; n < NPT; ++n)
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg)
+          *(unsigned short*)(p_lds + li * PROW_B +
+                             (16 * n + hi * 4 + reg) * 2) = f2bf(p[n][reg]);
+    }
+    __syncthreads();
+    bf16x8 pfrag = as_bf16x8(
+        *reinterpret_cast<const short8*>(p_lds + li * PROW_B + hi * 16));
+#pragma unroll
+    for (int n = 0; n < NDT_W; ++n)
+      acc_o[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vfrag[n], pfrag,
+                                                         acc_o[n], 0, 0, 0);
+  }
+
+  // d-partitioned output: every wave writes its own 128-d slice; (m, l) are
+  // identical across waves by construction
+  if (n_split == 1) {
+    if (li < Gl) {
+      const float inv = (l > 0.f) ? 1.f / l : 0.f;
+      unsigned short* orow =
+          out + (long)b * out_sb + (long)(kvh * G + g0 + li) * out_sh + d0;
+#pragma unroll
+      for (int n = 0; n < NDT_W; ++n) {
+        short4v o4;
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg) o4[reg] = f2bf(acc_o[n][reg] * inv);
+        *reinterpret_cast<short4v*>(orow + 16 * n + hi * 4) = o4;
+      }
+    }
+  } else {
+    const long pbase = ((long)bh * n_split + split);
+    if (li < Gl) {
+      float* prow = part_acc + (pbase * MAXG + li) * D + d0;
+#pragma unroll
+      for (int n = 0; n < NDT_W; ++n)
+        *reinterpret_cast<f32x4*>(prow + 16 * n + hi * 4) = acc_o[n];
+      if (hi == 0 && wave == 0) {
+        part_ml[(pbase * MAXG + li) * 2 + 0] = m2;
+        part_ml[(pbase * MAXG + li) * 2 + 1] = l;
+      }
+    }
+  }
+}

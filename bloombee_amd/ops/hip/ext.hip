@@ -204,12 +204,22 @@ static void attn_decode_launch(const torch::Tensor& q, const torch::Tensor& kp,
   dim3 grid(B * Hkv * nch, n_split);
   const int maxg = (G <= 4 && nch == 1) ? 4 : 16;
   auto launch_mfma = [&](auto mg) {
-    attn_decode_mfma_kernel<D, decltype(mg)::value>
-        <<<grid, 256, 0, cur_stream()>>>(
-            bf_ptr(q) + q_off, bf_ptr(kp), bf_ptr(vp), pt.data_ptr<int>(),
-            ctx.data_ptr<int>(), alibi_ptr, bf_ptr_mut(out),
-            pml.data_ptr<float>(), pacc.data_ptr<float>(), B, Hkv, G, nch, P,
-            maxp, n_split, window, scale, q_sb, q_sh, out_sb, out_sh);
+    if constexpr (D == 512) {
+      // wide-head path: d split across the 4 waves (gemma-4 global layers)
+      attn_decode_wide_kernel<decltype(mg)::value>
+          <<<grid, 256, 0, cur_stream()>>>(
+              bf_ptr(q) + q_off, bf_ptr(kp), bf_ptr(vp), pt.data_ptr<int>(),
+              ctx.data_ptr<int>(), alibi_ptr, bf_ptr_mut(out),
+              pml.data_ptr<float>(), pacc.data_ptr<float>(), B, Hkv, G, nch, P,
+              maxp, n_split, window, scale, q_sb, q_sh, out_sb, out_sh);
+    } else {
+      attn_decode_mfma_kernel<D, decltype(mg)::value>
+          <<<grid, 256, 0, cur_stream()>>>(
+              bf_ptr(q) + q_off, bf_ptr(kp), bf_ptr(vp), pt.data_ptr<int>(),
+              ctx.data_ptr<int>(), alibi_ptr, bf_ptr_mut(out),
+              pml.data_ptr<float>(), pacc.data_ptr<float>(), B, Hkv, G, nch, P,
+              maxp, n_split, window, scale, q_sb, q_sh, out_sb, out_sh);
+    }
   };
   if (maxg == 4) launch_mfma(std::integral_constant<int, 4>{});
   else launch_mfma(std::integral_constant<int, 16>{});
@@ -231,7 +241,7 @@ static torch::Tensor attn_decode_core(
   const int Hkv = k_pages.size(1), P = k_pages.size(2), maxp = page_table.size(1);
   const int G = Hq / Hkv;
   TORCH_CHECK(Hq % Hkv == 0);
-  TORCH_CHECK(D < 256 || G <= 4, "D=256 decode supports GQA group size <= 4");
+  TORCH_CHECK(D != 256 || G <= 4, "D=256 decode supports GQA group size <= 4");
   const int nch = (G + 15) / 16;  // MQA chunks (falcon G=71 -> 5)
   const float* alibi_ptr = nullptr;
   if (alibi.has_value()) {
@@ -266,6 +276,7 @@ static torch::Tensor attn_decode_core(
   if (D == 128) go(std::integral_constant<int, 128>{});
   else if (D == 64) go(std::integral_constant<int, 64>{});
   else if (D == 256) go(std::integral_constant<int, 256>{});
+  else if (D == 512) go(std::integral_constant<int, 512>{});
   else if (D == 32) go(std::integral_constant<int, 32>{});
   else TORCH_CHECK(false, "unsupported head_dim ", D);
   return out;

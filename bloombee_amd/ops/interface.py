@@ -141,10 +141,9 @@ def attn_decode(
                                    alibi_slopes=alibi_slopes)
     if _on_gpu(q):
         _require_ext()
-        if q.shape[-1] > 256:
-            # gemma-4 global layers (D=512): register pressure puts the MFMA
-            # kernel past the VGPR budget; run the gather-based torch
-            # composition on-device (kernel support tracked for round 2)
+        if q.shape[-1] > 256 and q.shape[-1] != 512:
+            # odd wide head dims: gather-based torch composition on-device
+            # (D=512 runs attn_decode_wide_kernel — d split across waves)
             return ref.attn_paged(q, k_pages, v_pages, page_table,
                                   ctx_lens.long() - 1, scale,
                                   sliding_window=window if window > 0 else None,

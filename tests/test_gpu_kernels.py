@@ -133,6 +133,9 @@ def _paged_setup(B, Hq, Hkv, T, D, P=16, seed=7, dtype=torch.bfloat16):
     (2, 8, 1, 57, 128),    # MQA G=8
     (1, 2, 2, 300, 64),    # D=64
     (1, 2, 1, 40, 256),    # D=256 (gemma-class)
+    (2, 8, 4, 77, 512),    # D=512 wide kernel (gemma-4 global, G=2)
+    (1, 4, 4, 130, 512),   # D=512 MHA, crosses page boundaries
+    (1, 32, 2, 50, 512),   # D=512 G=16 (MAXG=16 instantiation)
 ])
 def test_attn_decode_parity(shape):
     B, Hq, Hkv, T, D = shape
