@@ -68,6 +68,10 @@ class StackBackend:
                                  seed=seed, dtype=config.dtype, device=device)
                 self.pruner = create_pruner(method, head)
                 logger.info("mid-network tree pruner enabled: %s", method)
+        # sequence-chunk bound for long prefills (ref max_chunk_size_bytes)
+        import os as _os2
+        self.max_chunk_tokens = int(_os2.environ.get("BBAMD_MAX_CHUNK_TOKENS",
+                                                     "4096"))
         self.pool = TaskPool(name=f"worker[{start}:{end}]")
         self.sessions: Dict[str, SessionState] = {}
         self._lock = threading.Lock()
@@ -193,13 +197,32 @@ class StackBackend:
                     f"only {cur} tokens (gap)")
             elif speculative:
                 handle.rollback()  # drop any uncommitted previous tree
-            sp = torch.full((B,), start_pos, dtype=torch.int32, device=self.device)
-            handle.extend(T, speculative=speculative)
             pos = (position_ids.to(self.device).int()
                    if position_ids is not None else None)
             tm = tree_mask.to(self.device) if tree_mask is not None else None
-            out = self.stack.forward_inference(h, handle, sp, pos, tree_mask=tm,
-                                               deep_prompts=deep)
+            # long-prefill sequence chunking (ref backend.py:525-531,
+            # 839-845 max_chunk_size_bytes): bound activation memory by
+            # running the span over sequence slices; each chunk's KV lands
+            # in the pages before the next chunk attends (the prefill
+            # kernel's continuation path)
+            if (tm is None and pos is None and not speculative
+                    and T > self.max_chunk_tokens):
+                outs = []
+                for t0 in range(0, T, self.max_chunk_tokens):
+                    t1 = min(T, t0 + self.max_chunk_tokens)
+                    handle.extend(t1 - t0)
+                    sp = torch.full((B,), start_pos + t0, dtype=torch.int32,
+                                    device=self.device)
+                    outs.append(self.stack.forward_inference(
+                        h[:, t0:t1], handle, sp, deep_prompts=deep))
+                out = torch.cat(outs, dim=1)
+            else:
+                sp = torch.full((B,), start_pos, dtype=torch.int32,
+                                device=self.device)
+                handle.extend(T, speculative=speculative)
+                out = self.stack.forward_inference(h, handle, sp, pos,
+                                                   tree_mask=tm,
+                                                   deep_prompts=deep)
             from bloombee_amd.utils import activation_dumper
             if activation_dumper.enabled():
                 activation_dumper.capture_activation(

@@ -125,3 +125,35 @@ def test_idle_session_reaper():
     assert "s1" not in be.sessions
     assert be.kv_pool.tokens_left > left_before
     be.shutdown()
+
+
+def test_long_prefill_chunking_matches_unchunked(monkeypatch):
+    """Sequence-chunked prefill (BBAMD_MAX_CHUNK_TOKENS) must produce the
+    same hidden states as one whole-sequence pass (ref max_chunk_size_bytes
+    chunking, backend.py:525-531)."""
+    import torch
+
+    from bloombee_amd.models.base import resolve_config
+    from bloombee_amd.server.backend import StackBackend
+
+    cfg = resolve_config("llama-tiny")
+    gen = torch.Generator().manual_seed(3)
+    h = (torch.randn(2, 50, cfg.hidden_size, generator=gen) * 0.1).to(cfg.dtype)
+
+    be1 = StackBackend(cfg, 0, 2, device="cpu", seed=0, kv_max_tokens=1 << 12)
+    be1.open_session("a", batch_size=2, max_length=128)
+    whole = be1.inference_step("a", h, 0)
+    be1.shutdown()
+
+    monkeypatch.setenv("BBAMD_MAX_CHUNK_TOKENS", "16")
+    be2 = StackBackend(cfg, 0, 2, device="cpu", seed=0, kv_max_tokens=1 << 12)
+    assert be2.max_chunk_tokens == 16
+    be2.open_session("a", batch_size=2, max_length=128)
+    chunked = be2.inference_step("a", h, 0)
+    # decode continues correctly after a chunked prefill
+    nxt = (torch.randn(2, 1, cfg.hidden_size, generator=gen) * 0.1).to(cfg.dtype)
+    d1 = be2.inference_step("a", nxt, 50)
+    be2.shutdown()
+
+    assert torch.equal(whole, chunked)
+    assert torch.isfinite(d1.float()).all()
