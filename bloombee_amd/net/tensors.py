@@ -26,6 +26,16 @@ from typing import Any, Dict, List, Optional, Tuple
 import msgpack
 import torch
 
+# native byte-split + DEFLATE (ops/hip/wire.h) — same stream format as the
+# Python path below; used when the extension is importable (it builds for
+# CPU hosts too), Python zlib otherwise
+try:
+    from bloombee_amd.ops.interface import hip_ops as _native
+    if _native is None or not hasattr(_native, "wire_deflate"):
+        _native = None
+except Exception:  # pragma: no cover - import-order edge
+    _native = None
+
 _DTYPES = {
     "torch.bfloat16": torch.bfloat16,
     "torch.float16": torch.float16,
@@ -51,16 +61,37 @@ def _raw_bytes(t: torch.Tensor) -> bytes:
     return t.numpy().tobytes()
 
 
+def _deflate(raw: bytes, bsplit: bool) -> bytes:
+    if _native is not None:
+        t = torch.frombuffer(bytearray(raw), dtype=torch.uint8)
+        return bytes(_native.wire_deflate(t, bsplit, 1).numpy().tobytes())
+    if bsplit:
+        hi, lo = raw[1::2], raw[0::2]
+        return zlib.compress(hi + lo, level=1)
+    return zlib.compress(raw, level=1)
+
+
+import os as _os
+_MT_THREADS = min(32, (_os.cpu_count() or 8) * 2)
+
+
 def serialize_tensor(t: torch.Tensor, codec: str = "raw") -> Tuple[dict, bytes]:
     """-> (header dict, payload bytes)."""
     raw = _raw_bytes(t)
-    if codec == "zlib":
-        payload = zlib.compress(raw, level=1)
+    if codec == "bsplit+zlibmt" and t.dtype in _TWO_BYTE and _native is not None:
+        # multithreaded chunked DEFLATE (ops/hip/wire.h) — the native codec
+        # that makes compression viable at multi-MB activation payloads
+        r = torch.frombuffer(bytearray(raw), dtype=torch.uint8)
+        payload = bytes(_native.wire_deflate_mt(r, True, 1,
+                                                _MT_THREADS).numpy().tobytes())
+        if len(payload) >= len(raw):
+            codec, payload = "raw", raw
+    elif codec == "zlib":
+        payload = _deflate(raw, False)
         if len(payload) >= len(raw):  # min-gain gate (ref :167-186)
             codec, payload = "raw", raw
     elif codec == "bsplit+zlib" and t.dtype in _TWO_BYTE:
-        hi, lo = raw[1::2], raw[0::2]
-        payload = zlib.compress(hi + lo, level=1)
+        payload = _deflate(raw, True)
         if len(payload) >= len(raw):
             codec, payload = "raw", raw
     else:
@@ -81,12 +112,21 @@ def deserialize_tensor(head: dict, payload: bytes) -> torch.Tensor:
     codec = head["codec"]
     if codec == "zlib":
         raw = zlib.decompress(payload)
+    elif codec == "bsplit+zlibmt":
+        c = torch.frombuffer(bytearray(payload), dtype=torch.uint8)
+        nraw = int(torch.tensor(shape).prod()) * 2 if shape else 2
+        raw = bytes(_native.wire_inflate_mt(c, nraw, True).numpy().tobytes())
     elif codec == "bsplit+zlib":
-        raw = zlib.decompress(payload)
-        n = len(raw) // 2
-        hi, lo = raw[:n], raw[n:]
-        raw = bytes(b for pair in zip(lo, hi) for b in pair) if n < 1 << 12 else \
-            _interleave(lo, hi)
+        if _native is not None:
+            c = torch.frombuffer(bytearray(payload), dtype=torch.uint8)
+            nraw = int(torch.tensor(shape).prod()) * 2 if shape else 2
+            raw = bytes(_native.wire_inflate(c, nraw, True).numpy().tobytes())
+        else:
+            raw = zlib.decompress(payload)
+            n = len(raw) // 2
+            hi, lo = raw[:n], raw[n:]
+            raw = bytes(b for pair in zip(lo, hi) for b in pair) \
+                if n < 1 << 12 else _interleave(lo, hi)
     elif codec == "raw":
         raw = payload
     else:

@@ -19,6 +19,7 @@
 #include "gemm_skinny.hip"
 #include "mfma_selftest.hip"
 #include "quant4.hip"
+#include "wire.h"
 
 typedef __attribute__((ext_vector_type(8))) short short8_h;
 
@@ -571,6 +572,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("add_bf16", &add_bf16);
   m.def("kv_write", &kv_write);
   m.def("kv_gather", &kv_gather);
+  m.def("wire_deflate", &wire_deflate);
+  m.def("wire_inflate", &wire_inflate);
+  m.def("wire_deflate_mt", &wire_deflate_mt);
+  m.def("wire_inflate_mt", &wire_inflate_mt);
   m.def("attn_decode", &attn_decode);
   m.def("attn_decode_qkv", &attn_decode_qkv);
   m.def("attn_prefill_qkv", &attn_prefill_qkv);

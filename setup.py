@@ -50,7 +50,7 @@ def build(verbose: bool = True) -> Path:
         "g++", "-shared", str(obj),
         "-L", str(tlib), "-L", "/opt/rocm/lib",
         "-lc10", "-ltorch", "-ltorch_cpu", "-ltorch_python",
-        "-lamdhip64", "-lc10_hip", "-ltorch_hip",
+        "-lamdhip64", "-lc10_hip", "-ltorch_hip", "-lz",
         "-o", str(OUT_SO),
     ]
     for cmd in (compile_cmd, link_cmd):

@@ -119,3 +119,41 @@ def test_declare_and_spans():
     finally:
         n2.shutdown()
         boot.shutdown()
+
+
+def test_wire_codec_native_python_interop():
+    """The C++ wire codec (ops/hip/wire.h) and the pure-Python fallback must
+    produce mutually decodable streams (identical format)."""
+    import zlib as _zlib
+
+    import torch
+
+    from bloombee_amd.net import tensors as T
+
+    if T._native is None:
+        import pytest
+        pytest.skip("native codec not built")
+    t = (torch.randn(33, 17) * 0.1).bfloat16()
+    raw = T._raw_bytes(t)
+
+    # native-compress -> python-decompress
+    head, payload = T.serialize_tensor(t, codec="bsplit+zlib")
+    if head["codec"] == "bsplit+zlib":
+        dec = _zlib.decompress(payload)
+        n = len(dec) // 2
+        hi, lo = dec[:n], dec[n:]
+        joined = bytes(b for pair in zip(lo, hi) for b in pair)
+        assert joined == raw
+
+    # python-compress -> native-decompress (same header format)
+    hi, lo = raw[1::2], raw[0::2]
+    py_payload = _zlib.compress(hi + lo, level=1)
+    back = T.deserialize_tensor(
+        {"dtype": "torch.bfloat16", "shape": list(t.shape),
+         "codec": "bsplit+zlib", "nbytes": len(py_payload)}, py_payload)
+    assert torch.equal(back, t)
+
+    # scalar edge
+    s0 = torch.tensor(1.5, dtype=torch.bfloat16)
+    h0, p0 = T.serialize_tensor(s0, codec="bsplit+zlib")
+    assert torch.equal(T.deserialize_tensor(h0, p0), s0)

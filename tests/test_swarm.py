@@ -342,3 +342,15 @@ def test_deep_ptune_microbatch_split_matches_small_batch(swarm):
     out_b1 = model.generate(prompt[:1], max_new_tokens=4)
     assert torch.equal(out_b8[:1], out_b1), (out_b8[:1], out_b1)
     model.remote.manager.shutdown()
+
+
+def test_swarm_mt_compressed_wire_matches_local(swarm):
+    """The native multithreaded chunked codec is lossless end-to-end."""
+    boot, _ = swarm
+    model = _make_model(boot, wire_codec="bsplit+zlibmt")
+    gen = torch.Generator().manual_seed(5)
+    prompt = torch.randint(0, 1000, (2, 7), generator=gen)
+    out = model.generate(prompt, max_new_tokens=6)
+    expect = _local_tokens(prompt, 6)
+    assert torch.equal(out[:, 7:], expect)
+    model.remote.manager.shutdown()
