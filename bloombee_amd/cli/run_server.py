@@ -18,6 +18,16 @@ def parse_endpoint(s: str):
     return host, int(port)
 
 
+def _offload_policy(args):
+    if (args.weight_gpu_percent >= 100.0 and not args.compress_weight
+            and args.attn_sparsity >= 1.0):
+        return None
+    from bloombee_amd.offload import OffloadPolicy
+    return OffloadPolicy(weight_gpu_percent=args.weight_gpu_percent,
+                         compress_weight=args.compress_weight,
+                         attn_sparsity=args.attn_sparsity)
+
+
 def main():
     ap = argparse.ArgumentParser(description=__doc__)
     ap.add_argument("model", help="model preset name or local checkpoint dir")
@@ -37,7 +47,21 @@ def main():
                     help="per-block .npy weight dir (random init if absent)")
     ap.add_argument("--throughput", type=float, default=None,
                     help="announced rps (default: measured)")
+    ap.add_argument("--session-max-idle", type=float, default=600.0,
+                    help="reap inference sessions idle past this (s)")
+    ap.add_argument("--max-chunk-tokens", type=int, default=None,
+                    help="prefill sequence-chunk bound (BBAMD_MAX_CHUNK_TOKENS)")
+    ap.add_argument("--weight-gpu-percent", type=float, default=100.0,
+                    help="FlexGen-style offload: %% of blocks resident in HBM")
+    ap.add_argument("--compress-weight", action="store_true",
+                    help="4-bit group-quantize the host weight tier")
+    ap.add_argument("--attn-sparsity", type=float, default=1.0,
+                    help="top-k sparse decode attention fraction (1.0 = dense)")
     args = ap.parse_args()
+
+    if args.max_chunk_tokens is not None:
+        import os
+        os.environ["BBAMD_MAX_CHUNK_TOKENS"] = str(args.max_chunk_tokens)
 
     import torch
 
@@ -64,7 +88,8 @@ def main():
         block_indices=block_indices, num_blocks=args.num_blocks,
         device=device, seed=args.seed, kv_max_tokens=args.attn_cache_tokens,
         update_period=args.update_period, checkpoint_dir=args.checkpoint_dir,
-        throughput=throughput,
+        throughput=throughput, session_max_idle=args.session_max_idle,
+        offload_policy=_offload_policy(args),
     )
     server.run()
 
