@@ -113,9 +113,23 @@ def deserialize_tensor(head: dict, payload: bytes) -> torch.Tensor:
     if codec == "zlib":
         raw = zlib.decompress(payload)
     elif codec == "bsplit+zlibmt":
-        c = torch.frombuffer(bytearray(payload), dtype=torch.uint8)
         nraw = int(torch.tensor(shape).prod()) * 2 if shape else 2
-        raw = bytes(_native.wire_inflate_mt(c, nraw, True).numpy().tobytes())
+        if _native is not None:
+            c = torch.frombuffer(bytearray(payload), dtype=torch.uint8)
+            raw = bytes(_native.wire_inflate_mt(c, nraw, True).numpy().tobytes())
+        else:
+            # pure-Python decode of the chunked stream (same format)
+            nt, chunk = struct.unpack_from("<II", payload, 0)
+            lens = struct.unpack_from(f"<{nt}I", payload, 8)
+            off = 8 + 4 * nt
+            parts = []
+            for ln in lens:
+                if ln:
+                    parts.append(zlib.decompress(payload[off:off + ln]))
+                off += ln
+            dec = b"".join(parts)
+            n = len(dec) // 2
+            raw = _interleave(dec[n:], dec[:n])
     elif codec == "bsplit+zlib":
         if _native is not None:
             c = torch.frombuffer(bytearray(payload), dtype=torch.uint8)

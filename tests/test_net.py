@@ -157,3 +157,23 @@ def test_wire_codec_native_python_interop():
     s0 = torch.tensor(1.5, dtype=torch.bfloat16)
     h0, p0 = T.serialize_tensor(s0, codec="bsplit+zlib")
     assert torch.equal(T.deserialize_tensor(h0, p0), s0)
+
+
+def test_wire_mt_python_fallback_decode():
+    """A peer without the native extension must still decode mt streams."""
+    import torch
+
+    from bloombee_amd.net import tensors as T
+
+    if T._native is None:
+        import pytest
+        pytest.skip("native codec not built")
+    t = (torch.randn(600, 33) * 0.2).bfloat16()
+    head, payload = T.serialize_tensor(t, codec="bsplit+zlibmt")
+    saved = T._native
+    try:
+        T._native = None
+        back = T.deserialize_tensor(head, payload)
+    finally:
+        T._native = saved
+    assert torch.equal(back, t)
