@@ -32,6 +32,12 @@ __global__ __launch_bounds__(512) void attn_prefill_kernel(
     const int* __restrict__ page_table,          // (B, maxp)
     const int* __restrict__ q_start,             // (B,)
     const float* __restrict__ alibi,             // (Hq,) slopes or null
+    const unsigned char* __restrict__ tree_mask, // (B, Tq, Tq) bool or null:
+                                                 // spec-verify ancestor mask
+                                                 // over the NEW tokens (ref
+                                                 // backend.py:944-1047);
+                                                 // committed positions < qs
+                                                 // stay fully visible
     unsigned short* __restrict__ out,            // (B, Hq, Tq, D)
     int B, int Hq, int G, int Tq, int P, int maxp, int window, float scale,
     long q_sb, long q_st, long q_sh, long o_sb, long o_st, long o_sh) {
@@ -81,9 +87,10 @@ __global__ __launch_bounds__(512) void attn_prefill_kernel(
 #pragma unroll
   for (int n = 0; n < NDT; ++n) acc_o[n] = (f32x4){0.f, 0.f, 0.f, 0.f};
 
-  // key range this q-tile can see: [lo_min, kmax)
+  // key range this q-tile can see: [lo_min, kmax). Tree queries are not
+  // causal among themselves — every node may see any masked-visible node.
   const int qpos_max = qs + min(qbase + 127, Tq - 1);
-  const int kmax = min(ctx, qpos_max + 1);
+  const int kmax = (tree_mask != nullptr) ? ctx : min(ctx, qpos_max + 1);
   int kstart = 0;
   if (window > 0) kstart = max(0, ((qs + qbase) - window + 1) / KVBLK * KVBLK);
 
@@ -145,8 +152,16 @@ __global__ __launch_bounds__(512) void attn_prefill_kernel(
       for (int n = 0; n < KVBLK / 16; ++n) {
         const int kpos = kbase + li + 16 * n;
         float sv = s[n][reg] * sc2 + aslope * (float)kpos;
-        const bool dead = (qrow >= Tq) | (kpos > qpos) | (kpos >= ctx) |
-                          (window > 0 && kpos <= qpos - window);
+        bool dead = (qrow >= Tq) | (kpos >= ctx);
+        if (!dead) {
+          if (tree_mask != nullptr) {
+            if (kpos >= qs)
+              dead = !tree_mask[((long)b * Tq + qrow) * Tq + (kpos - qs)];
+          } else {
+            dead = (kpos > qpos) |
+                   (window > 0 && kpos <= qpos - window);
+          }
+        }
         sv = dead ? NEG_BIG : sv;
         p[n][reg] = sv;
         rm = fmaxf(rm, sv);

@@ -324,7 +324,8 @@ torch::Tensor attn_decode_qkv(torch::Tensor qkv, long Hq, torch::Tensor k_pages,
 torch::Tensor attn_prefill(torch::Tensor q, torch::Tensor k_pages,
                            torch::Tensor v_pages, torch::Tensor page_table,
                            torch::Tensor q_start, double scale, long window,
-                           c10::optional<torch::Tensor> alibi) {
+                           c10::optional<torch::Tensor> alibi,
+                           c10::optional<torch::Tensor> tree_mask) {
   CHECK_DEV(q); CHECK_BF16(q); CHECK_CONTIG(q);
   CHECK_DEV_ALL3(k_pages, page_table, q_start);
   if (alibi.has_value()) CHECK_DEV((*alibi));
@@ -337,13 +338,23 @@ torch::Tensor attn_prefill(torch::Tensor q, torch::Tensor k_pages,
     TORCH_CHECK(alibi->scalar_type() == at::kFloat && alibi->numel() == Hq);
     alibi_ptr = alibi->data_ptr<float>();
   }
+  const unsigned char* tm_ptr = nullptr;
+  if (tree_mask.has_value()) {
+    TORCH_CHECK(tree_mask->scalar_type() == at::kBool &&
+                tree_mask->is_contiguous() &&
+                tree_mask->sizes() == (at::IntArrayRef{B, Tq, Tq}),
+                "tree_mask must be contiguous bool (B, Tq, Tq)");
+    TORCH_CHECK(window == 0, "tree_mask + sliding window: use the torch path");
+    CHECK_DEV((*tree_mask));
+    tm_ptr = tree_mask->data_ptr<unsigned char>();
+  }
   auto out = torch::empty_like(q);
   dim3 grid((Tq + 127) / 128, B * Hq);
   auto launch = [&](auto d) {
     attn_prefill_kernel<decltype(d)::value><<<grid, 512, 0, cur_stream()>>>(
         bf_ptr(q), bf_ptr(k_pages), bf_ptr(v_pages), page_table.data_ptr<int>(),
-        q_start.data_ptr<int>(), alibi_ptr, bf_ptr_mut(out), B, Hq, G, Tq, P,
-        maxp, (int)window, (float)scale,
+        q_start.data_ptr<int>(), alibi_ptr, tm_ptr, bf_ptr_mut(out), B, Hq, G,
+        Tq, P, maxp, (int)window, (float)scale,
         (long)Hq * Tq * D, (long)D, (long)Tq * D,
         (long)Hq * Tq * D, (long)D, (long)Tq * D);
   };
@@ -375,8 +386,8 @@ torch::Tensor attn_prefill_qkv(torch::Tensor qkv, long Hq_,
   auto launch = [&](auto d) {
     attn_prefill_kernel<decltype(d)::value><<<grid, 512, 0, cur_stream()>>>(
         bf_ptr(qkv), bf_ptr(k_pages), bf_ptr(v_pages), page_table.data_ptr<int>(),
-        q_start.data_ptr<int>(), nullptr, bf_ptr_mut(out), B, Hq, G, Tq, P,
-        maxp, (int)window, (float)scale,
+        q_start.data_ptr<int>(), nullptr, nullptr, bf_ptr_mut(out), B, Hq, G,
+        Tq, P, maxp, (int)window, (float)scale,
         (long)Tq * X * D, (long)X * D, (long)D,
         (long)Tq * Hq * D, (long)Hq * D, (long)D);
   };

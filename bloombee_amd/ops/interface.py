@@ -9,6 +9,7 @@ from __future__ import annotations
 
 import importlib
 import math
+import os
 from typing import Optional, Tuple
 
 import torch
@@ -117,6 +118,7 @@ def kv_gather(k_pages, v_pages, page_table, ctx_len: int, batch_index: int):
 
 
 _attn_sparsity = 1.0  # Policy.attn_sparsity (ref flexgen policy :10-55)
+_TREE_KERNEL = os.environ.get("BBAMD_TREE_KERNEL", "") not in ("", "0")
 
 
 def set_attn_sparsity(frac: float) -> None:
@@ -176,7 +178,7 @@ def attn_prefill(
         al = (alibi_slopes.float().to(q.device).contiguous()
               if alibi_slopes is not None else None)
         return hip_ops.attn_prefill(q.contiguous(), k_pages, v_pages, page_table,
-                                    q_start.int(), scale, window, al)
+                                    q_start.int(), scale, window, al, None)
     return ref.attn_paged(q, k_pages, v_pages, page_table, q_start, scale,
                           sliding_window=window if window > 0 else None,
                           alibi_slopes=alibi_slopes)
@@ -244,11 +246,21 @@ def attn_paged(q, k_pages, v_pages, page_table, q_start, scale=None, window: int
                               alibi_slopes=alibi_slopes,
                               sliding_window=window if window > 0 else None)
     if tree_mask is not None:
-        # tree-attention (spec decode verify): device-agnostic torch
-        # composition (ref.attn_paged runs on GPU tensors directly; a fused
-        # tree-mask kernel is a round-2 item)
         if scale is None:
             scale = 1.0 / math.sqrt(q.shape[-1])
+        if (_on_gpu(q) and _TREE_KERNEL and window == 0
+                and q.shape[-1] <= 256 and alibi_slopes is None):
+            # fused tree-mask prefill kernel (spec verify). Shipped dark in
+            # round 1: BBAMD_TREE_KERNEL=1 enables it; numerics gated by
+            # tests/test_gpu_kernels.py -m treekernel before flipping the
+            # default in round 2.
+            _require_ext()
+            tm = tree_mask.to(q.device).contiguous().bool()
+            return hip_ops.attn_prefill(q.contiguous(), k_pages, v_pages,
+                                        page_table, q_start.int(), scale, 0,
+                                        None, tm)
+        # tree-attention (spec decode verify): device-agnostic torch
+        # composition (ref.attn_paged runs on GPU tensors directly)
         return ref.attn_paged(q, k_pages, v_pages, page_table,
                               q_start.to(q.device), scale,
                               tree_mask=tree_mask.to(q.device))
