@@ -37,3 +37,24 @@ def get_client(host: str, port: int) -> RpcClient:
         if key not in _clients:
             _clients[key] = RpcClient(host, int(port))
         return _clients[key]
+
+
+def shutdown_clients() -> None:
+    """Close every cached client's connection and reader task (quiet
+    interpreter shutdown — otherwise the loop is GC'd with pending
+    _read_loop tasks and asyncio logs 'Task was destroyed')."""
+    with _lock:
+        clients, loop = list(_clients.values()), _loop
+        _clients.clear()
+    if loop is None or loop.is_closed():
+        return
+    for c in clients:
+        try:
+            asyncio.run_coroutine_threadsafe(c.close(), loop).result(2)
+        except Exception:
+            pass
+
+
+import atexit  # noqa: E402
+
+atexit.register(shutdown_clients)

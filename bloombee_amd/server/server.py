@@ -181,7 +181,15 @@ class Server:
                 fut.result(timeout=5)
             except Exception:
                 pass
-            self._loop.call_soon_threadsafe(self._loop.stop)
+
+            def _drain_and_stop(loop=self._loop):
+                # cancel stragglers (e.g. s2s push clients' read loops) so
+                # the loop is not GC'd with pending tasks
+                for t in asyncio.all_tasks(loop):
+                    t.cancel()
+                loop.stop()
+
+            self._loop.call_soon_threadsafe(_drain_and_stop)
         if self._thread is not None:
             self._thread.join(timeout=5)
         self.backend.shutdown()
