@@ -73,3 +73,35 @@ def test_cli_dht_and_server_serve_requests(tmp_path):
             dht.wait(timeout=5)
         except subprocess.TimeoutExpired:
             dht.kill()
+
+
+def test_swarm_health_report():
+    """Health monitor: coverage + per-server rows against a live 2-server
+    loopback swarm (ref health website / DHT models registry)."""
+    import torch
+
+    from bloombee_amd.cli.health import swarm_health
+    from bloombee_amd.net.dht import Dht
+    from bloombee_amd.server import Server
+
+    boot = Dht()
+    s1 = Server("llama-tiny", initial_peers=[boot.endpoint],
+                block_indices=(0, 2), device="cpu", seed=0,
+                kv_max_tokens=1 << 12, update_period=5.0)
+    s2 = Server("llama-tiny", initial_peers=[boot.endpoint],
+                block_indices=(2, 4), device="cpu", seed=0,
+                kv_max_tokens=1 << 12, update_period=5.0)
+    s1.run_in_background()
+    s2.run_in_background()
+    try:
+        h = swarm_health("llama-tiny", [boot.endpoint])
+        assert h["complete"], h
+        assert h["blocks"] == [1, 1, 1, 1]
+        assert len(h["servers"]) == 2
+        for s in h["servers"].values():
+            assert s["rtt_ms"] is not None and s["rtt_ms"] < 2000
+            assert s["cache_tokens_left"] is not None
+    finally:
+        s1.shutdown()
+        s2.shutdown()
+        boot.shutdown()
