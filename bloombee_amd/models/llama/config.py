@@ -37,6 +37,11 @@ LLAMA_PRESETS = {
         num_key_value_heads=2, intermediate_size=512, vocab_size=1024,
         max_position_embeddings=2048, rope_theta=10000.0,
     ),
+    "llama-tiny-8l": dict(
+        hidden_size=256, num_hidden_layers=8, num_attention_heads=4,
+        num_key_value_heads=2, intermediate_size=512, vocab_size=1024,
+        max_position_embeddings=2048, rope_theta=10000.0,
+    ),
     "llama-mini-gpu": dict(
         hidden_size=1024, num_hidden_layers=4, num_attention_heads=8,
         num_key_value_heads=2, intermediate_size=2816, vocab_size=32000,
