@@ -63,12 +63,17 @@ class _SpanSession:
                 raise RpcError(f"failed to open session on {span.peer_id}")
             return stream
 
+        from bloombee_amd.utils.fault_injection import maybe_fail
+        maybe_fail("open_inference")
         stream = run_coroutine(open_(), timeout)
         return cls(span, stream, sid, quiet, codec)
 
     def step(self, hidden: torch.Tensor, pos: int, step: int,
              timeout: float,
              prompts: Optional[torch.Tensor] = None) -> Optional[torch.Tensor]:
+        from bloombee_amd.utils.fault_injection import maybe_fail
+        maybe_fail("inference_item")  # no-op unless fault injection is armed
+
         async def go():
             payload = [hidden] if prompts is None else [hidden, prompts]
             await self.stream.send({"pos": pos, "step": step}, payload,
@@ -211,14 +216,20 @@ class InferenceSession:
         span receives its block-range slice with the step item."""
         if self._closed:
             raise RuntimeError("session is closed")
-        if self.batch_size is None:
+        first_open = self.batch_size is None
+        if first_open:
             self.batch_size = hidden.shape[0]
-            self._open_chain(self.batch_size, replay=False,
-                             allow_push=prompts is None)
         pos = self.position
         attempt = 0
         while True:
             try:
+                # chain (re)open lives INSIDE the retry loop: a failure while
+                # establishing span streams must re-route like a step failure
+                if first_open or not self.spans:
+                    self._open_chain(self.batch_size,
+                                     replay=not first_open and bool(self.history),
+                                     allow_push=prompts is None)
+                    first_open = False
                 out = self._chain_step(hidden, pos, prompts=prompts)
                 for s in self.spans:
                     self.manager.on_request_success(s.span.peer_id)
@@ -227,6 +238,7 @@ class InferenceSession:
                 attempt += 1
                 for s in self.spans:
                     s.close()
+                self.spans = []
                 self._ban_dead_spans()
                 max_r = self.config.max_retries
                 if max_r is not None and attempt > max_r:
@@ -236,7 +248,7 @@ class InferenceSession:
                                "%.1fs (attempt %d)", e, delay, attempt)
                 time.sleep(delay)
                 self.manager.update()
-                self._open_chain(self.batch_size, replay=True)
+                # reopen happens at the top of the loop (also under retry)
         self.history.append((pos, hidden, prompts))
         self.position = pos + hidden.shape[1]
         self.step_count += 1

@@ -269,6 +269,8 @@ class RpcClient:
                    tensors: Optional[List[torch.Tensor]] = None,
                    codec: str = "raw", timeout: Optional[float] = 30.0,
                    ) -> Tuple[dict, List[torch.Tensor]]:
+        from bloombee_amd.utils.fault_injection import maybe_fail
+        maybe_fail(method)  # no-op unless BBAMD_FAULT_RPC_DROP is set
         conn = await self._ensure()
         cid = next(self._ids)
         fut: asyncio.Future = asyncio.get_event_loop().create_future()

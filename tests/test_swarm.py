@@ -354,3 +354,25 @@ def test_swarm_mt_compressed_wire_matches_local(swarm):
     expect = _local_tokens(prompt, 6)
     assert torch.equal(out[:, 7:], expect)
     model.remote.manager.shutdown()
+
+
+def test_generate_exact_under_injected_rpc_faults(swarm):
+    """With 20% of unary RPCs failing (deterministic injection), the retry
+    + re-route machinery must still produce the exact greedy output."""
+    from bloombee_amd.utils import fault_injection as fi
+
+    boot, _ = swarm
+    model = _make_model(boot, use_server_to_server=False,
+                        min_backoff=0.05, max_retries=None)
+    gen = torch.Generator().manual_seed(5)
+    prompt = torch.randint(0, 1000, (2, 7), generator=gen)
+    fi.configure(0.3, seed=7, max_faults=6)
+    try:
+        out = model.generate(prompt, max_new_tokens=6)
+    finally:
+        injected = fi.injected
+        fi.configure(0.0)
+    expect = _local_tokens(prompt, 6)
+    assert torch.equal(out[:, 7:], expect)
+    assert injected == 6, injected
+    model.remote.manager.shutdown()
