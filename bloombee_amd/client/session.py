@@ -236,10 +236,10 @@ class InferenceSession:
                 break
             except (RpcError, OSError, TimeoutError, ConnectionError) as e:
                 attempt += 1
+                self._ban_dead_spans()  # probe BEFORE dropping the spans
                 for s in self.spans:
                     s.close()
                 self.spans = []
-                self._ban_dead_spans()
                 max_r = self.config.max_retries
                 if max_r is not None and attempt > max_r:
                     raise
