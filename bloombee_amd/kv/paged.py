@@ -312,7 +312,8 @@ class SessionHandle:
 
     # -- host offload (session multiplexing) ------------------------------
     def swap_out(self, to_disk: bool = False,
-                 disk_dir: Optional[str] = None) -> None:
+                 disk_dir: Optional[str] = None,
+                 compress: bool = False) -> None:
         """Offload this session's KV to pinned host buffers and release its
         device pages (reference micro-batch KV offload: GPU working slots +
         CPU snapshots on dedicated streams, memory_cache_manager.py:944-1371).
@@ -323,6 +324,10 @@ class SessionHandle:
         RAM (the reference's third KV tier, TorchMixedDevice GPU/CPU/disk
         partition, pytorch_backend.py:1207-1237; here whole-session
         granularity under the same page-table-rewrite contract).
+
+        compress: 4-bit group-quantize the snapshot (Policy.compress_cache,
+        ref flexgen compression.py group quant) — ~4x smaller host/disk
+        footprint, lossy within the quant tolerance.
         """
         if getattr(self, "_swapped", None) is not None:
             return
@@ -351,6 +356,18 @@ class SessionHandle:
                 snap.append(per_layer)
         if on_gpu:
             stream.synchronize()
+        if compress:
+            from bloombee_amd import ops as _ops
+            comp = []
+            for per_layer in snap:
+                cl = []
+                for kh, vh in per_layer:
+                    kq = _ops.quant4_pack(kh.float().reshape(-1, 64)) + (kh.shape, kh.dtype)
+                    vq = (None if vh is None else
+                          _ops.quant4_pack(vh.float().reshape(-1, 64)) + (vh.shape, vh.dtype))
+                    cl.append((kq, vq))
+                comp.append(cl)
+            snap = ("q4", comp)
         if to_disk:
             import tempfile
             fd, path = tempfile.mkstemp(
@@ -378,6 +395,18 @@ class SessionHandle:
             path = snap[1]
             snap = torch.load(path, map_location="cpu", weights_only=False)
             os.unlink(path)
+        if isinstance(snap, tuple) and snap[0] == "q4":
+            from bloombee_amd import ops as _ops
+
+            def _deq(q):
+                if q is None:
+                    return None
+                packed, scale, zero, shape, dtype = q
+                return _ops.quant4_unpack(packed, scale, zero,
+                                          dtype=dtype).reshape(shape)
+
+            snap = [[(_deq(kq), _deq(vq)) for kq, vq in per_layer]
+                    for per_layer in snap[1]]
         cache = self.cache
         on_gpu = cache.device.type == "cuda"
         stream = torch.cuda.Stream(cache.device) if on_gpu else None

@@ -53,6 +53,8 @@ class StackBackend:
         if offload_policy is not None and offload_policy.attn_sparsity < 1.0:
             from bloombee_amd import ops as _ops
             _ops.set_attn_sparsity(offload_policy.attn_sparsity)
+        self.compress_swap = bool(offload_policy is not None
+                                  and offload_policy.compress_cache)
         self.kv_pool: PagedKVCache = self.stack.make_kv(kv_max_tokens)
         self.is_last_block = (end == config.num_hidden_layers)
         self.pruner = None
@@ -100,7 +102,9 @@ class StackBackend:
                 key=lambda kv: kv[1].last_activity)
         for sid, st in candidates[:1]:
             logger.info("KV pressure: swapping out idle session %s", sid[:8])
-            self.pool.submit(st.handle.swap_out, PRIORITY_TRAIN).result()
+            self.pool.submit(
+                lambda h=st.handle: h.swap_out(compress=self.compress_swap),
+                PRIORITY_TRAIN).result()
 
     def reap_idle_sessions(self, max_idle_s: float = 600.0) -> int:
         """Close sessions idle past max_idle_s (ref handler session GC —

@@ -200,3 +200,35 @@ def test_swap_out_to_disk_roundtrip(tmp_path):
         assert torch.equal(k, before[l][0])
         assert torch.equal(v, before[l][1])
     h.close()
+
+
+def test_swap_out_compressed_roundtrip(tmp_path):
+    """compress_cache tier: 4-bit quantized swap snapshots restore within the
+    group-quant tolerance (ref flexgen cache compression)."""
+    import torch
+
+    from bloombee_amd.kv.paged import PagedKVCache
+
+    cache = PagedKVCache(num_layers=1, num_kv_heads=2, head_dim=32,
+                         max_tokens=1 << 10, device="cpu",
+                         dtype=torch.float32)
+    h = cache.allocate(1, 128)
+    h.extend(48)
+    cache.k_pages(0).uniform_(-1, 1)
+    cache.v_pages(0).uniform_(-1, 1)
+    pages = torch.tensor(h.seqs[0].pages)
+    k0 = cache.k_pages(0)[pages].clone()
+    v0 = cache.v_pages(0)[pages].clone()
+    h.swap_out(to_disk=True, disk_dir=str(tmp_path), compress=True)
+    # compressed file must be much smaller than the raw snapshot
+    f = next(p for p in tmp_path.iterdir() if p.suffix == ".pt")
+    raw_bytes = (k0.numel() + v0.numel()) * 4
+    assert f.stat().st_size < raw_bytes * 0.5, (f.stat().st_size, raw_bytes)
+    cache.k_pages(0).zero_()
+    cache.v_pages(0).zero_()
+    h.swap_in()
+    k1 = cache.k_pages(0)[torch.tensor(h.seqs[0].pages)]
+    v1 = cache.v_pages(0)[torch.tensor(h.seqs[0].pages)]
+    assert (k1 - k0).abs().max() < 0.07  # 4-bit over [-1,1]: step ~ 2/15
+    assert (v1 - v0).abs().max() < 0.07
+    h.close()
