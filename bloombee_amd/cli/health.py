@@ -64,21 +64,37 @@ def main():
     ap.add_argument("--initial-peers", nargs="+", required=True)
     ap.add_argument("--num-blocks", type=int, default=None)
     ap.add_argument("--no-ping", action="store_true")
+    ap.add_argument("--json", action="store_true", help="machine-readable")
+    ap.add_argument("--watch", type=float, default=0.0, metavar="SECONDS",
+                    help="refresh every N seconds until interrupted")
     args = ap.parse_args()
 
     peers = []
     for p in args.initial_peers:
         host, _, port = p.rpartition(":")
         peers.append((host or "127.0.0.1", int(port)))
-    h = swarm_health(args.model, peers, args.num_blocks, ping=not args.no_ping)
-    print(f"model: {h['model']}  complete: {h['complete']}")
-    cov = h["blocks"]
-    print("block coverage:", " ".join(str(c) for c in cov),
-          f"({sum(1 for c in cov if c == 0)} uncovered)")
-    for peer, s in sorted(h["servers"].items()):
-        print(f"  {peer[:12]:12s} blocks {s['blocks']:>7s} {s['state']:8s} "
-              f"rps {s['throughput']:<8.3g} rtt {s['rtt_ms']} ms "
-              f"kv_left {s['cache_tokens_left']} @ {s['host']}")
+    import json as _json
+    import time as _time
+
+    def report():
+        h = swarm_health(args.model, peers, args.num_blocks,
+                         ping=not args.no_ping)
+        if args.json:
+            print(_json.dumps(h))
+            return
+        print(f"model: {h['model']}  complete: {h['complete']}")
+        cov = h["blocks"]
+        print("block coverage:", " ".join(str(c) for c in cov),
+              f"({sum(1 for c in cov if c == 0)} uncovered)")
+        for peer, s in sorted(h["servers"].items()):
+            print(f"  {peer[:12]:12s} blocks {s['blocks']:>7s} {s['state']:8s} "
+                  f"rps {s['throughput']:<8.3g} rtt {s['rtt_ms']} ms "
+                  f"kv_left {s['cache_tokens_left']} @ {s['host']}")
+
+    report()
+    while args.watch > 0:
+        _time.sleep(args.watch)
+        report()
 
 
 if __name__ == "__main__":
