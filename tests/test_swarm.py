@@ -376,3 +376,26 @@ def test_generate_exact_under_injected_rpc_faults(swarm):
     assert torch.equal(out[:, 7:], expect)
     assert injected == 6, injected
     model.remote.manager.shutdown()
+
+
+def test_concurrent_clients_exact(swarm):
+    """4 clients decoding different prompts concurrently against the same
+    2-server swarm must each match their single-process result (task-pool
+    interleaving + per-session KV isolation)."""
+    from concurrent.futures import ThreadPoolExecutor
+
+    boot, _ = swarm
+
+    def one(seed):
+        model = _make_model(boot)
+        gen = torch.Generator().manual_seed(100 + seed)
+        prompt = torch.randint(0, 1000, (1, 5 + seed), generator=gen)
+        out = model.generate(prompt, max_new_tokens=5)
+        model.remote.manager.shutdown()
+        return prompt, out
+
+    with ThreadPoolExecutor(4) as ex:
+        results = list(ex.map(one, range(4)))
+    for seed, (prompt, out) in enumerate(results):
+        expect = _local_tokens(prompt, 5)
+        assert torch.equal(out[:, prompt.shape[1]:], expect), seed
