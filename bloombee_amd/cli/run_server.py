@@ -57,6 +57,8 @@ def main():
                     help="4-bit group-quantize the host weight tier")
     ap.add_argument("--attn-sparsity", type=float, default=1.0,
                     help="top-k sparse decode attention fraction (1.0 = dense)")
+    ap.add_argument("--adapters", nargs="*", default=[], metavar="NAME=DIR",
+                    help="preload LoRA adapters: per-block subdirs block{i}/")
     args = ap.parse_args()
 
     if args.max_chunk_tokens is not None:
@@ -90,6 +92,7 @@ def main():
         update_period=args.update_period, checkpoint_dir=args.checkpoint_dir,
         throughput=throughput, session_max_idle=args.session_max_idle,
         offload_policy=_offload_policy(args),
+        adapters=dict(a.split("=", 1) for a in args.adapters) or None,
     )
     server.run()
 

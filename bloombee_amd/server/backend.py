@@ -139,7 +139,8 @@ class StackBackend:
                        position_ids: Optional[torch.Tensor] = None,
                        tree_mask: Optional[torch.Tensor] = None,
                        speculative: bool = False,
-                       batch_offset: Optional[int] = None) -> torch.Tensor:
+                       batch_offset: Optional[int] = None,
+                       adapter: Optional[str] = None) -> torch.Tensor:
         """One decode/prefill step for an open session.
 
         start_pos: absolute position of hidden[:, 0]. If the session has
@@ -154,8 +155,13 @@ class StackBackend:
         handle = state.handle
 
         def run():
+            # adapter selection must live on THIS worker thread: the
+            # handler's contextvar does not cross the task-pool boundary
+            from bloombee_amd.utils.peft import using_adapter
             from bloombee_amd.utils.trace import trace_range
-            with trace_range(f"infer[{self.start}:{self.end}] pos={start_pos}"):
+            with using_adapter(adapter), \
+                    trace_range(f"infer[{self.start}:{self.end}] "
+                                f"pos={start_pos}"):
                 return _compute()
 
         def _compute():

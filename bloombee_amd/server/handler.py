@@ -174,11 +174,9 @@ class ConnectionHandler:
                 if rest:
                     prompts = rest[0]
                 def _step():
-                    from bloombee_amd.utils.peft import using_adapter
-                    with using_adapter(adapter):
-                        return self.backend.inference_step(
-                            sid, hidden, pos, prompts, position_ids, tree_mask,
-                            speculative=spec)
+                    return self.backend.inference_step(
+                        sid, hidden, pos, prompts, position_ids, tree_mask,
+                        speculative=spec, adapter=adapter)
 
                 mbinfo = item_meta.get("mb")
                 if mbinfo is not None and not spec:
@@ -189,7 +187,8 @@ class ConnectionHandler:
                     off = int(mbinfo["offset"])
                     part = await loop.run_in_executor(
                         None, lambda: self.backend.inference_step(
-                            sid, hidden, pos, prompts, batch_offset=off))
+                            sid, hidden, pos, prompts, batch_offset=off,
+                            adapter=adapter))
                     part_cpu = part.cpu()
                     if push_to is not None:
                         asyncio.ensure_future(self._push_downstream(
@@ -224,7 +223,7 @@ class ConnectionHandler:
                             part = await loop.run_in_executor(
                                 None, lambda j=j: self.backend.inference_step(
                                     sid, hidden[j:j + mbs], pos, prompts,
-                                    batch_offset=j))
+                                    batch_offset=j, adapter=adapter))
                             part_cpu = part.cpu()
                             outs.append(part_cpu)
                             push_tasks.append(asyncio.ensure_future(

@@ -49,6 +49,7 @@ class Server:
         throughput: float = 1.0,
         model_name: Optional[str] = None,
         offload_policy=None,
+        adapters: Optional[dict] = None,
     ):
         self.config = model if isinstance(model, ModelConfig) else resolve_config(model)
         self.model_name = model_name or (model if isinstance(model, str)
@@ -79,6 +80,24 @@ class Server:
                                     kv_max_tokens=kv_max_tokens,
                                     checkpoint_dir=checkpoint_dir,
                                     offload_policy=offload_policy)
+        if adapters:
+            # preload LoRA adapters onto this range's blocks (ref
+            # run_server --adapters + utils/peft.py download-per-block;
+            # offline layout: {dir}/block{i}/ with *.lora_{a,b}.npy)
+            import os as _os
+
+            from bloombee_amd.utils.peft import (add_adapter_to_block,
+                                                 load_adapter)
+            for name, root in adapters.items():
+                n_loaded = 0
+                for i, blk in enumerate(self.backend.stack.blocks):
+                    bdir = _os.path.join(root, f"block{block_indices[0] + i}")
+                    if _os.path.isdir(bdir):
+                        add_adapter_to_block(blk, name, load_adapter(
+                            bdir, dtype=self.config.dtype))
+                        n_loaded += 1
+                logger.info("adapter %r: %d/%d blocks", name, n_loaded,
+                            len(self.backend.stack.blocks))
         self.rpc = RpcServer(host, port)
         self.handler = ConnectionHandler(self.backend, self.rpc)
         from bloombee_amd.server.reachability import attach_reachability

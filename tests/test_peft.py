@@ -49,3 +49,69 @@ def test_adapter_save_load_roundtrip(tmp_path):
     assert set(back) == set(sets)
     assert torch.allclose(back["o_w"].b.float(), sets["o_w"].b.float(), atol=1e-2)
     assert estimate_adapter_memory(stack.blocks[0], 4) > 0
+
+
+def test_server_adapter_preload_swarm(tmp_path):
+    """Adapters preloaded at server start (run_server --adapters layout):
+    a client selecting the adapter gets different-but-deterministic output;
+    without selection the base output is unchanged (exact)."""
+    import torch
+
+    from bloombee_amd.client import ClientConfig
+    from bloombee_amd.engine import BlockStack, LocalEngine
+    from bloombee_amd.models.auto import AutoDistributedModelForCausalLM
+    from bloombee_amd.models.base import resolve_config
+    from bloombee_amd.net.dht import Dht
+    from bloombee_amd.server import Server
+
+    cfg_m = resolve_config("llama-tiny")
+    # author an adapter per block, saved in the per-block layout
+    stack = BlockStack(cfg_m, 0, 4, device="cpu", seed=0)
+    for i, blk in enumerate(stack.blocks):
+        sets = create_lora_adapter(blk, rank=2, seed=40 + i)
+        for s in sets.values():
+            s.b.normal_(0, 0.05)  # nonzero so the delta actually bites
+        save_adapter(sets, str(tmp_path / f"block{i}"))
+
+    boot = Dht()
+    srv = {"adapters": {"demo": str(tmp_path)}}
+    s1 = Server("llama-tiny", initial_peers=[boot.endpoint],
+                block_indices=(0, 2), device="cpu", seed=0,
+                kv_max_tokens=1 << 14, update_period=5.0, **srv)
+    s2 = Server("llama-tiny", initial_peers=[boot.endpoint],
+                block_indices=(2, 4), device="cpu", seed=0,
+                kv_max_tokens=1 << 14, update_period=5.0, **srv)
+    s1.run_in_background()
+    s2.run_in_background()
+    try:
+        gen = torch.Generator().manual_seed(3)
+        prompt = torch.randint(0, 1000, (1, 6), generator=gen)
+
+        base_cfg = ClientConfig(initial_peers=[boot.endpoint])
+        base = AutoDistributedModelForCausalLM.from_pretrained(
+            "llama-tiny", client_config=base_cfg, seed=0)
+        out_base = base.generate(prompt, max_new_tokens=5)
+        base.remote.manager.shutdown()
+
+        eng = LocalEngine("llama-tiny", device="cpu", seed=0,
+                          kv_max_tokens=1 << 14)
+        kv = eng.kv_pool.allocate(1, 64)
+        toks = [eng.prefill(prompt, kv)]
+        for _ in range(4):
+            toks.append(eng.decode_step(toks[-1], kv))
+        kv.close()
+        assert torch.equal(out_base[:, 6:], torch.stack(toks, 1))
+
+        ad_cfg = ClientConfig(initial_peers=[boot.endpoint],
+                              active_adapter="demo")
+        ad = AutoDistributedModelForCausalLM.from_pretrained(
+            "llama-tiny", client_config=ad_cfg, seed=0)
+        out_ad = ad.generate(prompt, max_new_tokens=5)
+        out_ad2 = ad.generate(prompt, max_new_tokens=5)
+        assert torch.equal(out_ad, out_ad2)  # deterministic
+        assert not torch.equal(out_ad, out_base)  # adapter changes the dist
+        ad.remote.manager.shutdown()
+    finally:
+        s1.shutdown()
+        s2.shutdown()
+        boot.shutdown()
