@@ -26,11 +26,13 @@ MAX_TOKENS_IN_BATCH = 1024  # client sub-batch split (ref sequential_autograd.py
 
 def _call_forward(span: RemoteSpanInfo, hidden: torch.Tensor,
                   timeout: float,
-                  prompts: Optional[torch.Tensor] = None) -> torch.Tensor:
+                  prompts: Optional[torch.Tensor] = None,
+                  adapter: Optional[str] = None) -> torch.Tensor:
     client = get_client(span.server_info.host, span.server_info.port)
     payload = [hidden] if prompts is None else [hidden, prompts]
     meta, tensors = run_coroutine(
-        client.call("rpc_forward", {}, payload, timeout=timeout),
+        client.call("rpc_forward", {"adapter": adapter}, payload,
+                    timeout=timeout),
         timeout + 5)
     return tensors[0]
 
@@ -38,12 +40,14 @@ def _call_forward(span: RemoteSpanInfo, hidden: torch.Tensor,
 def _call_backward(span: RemoteSpanInfo, hidden_in: torch.Tensor,
                    grad_out: torch.Tensor, timeout: float,
                    prompts: Optional[torch.Tensor] = None,
+                   adapter: Optional[str] = None,
                    ) -> Tuple[torch.Tensor, Optional[torch.Tensor]]:
     client = get_client(span.server_info.host, span.server_info.port)
     payload = ([hidden_in, grad_out] if prompts is None
                else [hidden_in, grad_out, prompts])
     meta, tensors = run_coroutine(
-        client.call("rpc_backward", {}, payload, timeout=timeout),
+        client.call("rpc_backward", {"adapter": adapter}, payload,
+                    timeout=timeout),
         timeout + 5)
     return tensors[0], (tensors[1] if len(tensors) > 1 else None)
 
@@ -68,7 +72,9 @@ def sequential_forward(manager: RemoteSequenceManager, hidden: torch.Tensor,
         try:
             p = (prompts[span.start:span.end].detach()
                  if prompts is not None else None)
-            result = _call_forward(span, out, cfg.request_timeout, prompts=p)
+            result = _call_forward(span, out, cfg.request_timeout, prompts=p,
+                                   adapter=getattr(cfg, "active_adapter",
+                                                   None))
             manager.on_request_success(span.peer_id)
             saved.append((span, out))
             out = result
@@ -107,7 +113,10 @@ def sequential_backward(manager: RemoteSequenceManager, grad_out: torch.Tensor,
                 p = (prompts[span.start:span.end].detach()
                      if prompts is not None else None)
                 grad, gp = _call_backward(span, span_input, grad,
-                                          cfg.request_timeout, prompts=p)
+                                          cfg.request_timeout, prompts=p,
+                                          adapter=getattr(cfg,
+                                                          "active_adapter",
+                                                          None))
                 if gp is not None and grad_prompts is not None:
                     grad_prompts[span.start:span.end] += gp.to(grad_prompts.dtype)
                 manager.on_request_success(span.peer_id)

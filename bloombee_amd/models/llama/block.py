@@ -163,8 +163,19 @@ class LlamaBlock(torch.nn.Module):
             return (xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + cfg.rms_norm_eps)
                     ).to(x.dtype) * w
 
+        def lin(x_, w, name):
+            # training path honors active LoRA adapters too (the inference
+            # path routes through self._lin; keep both consistent)
+            y = F.linear(x_, w)
+            ld = getattr(self, "lora_delta", None)
+            if ld is not None:
+                d = ld(name, x_)
+                if d is not None:
+                    y = y + d
+            return y
+
         x = rms(hidden, self.input_norm_w)
-        qkv = F.linear(x, self.qkv_w).view(B, T, Hq + 2 * Hkv, D).permute(0, 2, 1, 3)
+        qkv = lin(x, self.qkv_w, "qkv_w").view(B, T, Hq + 2 * Hkv, D).permute(0, 2, 1, 3)
         q, k, v = qkv.split([Hq, Hkv, Hkv], dim=1)
         cos, sin = self.rope.get(hidden.device)
         pos = torch.arange(start_pos, start_pos + T, device=hidden.device)
@@ -185,7 +196,7 @@ class LlamaBlock(torch.nn.Module):
         p = torch.softmax(scores, dim=-1)
         attn = torch.matmul(p, v.float()).to(hidden.dtype)
         attn = attn.permute(0, 2, 1, 3).reshape(B, T, Hq * D)
-        h2 = hidden + F.linear(attn, self.o_w)
+        h2 = hidden + lin(attn, self.o_w, "o_w")
         y = rms(h2, self.post_norm_w)
         gu = F.linear(y, self.gate_up_w)
         g, u = gu.split([self.I, self.I], dim=-1)

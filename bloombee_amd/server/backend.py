@@ -261,33 +261,37 @@ class StackBackend:
                          PRIORITY_INFERENCE).result()
 
     def forward(self, hidden: torch.Tensor,
-                prompts: Optional[torch.Tensor] = None) -> torch.Tensor:
+                prompts: Optional[torch.Tensor] = None,
+                adapter: Optional[str] = None) -> torch.Tensor:
         """Training-path forward (no KV cache, full sequence). prompts:
         optional deep p-tune tensor (n_local_blocks, pre, H) added to the
         prompt positions at every block input (ref client/ptune.py deep)."""
 
         def run():
+            from bloombee_amd.utils.peft import using_adapter
             h = hidden.to(self.device).to(self.config.dtype)
             dp = prompts.to(self.device) if prompts is not None else None
-            with torch.no_grad():
+            with torch.no_grad(), using_adapter(adapter):
                 return self.stack.forward_train(h, deep_prompts=dp)
 
         return self.pool.submit(run, PRIORITY_TRAIN).result()
 
     def backward(self, hidden_in: torch.Tensor, grad_out: torch.Tensor,
                  prompts: Optional[torch.Tensor] = None,
+                 adapter: Optional[str] = None,
                  ) -> Tuple[torch.Tensor, Optional[torch.Tensor]]:
         """Re-forward + backward; returns (grad wrt span input, grad wrt deep
         prompts or None). Server weights are frozen — only input/prompt grads
         flow (ref backend.py:106-109, 427-462)."""
 
         def run():
+            from bloombee_amd.utils.peft import using_adapter
             h = hidden_in.to(self.device).to(self.config.dtype)
             h = h.detach().requires_grad_(True)
             dp = None
             if prompts is not None:
                 dp = prompts.to(self.device).detach().requires_grad_(True)
-            with torch.enable_grad():
+            with torch.enable_grad(), using_adapter(adapter):
                 out = self.stack.forward_train(h, deep_prompts=dp)
                 g = grad_out.to(out.device).to(out.dtype)
                 leaves = (h,) if dp is None else (h, dp)

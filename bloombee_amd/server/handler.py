@@ -85,8 +85,10 @@ class ConnectionHandler:
         # tensors: [hidden] or [hidden, deep_prompts(n_local_blocks, pre, H)]
         hidden = tensors[0]
         prompts = tensors[1] if len(tensors) > 1 else None
+        adapter = meta.get("adapter")
         out = await asyncio.get_event_loop().run_in_executor(
-            None, self.backend.forward, hidden, prompts)
+            None, lambda: self.backend.forward(hidden, prompts,
+                                               adapter=adapter))
         return {}, [out.cpu()]
 
     async def rpc_backward(self, meta, tensors):
@@ -94,8 +96,10 @@ class ConnectionHandler:
         # [grad_in] or [grad_in, grad_prompts]
         hidden_in, grad_out = tensors[0], tensors[1]
         prompts = tensors[2] if len(tensors) > 2 else None
+        adapter = meta.get("adapter")
         grad_in, grad_p = await asyncio.get_event_loop().run_in_executor(
-            None, self.backend.backward, hidden_in, grad_out, prompts)
+            None, lambda: self.backend.backward(hidden_in, grad_out, prompts,
+                                                adapter=adapter))
         outs = [grad_in.cpu()]
         if grad_p is not None:
             outs.append(grad_p.cpu())

@@ -115,3 +115,53 @@ def test_server_adapter_preload_swarm(tmp_path):
         s1.shutdown()
         s2.shutdown()
         boot.shutdown()
+
+
+def test_training_forward_respects_adapter(tmp_path):
+    """rpc_forward/backward run under the client's active adapter (the
+    training path, not just inference sessions)."""
+    import torch
+
+    from bloombee_amd.client import ClientConfig
+    from bloombee_amd.engine import BlockStack
+    from bloombee_amd.models.auto import AutoDistributedModelForCausalLM
+    from bloombee_amd.models.base import resolve_config
+    from bloombee_amd.net.dht import Dht
+    from bloombee_amd.server import Server
+
+    cfg_m = resolve_config("llama-tiny")
+    stack = BlockStack(cfg_m, 0, 4, device="cpu", seed=0)
+    for i, blk in enumerate(stack.blocks):
+        sets = create_lora_adapter(blk, rank=2, seed=50 + i)
+        for s in sets.values():
+            s.b.normal_(0, 0.05)
+        save_adapter(sets, str(tmp_path / f"block{i}"))
+
+    boot = Dht()
+    s1 = Server("llama-tiny", initial_peers=[boot.endpoint],
+                block_indices=(0, 4), device="cpu", seed=0,
+                kv_max_tokens=1 << 14, update_period=5.0,
+                adapters={"demo": str(tmp_path)})
+    s1.run_in_background()
+    try:
+        ids = torch.randint(0, 1000, (1, 5),
+                            generator=torch.Generator().manual_seed(2))
+        base = AutoDistributedModelForCausalLM.from_pretrained(
+            "llama-tiny",
+            client_config=ClientConfig(initial_peers=[boot.endpoint]), seed=0)
+        h_base = base.transformer.remote(base.transformer.embed(ids))
+        base.remote.manager.shutdown()
+
+        ad = AutoDistributedModelForCausalLM.from_pretrained(
+            "llama-tiny",
+            client_config=ClientConfig(initial_peers=[boot.endpoint],
+                                       active_adapter="demo"), seed=0)
+        h = ad.transformer.embed(ids).detach().requires_grad_(True)
+        out = ad.transformer.remote(h)
+        assert not torch.equal(out, h_base)  # adapter changes training fwd
+        out.float().square().mean().backward()
+        assert h.grad is not None and torch.isfinite(h.grad).all()
+        ad.remote.manager.shutdown()
+    finally:
+        s1.shutdown()
+        boot.shutdown()
