@@ -67,6 +67,7 @@ def test_server_adapter_preload_swarm(tmp_path):
     cfg_m = resolve_config("llama-tiny")
     # author an adapter per block, saved in the per-block layout
     stack = BlockStack(cfg_m, 0, 4, device="cpu", seed=0)
+    torch.manual_seed(41)
     for i, blk in enumerate(stack.blocks):
         sets = create_lora_adapter(blk, rank=2, seed=40 + i)
         for s in sets.values():
@@ -131,6 +132,7 @@ def test_training_forward_respects_adapter(tmp_path):
 
     cfg_m = resolve_config("llama-tiny")
     stack = BlockStack(cfg_m, 0, 4, device="cpu", seed=0)
+    torch.manual_seed(51)
     for i, blk in enumerate(stack.blocks):
         sets = create_lora_adapter(blk, rank=2, seed=50 + i)
         for s in sets.values():

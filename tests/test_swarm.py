@@ -273,6 +273,7 @@ def test_deep_ptune_grads_and_parity(swarm):
         "llama-tiny", client_config=cfg, seed=SEED,
         pre_seq_len=4, deep_ptune=True)
     with torch.no_grad():
+        torch.manual_seed(21)
         model.transformer.deep_prompts.normal_(0, 0.02)
     gen = torch.Generator().manual_seed(11)
     ids = torch.randint(0, 1000, (2, 6), generator=gen)
@@ -311,6 +312,7 @@ def test_deep_ptune_generation_uses_prompts(swarm):
         "llama-tiny", client_config=cfg, seed=SEED,
         pre_seq_len=4, deep_ptune=True)
     with torch.no_grad():
+        torch.manual_seed(22)
         model.transformer.deep_prompts.normal_(0, 0.5)
         model.transformer.prompt_embeds.normal_(0, 0.5)
     gen = torch.Generator().manual_seed(12)
@@ -334,6 +336,7 @@ def test_deep_ptune_microbatch_split_matches_small_batch(swarm):
         "llama-tiny", client_config=cfg, seed=SEED,
         pre_seq_len=4, deep_ptune=True)
     with torch.no_grad():
+        torch.manual_seed(23)
         model.transformer.deep_prompts.normal_(0, 0.3)
         model.transformer.prompt_embeds.normal_(0, 0.3)
     gen = torch.Generator().manual_seed(13)
