@@ -70,6 +70,17 @@ class DistributedLlamaModel(torch.nn.Module):
     def final_norm(self, hidden: torch.Tensor) -> torch.Tensor:
         return ops.rms_norm(hidden, self.norm_w, self.config.rms_norm_eps)
 
+    def trainable_parameters(self):
+        """The client-side fine-tune state (ref ptune.py: everything the
+        swarm trains lives on the client — shallow + deep prompts; server
+        blocks stay frozen). Feed to a torch optimizer directly."""
+        ps = []
+        if self.prompt_embeds is not None:
+            ps.append(self.prompt_embeds)
+        if self.deep_prompts is not None:
+            ps.append(self.deep_prompts)
+        return ps
+
     def forward(self, input_ids: torch.Tensor) -> torch.Tensor:
         """Training-path full forward (no KV): returns final hidden states."""
         h = self.embed(input_ids)
@@ -124,6 +135,9 @@ class DistributedLlamaForCausalLM(RemoteGenerationMixin, torch.nn.Module):
 
     def final_norm(self, hidden):
         return self.transformer.final_norm(hidden)
+
+    def trainable_parameters(self):
+        return self.transformer.trainable_parameters()
 
     def forward(self, input_ids: torch.Tensor) -> torch.Tensor:
         """Training-path logits (differentiable wrt prompts)."""

@@ -402,3 +402,30 @@ def test_concurrent_clients_exact(swarm):
     for seed, (prompt, out) in enumerate(results):
         expect = _local_tokens(prompt, 5)
         assert torch.equal(out[:, prompt.shape[1]:], expect), seed
+
+
+def test_deep_ptune_optimizer_step_reduces_loss(swarm):
+    """The p-tune trainer loop: optimizer over trainable_parameters()
+    (shallow + deep prompts) reduces a fixed-target loss over the swarm."""
+    boot, _ = swarm
+    cfg = ClientConfig(initial_peers=[boot.endpoint])
+    model = AutoDistributedModelForCausalLM.from_pretrained(
+        "llama-tiny", client_config=cfg, seed=SEED,
+        pre_seq_len=4, deep_ptune=True)
+    params = model.trainable_parameters()
+    assert len(params) == 2
+    opt = torch.optim.Adam(params, lr=5e-2)
+    gen = torch.Generator().manual_seed(31)
+    ids = torch.randint(0, 1000, (1, 6), generator=gen)
+    target = torch.randint(0, 1000, (1, 10), generator=gen)
+    losses = []
+    for _ in range(4):
+        logits = model(ids)
+        loss = torch.nn.functional.cross_entropy(
+            logits.float().view(-1, logits.shape[-1]), target.view(-1))
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
+        losses.append(float(loss))
+    assert losses[-1] < losses[0], losses
+    model.remote.manager.shutdown()
