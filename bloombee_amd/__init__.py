@@ -16,4 +16,5 @@ from bloombee_amd.models.auto import (  # noqa: F401
     AutoDistributedConfig,
     AutoDistributedModel,
     AutoDistributedModelForCausalLM,
+    AutoDistributedSpeculativeModel,
 )
