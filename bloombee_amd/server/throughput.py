@@ -20,7 +20,10 @@ from bloombee_amd.utils.logging import get_logger
 
 logger = get_logger(__name__)
 
-CACHE_FILE = os.path.expanduser("~/.cache/bloombee_amd/throughput_v1.json")
+CACHE_FILE = os.path.join(
+    os.environ.get("BBAMD_CACHE_DIR",
+                   os.path.expanduser("~/.cache/bloombee_amd")),
+    "throughput_v1.json")
 
 
 def measure_compute_rps(config: ModelConfig, device: str = "cpu",
