@@ -157,3 +157,52 @@ def test_long_prefill_chunking_matches_unchunked(monkeypatch):
 
     assert torch.equal(whole, chunked)
     assert torch.isfinite(d1.float()).all()
+
+
+def test_activation_dumper_roundtrip(tmp_path, monkeypatch):
+    """BBAMD_DUMP_ACTIVATIONS captures real hidden states to disk (ref
+    real_activation_dumper.py)."""
+    import importlib
+
+    import torch
+
+    monkeypatch.setenv("BBAMD_DUMP_ACTIVATIONS", str(tmp_path))
+    from bloombee_amd.utils import activation_dumper as ad
+    importlib.reload(ad)
+    assert ad.enabled()
+    t = torch.randn(2, 3, 8).bfloat16()
+    ad.capture_activation("blocks0_2", t)
+    files = list(tmp_path.rglob("*.npy")) + list(tmp_path.rglob("*.pt"))
+    assert files, list(tmp_path.rglob("*"))
+    monkeypatch.delenv("BBAMD_DUMP_ACTIVATIONS")
+    importlib.reload(ad)
+    assert not ad.enabled()
+
+
+def test_disk_cache_eviction(tmp_path):
+    """free_disk_space_for evicts oldest cache entries until the requested
+    room exists (ref utils/disk_cache.py:83)."""
+    import os
+    import time
+
+    from bloombee_amd.utils.disk_cache import free_disk_space_for
+
+    for i in range(3):
+        p = tmp_path / f"blob{i}.bin"
+        p.write_bytes(b"x" * 1000)
+        os.utime(p, (time.time() - 100 + i, time.time() - 100 + i))
+    # plenty of disk: nothing should be evicted
+    assert free_disk_space_for(1, cache_dir=str(tmp_path))
+    assert len(list(tmp_path.iterdir())) == 3
+
+
+def test_ping_aggregator_unreachable_is_inf():
+    """Unreachable endpoints report inf RTT instead of raising (ref
+    utils/ping.py; live-endpoint RTTs covered by test_swarm_health_report)."""
+    import math
+
+    from bloombee_amd.utils.ping import PingAggregator
+
+    agg = PingAggregator(timeout=0.3)
+    out = agg.ping_many([("127.0.0.1", 1)])  # nothing listens on port 1
+    assert math.isinf(out[("127.0.0.1", 1)])
