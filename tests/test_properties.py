@@ -139,3 +139,50 @@ def test_compute_spans_contiguity(ranges):
         # a span is the (first) maximal contiguous run of the peer's blocks
         assert (span.start, span.end) == claimed[peer]
         assert span.start < span.end
+
+
+# ---------------------------------------------------------------------------
+# Spec-tree accept: reorder_and_commit must compact exactly the kept rows,
+# in order, for every (region size, kept subset) — the subtlest KV op.
+# ---------------------------------------------------------------------------
+
+@settings(max_examples=80, deadline=None)
+@given(
+    prefix=st.integers(0, 40),
+    spec_n=st.integers(1, 30),
+    keep_mask=st.lists(st.booleans(), min_size=30, max_size=30),
+    seed=st.integers(0, 99),
+)
+def test_reorder_and_commit_compacts_kept_rows(prefix, spec_n, keep_mask,
+                                               seed):
+    cache = PagedKVCache(num_layers=2, num_kv_heads=1, head_dim=16,
+                         max_tokens=1 << 10, device="cpu",
+                         dtype=torch.float32)
+    h = cache.allocate(1, 128)
+    if prefix:
+        h.extend(prefix)  # committed region
+    h.extend(spec_n, speculative=True)
+    gen = torch.Generator().manual_seed(seed)
+    # stamp every speculative position with a recognizable value
+    P = cache.page_size
+    stamps = torch.randn(spec_n, 1, 16, generator=gen)
+    for j in range(spec_n):
+        pos = prefix + j
+        pg = h.seqs[0].pages[pos // P]
+        for l in range(2):
+            cache.k_pages(l)[pg, :, pos % P, :] = stamps[j] + l
+            cache.v_pages(l)[pg, :, :, pos % P] = stamps[j] + 10 + l
+
+    kept = [j for j in range(spec_n) if keep_mask[j]]
+    h.reorder_and_commit([kept])
+    assert h.seqs[0].l_acc == prefix + len(kept)
+    assert h.seqs[0].l_spec == prefix + len(kept)
+    for d, j in enumerate(kept):
+        pos = prefix + d
+        pg = h.seqs[0].pages[pos // P]
+        for l in range(2):
+            assert torch.equal(cache.k_pages(l)[pg, :, pos % P, :],
+                               stamps[j] + l), (d, j)
+            assert torch.equal(cache.v_pages(l)[pg, :, :, pos % P],
+                               stamps[j] + 10 + l), (d, j)
+    h.close()
