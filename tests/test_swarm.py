@@ -426,6 +426,6 @@ def test_deep_ptune_optimizer_step_reduces_loss(swarm):
         opt.zero_grad()
         loss.backward()
         opt.step()
-        losses.append(float(loss))
+        losses.append(float(loss.detach()))
     assert losses[-1] < losses[0], losses
     model.remote.manager.shutdown()
