@@ -171,6 +171,10 @@ class RpcServer:
             await self._server.wait_closed()
         for conn in list(self._conns):
             conn.writer.close()
+            try:
+                await asyncio.wait_for(conn.writer.wait_closed(), 1)
+            except Exception:
+                pass
 
     async def _on_conn(self, reader, writer):
         conn = _Conn(reader, writer)
