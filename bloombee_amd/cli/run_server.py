@@ -59,15 +59,31 @@ def main():
                     help="top-k sparse decode attention fraction (1.0 = dense)")
     ap.add_argument("--adapters", nargs="*", default=[], metavar="NAME=DIR",
                     help="preload LoRA adapters: per-block subdirs block{i}/")
+    ap.add_argument("--identity-path", default=None,
+                    help="persistent peer identity file (ref --identity_path)")
+    ap.add_argument("--max-batch-size", type=int, default=2048,
+                    help="max sequences per inference session")
+    ap.add_argument("--torch-dtype", default=None,
+                    choices=["bfloat16", "float32"],
+                    help="serve dtype (kernels are bf16; float32 = CPU path)")
+    ap.add_argument("--quant-type", default="none",
+                    choices=["none", "nf4"],
+                    help="nf4 maps to the 4-bit host weight tier "
+                         "(compress_weight); LLM.int8 is not supported")
     args = ap.parse_args()
 
     if args.max_chunk_tokens is not None:
         import os
         os.environ["BBAMD_MAX_CHUNK_TOKENS"] = str(args.max_chunk_tokens)
+    if args.quant_type == "nf4":
+        args.compress_weight = True
 
     import torch
 
     device = args.device or ("cuda:0" if torch.cuda.is_available() else "cpu")
+    if args.torch_dtype:
+        import os
+        os.environ["BBAMD_TORCH_DTYPE"] = args.torch_dtype
     block_indices = None
     if args.block_indices:
         a, b = args.block_indices.split(":")
@@ -93,6 +109,8 @@ def main():
         throughput=throughput, session_max_idle=args.session_max_idle,
         offload_policy=_offload_policy(args),
         adapters=dict(a.split("=", 1) for a in args.adapters) or None,
+        identity_path=args.identity_path,
+        max_batch_size=args.max_batch_size,
     )
     server.run()
 

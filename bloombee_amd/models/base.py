@@ -50,7 +50,10 @@ class ModelConfig:
 
     @property
     def dtype(self) -> torch.dtype:
-        return getattr(torch, self.torch_dtype)
+        # BBAMD_TORCH_DTYPE (run_server --torch-dtype) overrides the preset
+        import os
+        return getattr(torch, os.environ.get("BBAMD_TORCH_DTYPE",
+                                             self.torch_dtype))
 
     @classmethod
     def from_dict(cls, d: dict) -> "ModelConfig":

@@ -38,7 +38,8 @@ class StackBackend:
                  device: str = "cpu", seed: int = 0,
                  kv_max_tokens: int = 1 << 18,
                  checkpoint_dir: Optional[str] = None,
-                 offload_policy=None):
+                 offload_policy=None,
+                 max_batch_size: int = 2048):
         self.config = config
         self.start, self.end = start, end
         self.device = torch.device(device)
@@ -74,6 +75,7 @@ class StackBackend:
         import os as _os2
         self.max_chunk_tokens = int(_os2.environ.get("BBAMD_MAX_CHUNK_TOKENS",
                                                      "4096"))
+        self.max_batch_size = max_batch_size
         self.pool = TaskPool(name=f"worker[{start}:{end}]")
         self.sessions: Dict[str, SessionState] = {}
         self._lock = threading.Lock()
@@ -82,6 +84,11 @@ class StackBackend:
     def open_session(self, session_id: str, batch_size: int, max_length: int,
                      timeout: Optional[float] = 10.0) -> None:
         from bloombee_amd.kv.paged import AllocationFailed
+
+        if batch_size > self.max_batch_size:
+            raise ValueError(
+                f"batch {batch_size} > server max_batch_size "
+                f"{self.max_batch_size} (ref --max_batch_size)")
 
         try:
             handle = self.kv_pool.allocate(batch_size, max_length, timeout=0.5)

@@ -50,12 +50,29 @@ class Server:
         model_name: Optional[str] = None,
         offload_policy=None,
         adapters: Optional[dict] = None,
+        identity_path: Optional[str] = None,
+        max_batch_size: int = 2048,
     ):
         self.config = model if isinstance(model, ModelConfig) else resolve_config(model)
         self.model_name = model_name or (model if isinstance(model, str)
                                          else self.config.model_type)
         self.device = device
-        self.peer_id = uuid.uuid4().hex[:16]
+        # persistent identity (ref --identity_path): keeps routing history,
+        # bans and DHT records stable across restarts
+        if identity_path:
+            import os as _os
+            if _os.path.exists(identity_path):
+                with open(identity_path) as f:
+                    self.peer_id = f.read().strip()
+            else:
+                self.peer_id = uuid.uuid4().hex[:16]
+                _os.makedirs(_os.path.dirname(identity_path) or ".",
+                             exist_ok=True)
+                with open(identity_path, "w") as f:
+                    f.write(self.peer_id)
+        else:
+            self.peer_id = uuid.uuid4().hex[:16]
+        self.max_batch_size = max_batch_size
         self.update_period = update_period
         self.session_max_idle = session_max_idle
         self.expiration = expiration or max(60.0, 2 * update_period)
@@ -79,7 +96,8 @@ class Server:
                                     block_indices[1], device=device, seed=seed,
                                     kv_max_tokens=kv_max_tokens,
                                     checkpoint_dir=checkpoint_dir,
-                                    offload_policy=offload_policy)
+                                    offload_policy=offload_policy,
+                                    max_batch_size=max_batch_size)
         if adapters:
             # preload LoRA adapters onto this range's blocks (ref
             # run_server --adapters + utils/peft.py download-per-block;

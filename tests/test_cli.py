@@ -105,3 +105,33 @@ def test_swarm_health_report():
         s1.shutdown()
         s2.shutdown()
         boot.shutdown()
+
+
+def test_identity_persistence_and_batch_cap(tmp_path):
+    """--identity-path keeps the peer id stable across restarts; sessions
+    above --max-batch-size are rejected."""
+    import pytest
+    import torch
+
+    from bloombee_amd.net.dht import Dht
+    from bloombee_amd.server import Server
+
+    ident = tmp_path / "id"
+    boot = Dht()
+    s1 = Server("llama-tiny", initial_peers=[boot.endpoint],
+                block_indices=(0, 4), device="cpu", seed=0,
+                kv_max_tokens=1 << 12, identity_path=str(ident),
+                max_batch_size=4)
+    pid = s1.peer_id
+    with pytest.raises(ValueError, match="max_batch_size"):
+        s1.backend.open_session("big", batch_size=8, max_length=16)
+    s1.backend.open_session("ok", batch_size=4, max_length=16)
+    s1.backend.close_session("ok")
+    s1.shutdown()
+
+    s2 = Server("llama-tiny", initial_peers=[boot.endpoint],
+                block_indices=(0, 4), device="cpu", seed=0,
+                kv_max_tokens=1 << 12, identity_path=str(ident))
+    assert s2.peer_id == pid
+    s2.shutdown()
+    boot.shutdown()
