@@ -42,5 +42,9 @@ class AutoDistributedSpeculativeModel(_AutoModelBase):
     _attr = "speculative_cls"
 
 
+class AutoDistributedModelForSequenceClassification(_AutoModelBase):
+    _attr = "seq_cls_cls"
+
+
 def get_block_class(model_type: str):
     return get_family(model_type).block_cls

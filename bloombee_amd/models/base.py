@@ -82,6 +82,7 @@ class FamilyEntry:
     model_cls: Optional[type] = None       # client DistributedModel
     causal_lm_cls: Optional[type] = None   # client DistributedModelForCausalLM
     speculative_cls: Optional[type] = None # client ForSpeculativeGeneration
+    seq_cls_cls: Optional[type] = None     # client ForSequenceClassification
     presets: Dict[str, dict] = field(default_factory=dict)
 
 

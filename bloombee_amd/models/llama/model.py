@@ -153,3 +153,53 @@ class DistributedLlamaForCausalLM(RemoteGenerationMixin, torch.nn.Module):
         ccfg = client_config or ClientConfig(initial_peers=list(initial_peers))
         return cls(cfg, ccfg, model_name=name_or_path, seed=seed,
                    device=device, **kw)
+
+
+class DistributedLlamaForSequenceClassification(torch.nn.Module):
+    """Classification over the swarm (parity: reference
+    AutoDistributedModelForSequenceClassification — the SST-2 prompt-tuning
+    notebook's model class): frozen remote blocks + a TRAINABLE client-side
+    score head over the last token's hidden state; combine with
+    pre_seq_len/deep_ptune for prompt-tuned classification."""
+
+    def __init__(self, config: ModelConfig, client_config: ClientConfig,
+                 model_name: str, num_labels: int = 2, seed: int = 0,
+                 device: str = "cpu",
+                 manager: Optional[RemoteSequenceManager] = None,
+                 pre_seq_len: int = 0, deep_ptune: bool = False):
+        super().__init__()
+        self.config = config
+        self.num_labels = num_labels
+        self.transformer = DistributedLlamaModel(config, client_config,
+                                                 model_name, seed=seed,
+                                                 device=device,
+                                                 manager=manager,
+                                                 pre_seq_len=pre_seq_len,
+                                                 deep_ptune=deep_ptune)
+        gen = torch.Generator().manual_seed(seed + 7)
+        self.score = torch.nn.Parameter(
+            torch.randn(num_labels, config.hidden_size, generator=gen)
+            .mul_(0.02).float().to(device))
+
+    @property
+    def remote(self) -> RemoteSequential:
+        return self.transformer.remote
+
+    def trainable_parameters(self):
+        return self.transformer.trainable_parameters() + [self.score]
+
+    def forward(self, input_ids: torch.Tensor) -> torch.Tensor:
+        """-> (B, num_labels) logits from the LAST token's hidden state."""
+        h = self.transformer(input_ids)
+        return F.linear(h[:, -1].float(), self.score)
+
+    @classmethod
+    def from_pretrained(cls, name_or_path: str,
+                        config: Optional[ModelConfig] = None,
+                        client_config: Optional[ClientConfig] = None,
+                        initial_peers: Sequence[Tuple[str, int]] = (),
+                        seed: int = 0, device: str = "cpu", **kw):
+        cfg = config or resolve_config(name_or_path)
+        ccfg = client_config or ClientConfig(initial_peers=list(initial_peers))
+        return cls(cfg, ccfg, model_name=name_or_path, seed=seed,
+                   device=device, **kw)

@@ -429,3 +429,33 @@ def test_deep_ptune_optimizer_step_reduces_loss(swarm):
         losses.append(float(loss.detach()))
     assert losses[-1] < losses[0], losses
     model.remote.manager.shutdown()
+
+
+def test_sequence_classification_trains(swarm):
+    """AutoDistributedModelForSequenceClassification: prompt-tuned
+    classification head trains over the swarm (ref SST-2 notebook model)."""
+    from bloombee_amd.models.auto import \
+        AutoDistributedModelForSequenceClassification
+
+    boot, _ = swarm
+    cfg = ClientConfig(initial_peers=[boot.endpoint])
+    model = AutoDistributedModelForSequenceClassification.from_pretrained(
+        "llama-tiny", client_config=cfg, seed=SEED, num_labels=2,
+        pre_seq_len=4, deep_ptune=True)
+    params = model.trainable_parameters()
+    assert len(params) == 3  # prompts + deep prompts + score head
+    opt = torch.optim.Adam(params, lr=5e-2)
+    gen = torch.Generator().manual_seed(17)
+    ids = torch.randint(0, 1000, (4, 6), generator=gen)
+    labels = torch.tensor([0, 1, 0, 1])
+    losses = []
+    for _ in range(4):
+        logits = model(ids)
+        assert logits.shape == (4, 2)
+        loss = torch.nn.functional.cross_entropy(logits, labels)
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
+        losses.append(float(loss.detach()))
+    assert losses[-1] < losses[0], losses
+    model.remote.manager.shutdown()
