@@ -59,6 +59,10 @@ def main():
                     help="top-k sparse decode attention fraction (1.0 = dense)")
     ap.add_argument("--adapters", nargs="*", default=[], metavar="NAME=DIR",
                     help="preload LoRA adapters: per-block subdirs block{i}/")
+    ap.add_argument("--announce-host", default=None,
+                    help="advertise this address instead of the bind host "
+                         "(NAT, ref --announce_maddrs)")
+    ap.add_argument("--announce-port", type=int, default=None)
     ap.add_argument("--identity-path", default=None,
                     help="persistent peer identity file (ref --identity_path)")
     ap.add_argument("--max-batch-size", type=int, default=2048,
@@ -111,6 +115,8 @@ def main():
         adapters=dict(a.split("=", 1) for a in args.adapters) or None,
         identity_path=args.identity_path,
         max_batch_size=args.max_batch_size,
+        announce_host=args.announce_host,
+        announce_port=args.announce_port,
     )
     server.run()
 

@@ -52,6 +52,8 @@ class Server:
         adapters: Optional[dict] = None,
         identity_path: Optional[str] = None,
         max_batch_size: int = 2048,
+        announce_host: Optional[str] = None,
+        announce_port: Optional[int] = None,
     ):
         self.config = model if isinstance(model, ModelConfig) else resolve_config(model)
         self.model_name = model_name or (model if isinstance(model, str)
@@ -78,6 +80,10 @@ class Server:
         self.expiration = expiration or max(60.0, 2 * update_period)
         self.throughput = throughput
         self.host = host
+        # NAT / multi-homed hosts (ref --announce_maddrs): advertise a
+        # different address than the bind address
+        self.announce_host = announce_host
+        self.announce_port = announce_port
         self.dht = Dht(initial_peers=list(initial_peers or []), host=host,
                        port=dht_port)
 
@@ -131,7 +137,8 @@ class Server:
     def _server_info(self) -> ServerInfo:
         return ServerInfo(
             state=ServerState.ONLINE,
-            host=self.endpoint[0], port=self.endpoint[1],
+            host=self.announce_host or self.endpoint[0],
+            port=self.announce_port or self.endpoint[1],
             throughput=self.throughput,
             cache_tokens_left=self.backend.kv_pool.tokens_left,
             torch_dtype=self.config.torch_dtype,
