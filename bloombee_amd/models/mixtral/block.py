@@ -70,16 +70,22 @@ class MixtralBlock(torch.nn.Module):
         weights, experts = logits.topk(self.topk, dim=-1)
         weights = torch.softmax(weights, dim=-1).to(y.dtype)
         out = torch.zeros_like(flat)
-        for e in range(self.E):
+        # expert-parallel mode (parallel/expert.py): this rank holds only
+        # experts [e0, e1); partial sums cross ranks via one all-reduce
+        e0, e1 = getattr(self, "ep_range", (0, self.E))
+        for le, e in enumerate(range(e0, e1)):
             sel = (experts == e)
             rows = sel.any(dim=-1).nonzero(as_tuple=True)[0]
             if rows.numel() == 0:
                 continue
             xe = flat[rows]
-            h = ops.linear(ops.swiglu(ops.linear(xe, self.expert_gate_up_w[e])),
-                           self.expert_down_w[e])
+            h = ops.linear(ops.swiglu(ops.linear(xe, self.expert_gate_up_w[le])),
+                           self.expert_down_w[le])
             w = (weights * sel.to(weights.dtype)).sum(-1)[rows]
             out.index_add_(0, rows, h * w.unsqueeze(-1))
+        if getattr(self, "ep_enabled", False):
+            from bloombee_amd.parallel.expert import ep_all_reduce
+            ep_all_reduce(out, self.ep_group)
         return out.view(B, T, H)
 
     @torch.no_grad()
@@ -137,19 +143,23 @@ class MixtralBlock(torch.nn.Module):
         weights, experts = logits.topk(self.topk, dim=-1)
         weights = torch.softmax(weights, dim=-1).to(y.dtype)
         out = torch.zeros_like(flat)
-        for e in range(self.E):
+        e0, e1 = getattr(self, "ep_range", (0, self.E))
+        for le, e in enumerate(range(e0, e1)):
             sel = (experts == e)
             rows = sel.any(dim=-1).nonzero(as_tuple=True)[0]
             if rows.numel() == 0:
                 continue
             xe = flat[rows]
-            gu = torch.nn.functional.linear(xe, self.expert_gate_up_w[e])
+            gu = torch.nn.functional.linear(xe, self.expert_gate_up_w[le])
             g, u = gu.split([self.I, self.I], dim=-1)
             h = torch.nn.functional.linear(
                 torch.nn.functional.silu(g.float()).to(u.dtype) * u,
-                self.expert_down_w[e])
+                self.expert_down_w[le])
             w = (weights * sel.to(weights.dtype)).sum(-1)[rows]
             out = out.index_add(0, rows, h * w.unsqueeze(-1))
+        if getattr(self, "ep_enabled", False):
+            from bloombee_amd.parallel.expert import ep_all_reduce
+            ep_all_reduce(out, self.ep_group)
         return h2 + out.view(B, T, H)
 
     def forward(self, *args, **kw):

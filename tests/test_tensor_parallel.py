@@ -148,3 +148,56 @@ def test_pipeline_with_tp(world, tp):
     kv.close()
     expect = torch.stack(expect, 1)
     assert torch.equal(got[:, :2], expect[:, :2]), (got, expect)
+
+
+def _ep_worker(rank, world, port, q):
+    import torch
+    import torch.distributed as dist
+
+    from bloombee_amd.models.base import resolve_config
+    from bloombee_amd.models.mixtral.block import MixtralBlock
+    from bloombee_amd.parallel.expert import shard_experts
+
+    dist.init_process_group(
+        "gloo", init_method=f"tcp://127.0.0.1:{port}", rank=rank,
+        world_size=world)
+    try:
+        cfg = resolve_config("mixtral-tiny")
+        blk = MixtralBlock(cfg, 0).init_random(seed=3)
+        ref_blk = MixtralBlock(cfg, 0).init_random(seed=3)
+        shard_experts(blk, rank, world)  # default group
+        x = (torch.randn(2, 5, cfg.hidden_size,
+                         generator=torch.Generator().manual_seed(9)) * 0.1
+             ).to(cfg.dtype)
+        got = blk._moe(x)
+        want = ref_blk._moe(x)
+        ok = torch.equal(got, want)
+        if rank == 0:
+            q.put(("ok", bool(ok),
+                   float((got - want).abs().max())))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_expert_parallel_matches_local():
+    """MoE expert parallelism (parallel/expert.py, beyond the reference):
+    experts sharded over 2 gloo ranks + partial-sum all-reduce must equal
+    the single-process expert loop exactly (top-k=2 => <=2 contributions
+    per row, so the cross-rank sum is order-insensitive)."""
+    import multiprocessing as mp
+
+    port = _free_port()
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_ep_worker, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    try:
+        tag, ok, err = q.get(timeout=120)
+        assert tag == "ok" and ok, f"max err {err}"
+    finally:
+        for p in procs:
+            p.join(timeout=30)
+            if p.is_alive():
+                p.terminate()
