@@ -21,6 +21,15 @@ from bloombee_amd.utils.logging import get_logger
 
 logger = get_logger(__name__)
 
+# On Python < 3.11, concurrent.futures.TimeoutError and asyncio.TimeoutError
+# are NOT the builtin TimeoutError — a step timeout must still trigger
+# failover (found by fault injection on the s2s push path)
+import asyncio as _asyncio  # noqa: E402
+import concurrent.futures as _cf  # noqa: E402
+
+_RETRYABLE = (OSError, TimeoutError, ConnectionError,
+              _cf.TimeoutError, _asyncio.TimeoutError)
+
 MAX_TOKENS_IN_BATCH = 1024  # client sub-batch split (ref sequential_autograd.py:22)
 
 
@@ -80,7 +89,7 @@ def sequential_forward(manager: RemoteSequenceManager, hidden: torch.Tensor,
             out = result
             cur = span.end
             attempt = 0
-        except (RpcError, OSError, TimeoutError) as e:
+        except (RpcError, *_RETRYABLE) as e:
             attempt += 1
             manager.on_request_failure(span.peer_id)
             if cfg.max_retries is not None and attempt > cfg.max_retries:
@@ -121,7 +130,7 @@ def sequential_backward(manager: RemoteSequenceManager, grad_out: torch.Tensor,
                     grad_prompts[span.start:span.end] += gp.to(grad_prompts.dtype)
                 manager.on_request_success(span.peer_id)
                 break
-            except (RpcError, OSError, TimeoutError) as e:
+            except (RpcError, *_RETRYABLE) as e:
                 attempt += 1
                 manager.on_request_failure(span.peer_id)
                 if cfg.max_retries is not None and attempt > cfg.max_retries:

@@ -26,6 +26,15 @@ from bloombee_amd.utils.logging import get_logger
 
 logger = get_logger(__name__)
 
+# On Python < 3.11, concurrent.futures.TimeoutError and asyncio.TimeoutError
+# are NOT the builtin TimeoutError — a step timeout must still trigger
+# failover (found by fault injection on the s2s push path)
+import asyncio as _asyncio  # noqa: E402
+import concurrent.futures as _cf  # noqa: E402
+
+_RETRYABLE = (OSError, TimeoutError, ConnectionError,
+              _cf.TimeoutError, _asyncio.TimeoutError)
+
 
 class _SpanSession:
     """Client end of one span's rpc_inference stream."""
@@ -234,7 +243,7 @@ class InferenceSession:
                 for s in self.spans:
                     self.manager.on_request_success(s.span.peer_id)
                 break
-            except (RpcError, OSError, TimeoutError, ConnectionError) as e:
+            except (RpcError, *_RETRYABLE) as e:
                 attempt += 1
                 self._ban_dead_spans()  # probe BEFORE dropping the spans
                 for s in self.spans:

@@ -284,6 +284,12 @@ class ConnectionHandler:
                                ) -> None:
         import time as _time
 
+        from bloombee_amd.utils.fault_injection import maybe_fail
+        try:
+            maybe_fail("s2s_push")
+        except Exception as e:
+            logger.warning("s2s push suppressed by fault injection: %s", e)
+            return
         host, port, down_sid = push_to[0], int(push_to[1]), push_to[2]
         key = (host, port)
         if key not in self._peers:

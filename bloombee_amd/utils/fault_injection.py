@@ -22,14 +22,17 @@ class InjectedRpcFault(ConnectionError):
 
 
 _max_faults = None  # stop injecting after this many (None = unlimited)
+_methods = None     # restrict injection to these fault points (None = all)
 
 
 def configure(drop_p: float, seed: int = 0,
-              max_faults: "int | None" = None) -> None:
-    global _drop_p, _seed, _counter, injected, _max_faults
+              max_faults: "int | None" = None,
+              methods=None) -> None:
+    global _drop_p, _seed, _counter, injected, _max_faults, _methods
     with _lock:
         _drop_p, _seed, _counter, injected = float(drop_p), seed, 0, 0
         _max_faults = max_faults
+        _methods = tuple(methods) if methods else None
 
 
 def maybe_fail(method: str) -> None:
@@ -37,6 +40,8 @@ def maybe_fail(method: str) -> None:
     probability (never for rpc_info so liveness probes stay truthful)."""
     global _counter, injected
     if _drop_p <= 0.0 or method == "rpc_info":
+        return
+    if _methods is not None and method not in _methods:
         return
     with _lock:
         if _max_faults is not None and injected >= _max_faults:

@@ -459,3 +459,44 @@ def test_sequence_classification_trains(swarm):
         losses.append(float(loss.detach()))
     assert losses[-1] < losses[0], losses
     model.remote.manager.shutdown()
+
+
+def test_push_drop_recovers_exact():
+    """Dropped server-to-server pushes (injected) must be absorbed by the
+    client's step-timeout rebuild + history replay, still emitting the
+    exact greedy tokens."""
+    from bloombee_amd.utils import fault_injection as fi
+
+    boot = Dht()
+    s1 = Server(MODEL, initial_peers=[boot.endpoint], block_indices=(0, 2),
+                device="cpu", seed=SEED, kv_max_tokens=1 << 14,
+                update_period=2.0)
+    s2 = Server(MODEL, initial_peers=[boot.endpoint], block_indices=(2, 4),
+                device="cpu", seed=SEED, kv_max_tokens=1 << 14,
+                update_period=2.0)
+    s1.run_in_background()
+    s2.run_in_background()
+    try:
+        cfg = ClientConfig(initial_peers=[boot.endpoint],
+                           use_server_to_server=True,
+                           step_timeout=3.0, min_backoff=0.1,
+                           max_retries=None)
+        model = AutoDistributedModelForCausalLM.from_pretrained(
+            MODEL, client_config=cfg, seed=SEED)
+        gen = torch.Generator().manual_seed(5)
+        prompt = torch.randint(0, 1000, (2, 7), generator=gen)
+        fi.configure(0.9, seed=3, max_faults=2,
+                     methods=("s2s_push",))
+        try:
+            out = model.generate(prompt, max_new_tokens=6)
+        finally:
+            injected = fi.injected
+            fi.configure(0.0)
+        expect = _local_tokens(prompt, 6)
+        assert torch.equal(out[:, 7:], expect)
+        assert injected >= 1
+        model.remote.manager.shutdown()
+    finally:
+        s1.shutdown()
+        s2.shutdown()
+        boot.shutdown()
