@@ -273,36 +273,90 @@ class InferenceSession:
         stream span by span)."""
         if self.batch_size is None:
             self.batch_size = hidden.shape[0]
-            self._open_chain(self.batch_size, replay=False)
         t = self.config.step_timeout
-        out = hidden
-        keep = None
-        for s in self.spans:
-            async def go(s=s, out=out):
-                await s.stream.send({"pos": self.position, "spec": True,
-                                     "step": self.step_count, "tree": tree},
-                                    [out, position_ids.int(), tree_mask])
-                item = await s.stream.recv()
-                if item is None:
-                    raise RpcError(f"stream closed by {s.span.peer_id}")
-                return item[0].get("keep"), item[1][0]
+        attempt = 0
+        while True:
+            try:
+                if not self.spans:
+                    # spec rides the client stream span by span
+                    self.allow_push = False
+                    self._open_chain(self.batch_size,
+                                     replay=bool(self.history))
+                out = hidden
+                keep = None
+                from bloombee_amd.utils.fault_injection import maybe_fail
+                for s in self.spans:
+                    maybe_fail("spec_item")
 
-            keep, out = run_coroutine(go(), t)
-        self.step_count += 1
-        return out, keep
+                    async def go(s=s, out=out):
+                        await s.stream.send(
+                            {"pos": self.position, "spec": True,
+                             "step": self.step_count, "tree": tree},
+                            [out, position_ids.int(), tree_mask])
+                        item = await s.stream.recv()
+                        if item is None:
+                            raise RpcError(f"stream closed by {s.span.peer_id}")
+                        return item[0].get("keep"), item[1][0]
+
+                    keep, out = run_coroutine(go(), t)
+                # kept for spec_commit's history entry (failover replay)
+                self._last_spec_hidden = hidden
+                self.step_count += 1
+                return out, keep
+            except (RpcError, *_RETRYABLE) as e:
+                # rebuilt sessions hold only committed tokens (replayed from
+                # history), so the WHOLE tree step can simply be redone
+                attempt += 1
+                self._ban_dead_spans()
+                for s in self.spans:
+                    s.close()
+                self.spans = []
+                max_r = self.config.max_retries
+                if max_r is not None and attempt > max_r:
+                    raise
+                delay = self.manager.get_retry_delay(attempt)
+                logger.warning("spec step failed (%s); rebuilding route in "
+                               "%.1fs", e, delay)
+                time.sleep(delay)
+                self.manager.update()
 
     def spec_commit(self, keep) -> int:
         """Accept tree nodes `keep[b]` (ascending linear indices); every span
-        compacts + commits its KV. Returns the accepted length."""
+        compacts + commits its KV. Returns the accepted length.
+
+        The accepted rows' INPUT hiddens enter the session history first, so
+        a failover (here, or in any later step) replays them as ordinary
+        committed tokens — positions are sequential after compaction, which
+        is exactly the replayed-step semantics."""
+        lasth = getattr(self, "_last_spec_hidden", None)
+        if lasth is not None and keep and len(keep[0]) > 0:
+            kept = lasth[:, list(keep[0])]
+            self.history.append((self.position, kept, None))
         t = self.config.step_timeout
-        for s in self.spans:
+        remaining = list(self.spans)
+        while remaining:
+            s = remaining[0]
+
             async def go(s=s):
                 await s.stream.send({"spec_commit": [list(k) for k in keep]})
                 item = await s.stream.recv()
                 if item is None or not item[0].get("committed"):
                     raise RpcError(f"spec commit failed on {s.span.peer_id}")
 
-            run_coroutine(go(), t)
+            try:
+                run_coroutine(go(), t)
+                remaining.pop(0)
+            except (RpcError, *_RETRYABLE) as e:
+                # rebuild: the replay (which now includes the kept rows)
+                # leaves EVERY new span consistent — nothing left to commit
+                logger.warning("spec commit failed (%s); rebuilding chain", e)
+                self._ban_dead_spans()
+                for sp in self.spans:
+                    sp.close()
+                self.spans = []
+                self.manager.update()
+                self._open_chain(self.batch_size, replay=True)
+                remaining = []
         n = len(keep[0])
         self.position += n
         return n

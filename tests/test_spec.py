@@ -245,3 +245,53 @@ def test_pruner_checkpoint_roundtrip(tmp_path):
     k1 = p.keep_indices(hidden, [1, 2, 3, 4, 5], [-1, 0, 0, 1, 1])
     k2 = p2.keep_indices(hidden, [1, 2, 3, 4, 5], [-1, 0, 0, 1, 1])
     assert k1 == k2
+
+
+def test_speculative_survives_injected_faults():
+    """Spec sessions must failover too: with injected stream faults the
+    tree rounds retry on rebuilt chains (history now covers committed
+    spec tokens) and the greedy invariant still holds exactly."""
+    from bloombee_amd.client import ClientConfig
+    from bloombee_amd.engine import LocalEngine
+    from bloombee_amd.models.llama.speculative import \
+        DistributedLlamaForSpeculativeGeneration
+    from bloombee_amd.net.dht import Dht
+    from bloombee_amd.server import Server
+    from bloombee_amd.utils import fault_injection as fi
+
+    boot = Dht()
+    s1 = Server("llama-tiny", initial_peers=[boot.endpoint],
+                block_indices=(0, 2), device="cpu", seed=0,
+                kv_max_tokens=1 << 14, update_period=2.0)
+    s2 = Server("llama-tiny", initial_peers=[boot.endpoint],
+                block_indices=(2, 4), device="cpu", seed=0,
+                kv_max_tokens=1 << 14, update_period=2.0)
+    s1.run_in_background()
+    s2.run_in_background()
+    try:
+        cfg = ClientConfig(initial_peers=[boot.endpoint], min_backoff=0.1,
+                           max_retries=None, step_timeout=5.0)
+        model = DistributedLlamaForSpeculativeGeneration.from_pretrained(
+            "llama-tiny", client_config=cfg, seed=0)
+        draft = LocalEngine("llama-tiny", device="cpu", seed=7,
+                            kv_max_tokens=1 << 13)
+        model.set_drafter(draft, node_budget=6, max_depth=3)
+        gen = torch.Generator().manual_seed(5)
+        prompt = torch.randint(0, 1000, (1, 7), generator=gen)
+        fi.configure(0.25, seed=11, max_faults=3,
+                     methods=("inference_item", "open_inference", "spec_item"))
+        try:
+            out = model.generate_speculative(prompt, max_new_tokens=8)
+        finally:
+            injected = fi.injected
+            fi.configure(0.0)
+        eng = LocalEngine("llama-tiny", device="cpu", seed=0,
+                          kv_max_tokens=1 << 14)
+        expect = eng.generate_greedy(prompt, 8)
+        assert torch.equal(out, expect), (out, expect, injected)
+        assert injected >= 1
+        model.remote.manager.shutdown()
+    finally:
+        s1.shutdown()
+        s2.shutdown()
+        boot.shutdown()
