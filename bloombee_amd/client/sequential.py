@@ -13,7 +13,9 @@ from typing import List, Optional, Tuple
 import torch
 
 from bloombee_amd.client.config import ClientConfig
-from bloombee_amd.client.routing import RemoteSequenceManager, RemoteSpanInfo
+from bloombee_amd.client.routing import (MissingBlocksError,
+                                         RemoteSequenceManager,
+                                         RemoteSpanInfo)
 from bloombee_amd.client.session import InferenceSession
 from bloombee_amd.client.worker import get_client, run_coroutine
 from bloombee_amd.net.rpc import RpcError
@@ -76,9 +78,12 @@ def sequential_forward(manager: RemoteSequenceManager, hidden: torch.Tensor,
     out = hidden
     attempt = 0
     while cur < end:
-        route = manager.make_sequence(cur, end)
-        span = route[0]
+        span = None
         try:
+            # transient MissingBlocks (e.g. every server briefly banned)
+            # must wait out the ban backoff, not abort the pass
+            route = manager.make_sequence(cur, end)
+            span = route[0]
             p = (prompts[span.start:span.end].detach()
                  if prompts is not None else None)
             result = _call_forward(span, out, cfg.request_timeout, prompts=p,
@@ -89,15 +94,15 @@ def sequential_forward(manager: RemoteSequenceManager, hidden: torch.Tensor,
             out = result
             cur = span.end
             attempt = 0
-        except (RpcError, *_RETRYABLE) as e:
+        except (RpcError, MissingBlocksError, *_RETRYABLE) as e:
             attempt += 1
-            manager.on_request_failure(span.peer_id)
+            if span is not None:
+                manager.on_request_failure(span.peer_id)
             if cfg.max_retries is not None and attempt > cfg.max_retries:
                 raise
             delay = manager.get_retry_delay(attempt)
-            logger.warning("forward span [%d:%d) on %s failed (%s); retrying "
-                           "in %.1fs", span.start, span.end, span.peer_id, e,
-                           delay)
+            logger.warning("forward pass at block %d failed (%s); retrying "
+                           "in %.1fs", cur, e, delay)
             time.sleep(delay)
             manager.update()
     return out, saved
@@ -130,7 +135,7 @@ def sequential_backward(manager: RemoteSequenceManager, grad_out: torch.Tensor,
                     grad_prompts[span.start:span.end] += gp.to(grad_prompts.dtype)
                 manager.on_request_success(span.peer_id)
                 break
-            except (RpcError, *_RETRYABLE) as e:
+            except (RpcError, MissingBlocksError, *_RETRYABLE) as e:
                 attempt += 1
                 manager.on_request_failure(span.peer_id)
                 if cfg.max_retries is not None and attempt > cfg.max_retries:
@@ -138,8 +143,13 @@ def sequential_backward(manager: RemoteSequenceManager, grad_out: torch.Tensor,
                 time.sleep(manager.get_retry_delay(attempt))
                 manager.update()
                 # find a replacement covering exactly this span's range; the
-                # replacement recomputes the forward internally in rpc_backward
-                route = manager.make_sequence(span.start, span.end)
+                # replacement recomputes the forward internally in
+                # rpc_backward. A still-banned route raises MissingBlocks
+                # again -> caught above on the next spin.
+                try:
+                    route = manager.make_sequence(span.start, span.end)
+                except MissingBlocksError:
+                    continue
                 if len(route) == 1:
                     span = route[0]
                 else:

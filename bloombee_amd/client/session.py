@@ -19,7 +19,9 @@ from typing import List, Optional, Tuple
 import torch
 
 from bloombee_amd.client.config import ClientConfig
-from bloombee_amd.client.routing import RemoteSequenceManager, RemoteSpanInfo
+from bloombee_amd.client.routing import (MissingBlocksError,
+                                         RemoteSequenceManager,
+                                         RemoteSpanInfo)
 from bloombee_amd.client.worker import get_client, run_coroutine
 from bloombee_amd.net.rpc import RpcError, Stream
 from bloombee_amd.utils.logging import get_logger
@@ -243,7 +245,7 @@ class InferenceSession:
                 for s in self.spans:
                     self.manager.on_request_success(s.span.peer_id)
                 break
-            except (RpcError, *_RETRYABLE) as e:
+            except (RpcError, MissingBlocksError, *_RETRYABLE) as e:
                 attempt += 1
                 self._ban_dead_spans()  # probe BEFORE dropping the spans
                 for s in self.spans:
@@ -303,7 +305,7 @@ class InferenceSession:
                 self._last_spec_hidden = hidden
                 self.step_count += 1
                 return out, keep
-            except (RpcError, *_RETRYABLE) as e:
+            except (RpcError, MissingBlocksError, *_RETRYABLE) as e:
                 # rebuilt sessions hold only committed tokens (replayed from
                 # history), so the WHOLE tree step can simply be redone
                 attempt += 1
@@ -346,7 +348,7 @@ class InferenceSession:
             try:
                 run_coroutine(go(), t)
                 remaining.pop(0)
-            except (RpcError, *_RETRYABLE) as e:
+            except (RpcError, MissingBlocksError, *_RETRYABLE) as e:
                 # rebuild: the replay (which now includes the kept rows)
                 # leaves EVERY new span consistent — nothing left to commit
                 logger.warning("spec commit failed (%s); rebuilding chain", e)

@@ -500,3 +500,30 @@ def test_push_drop_recovers_exact():
         s1.shutdown()
         s2.shutdown()
         boot.shutdown()
+
+
+def test_training_grads_exact_under_faults(swarm):
+    """rpc_forward/rpc_backward retries (injected faults) must not change
+    the computed gradients at all."""
+    from bloombee_amd.utils import fault_injection as fi
+
+    boot, _ = swarm
+    model = _make_model(boot, min_backoff=0.05, max_retries=None,
+                        ban_timeout=0.2)
+    gen = torch.Generator().manual_seed(41)
+    ids = torch.randint(0, 1000, (2, 5), generator=gen)
+
+    h1 = model.transformer.embed(ids).detach().requires_grad_(True)
+    model.transformer.remote(h1).float().square().mean().backward()
+
+    h2 = model.transformer.embed(ids).detach().requires_grad_(True)
+    fi.configure(0.8, seed=13, max_faults=3,
+                 methods=("rpc_forward", "rpc_backward"))
+    try:
+        model.transformer.remote(h2).float().square().mean().backward()
+    finally:
+        injected = fi.injected
+        fi.configure(0.0)
+    assert injected >= 1
+    assert torch.equal(h1.grad, h2.grad)
+    model.remote.manager.shutdown()
