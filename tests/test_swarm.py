@@ -527,3 +527,26 @@ def test_training_grads_exact_under_faults(swarm):
     assert injected >= 1
     assert torch.equal(h1.grad, h2.grad)
     model.remote.manager.shutdown()
+
+
+def test_microbatch_push_drop_recovers_exact(swarm):
+    """B>=8 prefill triggers the server-side micro-batch split with per-MB
+    downstream pushes; dropped MB pushes must still converge to the exact
+    greedy output (client-stream fallback + rebuild)."""
+    from bloombee_amd.utils import fault_injection as fi
+
+    boot, _ = swarm
+    model = _make_model(boot, step_timeout=3.0, min_backoff=0.1,
+                        max_retries=None, ban_timeout=0.2)
+    gen = torch.Generator().manual_seed(51)
+    prompt = torch.randint(0, 1000, (8, 6), generator=gen)
+    fi.configure(0.5, seed=29, max_faults=2, methods=("s2s_push",))
+    try:
+        out = model.generate(prompt, max_new_tokens=4)
+    finally:
+        injected = fi.injected
+        fi.configure(0.0)
+    expect = _local_tokens(prompt, 4)
+    assert torch.equal(out[:, 6:], expect)
+    assert injected >= 1
+    model.remote.manager.shutdown()
