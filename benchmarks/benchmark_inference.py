@@ -34,11 +34,15 @@ def main():
     ap.add_argument("--batch", type=int, default=1)
     ap.add_argument("--warmup", type=int, default=1)
     ap.add_argument("--seed", type=int, default=0)
+    ap.add_argument("--client-device", default="cpu",
+                    help="device for the client-side embed/LM-head (ref "
+                         "README: use cuda for 150k+ vocab families)")
     args = ap.parse_args()
 
     cfg = ClientConfig(initial_peers=[parse_endpoint(p) for p in args.initial_peers])
     model = AutoDistributedModelForCausalLM.from_pretrained(
-        args.model, client_config=cfg, seed=args.seed)
+        args.model, client_config=cfg, seed=args.seed,
+        device=args.client_device)
     V = model.config.vocab_size
     gen = torch.Generator().manual_seed(args.seed)
     prompt = torch.randint(0, V, (args.batch, args.prompt_len), generator=gen)
