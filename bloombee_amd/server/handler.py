@@ -196,7 +196,8 @@ class ConnectionHandler:
                     part_cpu = part.cpu()
                     if push_to is not None:
                         asyncio.ensure_future(self._push_downstream(
-                            push_to, pos, part_cpu, item_meta, mb=mbinfo))
+                            push_to, pos, part_cpu, item_meta, mb=mbinfo,
+                            codec=codec))
                     buf = mb_buffers.setdefault(pos, {})
                     buf[off] = part_cpu
                     got = sum(t.shape[0] for t in buf.values())
@@ -233,7 +234,8 @@ class ConnectionHandler:
                             push_tasks.append(asyncio.ensure_future(
                                 self._push_downstream(
                                     push_to, pos, part_cpu, item_meta,
-                                    mb={"offset": j, "total": B})))
+                                    mb={"offset": j, "total": B},
+                                    codec=codec)))
                         out_cpu = torch.cat(outs, dim=0)
                         for t_ in push_tasks:
                             await t_
@@ -258,7 +260,7 @@ class ConnectionHandler:
                 if push_to is not None:
                     with times.span("push"):
                         await self._push_downstream(push_to, pos, out_cpu,
-                                                    item_meta)
+                                                    item_meta, codec=codec)
                 if not quiet:
                     with times.span("reply"):
                         await stream.send(
@@ -281,7 +283,7 @@ class ConnectionHandler:
 
     async def _push_downstream(self, push_to, pos: int, hidden: torch.Tensor,
                                item_meta: dict, mb: Optional[dict] = None,
-                               ) -> None:
+                               codec: str = "raw") -> None:
         import time as _time
 
         from bloombee_amd.utils.fault_injection import maybe_fail
@@ -302,7 +304,7 @@ class ConnectionHandler:
                     "rpc_push", {"session_id": down_sid, "pos": pos,
                                  "step": item_meta.get("step"),
                                  "mb": mb}, [hidden],
-                    timeout=30)
+                    codec=codec, timeout=30)
         except Exception as e:  # noqa: BLE001 — client will fall back
             ok = False
             logger.warning("s2s push to %s failed: %s", push_to, e)
