@@ -126,6 +126,9 @@ ALL = {
 }
 
 if __name__ == "__main__":
+    if not torch.cuda.is_available():
+        sys.exit("bench_kernels measures the gfx950 HIP kernels — "
+                 "run on a GPU box (e.g. via gpurun)")
     which = sys.argv[1:] or list(ALL)
     for name in which:
         ALL[name]()
