@@ -73,7 +73,8 @@ def test_paged_kv_invariants(ops, batch):
 @given(
     shape=st.lists(st.integers(1, 9), min_size=1, max_size=3),
     dtype=st.sampled_from([torch.float32, torch.bfloat16, torch.int32]),
-    codec=st.sampled_from(["raw", "zlib", "bsplit+zlib"]),
+    codec=st.sampled_from(["raw", "zlib", "bsplit+zlib",
+                           "bsplit+zlibmt"]),
     seed=st.integers(0, 2 ** 16),
 )
 def test_wire_codec_roundtrip(shape, dtype, codec, seed):
