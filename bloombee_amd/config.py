@@ -45,8 +45,8 @@ class MicrobatchConfig:
     """Micro-batch pipelining policy (reference microbatch_config.py:27-123)."""
 
     enabled: bool = field(default_factory=lambda: _env("MICROBATCH", True))
-    micro_batch_size: int = field(default_factory=lambda: _env("MICRO_BATCH_SIZE", 8))
-    min_batch_to_split: int = field(default_factory=lambda: _env("MIN_BATCH_TO_SPLIT", 16))
+    micro_batch_size: int = field(default_factory=lambda: _env("MICRO_BATCH_SIZE", 4))
+    min_batch_to_split: int = field(default_factory=lambda: _env("MIN_BATCH_TO_SPLIT", 8))
 
 
 @dataclass

@@ -61,10 +61,14 @@ class AdaptivePushConcurrency:
 
 
 class ConnectionHandler:
-    MICROBATCH_MIN_BATCH = 8     # split threshold (ref should_split_batch)
-    MICROBATCH_SIZE = 4          # sequences per micro-batch (ref default 2-4)
-
     def __init__(self, backend: StackBackend, server: RpcServer):
+        # micro-batch policy from the typed config (BBAMD_MICROBATCH /
+        # _MICRO_BATCH_SIZE / _MIN_BATCH_TO_SPLIT — ref microbatch_config.py)
+        from bloombee_amd.config import get_config
+        mb = get_config().microbatch
+        self.MICROBATCH_ENABLED = mb.enabled
+        self.MICROBATCH_MIN_BATCH = mb.min_batch_to_split
+        self.MICROBATCH_SIZE = mb.micro_batch_size
         self.backend = backend
         self.rpc = server
         # (session_id, step) -> queued pushed inputs awaiting the local stream
@@ -212,7 +216,8 @@ class ConnectionHandler:
                                  "keep": None}, [out_cpu], codec=codec)
                     continue
                 B = hidden.shape[0]
-                can_split = (not spec and push_to is not None
+                can_split = (self.MICROBATCH_ENABLED and not spec
+                             and push_to is not None
                              and B >= self.MICROBATCH_MIN_BATCH
                              and B % self.MICROBATCH_SIZE == 0)
                 if can_split:
