@@ -61,18 +61,31 @@ def _raw_bytes(t: torch.Tensor) -> bytes:
     return t.numpy().tobytes()
 
 
+def _wire_cfg():
+    from bloombee_amd.config import get_config
+    return get_config().compression
+
+
 def _deflate(raw: bytes, bsplit: bool) -> bytes:
+    level = _wire_cfg().level
     if _native is not None:
         t = torch.frombuffer(bytearray(raw), dtype=torch.uint8)
-        return bytes(_native.wire_deflate(t, bsplit, 1).numpy().tobytes())
+        return bytes(_native.wire_deflate(t, bsplit, level).numpy().tobytes())
     if bsplit:
         hi, lo = raw[1::2], raw[0::2]
-        return zlib.compress(hi + lo, level=1)
-    return zlib.compress(raw, level=1)
+        return zlib.compress(hi + lo, level=level)
+    return zlib.compress(raw, level=level)
 
 
 import os as _os
 _MT_THREADS = min(32, (_os.cpu_count() or 8) * 2)
+
+
+def _worth_it(raw: bytes, payload: bytes) -> bool:
+    # min-size / min-gain gates (ref lossless_transport.py:167-186)
+    cfg = _wire_cfg()
+    return (len(raw) >= cfg.min_size_bytes
+            and len(payload) <= len(raw) * (1.0 - cfg.min_gain))
 
 
 def serialize_tensor(t: torch.Tensor, codec: str = "raw") -> Tuple[dict, bytes]:
@@ -82,17 +95,17 @@ def serialize_tensor(t: torch.Tensor, codec: str = "raw") -> Tuple[dict, bytes]:
         # multithreaded chunked DEFLATE (ops/hip/wire.h) — the native codec
         # that makes compression viable at multi-MB activation payloads
         r = torch.frombuffer(bytearray(raw), dtype=torch.uint8)
-        payload = bytes(_native.wire_deflate_mt(r, True, 1,
-                                                _MT_THREADS).numpy().tobytes())
-        if len(payload) >= len(raw):
+        payload = bytes(_native.wire_deflate_mt(
+            r, True, _wire_cfg().level, _MT_THREADS).numpy().tobytes())
+        if not _worth_it(raw, payload):
             codec, payload = "raw", raw
     elif codec == "zlib":
         payload = _deflate(raw, False)
-        if len(payload) >= len(raw):  # min-gain gate (ref :167-186)
+        if not _worth_it(raw, payload):
             codec, payload = "raw", raw
     elif codec == "bsplit+zlib" and t.dtype in _TWO_BYTE:
         payload = _deflate(raw, True)
-        if len(payload) >= len(raw):
+        if not _worth_it(raw, payload):
             codec, payload = "raw", raw
     else:
         codec, payload = "raw", raw

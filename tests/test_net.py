@@ -31,7 +31,8 @@ def test_frame_roundtrip(codec, dtype):
 
 
 def test_bsplit_compresses_smooth_bf16():
-    t = torch.linspace(0, 1, 4096).to(torch.bfloat16)
+    # above the BBAMD_WIRE_MIN_SIZE gate (64 KiB): small tensors ride raw
+    t = torch.linspace(0, 1, 65536).to(torch.bfloat16)
     raw = pack_frame({}, [t], "raw")
     comp = pack_frame({}, [t], "bsplit+zlib")
     assert len(comp) < len(raw) * 0.8
@@ -177,3 +178,13 @@ def test_wire_mt_python_fallback_decode():
     finally:
         T._native = saved
     assert torch.equal(back, t)
+
+
+def test_wire_min_size_gate_sends_small_raw():
+    """Tensors below BBAMD_WIRE_MIN_SIZE skip compression entirely (ref
+    lossless_transport min-size gate)."""
+    from bloombee_amd.net.tensors import serialize_tensor
+
+    t = torch.linspace(0, 1, 512).to(torch.bfloat16)  # 1 KiB << 64 KiB
+    head, _ = serialize_tensor(t, codec="bsplit+zlib")
+    assert head["codec"] == "raw"

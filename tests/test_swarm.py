@@ -37,6 +37,26 @@ def swarm():
     boot.shutdown()
 
 
+import contextlib
+
+
+@contextlib.contextmanager
+def _no_wire_gates():
+    """Drop the codec min-size/min-gain gates so tiny test tensors actually
+    exercise compression end-to-end."""
+    import dataclasses
+
+    from bloombee_amd import config as bconf
+    cfg0 = bconf.get_config()
+    comp = dataclasses.replace(cfg0.compression, min_size_bytes=1,
+                               min_gain=0.0)
+    bconf.set_config(dataclasses.replace(cfg0, compression=comp))
+    try:
+        yield
+    finally:
+        bconf.set_config(cfg0)
+
+
 def _local_tokens(prompt, new_tokens):
     eng = LocalEngine(MODEL, device="cpu", seed=SEED, kv_max_tokens=1 << 14)
     kv = eng.kv_pool.allocate(prompt.shape[0], 64)
@@ -177,7 +197,12 @@ def test_bloom_swarm_matches_local():
 
 def test_swarm_compressed_wire_matches_local(swarm):
     boot, _ = swarm
-    model = _make_model(boot, wire_codec="bsplit+zlib")
+    with _no_wire_gates():
+        _compressed_roundtrip(boot, "bsplit+zlib")
+
+
+def _compressed_roundtrip(boot, codec):
+    model = _make_model(boot, wire_codec=codec)
     gen = torch.Generator().manual_seed(5)
     prompt = torch.randint(0, 1000, (2, 7), generator=gen)
     out = model.generate(prompt, max_new_tokens=6)
@@ -350,13 +375,8 @@ def test_deep_ptune_microbatch_split_matches_small_batch(swarm):
 def test_swarm_mt_compressed_wire_matches_local(swarm):
     """The native multithreaded chunked codec is lossless end-to-end."""
     boot, _ = swarm
-    model = _make_model(boot, wire_codec="bsplit+zlibmt")
-    gen = torch.Generator().manual_seed(5)
-    prompt = torch.randint(0, 1000, (2, 7), generator=gen)
-    out = model.generate(prompt, max_new_tokens=6)
-    expect = _local_tokens(prompt, 6)
-    assert torch.equal(out[:, 7:], expect)
-    model.remote.manager.shutdown()
+    with _no_wire_gates():
+        _compressed_roundtrip(boot, "bsplit+zlibmt")
 
 
 def test_generate_exact_under_injected_rpc_faults(swarm):
