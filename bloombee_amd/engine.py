@@ -53,11 +53,18 @@ class BlockStack(torch.nn.Module):
         else:
             hkv = self.config.num_key_value_heads
             dims = self.config.head_dim
+        from bloombee_amd.config import get_config
+        P = get_config().kv.page_size
+        # the decode kernels walk KV in 32-position tiles and require page
+        # boundaries to align (attn_decode_mfma.hip: "P | 32")
+        if 32 % P != 0:
+            raise ValueError(f"BBAMD_KV_PAGE_SIZE must divide 32, got {P}")
         return PagedKVCache(
             num_layers=len(self.blocks),
             num_kv_heads=hkv,
             head_dim=dims,
             max_tokens=max_tokens,
+            page_size=P,
             device=self.device,
             dtype=self.config.dtype,
         )
