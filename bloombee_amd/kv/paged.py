@@ -356,6 +356,10 @@ class SessionHandle:
                 snap.append(per_layer)
         if on_gpu:
             stream.synchronize()
+        from bloombee_amd.utils.logging import debug_log, get_logger
+        debug_log("kv", get_logger(__name__),
+                  "swap_out handle=%d to_disk=%s compress=%s",
+                  self.handle_id, to_disk, compress)
         if compress:
             from bloombee_amd import ops as _ops
             comp = []

@@ -84,8 +84,14 @@ _MT_THREADS = min(32, (_os.cpu_count() or 8) * 2)
 def _worth_it(raw: bytes, payload: bytes) -> bool:
     # min-size / min-gain gates (ref lossless_transport.py:167-186)
     cfg = _wire_cfg()
-    return (len(raw) >= cfg.min_size_bytes
-            and len(payload) <= len(raw) * (1.0 - cfg.min_gain))
+    ok = (len(raw) >= cfg.min_size_bytes
+          and len(payload) <= len(raw) * (1.0 - cfg.min_gain))
+    if not ok:
+        from bloombee_amd.utils.logging import debug_log, get_logger
+        debug_log("compression", get_logger(__name__),
+                  "gate: %d raw vs %d compressed -> raw", len(raw),
+                  len(payload))
+    return ok
 
 
 def serialize_tensor(t: torch.Tensor, codec: str = "raw") -> Tuple[dict, bytes]:

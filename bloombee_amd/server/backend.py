@@ -224,6 +224,10 @@ class StackBackend:
             # kernel's continuation path)
             if (tm is None and pos is None and not speculative
                     and T > self.max_chunk_tokens):
+                from bloombee_amd.utils.logging import debug_log
+                debug_log("inference", logger,
+                          "chunking prefill T=%d by %d", T,
+                          self.max_chunk_tokens)
                 outs = []
                 for t0 in range(0, T, self.max_chunk_tokens):
                     t1 = min(T, t0 + self.max_chunk_tokens)

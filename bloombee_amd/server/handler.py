@@ -221,6 +221,10 @@ class ConnectionHandler:
                              and B >= self.MICROBATCH_MIN_BATCH
                              and B % self.MICROBATCH_SIZE == 0)
                 if can_split:
+                    from bloombee_amd.utils.logging import debug_log
+                    debug_log("microbatch", logger,
+                              "splitting B=%d into %d-row micro-batches", B,
+                              self.MICROBATCH_SIZE)
                     # reference micro-batch overlap (block_functions.py:
                     # 2055-2460): compute micro-batch j while j-1 is in
                     # flight downstream; the client reply carries the merged
