@@ -164,12 +164,27 @@ class StackBackend:
         def run():
             # adapter selection must live on THIS worker thread: the
             # handler's contextvar does not cross the task-pool boundary
+            from bloombee_amd.config import get_config
             from bloombee_amd.utils.peft import using_adapter
             from bloombee_amd.utils.trace import trace_range
             with using_adapter(adapter), \
                     trace_range(f"infer[{self.start}:{self.end}] "
                                 f"pos={start_pos}"):
-                return _compute()
+                if not get_config().step_profile:
+                    return _compute()
+                # BBAMD_STEP_PROFILE (ref BLOOMBEE_STEP_PROFILE,
+                # backend.py:59-60): wall-clock the step with device sync
+                if self.device.type == "cuda":
+                    torch.cuda.synchronize(self.device)
+                t0 = time.monotonic()
+                out_ = _compute()
+                if self.device.type == "cuda":
+                    torch.cuda.synchronize(self.device)
+                logger.info("[STEP_PROFILE] blocks[%d:%d] pos=%d T=%d "
+                            "%.3f ms", self.start, self.end, start_pos,
+                            hidden.shape[1],
+                            (time.monotonic() - t0) * 1e3)
+                return out_
 
         def _compute():
             if handle.is_swapped:
