@@ -206,3 +206,36 @@ def test_ping_aggregator_unreachable_is_inf():
     agg = PingAggregator(timeout=0.3)
     out = agg.ping_many([("127.0.0.1", 1)])  # nothing listens on port 1
     assert math.isinf(out[("127.0.0.1", 1)])
+
+
+def test_checkpoint_layout_roundtrip(tmp_path):
+    """North-star per-block .npy layout (BASELINE.json): save a stack's
+    blocks, reload into a fresh random-init server backend, bitwise-equal
+    weights and identical decode output."""
+    import torch
+
+    from bloombee_amd.engine import BlockStack, LocalEngine
+    from bloombee_amd.models.base import resolve_config
+    from bloombee_amd.server.backend import StackBackend
+    from bloombee_amd.server.from_pretrained import (SENTINEL,
+                                                     is_converted,
+                                                     save_block_weights)
+
+    cfg = resolve_config("llama-tiny")
+    src = BlockStack(cfg, 0, 4, device="cpu", seed=123)
+    ck = tmp_path / "llama-tiny-np"
+    for i, blk in enumerate(src.blocks):
+        save_block_weights(blk, str(ck), i)
+    (ck / SENTINEL).touch()
+    assert is_converted(str(ck))
+    assert any(p.suffix == ".npy" for p in ck.iterdir())
+
+    # a backend seeded DIFFERENTLY must come out identical after loading
+    be = StackBackend(cfg, 0, 4, device="cpu", seed=999,
+                      kv_max_tokens=1 << 12, checkpoint_dir=str(ck))
+    for b_src, b_dst in zip(src.blocks, be.stack.blocks):
+        for (n1, p1), (n2, p2) in zip(b_src.named_parameters(),
+                                      b_dst.named_parameters()):
+            assert n1 == n2
+            assert torch.equal(p1, p2), n1
+    be.shutdown()
