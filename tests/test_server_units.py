@@ -239,3 +239,19 @@ def test_checkpoint_layout_roundtrip(tmp_path):
             assert n1 == n2
             assert torch.equal(p1, p2), n1
     be.shutdown()
+
+
+def test_client_weights_roundtrip(tmp_path):
+    import torch
+
+    from bloombee_amd.server.from_pretrained import (load_client_weights,
+                                                     save_client_weights)
+
+    e = torch.randn(100, 16).bfloat16()
+    n = torch.randn(16).bfloat16()
+    h = torch.randn(100, 16).bfloat16()
+    save_client_weights(str(tmp_path), e, n, h)
+    back = load_client_weights(str(tmp_path))
+    assert torch.equal(back["embed"], e)
+    assert torch.equal(back["final_norm"], n)
+    assert torch.equal(back["lm_head"], h)
