@@ -187,3 +187,38 @@ def test_reorder_and_commit_compacts_kept_rows(prefix, spec_n, keep_mask,
             assert torch.equal(cache.v_pages(l)[pg, :, :, pos % P],
                                stamps[j] + 10 + l), (d, j)
     h.close()
+
+
+# ---------------------------------------------------------------------------
+# Frames: arbitrary msgpack-able metadata + several mixed-dtype tensors must
+# survive a frame roundtrip regardless of codec.
+# ---------------------------------------------------------------------------
+
+meta_strategy = st.recursive(
+    st.one_of(st.integers(-2**40, 2**40), st.text(max_size=12),
+              st.booleans(), st.none(),
+              st.floats(allow_nan=False, allow_infinity=False)),
+    lambda children: st.one_of(
+        st.lists(children, max_size=4),
+        st.dictionaries(st.text(max_size=8), children, max_size=4)),
+    max_leaves=12)
+
+
+@settings(max_examples=40, deadline=None)
+@given(meta=st.dictionaries(st.text(min_size=1, max_size=8), meta_strategy,
+                            max_size=4),
+       n_tensors=st.integers(0, 3), seed=st.integers(0, 999),
+       codec=st.sampled_from(["raw", "bsplit+zlib"]))
+def test_frame_roundtrip_meta_and_tensors(meta, n_tensors, seed, codec):
+    gen = torch.Generator().manual_seed(seed)
+    dts = [torch.float32, torch.bfloat16, torch.int64]
+    ts = [torch.randn(3, 5, generator=gen).to(dts[i % 3])
+          if dts[i % 3] != torch.int64 else
+          torch.randint(0, 9, (3, 5), generator=gen)
+          for i in range(n_tensors)]
+    blob = pack_frame(meta, ts, codec=codec)
+    m2, ts2 = unpack_frame(blob)
+    assert m2 == meta
+    assert len(ts2) == len(ts)
+    for a, b in zip(ts, ts2):
+        assert a.dtype == b.dtype and torch.equal(a, b)
