@@ -134,12 +134,14 @@ def test_wire_codec_native_python_interop():
     if T._native is None:
         import pytest
         pytest.skip("native codec not built")
-    t = (torch.randn(33, 17) * 0.1).bfloat16()
+    # above the min-size gate so compression actually runs
+    t = (torch.randn(330, 170) * 0.1).bfloat16()
     raw = T._raw_bytes(t)
 
     # native-compress -> python-decompress
     head, payload = T.serialize_tensor(t, codec="bsplit+zlib")
-    if head["codec"] == "bsplit+zlib":
+    assert head["codec"] == "bsplit+zlib"
+    if True:
         dec = _zlib.decompress(payload)
         n = len(dec) // 2
         hi, lo = dec[:n], dec[n:]
@@ -169,8 +171,9 @@ def test_wire_mt_python_fallback_decode():
     if T._native is None:
         import pytest
         pytest.skip("native codec not built")
-    t = (torch.randn(600, 33) * 0.2).bfloat16()
+    t = (torch.randn(600, 66) * 0.2).bfloat16()  # > min-size gate
     head, payload = T.serialize_tensor(t, codec="bsplit+zlibmt")
+    assert head["codec"] == "bsplit+zlibmt"  # gate must not have bypassed
     saved = T._native
     try:
         T._native = None
@@ -188,3 +191,26 @@ def test_wire_min_size_gate_sends_small_raw():
     t = torch.linspace(0, 1, 512).to(torch.bfloat16)  # 1 KiB << 64 KiB
     head, _ = serialize_tensor(t, codec="bsplit+zlib")
     assert head["codec"] == "raw"
+
+
+def test_dht_record_expiry():
+    """Expired announcements disappear from reads (ref DHT record TTL —
+    dead servers age out of routing within 2x update_period)."""
+    import time as _time
+
+    from bloombee_amd.data_structures import (ServerInfo,
+                                              get_remote_module_infos)
+    from bloombee_amd.net.dht import Dht
+
+    boot = Dht()
+    try:
+        info = ServerInfo(host="h", port=1).to_dict()
+        boot.store("m.0", "peerA", info, _time.time() + 0.3)
+        boot.store("m.0", "peerB", info, _time.time() + 30)
+        out = get_remote_module_infos(boot, ["m.0"])[0]
+        assert set(out.servers) == {"peerA", "peerB"}
+        _time.sleep(0.4)
+        out = get_remote_module_infos(boot, ["m.0"])[0]
+        assert set(out.servers) == {"peerB"}
+    finally:
+        boot.shutdown()
