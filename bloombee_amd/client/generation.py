@@ -25,7 +25,8 @@ class RemoteGenerationMixin:
     def generate(self, input_ids: torch.Tensor, max_new_tokens: int = 20,
                  do_sample: bool = False, temperature: float = 1.0,
                  top_k: Optional[int] = None, top_p: Optional[float] = None,
-                 session=None) -> torch.Tensor:
+                 session=None,
+                 generator: Optional[torch.Generator] = None) -> torch.Tensor:
         B, T = input_ids.shape
         # p-tuning at inference (ref remote_generation + ptune): trained
         # prompt embeds are prepended once at prefill; deep per-block prompts
@@ -48,13 +49,15 @@ class RemoteGenerationMixin:
             hidden = session.step(
                 hidden, prompts=dp.detach() if dp is not None else None)
             logits = self.lm_head(self.final_norm(hidden[:, -1:]))[:, -1]
-            next_tok = self._pick(logits, do_sample, temperature, top_k, top_p)
+            next_tok = self._pick(logits, do_sample, temperature, top_k,
+                                  top_p, generator)
             out_tokens.append(next_tok.view(B, 1))
             for _ in range(max_new_tokens - 1):
                 hidden = self.embed(next_tok.view(B, 1))
                 hidden = session.step(hidden)
                 logits = self.lm_head(self.final_norm(hidden[:, -1:]))[:, -1]
-                next_tok = self._pick(logits, do_sample, temperature, top_k, top_p)
+                next_tok = self._pick(logits, do_sample, temperature, top_k,
+                                  top_p, generator)
                 out_tokens.append(next_tok.view(B, 1))
         finally:
             if own_session:
@@ -63,7 +66,8 @@ class RemoteGenerationMixin:
 
     @staticmethod
     def _pick(logits: torch.Tensor, do_sample: bool, temperature: float,
-              top_k: Optional[int], top_p: Optional[float]) -> torch.Tensor:
+              top_k: Optional[int], top_p: Optional[float],
+              generator: Optional[torch.Generator] = None) -> torch.Tensor:
         if not do_sample:
             return logits.argmax(dim=-1)
         logits = logits.float() / max(temperature, 1e-6)
@@ -79,4 +83,5 @@ class RemoteGenerationMixin:
             mask = torch.zeros_like(logits, dtype=torch.bool).scatter_(
                 -1, idx, kill)
             logits = logits.masked_fill(mask, float("-inf"))
-        return torch.multinomial(logits.softmax(-1), 1).squeeze(-1)
+        return torch.multinomial(logits.softmax(-1), 1,
+                                 generator=generator).squeeze(-1)

@@ -570,3 +570,20 @@ def test_microbatch_push_drop_recovers_exact(swarm):
     assert torch.equal(out[:, 6:], expect)
     assert injected >= 1
     model.remote.manager.shutdown()
+
+
+def test_sampled_generation_seedable(swarm):
+    """do_sample with an explicit generator is reproducible (and actually
+    samples: different seeds should usually diverge)."""
+    boot, _ = swarm
+    model = _make_model(boot)
+    gen = torch.Generator().manual_seed(5)
+    prompt = torch.randint(0, 1000, (1, 6), generator=gen)
+    a = model.generate(prompt, max_new_tokens=8, do_sample=True,
+                       temperature=1.5,
+                       generator=torch.Generator().manual_seed(7))
+    b = model.generate(prompt, max_new_tokens=8, do_sample=True,
+                       temperature=1.5,
+                       generator=torch.Generator().manual_seed(7))
+    assert torch.equal(a, b)
+    model.remote.manager.shutdown()
