@@ -121,10 +121,21 @@ class ConnectionHandler:
 
     # ------------------------------------------------------------------
     async def rpc_inference(self, meta, tensors, stream: Stream):
-        """Bidirectional decode stream. Open meta: {session_id?, max_length,
-        batch_size, push_to?: [host, port, session_id]}. Items: {pos} +
-        [hidden] (or empty tensors in push-only mode — inputs arrive via
-        rpc_push). Replies: {pos} + [hidden_out]."""
+        """Bidirectional decode stream.
+
+        Open meta: {session_id?, max_length, batch_size, adapter?, codec?,
+        push_to?: [host, port, session_id], push_only_recv?, quiet?}.
+
+        Item wire format (meta + tensors):
+          * plain step:  {pos, step} + [hidden]           (+ optional deep
+            p-tune prompts tensor appended: (n_local_blocks, pre, H))
+          * spec step:   {pos, step, spec: true, tree?} +
+                         [hidden, position_ids, tree_mask]
+          * commit:      {spec_commit: keep}              (no tensors)
+          * close:       {close: true}
+          * pushed micro-batch (via rpc_push): adds mb: {offset, total}
+        Replies: {pos, step, keep?} + [hidden_out]; quiet middle spans in
+        push mode reply nothing (the LAST span answers the client)."""
         sid = meta.get("session_id") or uuid.uuid4().hex
         adapter = meta.get("adapter")
         codec = meta.get("codec", "raw")
