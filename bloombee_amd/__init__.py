@@ -16,5 +16,6 @@ from bloombee_amd.models.auto import (  # noqa: F401
     AutoDistributedConfig,
     AutoDistributedModel,
     AutoDistributedModelForCausalLM,
+    AutoDistributedModelForSequenceClassification,
     AutoDistributedSpeculativeModel,
 )
