@@ -167,3 +167,56 @@ def test_training_forward_respects_adapter(tmp_path):
     finally:
         s1.shutdown()
         boot.shutdown()
+
+
+def test_adapter_plus_deep_ptune_compose(tmp_path):
+    """LoRA adapters (server-side, frozen) and deep prompt-tuning
+    (client-side, trainable) must compose: prompts train against the
+    adapter-modified stack."""
+    import torch
+
+    from bloombee_amd.client import ClientConfig
+    from bloombee_amd.engine import BlockStack
+    from bloombee_amd.models.auto import AutoDistributedModelForCausalLM
+    from bloombee_amd.models.base import resolve_config
+    from bloombee_amd.net.dht import Dht
+    from bloombee_amd.server import Server
+
+    cfg_m = resolve_config("llama-tiny")
+    stack = BlockStack(cfg_m, 0, 4, device="cpu", seed=0)
+    torch.manual_seed(61)
+    for i, blk in enumerate(stack.blocks):
+        sets = create_lora_adapter(blk, rank=2, seed=60 + i)
+        for s in sets.values():
+            s.b.normal_(0, 0.05)
+        save_adapter(sets, str(tmp_path / f"block{i}"))
+
+    boot = Dht()
+    srv = Server("llama-tiny", initial_peers=[boot.endpoint],
+                 block_indices=(0, 4), device="cpu", seed=0,
+                 kv_max_tokens=1 << 14, adapters={"demo": str(tmp_path)})
+    srv.run_in_background()
+    try:
+        model = AutoDistributedModelForCausalLM.from_pretrained(
+            "llama-tiny",
+            client_config=ClientConfig(initial_peers=[boot.endpoint],
+                                       active_adapter="demo"),
+            seed=0, pre_seq_len=4, deep_ptune=True)
+        opt = torch.optim.Adam(model.trainable_parameters(), lr=5e-2)
+        gen = torch.Generator().manual_seed(62)
+        ids = torch.randint(0, 1000, (1, 6), generator=gen)
+        target = torch.randint(0, 1000, (1, 10), generator=gen)
+        losses = []
+        for _ in range(3):
+            logits = model(ids)
+            loss = torch.nn.functional.cross_entropy(
+                logits.float().view(-1, logits.shape[-1]), target.view(-1))
+            opt.zero_grad()
+            loss.backward()
+            opt.step()
+            losses.append(float(loss.detach()))
+        assert losses[-1] < losses[0], losses
+        model.remote.manager.shutdown()
+    finally:
+        srv.shutdown()
+        boot.shutdown()
