@@ -123,10 +123,16 @@ class RemoteSequenceManager:
                 logger.warning("swarm map refresh failed: %s", e)
 
     def _usable_spans(self, start_index: int, end_index: int) -> List[RemoteSpanInfo]:
-        """Every contiguous span any server offers within [start, end)."""
+        """Every contiguous span any server offers within [start, end).
+
+        A server advertising DISJOINT ranges (e.g. [0,2) and [5,8)) yields
+        one candidate span per contiguous segment — keying on peer_id alone
+        would drop every segment after the first."""
         with self._lock:
             infos = self.infos
-        spans: Dict[str, RemoteSpanInfo] = {}
+        # (peer_id -> last open segment); finished segments move to `out`
+        open_spans: Dict[str, RemoteSpanInfo] = {}
+        out: List[RemoteSpanInfo] = []
         for i in range(start_index, end_index):
             for peer_id, srv in infos[i].servers.items():
                 if srv.state != ServerState.ONLINE:
@@ -139,12 +145,15 @@ class RemoteSequenceManager:
                 if (self.config.blocked_servers is not None
                         and peer_id in self.config.blocked_servers):
                     continue
-                sp = spans.get(peer_id)
+                sp = open_spans.get(peer_id)
                 if sp is not None and sp.end == i:
                     sp.end = i + 1
-                elif sp is None:
-                    spans[peer_id] = RemoteSpanInfo(peer_id, i, i + 1, srv)
-        return list(spans.values())
+                else:
+                    if sp is not None:
+                        out.append(sp)  # gap: close the previous segment
+                    open_spans[peer_id] = RemoteSpanInfo(peer_id, i, i + 1, srv)
+        out.extend(open_spans.values())
+        return out
 
     # -- routes -----------------------------------------------------------
     def make_sequence(self, start_index: int = 0, end_index: Optional[int] = None,

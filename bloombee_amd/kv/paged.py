@@ -229,6 +229,8 @@ class SessionHandle:
         """Make room for `num_tokens` new tokens on every sequence, allocating
         pages as needed. If `speculative`, the tokens sit above l_acc and can
         be rolled back (ref paged_kv.py track/commit/rollback:206-261)."""
+        if self.is_swapped:
+            raise PagedKVError("extend on a swapped-out session; swap_in first")
         P = self.cache.page_size
         need_total = 0
         per_seq_need = []
@@ -382,9 +384,14 @@ class SessionHandle:
         else:
             self._swapped = snap
         freed = []
+        # record per-seq page COUNTS and clear the id lists: stale ids left
+        # in seq.pages would be freed AGAIN by _release/rollback/truncate
+        # while swapped, putting duplicates in the free list so two sessions
+        # could share a physical page
+        self._swapped_counts = [len(s.pages) for s in self.seqs]
         for s in self.seqs:
             freed.extend(s.pages)
-            s.pages = list(s.pages)  # keep COUNT for re-alloc; ids stale
+            s.pages.clear()
         # release the physical pages but keep the reservation (the session
         # still owns its token budget)
         self.cache._give_pages(freed)
@@ -416,7 +423,7 @@ class SessionHandle:
         stream = torch.cuda.Stream(cache.device) if on_gpu else None
         with torch.cuda.stream(stream) if on_gpu else contextlib.nullcontext():
             for b, s in enumerate(self.seqs):
-                n = len(s.pages)
+                n = self._swapped_counts[b]
                 new_pages = self.cache._take_pages(n)
                 s.pages = new_pages
                 for j, pg in enumerate(new_pages):
@@ -434,6 +441,17 @@ class SessionHandle:
             stream.synchronize()
         self._table_dirty = True
         self._swapped = None
+        del self._swapped_counts
+        # a truncate/rollback issued while swapped shrank l_spec without
+        # touching pages (there were none); trim the restored excess now
+        P = cache.page_size
+        freed: List[int] = []
+        for s in self.seqs:
+            keep = (s.l_spec + P - 1) // P
+            while len(s.pages) > keep:
+                freed.append(s.pages.pop())
+        if freed:
+            self.cache._give_pages(freed)
 
     @property
     def is_swapped(self) -> bool:

@@ -112,9 +112,15 @@ class _Conn:
         cid = meta.get("id")
         kind = meta.get("kind")
         if cid in self.streams and kind in ("item", "end", "err"):
-            self.streams[cid]._feed((meta, tensors))
+            stream = self.streams[cid]
+            stream._feed((meta, tensors))
             if kind in ("end", "err"):
-                self.streams[cid]._feed(None) if kind == "err" else None
+                if kind == "err":
+                    stream._feed(None)
+                # unregister: no further frames arrive under this id, and
+                # long-lived shared per-peer connections must not accumulate
+                # one Stream + queue per finished session
+                self.streams.pop(cid, None)
             return True
         if cid in self.pending and kind in ("resp", "err"):
             fut = self.pending.pop(cid)

@@ -147,6 +147,12 @@ class ConnectionHandler:
         loop = asyncio.get_event_loop()
         times = StageTimes()
         mb_buffers: Dict[int, dict] = {}  # pos -> {offset: out_cpu}
+        # persistent reader futures for push_only_recv: cancelling a q.get()
+        # that already dequeued an item silently drops that frame (lost step
+        # or spec_commit -> stalled session). Instead the
+        # losing future stays pending and is reused next iteration.
+        get_push: Optional[asyncio.Future] = None
+        get_cli: Optional[asyncio.Future] = None
         await loop.run_in_executor(
             None, lambda: self.backend.open_session(sid, batch_size, max_length))
         try:
@@ -154,16 +160,18 @@ class ConnectionHandler:
             while True:
                 if push_only_recv:
                     q = self._push_q.setdefault(sid, asyncio.Queue())
-                    get_push = asyncio.ensure_future(q.get())
-                    get_cli = asyncio.ensure_future(stream.recv())
-                    done, pending = await asyncio.wait(
-                        {get_push, get_cli}, return_when=asyncio.FIRST_COMPLETED)
-                    if get_push in done:
-                        get_cli.cancel()
+                    if get_push is None:
+                        get_push = asyncio.ensure_future(q.get())
+                    if get_cli is None:
+                        get_cli = asyncio.ensure_future(stream.recv())
+                    await asyncio.wait({get_push, get_cli},
+                                       return_when=asyncio.FIRST_COMPLETED)
+                    if get_push.done():  # push preferred (ref merge order)
                         item_meta, item_tensors = get_push.result()
+                        get_push = None
                     else:
-                        get_push.cancel()
                         item = get_cli.result()
+                        get_cli = None
                         if item is None:
                             break
                         item_meta, item_tensors = item
@@ -291,6 +299,9 @@ class ConnectionHandler:
                 # reads outputs from the LAST span only (push_only_downstream,
                 # ref inference_session.py:178-196)
         finally:
+            for fut in (get_push, get_cli):
+                if fut is not None:
+                    fut.cancel()
             if times.steps:
                 logger.info("session %s closed\n%s", sid[:8], times.table())
             await loop.run_in_executor(None,

@@ -59,7 +59,9 @@ class MultiDrafter:
                 probs0 = torch.softmax(logits0, -1)
                 w0 = max(1, widths[0])
                 top = probs0.topk(w0)
-                roots = [tree.add(int(t), -1, float(p))
+                # full draft dist stored per node: the exact SpecInfer
+                # rejection residual (p_target - p_draft)+ needs it
+                roots = [tree.add(int(t), -1, float(p), dist=probs0)
                          for t, p in zip(top.indices, top.values)]
 
                 lock = threading.Lock()
@@ -83,7 +85,8 @@ class MultiDrafter:
                         p = torch.softmax(lg, -1)
                         t = int(p.argmax())
                         with lock:
-                            chain_parent = tree.add(t, chain_parent, float(p[t]))
+                            chain_parent = tree.add(t, chain_parent,
+                                                    float(p[t]), dist=p)
                         chain_tok = t
                         local.append(t)
                     return local

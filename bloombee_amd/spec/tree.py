@@ -19,13 +19,19 @@ class TokenTree:
     parents: List[int] = field(default_factory=list)
     # optional draft probabilities q(token | parent ctx) for stochastic verify
     probs: List[float] = field(default_factory=list)
+    # optional FULL draft distributions q(. | parent ctx) per node — needed
+    # for the exact SpecInfer rejection residual (p_target - p_draft)+ in
+    # verify_tree_sampling; None entries fall back to the scalar approximation
+    dists: List[Optional[torch.Tensor]] = field(default_factory=list)
 
-    def add(self, token: int, parent: int, prob: float = 1.0) -> int:
+    def add(self, token: int, parent: int, prob: float = 1.0,
+            dist: Optional[torch.Tensor] = None) -> int:
         if not (-1 <= parent < len(self.tokens)):
             raise ValueError(f"bad parent {parent}")
         self.tokens.append(int(token))
         self.parents.append(int(parent))
         self.probs.append(float(prob))
+        self.dists.append(dist)
         return len(self.tokens) - 1
 
     def __len__(self) -> int:

@@ -48,9 +48,13 @@ def verify_tree_sampling(tree: TokenTree, logits: torch.Tensor,
 
     At each node, children are tried in order: child c with draft prob q_c
     and target prob p_c is accepted with min(1, p_c / q_c); on rejection the
-    target residual is renormalized (p <- max(p - q, 0)) and the next child
-    tried; if all children are rejected, the bonus token is sampled from the
-    residual distribution.
+    target residual is the reference's residual_distribution — when the tree
+    carries full draft distributions (tree.dists, populated by the drafter)
+    the residual is normalize(clamp(p_target - p_draft, 0)), which makes the
+    output distribution-equivalent to sampling the target directly. Trees
+    without stored distributions fall back to subtracting only the rejected
+    token's scalar draft prob (p[tok] <- max(p[tok] - q_c, 0)): a biased
+    approximation that over-weights tokens the draft liked but didn't pick.
     """
     accepted: List[int] = []
     cur_children = tree.children(start) if start is not None else tree.roots()
@@ -65,8 +69,13 @@ def verify_tree_sampling(tree: TokenTree, logits: torch.Tensor,
             if r < min(1.0, float(p[tok]) / q):
                 matched = c
                 break
-            # reject: renormalize the residual
-            p[tok] = torch.clamp(p[tok] - q, min=0.0)
+            # reject: renormalize the residual (full-dist subtraction when
+            # the drafter recorded q(.|ctx); scalar fallback otherwise)
+            q_dist = tree.dists[c] if c < len(tree.dists) else None
+            if q_dist is not None:
+                p = torch.clamp(p - q_dist.float().to(p.device), min=0.0)
+            else:
+                p[tok] = torch.clamp(p[tok] - q, min=0.0)
             s = p.sum()
             if s <= 0:
                 p = torch.ones_like(p) / p.numel()

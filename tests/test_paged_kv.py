@@ -232,3 +232,50 @@ def test_swap_out_compressed_roundtrip(tmp_path):
     assert (k1 - k0).abs().max() < 0.07  # 4-bit over [-1,1]: step ~ 2/15
     assert (v1 - v0).abs().max() < 0.07
     h.close()
+
+
+def test_swap_out_close_no_double_free():
+    """Closing a swapped-out session must not put its (already freed) page
+    ids back in the free list — duplicates would let two sessions share a
+    physical page."""
+    c = make_cache(max_tokens=128, page=16)  # 8-page pool
+    h = c.allocate(1, 64)
+    h.extend(33)  # 3 pages
+    h.swap_out()
+    assert sorted(c._free_pages) == list(range(c.n_pages))
+    h.close()  # must NOT re-free the swapped pages
+    assert sorted(c._free_pages) == list(range(c.n_pages))
+    assert len(c._free_pages) == c.n_pages
+
+
+def test_swap_out_rollback_truncate_then_swap_in():
+    c = make_cache(max_tokens=256, page=16)
+    h = c.allocate(1, 128)
+    h.extend(40)
+    k0 = h.k_pages(0)[int(h.page_table_host()[0, 0]), 0, 3, :].clone()
+    h.swap_out()
+    # truncate while swapped: no pages held, must not free anything twice
+    h.truncate([20])
+    free_before = sorted(c._free_pages)
+    assert len(free_before) == len(set(free_before))
+    h.swap_in()
+    # only ceil(20/16)=2 pages retained after the trim
+    assert len(h.seqs[0].pages) == 2
+    assert h.lengths == [20]
+    k1 = h.k_pages(0)[int(h.page_table_host()[0, 0]), 0, 3, :]
+    assert torch.equal(k0, k1)
+    free_now = c._free_pages
+    assert len(free_now) == len(set(free_now))
+    h.close()
+    assert sorted(c._free_pages) == list(range(c.n_pages))
+
+
+def test_swap_out_extend_guarded():
+    from bloombee_amd.kv.paged import PagedKVError
+    c = make_cache(max_tokens=128)
+    h = c.allocate(1, 64)
+    h.extend(16)
+    h.swap_out()
+    with pytest.raises(PagedKVError):
+        h.extend(1)
+    h.close()

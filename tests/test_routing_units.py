@@ -69,3 +69,33 @@ def test_route_none_when_gap():
     left = _span("left", 0, 2)
     assert mgr._route_max_throughput([left], 0, 4) is None
     assert mgr._route_min_latency([left], 0, 4, None) is None
+
+
+def test_disjoint_ranges_one_span_per_segment():
+    """A server advertising [0,2) and [3,5) must contribute BOTH segments as
+    candidate spans (keying by peer alone dropped the second)."""
+    from bloombee_amd.client.config import ClientConfig
+    from bloombee_amd.data_structures import RemoteModuleInfo
+    mgr = _mgr(num_blocks=5)
+    mgr.config = ClientConfig(initial_peers=[])
+    from bloombee_amd.client.routing import Blacklist
+    mgr.blacklist = Blacklist(1.0)
+    import threading
+    mgr._lock = threading.Lock()
+    si = ServerInfo(host="h", port=1, throughput=1.0)
+    infos = []
+    for i in range(5):
+        info = RemoteModuleInfo(uid=f"m.{i}")
+        if i in (0, 1, 3, 4):
+            info.servers["pA"] = si
+        if i == 2:
+            info.servers["pB"] = si
+        infos.append(info)
+    mgr.infos = infos
+    spans = mgr._usable_spans(0, 5)
+    a_spans = sorted((s.start, s.end) for s in spans if s.peer_id == "pA")
+    assert a_spans == [(0, 2), (3, 5)]
+    # and a route exists across pA(0,2) + pB(2,3) + pA(3,5)
+    route = mgr._route_max_throughput(spans, 0, 5)
+    assert route is not None
+    assert [b for s in route for b in range(s.start, s.end)] == list(range(5))
