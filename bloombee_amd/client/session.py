@@ -51,12 +51,17 @@ class _SpanSession:
 
     @classmethod
     def create(cls, span: RemoteSpanInfo, batch_size: int, max_length: int,
-               push_to: Optional[Tuple[str, int, str]] = None,
+               push_to: Optional[Tuple] = None,
                push_only_recv: bool = False, quiet: bool = False,
                timeout: float = 30.0, adapter: Optional[str] = None,
                codec: str = "raw") -> "_SpanSession":
         client = get_client(span.server_info.host, span.server_info.port)
         sid = uuid.uuid4().hex
+        # device data plane: if this server shares our torch.distributed
+        # world, step payloads ride RCCL/xGMI to its rank; "drank" tells it
+        # where to send replies (net/channels.py)
+        from bloombee_amd.net.channels import channels
+        span_rank = (span.server_info.dist_rank if channels.enabled else None)
 
         async def open_():
             stream = await client.open_stream("rpc_inference", {
@@ -68,7 +73,8 @@ class _SpanSession:
                 "quiet": quiet,
                 "adapter": adapter,
                 "codec": codec,
-            })
+                "drank": channels.rank,
+            }, dist_rank=span_rank)
             first = await stream.recv()
             if first is None or not first[0].get("ok"):
                 raise RpcError(f"failed to open session on {span.peer_id}")
@@ -170,7 +176,8 @@ class InferenceSession:
                 codec=getattr(self.config, "wire_codec", "raw"))
             sessions[i] = s
             push_to = (route[i].server_info.host, route[i].server_info.port,
-                       s.session_id) if use_push else None
+                       s.session_id,
+                       route[i].server_info.dist_rank) if use_push else None
         self.spans = sessions  # type: ignore[assignment]
         self._push_mode = use_push
         if replay and self.history:

@@ -42,6 +42,10 @@ class ServerInfo:
     device: Optional[str] = None       # e.g. "cuda:0" / "cpu"
     start_block: Optional[int] = None
     end_block: Optional[int] = None
+    # torch.distributed rank when this server is part of a shared world
+    # (one process per GPU on a node): lets peers route activation payloads
+    # over the RCCL/xGMI device plane (net/channels.py) instead of TCP
+    dist_rank: Optional[int] = None
 
     def to_dict(self) -> dict:
         d = dataclasses.asdict(self)

@@ -23,7 +23,8 @@ from typing import Any, AsyncIterator, Awaitable, Callable, Dict, List, Optional
 
 import torch
 
-from bloombee_amd.net.tensors import pack_frame, unpack_frame
+from bloombee_amd.net.tensors import (PendingTensor, pack_frame,
+                                      resolve_frame_tensors, unpack_frame)
 from bloombee_amd.utils.logging import get_logger
 
 logger = get_logger(__name__)
@@ -48,18 +49,24 @@ def _write_frame(writer: asyncio.StreamWriter, buf: bytes) -> None:
 
 
 class Stream:
-    """One end of a bidirectional stream."""
+    """One end of a bidirectional stream.
+
+    dist_rank: when set, tensors sent on this stream ride the device data
+    plane to that rank (RCCL over xGMI for same-node peers) — negotiated at
+    stream open (client learns the server's rank from routing, the server
+    learns the client's from the open metadata's "drank")."""
 
     def __init__(self, conn: "_Conn", call_id: int):
         self._conn = conn
         self.call_id = call_id
         self._rx: asyncio.Queue = asyncio.Queue()
         self.closed = False
+        self.dist_rank: Optional[int] = None
 
     async def send(self, meta: dict, tensors: Optional[List[torch.Tensor]] = None,
                    codec: str = "raw") -> None:
         await self._conn.send_env({"id": self.call_id, "kind": "item", **meta},
-                                  tensors, codec)
+                                  tensors, codec, dist_rank=self.dist_rank)
 
     async def send_end(self, meta: Optional[dict] = None) -> None:
         await self._conn.send_env({"id": self.call_id, "kind": "end",
@@ -101,8 +108,12 @@ class _Conn:
         self._wlock = asyncio.Lock()
         self.closed = asyncio.Event()
 
-    async def send_env(self, meta: dict, tensors=None, codec: str = "raw") -> None:
-        buf = pack_frame(meta, tensors, codec)
+    async def send_env(self, meta: dict, tensors=None, codec: str = "raw",
+                       dist_rank: Optional[int] = None) -> None:
+        # pack_frame POSTS the device sends (when dist_rank is set) before
+        # the descriptor frame hits the wire, so the receiver's matching
+        # irecv is never the first op of the pair
+        buf = pack_frame(meta, tensors, codec, dist_rank=dist_rank)
         async with self._wlock:
             _write_frame(self.writer, buf)
             await self.writer.drain()
@@ -189,6 +200,11 @@ class RpcServer:
             while True:
                 buf = await _read_frame(reader)
                 meta, tensors = unpack_frame(buf)
+                if any(isinstance(t, PendingTensor) for t in tensors):
+                    # device-plane payloads: irecvs are already posted (in
+                    # frame order); await the data before dispatch so
+                    # handlers only ever see real tensors
+                    tensors = await resolve_frame_tensors(tensors)
                 if conn.dispatch(meta, tensors):
                     continue
                 if meta.get("kind") == "req":
@@ -216,8 +232,12 @@ class RpcServer:
             if fn is None:
                 raise RpcError(f"no method {method!r}")
             out_meta, out_tensors = await fn(meta, tensors)
+            # "drank" in the request = the caller's dist rank: response
+            # tensors ride the device plane back (set only by callers that
+            # have channels enabled)
             await conn.send_env({"id": cid, "kind": "resp", **out_meta},
-                                out_tensors, meta.get("codec", "raw"))
+                                out_tensors, meta.get("codec", "raw"),
+                                dist_rank=meta.get("drank"))
         except Exception as e:  # noqa: BLE001 — reported to the peer
             logger.debug("rpc handler error in %s: %s", method, e)
             try:
@@ -269,6 +289,8 @@ class RpcClient:
             while True:
                 buf = await _read_frame(conn.reader)
                 meta, tensors = unpack_frame(buf)
+                if any(isinstance(t, PendingTensor) for t in tensors):
+                    tensors = await resolve_frame_tensors(tensors)
                 conn.dispatch(meta, tensors)
         except (asyncio.IncompleteReadError, ConnectionError, OSError):
             pass
@@ -278,6 +300,7 @@ class RpcClient:
     async def call(self, method: str, meta: Optional[dict] = None,
                    tensors: Optional[List[torch.Tensor]] = None,
                    codec: str = "raw", timeout: Optional[float] = 30.0,
+                   dist_rank: Optional[int] = None,
                    ) -> Tuple[dict, List[torch.Tensor]]:
         from bloombee_amd.utils.fault_injection import maybe_fail
         maybe_fail(method)  # no-op unless BBAMD_FAULT_RPC_DROP is set
@@ -286,19 +309,22 @@ class RpcClient:
         fut: asyncio.Future = asyncio.get_event_loop().create_future()
         conn.pending[cid] = fut
         await conn.send_env({"id": cid, "kind": "req", "method": method,
-                             "codec": codec, **(meta or {})}, tensors, codec)
+                             "codec": codec, **(meta or {})}, tensors, codec,
+                            dist_rank=dist_rank)
         return await asyncio.wait_for(fut, timeout)
 
     async def open_stream(self, method: str, meta: Optional[dict] = None,
                           tensors: Optional[List[torch.Tensor]] = None,
-                          codec: str = "raw") -> Stream:
+                          codec: str = "raw",
+                          dist_rank: Optional[int] = None) -> Stream:
         conn = await self._ensure()
         cid = next(self._ids)
         stream = Stream(conn, cid)
+        stream.dist_rank = dist_rank
         conn.streams[cid] = stream
         await conn.send_env({"id": cid, "kind": "req", "method": method,
                              "stream": True, "codec": codec, **(meta or {})},
-                            tensors, codec)
+                            tensors, codec, dist_rank=dist_rank)
         return stream
 
     async def close(self):

@@ -186,9 +186,28 @@ def _interleave(lo: bytes, hi: bytes) -> bytes:
 
 
 def pack_frame(meta: Dict[str, Any], tensors: Optional[List[torch.Tensor]] = None,
-               codec: str = "raw") -> bytes:
+               codec: str = "raw", dist_rank: Optional[int] = None) -> bytes:
+    """dist_rank: when set (and the process's dist channels are enabled),
+    tensor payloads ride the device data plane (RCCL send/recv over xGMI for
+    same-node peers, net/channels.py) and the frame carries only per-tensor
+    descriptors {via, seq, src, dtype, shape} — the GPU→CPU→wire→GPU hop the
+    reference pays per activation (handler.py:1584-1605) disappears."""
     tensors = tensors or []
-    heads, payloads = [], []
+    heads: List[dict] = []
+    payloads: List[bytes] = []
+    if dist_rank is not None and tensors:
+        from bloombee_amd.net.channels import channels
+        if channels.enabled:
+            for t in tensors:
+                seq = channels.send(t, dist_rank)
+                heads.append({
+                    "via": "dist", "seq": seq, "src": channels.rank,
+                    "dtype": str(t.dtype), "shape": list(t.shape),
+                    "nbytes": 0, "codec": "dist",
+                    "requires_grad": bool(t.requires_grad),
+                })
+            header = msgpack.packb({"meta": meta, "tensors": heads})
+            return struct.pack("<I", len(header)) + header
     for t in tensors:
         h, p = serialize_tensor(t, codec)
         heads.append(h)
@@ -197,12 +216,62 @@ def pack_frame(meta: Dict[str, Any], tensors: Optional[List[torch.Tensor]] = Non
     return b"".join([struct.pack("<I", len(header)), header, *payloads])
 
 
-def unpack_frame(buf: bytes) -> Tuple[Dict[str, Any], List[torch.Tensor]]:
+class PendingTensor:
+    """Placeholder for a tensor in flight on the device data plane; the read
+    loop resolves it (resolve_frame_tensors) before dispatching the frame."""
+
+    __slots__ = ("fut", "dtype", "requires_grad")
+
+    def __init__(self, fut, dtype: torch.dtype, requires_grad: bool):
+        self.fut = fut
+        self.dtype = dtype
+        self.requires_grad = requires_grad
+
+    def resolve(self, timeout: Optional[float] = None) -> torch.Tensor:
+        t = self.fut.result(timeout)
+        if t.dtype != self.dtype:  # bool rides the wire as uint8
+            t = t.to(self.dtype)
+        if self.requires_grad:
+            t.requires_grad_(True)
+        return t
+
+
+def unpack_frame(buf: bytes) -> Tuple[Dict[str, Any], List]:
+    """Tensors sent via the device data plane come back as PendingTensor —
+    the irecv is POSTED here (in frame order, preserving the per-src sequence
+    contract) but completion is awaited later by resolve_frame_tensors."""
     (hlen,) = struct.unpack_from("<I", buf, 0)
     header = msgpack.unpackb(buf[4:4 + hlen])
     off = 4 + hlen
-    tensors = []
+    tensors: List = []
     for h in header["tensors"]:
+        if h.get("via") == "dist":
+            from bloombee_amd.net.channels import channels
+            dtype = _DTYPES[h["dtype"]]
+            fut = channels.recv(h["shape"], dtype, h["src"], h["seq"])
+            tensors.append(PendingTensor(fut, dtype,
+                                         bool(h.get("requires_grad"))))
+            continue
         tensors.append(deserialize_tensor(h, buf[off:off + h["nbytes"]]))
         off += h["nbytes"]
     return header["meta"], tensors
+
+
+async def resolve_frame_tensors(tensors: List, timeout: float = 60.0) -> List[torch.Tensor]:
+    """Await any PendingTensor entries (no-op for plain frames)."""
+    out = []
+    for t in tensors:
+        out.append(await _resolve_one(t, timeout)
+                   if isinstance(t, PendingTensor) else t)
+    return out
+
+
+async def _resolve_one(p: PendingTensor, timeout: float) -> torch.Tensor:
+    import asyncio
+
+    t = await asyncio.wait_for(asyncio.wrap_future(p.fut), timeout)
+    if t.dtype != p.dtype:
+        t = t.to(p.dtype)
+    if p.requires_grad:
+        t.requires_grad_(True)
+    return t

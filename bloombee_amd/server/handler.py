@@ -83,7 +83,10 @@ class ConnectionHandler:
 
     # ------------------------------------------------------------------
     async def rpc_info(self, meta, tensors):
-        return self.backend.info(), []
+        from bloombee_amd.net.channels import channels
+        info = self.backend.info()
+        info["dist_rank"] = channels.rank
+        return info, []
 
     async def rpc_forward(self, meta, tensors):
         # tensors: [hidden] or [hidden, deep_prompts(n_local_blocks, pre, H)]
@@ -141,9 +144,22 @@ class ConnectionHandler:
         codec = meta.get("codec", "raw")
         max_length = int(meta["max_length"])
         batch_size = int(meta["batch_size"])
-        push_to = meta.get("push_to")  # downstream [host, port]
+        # downstream [host, port, session_id, dist_rank?]
+        push_to = meta.get("push_to")
         push_only_recv = bool(meta.get("push_only_recv"))  # inputs via rpc_push
         quiet = bool(meta.get("quiet"))  # don't echo outputs to the client
+        # device data plane: reply to the client's dist rank / push to the
+        # downstream's dist rank over RCCL when a shared world exists
+        from bloombee_amd.net.channels import channels
+        if channels.enabled:
+            stream.dist_rank = meta.get("drank")
+        push_rank = (push_to[3] if push_to and len(push_to) > 3
+                     and channels.enabled else None)
+        # outputs stay on-device when every consumer rides the device plane;
+        # otherwise hop to CPU once and serialize from there
+        keep_dev = (stream.dist_rank is not None or quiet) and (
+            push_to is None or push_rank is not None)
+        move = (lambda t: t) if keep_dev else (lambda t: t.cpu())
         loop = asyncio.get_event_loop()
         times = StageTimes()
         mb_buffers: Dict[int, dict] = {}  # pos -> {offset: out_cpu}
@@ -216,11 +232,11 @@ class ConnectionHandler:
                         None, lambda: self.backend.inference_step(
                             sid, hidden, pos, prompts, batch_offset=off,
                             adapter=adapter))
-                    part_cpu = part.cpu()
+                    part_cpu = move(part)
                     if push_to is not None:
                         asyncio.ensure_future(self._push_downstream(
                             push_to, pos, part_cpu, item_meta, mb=mbinfo,
-                            codec=codec))
+                            codec=codec, dist_rank=push_rank))
                     buf = mb_buffers.setdefault(pos, {})
                     buf[off] = part_cpu
                     got = sum(t.shape[0] for t in buf.values())
@@ -257,13 +273,13 @@ class ConnectionHandler:
                                 None, lambda j=j: self.backend.inference_step(
                                     sid, hidden[j:j + mbs], pos, prompts,
                                     batch_offset=j, adapter=adapter))
-                            part_cpu = part.cpu()
+                            part_cpu = move(part)
                             outs.append(part_cpu)
                             push_tasks.append(asyncio.ensure_future(
                                 self._push_downstream(
                                     push_to, pos, part_cpu, item_meta,
                                     mb={"offset": j, "total": B},
-                                    codec=codec)))
+                                    codec=codec, dist_rank=push_rank)))
                         out_cpu = torch.cat(outs, dim=0)
                         for t_ in push_tasks:
                             await t_
@@ -283,12 +299,13 @@ class ConnectionHandler:
                     out, keep = await loop.run_in_executor(
                         None, lambda: self.backend.prune_tree(
                             out, tree["tokens"], tree["parents"]))
-                out_cpu = out.cpu()
+                out_cpu = move(out)
                 times.bump_step()
                 if push_to is not None:
                     with times.span("push"):
                         await self._push_downstream(push_to, pos, out_cpu,
-                                                    item_meta, codec=codec)
+                                                    item_meta, codec=codec,
+                                                    dist_rank=push_rank)
                 if not quiet:
                     with times.span("reply"):
                         await stream.send(
@@ -314,7 +331,8 @@ class ConnectionHandler:
 
     async def _push_downstream(self, push_to, pos: int, hidden: torch.Tensor,
                                item_meta: dict, mb: Optional[dict] = None,
-                               codec: str = "raw") -> None:
+                               codec: str = "raw",
+                               dist_rank: Optional[int] = None) -> None:
         import time as _time
 
         from bloombee_amd.utils.fault_injection import maybe_fail
@@ -335,7 +353,7 @@ class ConnectionHandler:
                     "rpc_push", {"session_id": down_sid, "pos": pos,
                                  "step": item_meta.get("step"),
                                  "mb": mb}, [hidden],
-                    codec=codec, timeout=30)
+                    codec=codec, timeout=30, dist_rank=dist_rank)
         except Exception as e:  # noqa: BLE001 — client will fall back
             ok = False
             logger.warning("s2s push to %s failed: %s", push_to, e)

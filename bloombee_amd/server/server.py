@@ -135,6 +135,7 @@ class Server:
 
     # ------------------------------------------------------------------
     def _server_info(self) -> ServerInfo:
+        from bloombee_amd.net.channels import channels
         return ServerInfo(
             state=ServerState.ONLINE,
             host=self.announce_host or self.endpoint[0],
@@ -145,6 +146,7 @@ class Server:
             device=self.device,
             start_block=self.block_range[0], end_block=self.block_range[1],
             next_pings=dict(self._next_pings),
+            dist_rank=channels.rank,
         )
 
     def _measure_next_pings(self) -> None:
