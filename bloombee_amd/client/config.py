@@ -32,3 +32,7 @@ class ClientConfig:
     wire_codec: str = "raw"
     use_server_to_server: bool = True      # s2s activation push during decode
     push_only_downstream_decode: bool = True
+    # keep per-step span-0 inputs for failover history replay (ref
+    # inference_session.py:71,139-150). Benchmarks on a healthy single node
+    # turn this off: the history pins every step's activations in memory.
+    keep_history: bool = True

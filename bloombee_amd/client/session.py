@@ -267,7 +267,8 @@ class InferenceSession:
                 time.sleep(delay)
                 self.manager.update()
                 # reopen happens at the top of the loop (also under retry)
-        self.history.append((pos, hidden, prompts))
+        if self.config.keep_history:
+            self.history.append((pos, hidden, prompts))
         self.position = pos + hidden.shape[1]
         self.step_count += 1
         return out

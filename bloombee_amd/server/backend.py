@@ -79,6 +79,16 @@ class StackBackend:
         self.pool = TaskPool(name=f"worker[{start}:{end}]")
         self.sessions: Dict[str, SessionState] = {}
         self._lock = threading.Lock()
+        # hipGraph decode capture (the server backend
+        # step was not captured — only the single-rank pipeline was). The
+        # whole local block range replays as one graph per (session, B);
+        # dynamic state (token position, page table, KV) lives in device
+        # tensors so the same graph serves every decode step.
+        from bloombee_amd.config import get_config
+        self._use_graphs = (get_config().use_hip_graphs
+                            and self.device.type == "cuda"
+                            and type(self.stack).__name__ == "BlockStack")
+        self._graphs: Dict[str, dict] = {}
 
     # -- sessions ---------------------------------------------------------
     def open_session(self, session_id: str, batch_size: int, max_length: int,
@@ -129,6 +139,7 @@ class StackBackend:
     def close_session(self, session_id: str) -> None:
         with self._lock:
             state = self.sessions.pop(session_id, None)
+        self._graphs.pop(session_id, None)
         if state is not None:
             state.handle.close()
 
@@ -229,6 +240,11 @@ class StackBackend:
                     f"only {cur} tokens (gap)")
             elif speculative:
                 handle.rollback()  # drop any uncommitted previous tree
+            if (self._use_graphs and T == 1 and not speculative
+                    and prompts is None and position_ids is None
+                    and tree_mask is None and adapter is None
+                    and start_pos == cur):
+                return self._graphed_decode(session_id, handle, h, start_pos)
             pos = (position_ids.to(self.device).int()
                    if position_ids is not None else None)
             tm = tree_mask.to(self.device) if tree_mask is not None else None
@@ -266,6 +282,44 @@ class StackBackend:
             return out
 
         return self.pool.submit(run, PRIORITY_INFERENCE).result()
+
+    def _graphed_decode(self, session_id: str, handle, h: torch.Tensor,
+                        start_pos: int) -> torch.Tensor:
+        """Plain decode step (T=1, committed position) as a hipGraph replay.
+
+        Runs ON the worker thread. All host-varying inputs are copied into
+        persistent device buffers before replay; kv growth (extend +
+        page-table flush) happens outside the graph, exactly like the
+        single-rank pipeline capture (parallel/pipeline.py:151-175).
+        capture_error_mode="thread_local" keeps the channels pump (RCCL ops
+        on other threads) legal during capture."""
+        B = h.shape[0]
+        st = self._graphs.get(session_id)
+        if st is None or st["B"] != B:
+            st = {"B": B,
+                  "in": torch.empty_like(h),
+                  "pos": torch.empty(B, dtype=torch.int32,
+                                     device=self.device),
+                  "out": None, "graph": None, "warm": 0}
+            self._graphs[session_id] = st
+        st["in"].copy_(h)
+        st["pos"].fill_(start_pos)
+        handle.extend(1)
+        handle.page_table()
+        if st["graph"] is None:
+            if st["warm"] < 2:  # warm the allocator/kernels before capture
+                st["warm"] += 1
+                return self.stack.forward_inference(st["in"], handle,
+                                                    st["pos"])
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g, capture_error_mode="thread_local"):
+                st["out"] = self.stack.forward_inference(st["in"], handle,
+                                                         st["pos"])
+            st["graph"] = g
+            logger.info("decode step blocks[%d:%d] B=%d captured as hipGraph",
+                        self.start, self.end, B)
+        st["graph"].replay()
+        return st["out"]
 
     def prune_tree(self, hidden: torch.Tensor, tokens: list,
                    parents: list):

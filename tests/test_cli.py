@@ -135,3 +135,25 @@ def test_identity_persistence_and_batch_cap(tmp_path):
     assert s2.peer_id == pid
     s2.shutdown()
     boot.shutdown()
+
+
+@pytest.mark.timeout(300)
+def test_bench_swarm_mode_cpu_contract():
+    """bench.py --mode swarm runs the REAL serving stack (Server + session +
+    dist channels) and prints the driver's JSON contract line."""
+    import json
+    import subprocess
+    import sys
+
+    r = subprocess.run(
+        [sys.executable, "bench.py", "--mode", "swarm", "--model",
+         "llama-tiny", "--gpus", "1", "--steps", "3", "--warmup", "1",
+         "--batch-per-gpu", "2", "--prompt", "16", "--device", "cpu"],
+        capture_output=True, text=True, timeout=240,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    out = json.loads(line)
+    assert out["unit"] == "tokens/s" and out["value"] > 0
+    assert "serving stack" in out["metric"]
+    assert out["config"]["parallelism"] == "swarm-pp1"
