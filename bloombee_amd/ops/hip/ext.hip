@@ -346,7 +346,9 @@ torch::Tensor attn_prefill(torch::Tensor q, torch::Tensor k_pages,
                 "tree_mask must be contiguous bool (B, Tq, Tq)");
     TORCH_CHECK(window == 0, "tree_mask + sliding window: use the torch path");
     CHECK_DEV((*tree_mask));
-    tm_ptr = tree_mask->data_ptr<unsigned char>();
+    // kBool tensors dispatch data_ptr<bool>, not <unsigned char> (Byte)
+    tm_ptr = reinterpret_cast<const unsigned char*>(
+        tree_mask->data_ptr<bool>());
   }
   auto out = torch::empty_like(q);
   dim3 grid((Tq + 127) / 128, B * Hq);

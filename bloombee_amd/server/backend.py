@@ -89,6 +89,7 @@ class StackBackend:
                             and self.device.type == "cuda"
                             and type(self.stack).__name__ == "BlockStack")
         self._graphs: Dict[str, dict] = {}
+        self.graphs_captured = 0  # lifetime count (survives session close)
 
     # -- sessions ---------------------------------------------------------
     def open_session(self, session_id: str, batch_size: int, max_length: int,
@@ -316,6 +317,7 @@ class StackBackend:
                 st["out"] = self.stack.forward_inference(st["in"], handle,
                                                          st["pos"])
             st["graph"] = g
+            self.graphs_captured += 1
             logger.info("decode step blocks[%d:%d] B=%d captured as hipGraph",
                         self.start, self.end, B)
         st["graph"].replay()
