@@ -57,6 +57,16 @@ class ModelConfig:
 
     @classmethod
     def from_dict(cls, d: dict) -> "ModelConfig":
+        d = dict(d)
+        # transformers >=5 config spellings: "dtype" replaced "torch_dtype",
+        # rope params moved under "rope_parameters"
+        if "dtype" in d and "torch_dtype" not in d:
+            d["torch_dtype"] = d.pop("dtype")
+        rp = d.get("rope_parameters")
+        if isinstance(rp, dict):
+            d.setdefault("rope_theta", rp.get("rope_theta", 10000.0))
+            if rp.get("rope_type", "default") not in ("default", None):
+                d.setdefault("rope_scaling", rp)
         known = {f.name for f in dataclasses.fields(cls)}
         kw = {k: v for k, v in d.items() if k in known}
         extras = {k: v for k, v in d.items() if k not in known}

@@ -26,7 +26,8 @@ class DistributedLlamaModel(torch.nn.Module):
                  model_name: str, seed: int = 0, device: str = "cpu",
                  manager: Optional[RemoteSequenceManager] = None,
                  pre_seq_len: int = 0,
-                 deep_ptune: bool = False):
+                 deep_ptune: bool = False,
+                 checkpoint_dir: Optional[str] = None):
         super().__init__()
         self.config = config
         self.device_ = torch.device(device)
@@ -41,6 +42,19 @@ class DistributedLlamaModel(torch.nn.Module):
         self.norm_w = torch.nn.Parameter(
             torch.ones(config.hidden_size, dtype=dt, device=device),
             requires_grad=False)
+        # real weights: the client's share of a converted checkpoint
+        # (embed + final norm; the LM head loads its own file)
+        self.checkpoint_dir = checkpoint_dir
+        if checkpoint_dir is not None:
+            from bloombee_amd.server.from_pretrained import load_client_weights
+            cw = load_client_weights(checkpoint_dir)
+            if cw["embed"] is None or cw["final_norm"] is None:
+                raise FileNotFoundError(
+                    f"no client weights (embed_tokens/norm) under "
+                    f"{checkpoint_dir}")
+            with torch.no_grad():
+                self.embed_tokens.copy_(cw["embed"].to(dt))
+                self.norm_w.copy_(cw["final_norm"].to(dt))
         self.remote = RemoteSequential(client_config, model_name,
                                        config.num_hidden_layers,
                                        manager=manager)
@@ -96,7 +110,8 @@ class LMHead(torch.nn.Module):
     """Client-side vocab projection (ref client/lm_head.py)."""
 
     def __init__(self, config: ModelConfig, embed: torch.nn.Parameter,
-                 gen: torch.Generator, device: str = "cpu"):
+                 gen: torch.Generator, device: str = "cpu",
+                 checkpoint_dir: Optional[str] = None):
         super().__init__()
         if config.tie_word_embeddings:
             self.weight = embed
@@ -104,6 +119,16 @@ class LMHead(torch.nn.Module):
             self.weight = torch.nn.Parameter(
                 torch.randn(config.vocab_size, config.hidden_size, generator=gen)
                 .mul_(0.02).to(config.dtype).to(device), requires_grad=False)
+            if checkpoint_dir is not None:
+                from bloombee_amd.server.from_pretrained import \
+                    load_client_weights
+                w = load_client_weights(checkpoint_dir)["lm_head"]
+                if w is None:
+                    raise FileNotFoundError(
+                        f"untied model but no lm_head.weight under "
+                        f"{checkpoint_dir}")
+                with torch.no_grad():
+                    self.weight.copy_(w.to(self.weight.dtype))
 
     def forward(self, hidden: torch.Tensor) -> torch.Tensor:
         return ops.linear(hidden, self.weight)
@@ -114,16 +139,19 @@ class DistributedLlamaForCausalLM(RemoteGenerationMixin, torch.nn.Module):
                  model_name: str, seed: int = 0, device: str = "cpu",
                  manager: Optional[RemoteSequenceManager] = None,
                  pre_seq_len: int = 0,
-                 deep_ptune: bool = False):
+                 deep_ptune: bool = False,
+                 checkpoint_dir: Optional[str] = None):
         super().__init__()
         self.config = config
         self.transformer = DistributedLlamaModel(config, client_config,
                                                  model_name, seed=seed,
                                                  device=device, manager=manager,
                                                  pre_seq_len=pre_seq_len,
-                                                 deep_ptune=deep_ptune)
+                                                 deep_ptune=deep_ptune,
+                                                 checkpoint_dir=checkpoint_dir)
         self.lm_head = LMHead(config, self.transformer.embed_tokens,
-                              gen=self.transformer._gen, device=device)
+                              gen=self.transformer._gen, device=device,
+                              checkpoint_dir=checkpoint_dir)
 
     # RemoteGenerationMixin hooks
     @property
@@ -151,6 +179,12 @@ class DistributedLlamaForCausalLM(RemoteGenerationMixin, torch.nn.Module):
                         seed: int = 0, device: str = "cpu", **kw):
         cfg = config or resolve_config(name_or_path)
         ccfg = client_config or ClientConfig(initial_peers=list(initial_peers))
+        # a converted checkpoint directory serves real client weights
+        # (ref client/from_pretrained.py: the client loads only its share)
+        if "checkpoint_dir" not in kw:
+            from bloombee_amd.server.from_pretrained import is_converted
+            if is_converted(name_or_path):
+                kw["checkpoint_dir"] = name_or_path
         return cls(cfg, ccfg, model_name=name_or_path, seed=seed,
                    device=device, **kw)
 

@@ -58,6 +58,11 @@ class Server:
         self.config = model if isinstance(model, ModelConfig) else resolve_config(model)
         self.model_name = model_name or (model if isinstance(model, str)
                                          else self.config.model_type)
+        # serving a converted real checkpoint: the model path IS the weight dir
+        if checkpoint_dir is None and isinstance(model, str):
+            from bloombee_amd.server.from_pretrained import is_converted
+            if is_converted(model):
+                checkpoint_dir = model
         self.device = device
         # persistent identity (ref --identity_path): keeps routing history,
         # bans and DHT records stable across restarts
