@@ -69,10 +69,34 @@ class MixtralBlock(torch.nn.Module):
         logits = ops.linear(flat, self.router_w).float()
         weights, experts = logits.topk(self.topk, dim=-1)
         weights = torch.softmax(weights, dim=-1).to(y.dtype)
+        e0, e1 = getattr(self, "ep_range", (0, self.E))
+        if (flat.is_cuda and ops.HAVE_HIP_OPS
+                and flat.dtype == torch.bfloat16
+                and (e0, e1) == (0, self.E) and flat.shape[0] <= 1024):
+            # grouped decode path: expert-sort the (token, choice) slots on
+            # device and run BOTH expert GEMMs as single grouped launches
+            # (ops/hip/moe_gemm.hip) — no per-expert host sync, no
+            # per-expert kernel launches (BASELINE config 4)
+            Tk = flat.shape[0] * self.topk
+            fe = experts.reshape(-1)
+            order = fe.argsort(stable=True)
+            counts = torch.bincount(fe, minlength=self.E)
+            off = torch.zeros(self.E + 1, dtype=torch.int32,
+                              device=flat.device)
+            off[1:] = counts.cumsum(0).int()
+            tok = (order // self.topk).int()
+            wsorted = weights.reshape(-1)[order].float()
+            mch = (flat.shape[0] + 31) // 32
+            gu = ops.moe_gemm_grouped(flat, self.expert_gate_up_w, off,
+                                      rowmap=tok, S=Tk, mchunks=mch)
+            dn = ops.moe_gemm_grouped(ops.swiglu(gu), self.expert_down_w,
+                                      off, scale=wsorted, S=Tk, mchunks=mch)
+            out = torch.zeros_like(flat)
+            out.index_add_(0, tok.long(), dn)
+            return out.view(B, T, H)
         out = torch.zeros_like(flat)
         # expert-parallel mode (parallel/expert.py): this rank holds only
         # experts [e0, e1); partial sums cross ranks via one all-reduce
-        e0, e1 = getattr(self, "ep_range", (0, self.E))
         for le, e in enumerate(range(e0, e1)):
             sel = (experts == e)
             rows = sel.any(dim=-1).nonzero(as_tuple=True)[0]

@@ -18,6 +18,7 @@ from bloombee_amd.ops.interface import (  # noqa: F401
     layer_norm,
     linear,
     mfma_selftest,
+    moe_gemm_grouped,
     quant4_pack,
     quant4_unpack,
     rms_norm,

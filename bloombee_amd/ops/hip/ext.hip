@@ -17,6 +17,7 @@
 #include "attn_decode_mfma.hip"
 #include "attn_prefill.hip"
 #include "gemm_skinny.hip"
+#include "moe_gemm.hip"
 #include "mfma_selftest.hip"
 #include "quant4.hip"
 #include "wire.h"
@@ -428,6 +429,47 @@ void rope_kv_write_(torch::Tensor qkv, long Hq_, long Hkv_, torch::Tensor cos_t,
       (int)cos_t.size(0));
 }
 
+
+// C(S,N) = A[row(s)] @ W[expert(s)]^T — grouped decode MoE (moe_gemm.hip).
+// off: (E+1,) int32 exclusive prefix sums of per-expert slot counts;
+// rowmap: optional (S,) int32 slot->A-row gather; scale: optional (S,) f32
+// per-slot epilogue scale (routing weights). S and mchunks are host-side
+// upper-bound shapes so no device sync is needed.
+torch::Tensor moe_gemm(torch::Tensor A, torch::Tensor W, torch::Tensor off,
+                       c10::optional<torch::Tensor> rowmap,
+                       c10::optional<torch::Tensor> scale,
+                       long S, long mchunks) {
+  CHECK_DEV(A); CHECK_BF16(A); CHECK_CONTIG(A);
+  CHECK_DEV(W); CHECK_BF16(W); CHECK_CONTIG(W);
+  CHECK_DEV(off);
+  TORCH_CHECK(W.dim() == 3, "W must be (E, N, K)");
+  const int E = W.size(0), N = W.size(1), K = W.size(2);
+  TORCH_CHECK(A.size(-1) == K, "A/W K mismatch");
+  TORCH_CHECK(K % 32 == 0 && N % 64 == 0, "K%32, N%64 required");
+  TORCH_CHECK(off.scalar_type() == at::kInt && off.numel() == E + 1);
+  const int* rmp = nullptr;
+  if (rowmap.has_value()) {
+    TORCH_CHECK(rowmap->scalar_type() == at::kInt &&
+                rowmap->is_contiguous() && rowmap->numel() == S);
+    CHECK_DEV((*rowmap));
+    rmp = rowmap->data_ptr<int>();
+  }
+  const float* sp = nullptr;
+  if (scale.has_value()) {
+    TORCH_CHECK(scale->scalar_type() == at::kFloat &&
+                scale->is_contiguous() && scale->numel() == S);
+    CHECK_DEV((*scale));
+    sp = scale->data_ptr<float>();
+  }
+  auto C = torch::empty({S, (long)N}, A.options());
+  TORCH_CHECK(mchunks >= 1);
+  dim3 grid(N / 64, (unsigned)(E * mchunks));
+  moe_gemm_kernel<2><<<grid, 256, 0, cur_stream()>>>(
+      bf_ptr(A), bf_ptr(W), off.data_ptr<int>(), rmp, sp, bf_ptr_mut(C),
+      N, K, (int)mchunks);
+  return C;
+}
+
 // ---------------------------------------------------------------------------
 // skinny-M GEMM (decode projections)
 // ---------------------------------------------------------------------------
@@ -595,6 +637,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope_kv_write_", &rope_kv_write_);
   m.def("attn_prefill", &attn_prefill);
   m.def("gemm_skinny", &gemm_skinny);
+  m.def("moe_gemm", &moe_gemm);
   m.def("quant4_pack", &quant4_pack);
   m.def("quant4_unpack", &quant4_unpack);
   m.def("mfma_selftest", &mfma_selftest);

@@ -296,6 +296,34 @@ def linear(x: torch.Tensor, w: torch.Tensor, residual: Optional[torch.Tensor] = 
     return y
 
 
+def moe_gemm_grouped(A: torch.Tensor, W: torch.Tensor, off: torch.Tensor,
+                     rowmap: Optional[torch.Tensor] = None,
+                     scale: Optional[torch.Tensor] = None,
+                     S: Optional[int] = None,
+                     mchunks: Optional[int] = None) -> torch.Tensor:
+    """Grouped expert GEMM, one launch over all experts (moe_gemm.hip):
+    C[s] = A[rowmap[s] if rowmap else s] @ W[expert(s)]^T (* scale[s]).
+
+    Slots are expert-sorted: [off[e], off[e+1]) belong to expert e; off is
+    the int32 exclusive prefix sum ON DEVICE (no host sync — hipGraph-safe,
+    unlike the per-expert nonzero() loop). S = total slots, mchunks = host
+    upper bound on ceil(max rows per expert / 32) (worst case: every token
+    routed to one expert). Replaces the per-expert skinny GEMM loop for
+    Mixtral decode (BASELINE config 4 "MoE grouped GEMM").
+    """
+    E, N, K = W.shape
+    if S is None:
+        S = int(rowmap.numel()) if rowmap is not None else int(A.shape[0])
+    if mchunks is None:
+        mchunks = (int(A.shape[0]) + 31) // 32
+    if (_on_gpu(A) and A.dtype == torch.bfloat16 and K % 32 == 0
+            and N % 64 == 0):
+        _require_ext()
+        return hip_ops.moe_gemm(A.contiguous(), W.contiguous(),
+                                off.int(), rowmap, scale, S, mchunks)
+    return ref.moe_gemm_grouped(A, W, off, rowmap, scale, S)
+
+
 def quant4_pack(x: torch.Tensor, group_size: int = 64):
     if _on_gpu(x):
         _require_ext()

@@ -405,3 +405,28 @@ def attn_paged_mixed(
         denom = l_hd.unsqueeze(-1) * c_h + l_d.unsqueeze(-1) * c_d
         out[b, :, 0] = ((o_hd * c_h + o_d * c_d) / denom).to(q.dtype)
     return out
+
+
+def moe_gemm_grouped(A: torch.Tensor, W: torch.Tensor, off: torch.Tensor,
+                     rowmap: Optional[torch.Tensor] = None,
+                     scale: Optional[torch.Tensor] = None,
+                     S: Optional[int] = None) -> torch.Tensor:
+    """Reference for the grouped expert GEMM (ops/hip/moe_gemm.hip):
+    per-slot C[s] = A[row(s)] @ W[expert(s)]^T * scale[s]; slots grouped by
+    expert via the off prefix sums. fp32 accumulation like the MFMA path."""
+    E, N, K = W.shape
+    offs = off.tolist()
+    if S is None:
+        S = offs[-1]
+    C = torch.zeros(S, N, dtype=A.dtype, device=A.device)
+    for e in range(E):
+        a, b = offs[e], offs[e + 1]
+        if b <= a:
+            continue
+        slots = torch.arange(a, b, device=A.device)
+        rows = rowmap[slots].long() if rowmap is not None else slots
+        y = A[rows].float() @ W[e].float().t()
+        if scale is not None:
+            y = y * scale[slots].float().unsqueeze(1)
+        C[slots] = y.to(A.dtype)
+    return C
