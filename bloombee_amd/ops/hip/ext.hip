@@ -464,9 +464,14 @@ torch::Tensor moe_gemm(torch::Tensor A, torch::Tensor W, torch::Tensor off,
   auto C = torch::empty({S, (long)N}, A.options());
   TORCH_CHECK(mchunks >= 1);
   dim3 grid(N / 64, (unsigned)(E * mchunks));
-  moe_gemm_kernel<2><<<grid, 256, 0, cur_stream()>>>(
-      bf_ptr(A), bf_ptr(W), off.data_ptr<int>(), rmp, sp, bf_ptr_mut(C),
-      N, K, (int)mchunks);
+  if (K % 256 == 0)
+    moe_gemm_v2_kernel<2><<<grid, 256, 0, cur_stream()>>>(
+        bf_ptr(A), bf_ptr(W), off.data_ptr<int>(), rmp, sp, bf_ptr_mut(C),
+        N, K, (int)mchunks);
+  else
+    moe_gemm_kernel<2><<<grid, 256, 0, cur_stream()>>>(
+        bf_ptr(A), bf_ptr(W), off.data_ptr<int>(), rmp, sp, bf_ptr_mut(C),
+        N, K, (int)mchunks);
   return C;
 }
 

@@ -118,7 +118,10 @@ def kv_gather(k_pages, v_pages, page_table, ctx_len: int, batch_index: int):
 
 
 _attn_sparsity = 1.0  # Policy.attn_sparsity (ref flexgen policy :10-55)
-_TREE_KERNEL = os.environ.get("BBAMD_TREE_KERNEL", "") not in ("", "0")
+# fused tree-mask attention: default ON since round 2 (numerics validated
+# on MI355X — tests/test_treekernel.py 3/3 after the bool-dispatch fix);
+# BBAMD_TREE_KERNEL=0 falls back to the torch composition
+_TREE_KERNEL = os.environ.get("BBAMD_TREE_KERNEL", "1") not in ("", "0")
 
 
 def set_attn_sparsity(frac: float) -> None:
@@ -250,10 +253,8 @@ def attn_paged(q, k_pages, v_pages, page_table, q_start, scale=None, window: int
             scale = 1.0 / math.sqrt(q.shape[-1])
         if (_on_gpu(q) and _TREE_KERNEL and window == 0
                 and q.shape[-1] <= 256 and alibi_slopes is None):
-            # fused tree-mask prefill kernel (spec verify). Shipped dark in
-            # round 1: BBAMD_TREE_KERNEL=1 enables it; numerics gated by
-            # tests/test_gpu_kernels.py -m treekernel before flipping the
-            # default in round 2.
+            # fused tree-mask prefill kernel (spec verify) — the default
+            # GPU path (tests/test_treekernel.py parity on MI355X)
             _require_ext()
             tm = tree_mask.to(q.device).contiguous().bool()
             return hip_ops.attn_prefill(q.contiguous(), k_pages, v_pages,
