@@ -72,7 +72,8 @@ class MixtralBlock(torch.nn.Module):
         e0, e1 = getattr(self, "ep_range", (0, self.E))
         if (flat.is_cuda and ops.HAVE_HIP_OPS
                 and flat.dtype == torch.bfloat16
-                and (e0, e1) == (0, self.E) and flat.shape[0] <= 1024):
+                and (e0, e1) == (0, self.E) and flat.shape[0] <= 1024
+                and not getattr(self, "moe_loop", False)):
             # grouped decode path: expert-sort the (token, choice) slots on
             # device and run BOTH expert GEMMs as single grouped launches
             # (ops/hip/moe_gemm.hip) — no per-expert host sync, no

@@ -465,42 +465,32 @@ def test_moe_gemm_grouped_parity():
 
 
 def test_mixtral_block_grouped_moe_decode():
-    """Mixtral block on GPU (grouped-MoE path) matches its own CPU
-    forward_inference (per-expert loop) bit-for-bit modulo bf16 tolerance."""
+    """The grouped-MoE wiring in the Mixtral block: same device, same
+    weights, grouped path vs per-expert loop path (moe_loop switch) — only
+    the MoE implementation differs, so tolerance is tight. (CPU-vs-GPU
+    whole-block noise is covered by the family parity suite.)"""
     from bloombee_amd.engine import BlockStack
     from bloombee_amd.models.base import resolve_config
 
     cfg = resolve_config("mixtral-tiny")
     torch.manual_seed(0)
-    cpu = BlockStack(cfg, 0, 1, device="cpu", seed=3)
     gpu = BlockStack(cfg, 0, 1, device=DEV, seed=3)
-    # widen router margins: with init-scale logits (~1e-3 apart) bf16
-    # rounding differences between the CPU and MFMA paths flip near-tied
-    # top-k picks, which compares two legitimately different expert mixes
-    rw = (torch.randn(cpu.blocks[0].router_w.shape,
-                      generator=torch.Generator().manual_seed(42)))
+    blk = gpu.blocks[0]
+    # widen router margins so near-ties cannot flip top-k between the two
+    # MoE implementations' (identical) routing math
+    rw = torch.randn(blk.router_w.shape,
+                     generator=torch.Generator().manual_seed(42))
     with torch.no_grad():
-        cpu.blocks[0].router_w.copy_(rw.to(cfg.dtype))
-        gpu.blocks[0].router_w.copy_(rw.to(cfg.dtype))
-    kvc = cpu.make_kv(1 << 12)
-    kvg = gpu.make_kv(1 << 12)
-    B, T = 4, 10
-    hc = kvc.allocate(B, 64)
-    hg = kvg.allocate(B, 64)
+        blk.router_w.copy_(rw.to(cfg.dtype))
     gen = torch.Generator().manual_seed(2)
-    x = (torch.randn(B, T, cfg.hidden_size, generator=gen) * 0.1).to(cfg.dtype)
-    sp = torch.zeros(B, dtype=torch.int32)
-    hc.extend(T)
-    hg.extend(T)
-    out_c = cpu.forward_inference(x.clone(), hc, sp)
-    out_g = gpu.forward_inference(x.to(DEV), hg, sp.to(DEV))
-    assert torch.allclose(out_g.cpu().float(), out_c.float(), atol=5e-2), \
-        (out_g.cpu().float() - out_c.float()).abs().max()
-    # decode step (the grouped path's home shape)
-    x1 = (torch.randn(B, 1, cfg.hidden_size, generator=gen) * 0.1).to(cfg.dtype)
-    spT = torch.full((B,), T, dtype=torch.int32)
-    hc.extend(1)
-    hg.extend(1)
-    out_c1 = cpu.forward_inference(x1.clone(), hc, spT)
-    out_g1 = gpu.forward_inference(x1.to(DEV), hg, spT.to(DEV))
-    assert torch.allclose(out_g1.cpu().float(), out_c1.float(), atol=5e-2)
+    for B, T in [(4, 10), (7, 1), (32, 1)]:
+        y = (torch.randn(B, T, cfg.hidden_size, generator=gen) * 0.1) \
+            .to(cfg.dtype).to(DEV)
+        blk.moe_loop = False
+        out_grp = blk._moe(y)
+        blk.moe_loop = True
+        out_loop = blk._moe(y)
+        blk.moe_loop = False
+        assert torch.allclose(out_grp.float(), out_loop.float(),
+                              atol=2e-2), \
+            (B, T, (out_grp - out_loop).abs().max().item())
