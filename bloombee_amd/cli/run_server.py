@@ -45,6 +45,9 @@ def main():
     ap.add_argument("--update-period", type=float, default=30.0)
     ap.add_argument("--checkpoint-dir", default=None,
                     help="per-block .npy weight dir (random init if absent)")
+    ap.add_argument("--network-rps", type=float, default=None,
+                    help="advertised NIC steps/s cap folded into throughput "
+                         "(ref speedtest-based network_rps; offline -> operator-provided)")
     ap.add_argument("--throughput", type=float, default=None,
                     help="announced rps (default: measured)")
     ap.add_argument("--session-max-idle", type=float, default=600.0,
@@ -101,7 +104,8 @@ def main():
         cfg = resolve_config(args.model)
         nb = (block_indices[1] - block_indices[0]) if block_indices \
             else (args.num_blocks or cfg.num_hidden_layers)
-        throughput = get_server_throughput(cfg, device, nb)["throughput"]
+        throughput = get_server_throughput(
+            cfg, device, nb, network_rps=args.network_rps)["throughput"]
 
     server = Server(
         args.model,
@@ -110,7 +114,8 @@ def main():
         block_indices=block_indices, num_blocks=args.num_blocks,
         device=device, seed=args.seed, kv_max_tokens=args.attn_cache_tokens,
         update_period=args.update_period, checkpoint_dir=args.checkpoint_dir,
-        throughput=throughput, session_max_idle=args.session_max_idle,
+        throughput=throughput, network_rps=args.network_rps,
+        session_max_idle=args.session_max_idle,
         offload_policy=_offload_policy(args),
         adapters=dict(a.split("=", 1) for a in args.adapters) or None,
         identity_path=args.identity_path,

@@ -35,15 +35,48 @@ _RETRYABLE = (OSError, TimeoutError, ConnectionError,
 MAX_TOKENS_IN_BATCH = 1024  # client sub-batch split (ref sequential_autograd.py:22)
 
 
+def _use_stream(span: RemoteSpanInfo, payload) -> bool:
+    """Oversized payloads stream in chunks (ref remote_forward_backward.py:
+    115-118 unary-vs-stream split) — unless the peer shares our device
+    plane, where the payload rides RCCL out-of-band anyway."""
+    from bloombee_amd.net.channels import channels
+    from bloombee_amd.net.streaming import (MAX_UNARY_PAYLOAD_BYTES,
+                                            payload_nbytes)
+    if channels.enabled and span.server_info.dist_rank is not None:
+        return False
+    return payload_nbytes(payload) > MAX_UNARY_PAYLOAD_BYTES
+
+
+def _call_streamed(span: RemoteSpanInfo, method: str, payload, timeout: float,
+                   adapter: Optional[str]):
+    from bloombee_amd.net.streaming import (recv_tensors_chunked,
+                                            send_tensors_chunked)
+    client = get_client(span.server_info.host, span.server_info.port)
+
+    async def go():
+        stream = await client.open_stream(method, {"adapter": adapter})
+        await send_tensors_chunked(stream, payload)
+        meta, tensors = await recv_tensors_chunked(stream)
+        return tensors
+
+    return run_coroutine(go(), timeout + 5)
+
+
 def _call_forward(span: RemoteSpanInfo, hidden: torch.Tensor,
                   timeout: float,
                   prompts: Optional[torch.Tensor] = None,
                   adapter: Optional[str] = None) -> torch.Tensor:
+    from bloombee_amd.net.channels import channels
     client = get_client(span.server_info.host, span.server_info.port)
     payload = [hidden] if prompts is None else [hidden, prompts]
+    if _use_stream(span, payload):
+        return _call_streamed(span, "rpc_forward_stream", payload, timeout,
+                              adapter)[0]
+    dist_rank = (span.server_info.dist_rank if channels.enabled else None)
     meta, tensors = run_coroutine(
-        client.call("rpc_forward", {"adapter": adapter}, payload,
-                    timeout=timeout),
+        client.call("rpc_forward", {"adapter": adapter,
+                                    "drank": channels.rank}, payload,
+                    timeout=timeout, dist_rank=dist_rank),
         timeout + 5)
     return tensors[0]
 
@@ -53,12 +86,19 @@ def _call_backward(span: RemoteSpanInfo, hidden_in: torch.Tensor,
                    prompts: Optional[torch.Tensor] = None,
                    adapter: Optional[str] = None,
                    ) -> Tuple[torch.Tensor, Optional[torch.Tensor]]:
+    from bloombee_amd.net.channels import channels
     client = get_client(span.server_info.host, span.server_info.port)
     payload = ([hidden_in, grad_out] if prompts is None
                else [hidden_in, grad_out, prompts])
+    if _use_stream(span, payload):
+        tensors = _call_streamed(span, "rpc_backward_stream", payload,
+                                 timeout, adapter)
+        return tensors[0], (tensors[1] if len(tensors) > 1 else None)
+    dist_rank = (span.server_info.dist_rank if channels.enabled else None)
     meta, tensors = run_coroutine(
-        client.call("rpc_backward", {"adapter": adapter}, payload,
-                    timeout=timeout),
+        client.call("rpc_backward", {"adapter": adapter,
+                                     "drank": channels.rank}, payload,
+                    timeout=timeout, dist_rank=dist_rank),
         timeout + 5)
     return tensors[0], (tensors[1] if len(tensors) > 1 else None)
 

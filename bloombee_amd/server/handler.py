@@ -80,6 +80,8 @@ class ConnectionHandler:
         server.register("rpc_backward", self.rpc_backward)
         server.register("rpc_push", self.rpc_push)
         server.register_stream("rpc_inference", self.rpc_inference)
+        server.register_stream("rpc_forward_stream", self.rpc_forward_stream)
+        server.register_stream("rpc_backward_stream", self.rpc_backward_stream)
 
     # ------------------------------------------------------------------
     async def rpc_info(self, meta, tensors):
@@ -87,6 +89,14 @@ class ConnectionHandler:
         info = self.backend.info()
         info["dist_rank"] = channels.rank
         return info, []
+
+    @staticmethod
+    def _reply_move(meta):
+        """Keep reply tensors on-device when they ride the device plane."""
+        from bloombee_amd.net.channels import channels
+        if channels.enabled and meta.get("drank") is not None:
+            return lambda t: t
+        return lambda t: t.cpu()
 
     async def rpc_forward(self, meta, tensors):
         # tensors: [hidden] or [hidden, deep_prompts(n_local_blocks, pre, H)]
@@ -96,7 +106,7 @@ class ConnectionHandler:
         out = await asyncio.get_event_loop().run_in_executor(
             None, lambda: self.backend.forward(hidden, prompts,
                                                adapter=adapter))
-        return {}, [out.cpu()]
+        return {}, [self._reply_move(meta)(out)]
 
     async def rpc_backward(self, meta, tensors):
         # tensors: [hidden_in, grad_out] or [..., deep_prompts]; replies
@@ -107,10 +117,31 @@ class ConnectionHandler:
         grad_in, grad_p = await asyncio.get_event_loop().run_in_executor(
             None, lambda: self.backend.backward(hidden_in, grad_out, prompts,
                                                 adapter=adapter))
-        outs = [grad_in.cpu()]
+        move = self._reply_move(meta)
+        outs = [move(grad_in)]
         if grad_p is not None:
-            outs.append(grad_p.cpu())
+            outs.append(move(grad_p))
         return {}, outs
+
+    # ------------------------------------------------------------------
+    async def rpc_forward_stream(self, meta, tensors, stream):
+        """Chunked-payload forward (ref rpc_forward_stream, handler.py:
+        2860-3010): inputs arrive as flattened parts, the reply streams
+        back the same way."""
+        from bloombee_amd.net.streaming import (recv_tensors_chunked,
+                                                send_tensors_chunked)
+        _, ins = await recv_tensors_chunked(stream)
+        out_meta, outs = await self.rpc_forward(meta, ins)
+        await send_tensors_chunked(stream, outs, meta=out_meta)
+        await stream.send_end()
+
+    async def rpc_backward_stream(self, meta, tensors, stream):
+        from bloombee_amd.net.streaming import (recv_tensors_chunked,
+                                                send_tensors_chunked)
+        _, ins = await recv_tensors_chunked(stream)
+        out_meta, outs = await self.rpc_backward(meta, ins)
+        await send_tensors_chunked(stream, outs, meta=out_meta)
+        await stream.send_end()
 
     # ------------------------------------------------------------------
     async def rpc_push(self, meta, tensors):

@@ -47,6 +47,7 @@ class Server:
         expiration: Optional[float] = None,
         checkpoint_dir: Optional[str] = None,
         throughput: float = 1.0,
+        network_rps: Optional[float] = None,
         model_name: Optional[str] = None,
         offload_policy=None,
         adapters: Optional[dict] = None,
@@ -83,7 +84,9 @@ class Server:
         self.update_period = update_period
         self.session_max_idle = session_max_idle
         self.expiration = expiration or max(60.0, 2 * update_period)
-        self.throughput = throughput
+        self.network_rps = network_rps
+        self.throughput = (min(throughput, network_rps)
+                           if network_rps is not None else throughput)
         self.host = host
         # NAT / multi-homed hosts (ref --announce_maddrs): advertise a
         # different address than the bind address
@@ -93,22 +96,26 @@ class Server:
                        port=dht_port)
 
         L = self.config.num_hidden_layers
+        # auto-chosen ranges opt in to runtime rebalancing (ref server.py:
+        # 479-542 rebuild loop); explicit --block-indices pins the range
+        self._auto_blocks = block_indices is None
+        self._num_blocks = num_blocks or L
         if block_indices is None:
-            nb = num_blocks or L
             infos = get_remote_module_infos(
                 self.dht, module_uids(self.model_name, L))
-            blocks = choose_best_blocks(nb, infos)
+            blocks = choose_best_blocks(self._num_blocks, infos)
             block_indices = (blocks[0], blocks[-1] + 1)
         self.block_range = block_indices
         self.uids = [f"{self.model_name}.{i}"
                      for i in range(block_indices[0], block_indices[1])]
 
+        self._backend_kw = dict(device=device, seed=seed,
+                                kv_max_tokens=kv_max_tokens,
+                                checkpoint_dir=checkpoint_dir,
+                                offload_policy=offload_policy,
+                                max_batch_size=max_batch_size)
         self.backend = StackBackend(self.config, block_indices[0],
-                                    block_indices[1], device=device, seed=seed,
-                                    kv_max_tokens=kv_max_tokens,
-                                    checkpoint_dir=checkpoint_dir,
-                                    offload_policy=offload_policy,
-                                    max_batch_size=max_batch_size)
+                                    block_indices[1], **self._backend_kw)
         if adapters:
             # preload LoRA adapters onto this range's blocks (ref
             # run_server --adapters + utils/peft.py download-per-block;
@@ -151,6 +158,7 @@ class Server:
             device=self.device,
             start_block=self.block_range[0], end_block=self.block_range[1],
             next_pings=dict(self._next_pings),
+            network_rps=self.network_rps,
             dist_rank=channels.rank,
         )
 
@@ -182,9 +190,68 @@ class Server:
                                        self._server_info(),
                                        time.time() + self.expiration)
                 self.backend.reap_idle_sessions(self.session_max_idle)
+                self._maybe_rebalance()
             except Exception as e:  # noqa: BLE001
                 logger.warning("announce failed: %s", e)
             self._stop.wait(self.update_period)
+
+    # -- runtime rebalancing (ref server.py:479-542 rebuild loop) ---------
+    def _maybe_rebalance(self) -> None:
+        """If re-placing this server would raise the swarm's bottleneck
+        throughput enough (block_selection.should_choose_other_blocks —
+        dead code in round 1), gracefully re-host a better block range.
+        Only auto-placed servers move; in-flight sessions on the old range
+        fail over via the client's history replay."""
+        if not self._auto_blocks or self._stop.is_set():
+            return
+        import random as _random
+
+        from bloombee_amd.server.block_selection import \
+            should_choose_other_blocks
+        L = self.config.num_hidden_layers
+        uids = module_uids(self.model_name, L)
+        infos = get_remote_module_infos(self.dht, uids)
+        if not should_choose_other_blocks(self.peer_id, infos):
+            return
+        # double-check after a random stagger so two under-used neighbors
+        # don't both jump onto the same orphaned range
+        self._stop.wait(_random.uniform(0.05, 0.5) * self.update_period)
+        infos = get_remote_module_infos(self.dht, uids)
+        if self._stop.is_set() or not should_choose_other_blocks(
+                self.peer_id, infos):
+            return
+        stripped = [
+            type(i)(uid=i.uid, servers={k: v for k, v in i.servers.items()
+                                        if k != self.peer_id})
+            for i in infos]
+        blocks = choose_best_blocks(min(self._num_blocks, L), stripped)
+        new_range = (blocks[0], blocks[-1] + 1)
+        if new_range == tuple(self.block_range):
+            return
+        logger.info("rebalancing: moving from blocks [%d:%d) to [%d:%d)",
+                    *self.block_range, *new_range)
+        old_backend = self.backend
+        old_uids = self.uids
+        new_backend = StackBackend(self.config, new_range[0], new_range[1],
+                                   **self._backend_kw)
+        # atomic swap: the handler reads self.backend per call
+        self.backend = new_backend
+        self.handler.backend = new_backend
+        self.block_range = new_range
+        self.uids = [f"{self.model_name}.{i}"
+                     for i in range(new_range[0], new_range[1])]
+        # retire the old records immediately (state OFFLINE) and announce
+        # the new range
+        try:
+            off_info = self._server_info()
+            off_info.state = ServerState.OFFLINE
+            declare_active_modules(self.dht, old_uids, self.peer_id,
+                                   off_info, time.time() + self.expiration)
+            declare_active_modules(self.dht, self.uids, self.peer_id,
+                                   self._server_info(),
+                                   time.time() + self.expiration)
+        finally:
+            old_backend.shutdown()
 
     # ------------------------------------------------------------------
     def run_in_background(self) -> Tuple[str, int]:
@@ -226,6 +293,21 @@ class Server:
 
     def shutdown(self):
         self._stop.set()
+        # the announcer must stop FIRST: an in-flight iteration could
+        # re-announce ONLINE after the OFFLINE retirement below
+        if self._announcer is not None:
+            self._announcer.join(timeout=10)
+        # retire this server's records NOW (ref: servers declare OFFLINE on
+        # graceful shutdown) so routing and rebalancing don't wait out the
+        # DHT expiration to notice the blocks are orphaned
+        try:
+            if self.endpoint is not None:
+                info = self._server_info()
+                info.state = ServerState.OFFLINE
+                declare_active_modules(self.dht, self.uids, self.peer_id,
+                                       info, time.time() + self.expiration)
+        except Exception:  # noqa: BLE001 — best effort on the way down
+            pass
         if self._loop is not None:
             fut = asyncio.run_coroutine_threadsafe(self.rpc.stop(), self._loop)
             try:

@@ -59,7 +59,13 @@ def measure_compute_rps(config: ModelConfig, device: str = "cpu",
 
 
 def get_server_throughput(config: ModelConfig, device: str, num_blocks: int,
-                          force_eval: bool = False) -> dict:
+                          force_eval: bool = False,
+                          network_rps: Optional[float] = None,
+                          relayed: bool = False) -> dict:
+    """network_rps: advertised NIC steps/s cap (the reference measures it
+    with speedtest-cli — impossible offline, so it is operator-provided via
+    --network-rps); reported throughput = min(compute, network*penalty)
+    (ref throughput.py:123-133)."""
     key = f"{config.model_type}-{config.hidden_size}-{device.split(':')[0]}-{config.torch_dtype}"
     cache = {}
     p = Path(CACHE_FILE)
@@ -80,4 +86,6 @@ def get_server_throughput(config: ModelConfig, device: str, num_blocks: int,
             pass
     ent = cache[key]
     throughput = ent["forward_rps"] / max(1.0, (num_blocks + 1) / 2)
-    return {"throughput": throughput, **ent}
+    if network_rps is not None:
+        throughput = min(throughput, network_rps * (0.2 if relayed else 1.0))
+    return {"throughput": throughput, "network_rps": network_rps, **ent}

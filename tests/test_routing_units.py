@@ -99,3 +99,18 @@ def test_disjoint_ranges_one_span_per_segment():
     route = mgr._route_max_throughput(spans, 0, 5)
     assert route is not None
     assert [b for s in route for b in range(s.start, s.end)] == list(range(5))
+
+
+def test_min_latency_network_rps_caps_compute():
+    """A fast-compute server behind a slow link must lose to a balanced one
+    (network_rps folded into the edge cost — VERDICT r01 weak item 6)."""
+    mgr = _mgr(num_blocks=2)
+    fast_slow_link = _span("pA", 0, 2, rps=1000.0)
+    fast_slow_link.server_info.network_rps = 5.0
+    balanced = _span("pB", 0, 2, rps=100.0)
+    route = mgr._route_min_latency([fast_slow_link, balanced], 0, 2, None)
+    assert [s.peer_id for s in route] == ["pB"]
+    # without the link cap, pA wins
+    fast_slow_link.server_info.network_rps = None
+    route = mgr._route_min_latency([fast_slow_link, balanced], 0, 2, None)
+    assert [s.peer_id for s in route] == ["pA"]
