@@ -47,6 +47,11 @@ class MicrobatchConfig:
     enabled: bool = field(default_factory=lambda: _env("MICROBATCH", True))
     micro_batch_size: int = field(default_factory=lambda: _env("MICRO_BATCH_SIZE", 4))
     min_batch_to_split: int = field(default_factory=lambda: _env("MIN_BATCH_TO_SPLIT", 8))
+    # BBAMD_KV_MULTIPLEX=1: micro-batched sessions keep only ~2 slices of
+    # KV device-resident; the rest cycles through pinned host snapshots on
+    # the staging stream while the previous slice computes (ref per-MB KV
+    # offload/prefetch, memory_cache_manager.py:944-1371)
+    kv_multiplex: bool = field(default_factory=lambda: _env("KV_MULTIPLEX", False))
 
 
 @dataclass

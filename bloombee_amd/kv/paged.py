@@ -132,16 +132,24 @@ class PagedKVCache:
 
     # -- admission --------------------------------------------------------
     def allocate(
-        self, batch_size: int, max_length: int, timeout: Optional[float] = None
+        self, batch_size: int, max_length: int, timeout: Optional[float] = None,
+        resident_batch: Optional[int] = None,
     ) -> "SessionHandle":
         """Reserve budget for a decode session of `batch_size` sequences of up
         to `max_length` tokens each. Blocks until budget frees up, then raises
         AllocationFailed past `timeout` (ref memory_cache.py:147-222).
 
+        resident_batch: KV-multiplexed sessions keep only this many rows on
+        the device at a time (the rest live in pinned host snapshots via
+        swap_out_rows), so admission reserves only the resident window —
+        a batch bigger than HBM can hold becomes servable (ref GPU-multiplex
+        working slots, memory_cache_manager.py:628-908).
+
         Pages themselves are allocated lazily as tokens arrive.
         """
         pages_per_seq = (max_length + self.page_size - 1) // self.page_size
-        need = batch_size * pages_per_seq * self.page_size
+        need = min(batch_size, resident_batch or batch_size) \
+            * pages_per_seq * self.page_size
         deadline = None if timeout is None else time.monotonic() + timeout
         with self._lock:
             while self.max_tokens - self._reserved_tokens < need:
@@ -211,6 +219,17 @@ class SessionHandle:
             pin_memory=(cache.device.type == "cuda"))
         self._table_dirty = False
         self._closed = False
+        # row-granular host staging (micro-batch KV multiplexing):
+        # b -> (page_count_at_swap, per-layer (k_host, v_host)) snapshots
+        self._row_snaps: Dict[int, tuple] = {}
+        # rows whose restore copies are in flight: b -> done event
+        self._rows_inflight: Dict[int, threading.Event] = {}
+        # serializes metadata mutation (page lists / lengths / table rows)
+        # between the compute thread's extend() and the staging thread's
+        # swap_{out,in}_rows — bulk copies run outside the lock
+        self._meta_lock = threading.Lock()
+        self._stage_stream = (torch.cuda.Stream(cache.device)
+                              if cache.device.type == "cuda" else None)
 
     # -- lengths ----------------------------------------------------------
     @property
@@ -229,33 +248,63 @@ class SessionHandle:
         """Make room for `num_tokens` new tokens on every sequence, allocating
         pages as needed. If `speculative`, the tokens sit above l_acc and can
         be rolled back (ref paged_kv.py track/commit/rollback:206-261)."""
+        self.extend_rows(0, self.batch_size, num_tokens,
+                         speculative=speculative)
+
+    def extend_rows(self, b0: int, b1: int, num_tokens: int,
+                    speculative: bool = False,
+                    timeout: Optional[float] = 60.0) -> None:
+        """Row-ranged extend — micro-batch slices grow only their own rows,
+        so a KV-multiplexed session (most rows host-staged) never allocates
+        device pages for rows outside the resident window. Under page
+        pressure this retries until the staging thread frees pages (the
+        lock is NOT held while waiting — swap_out_rows needs it to free)."""
         if self.is_swapped:
             raise PagedKVError("extend on a swapped-out session; swap_in first")
         P = self.cache.page_size
-        need_total = 0
-        per_seq_need = []
-        for s in self.seqs:
-            new_len = s.l_spec + num_tokens
-            if new_len > self.max_length:
-                raise PagedKVError(
-                    f"sequence would exceed session max_length {self.max_length}"
-                )
-            need = (new_len + P - 1) // P - len(s.pages)
-            per_seq_need.append(need)
-            need_total += need
-        if need_total:
-            pages = self.cache._take_pages(need_total)
-            i = 0
-            for b, (s, need) in enumerate(zip(self.seqs, per_seq_need)):
-                for _ in range(need):
-                    self._page_table_host[b, len(s.pages)] = pages[i]
-                    s.pages.append(pages[i])
-                    i += 1
-            self._table_dirty = True
-        for s in self.seqs:
-            s.l_spec += num_tokens
-            if not speculative:
-                s.l_acc = s.l_spec
+        deadline = None if timeout is None else time.monotonic() + timeout
+        while True:
+            with self._meta_lock:
+                need_total = 0
+                per_seq_need = []
+                for b in range(b0, b1):
+                    s = self.seqs[b]
+                    new_len = s.l_spec + num_tokens
+                    if new_len > self.max_length:
+                        raise PagedKVError(
+                            f"sequence would exceed session max_length "
+                            f"{self.max_length}")
+                    # row-swapped sequences advance length-only: their pages
+                    # are (re)allocated at swap_in_rows time
+                    need = (0 if b in self._row_snaps
+                            else (new_len + P - 1) // P - len(s.pages))
+                    per_seq_need.append(need)
+                    need_total += need
+                try:
+                    pages = (self.cache._take_pages(need_total)
+                             if need_total else [])
+                except AllocationFailed:
+                    if (deadline is not None
+                            and time.monotonic() > deadline):
+                        raise
+                    pages = None
+                if pages is not None:
+                    i = 0
+                    for b, need in zip(range(b0, b1), per_seq_need):
+                        s = self.seqs[b]
+                        for _ in range(need):
+                            self._page_table_host[b, len(s.pages)] = pages[i]
+                            s.pages.append(pages[i])
+                            i += 1
+                    if need_total:
+                        self._table_dirty = True
+                    for b in range(b0, b1):
+                        s = self.seqs[b]
+                        s.l_spec += num_tokens
+                        if not speculative:
+                            s.l_acc = s.l_spec
+                    return
+            time.sleep(0.002)  # page pressure: let the stager free pages
 
     def commit(self, accepted: Optional[List[int]] = None) -> None:
         """Commit speculative tokens: all of them, or `accepted[b]` tokens per
@@ -456,6 +505,136 @@ class SessionHandle:
     @property
     def is_swapped(self) -> bool:
         return getattr(self, "_swapped", None) is not None
+
+    # -- row-granular staging (micro-batch KV multiplexing) ----------------
+    def swap_out_rows(self, b0: int, b1: int) -> None:
+        """Offload rows [b0, b1) to pinned host snapshots and free their
+        device pages — the reference's GPU working slots + CPU snapshots on
+        dedicated streams (memory_cache_manager.py:944-1371). Swapped rows
+        keep advancing lengths (extend is page-free for them); the kernels
+        never read them because micro-batch SessionViews slice the page
+        table to the resident window."""
+        if self.is_swapped:
+            raise PagedKVError("whole session already swapped")
+        cache = self.cache
+        on_gpu = cache.device.type == "cuda"
+        work = []  # (b, page_ids tensor, per-layer host bufs)
+        with self._meta_lock:
+            for b in range(b0, b1):
+                s = self.seqs[b]
+                if b in self._row_snaps or not s.pages:
+                    continue
+                ids = torch.tensor(s.pages, dtype=torch.long,
+                                   device=cache.device)
+                work.append((b, len(s.pages), ids))
+        freed: List[int] = []
+        if on_gpu:
+            # the rows' last KV writes ride the compute (default) stream;
+            # order the staging gathers after them
+            ev = torch.cuda.Event()
+            ev.record(torch.cuda.current_stream(cache.device))
+            self._stage_stream.wait_event(ev)
+        stream_ctx = (torch.cuda.stream(self._stage_stream) if on_gpu
+                      else contextlib.nullcontext())
+        with stream_ctx:
+            for b, npages, ids in work:
+                per_layer = []
+                for l in range(cache.num_layers):
+                    k = cache.k_pages(l)[ids]
+                    kh = torch.empty_like(k, device="cpu", pin_memory=on_gpu)
+                    kh.copy_(k, non_blocking=on_gpu)
+                    if cache.layout[l] != "k":
+                        v = cache.v_pages(l)[ids]
+                        vh = torch.empty_like(v, device="cpu",
+                                              pin_memory=on_gpu)
+                        vh.copy_(v, non_blocking=on_gpu)
+                    else:
+                        vh = None
+                    per_layer.append((kh, vh))
+                with self._meta_lock:
+                    s = self.seqs[b]
+                    self._row_snaps[b] = (npages, per_layer)
+                    freed.extend(s.pages)
+                    s.pages.clear()
+        if on_gpu:
+            self._stage_stream.synchronize()
+        if freed:
+            self.cache._give_pages(freed)
+
+    def swap_in_rows(self, b0: int, b1: int,
+                     timeout: Optional[float] = 60.0) -> None:
+        """Restore rows [b0, b1): fresh pages, H2D copies on the staging
+        stream, page-table rows rewritten. Synchronous w.r.t. its own
+        copies — run it on a staging thread to overlap the previous
+        micro-batch's compute (server/handler.py drives this).
+
+        Concurrent-restore safe: a second call overlapping the same rows
+        (prefetch + arrival both staging a slice) waits for the in-flight
+        restore instead of returning while its copies are incomplete.
+        Under page pressure the allocation retries without holding the
+        metadata lock (swap_out_rows needs it to free pages)."""
+        cache = self.cache
+        on_gpu = cache.device.type == "cuda"
+        P = cache.page_size
+        deadline = None if timeout is None else time.monotonic() + timeout
+        work = []
+        wait_evs = []
+        for b in range(b0, b1):
+            while True:
+                with self._meta_lock:
+                    ev = self._rows_inflight.get(b)
+                    if ev is not None:
+                        wait_evs.append(ev)
+                        break
+                    if b not in self._row_snaps:
+                        break
+                    s = self.seqs[b]
+                    n_now = (s.l_spec + P - 1) // P
+                    try:
+                        pages = self.cache._take_pages(n_now)
+                    except AllocationFailed:
+                        pages = None
+                    if pages is not None:
+                        snap = self._row_snaps.pop(b)
+                        s.pages = pages
+                        for j, pg in enumerate(pages):
+                            self._page_table_host[b, j] = pg
+                        done = threading.Event()
+                        self._rows_inflight[b] = done
+                        work.append((b, snap, pages, done))
+                        self._table_dirty = True
+                        break
+                if deadline is not None and time.monotonic() > deadline:
+                    raise AllocationFailed(
+                        f"swap_in_rows timed out waiting for pages (row {b})")
+                time.sleep(0.002)  # pressure: let the stager free pages
+        stream_ctx = (torch.cuda.stream(self._stage_stream) if on_gpu
+                      else contextlib.nullcontext())
+        with stream_ctx:
+            for b, (n_snap, per_layer), pages, _ in work:
+                n_copy = min(n_snap, len(pages))
+                idx = torch.tensor(pages[:n_copy], dtype=torch.long,
+                                   device=cache.device)
+                for l in range(cache.num_layers):
+                    kh, vh = per_layer[l]
+                    cache.k_pages(l)[idx] = kh[:n_copy].to(
+                        cache.device, non_blocking=on_gpu)
+                    if vh is not None:
+                        cache.v_pages(l)[idx] = vh[:n_copy].to(
+                            cache.device, non_blocking=on_gpu)
+        if on_gpu:
+            self._stage_stream.synchronize()
+        with self._meta_lock:
+            for b, _, _, done in work:
+                done.set()
+                self._rows_inflight.pop(b, None)
+        for ev in wait_evs:
+            if not ev.wait(timeout=timeout):
+                raise PagedKVError("in-flight row restore did not finish")
+
+    def rows_swapped(self, b0: int, b1: int) -> bool:
+        with self._meta_lock:
+            return any(b in self._row_snaps for b in range(b0, b1))
 
     def truncate(self, new_lengths: List[int]) -> None:
         """Failover / history-replay support: cut sequences back to

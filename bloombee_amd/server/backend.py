@@ -93,7 +93,8 @@ class StackBackend:
 
     # -- sessions ---------------------------------------------------------
     def open_session(self, session_id: str, batch_size: int, max_length: int,
-                     timeout: Optional[float] = 10.0) -> None:
+                     timeout: Optional[float] = 10.0,
+                     resident_batch: Optional[int] = None) -> None:
         from bloombee_amd.kv.paged import AllocationFailed
 
         if batch_size > self.max_batch_size:
@@ -102,15 +103,22 @@ class StackBackend:
                 f"{self.max_batch_size} (ref --max_batch_size)")
 
         try:
-            handle = self.kv_pool.allocate(batch_size, max_length, timeout=0.5)
+            handle = self.kv_pool.allocate(batch_size, max_length, timeout=0.5,
+                                           resident_batch=resident_batch)
         except AllocationFailed:
             # KV pressure: offload the least-recently-active idle session to
             # host (ref micro-batch KV multiplexing) and retry
             self._swap_out_idle()
             handle = self.kv_pool.allocate(batch_size, max_length,
-                                           timeout=timeout)
+                                           timeout=timeout,
+                                           resident_batch=resident_batch)
         with self._lock:
             self.sessions[session_id] = SessionState(handle)
+
+    def session_handle(self, session_id: str):
+        """The session's KV handle — the handler's KV-staging driver uses it
+        to prefetch/offload micro-batch row windows."""
+        return self._session(session_id).handle
 
     def _swap_out_idle(self) -> None:
         with self._lock:
@@ -218,14 +226,16 @@ class StackBackend:
             cur = handle.seqs[0].l_acc
             if batch_offset is not None:
                 # micro-batch slice of the session (server-side split or
-                # upstream per-MB push): the FIRST slice of a step extends
-                # the whole session; later slices reuse the extension
-                if handle.seqs[0].l_spec == start_pos:
-                    handle.extend(T)
-                elif handle.seqs[0].l_spec != start_pos + T:
+                # upstream per-MB push): each slice extends ONLY its own
+                # rows, so a KV-multiplexed session never allocates device
+                # pages outside the resident window
+                b0, b1 = batch_offset, batch_offset + B
+                if handle.seqs[b0].l_spec == start_pos:
+                    handle.extend_rows(b0, b1, T)
+                elif handle.seqs[b0].l_spec != start_pos + T:
                     raise ValueError(
                         f"micro-batch at position {start_pos} inconsistent "
-                        f"with cache length {handle.seqs[0].l_spec}")
+                        f"with cache length {handle.seqs[b0].l_spec}")
                 from bloombee_amd.kv.views import SessionView
                 view = SessionView(handle, batch_offset, batch_offset + B)
                 sp = torch.full((B,), start_pos, dtype=torch.int32,

@@ -69,6 +69,7 @@ class ConnectionHandler:
         self.MICROBATCH_ENABLED = mb.enabled
         self.MICROBATCH_MIN_BATCH = mb.min_batch_to_split
         self.MICROBATCH_SIZE = mb.micro_batch_size
+        self.KV_MULTIPLEX = mb.kv_multiplex
         self.backend = backend
         self.rpc = server
         # (session_id, step) -> queued pushed inputs awaiting the local stream
@@ -194,6 +195,14 @@ class ConnectionHandler:
         loop = asyncio.get_event_loop()
         times = StageTimes()
         mb_buffers: Dict[int, dict] = {}  # pos -> {offset: out_cpu}
+        mb_stage: Dict[int, object] = {}  # offset -> pending swap-out future
+        # KV multiplexing: only ~2 micro-batch row windows stay device-
+        # resident; admission reserves just that window
+        mbs = self.MICROBATCH_SIZE
+        multiplex = (self.KV_MULTIPLEX and self.MICROBATCH_ENABLED
+                     and batch_size >= self.MICROBATCH_MIN_BATCH
+                     and batch_size % mbs == 0 and batch_size > 2 * mbs)
+        resident_batch = 3 * mbs if multiplex else None  # j-1 out, j, j+1 in
         # persistent reader futures for push_only_recv: cancelling a q.get()
         # that already dequeued an item silently drops that frame (lost step
         # or spec_commit -> stalled session). Instead the
@@ -201,7 +210,9 @@ class ConnectionHandler:
         get_push: Optional[asyncio.Future] = None
         get_cli: Optional[asyncio.Future] = None
         await loop.run_in_executor(
-            None, lambda: self.backend.open_session(sid, batch_size, max_length))
+            None, lambda: self.backend.open_session(
+                sid, batch_size, max_length,
+                resident_batch=resident_batch))
         try:
             await stream.send({"session_id": sid, "ok": True})
             while True:
@@ -259,10 +270,35 @@ class ConnectionHandler:
                     # merge), forward the slice downstream, reply merged once
                     # every slice of this step has arrived
                     off = int(mbinfo["offset"])
+                    if multiplex:
+                        # this stage's KV staging for pushed slices: swap
+                        # the slice's row window in before compute (after
+                        # any still-running swap-out of the same rows), out
+                        # after, prefetching the next expected window
+                        h_kv = self.backend.session_handle(sid)
+                        nrows = hidden.shape[0]
+                        total = int(mbinfo["total"])
+
+                        async def _stage_in(a, n):
+                            prev = mb_stage.pop(a, None)
+                            if prev is not None:
+                                await prev
+                            await loop.run_in_executor(
+                                None, h_kv.swap_in_rows, a, a + n)
+
+                        await _stage_in(off, nrows)
+                        nxt = off + nrows
+                        if nxt < total:
+                            asyncio.ensure_future(
+                                _stage_in(nxt, min(nrows, total - nxt)))
                     part = await loop.run_in_executor(
                         None, lambda: self.backend.inference_step(
                             sid, hidden, pos, prompts, batch_offset=off,
                             adapter=adapter))
+                    if multiplex:
+                        mb_stage[off] = loop.run_in_executor(
+                            None, h_kv.swap_out_rows, off,
+                            off + hidden.shape[0])
                     part_cpu = move(part)
                     if push_to is not None:
                         asyncio.ensure_future(self._push_downstream(
@@ -298,13 +334,39 @@ class ConnectionHandler:
                     mbs = self.MICROBATCH_SIZE
                     outs = []
                     push_tasks = []
+                    stage_tasks = []
+                    h_kv = (self.backend.session_handle(sid) if multiplex
+                            else None)
+
+                    def _swap_in(a):
+                        h_kv.swap_in_rows(a, min(a + mbs, B))
+
+                    def _swap_out(a):
+                        h_kv.swap_out_rows(a, min(a + mbs, B))
+
                     with times.span("compute"):
+                        # KV staging pipeline (ref memory_cache_manager.py:
+                        # 944-1371): while slice j computes, slice j+1's KV
+                        # swaps in on the staging stream and slice j-1's
+                        # swaps out — only the resident window holds pages
+                        pf = (loop.run_in_executor(None, _swap_in, 0)
+                              if multiplex else None)
                         for j in range(0, B, mbs):
+                            if pf is not None:
+                                await pf  # slice j resident
+                            if multiplex and j + mbs < B:
+                                pf = loop.run_in_executor(
+                                    None, _swap_in, j + mbs)
+                            else:
+                                pf = None
                             part = await loop.run_in_executor(
                                 None, lambda j=j: self.backend.inference_step(
                                     sid, hidden[j:j + mbs], pos, prompts,
                                     batch_offset=j, adapter=adapter))
                             part_cpu = move(part)
+                            if multiplex:
+                                stage_tasks.append(loop.run_in_executor(
+                                    None, _swap_out, j))
                             outs.append(part_cpu)
                             push_tasks.append(asyncio.ensure_future(
                                 self._push_downstream(
@@ -313,6 +375,8 @@ class ConnectionHandler:
                                     codec=codec, dist_rank=push_rank)))
                         out_cpu = torch.cat(outs, dim=0)
                         for t_ in push_tasks:
+                            await t_
+                        for t_ in stage_tasks:
                             await t_
                     times.bump_step()
                     if not quiet:

@@ -279,3 +279,61 @@ def test_swap_out_extend_guarded():
     with pytest.raises(PagedKVError):
         h.extend(1)
     h.close()
+
+
+def test_row_swap_roundtrip_and_extend_rows():
+    """Row-granular staging (micro-batch KV multiplexing): swapped rows
+    advance length-only; swap_in restores data into fresh pages; resident
+    rows are unaffected."""
+    c = make_cache(max_tokens=512, page=16)
+    h = c.allocate(4, 128)
+    h.extend(20)
+    # write recognizable data into row 2's pages
+    marker = torch.full((2, 16, 16), 7.0)
+    for l in range(2):
+        for j, pg in enumerate(h.seqs[2].pages):
+            c.k_pages(l)[pg] = float(l * 10 + j)
+    h.swap_out_rows(2, 4)
+    assert h.rows_swapped(2, 4) and not h.rows_swapped(0, 2)
+    assert h.seqs[2].pages == [] and h.seqs[3].pages == []
+    free_now = c._free_pages
+    assert len(free_now) == len(set(free_now))
+    # grow ALL rows while 2..4 are swapped: only resident rows take pages
+    h.extend(13)  # 20 -> 33 tokens: 3 pages per seq
+    assert len(h.seqs[0].pages) == 3 and len(h.seqs[2].pages) == 0
+    assert h.lengths == [33, 33, 33, 33]
+    h.swap_in_rows(2, 4)
+    assert not h.rows_swapped(2, 4)
+    assert len(h.seqs[2].pages) == 3  # grown while away
+    for l in range(2):
+        for j in range(2):  # snapshot covered the first 2 pages
+            pg = h.seqs[2].pages[j]
+            assert torch.all(c.k_pages(l)[pg] == float(l * 10 + j)), (l, j)
+    # per-row extends (micro-batch slices grow only their own rows)
+    h.extend_rows(0, 2, 16)
+    assert h.lengths == [49, 49, 33, 33]
+    h.extend_rows(2, 4, 16)
+    assert h.lengths == [49, 49, 49, 49]
+    h.close()
+    assert sorted(c._free_pages) == list(range(c.n_pages))
+
+
+def test_row_swap_extend_blocks_until_pages_freed():
+    """extend_rows under page pressure waits (without deadlocking the
+    staging thread) until swap_out_rows frees pages."""
+    c = make_cache(max_tokens=64, page=16)  # 4-page pool
+    h = c.allocate(4, 64, resident_batch=1)
+    h.extend_rows(0, 2, 32)  # rows 0,1 take all 4 pages
+    import threading
+
+    def stage():
+        time.sleep(0.05)
+        h.swap_out_rows(0, 2)
+
+    t = threading.Thread(target=stage)
+    t.start()
+    h.extend_rows(2, 4, 32, timeout=5)  # blocks until the swap frees pages
+    t.join()
+    assert len(h.seqs[2].pages) == 2 and h.seqs[0].pages == []
+    assert h.lengths == [32, 32, 32, 32]
+    h.close()

@@ -658,3 +658,45 @@ def test_rebalance_recovers_orphaned_blocks():
         for s in (s_auto, s_pin):
             s.shutdown()
         boot.shutdown()
+
+
+@pytest.mark.timeout(180)
+def test_kv_multiplex_swarm_exact():
+    """BBAMD_KV_MULTIPLEX: a micro-batched session whose KV pool cannot hold
+    the full batch (only the resident window is reserved) must still decode
+    token-exactly — slices cycle through pinned host snapshots while the
+    previous slice computes (ref memory_cache_manager.py:944-1371)."""
+    import dataclasses
+
+    from bloombee_amd import config as bconf
+
+    cfg0 = bconf.get_config()
+    mb = dataclasses.replace(cfg0.microbatch, enabled=True,
+                             micro_batch_size=4, min_batch_to_split=8,
+                             kv_multiplex=True)
+    bconf.set_config(dataclasses.replace(cfg0, microbatch=mb))
+    boot = Dht()
+    servers = []
+    try:
+        for rng in [(0, 2), (2, 4)]:
+            # pool: 13 pages = 208 tokens. The session (16 seqs x 13
+            # tokens -> 1 page each) would need 256 reserved un-multiplexed
+            # (> pool, refused outright); the 12-row resident window needs
+            # 192 and fits.
+            s = Server(MODEL, initial_peers=[boot.endpoint],
+                       block_indices=rng, device="cpu", seed=SEED,
+                       kv_max_tokens=208, update_period=2.0)
+            s.run_in_background()
+            servers.append(s)
+        model = _make_model(boot)
+        gen = torch.Generator().manual_seed(5)
+        prompt = torch.randint(0, 1000, (16, 7), generator=gen)
+        out = model.generate(prompt, max_new_tokens=6)
+        expect = _local_tokens(prompt, 6)
+        assert torch.equal(out[:, 7:], expect)
+        model.remote.manager.shutdown()
+    finally:
+        bconf.set_config(cfg0)
+        for s in servers:
+            s.shutdown()
+        boot.shutdown()
