@@ -50,7 +50,14 @@ DEVINL float quad16_reduce_sum(float x) {
 // (the S-form needed 208 VGPR + 32 AGPR and sat at 2); launch_bounds keeps
 // the floor at 2 WGs/CU. Host-side n_split sizes the grid for ~1024
 // workgroups = 4 resident WGs on all 256 CUs (profiles/r01_st_attention.md).
-template <int D, int MAXG>
+// NT: KV tiles (32 positions each) processed per wave iteration. NT=2
+// doubles the K/V bytes in flight per wave — the PMC diagnosis at 4
+// waves/SIMD was latency-bound (SQ_WAIT ~70%, profiles/r01_st_attention.md)
+// and more outstanding loads is the lever; it also halves the softmax
+// shuffle + LDS-relayout overhead per position. Costs ~40 VGPRs (V
+// fragments + score regs for the second tile); host falls back to NT=1 for
+// short splits.
+template <int D, int MAXG, int NT>
 __global__ __launch_bounds__(256, 2) void attn_decode_mfma_kernel(
     const unsigned short* __restrict__ q,        // strided, see q_sb/q_sh
     const unsigned short* __restrict__ k_pages,  // (np, Hkv, P, D)
@@ -66,7 +73,7 @@ __global__ __launch_bounds__(256, 2) void attn_decode_mfma_kernel(
   constexpr int NKK = D / 32;   // QK^T k-slices
   constexpr int NDT = D / 16;   // PV d-tiles
   constexpr int NPT = DKVBLK / 16;  // position tiles per KV tile
-  constexpr int PROW_B = DKVBLK * 2 + 16;  // P^T LDS row pitch, 16B-aligned
+  constexpr int PROW_B = NT * DKVBLK * 2 + 16;  // P^T LDS row pitch
 
   const int bh = blockIdx.x;
   const int split = blockIdx.y;
@@ -118,74 +125,85 @@ __global__ __launch_bounds__(256, 2) void attn_decode_mfma_kernel(
   const long head_slab_k = (long)kvh * P * D;  // same bytes, both layouts
   const int tile0 = (max(c0, lo) - c0) / DKVBLK;  // window skip, tile-aligned
 
-  for (int tb = c0 + (tile0 + wave) * DKVBLK; tb < c1; tb += 4 * DKVBLK) {
-    // ---- V^T fragments for this tile: direct A-layout loads, issued FIRST
-    // so the QK^T phase covers their latency. A[row=d][k=pos]: lane (li -> d
-    // row, hi -> position octet); 8 consecutive positions at fixed d are
-    // contiguous in the d-major pool. tb is 32-aligned and P | 32, so each
-    // octet sits in one page.
-    const int vpos = tb + hi * 8;
-    int vpage = -1;
-    if (vpos < c1) vpage = page_table[b * maxp + vpos / P];
-    const long vrow = ((long)vpage * Hkv + kvh) * D;
-    bf16x8 vfrag[NDT];
+  // wave w owns NT adjacent tiles starting at tile w*NT, stride 4*NT tiles
+  for (int tb = c0 + (tile0 + wave * NT) * DKVBLK; tb < c1;
+       tb += 4 * NT * DKVBLK) {
+    // ---- V^T fragments for ALL NT tiles: direct A-layout loads, issued
+    // FIRST so the QK^T phase covers their latency. A[row=d][k=pos]: lane
+    // (li -> d row, hi -> position octet); 8 consecutive positions at fixed
+    // d are contiguous in the d-major pool. tb is 32-aligned and P | 32, so
+    // each octet sits in one page.
+    bf16x8 vfrag[NT][NDT];
 #pragma unroll
-    for (int n = 0; n < NDT; ++n) {
-      vfrag[n] = as_bf16x8(short8{});
-      if (vpage >= 0)
-        vfrag[n] = as_bf16x8(__builtin_nontemporal_load(
-            reinterpret_cast<const short8*>(
-                v_pages + (vrow + li + 16 * n) * P + (vpos % P))));
+    for (int t = 0; t < NT; ++t) {
+      const int vpos = tb + t * DKVBLK + hi * 8;
+      int vpage = -1;
+      if (vpos < c1) vpage = page_table[b * maxp + vpos / P];
+      const long vrow = ((long)vpage * Hkv + kvh) * D;
+#pragma unroll
+      for (int n = 0; n < NDT; ++n) {
+        vfrag[t][n] = as_bf16x8(short8{});
+        if (vpage >= 0)
+          vfrag[t][n] = as_bf16x8(__builtin_nontemporal_load(
+              reinterpret_cast<const short8*>(
+                  v_pages + (vrow + li + 16 * n) * P + (vpos % P))));
+      }
     }
 
     // ---- S^T = K Q^T : K A-fragments straight from the paged pool ----
     // A[row=pos][k=d]: lane li -> position row, hi -> d octet.
-    f32x4 st[NPT];
+    f32x4 st[NT][NPT];
 #pragma unroll
-    for (int n = 0; n < NPT; ++n) {
-      const int pos = tb + li + 16 * n;
-      const int cpos = min(pos, c1 - 1);
-      const int page = page_table[b * maxp + cpos / P];
-      const unsigned short* krow = k_pages + ((long)page * Hkv) * P * D +
-                                   head_slab_k + (long)(cpos % P) * D;
-      st[n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+    for (int t = 0; t < NT; ++t)
 #pragma unroll
-      for (int kk = 0; kk < NKK; ++kk) {
-        // KV is read once per decode step: nontemporal keeps L2 for the
-        // GEMM weight streams that follow in the same step
-        bf16x8 kfrag = as_bf16x8(__builtin_nontemporal_load(
-            reinterpret_cast<const short8*>(krow + hi * 8 + 32 * kk)));
-        st[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kfrag, qfrag[kk],
-                                                        st[n], 0, 0, 0);
+      for (int n = 0; n < NPT; ++n) {
+        const int pos = tb + t * DKVBLK + li + 16 * n;
+        const int cpos = min(pos, c1 - 1);
+        const int page = page_table[b * maxp + cpos / P];
+        const unsigned short* krow = k_pages + ((long)page * Hkv) * P * D +
+                                     head_slab_k + (long)(cpos % P) * D;
+        st[t][n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int kk = 0; kk < NKK; ++kk) {
+          // KV is read once per decode step: nontemporal keeps L2 for the
+          // GEMM weight streams that follow in the same step
+          bf16x8 kfrag = as_bf16x8(__builtin_nontemporal_load(
+              reinterpret_cast<const short8*>(krow + hi * 8 + 32 * kk)));
+          st[t][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              kfrag, qfrag[kk], st[t][n], 0, 0, 0);
+        }
       }
-    }
 
     // ---- mask + online softmax over positions (all in-lane + 2 shuffles).
-    // st[n][reg] = score at position tb + 16n + hi*4 + reg for head li.
-    float p[NPT][4];
+    // st[t][n][reg] = score at tb + t*32 + 16n + hi*4 + reg for head li.
+    float p[NT][NPT][4];
     float rm = NEG_BIG;
 #pragma unroll
-    for (int n = 0; n < NPT; ++n)
+    for (int t = 0; t < NT; ++t)
 #pragma unroll
-      for (int reg = 0; reg < 4; ++reg) {
-        const int kpos = tb + 16 * n + hi * 4 + reg;
-        float sv = st[n][reg] * sc2 + aslope * (float)kpos;
-        const bool dead = (kpos >= c1) | (kpos < lo);
-        sv = dead ? NEG_BIG : sv;
-        p[n][reg] = sv;
-        rm = fmaxf(rm, sv);
-      }
+      for (int n = 0; n < NPT; ++n)
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg) {
+          const int kpos = tb + t * DKVBLK + 16 * n + hi * 4 + reg;
+          float sv = st[t][n][reg] * sc2 + aslope * (float)kpos;
+          const bool dead = (kpos >= c1) | (kpos < lo);
+          sv = dead ? NEG_BIG : sv;
+          p[t][n][reg] = sv;
+          rm = fmaxf(rm, sv);
+        }
     rm = quad16_reduce_max(rm);
     const float mn = fmaxf(m2, rm);
     const float corr = fast_exp2(m2 - mn);
     float psum = 0.f;
 #pragma unroll
-    for (int n = 0; n < NPT; ++n)
+    for (int t = 0; t < NT; ++t)
 #pragma unroll
-      for (int reg = 0; reg < 4; ++reg) {
-        p[n][reg] = fast_exp2(p[n][reg] - mn);
-        psum += p[n][reg];
-      }
+      for (int n = 0; n < NPT; ++n)
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg) {
+          p[t][n][reg] = fast_exp2(p[t][n][reg] - mn);
+          psum += p[t][n][reg];
+        }
     psum = quad16_reduce_sum(psum);
     l = l * corr + psum;
     m2 = mn;
@@ -201,19 +219,25 @@ __global__ __launch_bounds__(256, 2) void attn_decode_mfma_kernel(
     // ---- P^T through LDS into B-fragment layout (head-major rows, so the
     // B-frag load of 8 consecutive positions at head li is one 16 B read)
 #pragma unroll
-    for (int n = 0; n < NPT; ++n)
+    for (int t = 0; t < NT; ++t)
 #pragma unroll
-      for (int reg = 0; reg < 4; ++reg)
-        *(unsigned short*)(p_lds + li * PROW_B + (16 * n + hi * 4 + reg) * 2) =
-            f2bf(p[n][reg]);
-    bf16x8 pfrag = as_bf16x8(
-        *reinterpret_cast<const short8*>(p_lds + li * PROW_B + hi * 16));
+      for (int n = 0; n < NPT; ++n)
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg)
+          *(unsigned short*)(p_lds + li * PROW_B +
+                             (t * DKVBLK + 16 * n + hi * 4 + reg) * 2) =
+              f2bf(p[t][n][reg]);
 
     // ---- acc^T += V^T P^T (A-fragments were loaded at tile start) ----
 #pragma unroll
-    for (int n = 0; n < NDT; ++n)
-      acc_o[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vfrag[n], pfrag,
-                                                         acc_o[n], 0, 0, 0);
+    for (int t = 0; t < NT; ++t) {
+      bf16x8 pfrag = as_bf16x8(*reinterpret_cast<const short8*>(
+          p_lds + li * PROW_B + t * DKVBLK * 2 + hi * 16));
+#pragma unroll
+      for (int n = 0; n < NDT; ++n)
+        acc_o[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vfrag[t][n], pfrag,
+                                                           acc_o[n], 0, 0, 0);
+    }
   }
 
   // ---- merge the 4 waves ----

@@ -205,6 +205,14 @@ static void attn_decode_launch(const torch::Tensor& q, const torch::Tensor& kp,
                                long q_sb, long q_sh, long out_sb, long out_sh) {
   dim3 grid(B * Hkv * nch, n_split);
   const int maxg = (G <= 4 && nch == 1) ? 4 : 16;
+  // paired-tile variant (NT=2): 2x KV bytes in flight per wave — the
+  // latency-bound fix (profiles/r01_st_attention.md SQ_WAIT diagnosis).
+  // BBAMD_ATTN_NT=1 reverts; D>=256 keeps NT=1 (VGPR budget).
+  static const int nt_env = [] {
+    const char* e = std::getenv("BBAMD_ATTN_NT");
+    return e ? std::atoi(e) : 2;
+  }();
+  const bool nt2 = (nt_env >= 2) && (D <= 128);
   auto launch_mfma = [&](auto mg) {
     if constexpr (D == 512) {
       // wide-head path: d split across the 4 waves (gemma-4 global layers)
@@ -214,8 +222,15 @@ static void attn_decode_launch(const torch::Tensor& q, const torch::Tensor& kp,
               ctx.data_ptr<int>(), alibi_ptr, bf_ptr_mut(out),
               pml.data_ptr<float>(), pacc.data_ptr<float>(), B, Hkv, G, nch, P,
               maxp, n_split, window, scale, q_sb, q_sh, out_sb, out_sh);
+    } else if (D <= 128 && nt2) {
+      attn_decode_mfma_kernel<D, decltype(mg)::value, 2>
+          <<<grid, 256, 0, cur_stream()>>>(
+              bf_ptr(q) + q_off, bf_ptr(kp), bf_ptr(vp), pt.data_ptr<int>(),
+              ctx.data_ptr<int>(), alibi_ptr, bf_ptr_mut(out),
+              pml.data_ptr<float>(), pacc.data_ptr<float>(), B, Hkv, G, nch, P,
+              maxp, n_split, window, scale, q_sb, q_sh, out_sb, out_sh);
     } else {
-      attn_decode_mfma_kernel<D, decltype(mg)::value>
+      attn_decode_mfma_kernel<D, decltype(mg)::value, 1>
           <<<grid, 256, 0, cur_stream()>>>(
               bf_ptr(q) + q_off, bf_ptr(kp), bf_ptr(vp), pt.data_ptr<int>(),
               ctx.data_ptr<int>(), alibi_ptr, bf_ptr_mut(out),
@@ -252,12 +267,17 @@ static torch::Tensor attn_decode_core(
   }
   int n_split = (int)n_split_req;
   if (n_split <= 0) {
-    // measured optimum (benchmarks/bench_kernels.py attn_decode): the S^T
-    // kernel runs at 118 VGPRs -> 4 workgroups/CU resident, so ~1024 total
-    // workgroups = full residency (B32/ctx2048: ns=4 57us vs ns=2 63us);
-    // more splits only add merge traffic
+    // measured optimum (benchmarks/bench_kernels.py attn_decode): size the
+    // grid for full residency — NT=1 runs 118 VGPRs -> 4 WGs/CU (1024 WGs),
+    // the paired-tile NT=2 variant 158 VGPRs -> 3 WGs/CU (768 WGs); more
+    // splits only add merge traffic
+    static const int nt_env0 = [] {
+      const char* e = std::getenv("BBAMD_ATTN_NT");
+      return e ? std::atoi(e) : 2;
+    }();
+    const long target = (nt_env0 >= 2 && D <= 128) ? 768 : 1024;
     n_split = (int)std::max<long>(
-        1, std::min<long>(32, 1024 / std::max(1, B * Hkv * nch)));
+        1, std::min<long>(32, target / std::max(1, B * Hkv * nch)));
   }
   const int maxg = (G <= 4 && nch == 1) ? 4 : 16;
   auto fopt = torch::TensorOptions().device(q.device()).dtype(at::kFloat);
