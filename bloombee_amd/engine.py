@@ -114,7 +114,7 @@ class LocalEngine:
     """
 
     def __init__(self, config_or_name, device="cpu", seed: int = 0,
-                 kv_max_tokens: int = 1 << 16):
+                 kv_max_tokens: int = 1 << 16, quantize_q4: bool = False):
         cfg = (config_or_name if isinstance(config_or_name, ModelConfig)
                else resolve_config(config_or_name))
         self.config = cfg
@@ -140,6 +140,13 @@ class LocalEngine:
             self.lm_head_w = (torch.randn(cfg.vocab_size, cfg.hidden_size, generator=gen)
                               .mul_(0.02).to(dt).to(device))
         self.stack = BlockStack(cfg, 0, cfg.num_hidden_layers, device=device, seed=seed)
+        if quantize_q4:
+            # 4-bit weight-compressed decode (w4_gemm.hip): ~4x less weight
+            # memory + HBM traffic. Used for cheap same-weights draft models
+            # (speculative decoding) and compressed serving.
+            for blk in self.stack.blocks:
+                if hasattr(blk, "quantize_weights_q4"):
+                    blk.quantize_weights_q4()
         self.kv_pool = self.stack.make_kv(kv_max_tokens)
 
     @torch.no_grad()

@@ -94,9 +94,38 @@ class LlamaBlock(torch.nn.Module):
                                     device=dev).mul_(std).to(w.dtype))
         return self
 
+    @torch.no_grad()
+    def quantize_weights_q4(self, keep_full: bool = False):
+        """Pack the four projection weights into 4-bit group codes
+        (quant4_pack layout); decode GEMMs then stream ~3.5x fewer bytes
+        through w4_gemm.hip (FlexGen-style compression ON the compute
+        path — the reference only compresses weights at rest,
+        flexgen_utils/compression.py:94-210). keep_full=False drops the
+        bf16 weights (4x memory; the block becomes inference-only and
+        prefill-shaped GEMMs dequantize on the fly)."""
+        self._w4 = {}
+        for name in ("qkv_w", "o_w", "gate_up_w", "down_w"):
+            w = getattr(self, name).detach()
+            packed, scale, zero = ops.quant4_pack(w)
+            N = w.shape[0]
+            self._w4[name] = (packed.reshape(N, -1).contiguous(),
+                              scale.reshape(N, -1).half().contiguous(),
+                              zero.reshape(N, -1).half().contiguous(), N)
+            if not keep_full:
+                getattr(self, name).data = torch.empty(
+                    0, dtype=w.dtype, device=w.device)
+        return self
+
     def _lin(self, x, w, name, **kw):
-        """ops.linear + optional active-LoRA delta (utils/peft.py)."""
-        y = ops.linear(x, w, **kw)
+        """ops.linear + optional active-LoRA delta (utils/peft.py).
+        Quantized blocks (quantize_weights_q4) route through the 4-bit
+        weight-stream kernel instead."""
+        w4 = getattr(self, "_w4", None)
+        if w4 is not None and name in w4:
+            packed, scale, zero, N = w4[name]
+            y = ops.linear_w4(x, packed, scale, zero, N, **kw)
+        else:
+            y = ops.linear(x, w, **kw)
         ld = getattr(self, "lora_delta", None)
         if ld is not None:
             d = ld(name, x)
@@ -146,8 +175,8 @@ class LlamaBlock(torch.nn.Module):
         # h2 = hidden + a fused into the post-attention norm
         h2, y = ops.rms_norm_residual(a, hidden, self.post_norm_w, cfg.rms_norm_eps)
         # residual add fused into the down-projection epilogue
-        return ops.linear(ops.swiglu(ops.linear(y, self.gate_up_w)), self.down_w,
-                          residual=h2)
+        return self._lin(ops.swiglu(self._lin(y, self.gate_up_w, "gate_up_w")),
+                         self.down_w, "down_w", residual=h2)
 
     # ------------------------------------------------------------------
     # training path (differentiable; full sequence, no KV cache)

@@ -17,6 +17,7 @@ from bloombee_amd.ops.interface import (  # noqa: F401
     kv_write,
     layer_norm,
     linear,
+    linear_w4,
     mfma_selftest,
     moe_gemm_grouped,
     quant4_pack,

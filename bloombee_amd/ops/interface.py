@@ -325,6 +325,32 @@ def moe_gemm_grouped(A: torch.Tensor, W: torch.Tensor, off: torch.Tensor,
     return ref.moe_gemm_grouped(A, W, off, rowmap, scale, S)
 
 
+def linear_w4(x: torch.Tensor, packed: torch.Tensor, scale: torch.Tensor,
+              zero: torch.Tensor, N: int,
+              residual: Optional[torch.Tensor] = None,
+              bias: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """y = x @ dequant4(W)^T (+bias) (+residual) with W in quant4_pack
+    layout (N, K/2 codes + f16 scale/zero per 64-group). Decode shapes
+    (M<=32) run the 4-bit weight-stream kernel (w4_gemm.hip, ~3.5x less
+    HBM traffic than bf16); anything else dequantizes and uses the dense
+    path — W4 is a decode-bandwidth play, prefill is compute-bound."""
+    K = x.shape[-1]
+    M = x.numel() // K
+    if (_on_gpu(x) and x.dtype == torch.bfloat16 and M <= 32
+            and K % 128 == 0 and N % 64 == 0):
+        _require_ext()
+        return hip_ops.gemm_w4(x.contiguous(), packed.reshape(N, K // 2),
+                               scale.reshape(N, K // 64).half(),
+                               zero.reshape(N, K // 64).half(),
+                               residual, bias, N, 0)
+    w = quant4_unpack(packed.reshape(-1, 32), scale.reshape(-1),
+                      zero.reshape(-1), dtype=x.dtype).reshape(N, K)
+    y = torch.nn.functional.linear(x, w, bias)
+    if residual is not None:
+        y = y + residual.view_as(y)
+    return y
+
+
 def quant4_pack(x: torch.Tensor, group_size: int = 64):
     if _on_gpu(x):
         _require_ext()

@@ -494,3 +494,57 @@ def test_mixtral_block_grouped_moe_decode():
         assert torch.allclose(out_grp.float(), out_loop.float(),
                               atol=2e-2), \
             (B, T, (out_grp - out_loop).abs().max().item())
+
+
+def test_gemm_w4_parity():
+    """4-bit weight-stream GEMM (w4_gemm.hip) vs dequant + fp32 matmul, with
+    and without split-K and the residual/bias epilogue."""
+    from bloombee_amd.ops import interface as iface
+    torch.manual_seed(6)
+    for (M, N, K) in [(4, 128, 256), (32, 256, 512), (17, 64, 1024)]:
+        w = (torch.randn(N, K) * 0.5)
+        x = (torch.randn(M, K) * 0.5).to(torch.bfloat16)
+        packed, scale, zero = ref.quant4_pack(w)
+        wq = ref.quant4_unpack(packed, scale, zero,
+                               dtype=torch.float32).reshape(N, K)
+        want = x.float() @ wq.t()
+        pk = packed.reshape(N, K // 2).to(DEV)
+        sc = scale.reshape(N, K // 64).half().to(DEV)
+        zp = zero.reshape(N, K // 64).half().to(DEV)
+        for ks in (1, 2):
+            got = iface.hip_ops.gemm_w4(x.to(DEV), pk, sc, zp, None, None,
+                                        N, ks).cpu().float()
+            assert torch.allclose(got, want, atol=3e-2, rtol=2e-2), \
+                (M, N, K, ks, (got - want).abs().max())
+        r = (torch.randn(M, N) * 0.5).to(torch.bfloat16)
+        b = (torch.randn(N) * 0.5).to(torch.bfloat16)
+        got = iface.hip_ops.gemm_w4(x.to(DEV), pk, sc, zp, r.to(DEV),
+                                    b.to(DEV), N, 1).cpu().float()
+        assert torch.allclose(got, want + r.float() + b.float(),
+                              atol=5e-2, rtol=2e-2)
+
+
+def test_llama_quantized_engine_decode_gpu():
+    """Quantized llama engine on GPU (w4 kernel) tracks the CPU fallback
+    (dequant matmul) within bf16 tolerance — same codes, same math."""
+    from bloombee_amd.engine import BlockStack
+    from bloombee_amd.models.base import resolve_config
+
+    cfg = resolve_config("llama-tiny")
+    cpu = BlockStack(cfg, 0, 2, device="cpu", seed=5)
+    gpu = BlockStack(cfg, 0, 2, device=DEV, seed=5)
+    for b in cpu.blocks:
+        b.quantize_weights_q4()
+    for b in gpu.blocks:
+        b.quantize_weights_q4()
+    kvc, kvg = cpu.make_kv(1 << 10), gpu.make_kv(1 << 10)
+    hc, hg = kvc.allocate(2, 32), kvg.allocate(2, 32)
+    gen = torch.Generator().manual_seed(1)
+    x = (torch.randn(2, 1, cfg.hidden_size, generator=gen) * 0.1).to(cfg.dtype)
+    sp = torch.zeros(2, dtype=torch.int32)
+    hc.extend(1)
+    hg.extend(1)
+    yc = cpu.forward_inference(x.clone(), hc, sp)
+    yg = gpu.forward_inference(x.to(DEV), hg, sp.to(DEV))
+    assert torch.allclose(yg.cpu().float(), yc.float(), atol=3e-2), \
+        (yg.cpu().float() - yc.float()).abs().max()

@@ -226,3 +226,48 @@ def test_mixtral_moe_grouped_math_matches_loop():
     out_grp.index_add_(0, tok.long(), dn)
     assert torch.allclose(out_grp, out_loop, atol=1e-4), \
         (out_grp - out_loop).abs().max()
+
+
+def test_linear_w4_fallback_matches_dequant_matmul():
+    torch.manual_seed(3)
+    N, K, M = 64, 128, 4
+    w = torch.randn(N, K)
+    x = torch.randn(M, K)
+    packed, scale, zero = ref.quant4_pack(w)
+    from bloombee_amd import ops
+    y = ops.linear_w4(x, packed, scale, zero, N)
+    wq = ref.quant4_unpack(packed, scale, zero, dtype=torch.float32).reshape(N, K)
+    want = x @ wq.t()
+    assert torch.allclose(y, want, atol=1e-4)
+    # residual + bias epilogue
+    r = torch.randn(M, N)
+    b = torch.randn(N)
+    y2 = ops.linear_w4(x, packed, scale, zero, N, residual=r, bias=b)
+    assert torch.allclose(y2, want + r + b, atol=1e-4)
+
+
+def test_llama_block_quantized_decode_close_to_full():
+    """quantize_weights_q4 keeps the block's decode output within the 4-bit
+    quantization tolerance of the full-precision block (CPU fallback path;
+    the GPU kernel has its own parity test)."""
+    from bloombee_amd.engine import BlockStack
+    from bloombee_amd.models.base import resolve_config
+
+    cfg = resolve_config("llama-tiny")
+    full = BlockStack(cfg, 0, 1, device="cpu", seed=5)
+    quant = BlockStack(cfg, 0, 1, device="cpu", seed=5)
+    quant.blocks[0].quantize_weights_q4()
+    assert quant.blocks[0].qkv_w.numel() == 0  # bf16 weights dropped
+    kvf, kvq = full.make_kv(1 << 10), quant.make_kv(1 << 10)
+    hf_, hq_ = kvf.allocate(2, 32), kvq.allocate(2, 32)
+    gen = torch.Generator().manual_seed(1)
+    x = (torch.randn(2, 4, cfg.hidden_size, generator=gen) * 0.1).to(cfg.dtype)
+    sp = torch.zeros(2, dtype=torch.int32)
+    hf_.extend(4)
+    hq_.extend(4)
+    yf = full.forward_inference(x.clone(), hf_, sp)
+    yq = quant.forward_inference(x.clone(), hq_, sp)
+    # same direction, bounded quant error
+    num = (yf.float() - yq.float()).norm()
+    den = yf.float().norm().clamp_min(1e-6)
+    assert (num / den) < 0.2, (num / den)
