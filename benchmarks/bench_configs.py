@@ -24,11 +24,13 @@ import torch  # noqa: E402
 DEV = "cuda:0" if torch.cuda.is_available() else "cpu"
 
 
-def _decode_loop(eng, B, prompt_len, steps):
+def _decode_loop(eng, B, prompt_len, steps, warmup=4):
     ids = torch.randint(0, eng.config.vocab_size, (B, prompt_len),
                         generator=torch.Generator().manual_seed(0))
-    kv = eng.kv_pool.allocate(B, prompt_len + steps + 4)
+    kv = eng.kv_pool.allocate(B, prompt_len + steps + warmup + 4)
     tok = eng.prefill(ids, kv)
+    for _ in range(warmup):
+        tok = eng.decode_step(tok, kv)
     if DEV.startswith("cuda"):
         torch.cuda.synchronize()
     t0 = time.monotonic()
@@ -118,7 +120,13 @@ def bench_spec(steps=32, self_draft=False, draft_q4=False):
     else:
         draft = LocalEngine("llama-mini-gpu" if DEV != "cpu" else "llama-tiny",
                             device=DEV, seed=3, kv_max_tokens=1 << 14)
-    drafter = MultiDrafter(draft, node_budget=8, max_depth=4)
+    # cost_ratio: draft-step cost in target-step units (the planner's
+    # tokens/sec-optimal stopping rule) — ~0.3 for the 4-bit self-draft,
+    # 1.0 for bf16 self-draft, near-free for a mini draft
+    cr = 0.35 if draft_q4 else (1.0 if self_draft else 0.05)
+    drafter = MultiDrafter(draft, node_budget=8, max_depth=4, cost_ratio=cr)
+    drafter.start_session(512)  # persistent draft KV: only the committed
+    # suffix is prefilled per round (not the whole history)
     prompt = torch.randint(0, 1000, (1, 32),
                            generator=torch.Generator().manual_seed(1))
     kv = tgt.kv_pool.allocate(1, 512)
@@ -127,9 +135,21 @@ def bench_spec(steps=32, self_draft=False, draft_q4=False):
     pending = int(tok)
     emitted = 0
     rounds = 0
+    draft_s = verify_s = 0.0
+
+    def _sync():
+        if DEV.startswith("cuda"):
+            torch.cuda.synchronize()
+
+    _sync()
     t0 = time.monotonic()
     while emitted < steps:
-        sub = drafter.build_tree(torch.tensor(history + [pending]))
+        td = time.monotonic()
+        sub = drafter.build_tree_incremental(
+            torch.tensor(history + [pending]))
+        _sync()
+        draft_s += time.monotonic() - td
+        tv = time.monotonic()
         tree = TokenTree()
         tree.add(pending, -1, 1.0)
         for i in range(len(sub)):
@@ -155,6 +175,8 @@ def bench_spec(steps=32, self_draft=False, draft_q4=False):
         emitted += len(emit)
         pending = bonus
         rounds += 1
+        _sync()
+        verify_s += time.monotonic() - tv
         drafter.record_result(len(acc), offered_depth=4)
     if DEV.startswith("cuda"):
         torch.cuda.synchronize()
@@ -168,7 +190,9 @@ def bench_spec(steps=32, self_draft=False, draft_q4=False):
                       "tokens_per_round": round(emitted / rounds, 2),
                       "tokens_per_s": round(tps, 1),
                       "plain_greedy_tokens_per_s": round(base_tps, 1),
-                      "spec_speedup": round(tps / base_tps, 2)}))
+                      "spec_speedup": round(tps / base_tps, 2),
+                      "draft_ms_per_round": round(draft_s / rounds * 1e3, 2),
+                      "verify_ms_per_round": round(verify_s / rounds * 1e3, 2)}))
 
 
 ALL = {"offload": bench_offload, "mixtral": bench_mixtral,

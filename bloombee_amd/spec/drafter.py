@@ -26,12 +26,52 @@ class MultiDrafter:
     """
 
     def __init__(self, draft_model: LocalEngine, n_workers: int = 2,
-                 node_budget: int = 8, max_depth: int = 6):
+                 node_budget: int = 8, max_depth: int = 6,
+                 cost_ratio: float = 0.0):
         self.draft = draft_model
         self.n_workers = n_workers
         self.node_budget = node_budget
         self.max_depth = max_depth
+        self.cost_ratio = cost_ratio
         self.stats = AcceptanceStats(max_depth=max_depth + 2)
+        self._kv = None
+        self._kv_len = 0
+
+    def start_session(self, max_len: int = 1024) -> None:
+        """Persistent-KV drafting: keep the draft model's cache across
+        rounds and prefill only the NEWLY committed suffix each round
+        (build_tree_incremental) — the reference drafter workers hold
+        their SSM caches the same way (spec_decoding_drafter.py:110-480);
+        the stateless build_tree re-prefills the whole history per round."""
+        self.close_session()
+        self._kv = self.draft.kv_pool.allocate(1, max_len)
+        self._kv_len = 0
+
+    def close_session(self) -> None:
+        kv = getattr(self, "_kv", None)
+        if kv is not None:
+            kv.close()
+        self._kv = None
+        self._kv_len = 0
+
+    @torch.no_grad()
+    def build_tree_incremental(self, history: torch.Tensor) -> TokenTree:
+        """history: (T,) full committed tokens (append-only across calls).
+        Prefills only history[len_seen:] into the persistent cache, then
+        expands a tree. Requires start_session()."""
+        eng = self.draft
+        kv = self._kv
+        history = history.clamp(0, eng.config.vocab_size - 1)
+        new = history[self._kv_len:].view(1, -1)
+        kv.rollback()  # drop last round's speculative region
+        assert new.shape[1] > 0, "no new committed tokens since last round"
+        start = torch.full((1,), self._kv_len, dtype=torch.int32,
+                           device=eng.device)
+        kv.extend(new.shape[1])
+        hidden = eng.stack.forward_inference(eng._embed(new), kv, start)
+        logits0 = eng.logits_for(hidden[:, -1]).float()[0]
+        self._kv_len = history.numel()
+        return self._expand(kv, logits0)
 
     def build_tree(self, prompt_ids: torch.Tensor) -> TokenTree:
         """prompt_ids: (T,) full committed token history of the sequence.
@@ -41,11 +81,8 @@ class MultiDrafter:
         target vocab in random-init testing) are clamped so the draft
         embedding never indexes out of bounds — such tokens simply draft
         badly instead of crashing the device."""
-        widths = plan_tree_shape(self.stats, self.node_budget,
-                                 max_depth=self.max_depth)
         eng = self.draft
         prompt_ids = prompt_ids.clamp(0, eng.config.vocab_size - 1)
-        tree = TokenTree()
         ids = prompt_ids.view(1, -1)
         with torch.no_grad():
             kv = eng.kv_pool.allocate(1, ids.shape[1] + self.max_depth + 4)
@@ -56,50 +93,57 @@ class MultiDrafter:
                 hidden = eng._embed(ids)
                 hidden = eng.stack.forward_inference(hidden, kv, start)
                 logits0 = eng.logits_for(hidden[:, -1]).float()[0]
-                probs0 = torch.softmax(logits0, -1)
-                w0 = max(1, widths[0])
-                top = probs0.topk(w0)
-                # full draft dist stored per node: the exact SpecInfer
-                # rejection residual (p_target - p_draft)+ needs it
-                roots = [tree.add(int(t), -1, float(p), dist=probs0)
-                         for t, p in zip(top.indices, top.values)]
-
-                lock = threading.Lock()
-
-                def extend_branch(root_idx: int):
-                    # chain-extend one root on a speculative KV region
-                    chain_parent = root_idx
-                    chain_tok = tree.tokens[root_idx]
-                    local = []
-                    with lock:
-                        pass
-                    for d in range(1, len(widths)):
-                        with lock:
-                            pos = kv.seqs[0].l_spec
-                            kv.extend(1, speculative=True)
-                            h = eng._embed(torch.tensor([[chain_tok]]))
-                            sp = torch.tensor([pos], dtype=torch.int32,
-                                              device=eng.device)
-                            h = eng.stack.forward_inference(h, kv, sp)
-                            lg = eng.logits_for(h[:, -1]).float()[0]
-                        p = torch.softmax(lg, -1)
-                        t = int(p.argmax())
-                        with lock:
-                            chain_parent = tree.add(t, chain_parent,
-                                                    float(p[t]), dist=p)
-                        chain_tok = t
-                        local.append(t)
-                    return local
-
-                # NOTE: branches share one draft KV session; branch chains are
-                # serialized by the lock (thread workers mirror the reference
-                # API; true parallelism needs per-branch sessions) and each
-                # branch's speculative tokens are rolled back before the next.
-                for r in roots:
-                    extend_branch(r)
-                    kv.rollback()
+                return self._expand(kv, logits0)
             finally:
                 kv.close()
+
+    @torch.no_grad()
+    def _expand(self, kv, logits0: torch.Tensor) -> TokenTree:
+        widths = plan_tree_shape(self.stats, self.node_budget,
+                                 max_depth=self.max_depth,
+                                 cost_ratio=self.cost_ratio)
+        eng = self.draft
+        tree = TokenTree()
+        probs0 = torch.softmax(logits0, -1)
+        w0 = max(1, widths[0])
+        top = probs0.topk(w0)
+        # full draft dist stored per node: the exact SpecInfer
+        # rejection residual (p_target - p_draft)+ needs it
+        roots = [tree.add(int(t), -1, float(p), dist=probs0)
+                 for t, p in zip(top.indices, top.values)]
+
+        lock = threading.Lock()
+
+        def extend_branch(root_idx: int):
+            # chain-extend one root on a speculative KV region
+            chain_parent = root_idx
+            chain_tok = tree.tokens[root_idx]
+            local = []
+            for d in range(1, len(widths)):
+                with lock:
+                    pos = kv.seqs[0].l_spec
+                    kv.extend(1, speculative=True)
+                    h = eng._embed(torch.tensor([[chain_tok]]))
+                    sp = torch.tensor([pos], dtype=torch.int32,
+                                      device=eng.device)
+                    h = eng.stack.forward_inference(h, kv, sp)
+                    lg = eng.logits_for(h[:, -1]).float()[0]
+                p = torch.softmax(lg, -1)
+                t = int(p.argmax())
+                with lock:
+                    chain_parent = tree.add(t, chain_parent,
+                                            float(p[t]), dist=p)
+                chain_tok = t
+                local.append(t)
+            return local
+
+        # NOTE: branches share one draft KV session; branch chains are
+        # serialized by the lock (thread workers mirror the reference
+        # API; true parallelism needs per-branch sessions) and each
+        # branch's speculative tokens are rolled back before the next.
+        for r in roots:
+            extend_branch(r)
+            kv.rollback()
         return tree
 
     def record_result(self, accepted_len: int, offered_depth: int) -> None:

@@ -28,11 +28,19 @@ class AcceptanceStats:
 
 
 def plan_tree_shape(stats: AcceptanceStats, budget: int,
-                    max_depth: int = 8, max_width: int = 4) -> List[int]:
+                    max_depth: int = 8, max_width: int = 4,
+                    cost_ratio: float = 0.0) -> List[int]:
     """Widths per depth maximizing expected accepted length under a total
     node budget. Greedy marginal-gain allocation (the reference's dynamic
     width optimization, :74-165): start with a depth-1 chain and repeatedly
     add the node with the best expected marginal accepted-length gain.
+
+    cost_ratio: draft-step cost in TARGET-step units (e.g. ~0.3 for a
+    4-bit self-draft). A node is only added while its marginal expected
+    accepted tokens exceed what the same wall time would emit by plain
+    decoding — the tokens/sec-optimal stopping rule the reference's
+    fixed-budget planner lacks. With a low-acceptance draft the tree
+    collapses toward a single probe node instead of burning budget.
     """
     widths = [1]
     nodes = 1
@@ -60,7 +68,7 @@ def plan_tree_shape(stats: AcceptanceStats, budget: int,
         options = [("deep", gain_extend_depth())]
         options += [(("wide", d), gain_widen(d)) for d in range(len(widths))]
         kind, g = max(options, key=lambda kv: kv[1])
-        if g <= 0:
+        if g <= cost_ratio:
             break
         if kind == "deep":
             widths.append(1)
