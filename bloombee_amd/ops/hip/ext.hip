@@ -504,7 +504,9 @@ torch::Tensor gemm_w4(torch::Tensor A, torch::Tensor Wq, torch::Tensor scale,
                       torch::Tensor zero,
                       c10::optional<torch::Tensor> residual,
                       c10::optional<torch::Tensor> bias, long N_, long ksplit_req) {
-  CHECK_DEV(A); CHECK_BF16(A); CHECK_CONTIG(A);
+  CHECK_DEV(A); CHECK_CONTIG(A);
+  TORCH_CHECK(A.scalar_type() == at::kHalf,
+              "gemm_w4 takes fp16 activations (wrapper converts)");
   CHECK_DEV_ALL3(Wq, scale, zero);
   CHECK_CONTIG(Wq);
   TORCH_CHECK(Wq.scalar_type() == at::kByte && scale.scalar_type() == at::kHalf
@@ -518,7 +520,7 @@ torch::Tensor gemm_w4(torch::Tensor A, torch::Tensor Wq, torch::Tensor scale,
   TORCH_CHECK(K % 128 == 0 && N % 64 == 0, "K%128, N%64 required");
   auto sizes = A.sizes().vec();
   sizes.back() = N;
-  auto C = torch::empty(sizes, A.options());
+  auto C = torch::empty(sizes, A.options().dtype(at::kBFloat16));
   const unsigned short* rp = nullptr;
   if (residual.has_value()) {
     TORCH_CHECK(residual->is_contiguous() && residual->numel() == (long)M * N);
@@ -553,7 +555,8 @@ torch::Tensor gemm_w4(torch::Tensor A, torch::Tensor Wq, torch::Tensor scale,
   dim3 grid((N + 63) / 64, ksplit);
   auto launch = [&](auto mt) {
     gemm_skinny_w4_kernel<decltype(mt)::value><<<grid, 256, 0, cur_stream()>>>(
-        bf_ptr(A), Wq.data_ptr<unsigned char>(),
+        reinterpret_cast<const _Float16*>(A.data_ptr()),
+        Wq.data_ptr<unsigned char>(),
         reinterpret_cast<const __half*>(scale.data_ptr()),
         reinterpret_cast<const __half*>(zero.data_ptr()),
         ksplit == 1 ? rp : nullptr, ksplit == 1 ? bp : nullptr,

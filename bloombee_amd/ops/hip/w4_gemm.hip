@@ -2,40 +2,52 @@
 // M <= 32. Decode GEMMs are pure weight streams (gemm_skinny.hip); quantizing
 // W to 4-bit group-min/max codes (quant4.hip format: 2 codes/byte along K,
 // fp16 scale/zero per group of 64) cuts the streamed bytes ~3.5x, which is
-// the whole speedup — dequant runs as VALU work between the MFMAs.
+// the whole speedup — so the in-kernel dequant must be nearly free or the
+// kernel turns VALU-bound (the first, scalar-f2bf version measured 0.6 TB/s
+// effective = SLOWER than bf16).
 //
-// This is what makes W4 weight-compressed DECODE a compute path rather than
-// a storage trick (the reference's FlexGen compression only compresses
-// weights at rest, flexgen_utils/compression.py:94-210), and it makes
-// same-weights self-drafting for speculative decoding pay for itself: the
-// draft streams a quarter of the target's bytes.
+// Fast dequant (the classic w4 fp16 bit-trick):
+//   half(0x6400 | q) == 1024 + q        for q in 0..15
+// so two nibbles OR'd into a half2 give {1024+lo, 1024+hi} with 3 bit ops,
+// and one v_pk_fma_f16 applies w = q*sc + (zp - 1024*sc). A rides as fp16
+// (the host converts the tiny activation matrix) and the MFMA is the f16
+// variant — no per-element float conversions anywhere. ~5 VALU per pair
+// instead of ~20.
 //
-// Memory pattern: lane (li, hi) loads a 4-byte code word (8 nibbles) per
-// 32-k slice; the four hi lanes of one W row touch one 16 B window per
-// slice and adjacent slices walk the row sequentially, so L2 turns the
-// 16 B requests into full DRAM bursts (same locality argument as the bf16
-// kernel's 16 B fragments).
+// This puts FlexGen-style weight compression ON the compute path (the
+// reference compresses weights only at rest, flexgen_utils/compression.py:
+// 94-210) and makes same-weights self-drafting for speculative decoding
+// pay for itself: the draft streams a quarter of the target's bytes.
 
 #include "common.h"
 #include <hip/hip_fp16.h>
 
-DEVINL bf16x8 dq8_q4(unsigned int c, float sc, float zp) {
-  short8 s;
+typedef __attribute__((ext_vector_type(8))) _Float16 half8;
+typedef __attribute__((ext_vector_type(2))) _Float16 half2v;
+
+DEVINL half8 dq8_q4_f16(unsigned int c, half2v sc2, half2v zp2) {
+  half8 out;
 #pragma unroll
-  for (int j = 0; j < 8; ++j)
-    s[j] = (short)f2bf((float)((c >> (4 * j)) & 0xFu) * sc + zp);
-  return as_bf16x8(s);
+  for (int p = 0; p < 4; ++p) {
+    const unsigned byte = (c >> (8 * p)) & 0xFFu;
+    const unsigned h2 = 0x64006400u | (byte & 0xFu) | ((byte & 0xF0u) << 12);
+    half2v v = __builtin_bit_cast(half2v, h2);
+    v = v * sc2 + zp2;  // v_pk_fma_f16
+    out[2 * p] = v[0];
+    out[2 * p + 1] = v[1];
+  }
+  return out;
 }
 
 template <int MT>  // 16-row M-tiles (1: M<=16, 2: M<=32)
 __global__ __launch_bounds__(256) void gemm_skinny_w4_kernel(
-    const unsigned short* __restrict__ A,     // (M, K) bf16
+    const _Float16* __restrict__ A,           // (M, K) fp16
     const unsigned char* __restrict__ Wq,     // (N, K/2) packed nibbles
     const __half* __restrict__ scale,         // (N, K/64) f16
     const __half* __restrict__ zero,          // (N, K/64) f16
-    const unsigned short* __restrict__ R,     // (M, N) residual or null
-    const unsigned short* __restrict__ bias,  // (N,) or null
-    unsigned short* __restrict__ C,           // (M, N)       (ksplit == 1)
+    const unsigned short* __restrict__ R,     // (M, N) bf16 residual or null
+    const unsigned short* __restrict__ bias,  // (N,) bf16 or null
+    unsigned short* __restrict__ C,           // (M, N) bf16   (ksplit == 1)
     float* __restrict__ Cpart,                // (ksplit, M, N) (ksplit > 1)
     int M, int N, int K, int kchunk, int ksplit) {
   const int wave = threadIdx.x / WAVE;
@@ -70,26 +82,30 @@ __global__ __launch_bounds__(256) void gemm_skinny_w4_kernel(
 #pragma unroll
   for (int t = 0; t < MT; ++t) acc[t] = (f32x4){0.f, 0.f, 0.f, 0.f};
 
-  // two code words (2 x 32-k slices) per group of 64: load scale/zero once
-  // per 64-k step. The 4-deep unroll keeps ~8 loads in flight per lane.
+  // two code words (2 x 32-k slices) per group of 64: scale/zero loaded
+  // once per group; the 128-k step keeps ~4 code loads in flight per lane
   for (int k = k0; k < k1; k += 128) {
 #pragma unroll
     for (int g = 0; g < 2; ++g) {                 // two 64-k groups
       const int kg = k + g * 64;
-      const float sc = __half2float(srow[kg / 64]);
-      const float zp = __half2float(zrow[kg / 64]);
+      const float scf = __half2float(srow[kg / 64]);
+      const float zpf = __half2float(zrow[kg / 64]) - 1024.f * scf;
+      const _Float16 sch = (_Float16)scf;
+      const _Float16 zph = (_Float16)zpf;
+      const half2v sc2 = {sch, sch};
+      const half2v zp2 = {zph, zph};
 #pragma unroll
       for (int u = 0; u < 2; ++u) {               // two 32-k slices per group
         const int kk = kg + u * 32 + hi * 8;
         const unsigned int cw =
             *reinterpret_cast<const unsigned int*>(wrow + kk / 2);
-        bf16x8 bfrag = dq8_q4(cw, sc, zp);
+        half8 bfrag = dq8_q4_f16(cw, sc2, zp2);
 #pragma unroll
         for (int t = 0; t < MT; ++t) {
-          bf16x8 afrag = as_bf16x8(
-              *reinterpret_cast<const short8*>(A + (long)arow[t] * K + kk));
-          acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag,
-                                                           acc[t], 0, 0, 0);
+          half8 afrag = *reinterpret_cast<const half8*>(
+              A + (long)arow[t] * K + kk);
+          acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_f16(afrag, bfrag,
+                                                          acc[t], 0, 0, 0);
         }
       }
     }

@@ -339,7 +339,11 @@ def linear_w4(x: torch.Tensor, packed: torch.Tensor, scale: torch.Tensor,
     if (_on_gpu(x) and x.dtype == torch.bfloat16 and M <= 32
             and K % 128 == 0 and N % 64 == 0):
         _require_ext()
-        return hip_ops.gemm_w4(x.contiguous(), packed.reshape(N, K // 2),
+        # the kernel computes in f16 (fast nibble->half dequant + f16 MFMA);
+        # the activation conversion is M*K elements - noise next to the
+        # 4-bit weight stream it unlocks
+        return hip_ops.gemm_w4(x.half().contiguous(),
+                               packed.reshape(N, K // 2),
                                scale.reshape(N, K // 64).half(),
                                zero.reshape(N, K // 64).half(),
                                residual, bias, N, 0)
