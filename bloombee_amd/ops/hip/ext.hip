@@ -531,13 +531,37 @@ torch::Tensor gemm_w4(torch::Tensor A, torch::Tensor Wq, torch::Tensor scale,
     TORCH_CHECK(bias->is_contiguous() && bias->numel() == N);
     bp = bf_ptr(*bias);
   }
+  // M==1 (the speculative-draft GEMV shape): contiguous 1 KB code loads +
+  // v_dot2 — measured 2x the bf16 kernel (w4_probe.hip). rows_per_wave=4.
+  if (M == 1 && K % 2048 == 0 && K / 2048 <= 7) {
+    const int rpw = 4;
+    dim3 gv((N / rpw + 3) / 4);
+    auto lv = [&](auto nl) {
+      gemm_w4_gemv_kernel<decltype(nl)::value><<<gv, 256, 0, cur_stream()>>>(
+          reinterpret_cast<const _Float16*>(A.data_ptr()),
+          Wq.data_ptr<unsigned char>(),
+          reinterpret_cast<const __half*>(scale.data_ptr()),
+          reinterpret_cast<const __half*>(zero.data_ptr()),
+          rp, bp, bf_ptr_mut(C), N, K, rpw);
+    };
+    switch (K / 2048) {
+      case 1: lv(std::integral_constant<int, 1>{}); break;
+      case 2: lv(std::integral_constant<int, 2>{}); break;
+      case 3: lv(std::integral_constant<int, 3>{}); break;
+      case 4: lv(std::integral_constant<int, 4>{}); break;
+      case 5: lv(std::integral_constant<int, 5>{}); break;
+      case 6: lv(std::integral_constant<int, 6>{}); break;
+      default: lv(std::integral_constant<int, 7>{}); break;
+    }
+    return C;
+  }
   int ksplit = (int)ksplit_req;
   if (ksplit <= 0) {
-    // same residency logic as gemm_skinny, but the stream is ~3.5x shorter
-    if (N / 64 >= 256) ksplit = 1;
-    else ksplit = (int)std::min<long>(K / 1024,
-                                      std::max<long>(1, std::min<long>(8, 1024 / (N / 64))));
-    ksplit = std::max(1, ksplit);
+    // the w4 stream is LATENCY-bound at the bf16 kernel's grid size (1.2
+    // TB/s at 448 blocks); split K until the grid reaches ~1800 blocks
+    // (w4_probe.hip: ksplit=4 on N=28672 -> 2.25 TB/s, plateau beyond)
+    ksplit = (int)std::max<long>(1, std::min<long>(K / 512,
+                                                   1792 / std::max(1, (N + 63) / 64)));
   }
   // groups of 64 must not straddle splits; slices of 128 keep the k-loop
   // tail-free

@@ -501,7 +501,8 @@ def test_gemm_w4_parity():
     and without split-K and the residual/bias epilogue."""
     from bloombee_amd.ops import interface as iface
     torch.manual_seed(6)
-    for (M, N, K) in [(4, 128, 256), (32, 256, 512), (17, 64, 1024)]:
+    for (M, N, K) in [(4, 128, 256), (32, 256, 512), (17, 64, 1024),
+                      (1, 192, 2048), (1, 128, 4096)]:  # GEMV path (M=1)
         w = (torch.randn(N, K) * 0.5)
         x = (torch.randn(M, K) * 0.5).to(torch.bfloat16)
         packed, scale, zero = ref.quant4_pack(w)
