@@ -512,14 +512,15 @@ def test_gemm_w4_parity():
         pk = packed.reshape(N, K // 2).to(DEV)
         sc = scale.reshape(N, K // 64).half().to(DEV)
         zp = zero.reshape(N, K // 64).half().to(DEV)
+        xh = x.half().to(DEV)  # the kernel computes in f16
         for ks in (1, 2):
-            got = iface.hip_ops.gemm_w4(x.to(DEV), pk, sc, zp, None, None,
+            got = iface.hip_ops.gemm_w4(xh, pk, sc, zp, None, None,
                                         N, ks).cpu().float()
-            assert torch.allclose(got, want, atol=3e-2, rtol=2e-2), \
+            assert torch.allclose(got, want, atol=5e-2, rtol=2e-2), \
                 (M, N, K, ks, (got - want).abs().max())
         r = (torch.randn(M, N) * 0.5).to(torch.bfloat16)
         b = (torch.randn(N) * 0.5).to(torch.bfloat16)
-        got = iface.hip_ops.gemm_w4(x.to(DEV), pk, sc, zp, r.to(DEV),
+        got = iface.hip_ops.gemm_w4(xh, pk, sc, zp, r.to(DEV),
                                     b.to(DEV), N, 1).cpu().float()
         assert torch.allclose(got, want + r.float() + b.float(),
                               atol=5e-2, rtol=2e-2)
