@@ -47,6 +47,19 @@ LLAMA_PRESETS = {
         num_key_value_heads=2, intermediate_size=2816, vocab_size=32000,
         max_position_embeddings=4096, rope_theta=10000.0,
     ),
+    # trained speculative-decoding pair (benchmarks/spec_trained.py): both
+    # learn the same synthetic grammar so draft/target agreement reflects
+    # model quality, not random-init luck
+    "llama-spec-target": dict(
+        hidden_size=2048, num_hidden_layers=8, num_attention_heads=16,
+        num_key_value_heads=4, intermediate_size=5632, vocab_size=8192,
+        max_position_embeddings=2048, rope_theta=10000.0,
+    ),
+    "llama-spec-draft": dict(
+        hidden_size=512, num_hidden_layers=2, num_attention_heads=4,
+        num_key_value_heads=2, intermediate_size=1408, vocab_size=8192,
+        max_position_embeddings=2048, rope_theta=10000.0,
+    ),
 }
 
 
