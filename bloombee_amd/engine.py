@@ -200,3 +200,53 @@ class LocalEngine:
             return torch.stack(toks, dim=1)
         finally:
             kv.close()
+
+    def make_graphed_decoder(self, kv: SessionHandle):
+        """Single-token decode step (B=1) as a hipGraph replay against a
+        FIXED session handle: the ~n_layers*8-launch step otherwise costs
+        ~0.5 ms of host/launch overhead per token — fatal for small draft
+        models in speculative decoding (the drafter calls this per tree
+        node). Returns step(token:int, position:int) -> logits view.
+
+        The caller owns kv growth (extend/rollback) BEFORE each step; the
+        graph reads position from a device buffer and the page table from
+        its fixed device tensor. Falls back to an eager closure off-GPU."""
+        ids = torch.zeros(1, 1, dtype=torch.long, device=self.device)
+        pos = torch.zeros(1, dtype=torch.int32, device=self.device)
+
+        def _body():
+            h = self._embed(ids)
+            h = self.stack.forward_inference(h, kv, pos)
+            return self.logits_for(h[:, -1]).float()
+
+        from bloombee_amd.config import get_config
+        if self.device.type != "cuda" or not get_config().use_hip_graphs:
+            def step_eager(tok: int, position: int) -> torch.Tensor:
+                ids[0, 0] = tok
+                pos[0] = position
+                return _body()
+            return step_eager
+
+        # warm the allocator/kernels, then capture (thread_local keeps other
+        # threads' work — e.g. the channels pump — legal during capture)
+        for _ in range(2):
+            kv.extend(1, speculative=True)
+            kv.page_table()
+            _body()
+            kv.rollback()
+        kv.page_table()
+        graph = torch.cuda.CUDAGraph()
+        kv.extend(1, speculative=True)
+        kv.page_table()
+        with torch.cuda.graph(graph, capture_error_mode="thread_local"):
+            logits_buf = _body()
+        kv.rollback()
+
+        def step(tok: int, position: int) -> torch.Tensor:
+            ids[0, 0] = tok
+            pos[0] = position
+            kv.page_table()  # flush any new page ids
+            graph.replay()
+            return logits_buf
+
+        return step

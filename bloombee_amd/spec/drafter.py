@@ -42,10 +42,16 @@ class MultiDrafter:
         rounds and prefill only the NEWLY committed suffix each round
         (build_tree_incremental) — the reference drafter workers hold
         their SSM caches the same way (spec_decoding_drafter.py:110-480);
-        the stateless build_tree re-prefills the whole history per round."""
+        the stateless build_tree re-prefills the whole history per round.
+
+        On GPU the per-node decode step runs as a hipGraph replay against
+        the persistent cache (engine.make_graphed_decoder): the eager
+        ~n_layers*8-launch step costs ~0.5 ms of host overhead per node,
+        which dwarfs a small draft's actual compute."""
         self.close_session()
         self._kv = self.draft.kv_pool.allocate(1, max_len)
         self._kv_len = 0
+        self._gstep = self.draft.make_graphed_decoder(self._kv)
 
     def close_session(self) -> None:
         kv = getattr(self, "_kv", None)
@@ -53,6 +59,7 @@ class MultiDrafter:
             kv.close()
         self._kv = None
         self._kv_len = 0
+        self._gstep = None
 
     @torch.no_grad()
     def build_tree_incremental(self, history: torch.Tensor) -> TokenTree:
@@ -113,6 +120,8 @@ class MultiDrafter:
                  for t, p in zip(top.indices, top.values)]
 
         lock = threading.Lock()
+        gstep = (self._gstep if kv is getattr(self, "_kv", None)
+                 and getattr(self, "_gstep", None) is not None else None)
 
         def extend_branch(root_idx: int):
             # chain-extend one root on a speculative KV region
@@ -123,16 +132,21 @@ class MultiDrafter:
                 with lock:
                     pos = kv.seqs[0].l_spec
                     kv.extend(1, speculative=True)
-                    h = eng._embed(torch.tensor([[chain_tok]]))
-                    sp = torch.tensor([pos], dtype=torch.int32,
-                                      device=eng.device)
-                    h = eng.stack.forward_inference(h, kv, sp)
-                    lg = eng.logits_for(h[:, -1]).float()[0]
+                    if gstep is not None:
+                        lg = gstep(chain_tok, pos)[0]
+                    else:
+                        h = eng._embed(torch.tensor([[chain_tok]]))
+                        sp = torch.tensor([pos], dtype=torch.int32,
+                                          device=eng.device)
+                        h = eng.stack.forward_inference(h, kv, sp)
+                        lg = eng.logits_for(h[:, -1]).float()[0]
                 p = torch.softmax(lg, -1)
                 t = int(p.argmax())
                 with lock:
-                    chain_parent = tree.add(t, chain_parent,
-                                            float(p[t]), dist=p)
+                    # clone: the graphed path reuses the logits buffer
+                    chain_parent = tree.add(
+                        t, chain_parent, float(p[t]),
+                        dist=p.clone() if gstep is not None else p)
                 chain_tok = t
                 local.append(t)
             return local
