@@ -178,7 +178,7 @@ def main():
     history = prompt[0].tolist()
     pending = int(tok)
     emitted, rounds = 0, 0
-    draft_s = verify_s = 0.0
+    draft_s = verify_s = prefill_s = 0.0
 
     def _sync():
         if DEV.startswith("cuda"):
@@ -191,6 +191,7 @@ def main():
         sub = drafter.build_tree_incremental(torch.tensor(history + [pending]))
         _sync()
         draft_s += time.monotonic() - td
+        prefill_s += getattr(drafter, "t_prefill", 0.0)
         tv = time.monotonic()
         tree = TokenTree()
         tree.add(pending, -1, 1.0)
@@ -236,6 +237,7 @@ def main():
         "spec_tokens_per_s": round(spec_tps, 1),
         "spec_speedup": round(spec_tps / plain_tps, 2),
         "draft_ms_per_round": round(draft_s / rounds * 1e3, 2),
+        "draft_prefill_ms_per_round": round(prefill_s / rounds * 1e3, 2),
         "verify_ms_per_round": round(verify_s / rounds * 1e3, 2),
     }))
 
