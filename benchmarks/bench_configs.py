@@ -125,6 +125,7 @@ def bench_spec(steps=32, self_draft=False, draft_q4=False):
     # 1.0 for bf16 self-draft, near-free for a mini draft
     cr = 0.35 if draft_q4 else (1.0 if self_draft else 0.05)
     drafter = MultiDrafter(draft, node_budget=8, max_depth=4, cost_ratio=cr)
+    drafter.collect_dists = False  # greedy verify never reads them
     drafter.start_session(512)  # persistent draft KV: only the committed
     # suffix is prefilled per round (not the whole history)
     prompt = torch.randint(0, 1000, (1, 32),

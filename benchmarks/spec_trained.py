@@ -172,6 +172,7 @@ def main():
     # ---- speculative decode (tree verify on the target's HIP kernels) ----
     drafter = MultiDrafter(draft, node_budget=args.node_budget,
                            max_depth=args.max_depth, cost_ratio=0.10)
+    drafter.collect_dists = False  # greedy verify never reads them
     drafter.start_session(32 + args.decode + args.node_budget + 16)
     kv = tgt.kv_pool.allocate(1, 32 + args.decode + args.node_budget + 16)
     tok = tgt.prefill(prompt.to(DEV), kv)

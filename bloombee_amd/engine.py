@@ -202,29 +202,44 @@ class LocalEngine:
             kv.close()
 
     def make_graphed_decoder(self, kv: SessionHandle):
-        """Single-token decode step (B=1) as a hipGraph replay against a
-        FIXED session handle: the ~n_layers*8-launch step otherwise costs
-        ~0.5 ms of host/launch overhead per token — fatal for small draft
-        models in speculative decoding (the drafter calls this per tree
-        node). Returns step(token:int, position:int) -> logits view.
+        """Single-token GREEDY decode step (B=1) as a hipGraph replay
+        against a FIXED session handle. The whole step — embed, blocks,
+        logits, softmax, (prob, argmax) — is inside the graph; the host
+        writes (token, position) into one pinned staging buffer, replays,
+        and reads ONE 8-byte pair back. Measured motivation
+        (benchmarks/gstep_micro.py): the eager step costs ~350 µs/node
+        and even post-replay eager softmax/argmax ops add ~100 µs EACH —
+        fatal for small draft models in speculative decoding.
 
-        The caller owns kv growth (extend/rollback) BEFORE each step; the
-        graph reads position from a device buffer and the page table from
-        its fixed device tensor. Falls back to an eager closure off-GPU."""
-        ids = torch.zeros(1, 1, dtype=torch.long, device=self.device)
-        pos = torch.zeros(1, dtype=torch.int32, device=self.device)
+        Returns step(token:int, position:int) -> (next_tok:int,
+        prob:float, probs_buffer). probs_buffer is the graph-owned softmax
+        output — clone it before the next replay if kept. The caller owns
+        kv growth (extend/rollback) BEFORE each step. Falls back to an
+        eager closure off-GPU."""
+        from bloombee_amd.config import get_config
+
+        on_gpu = (self.device.type == "cuda"
+                  and get_config().use_hip_graphs)
+        staging = torch.zeros(2, dtype=torch.long,
+                              pin_memory=on_gpu)
+        dev_in = torch.zeros(2, dtype=torch.long, device=self.device)
 
         def _body():
+            ids = dev_in[0:1].view(1, 1)
+            pos = dev_in[1:2].to(torch.int32)
             h = self._embed(ids)
             h = self.stack.forward_inference(h, kv, pos)
-            return self.logits_for(h[:, -1]).float()
+            logits = self.logits_for(h[:, -1]).float()
+            p = torch.softmax(logits[0], -1)
+            pv, ti = p.max(-1)
+            return p, torch.stack((ti.to(torch.float32), pv))
 
-        from bloombee_amd.config import get_config
-        if self.device.type != "cuda" or not get_config().use_hip_graphs:
-            def step_eager(tok: int, position: int) -> torch.Tensor:
-                ids[0, 0] = tok
-                pos[0] = position
-                return _body()
+        if not on_gpu:
+            def step_eager(tok: int, position: int):
+                dev_in[0] = tok
+                dev_in[1] = position
+                p, pair = _body()
+                return int(pair[0]), float(pair[1]), p
             return step_eager
 
         # warm the allocator/kernels, then capture (thread_local keeps other
@@ -234,19 +249,21 @@ class LocalEngine:
             kv.page_table()
             _body()
             kv.rollback()
-        kv.page_table()
         graph = torch.cuda.CUDAGraph()
         kv.extend(1, speculative=True)
         kv.page_table()
         with torch.cuda.graph(graph, capture_error_mode="thread_local"):
-            logits_buf = _body()
+            p_buf, pair_buf = _body()
         kv.rollback()
+        pair_host = torch.zeros(2, dtype=torch.float32, pin_memory=True)
 
-        def step(tok: int, position: int) -> torch.Tensor:
-            ids[0, 0] = tok
-            pos[0] = position
+        def step(tok: int, position: int):
+            staging[0] = tok
+            staging[1] = position
+            dev_in.copy_(staging, non_blocking=True)
             kv.page_table()  # flush any new page ids
             graph.replay()
-            return logits_buf
+            pair_host.copy_(pair_buf, non_blocking=False)  # syncs
+            return int(pair_host[0]), float(pair_host[1]), p_buf
 
         return step

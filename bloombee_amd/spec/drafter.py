@@ -33,6 +33,9 @@ class MultiDrafter:
         self.node_budget = node_budget
         self.max_depth = max_depth
         self.cost_ratio = cost_ratio
+        # store full per-node draft distributions (needed only by the
+        # stochastic SpecInfer verify; greedy verify ignores them)
+        self.collect_dists = True
         self.stats = AcceptanceStats(max_depth=max_depth + 2)
         self._kv = None
         self._kv_len = 0
@@ -85,16 +88,18 @@ class MultiDrafter:
             base = self._kv_len
             for j, tok in enumerate(toks):
                 kv.extend(1)
-                logits0 = self._gstep(int(tok), base + j)[0]
+                _, _, p0 = self._gstep(int(tok), base + j)
+            probs0 = p0.clone()  # graph buffer: chain replays overwrite it
         else:
             start = torch.full((1,), self._kv_len, dtype=torch.int32,
                                device=eng.device)
             kv.extend(new.shape[1])
             hidden = eng.stack.forward_inference(eng._embed(new), kv, start)
-            logits0 = eng.logits_for(hidden[:, -1]).float()[0]
+            probs0 = torch.softmax(
+                eng.logits_for(hidden[:, -1]).float()[0], -1)
         self._kv_len = history.numel()
         self.t_prefill = _time.monotonic() - t0
-        tree = self._expand(kv, logits0)
+        tree = self._expand(kv, probs0)
         self.t_chain = _time.monotonic() - t0 - self.t_prefill
         return tree
 
@@ -117,19 +122,19 @@ class MultiDrafter:
                 kv.extend(ids.shape[1])
                 hidden = eng._embed(ids)
                 hidden = eng.stack.forward_inference(hidden, kv, start)
-                logits0 = eng.logits_for(hidden[:, -1]).float()[0]
-                return self._expand(kv, logits0)
+                probs0 = torch.softmax(
+                    eng.logits_for(hidden[:, -1]).float()[0], -1)
+                return self._expand(kv, probs0)
             finally:
                 kv.close()
 
     @torch.no_grad()
-    def _expand(self, kv, logits0: torch.Tensor) -> TokenTree:
+    def _expand(self, kv, probs0: torch.Tensor) -> TokenTree:
         widths = plan_tree_shape(self.stats, self.node_budget,
                                  max_depth=self.max_depth,
                                  cost_ratio=self.cost_ratio)
         eng = self.draft
         tree = TokenTree()
-        probs0 = torch.softmax(logits0, -1)
         w0 = max(1, widths[0])
         top = probs0.topk(w0)
         # full draft dist stored per node: the exact SpecInfer
@@ -151,23 +156,21 @@ class MultiDrafter:
                     pos = kv.seqs[0].l_spec
                     kv.extend(1, speculative=True)
                     if gstep is not None:
-                        lg = gstep(chain_tok, pos)[0]
+                        # whole step (incl. softmax/argmax/prob) inside the
+                        # graph; ONE 8-byte pinned readback per node
+                        t, pr, p = gstep(chain_tok, pos)
+                        dist = p.clone() if self.collect_dists else None
                     else:
                         h = eng._embed(torch.tensor([[chain_tok]]))
                         sp = torch.tensor([pos], dtype=torch.int32,
                                           device=eng.device)
                         h = eng.stack.forward_inference(h, kv, sp)
                         lg = eng.logits_for(h[:, -1]).float()[0]
-                p = torch.softmax(lg, -1)
-                # one host sync for (token, prob) instead of two .item()s
-                pv, ti = p.max(-1)
-                pair = torch.stack((ti.to(torch.float32), pv)).cpu()
-                t = int(pair[0])
-                with lock:
-                    # clone: the graphed path reuses the logits buffer
-                    chain_parent = tree.add(
-                        t, chain_parent, float(pair[1]),
-                        dist=p.clone() if gstep is not None else p)
+                        p = torch.softmax(lg, -1)
+                        pv, ti = p.max(-1)
+                        pair = torch.stack((ti.to(torch.float32), pv)).cpu()
+                        t, pr, dist = int(pair[0]), float(pair[1]), p
+                    chain_parent = tree.add(t, chain_parent, pr, dist=dist)
                 chain_tok = t
                 local.append(t)
             return local
