@@ -267,3 +267,65 @@ class LocalEngine:
             return int(pair_host[0]), float(pair_host[1]), p_buf
 
         return step
+
+    def make_graphed_chain(self, kv: SessionHandle, depth: int):
+        """A whole greedy draft CHAIN (depth single-token steps, each step's
+        argmax fed on-device into the next embed) as ONE hipGraph replay —
+        one host round-trip per drafting round instead of per node (the
+        per-node graphed step still costs ~150 µs of host floor,
+        benchmarks/gstep_micro.py).
+
+        Caller per round: rollback, extend(depth, speculative=True),
+        page_table(), then chain(token, position0) -> (tokens[depth],
+        probs[depth]) on the host. Off-GPU returns None (callers fall back
+        to per-node stepping)."""
+        from bloombee_amd.config import get_config
+
+        if self.device.type != "cuda" or not get_config().use_hip_graphs:
+            return None
+        staging = torch.zeros(2, dtype=torch.long, pin_memory=True)
+        dev_in = torch.zeros(2, dtype=torch.long, device=self.device)
+        out_host = torch.zeros(2, depth, dtype=torch.float32, pin_memory=True)
+
+        def _body():
+            toks = torch.zeros(depth, dtype=torch.float32,
+                               device=self.device)
+            prbs = torch.zeros(depth, dtype=torch.float32,
+                               device=self.device)
+            cur = dev_in[0:1]
+            pos0 = dev_in[1:2]
+            for i in range(depth):
+                h = self._embed(cur.view(1, 1))
+                sp = (pos0 + i).to(torch.int32)
+                h = self.stack.forward_inference(h, kv, sp)
+                logits = self.logits_for(h[:, -1]).float()
+                p = torch.softmax(logits[0], -1)
+                pv, ti = p.max(-1)
+                toks[i] = ti.to(torch.float32)
+                prbs[i] = pv
+                cur = ti.view(1)
+            return torch.stack((toks, prbs))
+
+        for _ in range(2):  # allocator/kernel warmup
+            kv.extend(depth, speculative=True)
+            kv.page_table()
+            _body()
+            kv.rollback()
+        graph = torch.cuda.CUDAGraph()
+        kv.extend(depth, speculative=True)
+        kv.page_table()
+        with torch.cuda.graph(graph, capture_error_mode="thread_local"):
+            out_buf = _body()
+        kv.rollback()
+
+        def chain(tok: int, position0: int):
+            staging[0] = tok
+            staging[1] = position0
+            dev_in.copy_(staging, non_blocking=True)
+            kv.page_table()
+            graph.replay()
+            out_host.copy_(out_buf, non_blocking=False)  # syncs
+            return ([int(t) for t in out_host[0]],
+                    [float(p) for p in out_host[1]])
+
+        return chain

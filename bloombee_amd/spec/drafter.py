@@ -55,6 +55,10 @@ class MultiDrafter:
         self._kv = self.draft.kv_pool.allocate(1, max_len)
         self._kv_len = 0
         self._gstep = self.draft.make_graphed_decoder(self._kv)
+        # whole-chain graph: one replay (and one host round-trip) per
+        # drafting round; None off-GPU
+        self._gchain = self.draft.make_graphed_chain(self._kv,
+                                                     self.max_depth)
 
     def close_session(self) -> None:
         kv = getattr(self, "_kv", None)
@@ -63,6 +67,7 @@ class MultiDrafter:
         self._kv = None
         self._kv_len = 0
         self._gstep = None
+        self._gchain = None
 
     @torch.no_grad()
     def build_tree_incremental(self, history: torch.Tensor) -> TokenTree:
@@ -135,6 +140,21 @@ class MultiDrafter:
                                  cost_ratio=self.cost_ratio)
         eng = self.draft
         tree = TokenTree()
+        gchain = (getattr(self, "_gchain", None)
+                  if kv is getattr(self, "_kv", None) else None)
+        if gchain is not None:
+            # chain-mode drafting: the root is argmax(probs0) and the whole
+            # depth-max chain comes back from ONE graph replay; the planner's
+            # widths are moot (greedy verify, single branch)
+            pv, ti = probs0.max(-1)
+            root = tree.add(int(ti), -1, float(pv))
+            pos0 = kv.seqs[0].l_spec
+            kv.extend(self.max_depth, speculative=True)
+            toks, prbs = gchain(tree.tokens[root], pos0)
+            parent = root
+            for t, pr in zip(toks, prbs):
+                parent = tree.add(int(t), parent, float(pr))
+            return tree
         w0 = max(1, widths[0])
         top = probs0.topk(w0)
         # full draft dist stored per node: the exact SpecInfer
