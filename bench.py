@@ -207,6 +207,9 @@ def main():
     ap.add_argument("--micro-batches", type=int, default=0)
     ap.add_argument("--tp", type=int, default=1,
                     help="tensor-parallel degree within the pipeline (world %% tp == 0)")
+    ap.add_argument("--tp-mode", default="tensor", choices=["tensor", "expert"],
+                    help="what the intra-stage group shards: tensor dims "
+                         "(llama TP) or MoE experts (mixtral EP)")
     ap.add_argument("--mode", default="pipeline", choices=["pipeline", "swarm"],
                     help="pipeline = raw RCCL pipeline stages (flagship); "
                          "swarm = the full decentralized serving stack "
@@ -240,7 +243,7 @@ def main():
     stage = PipelineStage(args.model, device, global_batch,
                           micro_batches=args.micro_batches,
                           kv_max_tokens=kv_tokens, max_session_len=session_len,
-                          tp=args.tp)
+                          tp=args.tp, tp_mode=args.tp_mode)
 
     gen = torch.Generator().manual_seed(42)
     V = stage.config.vocab_size
@@ -310,8 +313,9 @@ def main():
                 "model": args.model,
                 "global_batch": global_batch,
                 "seq_len": args.prompt,
-                "parallelism": (f"pp{world // args.tp}xtp{args.tp}" if args.tp > 1
-                                else f"pp{world}"),
+                "parallelism": (f"pp{world // args.tp}x"
+                                f"{'ep' if args.tp_mode == 'expert' else 'tp'}"
+                                f"{args.tp}" if args.tp > 1 else f"pp{world}"),
                 "micro_batches": stage.M,
                 "p50_step_ms": round(p50, 3),
             },

@@ -46,7 +46,7 @@ class PipelineStage:
     def __init__(self, config_or_name, device, global_batch: int,
                  micro_batches: int = 0, seed: int = 0,
                  kv_max_tokens: int = 1 << 17, max_session_len: int = 4096,
-                 tp: int = 1):
+                 tp: int = 1, tp_mode: str = "tensor"):
         cfg = (config_or_name if isinstance(config_or_name, ModelConfig)
                else resolve_config(config_or_name))
         self.config = cfg
@@ -85,7 +85,23 @@ class PipelineStage:
                                  self.pp_world)
         logger.info(f"rank {self.rank}/{self.world} (pp {self.pp_stage} tp "
                     f"{self.tp_rank}): layers [{start}, {end})")
-        if tp > 1:
+        if tp > 1 and tp_mode == "expert":
+            # expert parallelism inside a stage (BASELINE config 4 at scale;
+            # beyond the reference, which runs all experts locally): every
+            # rank keeps the full attention/router weights and ONLY its
+            # expert shard; each block's partial MoE output crosses the
+            # group with one RCCL all-reduce (parallel/expert.py). Non-MoE
+            # blocks compute replicated.
+            from bloombee_amd.parallel.expert import shard_experts
+            self.stack = BlockStack(cfg, start, end, device=self.device,
+                                    seed=seed)
+            n_moe = 0
+            for blk in self.stack.blocks:
+                if hasattr(blk, "expert_gate_up_w"):
+                    shard_experts(blk, self.tp_rank, tp, group=self.tp_group)
+                    n_moe += 1
+            logger.info("EP group %d-way: %d MoE blocks sharded", tp, n_moe)
+        elif tp > 1:
             from bloombee_amd.parallel.tensor import TPBlockStack
             self.stack = TPBlockStack(cfg, start, end, device=self.device,
                                       seed=seed, group=self.tp_group)
