@@ -196,7 +196,100 @@ def bench_spec(steps=32, self_draft=False, draft_q4=False):
                       "verify_ms_per_round": round(verify_s / rounds * 1e3, 2)}))
 
 
+def bench_kv_multiplex(steps=24):
+    """Micro-batch KV multiplexing on the REAL serving stack, 1 GPU: two
+    servers (llama-3-8b halves) whose pools hold only the resident window
+    (3 x micro-batch rows) — the full batch's KV lives in pinned host
+    snapshots and cycles through the device while slices compute
+    (kv/paged.py swap_{in,out}_rows; ref memory_cache_manager.py:944-1371).
+    This is a CAPACITY mode (sessions beyond HBM), so the number to read
+    is the achieved throughput under staging, not a speedup."""
+    import dataclasses
+
+    import torch.nn.functional as F
+
+    from bloombee_amd import config as bconf
+    from bloombee_amd import ops
+    from bloombee_amd.client.config import ClientConfig
+    from bloombee_amd.client.routing import RemoteSequenceManager
+    from bloombee_amd.client.session import InferenceSession
+    from bloombee_amd.models.base import resolve_config
+    from bloombee_amd.net.channels import channels
+    from bloombee_amd.net.dht import Dht
+    from bloombee_amd.server import Server
+
+    model = "llama-3-8b"
+    cfg = resolve_config(model)
+    B, prompt, mbs = 32, 256, 8
+    session_len = prompt + steps + 16
+    cfg0 = bconf.get_config()
+    mb = dataclasses.replace(cfg0.microbatch, enabled=True,
+                             micro_batch_size=mbs, min_batch_to_split=16,
+                             kv_multiplex=True)
+    bconf.set_config(dataclasses.replace(cfg0, microbatch=mb))
+    channels.enable(DEV)
+    boot = Dht()
+    servers = []
+    results = {}
+    try:
+        # pool: resident window (3*mbs rows) + slack; full B=32 would need
+        # 32/24x more
+        kv_tokens = 3 * mbs * ((session_len + 15) // 16 + 1) * 16 + 2048
+        L = cfg.num_hidden_layers
+        for rng in [(0, L // 2), (L // 2, L)]:
+            srv = Server(model, initial_peers=[boot.endpoint],
+                         block_indices=rng, device=DEV, seed=0,
+                         kv_max_tokens=kv_tokens, update_period=10.0,
+                         max_batch_size=2048)
+            srv.run_in_background()
+            servers.append(srv)
+        ccfg = ClientConfig(initial_peers=[boot.endpoint], keep_history=False,
+                            step_timeout=300.0)
+        mgr = RemoteSequenceManager(ccfg, model, L)
+        session = InferenceSession(mgr, max_length=session_len, config=ccfg)
+        gen = torch.Generator().manual_seed(0)
+        dev = torch.device(DEV)
+        embed = (torch.randn(cfg.vocab_size, cfg.hidden_size, generator=gen)
+                 .mul_(0.02).to(cfg.dtype).to(dev))
+        norm_w = torch.ones(cfg.hidden_size, dtype=cfg.dtype, device=dev)
+        head = (torch.randn(cfg.vocab_size, cfg.hidden_size, generator=gen)
+                .mul_(0.02).to(cfg.dtype).to(dev))
+        ids = torch.randint(0, cfg.vocab_size, (B, prompt), generator=gen)
+        out = session.step(F.embedding(ids.to(dev), embed))
+        tok = F.linear(ops.rms_norm(out[:, -1], norm_w, cfg.rms_norm_eps),
+                       head).float().argmax(-1)
+        for _ in range(4):
+            out = session.step(F.embedding(tok.view(-1, 1), embed))
+            tok = F.linear(ops.rms_norm(out[:, -1], norm_w, cfg.rms_norm_eps),
+                           head).float().argmax(-1)
+        torch.cuda.synchronize()
+        t0 = time.monotonic()
+        for _ in range(steps):
+            out = session.step(F.embedding(tok.view(-1, 1), embed))
+            tok = F.linear(ops.rms_norm(out[:, -1], norm_w, cfg.rms_norm_eps),
+                           head).float().argmax(-1)
+        torch.cuda.synchronize()
+        dt = time.monotonic() - t0
+        session.close()
+        mgr.shutdown()
+        results = {"config": "KV-multiplexed serving (resident window "
+                             f"{3 * mbs}/{B} rows on device)",
+                   "tokens_per_s": round(B * steps / dt, 1),
+                   "ms_per_step": round(dt / steps * 1e3, 2),
+                   "pool_tokens_per_server": kv_tokens,
+                   "full_batch_tokens_needed":
+                       B * ((session_len + 15) // 16 + 1) * 16}
+        print(json.dumps(results))
+    finally:
+        bconf.set_config(cfg0)
+        for srv in servers:
+            srv.shutdown()
+        boot.shutdown()
+        channels.disable()
+
+
 ALL = {"offload": bench_offload, "mixtral": bench_mixtral,
+       "kv_multiplex": bench_kv_multiplex,
        "spec": bench_spec,
        "spec_w4": lambda: bench_spec(draft_q4=True),
        "spec_selfdraft": lambda: bench_spec(self_draft=True)}
