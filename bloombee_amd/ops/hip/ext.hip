@@ -632,7 +632,13 @@ torch::Tensor gemm_skinny(torch::Tensor A, torch::Tensor W,
     // BW-bound — once N/64 covers the 256 CUs extra splits only add combine
     // traffic; small-N shapes are per-block-latency-bound and want
     // (N/64)*ksplit ~ 512-1024 blocks, capped at 8.
-    if (N / 64 >= 256) ksplit = 1;
+    // BBAMD_GEMM_KSPLIT_BIG: experiment knob for the in-context latency
+    // parking the PMC shows on the big MLP GEMMs (prof_ctx r2k).
+    static const int ks_big = [] {
+      const char* e = std::getenv("BBAMD_GEMM_KSPLIT_BIG");
+      return e ? std::atoi(e) : 1;
+    }();
+    if (N / 64 >= 256) ksplit = ks_big;
     else ksplit = (int)std::min<long>(K / 512,
                                       std::max<long>(1, std::min<long>(8, 1024 / (N / 64))));
     ksplit = std::max(1, ksplit);
