@@ -38,6 +38,11 @@ class KVConfig:
     # not set explicitly (the reference sizes from --attn_cache_tokens).
     max_tokens: Optional[int] = field(default_factory=lambda: _env("KV_MAX_TOKENS", None, int))
     alloc_timeout: float = field(default_factory=lambda: _env("KV_ALLOC_TIMEOUT", 60.0))
+    # BBAMD_MIXED_ATTN=1: host-swapped sessions resume DECODING with their
+    # committed KV left in host memory (mixed-device attention, exact
+    # log-sum-exp merge) instead of being restored to HBM first (ref
+    # _mixed_device_attention, pytorch_backend.py:969-1014)
+    mixed_attn: bool = field(default_factory=lambda: _env("MIXED_ATTN", False))
 
 
 @dataclass

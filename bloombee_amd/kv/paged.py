@@ -448,25 +448,9 @@ class SessionHandle:
     def swap_in(self) -> None:
         """Restore a swapped-out session: take fresh pages, copy the host
         snapshot back, rewrite the page table."""
-        snap = getattr(self, "_swapped", None)
-        if snap is None:
+        if not self.is_swapped:
             return
-        if isinstance(snap, tuple) and snap[0] == "disk":
-            path = snap[1]
-            snap = torch.load(path, map_location="cpu", weights_only=False)
-            os.unlink(path)
-        if isinstance(snap, tuple) and snap[0] == "q4":
-            from bloombee_amd import ops as _ops
-
-            def _deq(q):
-                if q is None:
-                    return None
-                packed, scale, zero, shape, dtype = q
-                return _ops.quant4_unpack(packed, scale, zero,
-                                          dtype=dtype).reshape(shape)
-
-            snap = [[(_deq(kq), _deq(vq)) for kq, vq in per_layer]
-                    for per_layer in snap[1]]
+        snap = self._materialize_snapshot()
         cache = self.cache
         on_gpu = cache.device.type == "cuda"
         stream = torch.cuda.Stream(cache.device) if on_gpu else None
@@ -505,6 +489,98 @@ class SessionHandle:
     @property
     def is_swapped(self) -> bool:
         return getattr(self, "_swapped", None) is not None
+
+    # -- mixed-device decode (host prefix + device recent segment) --------
+    @property
+    def pos_offset(self) -> int:
+        """Absolute position of the device pool's local position 0: after
+        swap_in_as_prefix the committed history lives in host tensors and
+        only tokens >= pos_offset occupy device pages. RoPE must use
+        absolute positions (local + pos_offset); the paged kernels use
+        local ones."""
+        return getattr(self, "_pos_offset", 0)
+
+    def host_prefix(self, layer: int):
+        """(k_host, v_host) as (B, Hkv, S_host, D) pinned CPU tensors for
+        `layer`, or None when the session is fully device-resident. The
+        attention path merges them with the device segment via the exact
+        log-sum-exp composition (ops.attn_paged_mixed)."""
+        hp = getattr(self, "_host_prefix", None)
+        return hp[layer] if hp is not None else None
+
+    def _materialize_snapshot(self):
+        """Resolve a whole-session snapshot (possibly disk-spilled or
+        4-bit-compressed) into per-seq per-layer (k_host, v_host) page
+        stacks — shared by swap_in and swap_in_as_prefix."""
+        snap = getattr(self, "_swapped", None)
+        if isinstance(snap, tuple) and snap[0] == "disk":
+            path = snap[1]
+            snap = torch.load(path, map_location="cpu", weights_only=False)
+            os.unlink(path)
+        if isinstance(snap, tuple) and snap[0] == "q4":
+            from bloombee_amd import ops as _ops
+
+            def _deq(q):
+                if q is None:
+                    return None
+                packed, scale, zero, shape, dtype = q
+                return _ops.quant4_unpack(packed, scale, zero,
+                                          dtype=dtype).reshape(shape)
+
+            snap = [[(_deq(kq), _deq(vq)) for kq, vq in per_layer]
+                    for per_layer in snap[1]]
+        return snap
+
+    def swap_in_as_prefix(self) -> None:
+        """Convert a WHOLE-swapped session into mixed-device form instead of
+        restoring it to HBM (ROUND2 item 7 / reference
+        _mixed_device_attention, pytorch_backend.py:969-1014): the
+        committed history becomes per-layer host tensors (B, Hkv, S, D);
+        the device pool restarts empty at local position 0 and only NEW
+        tokens take device pages. Decode steps then merge both segments
+        exactly (ops.attn_paged_mixed). Requires lockstep sequences and no
+        speculative tokens (both true for swapped idle sessions)."""
+        if not self.is_swapped:
+            return
+        cache = self.cache
+        S = self.seqs[0].l_acc
+        if any(s.l_acc != S or s.l_spec != S for s in self.seqs):
+            raise PagedKVError("mixed-device restore needs lockstep, "
+                               "committed-only sequences")
+        snap = self._materialize_snapshot()
+        counts = self._swapped_counts
+        host: List[tuple] = []
+        for l in range(cache.num_layers):
+            hkv = cache.num_kv_heads_per_layer[l]
+            d = cache.head_dim_per_layer[l]
+            P = cache.page_size
+            ks, vs = [], []
+            for b in range(self.batch_size):
+                kh, vh = snap[b][l]  # (npages, Hkv, P, D) / (npages, Hkv, D, P)
+                k_flat = kh.permute(1, 0, 2, 3).reshape(hkv, -1, d)[:, :S]
+                if vh is None:
+                    v_flat = k_flat
+                else:
+                    # (npages, Hkv, D, P) -> (Hkv, D, npages*P) -> (Hkv, S, D)
+                    v_flat = (vh.permute(1, 2, 0, 3)
+                              .reshape(hkv, d, -1)[:, :, :S]
+                              .transpose(1, 2).contiguous())
+                ks.append(k_flat)
+                vs.append(v_flat)
+            host.append((torch.stack(ks), torch.stack(vs)))
+        old = getattr(self, "_host_prefix", None)
+        if old is not None:
+            host = [(torch.cat([ok, nk], dim=2), torch.cat([ov, nv], dim=2))
+                    for (ok, ov), (nk, nv) in zip(old, host)]
+        self._host_prefix = host
+        self._pos_offset = self.pos_offset + S
+        for b, s in enumerate(self.seqs):
+            s.l_acc = 0
+            s.l_spec = 0
+            assert not s.pages
+        del self._swapped_counts
+        self._swapped = None
+        _ = counts  # page ids were freed at swap_out; counts informational
 
     # -- row-granular staging (micro-batch KV multiplexing) ----------------
     def swap_out_rows(self, b0: int, b1: int) -> None:

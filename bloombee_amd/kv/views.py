@@ -31,3 +31,13 @@ class SessionView:
 
     def v_pages(self, layer: int) -> torch.Tensor:
         return self.handle.v_pages(layer)
+
+    @property
+    def pos_offset(self) -> int:
+        return self.handle.pos_offset
+
+    def host_prefix(self, layer: int):
+        hp = self.handle.host_prefix(layer)
+        if hp is None:
+            return None
+        return hp[0][self.b0:self.b1], hp[1][self.b0:self.b1]

@@ -155,11 +155,31 @@ class LlamaBlock(torch.nn.Module):
         kp = kv.k_pages(self.layer_index)
         vp = kv.v_pages(self.layer_index)
         pt = kv.page_table()
+        # mixed-device sessions (host KV prefix + device recent segment,
+        # kv.swap_in_as_prefix): paged indices are LOCAL to the device pool
+        # but rotary positions must stay ABSOLUTE
+        off = getattr(kv, "pos_offset", 0)
+        if off and position_ids is None:
+            position_ids = (start_pos.view(B, 1) + off
+                            + torch.arange(T, device=hidden.device)
+                            .view(1, T)).int()
         # fused RoPE + paged KV write on the raw GEMM output, then attention
         # reads q in place and emits the O-projection input — no transposes
         ops.rope_kv_write_(qkv, Hq, Hkv, cos, sin, position_ids, kp, vp, pt,
                            start_pos)
-        if tree_mask is not None:
+        hp = (kv.host_prefix(self.layer_index)
+              if hasattr(kv, "host_prefix") else None)
+        if hp is not None:
+            # exact two-segment merge (ref _mixed_device_attention): the
+            # host prefix is attended on the CPU, the recent segment via
+            # the paged pool; decode-only (backend gates T == 1)
+            D_ = self.D
+            q = (qkv[..., : Hq * D_].view(B, T, Hq, D_)
+                 .permute(0, 2, 1, 3).contiguous())
+            attn = ops.attn_paged_mixed(q, kp, vp, pt, start_pos + T,
+                                        hp[0], hp[1], self.scale)
+            attn = attn.permute(0, 2, 1, 3).reshape(B, T, Hq * D_)
+        elif tree_mask is not None:
             # tree-structured verify step (spec decoding): unfused q view +
             # the tree-mask attention path
             q = (qkv[..., : Hq * D].view(B, T, Hq, D)

@@ -9,6 +9,7 @@ from bloombee_amd.ops.interface import (  # noqa: F401
     alibi_slopes_for,
     attn_decode,
     attn_paged,
+    attn_paged_mixed,
     attn_paged_qkv,
     attn_prefill,
     gelu_tanh,

@@ -297,6 +297,17 @@ def linear(x: torch.Tensor, w: torch.Tensor, residual: Optional[torch.Tensor] = 
     return y
 
 
+def attn_paged_mixed(q, k_pages, v_pages, page_table, ctx_lens,
+                     host_k, host_v, scale=None):
+    """Mixed-device decode attention: host-resident KV prefix (CPU fp32)
+    + device-resident recent segment, merged with the exact log-sum-exp
+    composition (ref _mixed_device_attention, pytorch_backend.py:969-1014).
+    Device-agnostic torch composition — the host segment's matmuls RUN on
+    the CPU by construction, which is the point of the mode."""
+    return ref.attn_paged_mixed(q, k_pages, v_pages, page_table, ctx_lens,
+                                host_k, host_v, scale)
+
+
 def moe_gemm_grouped(A: torch.Tensor, W: torch.Tensor, off: torch.Tensor,
                      rowmap: Optional[torch.Tensor] = None,
                      scale: Optional[torch.Tensor] = None,

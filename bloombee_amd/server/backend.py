@@ -207,8 +207,26 @@ class StackBackend:
                 return out_
 
         def _compute():
+            nonlocal start_pos
+            T_ = hidden.shape[1]
+            plain_decode = (T_ == 1 and not speculative and tree_mask is None
+                            and position_ids is None)
             if handle.is_swapped:
-                handle.swap_in()
+                from bloombee_amd.config import get_config as _gc
+                if _gc().kv.mixed_attn and plain_decode:
+                    # capacity mode: committed KV stays host-side; decode
+                    # merges it with the device-resident recent segment
+                    handle.swap_in_as_prefix()
+                else:
+                    handle.swap_in()
+            # host-prefixed sessions use pool-local positions (rotary stays
+            # absolute inside the blocks via kv.pos_offset)
+            if handle.pos_offset:
+                if hidden.shape[1] > 1 or speculative or tree_mask is not None:
+                    raise ValueError(
+                        "mixed-device (host-prefix) sessions are decode-only;"
+                        " disable BBAMD_MIXED_ATTN for this workload")
+                start_pos = start_pos - handle.pos_offset
             h = hidden.to(self.device, non_blocking=True)
             if h.dtype != self.config.dtype:
                 h = h.to(self.config.dtype)
@@ -254,7 +272,7 @@ class StackBackend:
             if (self._use_graphs and T == 1 and not speculative
                     and prompts is None and position_ids is None
                     and tree_mask is None and adapter is None
-                    and start_pos == cur):
+                    and start_pos == cur and handle.pos_offset == 0):
                 return self._graphed_decode(session_id, handle, h, start_pos)
             pos = (position_ids.to(self.device).int()
                    if position_ids is not None else None)

@@ -151,3 +151,47 @@ def test_mixed_device_attention_exact():
         hk, hv)
     assert torch.allclose(dense, mixed, atol=1e-4), \
         (dense - mixed).abs().max()
+
+
+def test_mixed_device_decode_after_prefix_swap():
+    """A host-swapped session resumes decoding WITHOUT restoring its KV to
+    the device: the committed history becomes per-layer host tensors and
+    every step merges host + device segments exactly (swap_in_as_prefix +
+    ops.attn_paged_mixed; ref _mixed_device_attention). Greedy tokens must
+    match the never-swapped run."""
+    from bloombee_amd.engine import LocalEngine
+
+    torch.manual_seed(0)
+    ref_eng = LocalEngine("llama-tiny", device="cpu", seed=11,
+                          kv_max_tokens=1 << 12)
+    gen = torch.Generator().manual_seed(3)
+    prompt = torch.randint(0, 1000, (2, 40), generator=gen)
+    kv = ref_eng.kv_pool.allocate(2, 64)
+    toks = [ref_eng.prefill(prompt, kv)]
+    for _ in range(5):
+        toks.append(ref_eng.decode_step(toks[-1], kv))
+    expect = torch.stack(toks, 1)
+    kv.close()
+
+    eng = LocalEngine("llama-tiny", device="cpu", seed=11,
+                      kv_max_tokens=1 << 12)
+    kv = eng.kv_pool.allocate(2, 64)
+    tok = eng.prefill(prompt, kv)
+    kv.swap_out()
+    assert kv.is_swapped
+    kv.swap_in_as_prefix()
+    assert not kv.is_swapped
+    assert kv.pos_offset == 40 and kv.lengths == [0, 0]
+    assert kv.host_prefix(0)[0].shape[2] == 40
+    got = [tok]
+    for _ in range(5):
+        got.append(eng.decode_step(got[-1], kv))
+    got = torch.stack(got, 1)
+    assert torch.equal(got, expect), (got, expect)
+    # a second swap cycle stacks onto the existing host prefix
+    kv.swap_out()
+    kv.swap_in_as_prefix()
+    assert kv.pos_offset == 45
+    nxt = eng.decode_step(got[:, -1], kv)
+    kv.close()
+    assert nxt.shape == (2,)
