@@ -85,7 +85,15 @@ class PipelineStage:
                                  self.pp_world)
         logger.info(f"rank {self.rank}/{self.world} (pp {self.pp_stage} tp "
                     f"{self.tp_rank}): layers [{start}, {end})")
-        if tp > 1 and tp_mode == "expert":
+        self.tp_mode = tp_mode
+        if tp > 1 and tp_mode == "context":
+            # context parallelism: full weights on every rank; PREFILL
+            # shards the sequence with ring attention (cp_prefill_llama)
+            # and all-gathers K/V so decode replicates locally
+            assert self.pp_world == 1, "cp mode is single-stage for now"
+            self.stack = BlockStack(cfg, start, end, device=self.device,
+                                    seed=seed)
+        elif tp > 1 and tp_mode == "expert":
             # expert parallelism inside a stage (BASELINE config 4 at scale;
             # beyond the reference, which runs all experts locally): every
             # rank keeps the full attention/router weights and ONLY its
@@ -250,8 +258,44 @@ class PipelineStage:
             return None
 
     @torch.no_grad()
+    def _prefill_cp(self, ids: Optional[torch.Tensor], T: int):
+        """Context-parallel one-shot prefill: the tp group shards the
+        SEQUENCE, runs ring attention per block, and all-gathers K/V into
+        every rank's pool (parallel/sequence.cp_prefill_llama)."""
+        from bloombee_amd.parallel.sequence import cp_prefill_llama
+
+        cfg = self.config
+        B = self.global_batch
+        assert self.kv.seqs[0].l_spec == 0, "CP prefill is one-shot"
+        assert T % self.tp == 0, "prompt length must divide by cp degree"
+        if self.is_client:
+            hid = F.embedding(ids.to(self.device), self.embed)
+        else:
+            hid = torch.empty(B, T, cfg.hidden_size, dtype=cfg.dtype,
+                              device=self.device)
+        if dist.is_initialized():
+            dist.broadcast(hid, src=0, group=self.tp_group)
+        Tl = T // self.tp
+        self.kv.extend(T)
+        shard = hid[:, self.tp_rank * Tl:(self.tp_rank + 1) * Tl].contiguous()
+        out_shard = cp_prefill_llama(self.stack, self.kv, shard,
+                                     self.tp_rank, self.tp, self.tp_group)
+        if self.tp > 1 and dist.is_initialized():
+            outs = [torch.empty_like(out_shard) for _ in range(self.tp)]
+            dist.all_gather(outs, out_shard.contiguous(),
+                            group=self.tp_group)
+            hidden = torch.cat(outs, dim=1)
+        else:
+            hidden = out_shard
+        if not self.is_client:
+            return None
+        return self._lm_head(hidden[:, -1])
+
+    @torch.no_grad()
     def prefill_round(self, ids: Optional[torch.Tensor], T: int) -> Optional[torch.Tensor]:
         """Prefill T prompt tokens (chunked by micro-batch over the batch dim)."""
+        if self.tp > 1 and self.tp_mode == "context":
+            return self._prefill_cp(ids, T)
         cfg = self.config
         B, M, mb = self.global_batch, self.M, self.mb
         pos0 = self.kv.seqs[0].l_spec

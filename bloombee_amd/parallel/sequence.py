@@ -46,8 +46,9 @@ def _fold(state, s, v, row_pos=None, col_pos=None):
 def _ring_exchange(t: torch.Tensor, rank: int, world: int,
                    group=None) -> torch.Tensor:
     """Send t to (rank+1) % world, receive the previous rank's tensor."""
+    t = t.contiguous()
     out = torch.empty_like(t)
-    send = dist.isend(t.contiguous(), (rank + 1) % world, group=group)
+    send = dist.isend(t, (rank + 1) % world, group=group)
     recv = dist.irecv(out, (rank - 1) % world, group=group)
     send.wait()
     recv.wait()
@@ -122,3 +123,104 @@ def ring_attention_local(q, k, v, world: int,
         m, l, acc = state
         outs.append((acc / l.clamp_min(1e-30)).to(q.dtype))
     return torch.cat(outs, dim=2)
+
+
+@torch.no_grad()
+def cp_prefill_llama(stack, kv_handle, hidden_shard: torch.Tensor,
+                     rank: int, world: int, group=None) -> torch.Tensor:
+    """Context-parallel ONE-SHOT prefill of a llama BlockStack (the serve
+    wiring for ring attention — ROUND2 item 8): the prompt's T_total =
+    world * T_loc tokens are sharded contiguously across the group; each
+    rank computes its shard's activations with ring attention (K/V blocks
+    rotate at Hkv width over xGMI and are GQA-expanded locally per fold),
+    and every layer's full K/V is all-gathered into THIS rank's paged pool
+    so decode proceeds locally afterwards (stage groups replicate decode
+    compute — PipelineStage tp_mode="context").
+
+    hidden_shard: (B, T_loc, H) embeddings of tokens
+    [rank*T_loc, (rank+1)*T_loc). The caller must have extended kv_handle
+    by T_total from position 0. Returns the rank's output hidden shard.
+    """
+    import torch.nn.functional as F
+
+    from bloombee_amd import ops
+    from bloombee_amd.ops import reference as refops
+
+    cfg = stack.config
+    B, T_loc, H = hidden_shard.shape
+    dev = hidden_shard.device
+    hidden = hidden_shard
+
+    def rms(x, w):
+        xf = x.float()
+        return (xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True)
+                                 + cfg.rms_norm_eps)).to(x.dtype) * w
+
+    for blk in stack.blocks:
+        if not (hasattr(blk, "qkv_w") and hasattr(blk, "gate_up_w")):
+            raise NotImplementedError(
+                "cp_prefill_llama supports llama-pattern blocks")
+        Hq, Hkv, D = blk.Hq, blk.Hkv, blk.D
+        G = Hq // Hkv
+        x = rms(hidden, blk.input_norm_w)
+        qkv = (F.linear(x, blk.qkv_w).view(B, T_loc, Hq + 2 * Hkv, D)
+               .permute(0, 2, 1, 3))
+        q, k, v = qkv.split([Hq, Hkv, Hkv], dim=1)
+        cos, sin = blk.rope.get(dev)
+        pos = (torch.arange(rank * T_loc, (rank + 1) * T_loc, device=dev)
+               .view(1, T_loc).expand(B, T_loc))
+        q, k = refops.rope_apply(q.contiguous(), k.contiguous(), cos, sin, pos)
+
+        # ring attention with Hkv-width exchange, GQA expansion per fold
+        qf = q.float()
+        state = (torch.full((B, Hq, T_loc, 1), float("-inf")),
+                 torch.zeros(B, Hq, T_loc, 1),
+                 torch.zeros(B, Hq, T_loc, D))
+        row_pos = torch.arange(rank * T_loc, (rank + 1) * T_loc)
+        k_cur = k.float().contiguous()
+        v_cur = v.float().contiguous()
+        src = rank
+        for step in range(world):
+            if src <= rank:
+                kx = k_cur.repeat_interleave(G, dim=1)
+                vx = v_cur.repeat_interleave(G, dim=1)
+                s = (qf @ kx.transpose(-1, -2)) * blk.scale
+                if src == rank:
+                    col_pos = torch.arange(src * T_loc, (src + 1) * T_loc)
+                    state = _fold(state, s, vx, row_pos, col_pos)
+                else:
+                    state = _fold(state, s, vx)
+            if step < world - 1:
+                k_cur = _ring_exchange(k_cur, rank, world, group)
+                v_cur = _ring_exchange(v_cur, rank, world, group)
+                src = (src - 1) % world
+        m, l, acc = state
+        attn = ((acc / l.clamp_min(1e-30)).to(hidden.dtype)
+                .permute(0, 2, 1, 3).reshape(B, T_loc, Hq * D))
+
+        h2 = hidden + F.linear(attn, blk.o_w)
+        y = rms(h2, blk.post_norm_w)
+        gu = F.linear(y, blk.gate_up_w)
+        g, u = gu.split([gu.shape[-1] // 2] * 2, dim=-1)
+        hidden = h2 + F.linear(
+            torch.nn.functional.silu(g.float()).to(u.dtype) * u, blk.down_w)
+
+        # full-sequence K/V into the local paged pool (decode runs locally)
+        if world > 1:
+            k = k.contiguous()
+            v = v.contiguous()
+            ks = [torch.empty_like(k) for _ in range(world)]
+            vs = [torch.empty_like(v) for _ in range(world)]
+            import torch.distributed as dist
+            dist.all_gather(ks, k, group=group)
+            dist.all_gather(vs, v, group=group)
+            k_full = torch.cat(ks, dim=2)
+            v_full = torch.cat(vs, dim=2)
+        else:
+            k_full, v_full = k, v
+        ops.kv_write(k_full.to(cfg.dtype), v_full.to(cfg.dtype),
+                     kv_handle.k_pages(blk.layer_index),
+                     kv_handle.v_pages(blk.layer_index),
+                     kv_handle.page_table(),
+                     torch.zeros(B, dtype=torch.int32, device=dev))
+    return hidden
