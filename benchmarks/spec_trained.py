@@ -132,7 +132,16 @@ def main():
     ap.add_argument("--draft", default="llama-spec-draft")
     ap.add_argument("--draft-q4", action="store_true",
                     help="additionally 4-bit-quantize the trained draft")
+    ap.add_argument("--swarm", action="store_true",
+                    help="run the spec loop THROUGH the serving stack: the "
+                         "trained target is saved to the per-block npy "
+                         "layout and served by two workers; the client "
+                         "drafts locally and verifies via "
+                         "InferenceSession.spec_step/spec_commit")
     args = ap.parse_args()
+
+    if args.swarm:
+        return run_swarm_spec(args)
 
     from bloombee_amd.spec.drafter import MultiDrafter
     from bloombee_amd.spec.tree import TokenTree
@@ -241,6 +250,149 @@ def main():
         "draft_prefill_ms_per_round": round(prefill_s / rounds * 1e3, 2),
         "verify_ms_per_round": round(verify_s / rounds * 1e3, 2),
     }))
+
+
+def run_swarm_spec(args):
+    """Speculative decoding over the REAL serving stack (ROUND3 item 5):
+    trained target weights are saved to the npy layout and served by two
+    workers (device data plane); the client drafts with its local trained
+    draft (whole-chain hipGraph) and verifies via the session's spec
+    protocol (tree rides the stream, paged KV commit/rollback per span)."""
+    import tempfile
+
+    import torch.nn.functional as F
+
+    from bloombee_amd import ops
+    from bloombee_amd.client.config import ClientConfig
+    from bloombee_amd.client.routing import RemoteSequenceManager
+    from bloombee_amd.client.session import InferenceSession
+    from bloombee_amd.models.base import resolve_config
+    from bloombee_amd.net.channels import channels
+    from bloombee_amd.net.dht import Dht
+    from bloombee_amd.server import Server
+    from bloombee_amd.server.from_pretrained import (save_block_weights,
+                                                     save_client_weights)
+    from bloombee_amd.spec.drafter import MultiDrafter
+    from bloombee_amd.spec.tree import TokenTree
+    from bloombee_amd.spec.verify import verify_tree_greedy
+
+    def _sync():
+        if DEV.startswith("cuda"):
+            torch.cuda.synchronize()
+
+    cfg = resolve_config(args.target)
+    grammar = Grammar(cfg.vocab_size, seed=0)
+    tgt_m = TrainableLM(args.target, DEV, seed=0)
+    drf_m = TrainableLM(args.draft, DEV, seed=1)
+    train_lm(tgt_m, grammar, args.steps_train, tag="target")
+    train_lm(drf_m, grammar, args.steps_train, tag="draft")
+
+    ckpt = tempfile.mkdtemp(suffix="-np")
+    for i, blk in enumerate(tgt_m.stack.blocks):
+        save_block_weights(blk, ckpt, i)
+    save_client_weights(ckpt, tgt_m.embed.data, tgt_m.norm_w.data,
+                        tgt_m.head.data)
+    embed = tgt_m.embed.data.clone()
+    norm_w = tgt_m.norm_w.data.clone()
+    head = tgt_m.head.data.clone()
+    draft = to_engine(drf_m, args.draft, 1 << 13,
+                      quantize_q4=args.draft_q4)
+    del tgt_m, drf_m
+
+    channels.enable(DEV)
+    boot = Dht()
+    servers = []
+    try:
+        L = cfg.num_hidden_layers
+        for rng in [(0, L // 2), (L // 2, L)]:
+            srv = Server(args.target, initial_peers=[boot.endpoint],
+                         block_indices=rng, device=DEV, seed=0,
+                         kv_max_tokens=1 << 14, update_period=10.0,
+                         checkpoint_dir=ckpt)
+            srv.run_in_background()
+            servers.append(srv)
+        ccfg = ClientConfig(initial_peers=[boot.endpoint],
+                            keep_history=False, step_timeout=120.0)
+        mgr = RemoteSequenceManager(ccfg, args.target, L)
+        maxlen = 32 + args.decode + args.node_budget + 32
+
+        def lm_head(hidden_last):
+            y = ops.rms_norm(hidden_last, norm_w, cfg.rms_norm_eps)
+            return F.linear(y, head).float()
+
+        prompt = grammar.sample(1, 32)
+        # ---- plain greedy over the swarm (the baseline spec must beat) --
+        session = InferenceSession(mgr, max_length=maxlen, config=ccfg)
+        out = session.step(F.embedding(prompt.to(DEV), embed))
+        tok = lm_head(out[:, -1]).argmax(-1)
+        for _ in range(4):
+            out = session.step(F.embedding(tok.view(1, 1), embed))
+            tok = lm_head(out[:, -1]).argmax(-1)
+        _sync()
+        t0 = time.monotonic()
+        for _ in range(args.decode):
+            out = session.step(F.embedding(tok.view(1, 1), embed))
+            tok = lm_head(out[:, -1]).argmax(-1)
+        _sync()
+        plain_tps = args.decode / (time.monotonic() - t0)
+        session.close()
+
+        # ---- speculative over the swarm ---------------------------------
+        session = InferenceSession(mgr, max_length=maxlen, config=ccfg,
+                                   allow_push=False)
+        drafter = MultiDrafter(draft, node_budget=args.node_budget,
+                               max_depth=args.max_depth, cost_ratio=0.10)
+        drafter.collect_dists = False
+        drafter.start_session(maxlen)
+        out = session.step(F.embedding(prompt.to(DEV), embed))
+        pending = int(lm_head(out[:, -1]).argmax(-1))
+        history = prompt[0].tolist()
+        emitted, rounds = 0, 0
+        _sync()
+        t0 = time.monotonic()
+        while emitted < args.decode:
+            sub = drafter.build_tree_incremental(
+                torch.tensor(history + [pending]))
+            tree = TokenTree()
+            tree.add(pending, -1, 1.0)
+            for i in range(len(sub)):
+                tree.add(sub.tokens[i],
+                         0 if sub.parents[i] == -1 else sub.parents[i] + 1,
+                         sub.probs[i])
+            toks = tree.token_tensor().view(1, -1)
+            pos = tree.position_ids(session.position).view(1, -1)
+            mask = tree.attention_mask().unsqueeze(0)
+            hid = F.embedding(toks.to(DEV), embed)
+            out, _keep = session.spec_step(hid, pos, mask)
+            logits = lm_head(out[0]).cpu()
+            acc, bonus = verify_tree_greedy(tree, logits, logits[0], start=0)
+            accepted = [0] + acc
+            session.spec_commit([accepted])
+            emit = [tree.tokens[i] for i in accepted]
+            history += emit
+            emitted += len(emit)
+            pending = bonus
+            rounds += 1
+            drafter.record_result(len(acc), offered_depth=args.max_depth)
+        _sync()
+        spec_tps = emitted / (time.monotonic() - t0)
+        session.close()
+        drafter.close_session()
+        mgr.shutdown()
+        print(json.dumps({
+            "config": "speculative decode OVER THE SWARM (trained pair, "
+                      "2 workers, device data plane)",
+            "tokens": emitted, "rounds": rounds,
+            "tokens_per_round": round(emitted / rounds, 2),
+            "plain_swarm_tokens_per_s": round(plain_tps, 1),
+            "spec_swarm_tokens_per_s": round(spec_tps, 1),
+            "spec_speedup": round(spec_tps / plain_tps, 2),
+        }))
+    finally:
+        for srv in servers:
+            srv.shutdown()
+        boot.shutdown()
+        channels.disable()
 
 
 if __name__ == "__main__":
