@@ -288,7 +288,40 @@ def bench_kv_multiplex(steps=24):
         channels.disable()
 
 
+def bench_families(steps=24):
+    """Per-family engine decode throughput on 1 GPU (coverage evidence:
+    every family runs its own HIP-kernel path — alibi for bloom, MQA
+    chunks for falcon, qk-norm for qwen3, grouped MoE for mixtral,
+    sliding/wide-head pools for gemma4)."""
+    from bloombee_amd.engine import LocalEngine
+
+    runs = [  # (preset, batch, prompt)
+        ("llama-2-7b", 32, 512),
+        ("qwen3-8b", 32, 512),
+        ("falcon-7b", 32, 512),
+        ("bloom-560m", 32, 512),
+        ("qwen3-0.6b", 32, 512),
+        ("gemma4-9b", 16, 512),
+        ("mixtral-8x7b-4l", 16, 512),
+    ]
+    for name, B, prompt in runs:
+        try:
+            eng = LocalEngine(name, device=DEV, seed=0,
+                              kv_max_tokens=B * (prompt + steps + 32) + 1024)
+            tps, ms = _decode_loop(eng, B, prompt, steps)
+            print(json.dumps({"family_bench": name, "batch": B,
+                              "prompt": prompt,
+                              "tokens_per_s": round(tps, 1),
+                              "ms_per_step": round(ms, 2)}))
+            del eng
+            if DEV.startswith("cuda"):
+                torch.cuda.empty_cache()
+        except Exception as e:  # noqa: BLE001 — report and continue
+            print(json.dumps({"family_bench": name, "error": str(e)[:200]}))
+
+
 ALL = {"offload": bench_offload, "mixtral": bench_mixtral,
+       "families": bench_families,
        "kv_multiplex": bench_kv_multiplex,
        "spec": bench_spec,
        "spec_w4": lambda: bench_spec(draft_q4=True),
