@@ -35,7 +35,7 @@ class Qwen3Block(LlamaBlock):
         cfg = self.config
 
         x = ops.rms_norm(hidden, self.input_norm_w, cfg.rms_norm_eps)
-        qkv = ops.linear(x, self.qkv_w)
+        qkv = self._lin(x, self.qkv_w, "qkv_w")
         # per-head q/k RMS norm on the fused buffer (contiguous D-sized rows)
         qk = qkv[..., :(Hq + Hkv) * D]
         q_flat = qkv[..., :Hq * D].reshape(-1, D)
@@ -51,10 +51,10 @@ class Qwen3Block(LlamaBlock):
         ops.rope_kv_write_(qkv, Hq, Hkv, cos, sin, position_ids, kp, vp, pt,
                            start_pos)
         attn = ops.attn_paged_qkv(qkv, Hq, Hkv, kp, vp, pt, start_pos, self.scale)
-        a = ops.linear(attn, self.o_w)
+        a = self._lin(attn, self.o_w, "o_w")
         h2, y = ops.rms_norm_residual(a, hidden, self.post_norm_w, cfg.rms_norm_eps)
-        return ops.linear(ops.swiglu(ops.linear(y, self.gate_up_w)), self.down_w,
-                          residual=h2)
+        return self._lin(ops.swiglu(self._lin(y, self.gate_up_w, "gate_up_w")),
+                         self.down_w, "down_w", residual=h2)
 
     def forward_train(self, hidden: torch.Tensor, start_pos: int = 0) -> torch.Tensor:
         B, T, H = hidden.shape

@@ -271,3 +271,29 @@ def test_llama_block_quantized_decode_close_to_full():
     num = (yf.float() - yq.float()).norm()
     den = yf.float().norm().clamp_min(1e-6)
     assert (num / den) < 0.2, (num / den)
+
+
+def test_qwen3_block_quantized_decode_close_to_full():
+    """Qwen3 inherits quantize_weights_q4 from LlamaBlock; its inference
+    path must route projections through _lin so the 4-bit weights are the
+    ones that run (regression: ops.linear calls bypassed the W4 table)."""
+    from bloombee_amd.engine import BlockStack
+    from bloombee_amd.models.base import resolve_config
+
+    cfg = resolve_config("qwen3-tiny")
+    full = BlockStack(cfg, 0, 1, device="cpu", seed=7)
+    quant = BlockStack(cfg, 0, 1, device="cpu", seed=7)
+    quant.blocks[0].quantize_weights_q4()
+    assert quant.blocks[0].qkv_w.numel() == 0
+    kvf, kvq = full.make_kv(1 << 10), quant.make_kv(1 << 10)
+    hf_, hq_ = kvf.allocate(2, 32), kvq.allocate(2, 32)
+    gen = torch.Generator().manual_seed(2)
+    x = (torch.randn(2, 4, cfg.hidden_size, generator=gen) * 0.1).to(cfg.dtype)
+    sp = torch.zeros(2, dtype=torch.int32)
+    hf_.extend(4)
+    hq_.extend(4)
+    yf = full.forward_inference(x.clone(), hf_, sp)
+    yq = quant.forward_inference(x.clone(), hq_, sp)
+    num = (yf.float() - yq.float()).norm()
+    den = yf.float().norm().clamp_min(1e-6)
+    assert (num / den) < 0.2, (num / den)
