@@ -1,27 +1,13 @@
 #!/bin/bash
-# UTCL2/TLB + L2 PMC probe for the in-context decode-GEMM parking question
+# UTCL1-TLB + L2 PMC probe for the in-context decode-GEMM parking question
 # (ROUND3.md item 1): same counters on (a) the standalone GEMM micro and
-# (b) the full bench.py decode step, compare gemm_skinny miss rates.
+# (b) the full bench.py decode step, compare gemm_skinny translation/L2
+# miss rates. Counter names verified against rocprofv3 -L on gfx950.
 set -x
 mkdir -p gpurun_out
 cd /root/repo
 
-rocprofv3 -L > gpurun_out/pmc_counters_list.txt 2>&1
-grep -oE '[A-Z][A-Z0-9_]+' gpurun_out/pmc_counters_list.txt | sort -u \
-  | grep -E 'UTCL|TCC_(REQ|MISS|HIT|EA_RDREQ|EA)' | head -40 \
-  > gpurun_out/pmc_candidates.txt
-cat gpurun_out/pmc_candidates.txt
-
-pick() {  # first candidate that matches regex $1
-  grep -m1 -E "$1" gpurun_out/pmc_candidates.txt
-}
-C_UTC_REQ=$(pick '^UTCL2.*(REQ|REQUEST)')
-C_UTC_MISS=$(pick '^UTCL2.*MISS')
-C_TCC_REQ=$(pick '^TCC_REQ')
-C_TCC_MISS=$(pick '^TCC_MISS')
-CTRS=$(echo "$C_UTC_REQ $C_UTC_MISS $C_TCC_REQ $C_TCC_MISS" | xargs)
-echo "chosen counters: $CTRS"
-[ -z "$CTRS" ] && { echo "no counters found"; exit 0; }
+CTRS="TCP_UTCL1_TRANSLATION_HIT TCP_UTCL1_TRANSLATION_MISS TCC_REQ TCC_MISS"
 
 echo "=== standalone GEMM micro ==="
 timeout 300 rocprofv3 --pmc $CTRS --output-format csv \
@@ -41,7 +27,7 @@ import csv, glob, collections
 def summarize(pat):
     agg = collections.defaultdict(lambda: collections.defaultdict(float))
     n = collections.Counter()
-    for f in glob.glob(pat):
+    for f in glob.glob(pat, recursive=True):
         with open(f) as fh:
             for row in csv.DictReader(fh):
                 kn = row.get("Kernel_Name") or row.get("Kernel Name") or ""
@@ -62,14 +48,18 @@ for tag, pat in [("micro", "gpurun_out/pmc_gemm_micro/**/*counter*.csv"),
             continue
         line = " ".join(f"{c}={v:.3e}" for c, v in sorted(ctrs.items()))
         print(f"{k}: n={n[k]} {line}")
-        for rq, ms in (("UTCL2", "UTCL2"), ("TCC", "TCC")):
-            reqs = [v for c, v in ctrs.items() if c.startswith(rq) and ("REQ" in c)]
-            miss = [v for c, v in ctrs.items() if c.startswith(ms) and "MISS" in c]
-            if reqs and miss and reqs[0] > 0:
-                print(f"    {rq} miss-rate: {miss[0]/reqs[0]*100:.2f}%")
+        th = ctrs.get("TCP_UTCL1_TRANSLATION_HIT", 0.0)
+        tm = ctrs.get("TCP_UTCL1_TRANSLATION_MISS", 0.0)
+        if th + tm > 0:
+            print(f"    UTCL1 translation miss-rate: {tm/(th+tm)*100:.3f}%")
+        tr = ctrs.get("TCC_REQ", 0.0)
+        tc = ctrs.get("TCC_MISS", 0.0)
+        if tr > 0:
+            print(f"    TCC (L2) miss-rate: {tc/tr*100:.2f}%")
 PYEOF
 cat gpurun_out/pmc_utcl2_summary.txt
-find gpurun_out/pmc_gemm_micro gpurun_out/pmc_bench -name '*.csv' -size +8M -delete
+find gpurun_out/pmc_gemm_micro gpurun_out/pmc_bench \
+  \( -name '*.csv' -size +8M \) -delete 2>/dev/null
 find gpurun_out/pmc_gemm_micro gpurun_out/pmc_bench -name '*.db' -delete 2>/dev/null
 du -sh gpurun_out
 echo PMC DONE
