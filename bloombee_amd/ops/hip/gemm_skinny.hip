@@ -124,11 +124,21 @@ __global__ __launch_bounds__(256) void gemm_skinny_v2_kernel(
     const unsigned short* __restrict__ bias,  // (N,) bf16 or null
     unsigned short* __restrict__ C,         // (M, N) bf16   (ksplit == 1)
     float* __restrict__ Cpart,              // (ksplit, M, N) f32 (ksplit > 1)
-    int M, int N, int K, int kchunk, int ksplit) {
+    int M, int N, int K, int kchunk, int ksplit,
+    const unsigned short* __restrict__ nw,  // (K,) rmsnorm weight or null
+    float eps) {
   constexpr int U = 4;                     // k-slices per pipeline set (128 k)
   constexpr int KSTEP = 256;               // A elements staged per stage
   constexpr int RSTRIDE = KSTEP + 8;       // padded LDS row stride (elements)
   __shared__ __attribute__((aligned(16))) unsigned short atile[32 * RSTRIDE];
+  // Fused rmsnorm of A (nw != null): removes the separate rms_norm launch +
+  // its read/write from the decode step (launch-count reduction — the PMC
+  // work in profiles/r02 §11 showed the in-context GEMM cost is boundary/
+  // drain, not cache, so fewer kernels is the lever). Each WG redundantly
+  // streams A (M*K bf16, L2-resident after the first WG) to compute per-row
+  // inv-rms, then scales A fragments by invr[m]*nw[k] as they stage to LDS.
+  __shared__ float ssp[256];
+  __shared__ float invr[32];
 
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
@@ -162,6 +172,7 @@ __global__ __launch_bounds__(256) void gemm_skinny_v2_kernel(
   // Thread piece j: row = (tid + j*256)/32 clamped, 8 elements at
   // ((tid + j*256)%32)*8 within the stage.
   short8 aregs[(32 * KSTEP) / (256 * 8)];  // 4 pieces per thread
+  short8 nregs[(32 * KSTEP) / (256 * 8)];  // matching norm-weight pieces
   auto load_a = [&](int ks) {
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
@@ -169,14 +180,24 @@ __global__ __launch_bounds__(256) void gemm_skinny_v2_kernel(
       const int row = min(flat / 32, M - 1);
       const int off = (flat % 32) * 8;
       aregs[j] = *reinterpret_cast<const short8*>(A + (long)row * K + ks + off);
+      if (nw)
+        nregs[j] = *reinterpret_cast<const short8*>(nw + ks + off);
     }
   };
   auto store_a = [&]() {
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
       const int flat = threadIdx.x + j * 256;
+      short8 v = aregs[j];
+      if (nw) {
+        const float ir = invr[flat / 32 >= M ? M - 1 : flat / 32];
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          v[e] = (short)f2bf(bf2f((unsigned short)v[e]) * ir *
+                             bf2f((unsigned short)nregs[j][e]));
+      }
       *reinterpret_cast<short8*>(atile + (flat / 32) * RSTRIDE + (flat % 32) * 8) =
-          aregs[j];
+          v;
     }
   };
 
