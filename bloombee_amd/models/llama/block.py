@@ -149,8 +149,19 @@ class LlamaBlock(torch.nn.Module):
         Hq, Hkv, D = self.Hq, self.Hkv, self.D
         cfg = self.config
 
-        x = ops.rms_norm(hidden, self.input_norm_w, cfg.rms_norm_eps)
-        qkv = self._lin(x, self.qkv_w, "qkv_w")            # (B, T, (Hq+2Hkv)D)
+        # Decode-shaped GPU steps fold both per-layer rmsnorms into the
+        # skinny GEMM's A-operand stage (ops.linear norm=...): two fewer
+        # launches + elementwise passes per layer. W4/LoRA blocks and
+        # prefill keep the separate-norm path.
+        fuse_norm = (getattr(self, "_w4", None) is None
+                     and getattr(self, "lora_delta", None) is None
+                     and ops.fuse_norm_linear_ok(hidden, self.qkv_w))
+        if fuse_norm:
+            qkv = ops.linear(hidden, self.qkv_w,
+                             norm=(self.input_norm_w, cfg.rms_norm_eps))
+        else:
+            x = ops.rms_norm(hidden, self.input_norm_w, cfg.rms_norm_eps)
+            qkv = self._lin(x, self.qkv_w, "qkv_w")        # (B, T, (Hq+2Hkv)D)
         cos, sin = self.rope.get(hidden.device)
         kp = kv.k_pages(self.layer_index)
         vp = kv.v_pages(self.layer_index)
@@ -190,6 +201,13 @@ class LlamaBlock(torch.nn.Module):
         else:
             attn = ops.attn_paged_qkv(qkv, Hq, Hkv, kp, vp, pt, start_pos,
                                       self.scale)
+        if fuse_norm:
+            # h2 = hidden + o(attn) via the GEMM's residual epilogue; the
+            # post-attention rmsnorm folds into the gate_up A-stage
+            h2 = ops.linear(attn, self.o_w, residual=hidden)
+            gu = ops.linear(h2, self.gate_up_w,
+                            norm=(self.post_norm_w, cfg.rms_norm_eps))
+            return ops.linear(ops.swiglu(gu), self.down_w, residual=h2)
         a = self._lin(attn, self.o_w, "o_w")
 
         # h2 = hidden + a fused into the post-attention norm

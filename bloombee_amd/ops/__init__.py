@@ -12,6 +12,7 @@ from bloombee_amd.ops.interface import (  # noqa: F401
     attn_paged_mixed,
     attn_paged_qkv,
     attn_prefill,
+    fuse_norm_linear_ok,
     gelu_tanh,
     hip_ops,
     kv_gather,

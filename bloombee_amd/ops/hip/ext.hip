@@ -604,7 +604,8 @@ torch::Tensor gemm_w4(torch::Tensor A, torch::Tensor Wq, torch::Tensor scale,
 // C(M,N) = A(M,K) @ W(N,K)^T [+ residual (M,N)] [+ bias (N,)], M <= 32.
 torch::Tensor gemm_skinny(torch::Tensor A, torch::Tensor W,
                           c10::optional<torch::Tensor> residual,
-                          c10::optional<torch::Tensor> bias, long ksplit_req) {
+                          c10::optional<torch::Tensor> bias, long ksplit_req,
+                          c10::optional<torch::Tensor> norm_w, double eps) {
   CHECK_DEV(A); CHECK_BF16(A); CHECK_CONTIG(A);
   CHECK_DEV(W); CHECK_BF16(W); CHECK_CONTIG(W);
   const int K = A.size(-1);
@@ -613,6 +614,14 @@ torch::Tensor gemm_skinny(torch::Tensor A, torch::Tensor W,
   TORCH_CHECK(M <= 32, "gemm_skinny is for M <= 32, got ", M);
   TORCH_CHECK(W.size(1) == K, "A/W K mismatch");
   TORCH_CHECK(K % 32 == 0 && N % 64 == 0, "K%32, N%64 required");
+  const unsigned short* nwp = nullptr;
+  if (norm_w.has_value()) {
+    TORCH_CHECK(K % 256 == 0, "fused rmsnorm needs the v2 (K%256) kernel");
+    TORCH_CHECK(norm_w->is_contiguous() && norm_w->numel() == K &&
+                norm_w->scalar_type() == at::kBFloat16,
+                "norm_w must be contiguous bf16 of length K");
+    nwp = bf_ptr(*norm_w);
+  }
   auto sizes = A.sizes().vec();
   sizes.back() = N;
   auto C = torch::empty(sizes, A.options());
@@ -665,7 +674,8 @@ torch::Tensor gemm_skinny(torch::Tensor A, torch::Tensor W,
     if (v2)
       gemm_skinny_v2_kernel<decltype(mt)::value><<<grid, 256, 0, cur_stream()>>>(
           bf_ptr(A), bf_ptr(W), ksplit == 1 ? rp : nullptr,
-          ksplit == 1 ? bp : nullptr, bf_ptr_mut(C), pp, M, N, K, kchunk, ksplit);
+          ksplit == 1 ? bp : nullptr, bf_ptr_mut(C), pp, M, N, K, kchunk, ksplit,
+          nwp, (float)eps);
     else
       gemm_skinny_kernel<decltype(mt)::value><<<grid, 256, 0, cur_stream()>>>(
           bf_ptr(A), bf_ptr(W), ksplit == 1 ? rp : nullptr,
@@ -769,7 +779,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_prefill_qkv", &attn_prefill_qkv);
   m.def("rope_kv_write_", &rope_kv_write_);
   m.def("attn_prefill", &attn_prefill);
-  m.def("gemm_skinny", &gemm_skinny);
+  m.def("gemm_skinny", &gemm_skinny, py::arg("A"), py::arg("W"),
+        py::arg("residual") = c10::nullopt, py::arg("bias") = c10::nullopt,
+        py::arg("ksplit") = 0, py::arg("norm_w") = c10::nullopt,
+        py::arg("eps") = 0.0);
   m.def("moe_gemm", &moe_gemm);
   m.def("gemm_w4", &gemm_w4);
   m.def("quant4_pack", &quant4_pack);

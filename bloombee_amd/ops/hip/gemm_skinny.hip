@@ -229,12 +229,46 @@ __global__ __launch_bounds__(256) void gemm_skinny_v2_kernel(
       }
   };
 
-  // Pipeline: stage s's A sits in LDS while its two B register sets stream;
-  // stage s+1's A loads issue right after the barrier that frees them.
-  load_a(k0);
-  issue0(k0);
-  store_a();
-  __syncthreads();
+  if (nw) {
+    // Per-row sum of squares over FULL K (every split needs the global rms,
+    // not its chunk's). 256 threads = 32 rows x 8 sub-readers; A is
+    // L2-resident after the first WG so this prepass costs L2 bandwidth,
+    // not HBM. The first W-register set is issued first so the weight
+    // stream is already in flight while the prepass reduces.
+    const int row = threadIdx.x >> 3, sub = threadIdx.x & 7;
+    issue0(k0);
+    float ss = 0.f;
+    if (row < M) {
+      const unsigned short* p = A + (long)row * K + sub * (K / 8);
+      for (int x = 0; x < K / 8; x += 8) {
+        short8 v = *reinterpret_cast<const short8*>(p + x);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const float f = bf2f((unsigned short)v[e]);
+          ss += f * f;
+        }
+      }
+    }
+    ssp[threadIdx.x] = ss;
+    __syncthreads();
+    if (sub == 0 && row < M) {
+      float t = 0.f;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) t += ssp[(row << 3) + e];
+      invr[row] = rsqrtf(t / K + eps);
+    }
+    __syncthreads();
+    load_a(k0);
+    store_a();
+    __syncthreads();
+  } else {
+    // Pipeline: stage s's A sits in LDS while its two B register sets
+    // stream; stage s+1's A loads issue right after the barrier frees them.
+    load_a(k0);
+    issue0(k0);
+    store_a();
+    __syncthreads();
+  }
   for (int k = k0; k < k1; k += KSTEP) {
     if (k + KSTEP < k1) load_a(k + KSTEP);
     issue1(k + 128);

@@ -274,11 +274,15 @@ def attn_paged(q, k_pages, v_pages, page_table, q_start, scale=None, window: int
 
 
 def linear(x: torch.Tensor, w: torch.Tensor, residual: Optional[torch.Tensor] = None,
-           bias: Optional[torch.Tensor] = None) -> torch.Tensor:
-    """y = x @ w^T (+bias) (+residual). Decode-shaped GEMMs (<=32 rows) on GPU
-    go through the skinny-M MFMA kernel (gemm_skinny.hip) — hipBLASLt leaves
-    ~2x weight-stream bandwidth on the table at M<=32; everything else uses
-    hipBLASLt via F.linear."""
+           bias: Optional[torch.Tensor] = None,
+           norm: Optional[tuple] = None) -> torch.Tensor:
+    """y = (rmsnorm(x) if norm else x) @ w^T (+bias) (+residual).
+    Decode-shaped GEMMs (<=32 rows) on GPU go through the skinny-M MFMA
+    kernel (gemm_skinny.hip) — hipBLASLt leaves ~2x weight-stream bandwidth
+    on the table at M<=32; everything else uses hipBLASLt via F.linear.
+    norm=(weight, eps) folds the input rmsnorm into the kernel's A-operand
+    stage (launch-count reduction; the separate rms_norm launch + its
+    read/write disappear from the decode step)."""
     M = x.numel() // x.shape[-1]
     N, K = w.shape[0], x.shape[-1]
     # All decode-shaped GEMMs route to the skinny kernel: although hipBLASLt
@@ -290,11 +294,29 @@ def linear(x: torch.Tensor, w: torch.Tensor, residual: Optional[torch.Tensor] = 
     if (_on_gpu(x) and M <= 32 and x.dtype == torch.bfloat16
             and K % 256 == 0 and N % 64 == 0):
         _require_ext()
+        if norm is not None:
+            return hip_ops.gemm_skinny(x.contiguous(), w, residual, bias, 0,
+                                       norm[0].contiguous(), float(norm[1]))
         return hip_ops.gemm_skinny(x.contiguous(), w, residual, bias, 0)
+    if norm is not None:
+        x = rms_norm(x, norm[0], norm[1])
     y = torch.nn.functional.linear(x, w, bias)
     if residual is not None:
         y = y + residual.view_as(y)
     return y
+
+
+_FUSE_NORM = os.environ.get("BBAMD_FUSE_NORM", "1") != "0"
+
+
+def fuse_norm_linear_ok(x: torch.Tensor, w: torch.Tensor) -> bool:
+    """True when linear(norm=...) will take the fused-kernel path (so the
+    caller can skip emitting a separate rms_norm). BBAMD_FUSE_NORM=0
+    reverts to the separate-launch path (A/B knob)."""
+    K = x.shape[-1]
+    return (_FUSE_NORM and _on_gpu(x) and x.numel() // K <= 32
+            and x.dtype == torch.bfloat16 and K % 256 == 0
+            and w.shape[0] % 64 == 0 and HAVE_HIP_OPS)
 
 
 def attn_paged_mixed(q, k_pages, v_pages, page_table, ctx_lens,

@@ -553,3 +553,52 @@ def test_llama_quantized_engine_decode_gpu():
     yg = gpu.forward_inference(x.to(DEV), hg, sp.to(DEV))
     assert torch.allclose(yg.cpu().float(), yc.float(), atol=3e-2), \
         (yg.cpu().float() - yc.float()).abs().max()
+
+
+@pytest.mark.gpu
+def test_gemm_skinny_fused_norm_parity():
+    """gemm_skinny norm_w path == rms_norm kernel + unfused gemm_skinny
+    (the fused path skips the intermediate bf16 write; tolerance covers
+    that rounding difference)."""
+    torch.manual_seed(11)
+    for M, N, K in [(1, 4096, 4096), (8, 6144, 4096), (32, 4096, 14336),
+                    (32, 28672, 4096), (17, 512, 512)]:
+        x = (torch.randn(M, K, dtype=torch.bfloat16, device=DEV) * 0.5)
+        w = torch.randn(N, K, dtype=torch.bfloat16, device=DEV) * 0.05
+        nw = torch.randn(K, dtype=torch.bfloat16, device=DEV)
+        r = torch.randn(M, N, dtype=torch.bfloat16, device=DEV)
+        want = ops.hip_ops.gemm_skinny(
+            ops.rms_norm(x, nw, 1e-5).contiguous(), w, r, None, 0)
+        got = ops.hip_ops.gemm_skinny(x, w, r, None, 0, nw, 1e-5)
+        err = (got.float() - want.float()).abs().max().item()
+        scale = want.float().abs().max().clamp_min(1.0).item()
+        assert err / scale < 2e-2, (M, N, K, err, scale)
+
+
+@pytest.mark.gpu
+def test_llama_block_fused_norm_matches_unfused():
+    """Whole-block decode with the fused-norm path vs BBAMD_FUSE_NORM=0
+    semantics (monkeypatched): same inputs, outputs within bf16 noise."""
+    from bloombee_amd.engine import BlockStack
+    from bloombee_amd.models.base import resolve_config
+    from bloombee_amd.ops import interface as iface
+
+    cfg = resolve_config("llama-mini-gpu")
+    stack = BlockStack(cfg, 0, 2, device=DEV, seed=9)
+    gen = torch.Generator().manual_seed(4)
+    x = (torch.randn(2, 1, cfg.hidden_size, generator=gen) * 0.1).to(cfg.dtype)
+    sp = torch.zeros(2, dtype=torch.int32)
+
+    outs = []
+    for fuse in (True, False):
+        kv = stack.make_kv(1 << 10)
+        h = kv.allocate(2, 32)
+        h.extend(1)
+        old = iface._FUSE_NORM
+        iface._FUSE_NORM = fuse
+        try:
+            outs.append(stack.forward_inference(x.to(DEV), h, sp.to(DEV)))
+        finally:
+            iface._FUSE_NORM = old
+    err = (outs[0].float() - outs[1].float()).abs().max().item()
+    assert err < 3e-2, err

@@ -297,3 +297,19 @@ def test_qwen3_block_quantized_decode_close_to_full():
     num = (yf.float() - yq.float()).norm()
     den = yf.float().norm().clamp_min(1e-6)
     assert (num / den) < 0.2, (num / den)
+
+
+def test_linear_norm_fallback_matches_separate():
+    """linear(norm=(w,eps)) on CPU == F.linear(rms_norm(x)) exactly (the
+    fused GPU kernel has its own parity test in test_gpu_kernels)."""
+    from bloombee_amd import ops
+    torch.manual_seed(3)
+    x = torch.randn(4, 256, dtype=torch.bfloat16)
+    w = torch.randn(64, 256, dtype=torch.bfloat16)
+    nw = torch.randn(256, dtype=torch.bfloat16)
+    want = torch.nn.functional.linear(ops.rms_norm(x, nw, 1e-5), w)
+    got = ops.linear(x, w, norm=(nw, 1e-5))
+    assert torch.equal(got, want)
+    r = torch.randn(4, 64, dtype=torch.bfloat16)
+    got2 = ops.linear(x, w, residual=r, norm=(nw, 1e-5))
+    assert torch.equal(got2, want + r)
