@@ -306,13 +306,17 @@ def linear(x: torch.Tensor, w: torch.Tensor, residual: Optional[torch.Tensor] = 
     return y
 
 
-_FUSE_NORM = os.environ.get("BBAMD_FUSE_NORM", "1") != "0"
+_FUSE_NORM = os.environ.get("BBAMD_FUSE_NORM", "0") == "1"
 
 
 def fuse_norm_linear_ok(x: torch.Tensor, w: torch.Tensor) -> bool:
     """True when linear(norm=...) will take the fused-kernel path (so the
-    caller can skip emitting a separate rms_norm). BBAMD_FUSE_NORM=0
-    reverts to the separate-launch path (A/B knob)."""
+    caller can skip emitting a separate rms_norm). DEFAULT OFF
+    (BBAMD_FUSE_NORM=1 enables): measured 4266 vs 5054 tok/s on the
+    flagship bench — the per-WG full-K variance prepass costs more than
+    the two saved launches once split-K multiplies the redundant A reads
+    (profiles/r02 §13). Kept as the numerics-verified base for the
+    epilogue-emitted-stats design (ROUND3.md item 2)."""
     K = x.shape[-1]
     return (_FUSE_NORM and _on_gpu(x) and x.numel() // K <= 32
             and x.dtype == torch.bfloat16 and K % 256 == 0
