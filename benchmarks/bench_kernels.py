@@ -132,3 +132,27 @@ if __name__ == "__main__":
     which = sys.argv[1:] or list(ALL)
     for name in which:
         ALL[name]()
+
+
+def bench_gemm_norm():
+    """Fused-rmsnorm GEMM A/B per llama decode shape: overhead of the
+    per-WG prepass+scale vs the separate rms_norm launch it replaces."""
+    for M, N, K, tag in [(32, 6144, 4096, "qkv"), (32, 4096, 4096, "o"),
+                         (32, 28672, 4096, "gate_up"), (32, 4096, 14336, "down")]:
+        x = torch.randn(M, K, dtype=torch.bfloat16, device=DEV)
+        w = torch.randn(N, K, dtype=torch.bfloat16, device=DEV)
+        nw = torch.randn(K, dtype=torch.bfloat16, device=DEV)
+        t0 = timeit(lambda: ops.hip_ops.gemm_skinny(x, w, None, None, 0))
+        t1 = timeit(lambda: ops.hip_ops.gemm_skinny(x, w, None, None, 0, nw, 1e-5))
+        tn = timeit(lambda: ops.hip_ops.rms_norm(x, nw, 1e-5))
+        print(f"gemm_norm {tag} {M}x{N}x{K}: plain {t0*1e6:6.1f} us  "
+              f"fused {t1*1e6:6.1f} us (+{(t1-t0)*1e6:5.1f})  "
+              f"rms_norm alone {tn*1e6:5.1f} us")
+    h = torch.randn(32, 4096, dtype=torch.bfloat16, device=DEV)
+    a = torch.randn(32, 4096, dtype=torch.bfloat16, device=DEV)
+    nw2 = torch.randn(4096, dtype=torch.bfloat16, device=DEV)
+    tr = timeit(lambda: ops.hip_ops.rms_norm_residual(a, h, nw2, 1e-5))
+    print(f"rms_norm_residual 32x4096: {tr*1e6:5.1f} us")
+
+
+ALL["gemm_norm"] = bench_gemm_norm
