@@ -116,23 +116,6 @@ def bench_gemm():
                       f"{N*K*2/ts/1e12:5.2f} TB/s(wt)")
 
 
-ALL = {
-    "attn_decode": bench_attn_decode,
-    "attn_prefill": bench_attn_prefill,
-    "kv_stream": bench_kv_stream,
-    "norms": bench_norms,
-    "swiglu": bench_swiglu,
-    "gemm": bench_gemm,
-}
-
-if __name__ == "__main__":
-    if not torch.cuda.is_available():
-        sys.exit("bench_kernels measures the gfx950 HIP kernels — "
-                 "run on a GPU box (e.g. via gpurun)")
-    which = sys.argv[1:] or list(ALL)
-    for name in which:
-        ALL[name]()
-
 
 def bench_gemm_norm():
     """Fused-rmsnorm GEMM A/B per llama decode shape: overhead of the
@@ -155,4 +138,21 @@ def bench_gemm_norm():
     print(f"rms_norm_residual 32x4096: {tr*1e6:5.1f} us")
 
 
-ALL["gemm_norm"] = bench_gemm_norm
+
+ALL = {
+    "attn_decode": bench_attn_decode,
+    "attn_prefill": bench_attn_prefill,
+    "kv_stream": bench_kv_stream,
+    "norms": bench_norms,
+    "swiglu": bench_swiglu,
+    "gemm": bench_gemm,
+    "gemm_norm": bench_gemm_norm,
+}
+
+if __name__ == "__main__":
+    if not torch.cuda.is_available():
+        sys.exit("bench_kernels measures the gfx950 HIP kernels — "
+                 "run on a GPU box (e.g. via gpurun)")
+    which = sys.argv[1:] or list(ALL)
+    for name in which:
+        ALL[name]()
