@@ -44,6 +44,14 @@ class BlockStack(torch.nn.Module):
             blk.init_random(seed=seed * 10_000 + i)
             blk.layer_index = i - start  # KV page index local to this stack's pool
             self.blocks.append(blk)
+        # chain blocks so each one's down-proj epilogue can leave row
+        # sum-of-squares for the next one's fused input rmsnorm
+        # (llama/block.py fuse_norm path); first block has no producer
+        # and keeps the separate-norm launch
+        for j in range(1, len(self.blocks)):
+            # bypass Module.__setattr__: a plain reference, not a submodule
+            object.__setattr__(self.blocks[j], "prev_block",
+                               self.blocks[j - 1])
         self.device = torch.device(device)
 
     def make_kv(self, max_tokens: int) -> PagedKVCache:

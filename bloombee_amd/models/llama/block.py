@@ -75,6 +75,12 @@ class LlamaBlock(torch.nn.Module):
         self.gate_up_w = p(2 * I, H)              # fused gate|up
         self.down_w = p(H, I)
         self.rope = rope if rope is not None else RopeTables(config)
+        # fused-norm chain state (forward_inference fuse_norm path):
+        # per-stripe row sum-of-squares buffers, (H/64, 32) f32, written by
+        # the o/down GEMM epilogues and consumed by the next fused rmsnorm
+        self._ss_h2 = None
+        self._ss_hidden = None
+        self._ss_valid = False
 
     @torch.no_grad()
     def init_random(self, seed: Optional[int] = None):
@@ -151,14 +157,26 @@ class LlamaBlock(torch.nn.Module):
 
         # Decode-shaped GPU steps fold both per-layer rmsnorms into the
         # skinny GEMM's A-operand stage (ops.linear norm=...): two fewer
-        # launches + elementwise passes per layer. W4/LoRA blocks and
-        # prefill keep the separate-norm path.
+        # launches + elementwise passes per layer. The row sum-of-squares
+        # the norm needs is produced by the PRECEDING GEMM's epilogue
+        # (ss buffers chained block-to-block via prev_block; re-streaming
+        # A instead costs +18-60 us/GEMM — profiles/r02 §13). W4/LoRA
+        # blocks and prefill keep the separate-norm path.
         fuse_norm = (getattr(self, "_w4", None) is None
                      and getattr(self, "lora_delta", None) is None
-                     and ops.fuse_norm_linear_ok(hidden, self.qkv_w))
-        if fuse_norm:
+                     and ops.fuse_norm_linear_ok(hidden, self.qkv_w)
+                     # all four GEMMs must hit the v2 kernel (K%256) with
+                     # stripe-aligned N so the ss chain stays on-device
+                     and self.I % 256 == 0 and (2 * self.I) % 64 == 0
+                     and (self.Hq * self.D) % 256 == 0
+                     and self.config.hidden_size % 64 == 0)
+        self._ss_valid = False
+        pb = getattr(self, "prev_block", None)
+        if fuse_norm and pb is not None and getattr(pb, "_ss_valid", False):
+            # previous block's down-proj left sum-of-squares for `hidden`
             qkv = ops.linear(hidden, self.qkv_w,
-                             norm=(self.input_norm_w, cfg.rms_norm_eps))
+                             norm=(self.input_norm_w, cfg.rms_norm_eps),
+                             ss_in=pb._ss_hidden)
         else:
             x = ops.rms_norm(hidden, self.input_norm_w, cfg.rms_norm_eps)
             qkv = self._lin(x, self.qkv_w, "qkv_w")        # (B, T, (Hq+2Hkv)D)
@@ -202,12 +220,24 @@ class LlamaBlock(torch.nn.Module):
             attn = ops.attn_paged_qkv(qkv, Hq, Hkv, kp, vp, pt, start_pos,
                                       self.scale)
         if fuse_norm:
-            # h2 = hidden + o(attn) via the GEMM's residual epilogue; the
-            # post-attention rmsnorm folds into the gate_up A-stage
-            h2 = ops.linear(attn, self.o_w, residual=hidden)
+            # h2 = hidden + o(attn) via the GEMM's residual epilogue, which
+            # also emits row sum-of-squares; the post-attention rmsnorm
+            # folds into the gate_up A-stage consuming them. down-proj
+            # leaves the stats for the NEXT block's input norm.
+            if self._ss_h2 is None or self._ss_h2.device != hidden.device:
+                st = self.config.hidden_size // 64
+                self._ss_h2 = torch.empty(st * 32, dtype=torch.float32,
+                                          device=hidden.device)
+                self._ss_hidden = torch.empty_like(self._ss_h2)
+            h2 = ops.linear(attn, self.o_w, residual=hidden,
+                            ss_out=self._ss_h2)
             gu = ops.linear(h2, self.gate_up_w,
-                            norm=(self.post_norm_w, cfg.rms_norm_eps))
-            return ops.linear(ops.swiglu(gu), self.down_w, residual=h2)
+                            norm=(self.post_norm_w, cfg.rms_norm_eps),
+                            ss_in=self._ss_h2)
+            out = ops.linear(ops.swiglu(gu), self.down_w, residual=h2,
+                             ss_out=self._ss_hidden)
+            self._ss_valid = True
+            return out
         a = self._lin(attn, self.o_w, "o_w")
 
         # h2 = hidden + a fused into the post-attention norm

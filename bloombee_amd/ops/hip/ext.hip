@@ -605,7 +605,9 @@ torch::Tensor gemm_w4(torch::Tensor A, torch::Tensor Wq, torch::Tensor scale,
 torch::Tensor gemm_skinny(torch::Tensor A, torch::Tensor W,
                           c10::optional<torch::Tensor> residual,
                           c10::optional<torch::Tensor> bias, long ksplit_req,
-                          c10::optional<torch::Tensor> norm_w, double eps) {
+                          c10::optional<torch::Tensor> norm_w, double eps,
+                          c10::optional<torch::Tensor> ss_in,
+                          c10::optional<torch::Tensor> ss_out) {
   CHECK_DEV(A); CHECK_BF16(A); CHECK_CONTIG(A);
   CHECK_DEV(W); CHECK_BF16(W); CHECK_CONTIG(W);
   const int K = A.size(-1);
@@ -621,6 +623,24 @@ torch::Tensor gemm_skinny(torch::Tensor A, torch::Tensor W,
                 norm_w->scalar_type() == at::kBFloat16,
                 "norm_w must be contiguous bf16 of length K");
     nwp = bf_ptr(*norm_w);
+  }
+  // ss_in: (nstripes, 32) f32 row sum-of-squares left by the producing
+  // GEMM's epilogue — lets the fused rmsnorm skip re-streaming A.
+  const float* ssinp = nullptr;
+  int nstripes = 0;
+  if (ss_in.has_value()) {
+    TORCH_CHECK(nwp, "ss_in only meaningful with norm_w");
+    TORCH_CHECK(ss_in->is_contiguous() && ss_in->scalar_type() == at::kFloat &&
+                ss_in->numel() % 32 == 0, "ss_in must be f32 (nstripes,32)");
+    ssinp = ss_in->data_ptr<float>();
+    nstripes = (int)(ss_in->numel() / 32);
+  }
+  float* ssoutp = nullptr;
+  if (ss_out.has_value()) {
+    TORCH_CHECK(ss_out->is_contiguous() && ss_out->scalar_type() == at::kFloat &&
+                ss_out->numel() == (long)(N / 64) * 32,
+                "ss_out must be f32 of numel (N/64)*32");
+    ssoutp = ss_out->data_ptr<float>();
   }
   auto sizes = A.sizes().vec();
   sizes.back() = N;
@@ -675,19 +695,28 @@ torch::Tensor gemm_skinny(torch::Tensor A, torch::Tensor W,
       gemm_skinny_v2_kernel<decltype(mt)::value><<<grid, 256, 0, cur_stream()>>>(
           bf_ptr(A), bf_ptr(W), ksplit == 1 ? rp : nullptr,
           ksplit == 1 ? bp : nullptr, bf_ptr_mut(C), pp, M, N, K, kchunk, ksplit,
-          nwp, (float)eps);
-    else
+          nwp, (float)eps, ssinp, nstripes,
+          ksplit == 1 ? ssoutp : nullptr);
+    else {
+      TORCH_CHECK(!nwp && (!ssoutp || ksplit > 1),
+                  "fused norm / ss_out need the v2 (K%256) kernel");
       gemm_skinny_kernel<decltype(mt)::value><<<grid, 256, 0, cur_stream()>>>(
           bf_ptr(A), bf_ptr(W), ksplit == 1 ? rp : nullptr,
           ksplit == 1 ? bp : nullptr, bf_ptr_mut(C), pp, M, N, K, ksplit);
+    }
   };
   if (M <= 16) launch(std::integral_constant<int, 1>{});
   else launch(std::integral_constant<int, 2>{});
   if (ksplit > 1) {
-    const long total = (long)M * N;
-    const int cgrid = (int)std::min<long>((total + 255) / 256, 2048);
-    gemm_skinny_combine_kernel<<<cgrid, 256, 0, cur_stream()>>>(
-        pp, rp, bp, bf_ptr_mut(C), M, N, ksplit);
+    if (ssoutp) {
+      gemm_skinny_combine_ss_kernel<<<N / 64, 256, 0, cur_stream()>>>(
+          pp, rp, bp, bf_ptr_mut(C), ssoutp, M, N, ksplit);
+    } else {
+      const long total = (long)M * N;
+      const int cgrid = (int)std::min<long>((total + 255) / 256, 2048);
+      gemm_skinny_combine_kernel<<<cgrid, 256, 0, cur_stream()>>>(
+          pp, rp, bp, bf_ptr_mut(C), M, N, ksplit);
+    }
   }
   return C;
 }
@@ -782,7 +811,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_skinny", &gemm_skinny, py::arg("A"), py::arg("W"),
         py::arg("residual") = c10::nullopt, py::arg("bias") = c10::nullopt,
         py::arg("ksplit") = 0, py::arg("norm_w") = c10::nullopt,
-        py::arg("eps") = 0.0);
+        py::arg("eps") = 0.0, py::arg("ss_in") = c10::nullopt,
+        py::arg("ss_out") = c10::nullopt);
   m.def("moe_gemm", &moe_gemm);
   m.def("gemm_w4", &gemm_w4);
   m.def("quant4_pack", &quant4_pack);

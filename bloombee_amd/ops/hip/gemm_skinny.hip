@@ -126,7 +126,10 @@ __global__ __launch_bounds__(256) void gemm_skinny_v2_kernel(
     float* __restrict__ Cpart,              // (ksplit, M, N) f32 (ksplit > 1)
     int M, int N, int K, int kchunk, int ksplit,
     const unsigned short* __restrict__ nw,  // (K,) rmsnorm weight or null
-    float eps) {
+    float eps,
+    const float* __restrict__ ssin,   // (nstripes, 32) producer row sum-sq
+    int nstripes,
+    float* __restrict__ ssout) {      // (N/64, 32) this GEMM's row sum-sq
   constexpr int U = 4;                     // k-slices per pipeline set (128 k)
   constexpr int KSTEP = 256;               // A elements staged per stage
   constexpr int RSTRIDE = KSTEP + 8;       // padded LDS row stride (elements)
@@ -231,14 +234,20 @@ __global__ __launch_bounds__(256) void gemm_skinny_v2_kernel(
 
   if (nw) {
     // Per-row sum of squares over FULL K (every split needs the global rms,
-    // not its chunk's). 256 threads = 32 rows x 8 sub-readers; A is
-    // L2-resident after the first WG so this prepass costs L2 bandwidth,
-    // not HBM. The first W-register set is issued first so the weight
-    // stream is already in flight while the prepass reduces.
+    // not its chunk's). Cheap path: the PRODUCING GEMM's epilogue left
+    // per-stripe row sums (ssin) — summing nstripes*32 f32 from L2 costs
+    // ~nothing (profiles/r02 §13: the A-restream fallback below costs
+    // +18..60 us/GEMM under split-K and is only for callers with no
+    // producer stats). 256 threads = 32 rows x 8 sub-readers; the first
+    // W-register set is issued first so the weight stream is already in
+    // flight while the prepass reduces.
     const int row = threadIdx.x >> 3, sub = threadIdx.x & 7;
     issue0(k0);
     float ss = 0.f;
-    if (row < M) {
+    if (ssin) {
+      if (row < M)
+        for (int s = sub; s < nstripes; s += 8) ss += ssin[s * 32 + row];
+    } else if (row < M) {
       const unsigned short* p = A + (long)row * K + sub * (K / 8);
       for (int x = 0; x < K / 8; x += 8) {
         short8 v = *reinterpret_cast<const short8*>(p + x);
@@ -283,17 +292,46 @@ __global__ __launch_bounds__(256) void gemm_skinny_v2_kernel(
   }
 
   if (ksplit == 1) {
+    if (ssout == nullptr) {
 #pragma unroll
-    for (int t = 0; t < MT; ++t)
+      for (int t = 0; t < MT; ++t)
 #pragma unroll
-      for (int reg = 0; reg < 4; ++reg) {
-        const int m = t * 16 + hi * 4 + reg;
-        if (m >= M) continue;
-        float v = acc[t][reg];
-        if (bias) v += bf2f(bias[n0 + li]);
-        if (R) v += bf2f(R[(long)m * N + n0 + li]);
-        C[(long)m * N + n0 + li] = f2bf(v);
-      }
+        for (int reg = 0; reg < 4; ++reg) {
+          const int m = t * 16 + hi * 4 + reg;
+          if (m >= M) continue;
+          float v = acc[t][reg];
+          if (bias) v += bf2f(bias[n0 + li]);
+          if (R) v += bf2f(R[(long)m * N + n0 + li]);
+          C[(long)m * N + n0 + li] = f2bf(v);
+        }
+    } else {
+      // Emit per-stripe row sum-of-squares alongside the store: li-group
+      // shfl reduce (16 cols per wave), cross-wave fold via ssp. The next
+      // GEMM's fused rmsnorm sums these instead of re-streaming A.
+      __syncthreads();  // atile is dead; reuse ssp[wave*32 + m]
+#pragma unroll
+      for (int t = 0; t < MT; ++t)
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg) {
+          const int m = t * 16 + hi * 4 + reg;
+          float vs = 0.f;
+          if (m < M) {
+            float v = acc[t][reg];
+            if (bias) v += bf2f(bias[n0 + li]);
+            if (R) v += bf2f(R[(long)m * N + n0 + li]);
+            C[(long)m * N + n0 + li] = f2bf(v);
+            vs = v * v;
+          }
+#pragma unroll
+          for (int msk = 1; msk < 16; msk <<= 1) vs += __shfl_xor(vs, msk);
+          if (li == 0) ssp[wave * 32 + m] = vs;
+        }
+      __syncthreads();
+      if (threadIdx.x < 32)
+        ssout[(long)tile * 32 + threadIdx.x] =
+            ssp[threadIdx.x] + ssp[32 + threadIdx.x] +
+            ssp[64 + threadIdx.x] + ssp[96 + threadIdx.x];
+    }
   } else {
     float* dst = Cpart + (long)split * M * N;
 #pragma unroll
@@ -305,6 +343,34 @@ __global__ __launch_bounds__(256) void gemm_skinny_v2_kernel(
         dst[(long)m * N + n0 + li] = acc[t][reg];
       }
   }
+}
+
+// Split-K combine that also emits per-stripe row sum-of-squares (ssout
+// layout (N/64, 32)): grid = N/64, 4 waves = 4 row-groups x 64 cols.
+// Wave w owns rows m ≡ w (mod 4); lane = col within the stripe.
+__global__ __launch_bounds__(256) void gemm_skinny_combine_ss_kernel(
+    const float* __restrict__ Cpart, const unsigned short* __restrict__ R,
+    const unsigned short* __restrict__ bias, unsigned short* __restrict__ C,
+    float* __restrict__ ssout, int M, int N, int ksplit) {
+  __shared__ float ssl[32];
+  const int n0 = blockIdx.x * 64;
+  const int col = threadIdx.x & 63;
+  const int w = threadIdx.x >> 6;
+  const long total = (long)M * N;
+  const float b = bias ? bf2f(bias[n0 + col]) : 0.f;
+  for (int m = w; m < M; m += 4) {
+    const long i = (long)m * N + n0 + col;
+    float v = b;
+    for (int s = 0; s < ksplit; ++s) v += Cpart[s * total + i];
+    if (R) v += bf2f(R[i]);
+    C[i] = f2bf(v);
+    const float vs = wave_reduce_sum(v * v);
+    if (col == 0) ssl[m] = vs;
+  }
+  __syncthreads();
+  if (threadIdx.x < 32)
+    ssout[(long)blockIdx.x * 32 + threadIdx.x] =
+        threadIdx.x < M ? ssl[threadIdx.x] : 0.f;
 }
 
 __global__ void gemm_skinny_combine_kernel(

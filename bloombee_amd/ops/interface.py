@@ -275,14 +275,19 @@ def attn_paged(q, k_pages, v_pages, page_table, q_start, scale=None, window: int
 
 def linear(x: torch.Tensor, w: torch.Tensor, residual: Optional[torch.Tensor] = None,
            bias: Optional[torch.Tensor] = None,
-           norm: Optional[tuple] = None) -> torch.Tensor:
+           norm: Optional[tuple] = None,
+           ss_in: Optional[torch.Tensor] = None,
+           ss_out: Optional[torch.Tensor] = None) -> torch.Tensor:
     """y = (rmsnorm(x) if norm else x) @ w^T (+bias) (+residual).
     Decode-shaped GEMMs (<=32 rows) on GPU go through the skinny-M MFMA
     kernel (gemm_skinny.hip) — hipBLASLt leaves ~2x weight-stream bandwidth
     on the table at M<=32; everything else uses hipBLASLt via F.linear.
     norm=(weight, eps) folds the input rmsnorm into the kernel's A-operand
     stage (launch-count reduction; the separate rms_norm launch + its
-    read/write disappear from the decode step)."""
+    read/write disappear from the decode step). ss_in is the producing
+    GEMM's per-stripe row sum-of-squares (its ss_out) so the fused norm
+    skips re-streaming A; ss_out asks this GEMM's epilogue to emit the
+    same stats for ITS consumer (f32 (N/64, 32), overwritten — graph-safe)."""
     M = x.numel() // x.shape[-1]
     N, K = w.shape[0], x.shape[-1]
     # All decode-shaped GEMMs route to the skinny kernel: although hipBLASLt
@@ -294,15 +299,20 @@ def linear(x: torch.Tensor, w: torch.Tensor, residual: Optional[torch.Tensor] = 
     if (_on_gpu(x) and M <= 32 and x.dtype == torch.bfloat16
             and K % 256 == 0 and N % 64 == 0):
         _require_ext()
-        if norm is not None:
+        if norm is not None or ss_out is not None:
+            nw = norm[0].contiguous() if norm is not None else None
+            eps = float(norm[1]) if norm is not None else 0.0
             return hip_ops.gemm_skinny(x.contiguous(), w, residual, bias, 0,
-                                       norm[0].contiguous(), float(norm[1]))
+                                       nw, eps, ss_in, ss_out)
         return hip_ops.gemm_skinny(x.contiguous(), w, residual, bias, 0)
     if norm is not None:
         x = rms_norm(x, norm[0], norm[1])
     y = torch.nn.functional.linear(x, w, bias)
     if residual is not None:
         y = y + residual.view_as(y)
+    if ss_out is not None:
+        raise RuntimeError("ss_out requires the GPU skinny-GEMM path "
+                           "(callers gate on fuse_norm_linear_ok)")
     return y
 
 

@@ -602,3 +602,34 @@ def test_llama_block_fused_norm_matches_unfused():
             iface._FUSE_NORM = old
     err = (outs[0].float() - outs[1].float()).abs().max().item()
     assert err < 3e-2, err
+
+
+@pytest.mark.gpu
+def test_gemm_skinny_ss_chain():
+    """Epilogue-emitted row sum-of-squares: ss_out matches the row sums of
+    the produced C (both the ksplit==1 in-kernel path and the split-K
+    combine_ss path), and a consumer GEMM using ss_in matches one that
+    re-streams A for its variance."""
+    torch.manual_seed(13)
+    for M, N, K in [(8, 512, 512),        # ksplit==1 epilogue path
+                    (32, 4096, 4096),     # split-K combine_ss path
+                    (32, 4096, 14336)]:
+        x = torch.randn(M, K, dtype=torch.bfloat16, device=DEV) * 0.3
+        w = torch.randn(N, K, dtype=torch.bfloat16, device=DEV) * 0.05
+        r = torch.randn(M, N, dtype=torch.bfloat16, device=DEV)
+        ss = torch.full((N // 64 * 32,), float("nan"), device=DEV)
+        c = ops.hip_ops.gemm_skinny(x, w, r, None, 0, None, 0.0, None, ss)
+        want_ss = c.float().pow(2).sum(-1)
+        got_ss = ss.view(-1, 32)[:, :M].sum(0)
+        rel = ((got_ss - want_ss).abs() / want_ss.clamp_min(1e-3)).max().item()
+        assert rel < 2e-2, (M, N, K, rel)
+
+        # consumer: fused norm via ss_in == fused norm via A re-stream
+        w2 = torch.randn(256, N, dtype=torch.bfloat16, device=DEV) * 0.05
+        nw = torch.randn(N, dtype=torch.bfloat16, device=DEV)
+        y_self = ops.hip_ops.gemm_skinny(c, w2, None, None, 0, nw, 1e-5)
+        y_ssin = ops.hip_ops.gemm_skinny(c, w2, None, None, 0, nw, 1e-5,
+                                         ss.contiguous(), None)
+        err = (y_self.float() - y_ssin.float()).abs().max().item()
+        sc = y_self.float().abs().max().clamp_min(1.0).item()
+        assert err / sc < 1e-2, (M, N, K, err)
