@@ -101,6 +101,45 @@ class LlamaBlock(torch.nn.Module):
         return self
 
     @torch.no_grad()
+    def fold_norm_weights(self):
+        """Reparameterize for the fused-norm decode path: qkv_w <- qkv_w *
+        input_norm_w, gate_up_w <- gate_up_w * post_norm_w (f32 product,
+        one bf16 round), norm weights <- 1. rms_norm with unit weight is
+        mathematically unchanged on EVERY path (train, prefill, CPU), and
+        the decode GEMM then only applies the per-row 1/rms in its
+        epilogue (norm_mode 2) instead of per-element weight scaling
+        (profiles/r02 §13: the per-stage scale cost +4-8 us/GEMM)."""
+        if getattr(self, "_norm_folded", False):
+            return
+        self._norm_fold_orig = (self.input_norm_w.detach().clone(),
+                                self.post_norm_w.detach().clone())
+        self.qkv_w.data = (self.qkv_w.float()
+                           * self.input_norm_w.float()).to(self.qkv_w.dtype)
+        self.gate_up_w.data = (self.gate_up_w.float()
+                               * self.post_norm_w.float()).to(self.gate_up_w.dtype)
+        self.input_norm_w.data.fill_(1.0)
+        self.post_norm_w.data.fill_(1.0)
+        self._norm_folded = True
+
+    @torch.no_grad()
+    def unfold_norm_weights(self):
+        """Best-effort inverse of fold_norm_weights (division re-rounds:
+        ~1 ulp drift on the projections). Needed before attaching LoRA
+        adapters trained against the original parameterization."""
+        if not getattr(self, "_norm_folded", False):
+            return
+        inw, pnw = self._norm_fold_orig
+        self.input_norm_w.data.copy_(inw)
+        self.post_norm_w.data.copy_(pnw)
+        den_i = inw.float().abs().clamp_min(1e-6) * inw.float().sign().where(
+            inw.float() != 0, torch.ones_like(inw.float()))
+        den_p = pnw.float().abs().clamp_min(1e-6) * pnw.float().sign().where(
+            pnw.float() != 0, torch.ones_like(pnw.float()))
+        self.qkv_w.data = (self.qkv_w.float() / den_i).to(self.qkv_w.dtype)
+        self.gate_up_w.data = (self.gate_up_w.float() / den_p).to(self.gate_up_w.dtype)
+        self._norm_folded = False
+
+    @torch.no_grad()
     def quantize_weights_q4(self, keep_full: bool = False):
         """Pack the four projection weights into 4-bit group codes
         (quant4_pack layout); decode GEMMs then stream ~3.5x fewer bytes
@@ -171,11 +210,14 @@ class LlamaBlock(torch.nn.Module):
                      and (self.Hq * self.D) % 256 == 0
                      and self.config.hidden_size % 64 == 0)
         self._ss_valid = False
+        if fuse_norm and not getattr(self, "_norm_folded", False):
+            self.fold_norm_weights()   # one-time reparameterization
         pb = getattr(self, "prev_block", None)
         if fuse_norm and pb is not None and getattr(pb, "_ss_valid", False):
-            # previous block's down-proj left sum-of-squares for `hidden`
+            # previous block's down-proj left sum-of-squares for `hidden`;
+            # norm weight is folded into qkv_w -> invr-only norm (mode 2)
             qkv = ops.linear(hidden, self.qkv_w,
-                             norm=(self.input_norm_w, cfg.rms_norm_eps),
+                             norm=(None, cfg.rms_norm_eps),
                              ss_in=pb._ss_hidden)
         else:
             x = ops.rms_norm(hidden, self.input_norm_w, cfg.rms_norm_eps)
@@ -232,7 +274,7 @@ class LlamaBlock(torch.nn.Module):
             h2 = ops.linear(attn, self.o_w, residual=hidden,
                             ss_out=self._ss_h2)
             gu = ops.linear(h2, self.gate_up_w,
-                            norm=(self.post_norm_w, cfg.rms_norm_eps),
+                            norm=(None, cfg.rms_norm_eps),
                             ss_in=self._ss_h2)
             out = ops.linear(ops.swiglu(gu), self.down_w, residual=h2,
                              ss_out=self._ss_hidden)

@@ -607,7 +607,8 @@ torch::Tensor gemm_skinny(torch::Tensor A, torch::Tensor W,
                           c10::optional<torch::Tensor> bias, long ksplit_req,
                           c10::optional<torch::Tensor> norm_w, double eps,
                           c10::optional<torch::Tensor> ss_in,
-                          c10::optional<torch::Tensor> ss_out) {
+                          c10::optional<torch::Tensor> ss_out,
+                          long norm_mode_req) {
   CHECK_DEV(A); CHECK_BF16(A); CHECK_CONTIG(A);
   CHECK_DEV(W); CHECK_BF16(W); CHECK_CONTIG(W);
   const int K = A.size(-1);
@@ -616,14 +617,20 @@ torch::Tensor gemm_skinny(torch::Tensor A, torch::Tensor W,
   TORCH_CHECK(M <= 32, "gemm_skinny is for M <= 32, got ", M);
   TORCH_CHECK(W.size(1) == K, "A/W K mismatch");
   TORCH_CHECK(K % 32 == 0 && N % 64 == 0, "K%32, N%64 required");
+  // norm_mode: 0 none; 1 scale A by invr*norm_w per stage; 2 norm weight
+  // pre-folded into W — only the invr accumulator scale remains (cheap).
+  int norm_mode = (int)norm_mode_req;
   const unsigned short* nwp = nullptr;
   if (norm_w.has_value()) {
-    TORCH_CHECK(K % 256 == 0, "fused rmsnorm needs the v2 (K%256) kernel");
     TORCH_CHECK(norm_w->is_contiguous() && norm_w->numel() == K &&
                 norm_w->scalar_type() == at::kBFloat16,
                 "norm_w must be contiguous bf16 of length K");
     nwp = bf_ptr(*norm_w);
+    if (norm_mode == 0) norm_mode = 1;
   }
+  TORCH_CHECK(norm_mode != 1 || nwp, "norm_mode 1 needs norm_w");
+  if (norm_mode)
+    TORCH_CHECK(K % 256 == 0, "fused rmsnorm needs the v2 (K%256) kernel");
   // ss_in: (nstripes, 32) f32 row sum-of-squares left by the producing
   // GEMM's epilogue — lets the fused rmsnorm skip re-streaming A.
   const float* ssinp = nullptr;
@@ -695,10 +702,10 @@ torch::Tensor gemm_skinny(torch::Tensor A, torch::Tensor W,
       gemm_skinny_v2_kernel<decltype(mt)::value><<<grid, 256, 0, cur_stream()>>>(
           bf_ptr(A), bf_ptr(W), ksplit == 1 ? rp : nullptr,
           ksplit == 1 ? bp : nullptr, bf_ptr_mut(C), pp, M, N, K, kchunk, ksplit,
-          nwp, (float)eps, ssinp, nstripes,
+          nwp, (float)eps, norm_mode, ssinp, nstripes,
           ksplit == 1 ? ssoutp : nullptr);
     else {
-      TORCH_CHECK(!nwp && (!ssoutp || ksplit > 1),
+      TORCH_CHECK(!norm_mode && (!ssoutp || ksplit > 1),
                   "fused norm / ss_out need the v2 (K%256) kernel");
       gemm_skinny_kernel<decltype(mt)::value><<<grid, 256, 0, cur_stream()>>>(
           bf_ptr(A), bf_ptr(W), ksplit == 1 ? rp : nullptr,
@@ -709,7 +716,8 @@ torch::Tensor gemm_skinny(torch::Tensor A, torch::Tensor W,
   else launch(std::integral_constant<int, 2>{});
   if (ksplit > 1) {
     if (ssoutp) {
-      gemm_skinny_combine_ss_kernel<<<N / 64, 256, 0, cur_stream()>>>(
+      dim3 cg(N / 64, (M + 3) / 4);
+      gemm_skinny_combine_ss_kernel<<<cg, 256, 0, cur_stream()>>>(
           pp, rp, bp, bf_ptr_mut(C), ssoutp, M, N, ksplit);
     } else {
       const long total = (long)M * N;
@@ -812,7 +820,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("residual") = c10::nullopt, py::arg("bias") = c10::nullopt,
         py::arg("ksplit") = 0, py::arg("norm_w") = c10::nullopt,
         py::arg("eps") = 0.0, py::arg("ss_in") = c10::nullopt,
-        py::arg("ss_out") = c10::nullopt);
+        py::arg("ss_out") = c10::nullopt, py::arg("norm_mode") = 0);
   m.def("moe_gemm", &moe_gemm);
   m.def("gemm_w4", &gemm_w4);
   m.def("quant4_pack", &quant4_pack);

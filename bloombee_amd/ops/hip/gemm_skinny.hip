@@ -125,8 +125,11 @@ __global__ __launch_bounds__(256) void gemm_skinny_v2_kernel(
     unsigned short* __restrict__ C,         // (M, N) bf16   (ksplit == 1)
     float* __restrict__ Cpart,              // (ksplit, M, N) f32 (ksplit > 1)
     int M, int N, int K, int kchunk, int ksplit,
-    const unsigned short* __restrict__ nw,  // (K,) rmsnorm weight or null
-    float eps,
+    const unsigned short* __restrict__ nw,  // (K,) rmsnorm weight (mode 1)
+    float eps, int norm_mode,  // 0: none; 1: scale A by invr*nw (stage path);
+                               // 2: weight pre-folded into W — scale the
+                               //    ACCUMULATOR by invr in the epilogue
+                               //    (zero per-stage cost; the production path)
     const float* __restrict__ ssin,   // (nstripes, 32) producer row sum-sq
     int nstripes,
     float* __restrict__ ssout) {      // (N/64, 32) this GEMM's row sum-sq
@@ -183,7 +186,7 @@ __global__ __launch_bounds__(256) void gemm_skinny_v2_kernel(
       const int row = min(flat / 32, M - 1);
       const int off = (flat % 32) * 8;
       aregs[j] = *reinterpret_cast<const short8*>(A + (long)row * K + ks + off);
-      if (nw)
+      if (norm_mode == 1)
         nregs[j] = *reinterpret_cast<const short8*>(nw + ks + off);
     }
   };
@@ -192,7 +195,7 @@ __global__ __launch_bounds__(256) void gemm_skinny_v2_kernel(
     for (int j = 0; j < 4; ++j) {
       const int flat = threadIdx.x + j * 256;
       short8 v = aregs[j];
-      if (nw) {
+      if (norm_mode == 1) {
         const float ir = invr[flat / 32 >= M ? M - 1 : flat / 32];
 #pragma unroll
         for (int e = 0; e < 8; ++e)
@@ -232,7 +235,7 @@ __global__ __launch_bounds__(256) void gemm_skinny_v2_kernel(
       }
   };
 
-  if (nw) {
+  if (norm_mode) {
     // Per-row sum of squares over FULL K (every split needs the global rms,
     // not its chunk's). Cheap path: the PRODUCING GEMM's epilogue left
     // per-stripe row sums (ssin) — summing nstripes*32 f32 from L2 costs
@@ -300,6 +303,7 @@ __global__ __launch_bounds__(256) void gemm_skinny_v2_kernel(
           const int m = t * 16 + hi * 4 + reg;
           if (m >= M) continue;
           float v = acc[t][reg];
+          if (norm_mode == 2) v *= invr[m];
           if (bias) v += bf2f(bias[n0 + li]);
           if (R) v += bf2f(R[(long)m * N + n0 + li]);
           C[(long)m * N + n0 + li] = f2bf(v);
@@ -317,6 +321,7 @@ __global__ __launch_bounds__(256) void gemm_skinny_v2_kernel(
           float vs = 0.f;
           if (m < M) {
             float v = acc[t][reg];
+            if (norm_mode == 2) v *= invr[m];
             if (bias) v += bf2f(bias[n0 + li]);
             if (R) v += bf2f(R[(long)m * N + n0 + li]);
             C[(long)m * N + n0 + li] = f2bf(v);
@@ -340,37 +345,34 @@ __global__ __launch_bounds__(256) void gemm_skinny_v2_kernel(
       for (int reg = 0; reg < 4; ++reg) {
         const int m = t * 16 + hi * 4 + reg;
         if (m >= M) continue;
-        dst[(long)m * N + n0 + li] = acc[t][reg];
+        float v = acc[t][reg];
+        if (norm_mode == 2) v *= invr[m];
+        dst[(long)m * N + n0 + li] = v;
       }
   }
 }
 
 // Split-K combine that also emits per-stripe row sum-of-squares (ssout
-// layout (N/64, 32)): grid = N/64, 4 waves = 4 row-groups x 64 cols.
-// Wave w owns rows m ≡ w (mod 4); lane = col within the stripe.
+// layout (N/64, 32)): grid (N/64, ceil(M/4)), 256 threads = 4 rows x 64
+// cols, wave = row — each (stripe, m) handled by exactly one wave, the
+// lane dimension is the 64 columns, so the row sum is one wave_reduce
+// and no LDS/atomics. Same workgroup count as the flat combine.
 __global__ __launch_bounds__(256) void gemm_skinny_combine_ss_kernel(
     const float* __restrict__ Cpart, const unsigned short* __restrict__ R,
     const unsigned short* __restrict__ bias, unsigned short* __restrict__ C,
     float* __restrict__ ssout, int M, int N, int ksplit) {
-  __shared__ float ssl[32];
   const int n0 = blockIdx.x * 64;
   const int col = threadIdx.x & 63;
-  const int w = threadIdx.x >> 6;
+  const int m = blockIdx.y * 4 + (threadIdx.x >> 6);
+  if (m >= M) return;
   const long total = (long)M * N;
-  const float b = bias ? bf2f(bias[n0 + col]) : 0.f;
-  for (int m = w; m < M; m += 4) {
-    const long i = (long)m * N + n0 + col;
-    float v = b;
-    for (int s = 0; s < ksplit; ++s) v += Cpart[s * total + i];
-    if (R) v += bf2f(R[i]);
-    C[i] = f2bf(v);
-    const float vs = wave_reduce_sum(v * v);
-    if (col == 0) ssl[m] = vs;
-  }
-  __syncthreads();
-  if (threadIdx.x < 32)
-    ssout[(long)blockIdx.x * 32 + threadIdx.x] =
-        threadIdx.x < M ? ssl[threadIdx.x] : 0.f;
+  const long i = (long)m * N + n0 + col;
+  float v = bias ? bf2f(bias[n0 + col]) : 0.f;
+  for (int s = 0; s < ksplit; ++s) v += Cpart[s * total + i];
+  if (R) v += bf2f(R[i]);
+  C[i] = f2bf(v);
+  const float vs = wave_reduce_sum(v * v);
+  if (col == 0) ssout[(long)blockIdx.x * 32 + m] = vs;
 }
 
 __global__ void gemm_skinny_combine_kernel(

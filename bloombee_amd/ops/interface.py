@@ -300,13 +300,24 @@ def linear(x: torch.Tensor, w: torch.Tensor, residual: Optional[torch.Tensor] = 
             and K % 256 == 0 and N % 64 == 0):
         _require_ext()
         if norm is not None or ss_out is not None:
-            nw = norm[0].contiguous() if norm is not None else None
-            eps = float(norm[1]) if norm is not None else 0.0
+            nw = None
+            eps = 0.0
+            mode = 0
+            if norm is not None:
+                eps = float(norm[1])
+                if norm[0] is None:
+                    mode = 2  # weight pre-folded into W: invr-only scaling
+                else:
+                    nw = norm[0].contiguous()
+                    mode = 1
             return hip_ops.gemm_skinny(x.contiguous(), w, residual, bias, 0,
-                                       nw, eps, ss_in, ss_out)
+                                       nw, eps, ss_in, ss_out, mode)
         return hip_ops.gemm_skinny(x.contiguous(), w, residual, bias, 0)
     if norm is not None:
-        x = rms_norm(x, norm[0], norm[1])
+        nw = norm[0]
+        if nw is None:
+            nw = torch.ones(x.shape[-1], dtype=x.dtype, device=x.device)
+        x = rms_norm(x, nw, norm[1])
     y = torch.nn.functional.linear(x, w, bias)
     if residual is not None:
         y = y + residual.view_as(y)

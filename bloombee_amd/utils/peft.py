@@ -130,5 +130,9 @@ def add_adapter_to_block(block: torch.nn.Module, name: str,
     if not isinstance(block, LoraAdapterMixin):
         block.__class__ = type(block.__class__.__name__ + "WithLora",
                                (LoraAdapterMixin, block.__class__), {})
+    # adapters are trained against the unfolded parameterization — undo the
+    # fused-norm weight fold (llama/block.py fold_norm_weights) if applied
+    if getattr(block, "_norm_folded", False):
+        block.unfold_norm_weights()
     block.init_lora()
     block.add_adapter(name, sets)
