@@ -636,7 +636,7 @@ torch::Tensor gemm_skinny(torch::Tensor A, torch::Tensor W,
   const float* ssinp = nullptr;
   int nstripes = 0;
   if (ss_in.has_value()) {
-    TORCH_CHECK(nwp, "ss_in only meaningful with norm_w");
+    TORCH_CHECK(norm_mode != 0, "ss_in only meaningful with a norm mode");
     TORCH_CHECK(ss_in->is_contiguous() && ss_in->scalar_type() == at::kFloat &&
                 ss_in->numel() % 32 == 0, "ss_in must be f32 (nstripes,32)");
     ssinp = ss_in->data_ptr<float>();
