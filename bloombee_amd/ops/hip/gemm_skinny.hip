@@ -244,32 +244,47 @@ __global__ __launch_bounds__(256) void gemm_skinny_v2_kernel(
     // producer stats). 256 threads = 32 rows x 8 sub-readers; the first
     // W-register set is issued first so the weight stream is already in
     // flight while the prepass reduces.
-    const int row = threadIdx.x >> 3, sub = threadIdx.x & 7;
     issue0(k0);
-    float ss = 0.f;
     if (ssin) {
-      if (row < M)
-        for (int s = sub; s < nstripes; s += 8) ss += ssin[s * 32 + row];
-    } else if (row < M) {
-      const unsigned short* p = A + (long)row * K + sub * (K / 8);
-      for (int x = 0; x < K / 8; x += 8) {
-        short8 v = *reinterpret_cast<const short8*>(p + x);
+      // Coalesced stripe reduce: lanes sweep the contiguous 32-float rows
+      // of each stripe (one 128 B line per stripe), 8 stripe-groups across
+      // the workgroup, folded in LDS.
+      const int row = threadIdx.x & 31, grp = threadIdx.x >> 5;
+      float ss = 0.f;
+      for (int s = grp; s < nstripes; s += 8) ss += ssin[(long)s * 32 + row];
+      ssp[threadIdx.x] = ss;
+      __syncthreads();
+      if (threadIdx.x < 32) {
+        float t = 0.f;
 #pragma unroll
-        for (int e = 0; e < 8; ++e) {
-          const float f = bf2f((unsigned short)v[e]);
-          ss += f * f;
+        for (int g = 0; g < 8; ++g) t += ssp[g * 32 + threadIdx.x];
+        invr[threadIdx.x] = rsqrtf(t / K + eps);
+      }
+      __syncthreads();
+    } else {
+      const int row = threadIdx.x >> 3, sub = threadIdx.x & 7;
+      float ss = 0.f;
+      if (row < M) {
+        const unsigned short* p = A + (long)row * K + sub * (K / 8);
+        for (int x = 0; x < K / 8; x += 8) {
+          short8 v = *reinterpret_cast<const short8*>(p + x);
+#pragma unroll
+          for (int e = 0; e < 8; ++e) {
+            const float f = bf2f((unsigned short)v[e]);
+            ss += f * f;
+          }
         }
       }
-    }
-    ssp[threadIdx.x] = ss;
-    __syncthreads();
-    if (sub == 0 && row < M) {
-      float t = 0.f;
+      ssp[threadIdx.x] = ss;
+      __syncthreads();
+      if (sub == 0 && row < M) {
+        float t = 0.f;
 #pragma unroll
-      for (int e = 0; e < 8; ++e) t += ssp[(row << 3) + e];
-      invr[row] = rsqrtf(t / K + eps);
+        for (int e = 0; e < 8; ++e) t += ssp[(row << 3) + e];
+        invr[row] = rsqrtf(t / K + eps);
+      }
+      __syncthreads();
     }
-    __syncthreads();
     load_a(k0);
     store_a();
     __syncthreads();
