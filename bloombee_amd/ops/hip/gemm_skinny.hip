@@ -235,24 +235,23 @@ __global__ __launch_bounds__(256) void gemm_skinny_v2_kernel(
       }
   };
 
-  if (norm_mode) {
-    // Per-row sum of squares over FULL K (every split needs the global rms,
-    // not its chunk's). Cheap path: the PRODUCING GEMM's epilogue left
-    // per-stripe row sums (ssin) — summing nstripes*32 f32 from L2 costs
-    // ~nothing (profiles/r02 §13: the A-restream fallback below costs
-    // +18..60 us/GEMM under split-K and is only for callers with no
-    // producer stats). 256 threads = 32 rows x 8 sub-readers; the first
-    // W-register set is issued first so the weight stream is already in
-    // flight while the prepass reduces.
-    issue0(k0);
+  // invr reduce from producer stats (coalesced: lanes sweep each stripe's
+  // contiguous 32 floats; 4 accumulator chains keep loads in flight) or
+  // from re-streaming A (fallback for callers with no producer stats —
+  // +18-60 us/GEMM under split-K, profiles/r02 §13).
+  auto reduce_invr = [&]() {
     if (ssin) {
-      // Coalesced stripe reduce: lanes sweep the contiguous 32-float rows
-      // of each stripe (one 128 B line per stripe), 8 stripe-groups across
-      // the workgroup, folded in LDS.
       const int row = threadIdx.x & 31, grp = threadIdx.x >> 5;
-      float ss = 0.f;
-      for (int s = grp; s < nstripes; s += 8) ss += ssin[(long)s * 32 + row];
-      ssp[threadIdx.x] = ss;
+      float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
+      int s = grp;
+      for (; s + 24 < nstripes; s += 32) {
+        s0 += ssin[(long)s * 32 + row];
+        s1 += ssin[(long)(s + 8) * 32 + row];
+        s2 += ssin[(long)(s + 16) * 32 + row];
+        s3 += ssin[(long)(s + 24) * 32 + row];
+      }
+      for (; s < nstripes; s += 8) s0 += ssin[(long)s * 32 + row];
+      ssp[threadIdx.x] = (s0 + s1) + (s2 + s3);
       __syncthreads();
       if (threadIdx.x < 32) {
         float t = 0.f;
@@ -285,12 +284,21 @@ __global__ __launch_bounds__(256) void gemm_skinny_v2_kernel(
       }
       __syncthreads();
     }
+  };
+
+  if (norm_mode == 1) {
+    // mode 1 scales A as it stages: invr must exist BEFORE the pipeline.
+    // Issue the first W-register set first so the stream is in flight.
+    issue0(k0);
+    reduce_invr();
     load_a(k0);
     store_a();
     __syncthreads();
   } else {
     // Pipeline: stage s's A sits in LDS while its two B register sets
     // stream; stage s+1's A loads issue right after the barrier frees them.
+    // mode 2 defers its invr reduce to just before the epilogue — the
+    // weight stream starts with zero added latency.
     load_a(k0);
     issue0(k0);
     store_a();
@@ -308,6 +316,8 @@ __global__ __launch_bounds__(256) void gemm_skinny_v2_kernel(
       __syncthreads();
     }
   }
+
+  if (norm_mode == 2) reduce_invr();
 
   if (ksplit == 1) {
     if (ssout == nullptr) {
