@@ -327,17 +327,17 @@ def linear(x: torch.Tensor, w: torch.Tensor, residual: Optional[torch.Tensor] = 
     return y
 
 
-_FUSE_NORM = os.environ.get("BBAMD_FUSE_NORM", "0") == "1"
+_FUSE_NORM = os.environ.get("BBAMD_FUSE_NORM", "1") != "0"
 
 
 def fuse_norm_linear_ok(x: torch.Tensor, w: torch.Tensor) -> bool:
     """True when linear(norm=...) will take the fused-kernel path (so the
-    caller can skip emitting a separate rms_norm). DEFAULT OFF
-    (BBAMD_FUSE_NORM=1 enables): measured 4266 vs 5054 tok/s on the
-    flagship bench — the per-WG full-K variance prepass costs more than
-    the two saved launches once split-K multiplies the redundant A reads
-    (profiles/r02 §13). Kept as the numerics-verified base for the
-    epilogue-emitted-stats design (ROUND3.md item 2)."""
+    caller can skip emitting a separate rms_norm). Default ON
+    (BBAMD_FUSE_NORM=0 reverts): +2.9% on the flagship decode bench
+    (5289 vs 5140 tok/s) once the row stats ride the producer GEMM's
+    epilogue and the norm weights are folded into the projections
+    (profiles/r02 §13; the naive per-WG A-restream variant measured 16%
+    SLOWER — the chain is what makes it pay)."""
     K = x.shape[-1]
     return (_FUSE_NORM and _on_gpu(x) and x.numel() // K <= 32
             and x.dtype == torch.bfloat16 and K % 256 == 0
