@@ -314,10 +314,15 @@ def test_linear_dispatch_matches_f_linear():
 
 
 @pytest.mark.parametrize("model", ["llama-tiny", "qwen3-tiny", "mixtral-tiny"])
-def test_family_engine_gpu_matches_cpu_tokens(model):
+def test_family_engine_gpu_matches_cpu_tokens(model, monkeypatch):
     """Families whose decode path runs the HIP kernels must produce the same
-    greedy tokens as the CPU reference engine (identical weights, copied)."""
+    greedy tokens as the CPU reference engine (identical weights, copied).
+    Fused-norm is pinned OFF here: it reparameterizes the weights (norm
+    fold), which legitimately shifts near-flat random-init logits — its
+    own parity is covered by test_llama_block_fused_norm_matches_unfused."""
     from bloombee_amd.engine import LocalEngine
+    from bloombee_amd.ops import interface as iface
+    monkeypatch.setattr(iface, "_FUSE_NORM", False)
 
     ids = torch.randint(0, 900, (2, 12), generator=torch.Generator().manual_seed(2))
     cpu = LocalEngine(model, device="cpu", seed=0, kv_max_tokens=8192)
