@@ -57,7 +57,16 @@ DEVINL float quad16_reduce_sum(float x) {
 // shuffle + LDS-relayout overhead per position. Costs ~40 VGPRs (V
 // fragments + score regs for the second tile); host falls back to NT=1 for
 // short splits.
-template <int D, int MAXG, int NT>
+// PIPE=1 (NT must be 1): register double-buffered software pipeline — the
+// next tile's K AND V loads are issued before the current tile's MFMA/
+// softmax consume their buffers, so a full iteration of compute covers each
+// load's latency. The r02 PMC diagnosis (WAIT_ANY 82%, ACTIVE 12%, no DRAM
+// credit stalls, and n_split>0 makes things WORSE) shows the kernel is
+// per-wave-latency-bound, not occupancy- or bandwidth-bound: the serial
+// K-wait -> softmax -> PV chain leaves ~2.5 us/iter of HBM idle per wave.
+// Costs ~96 VGPRs of double buffers — free at this grid (B*Hkv WGs = 1
+// WG/CU: occupancy is grid-bound, not VGPR-bound).
+template <int D, int MAXG, int NT, int PIPE = 0>
 __global__ __launch_bounds__(256, 2) void attn_decode_mfma_kernel(
     const unsigned short* __restrict__ q,        // strided, see q_sb/q_sh
     const unsigned short* __restrict__ k_pages,  // (np, Hkv, P, D)
@@ -125,28 +134,68 @@ __global__ __launch_bounds__(256, 2) void attn_decode_mfma_kernel(
   const long head_slab_k = (long)kvh * P * D;  // same bytes, both layouts
   const int tile0 = (max(c0, lo) - c0) / DKVBLK;  // window skip, tile-aligned
 
-  // wave w owns NT adjacent tiles starting at tile w*NT, stride 4*NT tiles
-  for (int tb = c0 + (tile0 + wave * NT) * DKVBLK; tb < c1;
-       tb += 4 * NT * DKVBLK) {
-    // ---- V^T fragments for ALL NT tiles: direct A-layout loads, issued
-    // FIRST so the QK^T phase covers their latency. A[row=d][k=pos]: lane
-    // (li -> d row, hi -> position octet); 8 consecutive positions at fixed
-    // d are contiguous in the d-major pool. tb is 32-aligned and P | 32, so
-    // each octet sits in one page.
-    bf16x8 vfrag[NT][NDT];
-#pragma unroll
-    for (int t = 0; t < NT; ++t) {
-      const int vpos = tb + t * DKVBLK + hi * 8;
+  // PIPE double buffers (size-1 arrays when unused so the compiler drops
+  // them); loads target buffer `bi`, the MFMAs consume buffer `cur`.
+  bf16x8 vbuf[PIPE ? 2 : 1][PIPE ? NDT : 1];
+  bf16x8 kbuf[PIPE ? 2 : 1][PIPE ? NPT : 1][PIPE ? NKK : 1];
+  auto load_tile_pipe = [&](int bi, int tb) {
+    if constexpr (PIPE) {
+      const int vpos = tb + hi * 8;
       int vpage = -1;
       if (vpos < c1) vpage = page_table[b * maxp + vpos / P];
       const long vrow = ((long)vpage * Hkv + kvh) * D;
 #pragma unroll
       for (int n = 0; n < NDT; ++n) {
-        vfrag[t][n] = as_bf16x8(short8{});
+        vbuf[bi][n] = as_bf16x8(short8{});
         if (vpage >= 0)
-          vfrag[t][n] = as_bf16x8(__builtin_nontemporal_load(
+          vbuf[bi][n] = as_bf16x8(__builtin_nontemporal_load(
               reinterpret_cast<const short8*>(
                   v_pages + (vrow + li + 16 * n) * P + (vpos % P))));
+      }
+#pragma unroll
+      for (int n = 0; n < NPT; ++n) {
+        const int pos = tb + li + 16 * n;
+        const int cpos = min(pos, c1 - 1);
+        const int page = page_table[b * maxp + cpos / P];
+        const unsigned short* krow = k_pages + ((long)page * Hkv) * P * D +
+                                     head_slab_k + (long)(cpos % P) * D;
+#pragma unroll
+        for (int kk = 0; kk < NKK; ++kk)
+          kbuf[bi][n][kk] = as_bf16x8(__builtin_nontemporal_load(
+              reinterpret_cast<const short8*>(krow + hi * 8 + 32 * kk)));
+      }
+    }
+  };
+
+  constexpr int TSTRIDE = 4 * NT * DKVBLK;
+  const int tbs = c0 + (tile0 + wave * NT) * DKVBLK;
+  int cur = 0;
+  if (PIPE && tbs < c1) load_tile_pipe(0, tbs);
+
+  // wave w owns NT adjacent tiles starting at tile w*NT, stride 4*NT tiles
+  for (int tb = tbs; tb < c1; tb += TSTRIDE) {
+    if (PIPE && tb + TSTRIDE < c1) load_tile_pipe(cur ^ 1, tb + TSTRIDE);
+    // ---- V^T fragments for ALL NT tiles: direct A-layout loads, issued
+    // FIRST so the QK^T phase covers their latency. A[row=d][k=pos]: lane
+    // (li -> d row, hi -> position octet); 8 consecutive positions at fixed
+    // d are contiguous in the d-major pool. tb is 32-aligned and P | 32, so
+    // each octet sits in one page. (PIPE path: already resident in vbuf.)
+    bf16x8 vfrag[NT][NDT];
+    if constexpr (!PIPE) {
+#pragma unroll
+      for (int t = 0; t < NT; ++t) {
+        const int vpos = tb + t * DKVBLK + hi * 8;
+        int vpage = -1;
+        if (vpos < c1) vpage = page_table[b * maxp + vpos / P];
+        const long vrow = ((long)vpage * Hkv + kvh) * D;
+#pragma unroll
+        for (int n = 0; n < NDT; ++n) {
+          vfrag[t][n] = as_bf16x8(short8{});
+          if (vpage >= 0)
+            vfrag[t][n] = as_bf16x8(__builtin_nontemporal_load(
+                reinterpret_cast<const short8*>(
+                    v_pages + (vrow + li + 16 * n) * P + (vpos % P))));
+        }
       }
     }
 
@@ -157,20 +206,27 @@ __global__ __launch_bounds__(256, 2) void attn_decode_mfma_kernel(
     for (int t = 0; t < NT; ++t)
 #pragma unroll
       for (int n = 0; n < NPT; ++n) {
-        const int pos = tb + t * DKVBLK + li + 16 * n;
-        const int cpos = min(pos, c1 - 1);
-        const int page = page_table[b * maxp + cpos / P];
-        const unsigned short* krow = k_pages + ((long)page * Hkv) * P * D +
-                                     head_slab_k + (long)(cpos % P) * D;
         st[t][n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+        if constexpr (PIPE) {
 #pragma unroll
-        for (int kk = 0; kk < NKK; ++kk) {
-          // KV is read once per decode step: nontemporal keeps L2 for the
-          // GEMM weight streams that follow in the same step
-          bf16x8 kfrag = as_bf16x8(__builtin_nontemporal_load(
-              reinterpret_cast<const short8*>(krow + hi * 8 + 32 * kk)));
-          st[t][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              kfrag, qfrag[kk], st[t][n], 0, 0, 0);
+          for (int kk = 0; kk < NKK; ++kk)
+            st[t][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                kbuf[cur][n][kk], qfrag[kk], st[t][n], 0, 0, 0);
+        } else {
+          const int pos = tb + t * DKVBLK + li + 16 * n;
+          const int cpos = min(pos, c1 - 1);
+          const int page = page_table[b * maxp + cpos / P];
+          const unsigned short* krow = k_pages + ((long)page * Hkv) * P * D +
+                                       head_slab_k + (long)(cpos % P) * D;
+#pragma unroll
+          for (int kk = 0; kk < NKK; ++kk) {
+            // KV is read once per decode step: nontemporal keeps L2 for the
+            // GEMM weight streams that follow in the same step
+            bf16x8 kfrag = as_bf16x8(__builtin_nontemporal_load(
+                reinterpret_cast<const short8*>(krow + hi * 8 + 32 * kk)));
+            st[t][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                kfrag, qfrag[kk], st[t][n], 0, 0, 0);
+          }
         }
       }
 
@@ -234,10 +290,15 @@ __global__ __launch_bounds__(256, 2) void attn_decode_mfma_kernel(
       bf16x8 pfrag = as_bf16x8(*reinterpret_cast<const short8*>(
           p_lds + li * PROW_B + t * DKVBLK * 2 + hi * 16));
 #pragma unroll
-      for (int n = 0; n < NDT; ++n)
-        acc_o[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vfrag[t][n], pfrag,
+      for (int n = 0; n < NDT; ++n) {
+        bf16x8 vv;
+        if constexpr (PIPE) vv = vbuf[cur][n];
+        else vv = vfrag[t][n];
+        acc_o[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vv, pfrag,
                                                            acc_o[n], 0, 0, 0);
+      }
     }
+    cur ^= 1;
   }
 
   // ---- merge the 4 waves ----

@@ -213,11 +213,25 @@ static void attn_decode_launch(const torch::Tensor& q, const torch::Tensor& kp,
     const char* e = std::getenv("BBAMD_ATTN_NT");
     return e ? std::atoi(e) : 2;
   }();
-  const bool nt2 = (nt_env >= 2) && (D <= 128);
+  // BBAMD_ATTN_PF=1: software-pipelined NT=1 variant (next tile's K+V
+  // double-buffered in registers; r02 WAIT_ANY diagnosis)
+  static const bool pf_env = [] {
+    const char* e = std::getenv("BBAMD_ATTN_PF");
+    return e && std::atoi(e) != 0;
+  }();
+  const bool pf = pf_env && (D <= 128);
+  const bool nt2 = (nt_env >= 2) && (D <= 128) && !pf;
   auto launch_mfma = [&](auto mg) {
     if constexpr (D == 512) {
       // wide-head path: d split across the 4 waves (gemma-4 global layers)
       attn_decode_wide_kernel<decltype(mg)::value>
+          <<<grid, 256, 0, cur_stream()>>>(
+              bf_ptr(q) + q_off, bf_ptr(kp), bf_ptr(vp), pt.data_ptr<int>(),
+              ctx.data_ptr<int>(), alibi_ptr, bf_ptr_mut(out),
+              pml.data_ptr<float>(), pacc.data_ptr<float>(), B, Hkv, G, nch, P,
+              maxp, n_split, window, scale, q_sb, q_sh, out_sb, out_sh);
+    } else if (D <= 128 && pf) {
+      attn_decode_mfma_kernel<D, decltype(mg)::value, 1, 1>
           <<<grid, 256, 0, cur_stream()>>>(
               bf_ptr(q) + q_off, bf_ptr(kp), bf_ptr(vp), pt.data_ptr<int>(),
               ctx.data_ptr<int>(), alibi_ptr, bf_ptr_mut(out),
