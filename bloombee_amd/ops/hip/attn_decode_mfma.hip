@@ -134,21 +134,30 @@ __global__ __launch_bounds__(256, 2) void attn_decode_mfma_kernel(
   const long head_slab_k = (long)kvh * P * D;  // same bytes, both layouts
   const int tile0 = (max(c0, lo) - c0) / DKVBLK;  // window skip, tile-aligned
 
-  // PIPE double buffers (size-1 arrays when unused so the compiler drops
-  // them); loads target buffer `bi`, the MFMAs consume buffer `cur`.
-  bf16x8 vbuf[PIPE ? 2 : 1][PIPE ? NDT : 1];
-  bf16x8 kbuf[PIPE ? 2 : 1][PIPE ? NPT : 1][PIPE ? NKK : 1];
-  auto load_tile_pipe = [&](int bi, int tb) {
-    if constexpr (PIPE) {
+  // ---------------------------------------------------------------------
+  // PIPE=1: statically-unrolled register double buffer (two tiles per loop
+  // iteration). Dynamic buffer indexing (buf[cur]) demotes the arrays to
+  // scratch memory on amdgcn — the buffers must be distinct named arrays
+  // with compile-time indices, and sched_barrier pins each prefetch above
+  // the compute phase it overlaps (the scheduler otherwise sinks loads to
+  // their uses and the pipeline degenerates to the serial form).
+  constexpr int TSTRIDE = 4 * NT * DKVBLK;
+  const int tbs = c0 + (tile0 + wave * NT) * DKVBLK;
+
+  if constexpr (PIPE) {
+    static_assert(NT == 1, "PIPE implies NT==1");
+    bf16x8 vb0[NDT], vb1[NDT];
+    bf16x8 kb0[NPT][NKK], kb1[NPT][NKK];
+    auto load_tile = [&](bf16x8 (&vb)[NDT], bf16x8 (&kb)[NPT][NKK], int tb) {
       const int vpos = tb + hi * 8;
       int vpage = -1;
       if (vpos < c1) vpage = page_table[b * maxp + vpos / P];
       const long vrow = ((long)vpage * Hkv + kvh) * D;
 #pragma unroll
       for (int n = 0; n < NDT; ++n) {
-        vbuf[bi][n] = as_bf16x8(short8{});
+        vb[n] = as_bf16x8(short8{});
         if (vpage >= 0)
-          vbuf[bi][n] = as_bf16x8(__builtin_nontemporal_load(
+          vb[n] = as_bf16x8(__builtin_nontemporal_load(
               reinterpret_cast<const short8*>(
                   v_pages + (vrow + li + 16 * n) * P + (vpos % P))));
       }
@@ -161,41 +170,99 @@ __global__ __launch_bounds__(256, 2) void attn_decode_mfma_kernel(
                                      head_slab_k + (long)(cpos % P) * D;
 #pragma unroll
         for (int kk = 0; kk < NKK; ++kk)
-          kbuf[bi][n][kk] = as_bf16x8(__builtin_nontemporal_load(
+          kb[n][kk] = as_bf16x8(__builtin_nontemporal_load(
               reinterpret_cast<const short8*>(krow + hi * 8 + 32 * kk)));
       }
+    };
+    auto compute_tile = [&](const bf16x8 (&vb)[NDT],
+                            const bf16x8 (&kb)[NPT][NKK], int tb) {
+      f32x4 st[NPT];
+#pragma unroll
+      for (int n = 0; n < NPT; ++n) {
+        st[n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int kk = 0; kk < NKK; ++kk)
+          st[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              kb[n][kk], qfrag[kk], st[n], 0, 0, 0);
+      }
+      float p[NPT][4];
+      float rm = NEG_BIG;
+#pragma unroll
+      for (int n = 0; n < NPT; ++n)
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg) {
+          const int kpos = tb + 16 * n + hi * 4 + reg;
+          float sv = st[n][reg] * sc2 + aslope * (float)kpos;
+          const bool dead = (kpos >= c1) | (kpos < lo);
+          sv = dead ? NEG_BIG : sv;
+          p[n][reg] = sv;
+          rm = fmaxf(rm, sv);
+        }
+      rm = quad16_reduce_max(rm);
+      const float mn = fmaxf(m2, rm);
+      const float corr = fast_exp2(m2 - mn);
+      float psum = 0.f;
+#pragma unroll
+      for (int n = 0; n < NPT; ++n)
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg) {
+          p[n][reg] = fast_exp2(p[n][reg] - mn);
+          psum += p[n][reg];
+        }
+      psum = quad16_reduce_sum(psum);
+      l = l * corr + psum;
+      m2 = mn;
+      if (__builtin_amdgcn_ballot_w64(corr < 0.9999f)) {
+#pragma unroll
+        for (int n = 0; n < NDT; ++n)
+#pragma unroll
+          for (int reg = 0; reg < 4; ++reg) acc_o[n][reg] *= corr;
+      }
+#pragma unroll
+      for (int n = 0; n < NPT; ++n)
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg)
+          *(unsigned short*)(p_lds + li * PROW_B +
+                             (16 * n + hi * 4 + reg) * 2) = f2bf(p[n][reg]);
+      bf16x8 pfrag = as_bf16x8(
+          *reinterpret_cast<const short8*>(p_lds + li * PROW_B + hi * 16));
+#pragma unroll
+      for (int n = 0; n < NDT; ++n)
+        acc_o[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vb[n], pfrag,
+                                                           acc_o[n], 0, 0, 0);
+    };
+    if (tbs < c1) load_tile(vb0, kb0, tbs);
+    for (int tb = tbs; tb < c1; tb += 2 * TSTRIDE) {
+      if (tb + TSTRIDE < c1) load_tile(vb1, kb1, tb + TSTRIDE);
+      __builtin_amdgcn_sched_barrier(0);
+      compute_tile(vb0, kb0, tb);
+      if (tb + TSTRIDE >= c1) break;
+      if (tb + 2 * TSTRIDE < c1) load_tile(vb0, kb0, tb + 2 * TSTRIDE);
+      __builtin_amdgcn_sched_barrier(0);
+      compute_tile(vb1, kb1, tb + TSTRIDE);
     }
-  };
-
-  constexpr int TSTRIDE = 4 * NT * DKVBLK;
-  const int tbs = c0 + (tile0 + wave * NT) * DKVBLK;
-  int cur = 0;
-  if (PIPE && tbs < c1) load_tile_pipe(0, tbs);
-
+  } else {
   // wave w owns NT adjacent tiles starting at tile w*NT, stride 4*NT tiles
   for (int tb = tbs; tb < c1; tb += TSTRIDE) {
-    if (PIPE && tb + TSTRIDE < c1) load_tile_pipe(cur ^ 1, tb + TSTRIDE);
     // ---- V^T fragments for ALL NT tiles: direct A-layout loads, issued
     // FIRST so the QK^T phase covers their latency. A[row=d][k=pos]: lane
     // (li -> d row, hi -> position octet); 8 consecutive positions at fixed
     // d are contiguous in the d-major pool. tb is 32-aligned and P | 32, so
-    // each octet sits in one page. (PIPE path: already resident in vbuf.)
+    // each octet sits in one page.
     bf16x8 vfrag[NT][NDT];
-    if constexpr (!PIPE) {
 #pragma unroll
-      for (int t = 0; t < NT; ++t) {
-        const int vpos = tb + t * DKVBLK + hi * 8;
-        int vpage = -1;
-        if (vpos < c1) vpage = page_table[b * maxp + vpos / P];
-        const long vrow = ((long)vpage * Hkv + kvh) * D;
+    for (int t = 0; t < NT; ++t) {
+      const int vpos = tb + t * DKVBLK + hi * 8;
+      int vpage = -1;
+      if (vpos < c1) vpage = page_table[b * maxp + vpos / P];
+      const long vrow = ((long)vpage * Hkv + kvh) * D;
 #pragma unroll
-        for (int n = 0; n < NDT; ++n) {
-          vfrag[t][n] = as_bf16x8(short8{});
-          if (vpage >= 0)
-            vfrag[t][n] = as_bf16x8(__builtin_nontemporal_load(
-                reinterpret_cast<const short8*>(
-                    v_pages + (vrow + li + 16 * n) * P + (vpos % P))));
-        }
+      for (int n = 0; n < NDT; ++n) {
+        vfrag[t][n] = as_bf16x8(short8{});
+        if (vpage >= 0)
+          vfrag[t][n] = as_bf16x8(__builtin_nontemporal_load(
+              reinterpret_cast<const short8*>(
+                  v_pages + (vrow + li + 16 * n) * P + (vpos % P))));
       }
     }
 
@@ -206,27 +273,20 @@ __global__ __launch_bounds__(256, 2) void attn_decode_mfma_kernel(
     for (int t = 0; t < NT; ++t)
 #pragma unroll
       for (int n = 0; n < NPT; ++n) {
+        const int pos = tb + t * DKVBLK + li + 16 * n;
+        const int cpos = min(pos, c1 - 1);
+        const int page = page_table[b * maxp + cpos / P];
+        const unsigned short* krow = k_pages + ((long)page * Hkv) * P * D +
+                                     head_slab_k + (long)(cpos % P) * D;
         st[t][n] = (f32x4){0.f, 0.f, 0.f, 0.f};
-        if constexpr (PIPE) {
 #pragma unroll
-          for (int kk = 0; kk < NKK; ++kk)
-            st[t][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                kbuf[cur][n][kk], qfrag[kk], st[t][n], 0, 0, 0);
-        } else {
-          const int pos = tb + t * DKVBLK + li + 16 * n;
-          const int cpos = min(pos, c1 - 1);
-          const int page = page_table[b * maxp + cpos / P];
-          const unsigned short* krow = k_pages + ((long)page * Hkv) * P * D +
-                                       head_slab_k + (long)(cpos % P) * D;
-#pragma unroll
-          for (int kk = 0; kk < NKK; ++kk) {
-            // KV is read once per decode step: nontemporal keeps L2 for the
-            // GEMM weight streams that follow in the same step
-            bf16x8 kfrag = as_bf16x8(__builtin_nontemporal_load(
-                reinterpret_cast<const short8*>(krow + hi * 8 + 32 * kk)));
-            st[t][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                kfrag, qfrag[kk], st[t][n], 0, 0, 0);
-          }
+        for (int kk = 0; kk < NKK; ++kk) {
+          // KV is read once per decode step: nontemporal keeps L2 for the
+          // GEMM weight streams that follow in the same step
+          bf16x8 kfrag = as_bf16x8(__builtin_nontemporal_load(
+              reinterpret_cast<const short8*>(krow + hi * 8 + 32 * kk)));
+          st[t][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              kfrag, qfrag[kk], st[t][n], 0, 0, 0);
         }
       }
 
@@ -290,15 +350,11 @@ __global__ __launch_bounds__(256, 2) void attn_decode_mfma_kernel(
       bf16x8 pfrag = as_bf16x8(*reinterpret_cast<const short8*>(
           p_lds + li * PROW_B + t * DKVBLK * 2 + hi * 16));
 #pragma unroll
-      for (int n = 0; n < NDT; ++n) {
-        bf16x8 vv;
-        if constexpr (PIPE) vv = vbuf[cur][n];
-        else vv = vfrag[t][n];
-        acc_o[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vv, pfrag,
+      for (int n = 0; n < NDT; ++n)
+        acc_o[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vfrag[t][n], pfrag,
                                                            acc_o[n], 0, 0, 0);
-      }
     }
-    cur ^= 1;
+  }
   }
 
   // ---- merge the 4 waves ----
