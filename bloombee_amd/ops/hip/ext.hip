@@ -213,11 +213,13 @@ static void attn_decode_launch(const torch::Tensor& q, const torch::Tensor& kp,
     const char* e = std::getenv("BBAMD_ATTN_NT");
     return e ? std::atoi(e) : 2;
   }();
-  // BBAMD_ATTN_PF=1: software-pipelined NT=1 variant (next tile's K+V
-  // double-buffered in registers; r02 WAIT_ANY diagnosis)
+  // Software-pipelined variant (next tile's K+V double-buffered in
+  // registers; r02 WAIT_ANY diagnosis): DEFAULT — 50 us / 5.4 TB/s at B32
+  // ctx2048 and 6.3 TB/s (achievable-HBM ceiling) at ctx8192 vs 59-62 us
+  // for the NT variants. BBAMD_ATTN_PF=0 reverts to the NT path.
   static const bool pf_env = [] {
     const char* e = std::getenv("BBAMD_ATTN_PF");
-    return e && std::atoi(e) != 0;
+    return !e || std::atoi(e) != 0;
   }();
   const bool pf = pf_env && (D <= 128);
   const bool nt2 = (nt_env >= 2) && (D <= 128) && !pf;
@@ -290,7 +292,14 @@ static torch::Tensor attn_decode_core(
       const char* e = std::getenv("BBAMD_ATTN_NT");
       return e ? std::atoi(e) : 2;
     }();
-    const long target = (nt_env0 >= 2 && D <= 128) ? 768 : 1024;
+    static const bool pf_env0 = [] {
+      const char* e = std::getenv("BBAMD_ATTN_PF");
+      return !e || std::atoi(e) != 0;
+    }();
+    // pipelined variant: 254 VGPRs -> 2 waves/SIMD -> 512 WGs fill the chip
+    // (measured: B32 ns=2 50 us / 5.4 TB/s vs 59 at 768-1024 WGs)
+    const long target = (pf_env0 && D <= 128) ? 512
+                        : (nt_env0 >= 2 && D <= 128) ? 768 : 1024;
     n_split = (int)std::max<long>(
         1, std::min<long>(32, target / std::max(1, B * Hkv * nch)));
   }
