@@ -100,6 +100,24 @@ class LlamaBlock(torch.nn.Module):
                                     device=dev).mul_(std).to(w.dtype))
         return self
 
+    def _fuse_norm_gate(self, hidden) -> bool:
+        """All four decode GEMMs must hit the v2 kernel (K%256) with
+        stripe-aligned N so the fused-norm ss chain stays on-device;
+        W4/LoRA blocks keep the separate-norm path."""
+        return (getattr(self, "_w4", None) is None
+                and getattr(self, "lora_delta", None) is None
+                and ops.fuse_norm_linear_ok(hidden, self.qkv_w)
+                and self.I % 256 == 0 and (2 * self.I) % 64 == 0
+                and (self.Hq * self.D) % 256 == 0
+                and self.config.hidden_size % 64 == 0)
+
+    def _ensure_ss_bufs(self, device) -> None:
+        if self._ss_h2 is None or self._ss_h2.device != device:
+            st = self.config.hidden_size // 64
+            self._ss_h2 = torch.empty(st * 32, dtype=torch.float32,
+                                      device=device)
+            self._ss_hidden = torch.empty_like(self._ss_h2)
+
     @torch.no_grad()
     def fold_norm_weights(self):
         """Reparameterize for the fused-norm decode path: qkv_w <- qkv_w *
@@ -201,14 +219,7 @@ class LlamaBlock(torch.nn.Module):
         # (ss buffers chained block-to-block via prev_block; re-streaming
         # A instead costs +18-60 us/GEMM — profiles/r02 §13). W4/LoRA
         # blocks and prefill keep the separate-norm path.
-        fuse_norm = (getattr(self, "_w4", None) is None
-                     and getattr(self, "lora_delta", None) is None
-                     and ops.fuse_norm_linear_ok(hidden, self.qkv_w)
-                     # all four GEMMs must hit the v2 kernel (K%256) with
-                     # stripe-aligned N so the ss chain stays on-device
-                     and self.I % 256 == 0 and (2 * self.I) % 64 == 0
-                     and (self.Hq * self.D) % 256 == 0
-                     and self.config.hidden_size % 64 == 0)
+        fuse_norm = self._fuse_norm_gate(hidden)
         self._ss_valid = False
         if fuse_norm and not getattr(self, "_norm_folded", False):
             self.fold_norm_weights()   # one-time reparameterization
@@ -266,11 +277,7 @@ class LlamaBlock(torch.nn.Module):
             # also emits row sum-of-squares; the post-attention rmsnorm
             # folds into the gate_up A-stage consuming them. down-proj
             # leaves the stats for the NEXT block's input norm.
-            if self._ss_h2 is None or self._ss_h2.device != hidden.device:
-                st = self.config.hidden_size // 64
-                self._ss_h2 = torch.empty(st * 32, dtype=torch.float32,
-                                          device=hidden.device)
-                self._ss_hidden = torch.empty_like(self._ss_h2)
+            self._ensure_ss_bufs(hidden.device)
             h2 = ops.linear(attn, self.o_w, residual=hidden,
                             ss_out=self._ss_h2)
             gu = ops.linear(h2, self.gate_up_w,

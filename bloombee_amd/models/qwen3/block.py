@@ -34,8 +34,21 @@ class Qwen3Block(LlamaBlock):
         Hq, Hkv, D = self.Hq, self.Hkv, self.D
         cfg = self.config
 
-        x = ops.rms_norm(hidden, self.input_norm_w, cfg.rms_norm_eps)
-        qkv = self._lin(x, self.qkv_w, "qkv_w")
+        # fused-norm chain (see llama/block.py): both rmsnorms ride the
+        # GEMMs; the per-head q/k norms below act on the GEMM OUTPUT and
+        # are unaffected by the input-norm weight fold
+        fuse_norm = self._fuse_norm_gate(hidden)
+        self._ss_valid = False
+        if fuse_norm and not getattr(self, "_norm_folded", False):
+            self.fold_norm_weights()
+        pb = getattr(self, "prev_block", None)
+        if fuse_norm and pb is not None and getattr(pb, "_ss_valid", False):
+            qkv = ops.linear(hidden, self.qkv_w,
+                             norm=(None, cfg.rms_norm_eps),
+                             ss_in=pb._ss_hidden)
+        else:
+            x = ops.rms_norm(hidden, self.input_norm_w, cfg.rms_norm_eps)
+            qkv = self._lin(x, self.qkv_w, "qkv_w")
         # per-head q/k RMS norm on the fused buffer (contiguous D-sized rows)
         qk = qkv[..., :(Hq + Hkv) * D]
         q_flat = qkv[..., :Hq * D].reshape(-1, D)
@@ -51,6 +64,16 @@ class Qwen3Block(LlamaBlock):
         ops.rope_kv_write_(qkv, Hq, Hkv, cos, sin, position_ids, kp, vp, pt,
                            start_pos)
         attn = ops.attn_paged_qkv(qkv, Hq, Hkv, kp, vp, pt, start_pos, self.scale)
+        if fuse_norm:
+            self._ensure_ss_bufs(hidden.device)
+            h2 = ops.linear(attn, self.o_w, residual=hidden,
+                            ss_out=self._ss_h2)
+            gu = ops.linear(h2, self.gate_up_w,
+                            norm=(None, cfg.rms_norm_eps), ss_in=self._ss_h2)
+            out = ops.linear(ops.swiglu(gu), self.down_w, residual=h2,
+                             ss_out=self._ss_hidden)
+            self._ss_valid = True
+            return out
         a = self._lin(attn, self.o_w, "o_w")
         h2, y = ops.rms_norm_residual(a, hidden, self.post_norm_w, cfg.rms_norm_eps)
         return self._lin(ops.swiglu(self._lin(y, self.gate_up_w, "gate_up_w")),

@@ -638,3 +638,31 @@ def test_gemm_skinny_ss_chain():
         err = (y_self.float() - y_ssin.float()).abs().max().item()
         sc = y_self.float().abs().max().clamp_min(1.0).item()
         assert err / sc < 1e-2, (M, N, K, err)
+
+
+@pytest.mark.gpu
+def test_qwen3_block_fused_norm_matches_unfused():
+    """Qwen3 inherits the fused-norm chain (q/k per-head norms act on the
+    GEMM output and are unaffected by the input-norm fold)."""
+    from bloombee_amd.engine import BlockStack
+    from bloombee_amd.models.base import resolve_config
+    from bloombee_amd.ops import interface as iface
+
+    cfg = resolve_config("qwen3-tiny")
+    stack = BlockStack(cfg, 0, 2, device=DEV, seed=21)
+    gen = torch.Generator().manual_seed(5)
+    x = (torch.randn(2, 1, cfg.hidden_size, generator=gen) * 0.1).to(cfg.dtype)
+    sp = torch.zeros(2, dtype=torch.int32)
+    outs = []
+    for fuse in (True, False):
+        kv = stack.make_kv(1 << 10)
+        h = kv.allocate(2, 32)
+        h.extend(1)
+        old = iface._FUSE_NORM
+        iface._FUSE_NORM = fuse
+        try:
+            outs.append(stack.forward_inference(x.to(DEV), h, sp.to(DEV)))
+        finally:
+            iface._FUSE_NORM = old
+    err = (outs[0].float() - outs[1].float()).abs().max().item()
+    assert err < 3e-2, err
