@@ -307,12 +307,8 @@ __global__ __launch_bounds__(256) void gemm_skinny_v2_kernel(
   for (int k = k0; k < k1; k += KSTEP) {
     if (k + KSTEP < k1) load_a(k + KSTEP);
     issue1(k + 128);
-    // pin the W-register-set issues above the MFMA phase that covers them
-    // (same amdgpu-scheduler sinking found in attn PIPE, r02 §14)
-    __builtin_amdgcn_sched_barrier(0);
     mfma_set(bB0, 0);
     if (k + KSTEP < k1) issue0(k + KSTEP);
-    __builtin_amdgcn_sched_barrier(0);
     mfma_set(bB1, 1);
     if (k + KSTEP < k1) {
       __syncthreads();  // everyone done reading stage s

@@ -218,11 +218,8 @@ __global__ __launch_bounds__(256) void moe_gemm_v2_kernel(
   for (int k = 0; k < K; k += KSTEP) {
     if (k + KSTEP < K) load_a(k + KSTEP);
     issue1(k + 128);
-    // keep the W issues above the MFMA phase (attn PIPE lesson, r02 §14)
-    __builtin_amdgcn_sched_barrier(0);
     mfma_set(bB0, 0);
     if (k + KSTEP < K) issue0(k + KSTEP);
-    __builtin_amdgcn_sched_barrier(0);
     mfma_set(bB1, 1);
     if (k + KSTEP < K) {
       __syncthreads();
