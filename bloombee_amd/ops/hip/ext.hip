@@ -554,18 +554,29 @@ torch::Tensor gemm_w4(torch::Tensor A, torch::Tensor Wq, torch::Tensor scale,
     TORCH_CHECK(bias->is_contiguous() && bias->numel() == N);
     bp = bf_ptr(*bias);
   }
-  // M==1 (the speculative-draft GEMV shape): contiguous 1 KB code loads +
-  // v_dot2 — measured 2x the bf16 kernel (w4_probe.hip). rows_per_wave=4.
-  if (M == 1 && K % 2048 == 0 && K / 2048 <= 7) {
+  // M<=4 (speculative-draft / small-batch serving): contiguous 1 KB code
+  // loads + v_dot2, the dequant amortized over the M rows — measured 2x
+  // the bf16 kernel at M=1 (w4_probe.hip); the MFMA form only reaches
+  // 1.35x, so the GEMV form carries M<=4 too (ROUND3 item 4).
+  if (M <= 4 && K % 2048 == 0 && K / 2048 <= 7) {
     const int rpw = 4;
     dim3 gv((N / rpw + 3) / 4);
+    auto lv2 = [&](auto nl, auto mr) {
+      gemm_w4_gemv_kernel<decltype(nl)::value, decltype(mr)::value>
+          <<<gv, 256, 0, cur_stream()>>>(
+              reinterpret_cast<const _Float16*>(A.data_ptr()),
+              Wq.data_ptr<unsigned char>(),
+              reinterpret_cast<const __half*>(scale.data_ptr()),
+              reinterpret_cast<const __half*>(zero.data_ptr()),
+              rp, bp, bf_ptr_mut(C), N, K, rpw);
+    };
     auto lv = [&](auto nl) {
-      gemm_w4_gemv_kernel<decltype(nl)::value><<<gv, 256, 0, cur_stream()>>>(
-          reinterpret_cast<const _Float16*>(A.data_ptr()),
-          Wq.data_ptr<unsigned char>(),
-          reinterpret_cast<const __half*>(scale.data_ptr()),
-          reinterpret_cast<const __half*>(zero.data_ptr()),
-          rp, bp, bf_ptr_mut(C), N, K, rpw);
+      switch (M) {
+        case 1: lv2(nl, std::integral_constant<int, 1>{}); break;
+        case 2: lv2(nl, std::integral_constant<int, 2>{}); break;
+        case 3: lv2(nl, std::integral_constant<int, 3>{}); break;
+        default: lv2(nl, std::integral_constant<int, 4>{}); break;
+      }
     };
     switch (K / 2048) {
       case 1: lv(std::integral_constant<int, 1>{}); break;

@@ -53,31 +53,37 @@ DEVINL half8 dq8_q4_f16(unsigned int c, half2v sc2, half2v zp2) {
 // PRELOAD: K <= 4096 keeps the whole A row in registers (K/64 half2/lane);
 // larger K re-reads A fragments from L1 inside the loop.
 // ---------------------------------------------------------------------------
-template <int NLOADS>  // K / 2048; <=2 preloads A into registers
+// MROWS (1..4): batch rows sharing one code stream — the dequant (the
+// issue-bound part, see profiles/r02 §4 PMC) is amortized over MROWS
+// v_dot2 chains, so M<=4 runs near the M=1 byte rate instead of falling
+// back to the 2.25 TB/s MFMA form (ROUND3 item 4).
+template <int NLOADS, int MROWS = 1>  // NLOADS = K / 2048
 __global__ __launch_bounds__(256) void gemm_w4_gemv_kernel(
-    const _Float16* __restrict__ A,           // (1, K) fp16
+    const _Float16* __restrict__ A,           // (MROWS, K) fp16
     const unsigned char* __restrict__ Wq,     // (N, K/2)
     const __half* __restrict__ scale,         // (N, K/64)
     const __half* __restrict__ zero,          // (N, K/64)
-    const unsigned short* __restrict__ R,     // (1, N) bf16 or null
+    const unsigned short* __restrict__ R,     // (MROWS, N) bf16 or null
     const unsigned short* __restrict__ bias,  // (N,) bf16 or null
-    unsigned short* __restrict__ C,           // (1, N) bf16
+    unsigned short* __restrict__ C,           // (MROWS, N) bf16
     int N, int K, int rows_per_wave) {
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int n_base = (blockIdx.x * 4 + wave) * rows_per_wave;
   if (n_base >= N) return;
-  constexpr bool PRELOAD = (NLOADS <= 2);
+  constexpr bool PRELOAD = (MROWS == 1 ? NLOADS <= 2 : MROWS * NLOADS <= 8);
   const half2v magic = {(_Float16)1024.f, (_Float16)1024.f};
 
-  half2v areg[PRELOAD ? NLOADS * 16 : 1];
+  half2v areg[PRELOAD ? MROWS * NLOADS * 16 : 1];
   if (PRELOAD) {
 #pragma unroll
-    for (int h = 0; h < NLOADS; ++h)
+    for (int m = 0; m < MROWS; ++m)
 #pragma unroll
-      for (int j = 0; j < 16; ++j)
-        areg[h * 16 + j] = *reinterpret_cast<const half2v*>(
-            A + h * 2048 + lane * 32 + 2 * j);
+      for (int h = 0; h < NLOADS; ++h)
+#pragma unroll
+        for (int j = 0; j < 16; ++j)
+          areg[(m * NLOADS + h) * 16 + j] = *reinterpret_cast<const half2v*>(
+              A + (long)m * K + h * 2048 + lane * 32 + 2 * j);
   }
 
   uint4v buf[2][NLOADS];  // [row parity][k-chunk]
@@ -95,7 +101,9 @@ __global__ __launch_bounds__(256) void gemm_w4_gemv_kernel(
     if (r + 1 < rows_per_wave && n + 1 < N) ldrow((r + 1) & 1, n + 1);
     const __half* srow = scale + (long)n * (K / 64);
     const __half* zrow = zero + (long)n * (K / 64);
-    float acc = 0.f;
+    float acc[MROWS];
+#pragma unroll
+    for (int m = 0; m < MROWS; ++m) acc[m] = 0.f;
 #pragma unroll
     for (int h = 0; h < NLOADS; ++h) {
       const int g = (h * 2048 + lane * 32) / 64;
@@ -114,23 +122,30 @@ __global__ __launch_bounds__(256) void gemm_w4_gemv_kernel(
               0x64006400u | (byte & 0xFu) | ((byte & 0xF0u) << 12);
           half2v v = __builtin_bit_cast(half2v, hh);
           v = (v - magic) * sc2 + zp2;
-          half2v a;
-          if constexpr (PRELOAD) {
-            a = areg[h * 16 + d * 4 + p];
-          } else {
-            a = *reinterpret_cast<const half2v*>(
-                A + h * 2048 + lane * 32 + (d * 4 + p) * 2);
+#pragma unroll
+          for (int m = 0; m < MROWS; ++m) {
+            half2v a;
+            if constexpr (PRELOAD) {
+              a = areg[(m * NLOADS + h) * 16 + d * 4 + p];
+            } else {
+              a = *reinterpret_cast<const half2v*>(
+                  A + (long)m * K + h * 2048 + lane * 32 + (d * 4 + p) * 2);
+            }
+            acc[m] = __builtin_amdgcn_fdot2(a, v, acc[m], false);
           }
-          acc = __builtin_amdgcn_fdot2(a, v, acc, false);
         }
       }
     }
 #pragma unroll
-    for (int m = 1; m < WAVE; m <<= 1) acc += __shfl_xor(acc, m);
-    if (lane == 0) {
-      if (bias) acc += bf2f(bias[n]);
-      if (R) acc += bf2f(R[n]);
-      C[n] = f2bf(acc);
+    for (int m = 0; m < MROWS; ++m) {
+#pragma unroll
+      for (int x = 1; x < WAVE; x <<= 1) acc[m] += __shfl_xor(acc[m], x);
+      if (lane == 0) {
+        float av = acc[m];
+        if (bias) av += bf2f(bias[n]);
+        if (R) av += bf2f(R[(long)m * N + n]);
+        C[(long)m * N + n] = f2bf(av);
+      }
     }
   }
 }

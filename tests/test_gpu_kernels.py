@@ -666,3 +666,24 @@ def test_qwen3_block_fused_norm_matches_unfused():
             iface._FUSE_NORM = old
     err = (outs[0].float() - outs[1].float()).abs().max().item()
     assert err < 3e-2, err
+
+
+@pytest.mark.gpu
+def test_gemm_w4_gemv_small_m_parity():
+    """W4 GEMV M<=4 path (dequant amortized over rows) vs dequantized
+    F.linear reference."""
+    torch.manual_seed(17)
+    for M in (1, 2, 3, 4):
+        for N, K in [(512, 2048), (1024, 4096), (256, 14336)]:
+            w = torch.randn(N, K, dtype=torch.bfloat16, device=DEV) * 0.05
+            packed, scale, zero = ops.quant4_pack(w)
+            wd = ops.quant4_unpack(packed.reshape(-1, 32), scale.reshape(-1),
+                                   zero.reshape(-1),
+                                   dtype=torch.bfloat16).reshape(N, K)
+            x = torch.randn(M, K, dtype=torch.bfloat16, device=DEV) * 0.3
+            r = torch.randn(M, N, dtype=torch.bfloat16, device=DEV)
+            want = torch.nn.functional.linear(x.float(), wd.float()) + r.float()
+            got = ops.linear_w4(x, packed, scale, zero, N, residual=r)
+            err = (got.float() - want).abs().max().item()
+            sc = want.abs().max().clamp_min(1.0).item()
+            assert err / sc < 2e-2, (M, N, K, err / sc)
