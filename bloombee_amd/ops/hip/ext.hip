@@ -554,11 +554,13 @@ torch::Tensor gemm_w4(torch::Tensor A, torch::Tensor Wq, torch::Tensor scale,
     TORCH_CHECK(bias->is_contiguous() && bias->numel() == N);
     bp = bf_ptr(*bias);
   }
-  // M<=4 (speculative-draft / small-batch serving): contiguous 1 KB code
-  // loads + v_dot2, the dequant amortized over the M rows — measured 2x
-  // the bf16 kernel at M=1 (w4_probe.hip); the MFMA form only reaches
-  // 1.35x, so the GEMV form carries M<=4 too (ROUND3 item 4).
-  if (M <= 4 && K % 2048 == 0 && K / 2048 <= 7) {
+  // M==1 (the speculative-draft GEMV shape): contiguous 1 KB code loads +
+  // v_dot2 — measured 2x the bf16 kernel (w4_probe.hip). The MROWS<=4
+  // extension was measured and REJECTED for dispatch: amortizing the
+  // dequant over rows loses to the split-K MFMA form (M4 down-proj 131 us
+  // vs ~13 — the per-lane 4 B A loads scale with M and dominate); the
+  // kernel keeps MROWS for completeness (parity-tested).
+  if (M == 1 && K % 2048 == 0 && K / 2048 <= 7) {
     const int rpw = 4;
     dim3 gv((N / rpw + 3) / 4);
     auto lv2 = [&](auto nl, auto mr) {
