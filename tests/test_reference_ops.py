@@ -313,3 +313,42 @@ def test_linear_norm_fallback_matches_separate():
     r = torch.randn(4, 64, dtype=torch.bfloat16)
     got2 = ops.linear(x, w, residual=r, norm=(nw, 1e-5))
     assert torch.equal(got2, want + r)
+
+
+def test_norm_weight_fold_is_reparameterization():
+    """fold_norm_weights is math-preserving on every path: the folded block
+    (projections *= norm weights, norms = 1) produces the same training
+    forward as the original within bf16 fold-rounding, and unfold restores
+    the original weights to ~1 ulp."""
+    from bloombee_amd.engine import BlockStack
+    from bloombee_amd.models.base import resolve_config
+
+    cfg = resolve_config("llama-tiny")
+    stack = BlockStack(cfg, 0, 1, device="cpu", seed=11)
+    blk = stack.blocks[0]
+    # non-trivial norm weights (init_random sets them to 1)
+    gen = torch.Generator().manual_seed(3)
+    blk.input_norm_w.data = 1.0 + 0.3 * torch.randn(
+        cfg.hidden_size, generator=gen).to(cfg.dtype)
+    blk.post_norm_w.data = 1.0 + 0.3 * torch.randn(
+        cfg.hidden_size, generator=gen).to(cfg.dtype)
+    x = (torch.randn(2, 4, cfg.hidden_size, generator=gen) * 0.1).to(cfg.dtype)
+
+    want = blk.forward_train(x.clone())
+    orig_qkv = blk.qkv_w.detach().clone()
+    orig_inw = blk.input_norm_w.detach().clone()
+
+    blk.fold_norm_weights()
+    assert blk._norm_folded
+    assert torch.all(blk.input_norm_w == 1.0)
+    got = blk.forward_train(x.clone())
+    rel = ((got.float() - want.float()).norm()
+           / want.float().norm().clamp_min(1e-6)).item()
+    assert rel < 2e-2, rel
+
+    blk.unfold_norm_weights()
+    assert not blk._norm_folded
+    assert torch.equal(blk.input_norm_w, orig_inw)
+    drift = ((blk.qkv_w.float() - orig_qkv.float()).abs()
+             / orig_qkv.float().abs().clamp_min(1e-3)).max().item()
+    assert drift < 2e-2, drift
