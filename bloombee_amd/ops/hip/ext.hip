@@ -752,7 +752,7 @@ torch::Tensor gemm_skinny(torch::Tensor A, torch::Tensor W,
   else launch(std::integral_constant<int, 2>{});
   if (ksplit > 1) {
     if (ssoutp) {
-      dim3 cg(N / 64, (M + 3) / 4);
+      dim3 cg(N / 64, (M + 15) / 16);
       gemm_skinny_combine_ss_kernel<<<cg, 256, 0, cur_stream()>>>(
           pp, rp, bp, bf_ptr_mut(C), ssoutp, M, N, ksplit);
     } else {

@@ -378,26 +378,41 @@ __global__ __launch_bounds__(256) void gemm_skinny_v2_kernel(
 }
 
 // Split-K combine that also emits per-stripe row sum-of-squares (ssout
-// layout (N/64, 32)): grid (N/64, ceil(M/4)), 256 threads = 4 rows x 64
-// cols, wave = row — each (stripe, m) handled by exactly one wave, the
-// lane dimension is the 64 columns, so the row sum is one wave_reduce
-// and no LDS/atomics. Same workgroup count as the flat combine.
+// layout (N/64, 32)): grid (N/64, ceil(M/16)); a wave covers 4 rows x 16
+// float4 column-groups of the 64-col stripe, so partial loads are 16 B
+// vectors (16 lanes x 16 B = one coalesced 256 B line run per split per
+// row), the row sum is an in-lane fold + a 16-lane shfl reduce, and each
+// (stripe, m) is produced by exactly one lane — no LDS, no atomics.
 __global__ __launch_bounds__(256) void gemm_skinny_combine_ss_kernel(
     const float* __restrict__ Cpart, const unsigned short* __restrict__ R,
     const unsigned short* __restrict__ bias, unsigned short* __restrict__ C,
     float* __restrict__ ssout, int M, int N, int ksplit) {
   const int n0 = blockIdx.x * 64;
-  const int col = threadIdx.x & 63;
-  const int m = blockIdx.y * 4 + (threadIdx.x >> 6);
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int c4 = (lane & 15) * 4;            // first of this lane's 4 cols
+  const int m = blockIdx.y * 16 + (threadIdx.x >> 6) * 4 + (lane >> 4);
   if (m >= M) return;
   const long total = (long)M * N;
-  const long i = (long)m * N + n0 + col;
-  float v = bias ? bf2f(bias[n0 + col]) : 0.f;
-  for (int s = 0; s < ksplit; ++s) v += Cpart[s * total + i];
-  if (R) v += bf2f(R[i]);
-  C[i] = f2bf(v);
-  const float vs = wave_reduce_sum(v * v);
-  if (col == 0) ssout[(long)blockIdx.x * 32 + m] = vs;
+  const long i = (long)m * N + n0 + c4;
+  f32x4 v = {0.f, 0.f, 0.f, 0.f};
+  for (int s = 0; s < ksplit; ++s) {
+    const f32x4 p = *reinterpret_cast<const f32x4*>(Cpart + s * total + i);
+#pragma unroll
+    for (int e = 0; e < 4; ++e) v[e] += p[e];
+  }
+#pragma unroll
+  for (int e = 0; e < 4; ++e) {
+    if (bias) v[e] += bf2f(bias[n0 + c4 + e]);
+    if (R) v[e] += bf2f(R[i + e]);
+  }
+  short4v o4;
+#pragma unroll
+  for (int e = 0; e < 4; ++e) o4[e] = (short)f2bf(v[e]);
+  *reinterpret_cast<short4v*>(C + i) = o4;
+  float vs = v[0] * v[0] + v[1] * v[1] + v[2] * v[2] + v[3] * v[3];
+#pragma unroll
+  for (int msk = 1; msk < 16; msk <<= 1) vs += __shfl_xor(vs, msk);
+  if ((lane & 15) == 0) ssout[(long)blockIdx.x * 32 + m] = vs;
 }
 
 __global__ void gemm_skinny_combine_kernel(
