@@ -751,7 +751,10 @@ torch::Tensor gemm_skinny(torch::Tensor A, torch::Tensor W,
   if (M <= 16) launch(std::integral_constant<int, 1>{});
   else launch(std::integral_constant<int, 2>{});
   if (ksplit > 1) {
-    if (ssoutp) {
+    // the vectorized (float4-per-lane) combine serves both the ss-emitting
+    // and plain cases; the flat grid-stride combine remains the fallback
+    // for N % 64 != 0
+    if (ssoutp || N % 64 == 0) {
       dim3 cg(N / 64, (M + 15) / 16);
       gemm_skinny_combine_ss_kernel<<<cg, 256, 0, cur_stream()>>>(
           pp, rp, bp, bf_ptr_mut(C), ssoutp, M, N, ksplit);

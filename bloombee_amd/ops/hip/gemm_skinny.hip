@@ -409,10 +409,12 @@ __global__ __launch_bounds__(256) void gemm_skinny_combine_ss_kernel(
 #pragma unroll
   for (int e = 0; e < 4; ++e) o4[e] = (short)f2bf(v[e]);
   *reinterpret_cast<short4v*>(C + i) = o4;
-  float vs = v[0] * v[0] + v[1] * v[1] + v[2] * v[2] + v[3] * v[3];
+  if (ssout) {
+    float vs = v[0] * v[0] + v[1] * v[1] + v[2] * v[2] + v[3] * v[3];
 #pragma unroll
-  for (int msk = 1; msk < 16; msk <<= 1) vs += __shfl_xor(vs, msk);
-  if ((lane & 15) == 0) ssout[(long)blockIdx.x * 32 + m] = vs;
+    for (int msk = 1; msk < 16; msk <<= 1) vs += __shfl_xor(vs, msk);
+    if ((lane & 15) == 0) ssout[(long)blockIdx.x * 32 + m] = vs;
+  }
 }
 
 __global__ void gemm_skinny_combine_kernel(
