@@ -138,7 +138,7 @@ def run_swarm(args):
 
     def _lm_head(hidden_last):
         y = ops.rms_norm(hidden_last, norm_w, cfg.rms_norm_eps)
-        return ops.linear(y, lm_head_w).float().argmax(-1)
+        return F.linear(y, lm_head_w).float().argmax(-1)
 
     ids = _lm_head(out[:, -1])
     for _ in range(args.warmup):

@@ -168,10 +168,10 @@ class PipelineStage:
         from bloombee_amd import ops
 
         y = ops.rms_norm(h[:, -1], self.final_norm_w, self.config.rms_norm_eps)
-        # vocab projection through the skinny weight-stream kernel: at M<=32
-        # the 1.05 GB (128k-vocab) weight read runs ~5.8 TB/s vs ~2.6 for
-        # the library GEMM — ~150-250 us per decode step
-        nxt = ops.linear(y, self.lm_head_w).float().argmax(-1)
+        # vocab projection stays on hipBLASLt: at N=128k it streams the
+        # 1.05 GB weight at 5.5 TB/s vs 5.3 for the skinny kernel (measured
+        # — the skinny kernel only wins the small-N decode shapes)
+        nxt = F.linear(y, self.lm_head_w).float().argmax(-1)
         self._ids_buf.copy_(nxt)
         self._pos_buf += 1
 
@@ -352,7 +352,7 @@ class PipelineStage:
         from bloombee_amd import ops
 
         y = ops.rms_norm(hidden_last, self.final_norm_w, self.config.rms_norm_eps)
-        return ops.linear(y, self.lm_head_w).float().argmax(-1)
+        return F.linear(y, self.lm_head_w).float().argmax(-1)
 
 
 def init_distributed(device_type: str = "auto") -> str:
