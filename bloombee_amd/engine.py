@@ -203,50 +203,8 @@ class LocalEngine:
         kv = self.kv_pool.allocate(B, T + max_new_tokens + 1)
         try:
             toks = [self.prefill(input_ids, kv)]
-            from bloombee_amd.config import get_config
-
-            use_graph = (self.device.type == "cuda"
-                         and get_config().use_hip_graphs
-                         and max_new_tokens > 4)
-            if not use_graph:
-                for _ in range(max_new_tokens - 1):
-                    toks.append(self.decode_step(toks[-1], kv))
-                return torch.stack(toks, dim=1)
-            # hipGraph decode loop: the whole step (embed, blocks, logits,
-            # argmax, buffer updates) replays against persistent buffers;
-            # the host only grows KV pages per step (same pattern as
-            # make_graphed_decoder — eager B=1 decode measured 4.2 ms/step
-            # launch-bound, profiles/r02 §18)
-            ids_buf = toks[-1].clone()
-            pos_buf = torch.tensor([s.l_spec for s in kv.seqs],
-                                   dtype=torch.int32, device=self.device)
-
-            def _body():
-                h = self._embed(ids_buf.view(-1, 1))
-                h = self.stack.forward_inference(h, kv, pos_buf)
-                nxt = self.logits_for(h[:, -1]).argmax(-1)
-                ids_buf.copy_(nxt)
-                pos_buf.add_(1)
-                return nxt
-
-            n_warm = min(2, max_new_tokens - 1)
-            for _ in range(n_warm):
-                kv.extend(1)
-                kv.page_table()
-                toks.append(_body().clone())
-            if len(toks) < max_new_tokens:
-                graph = torch.cuda.CUDAGraph()
-                kv.extend(1, speculative=True)
-                kv.page_table()
-                with torch.cuda.graph(graph,
-                                      capture_error_mode="thread_local"):
-                    nxt_buf = _body()
-                kv.rollback()
-                for _ in range(max_new_tokens - len(toks)):
-                    kv.extend(1)
-                    kv.page_table()
-                    graph.replay()
-                    toks.append(nxt_buf.clone())
+            for _ in range(max_new_tokens - 1):
+                toks.append(self.decode_step(toks[-1], kv))
             return torch.stack(toks, dim=1)
         finally:
             kv.close()
