@@ -753,8 +753,9 @@ torch::Tensor gemm_skinny(torch::Tensor A, torch::Tensor W,
   if (ksplit > 1) {
     // the vectorized (float4-per-lane) combine serves both the ss-emitting
     // and plain cases; the flat grid-stride combine remains the fallback
-    // for N % 64 != 0
-    if (ssoutp || N % 64 == 0) {
+    // for N % 64 != 0 and for narrow N where (N/64, M/16) leaves the grid
+    // too small (gemma-class H=2304 -> 36 WGs)
+    if (ssoutp || (N % 64 == 0 && N / 64 >= 64)) {
       dim3 cg(N / 64, (M + 15) / 16);
       gemm_skinny_combine_ss_kernel<<<cg, 256, 0, cur_stream()>>>(
           pp, rp, bp, bf_ptr_mut(C), ssoutp, M, N, ksplit);
