@@ -27,7 +27,12 @@ class BloomBlock(torch.nn.Module):
         dt = config.dtype
         self.Hq, self.D, self.I = Hq, D, I
         self.scale = 1.0 / math.sqrt(D)
-        self.alibi = ops.alibi_slopes_for(Hq)
+        # buffer (not a plain attr) so .to(device) moves it once: a CPU
+        # tensor here means a pageable H2D copy per layer per step in the
+        # attention dispatch — wasteful eager and illegal under hipGraph
+        # capture (the bloom capture fault isolated in r02)
+        self.register_buffer("alibi", ops.alibi_slopes_for(Hq).float(),
+                             persistent=False)
 
         def p(*shape):
             return torch.nn.Parameter(torch.empty(*shape, dtype=dt),
