@@ -89,8 +89,9 @@ class PipelineStage:
         if tp > 1 and tp_mode == "context":
             # context parallelism: full weights on every rank; PREFILL
             # shards the sequence with ring attention (cp_prefill_llama)
-            # and all-gathers K/V so decode replicates locally
-            assert self.pp_world == 1, "cp mode is single-stage for now"
+            # and all-gathers K/V so decode replicates locally. Composes
+            # with pp: each stage's cp group ring-prefills its own layer
+            # range, full hidden flows leader-to-leader between stages.
             self.stack = BlockStack(cfg, start, end, device=self.device,
                                     seed=seed)
         elif tp > 1 and tp_mode == "expert":
@@ -262,22 +263,27 @@ class PipelineStage:
 
     @torch.no_grad()
     def _prefill_cp(self, ids: Optional[torch.Tensor], T: int):
-        """Context-parallel one-shot prefill: the tp group shards the
-        SEQUENCE, runs ring attention per block, and all-gathers K/V into
-        every rank's pool (parallel/sequence.cp_prefill_llama)."""
+        """Context-parallel one-shot prefill, composable with pp: each
+        stage's cp group shards the SEQUENCE, runs ring attention over the
+        stage's layers, and all-gathers K/V into every rank's pool
+        (parallel/sequence.cp_prefill_llama); the full hidden flows
+        leader-to-leader between stages like the generic prefill."""
         from bloombee_amd.parallel.sequence import cp_prefill_llama
 
         cfg = self.config
         B = self.global_batch
         assert self.kv.seqs[0].l_spec == 0, "CP prefill is one-shot"
         assert T % self.tp == 0, "prompt length must divide by cp degree"
-        if self.is_client:
-            hid = F.embedding(ids.to(self.device), self.embed)
-        else:
-            hid = torch.empty(B, T, cfg.hidden_size, dtype=cfg.dtype,
-                              device=self.device)
-        if dist.is_initialized():
-            dist.broadcast(hid, src=0, group=self.tp_group)
+        hid = torch.empty(B, T, cfg.hidden_size, dtype=cfg.dtype,
+                          device=self.device)
+        if self.pp_stage == 0:
+            if self.is_client:
+                hid = F.embedding(ids.to(self.device), self.embed)
+        elif self.is_leader:
+            dist.recv(hid, self.prev_rank)
+        if self.tp > 1 and dist.is_initialized():
+            dist.broadcast(hid, src=self.pp_stage * self.tp,
+                           group=self.tp_group)
         Tl = T // self.tp
         self.kv.extend(T)
         shard = hid[:, self.tp_rank * Tl:(self.tp_rank + 1) * Tl].contiguous()
@@ -290,6 +296,14 @@ class PipelineStage:
             hidden = torch.cat(outs, dim=1)
         else:
             hidden = out_shard
+        if self.pp_world > 1:
+            if self.is_leader:
+                dist.send(hidden.contiguous(), self.next_rank)
+            if self.is_client:
+                final = torch.empty_like(hid)
+                dist.recv(final, self.prev_rank)
+                return self._lm_head(final[:, -1])
+            return None
         if not self.is_client:
             return None
         return self._lm_head(hidden[:, -1])

@@ -45,11 +45,18 @@ def _fold(state, s, v, row_pos=None, col_pos=None):
 
 def _ring_exchange(t: torch.Tensor, rank: int, world: int,
                    group=None) -> torch.Tensor:
-    """Send t to (rank+1) % world, receive the previous rank's tensor."""
+    """Send t to (rank+1) % world, receive the previous rank's tensor.
+    rank/world are GROUP-local; p2p dst/src take GLOBAL ranks, so when the
+    group is not the world (pp x cp stages) they must be translated —
+    passing local ranks only works for the rank-0-based group."""
     t = t.contiguous()
     out = torch.empty_like(t)
-    send = dist.isend(t, (rank + 1) % world, group=group)
-    recv = dist.irecv(out, (rank - 1) % world, group=group)
+    nxt, prv = (rank + 1) % world, (rank - 1) % world
+    if group is not None:
+        nxt = dist.get_global_rank(group, nxt)
+        prv = dist.get_global_rank(group, prv)
+    send = dist.isend(t, nxt, group=group)
+    recv = dist.irecv(out, prv, group=group)
     send.wait()
     recv.wait()
     return out
@@ -166,6 +173,11 @@ def cp_prefill_llama(stack, kv_handle, hidden_shard: torch.Tensor,
         qkv = (F.linear(x, blk.qkv_w).view(B, T_loc, Hq + 2 * Hkv, D)
                .permute(0, 2, 1, 3))
         q, k, v = qkv.split([Hq, Hkv, Hkv], dim=1)
+        if hasattr(blk, "q_norm_w"):
+            # qwen3-pattern: per-head RMS on raw q/k before RoPE (the rms
+            # helper normalizes over the last dim = D, weight (D,))
+            q = rms(q, blk.q_norm_w)
+            k = rms(k, blk.k_norm_w)
         cos, sin = blk.rope.get(dev)
         pos = (torch.arange(rank * T_loc, (rank + 1) * T_loc, device=dev)
                .view(1, T_loc).expand(B, T_loc))

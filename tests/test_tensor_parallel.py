@@ -354,7 +354,7 @@ def test_expert_parallel_pipeline_matches_local():
     assert torch.equal(got, expect), (got, expect)
 
 
-def _cp_pipe_worker(rank, world, port, q):
+def _cp_pipe_worker(rank, world, port, q, cp=2, model="llama-tiny"):
     import os
 
     import torch.distributed as dist
@@ -366,9 +366,9 @@ def _cp_pipe_worker(rank, world, port, q):
     try:
         from bloombee_amd.parallel.pipeline import PipelineStage
 
-        stage = PipelineStage("llama-tiny", "cpu", global_batch=2,
+        stage = PipelineStage(model, "cpu", global_batch=2,
                               micro_batches=1, seed=0, kv_max_tokens=4096,
-                              max_session_len=64, tp=2, tp_mode="context")
+                              max_session_len=64, tp=cp, tp_mode="context")
         gen = torch.Generator().manual_seed(9)
         prompt = torch.randint(0, 1000, (2, 16), generator=gen)
         ids = stage.prefill_round(prompt if rank == 0 else None, 16)
@@ -406,6 +406,77 @@ def test_context_parallel_prefill_matches_single_shard():
         assert p.exitcode == 0
     # single-process reference through the same cp code path (world=1)
     stage = PipelineStage("llama-tiny", "cpu", global_batch=2,
+                          micro_batches=1, seed=0, kv_max_tokens=4096,
+                          max_session_len=64, tp=1, tp_mode="context")
+    stage.tp_mode = "context"
+    gen = torch.Generator().manual_seed(9)
+    prompt = torch.randint(0, 1000, (2, 16), generator=gen)
+    ids = stage._prefill_cp(prompt, 16)
+    toks = [ids.clone()]
+    for _ in range(4):
+        ids = stage.decode_round(ids)
+        toks.append(ids.clone())
+    expect = torch.stack(toks, 1)
+    assert torch.equal(got, expect), (got, expect)
+
+
+@pytest.mark.timeout(300)
+def test_context_parallel_composes_with_pp():
+    """pp2 x cp2 (world 4): each stage's cp group ring-prefills its OWN
+    layer range, full hidden flows leader-to-leader between stages, decode
+    replicates inside each group — tokens must match the single-process
+    cp code path exactly."""
+    import torch.multiprocessing as mp
+
+    from bloombee_amd.parallel.pipeline import PipelineStage
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = _free_port()
+    procs = [ctx.Process(target=_cp_pipe_worker, args=(r, 4, port, q, 2))
+             for r in range(4)]
+    for p in procs:
+        p.start()
+    got = q.get(timeout=240)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    stage = PipelineStage("llama-tiny", "cpu", global_batch=2,
+                          micro_batches=1, seed=0, kv_max_tokens=4096,
+                          max_session_len=64, tp=1, tp_mode="context")
+    stage.tp_mode = "context"
+    gen = torch.Generator().manual_seed(9)
+    prompt = torch.randint(0, 1000, (2, 16), generator=gen)
+    ids = stage._prefill_cp(prompt, 16)
+    toks = [ids.clone()]
+    for _ in range(4):
+        ids = stage.decode_round(ids)
+        toks.append(ids.clone())
+    expect = torch.stack(toks, 1)
+    assert torch.equal(got, expect), (got, expect)
+
+
+@pytest.mark.timeout(300)
+def test_context_parallel_qwen3_pattern():
+    """cp2 ring prefill on a qwen3-pattern stack (per-head q/k norms before
+    RoPE inside the ring path) matches the same code path at cp1."""
+    import torch.multiprocessing as mp
+
+    from bloombee_amd.parallel.pipeline import PipelineStage
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = _free_port()
+    procs = [ctx.Process(target=_cp_pipe_worker,
+                         args=(r, 2, port, q, 2, "qwen3-tiny"))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    got = q.get(timeout=240)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    stage = PipelineStage("qwen3-tiny", "cpu", global_batch=2,
                           micro_batches=1, seed=0, kv_max_tokens=4096,
                           max_session_len=64, tp=1, tp_mode="context")
     stage.tp_mode = "context"
