@@ -16,6 +16,15 @@ from dataclasses import dataclass, field
 from typing import Optional
 
 
+def _parse_mixed(raw: str) -> str:
+    raw = raw.strip().lower()
+    if raw in ("1", "on", "true", "yes"):
+        return "on"
+    if raw == "auto":
+        return "auto"
+    return "off"
+
+
 def _env(name: str, default, cast=None):
     raw = os.environ.get(f"BBAMD_{name}")
     if raw is None:
@@ -38,11 +47,17 @@ class KVConfig:
     # not set explicitly (the reference sizes from --attn_cache_tokens).
     max_tokens: Optional[int] = field(default_factory=lambda: _env("KV_MAX_TOKENS", None, int))
     alloc_timeout: float = field(default_factory=lambda: _env("KV_ALLOC_TIMEOUT", 60.0))
-    # BBAMD_MIXED_ATTN=1: host-swapped sessions resume DECODING with their
+    # BBAMD_MIXED_ATTN: host-swapped sessions resume DECODING with their
     # committed KV left in host memory (mixed-device attention, exact
     # log-sum-exp merge) instead of being restored to HBM first (ref
-    # _mixed_device_attention, pytorch_backend.py:969-1014)
-    mixed_attn: bool = field(default_factory=lambda: _env("MIXED_ATTN", False))
+    # _mixed_device_attention, pytorch_backend.py:969-1014).
+    #   off  (default): always restore (swap_in)
+    #   on/1: always decode mixed (capacity mode)
+    #   auto: restore when the device pool currently has room for the
+    #         session's pages, go mixed only when it does not — mixed-attn
+    #         as the measured capacity fallback rather than a blanket mode
+    mixed_attn: str = field(default_factory=lambda: _parse_mixed(
+        os.environ.get("BBAMD_MIXED_ATTN", "off")))
 
 
 @dataclass

@@ -181,6 +181,10 @@ class PagedKVCache:
             self._reserved_tokens -= handle.reserved_tokens
             self._lock.notify_all()
 
+    def free_page_count(self) -> int:
+        with self._lock:
+            return len(self._free_pages)
+
     def _take_pages(self, n: int) -> List[int]:
         with self._lock:
             if len(self._free_pages) < n:
@@ -489,6 +493,14 @@ class SessionHandle:
     @property
     def is_swapped(self) -> bool:
         return getattr(self, "_swapped", None) is not None
+
+    def swapped_pages_needed(self) -> int:
+        """Device pages a full swap_in would claim (0 when not swapped) —
+        the backend's mixed_attn="auto" policy compares this against the
+        pool's current free pages to pick restore vs mixed-device decode."""
+        if not self.is_swapped:
+            return 0
+        return sum(self._swapped_counts)
 
     # -- mixed-device decode (host prefix + device recent segment) --------
     @property

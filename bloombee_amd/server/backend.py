@@ -213,7 +213,12 @@ class StackBackend:
                             and position_ids is None)
             if handle.is_swapped:
                 from bloombee_amd.config import get_config as _gc
-                if _gc().kv.mixed_attn and plain_decode:
+                mode = _gc().kv.mixed_attn
+                use_mixed = plain_decode and (
+                    mode == "on"
+                    or (mode == "auto" and handle.swapped_pages_needed()
+                        > handle.cache.free_page_count()))
+                if use_mixed:
                     # capacity mode: committed KV stays host-side; decode
                     # merges it with the device-resident recent segment
                     handle.swap_in_as_prefix()

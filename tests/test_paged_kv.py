@@ -337,3 +337,31 @@ def test_row_swap_extend_blocks_until_pages_freed():
     assert len(h.seqs[2].pages) == 2 and h.seqs[0].pages == []
     assert h.lengths == [32, 32, 32, 32]
     h.close()
+
+
+def test_swap_auto_policy_quantities():
+    """mixed_attn="auto" decides restore-vs-mixed from swapped_pages_needed()
+    vs free_page_count(): after swap-out the pool reports the freed pages,
+    and page scarcity flips the decision to mixed-device decode."""
+    from bloombee_amd.kv.paged import PagedKVCache
+
+    cache = PagedKVCache(num_layers=1, num_kv_heads=2, head_dim=64,
+                         max_tokens=16 * 8, page_size=16, device="cpu",
+                         dtype=torch.bfloat16)
+    h = cache.allocate(2, 16 * 2)   # 2 seqs x up to 2 pages each
+    h.extend(20)                    # 2 pages per seq -> 4 pages used
+    used = 4
+    assert cache.free_page_count() == 8 - used
+    h.swap_out()
+    assert h.swapped_pages_needed() == used
+    assert cache.free_page_count() == 8   # pages returned to the pool
+    # fits -> auto restores
+    assert h.swapped_pages_needed() <= cache.free_page_count()
+    # page scarcity (e.g. other resident sessions) -> auto goes mixed
+    held = cache._take_pages(6)
+    assert cache.free_page_count() == 2
+    assert h.swapped_pages_needed() > cache.free_page_count()
+    cache._give_pages(held)
+    h.swap_in()
+    assert not h.is_swapped and h.swapped_pages_needed() == 0
+    h.close()
