@@ -215,7 +215,7 @@ class StackBackend:
                 from bloombee_amd.config import get_config as _gc
                 mode = _gc().kv.mixed_attn
                 use_mixed = plain_decode and (
-                    mode == "on"
+                    mode in ("on", True)  # bool tolerated for programmatic cfgs
                     or (mode == "auto" and handle.swapped_pages_needed()
                         > handle.cache.free_page_count()))
                 if use_mixed:
