@@ -255,3 +255,40 @@ def test_client_weights_roundtrip(tmp_path):
     assert torch.equal(back["embed"], e)
     assert torch.equal(back["final_norm"], n)
     assert torch.equal(back["lm_head"], h)
+
+
+def test_mixed_attn_auto_policy_backend():
+    """BBAMD_MIXED_ATTN=auto through the real backend: a swapped session
+    RESTORES when the pool has room and decodes MIXED (host prefix,
+    pos_offset > 0) under page scarcity."""
+    from bloombee_amd.config import RuntimeConfig, KVConfig, get_config, set_config
+    from bloombee_amd.models.base import resolve_config
+    from bloombee_amd.server.backend import StackBackend
+
+    cfg = resolve_config("llama-tiny")
+    be = StackBackend(cfg, 0, 2, device="cpu", seed=0,
+                            kv_max_tokens=16 * 16)
+    old_cfg = get_config()
+    set_config(RuntimeConfig(kv=KVConfig(mixed_attn="auto")))
+    try:
+        be.open_session("s", 1, 64)
+        h = torch.randn(1, 20, cfg.hidden_size).to(cfg.dtype) * 0.1
+        be.inference_step("s", h, 0)     # 20 tokens -> 2 pages
+        handle = be.session_handle("s")
+
+        # room in the pool -> auto restores
+        handle.swap_out()
+        assert handle.is_swapped and handle.swapped_pages_needed() == 2
+        be.inference_step("s", h[:, :1], 20)
+        assert not handle.is_swapped and handle.pos_offset == 0
+
+        # page scarcity (1 free < 2 needed, but enough for the recent
+        # segment) -> auto keeps the prefix host-side (mixed decode)
+        handle.swap_out()
+        held = be.kv_pool._take_pages(be.kv_pool.free_page_count() - 1)
+        be.inference_step("s", h[:, :1], 21)
+        assert handle.pos_offset > 0, "expected mixed-device decode"
+        be.kv_pool._give_pages(held)
+    finally:
+        set_config(old_cfg)
+        be.close_session("s")
