@@ -268,7 +268,8 @@ class PipelineStage:
         stage's layers, and all-gathers K/V into every rank's pool
         (parallel/sequence.cp_prefill_llama); the full hidden flows
         leader-to-leader between stages like the generic prefill."""
-        from bloombee_amd.parallel.sequence import cp_prefill_llama
+        from bloombee_amd.parallel.sequence import (cp_prefill_gemma4,
+                                                    cp_prefill_llama)
 
         cfg = self.config
         B = self.global_batch
@@ -287,8 +288,11 @@ class PipelineStage:
         Tl = T // self.tp
         self.kv.extend(T)
         shard = hid[:, self.tp_rank * Tl:(self.tp_rank + 1) * Tl].contiguous()
-        out_shard = cp_prefill_llama(self.stack, self.kv, shard,
-                                     self.tp_rank, self.tp, self.tp_group)
+        cp_fn = (cp_prefill_gemma4
+                 if hasattr(self.stack.blocks[0], "pre_ffn_norm_w")
+                 else cp_prefill_llama)
+        out_shard = cp_fn(self.stack, self.kv, shard,
+                          self.tp_rank, self.tp, self.tp_group)
         if self.tp > 1 and dist.is_initialized():
             outs = [torch.empty_like(out_shard) for _ in range(self.tp)]
             dist.all_gather(outs, out_shard.contiguous(),

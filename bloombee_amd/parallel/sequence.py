@@ -20,15 +20,21 @@ import torch
 import torch.distributed as dist
 
 
-def _fold(state, s, v, row_pos=None, col_pos=None):
+def _fold(state, s, v, row_pos=None, col_pos=None, window: int = 0):
     """Fold one score block into the online-softmax state.
 
     state: (m, l, acc) with m,l (B,H,T,1) and acc (B,H,T,D), all fp32.
     s: (B,H,T,S) scores; v: (B,H,S,D). If row_pos/col_pos given, mask
-    s[t, c] where col_pos[c] > row_pos[t] (causal)."""
+    s[t, c] where col_pos[c] > row_pos[t] (causal); window > 0 adds the
+    sliding lower bound col_pos[c] > row_pos[t] - window (gemma-4
+    alternating sliding layers)."""
     m, l, acc = state
     if row_pos is not None:
-        dead = col_pos.view(1, 1, 1, -1) > row_pos.view(1, 1, -1, 1)
+        cp = col_pos.view(1, 1, 1, -1)
+        rp = row_pos.view(1, 1, -1, 1)
+        dead = cp > rp
+        if window > 0:
+            dead = dead | (cp <= rp - window)
         s = s.masked_fill(dead, float("-inf"))
     m_new = torch.maximum(m, s.amax(-1, keepdim=True))
     # fully-masked rows keep m = -inf; exp(-inf - -inf) guards below
@@ -235,4 +241,115 @@ def cp_prefill_llama(stack, kv_handle, hidden_shard: torch.Tensor,
                      kv_handle.v_pages(blk.layer_index),
                      kv_handle.page_table(),
                      torch.zeros(B, dtype=torch.int32, device=dev))
+    return hidden
+
+
+def cp_prefill_gemma4(stack, kv_handle, hidden_shard: torch.Tensor,
+                      rank: int, world: int, group=None) -> torch.Tensor:
+    """Context-parallel one-shot prefill for a gemma-4 BlockStack: same
+    ring structure as cp_prefill_llama but with the family's specifics —
+    (1+w) rms norms, separate q/k/v projections with optional k==v and
+    shared-KV donor tail layers, per-layer partial rotary tables, per-layer
+    sliding window (ring folds carry the window lower bound), scale=1.0
+    scores and a gelu-tanh MLP with post-norms. Donor layers keep their
+    post-rope K/V shard so tail layers re-rotate the SAME tensors through
+    the ring; only donor layers all-gather into the paged pool."""
+    import torch.nn.functional as F
+
+    from bloombee_amd import ops
+    from bloombee_amd.models.gemma4.block import (_rms1p, _rms1p_headdim,
+                                                  _rope_partial)
+
+    cfg = stack.config
+    B, T_loc, H = hidden_shard.shape
+    dev = hidden_shard.device
+    hidden = hidden_shard
+    donor_kv = {}
+
+    for blk in stack.blocks:
+        eps = blk.config.rms_norm_eps
+        Hq, Hkv, D = blk.Hq, blk.Hkv, blk.D
+        G = Hq // Hkv
+        cos, sin = blk._tables(dev)
+        pos = (torch.arange(rank * T_loc, (rank + 1) * T_loc, device=dev)
+               .view(1, T_loc).expand(B, T_loc).long())
+
+        x = _rms1p(hidden, blk.input_norm_w, eps)
+        q = (F.linear(x, blk.q_w).view(B, T_loc, Hq, D)
+             .permute(0, 2, 1, 3))
+        q = _rms1p_headdim(q, blk.q_norm_w, eps)
+        q = _rope_partial(q, cos, sin, pos, blk.rot_dim).contiguous()
+
+        if blk.donor is None:
+            kraw = (F.linear(x, blk.k_w).view(B, T_loc, Hkv, D)
+                    .permute(0, 2, 1, 3))
+            k = _rms1p_headdim(kraw, blk.k_norm_w, eps)
+            k = _rope_partial(k, cos, sin, pos, blk.rot_dim).contiguous()
+            if blk.v_w is not None:
+                vraw = (F.linear(x, blk.v_w).view(B, T_loc, Hkv, D)
+                        .permute(0, 2, 1, 3))
+            else:
+                vraw = kraw                      # attention_k_eq_v
+            v = _rms1p_headdim(vraw, None, eps).contiguous()
+            donor_kv[blk.global_index] = (k, v)
+            write_layer = blk.layer_index
+        else:
+            if blk.donor not in donor_kv:
+                raise RuntimeError(
+                    f"gemma-4 shared-KV layer {blk.global_index} requires "
+                    f"donor {blk.donor} in the same stage for CP prefill")
+            k, v = donor_kv[blk.donor]           # shared-KV tail
+            write_layer = None
+
+        state = (torch.full((B, Hq, T_loc, 1), float("-inf")),
+                 torch.zeros(B, Hq, T_loc, 1),
+                 torch.zeros(B, Hq, T_loc, D))
+        row_pos = torch.arange(rank * T_loc, (rank + 1) * T_loc)
+        win = blk.window
+        k_cur = k.float().contiguous()
+        v_cur = v.float().contiguous()
+        src = rank
+        qf = q.float()
+        for step in range(world):
+            if src <= rank:
+                kx = k_cur.repeat_interleave(G, dim=1)
+                vx = v_cur.repeat_interleave(G, dim=1)
+                s = qf @ kx.transpose(-1, -2)    # gemma scale = 1.0
+                col_pos = torch.arange(src * T_loc, (src + 1) * T_loc)
+                if src == rank or win > 0:
+                    state = _fold(state, s, vx, row_pos, col_pos, window=win)
+                else:
+                    state = _fold(state, s, vx)
+            if step < world - 1:
+                k_cur = _ring_exchange(k_cur, rank, world, group)
+                v_cur = _ring_exchange(v_cur, rank, world, group)
+                src = (src - 1) % world
+        m, l, acc = state
+        attn = ((acc / l.clamp_min(1e-30)).to(hidden.dtype)
+                .permute(0, 2, 1, 3).reshape(B, T_loc, Hq * D))
+
+        a = F.linear(attn, blk.o_w)
+        a = _rms1p(a, blk.post_attn_norm_w, eps)
+        h = hidden + a
+        y = _rms1p(h, blk.pre_ffn_norm_w, eps)
+        gu = F.linear(y, blk.gate_up_w)
+        g, u = gu.split([blk.I, blk.I], dim=-1)
+        m_ = F.linear(ops.gelu_tanh(g) * u, blk.down_w)
+        hidden = h + _rms1p(m_, blk.post_ffn_norm_w, eps)
+
+        if write_layer is not None:
+            if world > 1:
+                ks = [torch.empty_like(k) for _ in range(world)]
+                vs = [torch.empty_like(v) for _ in range(world)]
+                dist.all_gather(ks, k, group=group)
+                dist.all_gather(vs, v, group=group)
+                k_full = torch.cat(ks, dim=2)
+                v_full = torch.cat(vs, dim=2)
+            else:
+                k_full, v_full = k, v
+            ops.kv_write(k_full.to(cfg.dtype), v_full.to(cfg.dtype),
+                         kv_handle.k_pages(write_layer),
+                         kv_handle.v_pages(write_layer),
+                         kv_handle.page_table(),
+                         torch.zeros(B, dtype=torch.int32, device=dev))
     return hidden

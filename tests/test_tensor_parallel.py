@@ -370,7 +370,8 @@ def _cp_pipe_worker(rank, world, port, q, cp=2, model="llama-tiny"):
                               micro_batches=1, seed=0, kv_max_tokens=4096,
                               max_session_len=64, tp=cp, tp_mode="context")
         gen = torch.Generator().manual_seed(9)
-        prompt = torch.randint(0, 1000, (2, 16), generator=gen)
+        hi = min(1000, stage.config.vocab_size - 1)
+        prompt = torch.randint(0, hi, (2, 16), generator=gen)
         ids = stage.prefill_round(prompt if rank == 0 else None, 16)
         toks = [ids.clone()] if rank == 0 else []
         for _ in range(4):
@@ -482,6 +483,43 @@ def test_context_parallel_qwen3_pattern():
     stage.tp_mode = "context"
     gen = torch.Generator().manual_seed(9)
     prompt = torch.randint(0, 1000, (2, 16), generator=gen)
+    ids = stage._prefill_cp(prompt, 16)
+    toks = [ids.clone()]
+    for _ in range(4):
+        ids = stage.decode_round(ids)
+        toks.append(ids.clone())
+    expect = torch.stack(toks, 1)
+    assert torch.equal(got, expect), (got, expect)
+
+
+@pytest.mark.timeout(300)
+def test_context_parallel_gemma4_pattern():
+    """cp2 ring prefill on a gemma-4 stack (sliding windows in the ring
+    folds, (1+w) norms, partial rotary, shared-KV donor tail) matches the
+    same code path at cp1."""
+    import torch.multiprocessing as mp
+
+    from bloombee_amd.parallel.pipeline import PipelineStage
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = _free_port()
+    procs = [ctx.Process(target=_cp_pipe_worker,
+                         args=(r, 2, port, q, 2, "gemma4-tiny"))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    got = q.get(timeout=240)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    stage = PipelineStage("gemma4-tiny", "cpu", global_batch=2,
+                          micro_batches=1, seed=0, kv_max_tokens=4096,
+                          max_session_len=64, tp=1, tp_mode="context")
+    stage.tp_mode = "context"
+    gen = torch.Generator().manual_seed(9)
+    hi = min(1000, stage.config.vocab_size - 1)
+    prompt = torch.randint(0, hi, (2, 16), generator=gen)
     ids = stage._prefill_cp(prompt, 16)
     toks = [ids.clone()]
     for _ in range(4):
