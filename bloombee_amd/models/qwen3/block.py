@@ -61,9 +61,26 @@ class Qwen3Block(LlamaBlock):
         kp = kv.k_pages(self.layer_index)
         vp = kv.v_pages(self.layer_index)
         pt = kv.page_table()
+        # mixed-device sessions (host KV prefix + device recent segment):
+        # pool positions are local, rotary stays absolute (see llama block)
+        off = getattr(kv, "pos_offset", 0)
+        if off and position_ids is None:
+            position_ids = (start_pos.view(B, 1) + off
+                            + torch.arange(T, device=hidden.device)
+                            .view(1, T)).int()
         ops.rope_kv_write_(qkv, Hq, Hkv, cos, sin, position_ids, kp, vp, pt,
                            start_pos)
-        attn = ops.attn_paged_qkv(qkv, Hq, Hkv, kp, vp, pt, start_pos, self.scale)
+        hp = (kv.host_prefix(self.layer_index)
+              if hasattr(kv, "host_prefix") else None)
+        if hp is not None:
+            q = (qkv[..., : Hq * D].view(B, T, Hq, D)
+                 .permute(0, 2, 1, 3).contiguous())
+            attn = ops.attn_paged_mixed(q, kp, vp, pt, start_pos + T,
+                                        hp[0], hp[1], self.scale)
+            attn = attn.permute(0, 2, 1, 3).reshape(B, T, Hq * D)
+        else:
+            attn = ops.attn_paged_qkv(qkv, Hq, Hkv, kp, vp, pt, start_pos,
+                                      self.scale)
         if fuse_norm:
             self._ensure_ss_bufs(hidden.device)
             h2 = ops.linear(attn, self.o_w, residual=hidden,

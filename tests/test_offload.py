@@ -195,3 +195,36 @@ def test_mixed_device_decode_after_prefix_swap():
     nxt = eng.decode_step(got[:, -1], kv)
     kv.close()
     assert nxt.shape == (2,)
+
+
+def test_mixed_device_decode_qwen3_pattern():
+    """Mixed-device decode also covers qwen3-pattern blocks (per-head q/k
+    norms; host-prefix branch added late round 2): token-exact vs the
+    never-swapped run."""
+    from bloombee_amd.engine import LocalEngine
+
+    torch.manual_seed(0)
+    ref_eng = LocalEngine("qwen3-tiny", device="cpu", seed=13,
+                          kv_max_tokens=1 << 12)
+    gen = torch.Generator().manual_seed(5)
+    prompt = torch.randint(0, 900, (2, 40), generator=gen)
+    kv = ref_eng.kv_pool.allocate(2, 64)
+    toks = [ref_eng.prefill(prompt, kv)]
+    for _ in range(5):
+        toks.append(ref_eng.decode_step(toks[-1], kv))
+    expect = torch.stack(toks, 1)
+    kv.close()
+
+    eng = LocalEngine("qwen3-tiny", device="cpu", seed=13,
+                      kv_max_tokens=1 << 12)
+    kv = eng.kv_pool.allocate(2, 64)
+    tok = eng.prefill(prompt, kv)
+    kv.swap_out()
+    kv.swap_in_as_prefix()
+    assert kv.pos_offset == 40
+    got = [tok]
+    for _ in range(5):
+        got.append(eng.decode_step(got[-1], kv))
+    kv.close()
+    got = torch.stack(got, 1)
+    assert torch.equal(got, expect), (got, expect)
